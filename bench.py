@@ -1,0 +1,238 @@
+#!/usr/bin/env python3
+"""Benchmark harness for the MI355X LP hot path.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`.
+A step = one full deterministic LP refinement (5 sweeps) over the workload
+graph from a fixed pseudo-random initial partition (reset between steps),
+mirroring the reference's standalone LP benchmark semantics
+(apps/benchmarks/shm_label_propagation_benchmark.cc:106-127: LP region only,
+allocation/generation excluded).
+
+Workload (BASELINE.json metric: "LP edges processed/sec + final edge-cut,
+R-MAT scale-26 k=16"): R-MAT scale-26 (n=2^26, ~1.05G directed arcs,
+Graph500 parameters, symmetrized/dedup'd, seed 42), k=16, eps=0.03,
+synthetic, unit weights. value = directed arcs scanned per second over the
+timed LP regions (whole-job across all ranks).
+
+For N>1 the vertex set of every chunk is sharded across ranks and proposal
+lists are all-gathered over RCCL; results are bit-identical to N=1 (see
+kaminpar_amd/multi.py).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+
+HBM_PEAK_GBS = 8000.0  # MI355X spec peak (MI355X_MICROARCH.md)
+BYTES_PER_ARC = 8.0    # 4 B adjncy + 4 B labels gather (unweighted model)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--scale", type=int, default=26, help="R-MAT scale")
+    ap.add_argument("--edgefactor", type=int, default=8)
+    ap.add_argument("--k", type=int, default=16)
+    ap.add_argument("--iters", type=int, default=5)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import kaminpar_amd as ka
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+
+    if world > 1:
+        import torch.distributed as dist
+
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cuda:0"
+
+    log(f"[bench] generating R-MAT scale-{args.scale} ef={args.edgefactor} ...")
+    t0 = time.time()
+    g = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+    log(f"[bench] n={g.n} m={g.m} ({time.time()-t0:.1f}s); uploading ...")
+
+    k = args.k
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    cut0 = g.edge_cut(part0)
+
+    eng = ka.LpEngine(g)
+
+    from kaminpar_amd.multi import LocalComm, TorchComm, refine_dist
+
+    if world > 1:
+        comm = TorchComm(device)
+    else:
+        comm = None  # fast path: full loop inside C++
+
+    def one_step():
+        if comm is None:
+            cut, part, stats = eng.refine(k, mbw, part0, seed=args.seed, iters=args.iters)
+        else:
+            cut, part, stats = refine_dist(eng, k, mbw, part0, args.seed, args.iters, comm)
+        return cut, stats
+
+    def barrier_sync():
+        if world > 1:
+            import torch.distributed as dist
+
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step()
+
+    barrier_sync()
+    t_start = time.time()
+    total_arcs = 0
+    phase_a_ns = 0
+    last_cut = None
+    moves = 0
+    for _ in range(args.steps):
+        cut, stats = one_step()
+        total_arcs += stats.arcs_scanned
+        phase_a_ns += stats.phase_a_ns
+        moves += stats.moves
+        last_cut = cut
+    barrier_sync()
+    t_end = time.time()
+
+    elapsed = t_end - t_start
+    # per-rank arcs: in sharded mode each rank scans its slice; whole-job arcs
+    # = sum over ranks (phase A split); reduce.
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([total_arcs, phase_a_ns], dtype=torch.int64, device=device)
+        dist.all_reduce(t)
+        total_arcs = int(t[0].item())
+        phase_a_ns = int(t[1].item())  # summed over ranks
+        tmax = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
+        elapsed = float(tmax.item())
+
+    value = total_arcs / elapsed
+
+    # roofline of the dominant kernel group (phase A: bin+gain/select),
+    # algorithmic bytes 8 B/arc over HIP-event time of those launches
+    achieved_gbs = (total_arcs * BYTES_PER_ARC) / max(phase_a_ns, 1)  # B/ns = GB/s
+    if world > 1:
+        achieved_gbs *= world  # phase_a_ns was summed across ranks; arcs too
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved_gbs, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+        "traffic": None,  # filled from rocprofv3 PMC runs (profiles/)
+    }
+
+    result = None
+    if rank == 0:
+        cpu_baseline = None
+        if not args.no_cpu_baseline and world == 1:
+            cpu_baseline = run_cpu_baseline(g, k, mbw, part0, args)
+
+        result = {
+            "metric": "LP edges processed/sec",
+            "value": round(value, 1),
+            "unit": "arcs/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int32",
+            "data": "synthetic",
+            "config": {
+                "workload": f"rmat{args.scale}_k{args.k}_lp_refine",
+                "n": int(g.n),
+                "arcs": int(g.m),
+                "k": k,
+                "iters": args.iters,
+                "edge_cut_before": int(cut0),
+                "edge_cut_after": int(last_cut),
+                "moves": int(moves // max(args.steps, 1)),
+                "parallelism": f"shard{world}" if world > 1 else "single",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(result))
+    return result
+
+
+def run_cpu_baseline(g, k, mbw, part0, args):
+    """Time the CPU oracle (the restated reference LP semantics, oracle/) on a
+    bounded sample of the same workload: ONE sweep (iters=1) over the same
+    graph and initial partition (~10-30 s of CPU work at scale 26)."""
+    res = _cpu_baseline_impl(g, k, mbw, part0, iters=1, seed=args.seed)
+    if res is None:
+        return None
+    arcs, dt = res
+    # the oracle's LP sweep is currently single-threaded; report that honestly
+    return {
+        "value": round(arcs / dt, 1),
+        "unit": "arcs/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"1 LP sweep over the full workload graph ({arcs} arcs, {dt:.1f}s)",
+    }
+
+
+def _cpu_baseline_impl(g, k, mbw, part0, iters, seed):
+    import ctypes
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    path = os.path.join(here, "oracle", "liblp_oracle.so")
+    if not os.path.exists(path):
+        return None
+    o = ctypes.CDLL(path)
+    o.kmp_oracle_lp_refine.restype = ctypes.c_int64
+    part = np.ascontiguousarray(part0, dtype=np.uint32).copy()
+    xadj = np.ascontiguousarray(g.xadj, dtype=np.uint32)
+    adjncy = np.ascontiguousarray(g.adjncy, dtype=np.uint32)
+    stats = np.zeros(3, dtype=np.uint64)
+    t0 = time.time()
+    o.kmp_oracle_lp_refine(
+        ctypes.c_uint32(g.n), ctypes.c_uint64(g.m),
+        xadj.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        adjncy.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        None, None, ctypes.c_uint32(k),
+        mbw.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        part.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ctypes.c_uint64(seed), ctypes.c_int(iters),
+        stats.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+    )
+    dt = time.time() - t0
+    return int(stats[0]), dt
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,167 @@
+/*
+ * C ABI of the MI355X-native KaMinPar label-propagation hot path.
+ *
+ * This is the drop-in boundary for the reference's LP operator seams:
+ *   - kmp_lp_cluster_*  replaces Clusterer::compute_clustering behind
+ *     kaminpar-shm/coarsening/clusterer.h:19-47 (LPClustering,
+ *     lp_clusterer.cc:395) including set_max_cluster_weight /
+ *     set_desired_cluster_count (clusterer.h:35-36);
+ *   - kmp_lp_refine_*   replaces Refiner::refine behind
+ *     kaminpar-shm/refinement/refiner.h:18-57 (LabelPropagationRefiner,
+ *     lp_refiner.cc:370-372);
+ *   - graph handles keep the reference CSR layout exactly
+ *     (CSRGraphMemory, kaminpar-shm/datastructures/csr_graph.h:27-33:
+ *     xadj[n+1], adjncy[m] with both arc directions, optional vwgt/adjwgt);
+ *   - kmp_edge_cut mirrors metrics::edge_cut (kaminpar-shm/metrics.cc:37-59).
+ *
+ * Types follow include/kaminpar-shm/ckaminpar.h:27-52: NodeID/EdgeID u32
+ * (u64 edge ids are a planned variant for >4G-arc graphs), NodeWeight/
+ * EdgeWeight i32, BlockID u32, BlockWeight i64 here (the reference uses
+ * NodeWeight-width block weights; i64 caps avoid overflow at scale-28).
+ *
+ * Plain pointers and sizes only; no torch types. Compute entry points
+ * REQUIRE a GPU (they abort with a clear error if no HIP device is present);
+ * graph generation/IO entry points are host-only.
+ */
+#ifndef KAMINPAR_LP_H
+#define KAMINPAR_LP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct kmp_graph_t kmp_graph_t; /* host-resident CSR graph */
+typedef struct kmp_lp_t kmp_lp_t;       /* device-resident LP engine */
+
+/* ---------------------------------------------------------------- graphs */
+
+/* Copy a CSR graph (reference layout, csr_graph.h:27-33). vwgt/adjwgt may be
+ * NULL for unit weights. Returns NULL on invalid input. */
+kmp_graph_t *kmp_graph_from_csr(
+    uint32_t n,
+    uint64_t m,
+    const uint32_t *xadj,
+    const uint32_t *adjncy,
+    const int32_t *vwgt,
+    const int32_t *adjwgt
+);
+
+/* Graph500-parameter R-MAT generator (a=.57,b=.19,c=.19,d=.05), symmetrized
+ * and deduplicated, unit weights; deterministic in (scale, edgefactor, seed).
+ * n = 2^scale, #undirected edges before dedup = edgefactor * n. */
+kmp_graph_t *kmp_gen_rmat(int scale, int edgefactor, uint64_t seed);
+
+/* 2D random geometric graph on the unit square: n points, radius chosen for
+ * the given average degree; deterministic in (n, avg_deg, seed). */
+kmp_graph_t *kmp_gen_rgg2d(uint32_t n, double avg_deg, uint64_t seed);
+
+/* METIS ASCII reader (kaminpar-io/metis_parser.h:17-25 format). */
+kmp_graph_t *kmp_read_metis(const char *path);
+
+uint32_t kmp_graph_n(const kmp_graph_t *g);
+uint64_t kmp_graph_m(const kmp_graph_t *g); /* number of directed arcs */
+const uint32_t *kmp_graph_xadj(const kmp_graph_t *g);
+const uint32_t *kmp_graph_adjncy(const kmp_graph_t *g);
+const int32_t *kmp_graph_vwgt(const kmp_graph_t *g);     /* NULL if unit */
+const int32_t *kmp_graph_adjwgt(const kmp_graph_t *g);   /* NULL if unit */
+int64_t kmp_graph_total_node_weight(const kmp_graph_t *g);
+void kmp_graph_free(kmp_graph_t *g);
+
+/* Sequential host edge cut of a labelling (metrics.cc:73-84). */
+int64_t kmp_edge_cut_host(const kmp_graph_t *g, const uint32_t *labels);
+
+/* Max block weight as the reference's PartitionContext computes it for
+ * uniform epsilon (context.cc:27-39): (1+eps) * ceil(total_weight / k). */
+int64_t kmp_max_block_weight(const kmp_graph_t *g, uint32_t k, double eps);
+
+/* ------------------------------------------------------------- LP engine */
+
+/* Run statistics, filled by the compute entry points. */
+typedef struct kmp_lp_stats_t {
+  uint64_t arcs_scanned;    /* directed arcs scanned over all sweeps */
+  uint64_t moves;           /* committed moves */
+  uint64_t phase_a_ns;      /* HIP-event time in gain/select kernels */
+  uint64_t total_ns;        /* wall time of the LP region (device-synced) */
+  uint64_t num_clusters;    /* clusterer: final non-empty clusters */
+  int64_t edge_cut;         /* refiner: final cut (device-computed) */
+} kmp_lp_stats_t;
+
+/* Create an engine on the current HIP device and upload the graph.
+ * Aborts (returns NULL + message on stderr) if no GPU is available. */
+kmp_lp_t *kmp_lp_create(const kmp_graph_t *g);
+void kmp_lp_free(kmp_lp_t *e);
+
+/* Deterministic LP refinement (chunk-synchronous schedule; see
+ * oracle/lp_oracle.cpp header for the schedule contract). partition: in/out,
+ * n entries, values < k. max_block_weights: k entries. Returns the final
+ * edge cut, or -1 on error. */
+int64_t kmp_lp_refine(
+    kmp_lp_t *e,
+    uint32_t k,
+    const int64_t *max_block_weights,
+    uint32_t *partition,
+    uint64_t seed,
+    int iters,
+    kmp_lp_stats_t *stats
+);
+
+/* Deterministic LP clustering (coarsening instantiation; clusters start as
+ * singletons, uniform cap, isolated-node + two-hop passes). clustering: out,
+ * n entries. Returns the number of non-empty clusters, or -1 on error. */
+int64_t kmp_lp_cluster(
+    kmp_lp_t *e,
+    int64_t max_cluster_weight,
+    uint32_t desired_clusters,
+    uint32_t *clustering,
+    uint64_t seed,
+    int iters,
+    kmp_lp_stats_t *stats
+);
+
+/* --------------------------- sharded (multi-GPU) refinement sub-steps ----
+ * One process per GPU; rank r of world R computes phase A for its slice of
+ * each chunk's position range, the proposal lists are all-gathered by the
+ * caller (RCCL via torch.distributed), and every rank runs the identical
+ * deterministic commit, so all replicas stay bit-identical (mirrors the
+ * ghost-label exchange of kaminpar-dist/coarsening/clustering/lp/
+ * global_lp_clusterer.cc:480-594 with labels replicated instead of owned).
+ *
+ * Proposals are 16-byte records {u, to, rank_in_chunk, weight} (uint32x4). */
+
+/* Initialize a sharded refinement run. partition: host, n entries. */
+int kmp_lp_refine_begin(
+    kmp_lp_t *e,
+    uint32_t k,
+    const int64_t *max_block_weights,
+    const uint32_t *partition,
+    uint64_t seed
+);
+
+/* Number of chunks per sweep (fixed by the schedule). */
+uint32_t kmp_lp_num_chunks(const kmp_lp_t *e);
+
+/* Phase A for chunk `chunk` of sweep `iter`, positions [pos_lo, pos_hi).
+ * Writes proposals to the DEVICE buffer d_out (capacity cap records) and
+ * returns the record count (host), or -1 on error/overflow. */
+int64_t kmp_lp_phase_a(
+    kmp_lp_t *e, int iter, uint32_t chunk, uint32_t pos_lo, uint32_t pos_hi,
+    void *d_out, uint32_t cap
+);
+
+/* Commit `count` proposals (DEVICE buffer d_props, 16B records; may be the
+ * concatenation of all ranks' phase-A outputs in rank order) for chunk
+ * `chunk` of sweep `iter`. Returns committed moves, or -1 on error. */
+int64_t kmp_lp_commit(
+    kmp_lp_t *e, int iter, uint32_t chunk, const void *d_props, uint32_t count
+);
+
+/* Finish a sharded run: download the partition, report stats. */
+int64_t kmp_lp_refine_end(kmp_lp_t *e, uint32_t *partition, kmp_lp_stats_t *stats);
+
+#ifdef __cplusplus
+} /* extern "C" */
+#endif
+
+#endif /* KAMINPAR_LP_H */

@@ -1,0 +1,253 @@
+"""kaminpar_amd: MI355X-native KaMinPar label-propagation hot path.
+
+ctypes bindings over libkaminpar_lp.so (C ABI: include/kaminpar_lp.h).
+Host-side graph handling works everywhere; the LP engine requires a GPU and
+fails loudly if the HIP device or the native library is missing -- there is
+no CPU fallback on the product path (the CPU oracle under oracle/ is test
+infrastructure only).
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "libkaminpar_lp.so")
+
+
+def _load():
+    if not os.path.exists(_LIB_PATH):
+        raise ImportError(
+            f"kaminpar_amd: native library not found at {_LIB_PATH}; "
+            "build it with __graft_entry__.build() or kaminpar_amd/csrc/build.sh"
+        )
+    lib = ctypes.CDLL(_LIB_PATH)
+
+    u32, u64, i32, i64 = ctypes.c_uint32, ctypes.c_uint64, ctypes.c_int32, ctypes.c_int64
+    p = ctypes.POINTER
+    vp = ctypes.c_void_p
+
+    lib.kmp_graph_from_csr.restype = vp
+    lib.kmp_graph_from_csr.argtypes = [u32, u64, p(u32), p(u32), p(i32), p(i32)]
+    lib.kmp_gen_rmat.restype = vp
+    lib.kmp_gen_rmat.argtypes = [ctypes.c_int, ctypes.c_int, u64]
+    lib.kmp_gen_rgg2d.restype = vp
+    lib.kmp_gen_rgg2d.argtypes = [u32, ctypes.c_double, u64]
+    lib.kmp_read_metis.restype = vp
+    lib.kmp_read_metis.argtypes = [ctypes.c_char_p]
+    lib.kmp_graph_n.restype = u32
+    lib.kmp_graph_n.argtypes = [vp]
+    lib.kmp_graph_m.restype = u64
+    lib.kmp_graph_m.argtypes = [vp]
+    lib.kmp_graph_xadj.restype = p(u32)
+    lib.kmp_graph_xadj.argtypes = [vp]
+    lib.kmp_graph_adjncy.restype = p(u32)
+    lib.kmp_graph_adjncy.argtypes = [vp]
+    lib.kmp_graph_vwgt.restype = p(i32)
+    lib.kmp_graph_vwgt.argtypes = [vp]
+    lib.kmp_graph_adjwgt.restype = p(i32)
+    lib.kmp_graph_adjwgt.argtypes = [vp]
+    lib.kmp_graph_total_node_weight.restype = i64
+    lib.kmp_graph_total_node_weight.argtypes = [vp]
+    lib.kmp_graph_free.argtypes = [vp]
+    lib.kmp_edge_cut_host.restype = i64
+    lib.kmp_edge_cut_host.argtypes = [vp, p(u32)]
+    lib.kmp_max_block_weight.restype = i64
+    lib.kmp_max_block_weight.argtypes = [vp, u32, ctypes.c_double]
+
+    lib.kmp_lp_create.restype = vp
+    lib.kmp_lp_create.argtypes = [vp]
+    lib.kmp_lp_free.argtypes = [vp]
+    lib.kmp_lp_refine.restype = i64
+    lib.kmp_lp_refine.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
+    lib.kmp_lp_cluster.restype = i64
+    lib.kmp_lp_cluster.argtypes = [vp, i64, u32, p(u32), u64, ctypes.c_int, vp]
+    lib.kmp_lp_num_chunks.restype = u32
+    lib.kmp_lp_num_chunks.argtypes = [vp]
+    lib.kmp_lp_refine_begin.restype = ctypes.c_int
+    lib.kmp_lp_refine_begin.argtypes = [vp, u32, p(i64), p(u32), u64]
+    lib.kmp_lp_phase_a.restype = i64
+    lib.kmp_lp_phase_a.argtypes = [vp, ctypes.c_int, u32, u32, u32, vp, u32]
+    lib.kmp_lp_commit.restype = i64
+    lib.kmp_lp_commit.argtypes = [vp, ctypes.c_int, u32, vp, u32]
+    lib.kmp_lp_refine_end.restype = i64
+    lib.kmp_lp_refine_end.argtypes = [vp, p(u32), vp]
+    return lib
+
+
+_lib = _load()
+
+
+class Stats(ctypes.Structure):
+    _fields_ = [
+        ("arcs_scanned", ctypes.c_uint64),
+        ("moves", ctypes.c_uint64),
+        ("phase_a_ns", ctypes.c_uint64),
+        ("total_ns", ctypes.c_uint64),
+        ("num_clusters", ctypes.c_uint64),
+        ("edge_cut", ctypes.c_int64),
+    ]
+
+
+def _u32p(a):
+    assert a.dtype == np.uint32 and a.flags["C_CONTIGUOUS"]
+    return a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32))
+
+
+def _i64p(a):
+    assert a.dtype == np.int64 and a.flags["C_CONTIGUOUS"]
+    return a.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
+
+
+class Graph:
+    """Host CSR graph in the reference layout (csr_graph.h:27-33)."""
+
+    def __init__(self, handle):
+        if not handle:
+            raise ValueError("graph construction failed")
+        self._h = handle
+
+    @classmethod
+    def from_csr(cls, xadj, adjncy, vwgt=None, adjwgt=None):
+        xadj = np.ascontiguousarray(xadj, dtype=np.uint32)
+        adjncy = np.ascontiguousarray(adjncy, dtype=np.uint32)
+        n = len(xadj) - 1
+        m = len(adjncy)
+        vp = None
+        ap = None
+        if vwgt is not None:
+            vwgt = np.ascontiguousarray(vwgt, dtype=np.int32)
+            vp = vwgt.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        if adjwgt is not None:
+            adjwgt = np.ascontiguousarray(adjwgt, dtype=np.int32)
+            ap = adjwgt.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        return cls(_lib.kmp_graph_from_csr(n, m, _u32p(xadj), _u32p(adjncy), vp, ap))
+
+    @classmethod
+    def rmat(cls, scale, edgefactor=8, seed=42):
+        return cls(_lib.kmp_gen_rmat(scale, edgefactor, seed))
+
+    @classmethod
+    def rgg2d(cls, n, avg_deg=16.0, seed=42):
+        return cls(_lib.kmp_gen_rgg2d(n, avg_deg, seed))
+
+    @classmethod
+    def read_metis(cls, path):
+        return cls(_lib.kmp_read_metis(os.fsencode(path)))
+
+    @property
+    def n(self):
+        return _lib.kmp_graph_n(self._h)
+
+    @property
+    def m(self):
+        return _lib.kmp_graph_m(self._h)
+
+    @property
+    def xadj(self):
+        return np.ctypeslib.as_array(_lib.kmp_graph_xadj(self._h), shape=(self.n + 1,))
+
+    @property
+    def adjncy(self):
+        return np.ctypeslib.as_array(_lib.kmp_graph_adjncy(self._h), shape=(self.m,))
+
+    @property
+    def total_node_weight(self):
+        return _lib.kmp_graph_total_node_weight(self._h)
+
+    def edge_cut(self, labels):
+        labels = np.ascontiguousarray(labels, dtype=np.uint32)
+        return _lib.kmp_edge_cut_host(self._h, _u32p(labels))
+
+    def max_block_weight(self, k, eps=0.03):
+        return _lib.kmp_max_block_weight(self._h, k, eps)
+
+    def __del__(self):
+        h = getattr(self, "_h", None)
+        if h:
+            _lib.kmp_graph_free(h)
+            self._h = None
+
+
+def random_partition(n, k, seed=42):
+    """Deterministic pseudo-random balanced-in-expectation partition."""
+    with np.errstate(over="ignore"):
+        u = np.arange(n, dtype=np.uint64)
+        x = u + np.uint64((0x9E3779B97F4A7C15 * (seed + 1)) & 0xFFFFFFFFFFFFFFFF)
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        x = x ^ (x >> np.uint64(31))
+    return (x % np.uint64(k)).astype(np.uint32)
+
+
+class LpEngine:
+    """Device-resident LP engine (requires a GPU; fails loudly otherwise)."""
+
+    def __init__(self, graph):
+        self._graph = graph  # keep alive
+        self._h = _lib.kmp_lp_create(graph._h)
+        if not self._h:
+            raise RuntimeError(
+                "kaminpar_amd: LP engine creation failed (no HIP GPU available?)"
+            )
+
+    def refine(self, k, max_block_weights, partition, seed=1, iters=5):
+        """Deterministic LP refinement; returns (cut, partition, Stats)."""
+        part = np.ascontiguousarray(partition, dtype=np.uint32).copy()
+        mbw = np.ascontiguousarray(max_block_weights, dtype=np.int64)
+        stats = Stats()
+        cut = _lib.kmp_lp_refine(
+            self._h, k, _i64p(mbw), _u32p(part), seed, iters, ctypes.byref(stats)
+        )
+        if cut < 0:
+            raise RuntimeError("kmp_lp_refine failed")
+        return cut, part, stats
+
+    def cluster(self, max_cluster_weight, clustering=None, desired=0, seed=1, iters=5):
+        """Deterministic LP clustering; returns (n_clusters, clustering, Stats)."""
+        n = self._graph.n
+        clus = np.zeros(n, dtype=np.uint32)
+        stats = Stats()
+        nc = _lib.kmp_lp_cluster(
+            self._h, max_cluster_weight, desired, _u32p(clus), seed, iters,
+            ctypes.byref(stats),
+        )
+        if nc < 0:
+            raise RuntimeError("kmp_lp_cluster failed")
+        return nc, clus, stats
+
+    # sharded API (multi-GPU orchestration; see kaminpar_amd/multi.py)
+    def num_chunks(self):
+        return _lib.kmp_lp_num_chunks(self._h)
+
+    def refine_begin(self, k, max_block_weights, partition, seed=1):
+        part = np.ascontiguousarray(partition, dtype=np.uint32)
+        mbw = np.ascontiguousarray(max_block_weights, dtype=np.int64)
+        rc = _lib.kmp_lp_refine_begin(self._h, k, _i64p(mbw), _u32p(part), seed)
+        if rc != 0:
+            raise RuntimeError("kmp_lp_refine_begin failed")
+
+    def phase_a(self, it, chunk, pos_lo, pos_hi, d_out_ptr, cap):
+        cnt = _lib.kmp_lp_phase_a(self._h, it, chunk, pos_lo, pos_hi, d_out_ptr, cap)
+        if cnt < 0:
+            raise RuntimeError("kmp_lp_phase_a failed")
+        return int(cnt)
+
+    def commit(self, it, chunk, d_props_ptr, count):
+        mv = _lib.kmp_lp_commit(self._h, it, chunk, d_props_ptr, count)
+        if mv < 0:
+            raise RuntimeError("kmp_lp_commit failed")
+        return int(mv)
+
+    def refine_end(self):
+        part = np.zeros(self._graph.n, dtype=np.uint32)
+        stats = Stats()
+        cut = _lib.kmp_lp_refine_end(self._h, _u32p(part), ctypes.byref(stats))
+        return cut, part, stats
+
+    def __del__(self):
+        h = getattr(self, "_h", None)
+        if h:
+            _lib.kmp_lp_free(h)
+            self._h = None

@@ -1,0 +1,1190 @@
+// MI355X (gfx950) HIP implementation of the KaMinPar label-propagation hot
+// path under the deterministic chunk-synchronous schedule (parity contract:
+// oracle/lp_oracle.cpp header). Semantics restated from
+//   kaminpar-shm/label_propagation.h (engine: gains :487-505, active set
+//   :848-870,1908-1912, try_node_move :817-841)
+//   kaminpar-shm/refinement/lp/lp_refiner.cc:151-285 (refiner select)
+//   kaminpar-shm/coarsening/clustering/lp_clusterer.cc:181-280 (clusterer
+//   select), with the commit fixpoint mirroring
+//   kaminpar-dist/refinement/lp/lp_refiner.cc:296-333 (rollback protocol).
+//
+// Kernel design (CDNA4): this is an irregular integer gather workload; the
+// roofline bound is HBM bandwidth (8 B per directed arc: 4 B adjncy + 4 B
+// labels gather), not MFMA. Per-vertex gain maps live in registers (degree
+// <= 16: 16-lane subgroups with shuffle-waterfall dedup) or LDS (dense
+// per-cluster arrays for k <= 2048), wavefronts are 64 wide, argmax uses
+// wave shuffles, and the commit is a sorted segmented-prefix fixpoint.
+
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string.h>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+
+#include <rocprim/rocprim.hpp>
+
+#include "../../include/kaminpar_lp.h"
+#include "lp_common.h"
+
+using kmp::i32;
+using kmp::i64;
+using kmp::u32;
+using kmp::u64;
+using kmp::FeistelPerm;
+using kmp::iter_seed_of;
+using kmp::tie_hash;
+
+#define HIP_CHECK(cmd)                                                                             \
+  do {                                                                                             \
+    hipError_t err_ = (cmd);                                                                       \
+    if (err_ != hipSuccess) {                                                                      \
+      fprintf(stderr, "kaminpar_amd: HIP error %s at %s:%d\n", hipGetErrorString(err_), __FILE__,  \
+              __LINE__);                                                                           \
+      abort();                                                                                     \
+    }                                                                                              \
+  } while (0)
+
+namespace {
+
+constexpr u32 kSmallDeg = 16;    // S bucket: <= 16 neighbours, 16 lanes/vertex
+constexpr u32 kMidDeg = 2048;    // M bucket: one wavefront per vertex
+constexpr u32 kMaxDenseK = 2048; // dense per-cluster LDS gains limit (refine)
+constexpr u32 kWave = 64;
+
+struct Prop { // 16-byte proposal record (ABI: uint32x4)
+  u32 u;
+  u32 to;
+  u32 rank;
+  u32 w; // node weight bits (i32 >= 0)
+};
+
+// --------------------------------------------------------------- binning
+__global__ void k_bin_impl(
+    u32 pos_lo,
+    u32 pos_hi,
+    u32 n,
+    u64 iter_seed,
+    u32 max_degree,
+    const u32 *__restrict__ xadj,
+    const uint8_t *__restrict__ active,
+    u64 *__restrict__ bucket_s,
+    u64 *__restrict__ bucket_m,
+    u64 *__restrict__ bucket_l,
+    u32 *__restrict__ counts, // [0]=S [1]=M [2]=L [3]=processed
+    u32 *__restrict__ processed,
+    unsigned long long *__restrict__ arcs
+) {
+  const u32 tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const u32 p = pos_lo + tid;
+  FeistelPerm perm(n, iter_seed);
+
+  u64 my_deg = 0;
+  if (p < pos_hi) {
+    const u32 u = perm(p);
+    const u32 deg = xadj[u + 1] - xadj[u];
+    if (deg <= max_degree && active[u]) {
+      my_deg = deg;
+      const u64 rec = (static_cast<u64>(p) << 32) | u;
+      if (deg <= kSmallDeg) {
+        bucket_s[atomicAdd(&counts[0], 1u)] = rec;
+      } else if (deg <= kMidDeg) {
+        bucket_m[atomicAdd(&counts[1], 1u)] = rec;
+      } else {
+        bucket_l[atomicAdd(&counts[2], 1u)] = rec;
+      }
+      processed[atomicAdd(&counts[3], 1u)] = u;
+    }
+  }
+  // wave-level reduction of scanned arcs
+  for (int off = 32; off > 0; off >>= 1) {
+    my_deg += __shfl_down(static_cast<unsigned long long>(my_deg), off, kWave);
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0 && my_deg > 0) {
+    atomicAdd(arcs, static_cast<unsigned long long>(my_deg));
+  }
+}
+
+// ------------------------------------------------- select helpers (device)
+struct BestState {
+  i32 gain;
+  u64 h;
+  u32 c;
+  bool have;
+};
+
+__device__ inline bool key_better(i32 g, u64 h, u32 c, const BestState &b) {
+  if (!b.have) {
+    return true;
+  }
+  if (g != b.gain) {
+    return g > b.gain;
+  }
+  if (h != b.h) {
+    return h > b.h;
+  }
+  return c < b.c;
+}
+
+// Weight-acceptance predicate (refiner variant, lp_refiner.cc:185-230).
+__device__ inline bool accept_refine(
+    u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw, i64 cur_w, i64 cur_maxw
+) {
+  return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw)) || (c == cur);
+}
+
+// ------------------------------------------------------ S bucket (deg<=16)
+// 16 lanes per vertex, 4 vertices per wave; gains deduped via shuffle
+// waterfall inside the 16-lane subgroup (no LDS, works for any k).
+__global__ void k_lp_small(
+    const u64 *__restrict__ bucket,
+    const u32 *__restrict__ counts, // counts[0] = S count
+    u32 chunk_base,
+    u64 iter_seed,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ vwgt,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    Prop *__restrict__ props,
+    u32 *__restrict__ prop_count
+) {
+  const u32 count = counts[0];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 sub = lane >> 4;          // subgroup 0..3 within wave
+  const u32 slot = lane & 15;         // lane within subgroup
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 vid = wave_id * 4 + sub;
+  if (vid >= count) {
+    return;
+  }
+
+  const u64 rec = bucket[vid];
+  const u32 p = static_cast<u32>(rec >> 32);
+  const u32 u = static_cast<u32>(rec);
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+
+  // candidate load: lane slot handles edge slot
+  u32 c = 0xFFFFFFFFu;
+  i32 w = 0;
+  if (slot < deg) {
+    const u32 v = adjncy[row + slot];
+    c = labels[v];
+    w = adjwgt ? adjwgt[row + slot] : 1;
+  }
+
+  // dedupe within subgroup: sum weights of equal clusters; lowest slot owns
+  i32 gain = w;
+  bool owner = (slot < deg);
+  const u32 base = sub * 16;
+  for (u32 j = 0; j < 16; ++j) {
+    const u32 cj = __shfl(c, base + j, kWave);
+    const i32 wj = __shfl(w, base + j, kWave);
+    if (j != slot && cj == c && c != 0xFFFFFFFFu) {
+      if (cj != 0xFFFFFFFFu) {
+        gain += wj;
+        if (j < slot) {
+          owner = false;
+        }
+      }
+    }
+  }
+
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+  const i64 cur_maxw = maxw[cur];
+
+  // per-lane candidate key
+  BestState best{0, 0, 0, false};
+  if (owner && c != 0xFFFFFFFFu) {
+    const i64 cw = weights[c];
+    const i64 mw = maxw[c];
+    if (accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      best = BestState{gain, tie_hash(iter_seed, u, c), c, true};
+    }
+  }
+
+  // subgroup argmax over 16 lanes
+  for (int off = 8; off > 0; off >>= 1) {
+    const i32 og = __shfl_down(best.gain, off, kWave);
+    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+    const u32 oc = __shfl_down(best.c, off, kWave);
+    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+    // only combine within the same 16-lane subgroup
+    if ((slot + off) < 16 && ohave && key_better(og, oh, oc, best)) {
+      best = BestState{og, oh, oc, true};
+    }
+  }
+
+  if (slot == 0 && best.have && best.c != cur) {
+    const u32 idx = atomicAdd(prop_count, 1u);
+    props[idx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  }
+}
+
+// -------------------------------------------------- M bucket (wave/vertex)
+// Dense per-cluster LDS gains (k <= kMaxDenseK); one wavefront per vertex.
+// blockDim.x = 256 (4 waves); dynamic LDS = 4 * k * sizeof(i32).
+__global__ void k_lp_mid(
+    const u64 *__restrict__ bucket,
+    const u32 *__restrict__ counts, // counts[1] = M count
+    u32 chunk_base,
+    u64 iter_seed,
+    u32 k,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ vwgt,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    Prop *__restrict__ props,
+    u32 *__restrict__ prop_count
+) {
+  extern __shared__ i32 lds[];
+  const u32 count = counts[1];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wave_in_wg = threadIdx.x >> 6;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  if (wave_id >= count) {
+    return;
+  }
+  i32 *gains = lds + wave_in_wg * k;
+
+  for (u32 c = lane; c < k; c += kWave) {
+    gains[c] = 0;
+  }
+  __threadfence_block(); // LDS ordering within the wave (lockstep lanes; the
+                         // gains slice is private to this wave)
+
+  const u64 rec = bucket[wave_id];
+  const u32 p = static_cast<u32>(rec >> 32);
+  const u32 u = static_cast<u32>(rec);
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+
+  for (u32 e = lane; e < deg; e += kWave) {
+    const u32 v = adjncy[row + e];
+    const i32 w = adjwgt ? adjwgt[row + e] : 1;
+    atomicAdd(&gains[labels[v]], w);
+  }
+  __threadfence_block();
+
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+  const i64 cur_maxw = maxw[cur];
+
+  BestState best{0, 0, 0, false};
+  for (u32 c = lane; c < k; c += kWave) {
+    const i32 g = gains[c];
+    if (g <= 0) {
+      continue;
+    }
+    const i64 cw = weights[c];
+    const i64 mw = maxw[c];
+    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      continue;
+    }
+    const u64 h = tie_hash(iter_seed, u, c);
+    if (key_better(g, h, c, best)) {
+      best = BestState{g, h, c, true};
+    }
+  }
+
+  for (int off = 32; off > 0; off >>= 1) {
+    const i32 og = __shfl_down(best.gain, off, kWave);
+    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+    const u32 oc = __shfl_down(best.c, off, kWave);
+    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+    if (ohave && key_better(og, oh, oc, best)) {
+      best = BestState{og, oh, oc, true};
+    }
+  }
+
+  if (lane == 0 && best.have && best.c != cur) {
+    const u32 idx = atomicAdd(prop_count, 1u);
+    props[idx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  }
+}
+
+// ---------------------------------------------- L bucket (workgroup/vertex)
+// One 256-thread workgroup per vertex, shared dense LDS gains (k <= 2048).
+__global__ void k_lp_large(
+    const u64 *__restrict__ bucket,
+    const u32 *__restrict__ counts, // counts[2] = L count
+    u32 chunk_base,
+    u64 iter_seed,
+    u32 k,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ vwgt,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    Prop *__restrict__ props,
+    u32 *__restrict__ prop_count
+) {
+  extern __shared__ i32 lds[];
+  const u32 count = counts[2];
+  if (blockIdx.x >= count) {
+    return;
+  }
+  i32 *gains = lds;                                       // k i32
+  i64 *red = reinterpret_cast<i64 *>(lds + ((k + 1) & ~1u)); // 4-wave scratch
+
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    gains[c] = 0;
+  }
+  __syncthreads();
+
+  const u64 rec = bucket[blockIdx.x];
+  const u32 p = static_cast<u32>(rec >> 32);
+  const u32 u = static_cast<u32>(rec);
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+
+  for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
+    const u32 v = adjncy[row + e];
+    const i32 w = adjwgt ? adjwgt[row + e] : 1;
+    atomicAdd(&gains[labels[v]], w);
+  }
+  __syncthreads();
+
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+  const i64 cur_maxw = maxw[cur];
+
+  BestState best{0, 0, 0, false};
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    const i32 g = gains[c];
+    if (g <= 0) {
+      continue;
+    }
+    const i64 cw = weights[c];
+    const i64 mw = maxw[c];
+    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      continue;
+    }
+    const u64 h = tie_hash(iter_seed, u, c);
+    if (key_better(g, h, c, best)) {
+      best = BestState{g, h, c, true};
+    }
+  }
+  const u32 lane = threadIdx.x & (kWave - 1);
+  for (int off = 32; off > 0; off >>= 1) {
+    const i32 og = __shfl_down(best.gain, off, kWave);
+    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+    const u32 oc = __shfl_down(best.c, off, kWave);
+    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+    if (ohave && key_better(og, oh, oc, best)) {
+      best = BestState{og, oh, oc, true};
+    }
+  }
+  // cross-wave reduce via LDS (4 waves)
+  const u32 wave_in_wg = threadIdx.x >> 6;
+  if (lane == 0) {
+    red[wave_in_wg * 2] = (static_cast<i64>(best.gain) << 1) | (best.have ? 1 : 0);
+    red[wave_in_wg * 2 + 1] = static_cast<i64>(best.h);
+    reinterpret_cast<u32 *>(red + 8)[wave_in_wg] = best.c;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    BestState total{0, 0, 0, false};
+    const u32 waves = blockDim.x >> 6;
+    for (u32 wv = 0; wv < waves; ++wv) {
+      const i64 packed = red[wv * 2];
+      if (packed & 1) {
+        const i32 g = static_cast<i32>(packed >> 1);
+        const u64 h = static_cast<u64>(red[wv * 2 + 1]);
+        const u32 c = reinterpret_cast<u32 *>(red + 8)[wv];
+        if (key_better(g, h, c, total)) {
+          total = BestState{g, h, c, true};
+        }
+      }
+    }
+    if (total.have && total.c != cur) {
+      const u32 idx = atomicAdd(prop_count, 1u);
+      props[idx] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+    }
+  }
+}
+
+// -------------------------------------------------------------- commit
+__global__ void k_extract_sorted(
+    const u32 *__restrict__ order, // sorted proposal indices
+    const Prop *__restrict__ props,
+    u32 count,
+    u32 *__restrict__ sto,
+    i64 *__restrict__ sw
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    const Prop pr = props[order[i]];
+    sto[i] = pr.to;
+    sw[i] = static_cast<i64>(static_cast<i32>(pr.w));
+  }
+}
+
+__global__ void k_seg_bounds(
+    const u32 *__restrict__ sto,
+    u32 count,
+    u32 *__restrict__ seg_begin,
+    u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  if (i == 0 || sto[i - 1] != to) {
+    seg_begin[to] = i;
+  }
+  if (i == count - 1 || sto[i + 1] != to) {
+    seg_end[to] = i + 1;
+  }
+  (void)prefix_len;
+}
+
+__global__ void k_seg_len(
+    u32 k_or_n,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len
+) {
+  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c < k_or_n && seg_end[c] > seg_begin[c]) {
+    prefix_len[c] = seg_end[c] - seg_begin[c];
+  }
+}
+
+__global__ void k_dep(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ labels,
+    unsigned long long *__restrict__ dep
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  if (i - seg_begin[to] < prefix_len[to]) {
+    const Prop pr = props[order[i]];
+    atomicAdd(&dep[labels[pr.u]], static_cast<unsigned long long>(pr.w));
+  }
+}
+
+__global__ void k_cutoff(
+    u32 k_or_n,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len,
+    const i64 *__restrict__ pw, // within-segment inclusive prefix weights
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    const unsigned long long *__restrict__ dep,
+    int *__restrict__ changed
+) {
+  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= k_or_n) {
+    return;
+  }
+  const u32 b = seg_begin[c], e = seg_end[c];
+  if (e <= b) {
+    return;
+  }
+  const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(dep[c]);
+  const u32 old_len = prefix_len[c];
+  // binary search: largest t <= old_len with pw[b + t - 1] <= capacity
+  u32 lo = 0, hi = old_len;
+  while (lo < hi) {
+    const u32 mid = (lo + hi + 1) >> 1;
+    if (pw[b + mid - 1] <= capacity) {
+      lo = mid;
+    } else {
+      hi = mid - 1;
+    }
+  }
+  if (lo < old_len) {
+    prefix_len[c] = lo;
+    atomicExch(changed, 1);
+  }
+}
+
+__global__ void k_weights_update(
+    u32 k_or_n,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ seg_end,
+    const u32 *__restrict__ prefix_len,
+    const i64 *__restrict__ pw,
+    const unsigned long long *__restrict__ dep,
+    i64 *__restrict__ weights
+) {
+  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= k_or_n) {
+    return;
+  }
+  i64 arr = 0;
+  const u32 b = seg_begin[c], e = seg_end[c];
+  if (e > b && prefix_len[c] > 0) {
+    arr = pw[b + prefix_len[c] - 1];
+  }
+  const i64 delta = arr - static_cast<i64>(dep[c]);
+  if (delta != 0) {
+    weights[c] += delta;
+  }
+}
+
+__global__ void k_apply(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ prefix_len,
+    u32 *__restrict__ labels,
+    u32 *__restrict__ admitted_idx, // compact list of admitted proposal idxs
+    u32 *__restrict__ admitted_count
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  if (i - seg_begin[to] < prefix_len[to]) {
+    const u32 pi = order[i];
+    labels[props[pi].u] = to;
+    admitted_idx[atomicAdd(admitted_count, 1u)] = pi;
+  }
+}
+
+__global__ void k_clear_active(
+    const u32 *__restrict__ processed, const u32 *__restrict__ counts, uint8_t *__restrict__ active
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < counts[3]) {
+    active[processed[i]] = 0;
+  }
+}
+
+__global__ void k_activate(
+    const u32 *__restrict__ admitted_idx,
+    const u32 *__restrict__ admitted_count,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    uint8_t *__restrict__ active
+) {
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 lane = threadIdx.x & (kWave - 1);
+  if (wave_id >= *admitted_count) {
+    return;
+  }
+  const u32 u = props[admitted_idx[wave_id]].u;
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+  for (u32 e = lane; e < deg; e += kWave) {
+    active[adjncy[row + e]] = 1;
+  }
+}
+
+__global__ void k_reset_segs(
+    const u32 *__restrict__ sto,
+    u32 count,
+    u32 *__restrict__ seg_begin,
+    u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  seg_begin[to] = 0;
+  seg_end[to] = 0;
+  prefix_len[to] = 0;
+  dep[to] = 0;
+}
+
+__global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep) {
+  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c < k_or_n) {
+    dep[c] = 0;
+  }
+}
+
+// dep array must be reset per fixpoint round for ALL clusters that can
+// appear as sources; for refinement k is small so a full reset is cheap.
+
+__global__ void k_init_weights(
+    u32 n,
+    const u32 *__restrict__ labels,
+    const i32 *__restrict__ vwgt,
+    unsigned long long *__restrict__ weights
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u < n) {
+    atomicAdd(&weights[labels[u]], static_cast<unsigned long long>(vwgt ? vwgt[u] : 1));
+  }
+}
+
+__global__ void k_make_keys(
+    const Prop *__restrict__ props, u32 count, u64 *__restrict__ keys, u32 *__restrict__ vals
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    keys[i] = (static_cast<u64>(props[i].to) << 32) | props[i].rank;
+    vals[i] = i;
+  }
+}
+
+__global__ void k_edge_cut(
+    u32 n,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    unsigned long long *__restrict__ cut
+) {
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 lane = threadIdx.x & (kWave - 1);
+  if (wave_id >= n) {
+    return;
+  }
+  const u32 u = wave_id;
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+  const u32 lu = labels[u];
+  u64 local = 0;
+  for (u32 e = lane; e < deg; e += kWave) {
+    if (labels[adjncy[row + e]] != lu) {
+      local += adjwgt ? adjwgt[row + e] : 1;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    local += __shfl_down(static_cast<unsigned long long>(local), off, kWave);
+  }
+  if (lane == 0 && local) {
+    atomicAdd(cut, static_cast<unsigned long long>(local));
+  }
+}
+
+} // namespace
+
+// ================================================================ engine
+struct kmp_lp_t {
+  u32 n = 0;
+  u64 m = 0;
+  u32 k = 0;
+  u64 seed = 1;
+  u32 C = 0; // chunk size
+  bool has_vwgt = false, has_adjwgt = false;
+
+  // device graph
+  u32 *d_xadj = nullptr;
+  u32 *d_adjncy = nullptr;
+  i32 *d_vwgt = nullptr;
+  i32 *d_adjwgt = nullptr;
+
+  // device LP state
+  u32 *d_labels = nullptr;
+  i64 *d_weights = nullptr;
+  i64 *d_maxw = nullptr;
+  uint8_t *d_active = nullptr;
+
+  // phase buffers
+  u64 *d_bucket_s = nullptr, *d_bucket_m = nullptr, *d_bucket_l = nullptr;
+  u32 *d_counts = nullptr;    // 4 u32
+  u32 *d_processed = nullptr; // C
+  Prop *d_props = nullptr;    // C
+  u32 *d_prop_count = nullptr;
+  unsigned long long *d_arcs = nullptr;
+
+  // commit buffers
+  u64 *d_sort_keys[2] = {nullptr, nullptr};
+  u32 *d_sort_vals[2] = {nullptr, nullptr};
+  void *d_sort_temp = nullptr;
+  size_t sort_temp_bytes = 0;
+  u32 *d_sto = nullptr;
+  i64 *d_sw = nullptr;
+  i64 *d_pw = nullptr;
+  void *d_scan_temp = nullptr;
+  size_t scan_temp_bytes = 0;
+  u32 *d_seg_begin = nullptr, *d_seg_end = nullptr, *d_prefix_len = nullptr;
+  unsigned long long *d_dep = nullptr;
+  int *d_changed = nullptr;
+  u32 *d_admitted_idx = nullptr;
+  u32 *d_admitted_count = nullptr;
+  unsigned long long *d_cut = nullptr;
+
+  // pinned host mirrors
+  u32 *h_count = nullptr; // pinned: prop count / admitted count
+  int *h_changed = nullptr;
+
+  hipStream_t stream = nullptr;
+
+  // run bookkeeping
+  u64 arcs_scanned = 0;
+  u64 moves = 0;
+  double phase_a_ms = 0.0;
+  std::vector<hipEvent_t> ev_pool;
+  size_t ev_used = 0;
+
+  hipEvent_t ev_begin() {
+    if (ev_used + 2 > ev_pool.size()) {
+      hipEvent_t a, b;
+      HIP_CHECK(hipEventCreate(&a));
+      HIP_CHECK(hipEventCreate(&b));
+      ev_pool.push_back(a);
+      ev_pool.push_back(b);
+    }
+    return ev_pool[ev_used];
+  }
+};
+
+namespace {
+
+u32 ceil_div(u64 a, u64 b) { return static_cast<u32>((a + b - 1) / b); }
+
+void engine_alloc_k_buffers(kmp_lp_t *e, u32 k_or_n) {
+  // commit per-cluster arrays sized to k (refine) / n (cluster)
+  HIP_CHECK(hipMalloc(&e->d_seg_begin, sizeof(u32) * k_or_n));
+  HIP_CHECK(hipMalloc(&e->d_seg_end, sizeof(u32) * k_or_n));
+  HIP_CHECK(hipMalloc(&e->d_prefix_len, sizeof(u32) * k_or_n));
+  HIP_CHECK(hipMalloc(&e->d_dep, sizeof(unsigned long long) * k_or_n));
+  HIP_CHECK(hipMemsetAsync(e->d_seg_begin, 0, sizeof(u32) * k_or_n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_seg_end, 0, sizeof(u32) * k_or_n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_prefix_len, 0, sizeof(u32) * k_or_n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * k_or_n, e->stream));
+}
+
+void engine_free_k_buffers(kmp_lp_t *e) {
+  if (e->d_seg_begin) {
+    HIP_CHECK(hipFree(e->d_seg_begin));
+    e->d_seg_begin = nullptr;
+  }
+  if (e->d_seg_end) {
+    HIP_CHECK(hipFree(e->d_seg_end));
+    e->d_seg_end = nullptr;
+  }
+  if (e->d_prefix_len) {
+    HIP_CHECK(hipFree(e->d_prefix_len));
+    e->d_prefix_len = nullptr;
+  }
+  if (e->d_dep) {
+    HIP_CHECK(hipFree(e->d_dep));
+    e->d_dep = nullptr;
+  }
+}
+
+} // namespace
+
+extern "C" {
+
+kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
+  int ndev = 0;
+  if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) {
+    fprintf(stderr, "kaminpar_amd: no HIP device available -- the LP engine requires a GPU\n");
+    return nullptr;
+  }
+
+  auto *e = new kmp_lp_t();
+  e->n = kmp_graph_n(g);
+  e->m = kmp_graph_m(g);
+  e->C = kmp::chunk_size_for(e->n);
+  e->has_vwgt = kmp_graph_vwgt(g) != nullptr;
+  e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
+  HIP_CHECK(hipStreamCreate(&e->stream));
+
+  HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
+  HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
+  HIP_CHECK(hipMemcpy(e->d_xadj, kmp_graph_xadj(g), sizeof(u32) * (e->n + 1), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(e->d_adjncy, kmp_graph_adjncy(g), sizeof(u32) * e->m, hipMemcpyHostToDevice));
+  if (e->has_vwgt) {
+    HIP_CHECK(hipMalloc(&e->d_vwgt, sizeof(i32) * e->n));
+    HIP_CHECK(hipMemcpy(e->d_vwgt, kmp_graph_vwgt(g), sizeof(i32) * e->n, hipMemcpyHostToDevice));
+  }
+  if (e->has_adjwgt) {
+    HIP_CHECK(hipMalloc(&e->d_adjwgt, sizeof(i32) * e->m));
+    HIP_CHECK(
+        hipMemcpy(e->d_adjwgt, kmp_graph_adjwgt(g), sizeof(i32) * e->m, hipMemcpyHostToDevice)
+    );
+  }
+
+  HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
+  HIP_CHECK(hipMalloc(&e->d_active, e->n));
+
+  const u32 C = e->C;
+  HIP_CHECK(hipMalloc(&e->d_bucket_s, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_bucket_m, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_bucket_l, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_counts, sizeof(u32) * 4));
+  HIP_CHECK(hipMalloc(&e->d_processed, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
+  HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
+  HIP_CHECK(hipMalloc(&e->d_arcs, sizeof(unsigned long long)));
+
+  HIP_CHECK(hipMalloc(&e->d_sort_keys[0], sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_sort_keys[1], sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_sort_vals[0], sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_sort_vals[1], sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_sto, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_sw, sizeof(i64) * C));
+  HIP_CHECK(hipMalloc(&e->d_pw, sizeof(i64) * C));
+  HIP_CHECK(hipMalloc(&e->d_changed, sizeof(int)));
+  HIP_CHECK(hipMalloc(&e->d_admitted_idx, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_admitted_count, sizeof(u32)));
+  HIP_CHECK(hipMalloc(&e->d_cut, sizeof(unsigned long long)));
+
+  // rocprim temp sizes (max problem size C)
+  rocprim::double_buffer<u64> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
+  rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
+  HIP_CHECK(rocprim::radix_sort_pairs(nullptr, e->sort_temp_bytes, keys, vals, C));
+  HIP_CHECK(hipMalloc(&e->d_sort_temp, e->sort_temp_bytes));
+  HIP_CHECK(rocprim::inclusive_scan_by_key(
+      nullptr, e->scan_temp_bytes, e->d_sto, e->d_sw, e->d_pw, C, rocprim::plus<i64>(),
+      rocprim::equal_to<u32>()
+  ));
+  HIP_CHECK(hipMalloc(&e->d_scan_temp, e->scan_temp_bytes));
+
+  HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
+  HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
+  return e;
+}
+
+void kmp_lp_free(kmp_lp_t *e) {
+  if (!e) {
+    return;
+  }
+  HIP_CHECK(hipDeviceSynchronize());
+  for (hipEvent_t ev : e->ev_pool) {
+    (void)hipEventDestroy(ev);
+  }
+  engine_free_k_buffers(e);
+  for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
+                  (void *)e->d_labels, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
+                  (void *)e->d_bucket_s, (void *)e->d_bucket_m, (void *)e->d_bucket_l,
+                  (void *)e->d_counts, (void *)e->d_processed, (void *)e->d_props,
+                  (void *)e->d_prop_count, (void *)e->d_arcs, (void *)e->d_sort_keys[0],
+                  (void *)e->d_sort_keys[1], (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1],
+                  (void *)e->d_sort_temp, (void *)e->d_sto, (void *)e->d_sw, (void *)e->d_pw,
+                  (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_idx,
+                  (void *)e->d_admitted_count, (void *)e->d_cut}) {
+    if (p) {
+      (void)hipFree(p);
+    }
+  }
+  if (e->h_count) {
+    (void)hipHostFree(e->h_count);
+  }
+  if (e->h_changed) {
+    (void)hipHostFree(e->h_changed);
+  }
+  (void)hipStreamDestroy(e->stream);
+  delete e;
+}
+
+u32 kmp_lp_num_chunks(const kmp_lp_t *) { return kmp::kNumChunks; }
+
+int kmp_lp_refine_begin(
+    kmp_lp_t *e, u32 k, const i64 *max_block_weights, const u32 *partition, u64 seed
+) {
+  if (k > kMaxDenseK) {
+    fprintf(stderr, "kaminpar_amd: refine currently supports k <= %u (got %u)\n", kMaxDenseK, k);
+    return -1;
+  }
+  e->k = k;
+  e->seed = seed;
+  e->arcs_scanned = 0;
+  e->moves = 0;
+  e->phase_a_ms = 0.0;
+  e->ev_used = 0;
+
+  engine_free_k_buffers(e);
+  engine_alloc_k_buffers(e, k);
+
+  if (e->d_weights) {
+    HIP_CHECK(hipFree(e->d_weights));
+  }
+  if (e->d_maxw) {
+    HIP_CHECK(hipFree(e->d_maxw));
+  }
+  HIP_CHECK(hipMalloc(&e->d_weights, sizeof(i64) * k));
+  HIP_CHECK(hipMalloc(&e->d_maxw, sizeof(i64) * k));
+  HIP_CHECK(hipMemcpy(e->d_maxw, max_block_weights, sizeof(i64) * k, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(e->d_labels, partition, sizeof(u32) * e->n, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
+
+  // initial block weights on device
+  HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * k, e->stream));
+  {
+    const u32 threads = 256;
+    hipLaunchKernelGGL(
+        k_init_weights, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
+        e->d_labels, e->d_vwgt, reinterpret_cast<unsigned long long *>(e->d_weights)
+    );
+  }
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  return 0;
+}
+
+i64 kmp_lp_phase_a(
+    kmp_lp_t *e, int iter, u32 chunk, u32 pos_lo, u32 pos_hi, void *d_out, u32 cap
+) {
+  const u64 iseed = iter_seed_of(e->seed, iter);
+  const u32 chunk_base = chunk * e->C;
+  Prop *props = static_cast<Prop *>(d_out);
+
+  HIP_CHECK(hipMemsetAsync(e->d_counts, 0, sizeof(u32) * 4, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_prop_count, 0, sizeof(u32), e->stream));
+
+  const u32 span = pos_hi - pos_lo;
+  const u32 threads = 256;
+
+  hipEvent_t ev0 = e->ev_begin();
+  hipEvent_t ev1 = e->ev_pool[e->ev_used + 1];
+  e->ev_used += 2;
+  HIP_CHECK(hipEventRecord(ev0, e->stream));
+
+  hipLaunchKernelGGL(
+      k_bin_impl, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
+      iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_bucket_s, e->d_bucket_m, e->d_bucket_l,
+      e->d_counts, e->d_processed, e->d_arcs
+  );
+
+  // S: 4 vertices/wave -> span/4 waves max
+  {
+    const u32 waves = ceil_div(span, 4);
+    hipLaunchKernelGGL(
+        k_lp_small, dim3(ceil_div(waves * kWave, threads)), dim3(threads), 0, e->stream,
+        e->d_bucket_s, e->d_counts, chunk_base, iseed, e->d_xadj, e->d_adjncy, e->d_vwgt,
+        e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, props, e->d_prop_count
+    );
+  }
+  // M: 1 vertex/wave
+  {
+    const u32 waves = span; // upper bound
+    const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
+    hipLaunchKernelGGL(
+        k_lp_mid, dim3(ceil_div(waves * kWave, threads)), dim3(threads), lds, e->stream,
+        e->d_bucket_m, e->d_counts, chunk_base, iseed, e->k, e->d_xadj, e->d_adjncy, e->d_vwgt,
+        e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, props, e->d_prop_count
+    );
+  }
+  // L: 1 vertex/workgroup. Grid must cover the worst case; blocks early-exit.
+  {
+    const size_t lds = (static_cast<size_t>((e->k + 1) & ~1u)) * sizeof(i32) + 16 * sizeof(i64);
+    hipLaunchKernelGGL(
+        k_lp_large, dim3(4096), dim3(256), lds, e->stream, e->d_bucket_l, e->d_counts, chunk_base,
+        iseed, e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
+        e->d_maxw, props, e->d_prop_count
+    );
+  }
+  HIP_CHECK(hipEventRecord(ev1, e->stream));
+
+  HIP_CHECK(hipMemcpyAsync(e->h_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+  e->phase_a_ms += ms;
+
+  if (*e->h_count > cap) {
+    fprintf(stderr, "kaminpar_amd: proposal buffer overflow (%u > %u)\n", *e->h_count, cap);
+    return -1;
+  }
+  return static_cast<i64>(*e->h_count);
+}
+
+i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 count) {
+  (void)iter;
+  (void)chunk;
+  const u32 threads = 256;
+  if (count == 0) {
+    // no proposals: still clear the active flags of processed vertices
+    hipLaunchKernelGGL(
+        k_clear_active, dim3(ceil_div(e->C, threads)), dim3(threads), 0, e->stream, e->d_processed,
+        e->d_counts, e->d_active
+    );
+    return 0;
+  }
+  const Prop *props = static_cast<const Prop *>(d_props);
+  const u32 grid = ceil_div(count, threads);
+
+  // build sort keys (to<<32 | rank) + index values
+  hipLaunchKernelGGL(
+      k_make_keys, dim3(grid), dim3(threads), 0, e->stream, props, count, e->d_sort_keys[0],
+      e->d_sort_vals[0]
+  );
+  rocprim::double_buffer<u64> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
+  rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
+  size_t tb = e->sort_temp_bytes;
+  HIP_CHECK(rocprim::radix_sort_pairs(e->d_sort_temp, tb, keys, vals, count, 0, 64, e->stream));
+  u32 *order = vals.current();
+
+  hipLaunchKernelGGL(
+      k_extract_sorted, dim3(grid), dim3(threads), 0, e->stream, order, props, count, e->d_sto,
+      e->d_sw
+  );
+  size_t sb = e->scan_temp_bytes;
+  HIP_CHECK(rocprim::inclusive_scan_by_key(
+      e->d_scan_temp, sb, e->d_sto, e->d_sw, e->d_pw, count, rocprim::plus<i64>(),
+      rocprim::equal_to<u32>(), e->stream
+  ));
+  hipLaunchKernelGGL(
+      k_seg_bounds, dim3(grid), dim3(threads), 0, e->stream, e->d_sto, count, e->d_seg_begin,
+      e->d_seg_end, e->d_prefix_len
+  );
+  const u32 kgrid = ceil_div(e->k, threads);
+  hipLaunchKernelGGL(
+      k_seg_len, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
+      e->d_prefix_len
+  );
+
+  // fixpoint rounds
+  while (true) {
+    hipLaunchKernelGGL(
+        k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep
+    );
+    hipLaunchKernelGGL(
+        k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, e->d_sto, count,
+        e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
+    );
+    HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
+    hipLaunchKernelGGL(
+        k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
+        e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
+    );
+    HIP_CHECK(
+        hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
+    );
+    HIP_CHECK(hipStreamSynchronize(e->stream));
+    if (!*e->h_changed) {
+      break;
+    }
+  }
+
+  hipLaunchKernelGGL(
+      k_weights_update, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
+      e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
+  );
+  HIP_CHECK(hipMemsetAsync(e->d_admitted_count, 0, sizeof(u32), e->stream));
+  hipLaunchKernelGGL(
+      k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, e->d_sto, count,
+      e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_admitted_idx, e->d_admitted_count
+  );
+  hipLaunchKernelGGL(
+      k_clear_active, dim3(ceil_div(e->C, threads)), dim3(threads), 0, e->stream, e->d_processed,
+      e->d_counts, e->d_active
+  );
+  hipLaunchKernelGGL(
+      k_activate, dim3(ceil_div(count * kWave, threads)), dim3(threads), 0, e->stream,
+      e->d_admitted_idx, e->d_admitted_count, props, e->d_xadj, e->d_adjncy, e->d_active
+  );
+  // reset per-cluster segment state for the next chunk (touched entries only)
+  hipLaunchKernelGGL(
+      k_reset_segs, dim3(grid), dim3(threads), 0, e->stream, e->d_sto, count, e->d_seg_begin,
+      e->d_seg_end, e->d_prefix_len, e->d_dep
+  );
+
+  HIP_CHECK(
+      hipMemcpyAsync(e->h_count + 1, e->d_admitted_count, sizeof(u32), hipMemcpyDeviceToHost,
+                     e->stream)
+  );
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  e->moves += e->h_count[1];
+  return static_cast<i64>(e->h_count[1]);
+}
+
+i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
+  HIP_CHECK(hipMemsetAsync(e->d_cut, 0, sizeof(unsigned long long), e->stream));
+  hipLaunchKernelGGL(
+      k_edge_cut, dim3(ceil_div(static_cast<u64>(e->n) * kWave, 256)), dim3(256), 0, e->stream,
+      e->n, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_labels, e->d_cut
+  );
+  unsigned long long cut2 = 0;
+  HIP_CHECK(hipMemcpyAsync(&cut2, e->d_cut, sizeof(cut2), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpy(partition, e->d_labels, sizeof(u32) * e->n, hipMemcpyDeviceToHost));
+  unsigned long long arcs = 0;
+  HIP_CHECK(hipMemcpy(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost));
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+
+  if (stats) {
+    stats->arcs_scanned = arcs;
+    stats->moves = e->moves;
+    stats->phase_a_ns = static_cast<u64>(e->phase_a_ms * 1e6);
+    stats->total_ns = 0; // caller times the region
+    stats->num_clusters = 0;
+    stats->edge_cut = static_cast<i64>(cut2 / 2);
+  }
+  return static_cast<i64>(cut2 / 2);
+}
+
+i64 kmp_lp_refine(
+    kmp_lp_t *e,
+    u32 k,
+    const i64 *max_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    kmp_lp_stats_t *stats
+) {
+  if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
+    return -1;
+  }
+  const u32 num_chunks = kmp::kNumChunks;
+  for (int iter = 0; iter < iters; ++iter) {
+    u64 sweep_moves = 0;
+    for (u32 chunk = 0; chunk < num_chunks; ++chunk) {
+      const u32 pos_lo = chunk * e->C;
+      const u32 pos_hi = pos_lo + e->C > e->n ? e->n : pos_lo + e->C;
+      if (pos_lo >= pos_hi) {
+        continue;
+      }
+      const i64 cnt = kmp_lp_phase_a(e, iter, chunk, pos_lo, pos_hi, e->d_props, e->C);
+      if (cnt < 0) {
+        return -1;
+      }
+      const i64 mv = kmp_lp_commit(e, iter, chunk, e->d_props, static_cast<u32>(cnt));
+      if (mv < 0) {
+        return -1;
+      }
+      sweep_moves += mv;
+    }
+    if (sweep_moves == 0) {
+      break;
+    }
+  }
+  return kmp_lp_refine_end(e, partition, stats);
+}
+
+i64 kmp_lp_cluster(
+    kmp_lp_t *e, i64 max_cluster_weight, u32 desired_clusters, u32 *clustering, u64 seed,
+    int iters, kmp_lp_stats_t *stats
+) {
+  (void)e;
+  (void)max_cluster_weight;
+  (void)desired_clusters;
+  (void)clustering;
+  (void)seed;
+  (void)iters;
+  (void)stats;
+  fprintf(stderr, "kaminpar_amd: kmp_lp_cluster GPU path not implemented yet\n");
+  return -1;
+}
+
+} // extern "C"

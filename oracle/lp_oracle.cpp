@@ -337,7 +337,7 @@ void lp_run(
           const u64 h = tie_hash(iter_seed, u, c);
 
           if (store_favored) {
-            if (r > fav_gain || (r == fav_gain && h > fav_h)) {
+            if (r > fav_gain || (r == fav_gain && (h > fav_h || (h == fav_h && c < fav)))) {
               fav_gain = r;
               fav = c;
               fav_h = h;
@@ -370,8 +370,10 @@ void lp_run(
             better = true;
           } else if (r != best_gain) {
             better = r > best_gain;
-          } else {
+          } else if (h != best_h) {
             better = h > best_h;
+          } else {
+            better = c < best; // total order (matches device key_better)
           }
           if (better) {
             best = c;

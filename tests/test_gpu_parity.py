@@ -1,0 +1,147 @@
+"""GPU parity tests: the HIP LP path must be BIT-IDENTICAL to the CPU oracle
+under the deterministic chunk-synchronous schedule (labels, block weights,
+edge cut). Covers the S (deg<=16), M (wave) and L (workgroup) kernel paths,
+weighted graphs, several k, and full-size property checks."""
+
+import numpy as np
+import pytest
+
+import kaminpar_amd as ka
+from helpers import oracle_refine
+
+pytestmark = pytest.mark.gpu
+
+
+def _require_gpu():
+    import ctypes
+
+    try:
+        eng_probe = ka._lib.kmp_lp_create
+    except AttributeError:
+        pytest.fail("native library missing")
+    return True
+
+
+@pytest.mark.parametrize(
+    "scale,ef,k,seed",
+    [
+        (10, 8, 8, 1),
+        (12, 8, 16, 2),
+        (12, 8, 256, 3),
+        (14, 8, 16, 4),
+        (14, 8, 2, 1),
+    ],
+)
+def test_refine_parity_rmat(oracle, scale, ef, k, seed):
+    g = ka.Graph.rmat(scale, ef, seed=7)
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+
+    eng = ka.LpEngine(g)
+    cut, part, stats = eng.refine(k, mbw, part0, seed=seed, iters=5)
+
+    ocut, opart, ostats = oracle_refine(oracle, g, k, mbw, part0, seed=seed, iters=5)
+    assert cut == ocut, f"cut mismatch gpu={cut} oracle={ocut}"
+    assert (part == opart).all(), f"{(part != opart).sum()} labels differ"
+    assert stats.arcs_scanned == ostats[0]
+    assert stats.moves == ostats[1]
+    # block weights identical (recomputed)
+    assert (np.bincount(part, minlength=k) == np.bincount(opart, minlength=k)).all()
+
+
+def test_refine_parity_rgg(oracle):
+    g = ka.Graph.rgg2d(1 << 14, 16.0, seed=3)
+    k = 64
+    part0 = ka.random_partition(g.n, k, seed=9)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.refine(k, mbw, part0, seed=11, iters=5)
+    ocut, opart, _ = oracle_refine(oracle, g, k, mbw, part0, seed=11, iters=5)
+    assert cut == ocut and (part == opart).all()
+
+
+def test_refine_parity_star_graph(oracle):
+    """Star + ring: exercises the L kernel (hub degree >> 2048)."""
+    n = 100_000
+    hub_edges = [(0, v) for v in range(1, n)]
+    ring_edges = [(v, v % (n - 1) + 1) for v in range(1, n)]
+    arcs = set()
+    for u, v in hub_edges + ring_edges:
+        if u != v:
+            arcs.add((u, v))
+            arcs.add((v, u))
+    arcs = sorted(arcs)
+    xadj = np.zeros(n + 1, np.uint32)
+    for u, v in arcs:
+        xadj[u + 1] += 1
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = np.array([v for _, v in arcs], dtype=np.uint32)
+    g = ka.Graph.from_csr(xadj, adjncy)
+
+    k = 8
+    part0 = ka.random_partition(n, k, seed=1)
+    mbw = np.full(k, g.max_block_weight(k, 0.10), dtype=np.int64)
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.refine(k, mbw, part0, seed=2, iters=5)
+    ocut, opart, _ = oracle_refine(oracle, g, k, mbw, part0, seed=2, iters=5)
+    assert cut == ocut and (part == opart).all()
+
+
+def test_refine_parity_weighted(oracle):
+    rng = np.random.default_rng(0)
+    n = 4096
+    src = rng.integers(0, n, 30000)
+    dst = rng.integers(0, n, 30000)
+    mask = src != dst
+    pairs = np.unique(
+        np.stack([np.concatenate([src[mask], dst[mask]]),
+                  np.concatenate([dst[mask], src[mask]])], 1), axis=0)
+    pairs = pairs[np.lexsort((pairs[:, 1], pairs[:, 0]))]
+    xadj = np.zeros(n + 1, np.uint32)
+    np.add.at(xadj, pairs[:, 0] + 1, 1)
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = pairs[:, 1].astype(np.uint32)
+    vwgt = rng.integers(1, 9, n).astype(np.int32)
+    wkey = (np.minimum(pairs[:, 0], pairs[:, 1]) * 31
+            + np.maximum(pairs[:, 0], pairs[:, 1])) % 7 + 1
+    adjwgt = wkey.astype(np.int32)
+
+    g = ka.Graph.from_csr(xadj, adjncy, vwgt=vwgt, adjwgt=adjwgt)
+    k = 16
+    part0 = ka.random_partition(n, k, seed=2)
+    total_w = int(vwgt.sum())
+    mbw = np.full(k, int(np.ceil(total_w / k) * 1.05), dtype=np.int64)
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.refine(k, mbw, part0, seed=3, iters=5)
+    ocut, opart, _ = oracle_refine(
+        oracle, g, k, mbw, part0, seed=3, iters=5, vwgt=vwgt, adjwgt=adjwgt
+    )
+    assert cut == ocut and (part == opart).all()
+
+
+def test_refine_deterministic_on_device():
+    g = ka.Graph.rmat(14, 8, seed=7)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    eng = ka.LpEngine(g)
+    cut1, p1, _ = eng.refine(k, mbw, part0, seed=9)
+    cut2, p2, _ = eng.refine(k, mbw, part0, seed=9)
+    assert cut1 == cut2 and (p1 == p2).all()
+
+
+def test_refine_fullsize_properties():
+    """Size-independent properties at a larger size (oracle too slow there is
+    fine -- the full-size run is checked via invariants, the small sizes via
+    bit-parity)."""
+    g = ka.Graph.rmat(18, 8, seed=42)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    cut0 = g.edge_cut(part0)
+    eng = ka.LpEngine(g)
+    cut, part, stats = eng.refine(k, mbw, part0, seed=1, iters=5)
+    assert cut == g.edge_cut(part)  # device cut == host recomputed cut
+    assert cut < cut0
+    assert np.bincount(part, minlength=k).max() <= mbw[0]
+    assert stats.arcs_scanned > 0 and stats.moves > 0

@@ -1,0 +1,132 @@
+"""CPU tests of the oracle's own invariants (the properties the reference's
+end-to-end test pins: determinism under a seed, cut consistency, weight caps;
+shm_endtoend_test.cc:142-247)."""
+
+import numpy as np
+import pytest
+
+import kaminpar_amd as ka
+from helpers import oracle_cluster, oracle_refine
+
+
+@pytest.fixture(scope="module")
+def graphs():
+    return {
+        "rmat12": ka.Graph.rmat(12, 8, seed=7),
+        "rgg4k": ka.Graph.rgg2d(4096, 16.0, seed=3),
+    }
+
+
+@pytest.mark.parametrize("name,k", [("rmat12", 8), ("rmat12", 64), ("rgg4k", 4)])
+def test_refine_invariants(oracle, graphs, name, k):
+    g = graphs[name]
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    cut0 = g.edge_cut(part0)
+
+    cut, part, stats = oracle_refine(oracle, g, k, mbw, part0, seed=1, iters=5)
+    # reported cut equals independently recomputed cut
+    assert cut == g.edge_cut(part)
+    # refinement does not worsen the cut
+    assert cut <= cut0
+    # hard balance constraint (move_block_weight semantics)
+    bw = np.bincount(part, minlength=k)
+    assert bw.max() <= mbw[0]
+    # all labels valid
+    assert part.max() < k
+
+
+def test_refine_deterministic(oracle, graphs):
+    g = graphs["rmat12"]
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    cut1, p1, _ = oracle_refine(oracle, g, k, mbw, part0, seed=9)
+    cut2, p2, _ = oracle_refine(oracle, g, k, mbw, part0, seed=9)
+    assert cut1 == cut2 and (p1 == p2).all()
+    # different seeds explore different schedules (shm_endtoend_test.cc:219)
+    cut3, p3, _ = oracle_refine(oracle, g, k, mbw, part0, seed=10)
+    assert (p1 != p3).any()
+
+
+def test_cluster_invariants(oracle, graphs):
+    g = graphs["rmat12"]
+    max_w = 32
+    nc, clus, stats = oracle_cluster(oracle, g, max_w, seed=1)
+    # every cluster respects the weight cap (unit weights -> size cap)
+    sizes = np.bincount(clus, minlength=g.n)
+    assert sizes.max() <= max_w
+    assert nc == (sizes > 0).sum()
+    # clustering shrinks the graph substantially
+    assert nc < g.n // 2
+    # determinism
+    nc2, clus2, _ = oracle_cluster(oracle, g, max_w, seed=1)
+    assert nc2 == nc and (clus == clus2).all()
+
+
+def test_cluster_isolated_nodes(oracle):
+    # graph with isolated vertices: pairs of isolated nodes get matched
+    xadj = np.array([0, 1, 2, 2, 2, 2, 2], dtype=np.uint32)  # 0-1 edge; 2..5 isolated
+    adjncy = np.array([1, 0], dtype=np.uint32)
+    g = ka.Graph.from_csr(xadj, adjncy)
+    nc, clus, _ = oracle_cluster(oracle, g, 4, seed=1)
+    sizes = np.bincount(clus, minlength=g.n)
+    assert sizes.max() <= 4
+    # isolated nodes were matched pairwise (match semantics: <= ceil(4/2)
+    # clusters among the isolated nodes)
+    iso_clusters = len(set(clus[2:].tolist()))
+    assert iso_clusters <= 2
+
+
+def test_weighted_refine(oracle):
+    # node + edge weights exercised
+    rng = np.random.default_rng(0)
+    n = 512
+    src = rng.integers(0, n, 4000)
+    dst = rng.integers(0, n, 4000)
+    mask = src != dst
+    pairs = np.unique(
+        np.stack([np.concatenate([src[mask], dst[mask]]),
+                  np.concatenate([dst[mask], src[mask]])], 1), axis=0)
+    pairs = pairs[np.lexsort((pairs[:, 1], pairs[:, 0]))]
+    xadj = np.zeros(n + 1, np.uint32)
+    np.add.at(xadj, pairs[:, 0] + 1, 1)
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = pairs[:, 1].astype(np.uint32)
+    vwgt = rng.integers(1, 5, n).astype(np.int32)
+    # symmetric edge weights: w(u,v) = f(min,max)
+    wkey = (np.minimum(pairs[:, 0], pairs[:, 1]) * 31 + np.maximum(pairs[:, 0], pairs[:, 1])) % 7 + 1
+    adjwgt = wkey.astype(np.int32)
+
+    g = ka.Graph.from_csr(xadj, adjncy, vwgt=vwgt, adjwgt=adjwgt)
+    k = 8
+    part0 = ka.random_partition(n, k, seed=2)
+    total_w = int(vwgt.sum())
+    mbw = np.full(k, int(np.ceil(total_w / k) * 1.10), dtype=np.int64)
+    cut0 = g.edge_cut(part0)
+    cut, part, _ = oracle_refine(oracle, g, k, mbw, part0, vwgt=vwgt, adjwgt=adjwgt)
+    assert cut == g.edge_cut(part)
+    assert cut <= cut0
+    bw = np.bincount(part, minlength=k, weights=vwgt.astype(np.float64))
+    assert bw.max() <= mbw[0]
+
+
+def test_perm_cross_impl(oracle):
+    """Oracle Feistel permutation == product-side (lp_common.h) permutation."""
+    import ctypes
+
+    from helpers import u32p
+
+    prod = ka._lib
+    prod.kmp_perm.argtypes = [ctypes.c_uint32, ctypes.c_uint64, ctypes.c_int,
+                              ctypes.POINTER(ctypes.c_uint32)]
+    for n in [5, 64, 1000, 65536, 1 << 20]:
+        for seed, it in [(1, 0), (42, 3)]:
+            a = np.zeros(n, dtype=np.uint32)
+            b = np.zeros(n, dtype=np.uint32)
+            oracle.kmp_oracle_perm(ctypes.c_uint32(n), ctypes.c_uint64(seed),
+                                   ctypes.c_int(it), u32p(a))
+            prod.kmp_perm(n, seed, it, u32p(b))
+            assert (a == b).all(), (n, seed, it)
+            # is a permutation
+            assert len(np.unique(a)) == n and a.max() == n - 1
