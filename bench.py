@@ -57,9 +57,8 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     n_gpus = max(args.gpus, world)
 
-    import torch
-
     if world > 1:
+        import torch
         import torch.distributed as dist
 
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
@@ -96,11 +95,14 @@ def main():
         return cut, stats
 
     def barrier_sync():
+        # single-GPU: eng.refine device-syncs internally; multi-GPU adds the
+        # rank barrier + torch-stream sync
         if world > 1:
+            import torch
             import torch.distributed as dist
 
             dist.barrier()
-        torch.cuda.synchronize()
+            torch.cuda.synchronize()
 
     # warmup
     for _ in range(args.warmup):
@@ -125,6 +127,7 @@ def main():
     # per-rank arcs: in sharded mode each rank scans its slice; whole-job arcs
     # = sum over ranks (phase A split); reduce.
     if world > 1:
+        import torch
         import torch.distributed as dist
 
         t = torch.tensor([total_arcs, phase_a_ns], dtype=torch.int64, device=device)
