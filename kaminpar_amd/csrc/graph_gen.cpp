@@ -434,12 +434,15 @@ i64 kmp_edge_cut_host(const kmp_graph_t *g, const u32 *labels) {
 // Expose the product-side Feistel permutation (lp_common.h, the same code
 // the GPU kernels execute) for cross-checking against the oracle's
 // independent restatement.
+// out must have kmp::pos_count(n) entries; values >= n mark skipped tail
+// positions of the last 64-vertex unit.
 void kmp_perm(u32 n, u64 seed, int iter, u32 *out) {
-  const kmp::FeistelPerm perm(n, kmp::iter_seed_of(seed, iter));
+  const kmp::BlockPerm perm(n, kmp::iter_seed_of(seed, iter));
+  const u32 P = kmp::pos_count(n);
 #ifdef _OPENMP
 #pragma omp parallel for schedule(static)
 #endif
-  for (long long p = 0; p < static_cast<long long>(n); ++p) {
+  for (long long p = 0; p < static_cast<long long>(P); ++p) {
     out[p] = perm(static_cast<u32>(p));
   }
 }

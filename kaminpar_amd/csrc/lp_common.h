@@ -83,6 +83,34 @@ struct FeistelPerm {
   }
 };
 
-KMP_HD inline u32 chunk_size_for(u32 n) { return (n + kNumChunks - 1) / kNumChunks; }
+// Block permutation: vertices are grouped into units of 64 consecutive ids
+// (the reference's own randomization granularity, kPermutationSize = 64,
+// label_propagation.h:52) and the UNITS are permuted by the Feistel network.
+// Consecutive ids inside a unit keep CSR reads coalesced on the GPU.
+// Position space: [0, num_blocks(n) * 64); a position may map to u >= n
+// (tail of the last unit) and is then skipped.
+constexpr u32 kUnit = 64;
+
+KMP_HD inline u32 num_units(u32 n) { return (n + kUnit - 1) / kUnit; }
+
+// positions per chunk (multiple of kUnit; kNumChunks chunks cover all units)
+KMP_HD inline u32 chunk_size_for(u32 n) {
+  const u32 nu = num_units(n);
+  return ((nu + kNumChunks - 1) / kNumChunks) * kUnit;
+}
+
+KMP_HD inline u32 pos_count(u32 n) { return num_units(n) * kUnit; }
+
+struct BlockPerm {
+  FeistelPerm fp;
+  u32 n;
+
+  KMP_HD BlockPerm(u32 n_, u64 seed) : fp(num_units(n_), seed), n(n_) {}
+
+  // maps position -> vertex id; result >= n means "no vertex" (skip)
+  KMP_HD inline u32 operator()(u32 p) const {
+    return fp(p / kUnit) * kUnit + (p % kUnit);
+  }
+};
 
 } // namespace kmp

@@ -8,12 +8,16 @@
 //   select), with the commit fixpoint mirroring
 //   kaminpar-dist/refinement/lp/lp_refiner.cc:296-333 (rollback protocol).
 //
-// Kernel design (CDNA4): this is an irregular integer gather workload; the
-// roofline bound is HBM bandwidth (8 B per directed arc: 4 B adjncy + 4 B
-// labels gather), not MFMA. Per-vertex gain maps live in registers (degree
-// <= 16: 16-lane subgroups with shuffle-waterfall dedup) or LDS (dense
-// per-cluster arrays for k <= 2048), wavefronts are 64 wide, argmax uses
-// wave shuffles, and the commit is a sorted segmented-prefix fixpoint.
+// Kernel design (CDNA4): irregular integer gather workload; the roofline
+// bound is HBM bandwidth (8 B per directed arc: 4 B adjncy + 4 B labels
+// gather), not MFMA. The schedule's 64-vertex units keep per-vertex state
+// reads coalesced. Phase A uses NO append atomics: every position owns a
+// 16-byte proposal slot, written by exactly one kernel (S: deg<=16 with
+// 16-lane subgroup shuffle-waterfall gains; M: one wavefront per vertex with
+// dense per-cluster LDS gains for k<=2048; L: one workgroup per high-degree
+// vertex), then rocprim::select compacts valid slots in position order
+// (stable), so the commit's stable 32-bit radix sort by target cluster
+// yields the deterministic (to, rank) admission order.
 
 #include <cstdio>
 #include <cstdlib>
@@ -32,7 +36,7 @@ using kmp::i32;
 using kmp::i64;
 using kmp::u32;
 using kmp::u64;
-using kmp::FeistelPerm;
+using kmp::BlockPerm;
 using kmp::iter_seed_of;
 using kmp::tie_hash;
 
@@ -48,63 +52,22 @@ using kmp::tie_hash;
 
 namespace {
 
-constexpr u32 kSmallDeg = 16;    // S bucket: <= 16 neighbours, 16 lanes/vertex
-constexpr u32 kMidDeg = 2048;    // M bucket: one wavefront per vertex
+constexpr u32 kSmallDeg = 16;    // S path: <= 16 neighbours, 16 lanes/vertex
+constexpr u32 kMidDeg = 2048;    // M path: one wavefront per vertex
 constexpr u32 kMaxDenseK = 2048; // dense per-cluster LDS gains limit (refine)
 constexpr u32 kWave = 64;
+constexpr u32 kInvalid = 0xFFFFFFFFu;
 
 struct Prop { // 16-byte proposal record (ABI: uint32x4)
   u32 u;
-  u32 to;
-  u32 rank;
-  u32 w; // node weight bits (i32 >= 0)
+  u32 to;   // kInvalid marks an empty slot (pre-compaction)
+  u32 rank; // position - chunk_base (admission order within the chunk)
+  u32 w;    // node weight bits (i32 >= 0)
 };
 
-// --------------------------------------------------------------- binning
-__global__ void k_bin_impl(
-    u32 pos_lo,
-    u32 pos_hi,
-    u32 n,
-    u64 iter_seed,
-    u32 max_degree,
-    const u32 *__restrict__ xadj,
-    const uint8_t *__restrict__ active,
-    u64 *__restrict__ bucket_s,
-    u64 *__restrict__ bucket_m,
-    u64 *__restrict__ bucket_l,
-    u32 *__restrict__ counts, // [0]=S [1]=M [2]=L [3]=processed
-    u32 *__restrict__ processed,
-    unsigned long long *__restrict__ arcs
-) {
-  const u32 tid = blockIdx.x * blockDim.x + threadIdx.x;
-  const u32 p = pos_lo + tid;
-  FeistelPerm perm(n, iter_seed);
-
-  u64 my_deg = 0;
-  if (p < pos_hi) {
-    const u32 u = perm(p);
-    const u32 deg = xadj[u + 1] - xadj[u];
-    if (deg <= max_degree && active[u]) {
-      my_deg = deg;
-      const u64 rec = (static_cast<u64>(p) << 32) | u;
-      if (deg <= kSmallDeg) {
-        bucket_s[atomicAdd(&counts[0], 1u)] = rec;
-      } else if (deg <= kMidDeg) {
-        bucket_m[atomicAdd(&counts[1], 1u)] = rec;
-      } else {
-        bucket_l[atomicAdd(&counts[2], 1u)] = rec;
-      }
-      processed[atomicAdd(&counts[3], 1u)] = u;
-    }
-  }
-  // wave-level reduction of scanned arcs
-  for (int off = 32; off > 0; off >>= 1) {
-    my_deg += __shfl_down(static_cast<unsigned long long>(my_deg), off, kWave);
-  }
-  if ((threadIdx.x & (kWave - 1)) == 0 && my_deg > 0) {
-    atomicAdd(arcs, static_cast<unsigned long long>(my_deg));
-  }
-}
+struct PropValid {
+  __host__ __device__ bool operator()(const Prop &p) const { return p.to != kInvalid; }
+};
 
 // ------------------------------------------------- select helpers (device)
 struct BestState {
@@ -134,14 +97,18 @@ __device__ inline bool accept_refine(
   return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw)) || (c == cur);
 }
 
-// ------------------------------------------------------ S bucket (deg<=16)
-// 16 lanes per vertex, 4 vertices per wave; gains deduped via shuffle
-// waterfall inside the 16-lane subgroup (no LDS, works for any k).
-__global__ void k_lp_small(
-    const u64 *__restrict__ bucket,
-    const u32 *__restrict__ counts, // counts[0] = S count
+// ------------------------------------------------------------ S path
+// Covers EVERY position of the slice: 4 positions per wave, 16 lanes each.
+// Owns the slot for: tail positions (u >= n), inactive or degree-filtered
+// vertices (invalid slot), and active deg <= 16 vertices (computed result).
+// Leaves deg in (16, inf) active slots for the M/L kernels.
+__global__ void k_phase_s(
+    u32 pos_lo,
+    u32 pos_hi,
     u32 chunk_base,
+    u32 n,
     u64 iter_seed,
+    u32 max_degree,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ vwgt,
@@ -149,27 +116,44 @@ __global__ void k_lp_small(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    Prop *__restrict__ props,
-    u32 *__restrict__ prop_count
+    const uint8_t *__restrict__ active,
+    Prop *__restrict__ slots
 ) {
-  const u32 count = counts[0];
   const u32 lane = threadIdx.x & (kWave - 1);
-  const u32 sub = lane >> 4;          // subgroup 0..3 within wave
-  const u32 slot = lane & 15;         // lane within subgroup
+  const u32 sub = lane >> 4;  // subgroup 0..3
+  const u32 slot = lane & 15; // lane within subgroup
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const u32 vid = wave_id * 4 + sub;
-  if (vid >= count) {
+  const u32 p = pos_lo + wave_id * 4 + sub;
+  if (p >= pos_hi) {
     return;
   }
 
-  const u64 rec = bucket[vid];
-  const u32 p = static_cast<u32>(rec >> 32);
-  const u32 u = static_cast<u32>(rec);
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
+  const BlockPerm perm(n, iter_seed);
+  const u32 u = perm(p);
+  const u32 sidx = p - pos_lo;
 
-  // candidate load: lane slot handles edge slot
-  u32 c = 0xFFFFFFFFu;
+  bool emit_invalid = false;
+  u32 row = 0, deg = 0;
+  if (u >= n) {
+    emit_invalid = true;
+  } else {
+    row = xadj[u];
+    deg = xadj[u + 1] - row;
+    if (!active[u] || deg > max_degree) {
+      emit_invalid = true;
+    } else if (deg > kSmallDeg) {
+      return; // M/L owns this slot
+    }
+  }
+  if (emit_invalid) {
+    if (slot == 0) {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
+    return;
+  }
+
+  // candidate load: lane handles one edge
+  u32 c = kInvalid;
   i32 w = 0;
   if (slot < deg) {
     const u32 v = adjncy[row + slot];
@@ -184,12 +168,10 @@ __global__ void k_lp_small(
   for (u32 j = 0; j < 16; ++j) {
     const u32 cj = __shfl(c, base + j, kWave);
     const i32 wj = __shfl(w, base + j, kWave);
-    if (j != slot && cj == c && c != 0xFFFFFFFFu) {
-      if (cj != 0xFFFFFFFFu) {
-        gain += wj;
-        if (j < slot) {
-          owner = false;
-        }
+    if (j != slot && c != kInvalid && cj == c) {
+      gain += wj;
+      if (j < slot) {
+        owner = false;
       }
     }
   }
@@ -199,9 +181,8 @@ __global__ void k_lp_small(
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
 
-  // per-lane candidate key
   BestState best{0, 0, 0, false};
-  if (owner && c != 0xFFFFFFFFu) {
+  if (owner && c != kInvalid) {
     const i64 cw = weights[c];
     const i64 mw = maxw[c];
     if (accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
@@ -215,26 +196,32 @@ __global__ void k_lp_small(
     const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
     const u32 oc = __shfl_down(best.c, off, kWave);
     const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
-    // only combine within the same 16-lane subgroup
     if ((slot + off) < 16 && ohave && key_better(og, oh, oc, best)) {
       best = BestState{og, oh, oc, true};
     }
   }
 
-  if (slot == 0 && best.have && best.c != cur) {
-    const u32 idx = atomicAdd(prop_count, 1u);
-    props[idx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  if (slot == 0) {
+    if (best.have && best.c != cur) {
+      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
   }
 }
 
-// -------------------------------------------------- M bucket (wave/vertex)
-// Dense per-cluster LDS gains (k <= kMaxDenseK); one wavefront per vertex.
+// ------------------------------------------------------------ M path
+// One wavefront per position; handles active vertices with
+// kSmallDeg < deg <= kMidDeg (dense per-wave LDS gains, k <= kMaxDenseK).
+// Active deg > kMidDeg vertices are appended to the L list (rare).
 // blockDim.x = 256 (4 waves); dynamic LDS = 4 * k * sizeof(i32).
-__global__ void k_lp_mid(
-    const u64 *__restrict__ bucket,
-    const u32 *__restrict__ counts, // counts[1] = M count
+__global__ void k_phase_m(
+    u32 pos_lo,
+    u32 pos_hi,
     u32 chunk_base,
+    u32 n,
     u64 iter_seed,
+    u32 max_degree,
     u32 k,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -243,30 +230,43 @@ __global__ void k_lp_mid(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    Prop *__restrict__ props,
-    u32 *__restrict__ prop_count
+    const uint8_t *__restrict__ active,
+    Prop *__restrict__ slots,
+    u64 *__restrict__ l_list, // (position << 32) | u
+    u32 *__restrict__ l_count
 ) {
   extern __shared__ i32 lds[];
-  const u32 count = counts[1];
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wave_in_wg = threadIdx.x >> 6;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  if (wave_id >= count) {
+  const u32 p = pos_lo + wave_id;
+  if (p >= pos_hi) {
     return;
   }
-  i32 *gains = lds + wave_in_wg * k;
 
+  const BlockPerm perm(n, iter_seed);
+  const u32 u = perm(p);
+  if (u >= n) {
+    return; // S wrote the slot
+  }
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+  if (deg <= kSmallDeg || deg > max_degree || !active[u]) {
+    return; // S wrote the slot
+  }
+  const u32 sidx = p - pos_lo;
+  if (deg > kMidDeg) {
+    if (lane == 0) {
+      l_list[atomicAdd(l_count, 1u)] = (static_cast<u64>(p) << 32) | u;
+    }
+    return; // L kernel writes the slot
+  }
+
+  i32 *gains = lds + wave_in_wg * k;
   for (u32 c = lane; c < k; c += kWave) {
     gains[c] = 0;
   }
-  __threadfence_block(); // LDS ordering within the wave (lockstep lanes; the
-                         // gains slice is private to this wave)
-
-  const u64 rec = bucket[wave_id];
-  const u32 p = static_cast<u32>(rec >> 32);
-  const u32 u = static_cast<u32>(rec);
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
+  __threadfence_block(); // LDS ordering; gains slice is private to this wave
 
   for (u32 e = lane; e < deg; e += kWave) {
     const u32 v = adjncy[row + e];
@@ -307,17 +307,20 @@ __global__ void k_lp_mid(
     }
   }
 
-  if (lane == 0 && best.have && best.c != cur) {
-    const u32 idx = atomicAdd(prop_count, 1u);
-    props[idx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  if (lane == 0) {
+    if (best.have && best.c != cur) {
+      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
   }
 }
 
-// ---------------------------------------------- L bucket (workgroup/vertex)
-// One 256-thread workgroup per vertex, shared dense LDS gains (k <= 2048).
-__global__ void k_lp_large(
-    const u64 *__restrict__ bucket,
-    const u32 *__restrict__ counts, // counts[2] = L count
+// ------------------------------------------------------------ L path
+// Grid-stride over the L list; one 256-thread workgroup per vertex, shared
+// dense LDS gains (k <= kMaxDenseK).
+__global__ void k_phase_l(
+    u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
     u32 k,
@@ -328,117 +331,118 @@ __global__ void k_lp_large(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    Prop *__restrict__ props,
-    u32 *__restrict__ prop_count
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
-  const u32 count = counts[2];
-  if (blockIdx.x >= count) {
-    return;
-  }
-  i32 *gains = lds;                                       // k i32
-  i64 *red = reinterpret_cast<i64 *>(lds + ((k + 1) & ~1u)); // 4-wave scratch
+  i32 *gains = lds;                                          // k i32 (padded)
+  i64 *red = reinterpret_cast<i64 *>(lds + ((k + 1) & ~1u)); // cross-wave scratch
 
-  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
-    gains[c] = 0;
-  }
-  __syncthreads();
-
-  const u64 rec = bucket[blockIdx.x];
-  const u32 p = static_cast<u32>(rec >> 32);
-  const u32 u = static_cast<u32>(rec);
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
-
-  for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
-    const u32 v = adjncy[row + e];
-    const i32 w = adjwgt ? adjwgt[row + e] : 1;
-    atomicAdd(&gains[labels[v]], w);
-  }
-  __syncthreads();
-
-  const u32 cur = labels[u];
-  const i32 u_w = vwgt ? vwgt[u] : 1;
-  const i64 cur_w = weights[cur];
-  const i64 cur_maxw = maxw[cur];
-
-  BestState best{0, 0, 0, false};
-  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
-    const i32 g = gains[c];
-    if (g <= 0) {
-      continue;
+  const u32 count = *l_count;
+  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+      gains[c] = 0;
     }
-    const i64 cw = weights[c];
-    const i64 mw = maxw[c];
-    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
-      continue;
+    __syncthreads();
+
+    const u64 rec = l_list[vid];
+    const u32 p = static_cast<u32>(rec >> 32);
+    const u32 u = static_cast<u32>(rec);
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+
+    for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
+      const u32 v = adjncy[row + e];
+      const i32 w = adjwgt ? adjwgt[row + e] : 1;
+      atomicAdd(&gains[labels[v]], w);
     }
-    const u64 h = tie_hash(iter_seed, u, c);
-    if (key_better(g, h, c, best)) {
-      best = BestState{g, h, c, true};
-    }
-  }
-  const u32 lane = threadIdx.x & (kWave - 1);
-  for (int off = 32; off > 0; off >>= 1) {
-    const i32 og = __shfl_down(best.gain, off, kWave);
-    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
-    const u32 oc = __shfl_down(best.c, off, kWave);
-    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
-    if (ohave && key_better(og, oh, oc, best)) {
-      best = BestState{og, oh, oc, true};
-    }
-  }
-  // cross-wave reduce via LDS (4 waves)
-  const u32 wave_in_wg = threadIdx.x >> 6;
-  if (lane == 0) {
-    red[wave_in_wg * 2] = (static_cast<i64>(best.gain) << 1) | (best.have ? 1 : 0);
-    red[wave_in_wg * 2 + 1] = static_cast<i64>(best.h);
-    reinterpret_cast<u32 *>(red + 8)[wave_in_wg] = best.c;
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    BestState total{0, 0, 0, false};
-    const u32 waves = blockDim.x >> 6;
-    for (u32 wv = 0; wv < waves; ++wv) {
-      const i64 packed = red[wv * 2];
-      if (packed & 1) {
-        const i32 g = static_cast<i32>(packed >> 1);
-        const u64 h = static_cast<u64>(red[wv * 2 + 1]);
-        const u32 c = reinterpret_cast<u32 *>(red + 8)[wv];
-        if (key_better(g, h, c, total)) {
-          total = BestState{g, h, c, true};
-        }
+    __syncthreads();
+
+    const u32 cur = labels[u];
+    const i32 u_w = vwgt ? vwgt[u] : 1;
+    const i64 cur_w = weights[cur];
+    const i64 cur_maxw = maxw[cur];
+
+    BestState best{0, 0, 0, false};
+    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+      const i32 g = gains[c];
+      if (g <= 0) {
+        continue;
+      }
+      const i64 cw = weights[c];
+      const i64 mw = maxw[c];
+      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+        continue;
+      }
+      const u64 h = tie_hash(iter_seed, u, c);
+      if (key_better(g, h, c, best)) {
+        best = BestState{g, h, c, true};
       }
     }
-    if (total.have && total.c != cur) {
-      const u32 idx = atomicAdd(prop_count, 1u);
-      props[idx] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+    const u32 lane = threadIdx.x & (kWave - 1);
+    for (int off = 32; off > 0; off >>= 1) {
+      const i32 og = __shfl_down(best.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+      const u32 oc = __shfl_down(best.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+      if (ohave && key_better(og, oh, oc, best)) {
+        best = BestState{og, oh, oc, true};
+      }
     }
+    const u32 wave_in_wg = threadIdx.x >> 6;
+    if (lane == 0) {
+      red[wave_in_wg * 2] = (static_cast<i64>(best.gain) << 1) | (best.have ? 1 : 0);
+      red[wave_in_wg * 2 + 1] = static_cast<i64>(best.h);
+      reinterpret_cast<u32 *>(red + 8)[wave_in_wg] = best.c;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      BestState total{0, 0, 0, false};
+      const u32 waves = blockDim.x >> 6;
+      for (u32 wv = 0; wv < waves; ++wv) {
+        const i64 packed = red[wv * 2];
+        if (packed & 1) {
+          const i32 g = static_cast<i32>(packed >> 1);
+          const u64 h = static_cast<u64>(red[wv * 2 + 1]);
+          const u32 c = reinterpret_cast<u32 *>(red + 8)[wv];
+          if (key_better(g, h, c, total)) {
+            total = BestState{g, h, c, true};
+          }
+        }
+      }
+      if (total.have && total.c != cur) {
+        slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
+      }
+    }
+    __syncthreads(); // gains/red reuse across grid-stride iterations
   }
 }
 
 // -------------------------------------------------------------- commit
-__global__ void k_extract_sorted(
-    const u32 *__restrict__ order, // sorted proposal indices
-    const Prop *__restrict__ props,
-    u32 count,
-    u32 *__restrict__ sto,
-    i64 *__restrict__ sw
+__global__ void k_make_keys(
+    const Prop *__restrict__ props, u32 count, u32 *__restrict__ keys, u32 *__restrict__ vals
 ) {
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < count) {
-    const Prop pr = props[order[i]];
-    sto[i] = pr.to;
-    sw[i] = static_cast<i64>(static_cast<i32>(pr.w));
+    keys[i] = props[i].to;
+    vals[i] = i;
+  }
+}
+
+__global__ void k_extract_w(
+    const u32 *__restrict__ order, const Prop *__restrict__ props, u32 count, i64 *__restrict__ sw
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    sw[i] = static_cast<i64>(static_cast<i32>(props[order[i]].w));
   }
 }
 
 __global__ void k_seg_bounds(
-    const u32 *__restrict__ sto,
-    u32 count,
-    u32 *__restrict__ seg_begin,
-    u32 *__restrict__ seg_end,
-    u32 *__restrict__ prefix_len
+    const u32 *__restrict__ sto, u32 count, u32 *__restrict__ seg_begin, u32 *__restrict__ seg_end
 ) {
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= count) {
@@ -451,7 +455,6 @@ __global__ void k_seg_bounds(
   if (i == count - 1 || sto[i + 1] != to) {
     seg_end[to] = i + 1;
   }
-  (void)prefix_len;
 }
 
 __global__ void k_seg_len(
@@ -508,7 +511,6 @@ __global__ void k_cutoff(
   }
   const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(dep[c]);
   const u32 old_len = prefix_len[c];
-  // binary search: largest t <= old_len with pw[b + t - 1] <= capacity
   u32 lo = 0, hi = old_len;
   while (lo < hi) {
     const u32 mid = (lo + hi + 1) >> 1;
@@ -556,44 +558,89 @@ __global__ void k_apply(
     const u32 *__restrict__ seg_begin,
     const u32 *__restrict__ prefix_len,
     u32 *__restrict__ labels,
-    u32 *__restrict__ admitted_idx, // compact list of admitted proposal idxs
-    u32 *__restrict__ admitted_count
+    u32 *__restrict__ admitted_flags, // per sorted index
+    unsigned long long *__restrict__ moves
 ) {
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= count) {
-    return;
+  u64 local = 0;
+  if (i < count) {
+    const u32 to = sto[i];
+    const bool admitted = (i - seg_begin[to]) < prefix_len[to];
+    admitted_flags[i] = admitted ? 1u : 0u;
+    if (admitted) {
+      labels[props[order[i]].u] = to;
+      local = 1;
+    }
   }
-  const u32 to = sto[i];
-  if (i - seg_begin[to] < prefix_len[to]) {
-    const u32 pi = order[i];
-    labels[props[pi].u] = to;
-    admitted_idx[atomicAdd(admitted_count, 1u)] = pi;
+  // wave-aggregated move count
+  for (int off = 32; off > 0; off >>= 1) {
+    local += __shfl_down(static_cast<unsigned long long>(local), off, kWave);
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0 && local) {
+    atomicAdd(moves, static_cast<unsigned long long>(local));
   }
 }
 
+// Clear active flags for processed vertices of the WHOLE chunk (identical on
+// every rank) and count scanned arcs (WG-aggregated).
 __global__ void k_clear_active(
-    const u32 *__restrict__ processed, const u32 *__restrict__ counts, uint8_t *__restrict__ active
+    u32 chunk_lo,
+    u32 chunk_hi,
+    u32 n,
+    u64 iter_seed,
+    u32 max_degree,
+    const u32 *__restrict__ xadj,
+    uint8_t *__restrict__ active,
+    unsigned long long *__restrict__ arcs
 ) {
-  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < counts[3]) {
-    active[processed[i]] = 0;
+  __shared__ unsigned long long wg_sum[4];
+  const u32 tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const u32 p = chunk_lo + tid;
+  const BlockPerm perm(n, iter_seed);
+
+  u64 my_deg = 0;
+  if (p < chunk_hi) {
+    const u32 u = perm(p);
+    if (u < n) {
+      const u32 deg = xadj[u + 1] - xadj[u];
+      if (deg <= max_degree && active[u]) {
+        my_deg = deg;
+        active[u] = 0;
+      }
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    my_deg += __shfl_down(static_cast<unsigned long long>(my_deg), off, kWave);
+  }
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wave_in_wg = threadIdx.x >> 6;
+  if (lane == 0) {
+    wg_sum[wave_in_wg] = my_deg;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned long long total = wg_sum[0] + wg_sum[1] + wg_sum[2] + wg_sum[3];
+    if (total) {
+      atomicAdd(arcs, total);
+    }
   }
 }
 
 __global__ void k_activate(
-    const u32 *__restrict__ admitted_idx,
-    const u32 *__restrict__ admitted_count,
+    const u32 *__restrict__ order,
+    const u32 *__restrict__ admitted_flags,
     const Prop *__restrict__ props,
+    u32 count,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     uint8_t *__restrict__ active
 ) {
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const u32 lane = threadIdx.x & (kWave - 1);
-  if (wave_id >= *admitted_count) {
+  if (wave_id >= count || !admitted_flags[wave_id]) {
     return;
   }
-  const u32 u = props[admitted_idx[wave_id]].u;
+  const u32 u = props[order[wave_id]].u;
   const u32 row = xadj[u];
   const u32 deg = xadj[u + 1] - row;
   for (u32 e = lane; e < deg; e += kWave) {
@@ -601,23 +648,23 @@ __global__ void k_activate(
   }
 }
 
+// Reset the touched per-cluster segment entries (reads the UNSORTED props'
+// target fields directly; zeroing per cluster is idempotent).
 __global__ void k_reset_segs(
-    const u32 *__restrict__ sto,
+    const Prop *__restrict__ props,
     u32 count,
     u32 *__restrict__ seg_begin,
     u32 *__restrict__ seg_end,
-    u32 *__restrict__ prefix_len,
-    unsigned long long *__restrict__ dep
+    u32 *__restrict__ prefix_len
 ) {
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= count) {
     return;
   }
-  const u32 to = sto[i];
+  const u32 to = props[i].to;
   seg_begin[to] = 0;
   seg_end[to] = 0;
   prefix_len[to] = 0;
-  dep[to] = 0;
 }
 
 __global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep) {
@@ -626,9 +673,6 @@ __global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep
     dep[c] = 0;
   }
 }
-
-// dep array must be reset per fixpoint round for ALL clusters that can
-// appear as sources; for refinement k is small so a full reset is cheap.
 
 __global__ void k_init_weights(
     u32 n,
@@ -639,16 +683,6 @@ __global__ void k_init_weights(
   const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
   if (u < n) {
     atomicAdd(&weights[labels[u]], static_cast<unsigned long long>(vwgt ? vwgt[u] : 1));
-  }
-}
-
-__global__ void k_make_keys(
-    const Prop *__restrict__ props, u32 count, u64 *__restrict__ keys, u32 *__restrict__ vals
-) {
-  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < count) {
-    keys[i] = (static_cast<u64>(props[i].to) << 32) | props[i].rank;
-    vals[i] = i;
   }
 }
 
@@ -691,7 +725,8 @@ struct kmp_lp_t {
   u64 m = 0;
   u32 k = 0;
   u64 seed = 1;
-  u32 C = 0; // chunk size
+  u32 C = 0; // positions per chunk
+  u32 P = 0; // total positions (pos_count)
   bool has_vwgt = false, has_adjwgt = false;
 
   // device graph
@@ -707,19 +742,21 @@ struct kmp_lp_t {
   uint8_t *d_active = nullptr;
 
   // phase buffers
-  u64 *d_bucket_s = nullptr, *d_bucket_m = nullptr, *d_bucket_l = nullptr;
-  u32 *d_counts = nullptr;    // 4 u32
-  u32 *d_processed = nullptr; // C
-  Prop *d_props = nullptr;    // C
+  Prop *d_slots = nullptr; // C
+  Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
+  u64 *d_l_list = nullptr; // C
+  u32 *d_l_count = nullptr;
   u32 *d_prop_count = nullptr;
   unsigned long long *d_arcs = nullptr;
+  unsigned long long *d_moves = nullptr;
 
   // commit buffers
-  u64 *d_sort_keys[2] = {nullptr, nullptr};
+  u32 *d_sort_keys[2] = {nullptr, nullptr};
   u32 *d_sort_vals[2] = {nullptr, nullptr};
   void *d_sort_temp = nullptr;
   size_t sort_temp_bytes = 0;
-  u32 *d_sto = nullptr;
+  void *d_select_temp = nullptr;
+  size_t select_temp_bytes = 0;
   i64 *d_sw = nullptr;
   i64 *d_pw = nullptr;
   void *d_scan_temp = nullptr;
@@ -727,32 +764,32 @@ struct kmp_lp_t {
   u32 *d_seg_begin = nullptr, *d_seg_end = nullptr, *d_prefix_len = nullptr;
   unsigned long long *d_dep = nullptr;
   int *d_changed = nullptr;
-  u32 *d_admitted_idx = nullptr;
-  u32 *d_admitted_count = nullptr;
+  u32 *d_admitted_flags = nullptr;
   unsigned long long *d_cut = nullptr;
 
   // pinned host mirrors
-  u32 *h_count = nullptr; // pinned: prop count / admitted count
+  u32 *h_count = nullptr;
   int *h_changed = nullptr;
+  unsigned long long *h_moves = nullptr; // [0]=before [1]=after
 
   hipStream_t stream = nullptr;
 
   // run bookkeeping
-  u64 arcs_scanned = 0;
-  u64 moves = 0;
   double phase_a_ms = 0.0;
   std::vector<hipEvent_t> ev_pool;
   size_t ev_used = 0;
 
-  hipEvent_t ev_begin() {
+  void ev_pair(hipEvent_t &a, hipEvent_t &b) {
     if (ev_used + 2 > ev_pool.size()) {
-      hipEvent_t a, b;
-      HIP_CHECK(hipEventCreate(&a));
-      HIP_CHECK(hipEventCreate(&b));
-      ev_pool.push_back(a);
-      ev_pool.push_back(b);
+      hipEvent_t x, y;
+      HIP_CHECK(hipEventCreate(&x));
+      HIP_CHECK(hipEventCreate(&y));
+      ev_pool.push_back(x);
+      ev_pool.push_back(y);
     }
-    return ev_pool[ev_used];
+    a = ev_pool[ev_used];
+    b = ev_pool[ev_used + 1];
+    ev_used += 2;
   }
 };
 
@@ -761,7 +798,6 @@ namespace {
 u32 ceil_div(u64 a, u64 b) { return static_cast<u32>((a + b - 1) / b); }
 
 void engine_alloc_k_buffers(kmp_lp_t *e, u32 k_or_n) {
-  // commit per-cluster arrays sized to k (refine) / n (cluster)
   HIP_CHECK(hipMalloc(&e->d_seg_begin, sizeof(u32) * k_or_n));
   HIP_CHECK(hipMalloc(&e->d_seg_end, sizeof(u32) * k_or_n));
   HIP_CHECK(hipMalloc(&e->d_prefix_len, sizeof(u32) * k_or_n));
@@ -806,6 +842,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   e->n = kmp_graph_n(g);
   e->m = kmp_graph_m(g);
   e->C = kmp::chunk_size_for(e->n);
+  e->P = kmp::pos_count(e->n);
   e->has_vwgt = kmp_graph_vwgt(g) != nullptr;
   e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
   HIP_CHECK(hipStreamCreate(&e->stream));
@@ -829,40 +866,41 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   HIP_CHECK(hipMalloc(&e->d_active, e->n));
 
   const u32 C = e->C;
-  HIP_CHECK(hipMalloc(&e->d_bucket_s, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_bucket_m, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_bucket_l, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_counts, sizeof(u32) * 4));
-  HIP_CHECK(hipMalloc(&e->d_processed, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
+  HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_l_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_arcs, sizeof(unsigned long long)));
+  HIP_CHECK(hipMalloc(&e->d_moves, sizeof(unsigned long long)));
 
-  HIP_CHECK(hipMalloc(&e->d_sort_keys[0], sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_sort_keys[1], sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_sort_keys[0], sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_sort_keys[1], sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_sort_vals[0], sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_sort_vals[1], sizeof(u32) * C));
-  HIP_CHECK(hipMalloc(&e->d_sto, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_sw, sizeof(i64) * C));
   HIP_CHECK(hipMalloc(&e->d_pw, sizeof(i64) * C));
   HIP_CHECK(hipMalloc(&e->d_changed, sizeof(int)));
-  HIP_CHECK(hipMalloc(&e->d_admitted_idx, sizeof(u32) * C));
-  HIP_CHECK(hipMalloc(&e->d_admitted_count, sizeof(u32)));
+  HIP_CHECK(hipMalloc(&e->d_admitted_flags, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_cut, sizeof(unsigned long long)));
 
-  // rocprim temp sizes (max problem size C)
-  rocprim::double_buffer<u64> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
+  rocprim::double_buffer<u32> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
   rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
-  HIP_CHECK(rocprim::radix_sort_pairs(nullptr, e->sort_temp_bytes, keys, vals, C));
+  HIP_CHECK(rocprim::radix_sort_pairs(nullptr, e->sort_temp_bytes, keys, vals, C, 0, 32));
   HIP_CHECK(hipMalloc(&e->d_sort_temp, e->sort_temp_bytes));
   HIP_CHECK(rocprim::inclusive_scan_by_key(
-      nullptr, e->scan_temp_bytes, e->d_sto, e->d_sw, e->d_pw, C, rocprim::plus<i64>(),
+      nullptr, e->scan_temp_bytes, e->d_sort_keys[0], e->d_sw, e->d_pw, C, rocprim::plus<i64>(),
       rocprim::equal_to<u32>()
   ));
   HIP_CHECK(hipMalloc(&e->d_scan_temp, e->scan_temp_bytes));
+  HIP_CHECK(rocprim::select(
+      nullptr, e->select_temp_bytes, e->d_slots, e->d_props, e->d_prop_count, C, PropValid()
+  ));
+  HIP_CHECK(hipMalloc(&e->d_select_temp, e->select_temp_bytes));
 
   HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
   HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
+  HIP_CHECK(hipHostMalloc(&e->h_moves, sizeof(unsigned long long) * 2));
   return e;
 }
 
@@ -877,13 +915,13 @@ void kmp_lp_free(kmp_lp_t *e) {
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
-                  (void *)e->d_bucket_s, (void *)e->d_bucket_m, (void *)e->d_bucket_l,
-                  (void *)e->d_counts, (void *)e->d_processed, (void *)e->d_props,
-                  (void *)e->d_prop_count, (void *)e->d_arcs, (void *)e->d_sort_keys[0],
-                  (void *)e->d_sort_keys[1], (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1],
-                  (void *)e->d_sort_temp, (void *)e->d_sto, (void *)e->d_sw, (void *)e->d_pw,
-                  (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_idx,
-                  (void *)e->d_admitted_count, (void *)e->d_cut}) {
+                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
+                  (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
+                  (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
+                  (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
+                  (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_flags,
+                  (void *)e->d_cut}) {
     if (p) {
       (void)hipFree(p);
     }
@@ -893,6 +931,9 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   if (e->h_changed) {
     (void)hipHostFree(e->h_changed);
+  }
+  if (e->h_moves) {
+    (void)hipHostFree(e->h_moves);
   }
   (void)hipStreamDestroy(e->stream);
   delete e;
@@ -909,8 +950,6 @@ int kmp_lp_refine_begin(
   }
   e->k = k;
   e->seed = seed;
-  e->arcs_scanned = 0;
-  e->moves = 0;
   e->phase_a_ms = 0.0;
   e->ev_used = 0;
 
@@ -929,8 +968,8 @@ int kmp_lp_refine_begin(
   HIP_CHECK(hipMemcpy(e->d_labels, partition, sizeof(u32) * e->n, hipMemcpyHostToDevice));
   HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
 
-  // initial block weights on device
   HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * k, e->stream));
   {
     const u32 threads = 256;
@@ -948,56 +987,55 @@ i64 kmp_lp_phase_a(
 ) {
   const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 chunk_base = chunk * e->C;
-  Prop *props = static_cast<Prop *>(d_out);
-
-  HIP_CHECK(hipMemsetAsync(e->d_counts, 0, sizeof(u32) * 4, e->stream));
-  HIP_CHECK(hipMemsetAsync(e->d_prop_count, 0, sizeof(u32), e->stream));
-
+  Prop *out = static_cast<Prop *>(d_out);
   const u32 span = pos_hi - pos_lo;
   const u32 threads = 256;
+  const u32 max_degree = 0xFFFFFFFFu;
 
-  hipEvent_t ev0 = e->ev_begin();
-  hipEvent_t ev1 = e->ev_pool[e->ev_used + 1];
-  e->ev_used += 2;
+  HIP_CHECK(hipMemsetAsync(e->d_l_count, 0, sizeof(u32), e->stream));
+
+  hipEvent_t ev0, ev1;
+  e->ev_pair(ev0, ev1);
   HIP_CHECK(hipEventRecord(ev0, e->stream));
 
+  // S: 4 positions/wave (default owner of every position's slot)
   hipLaunchKernelGGL(
-      k_bin_impl, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
-      iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_bucket_s, e->d_bucket_m, e->d_bucket_l,
-      e->d_counts, e->d_processed, e->d_arcs
+      k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+      dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->d_xadj,
+      e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
+      e->d_slots
   );
-
-  // S: 4 vertices/wave -> span/4 waves max
+  // M: 1 position/wave
   {
-    const u32 waves = ceil_div(span, 4);
-    hipLaunchKernelGGL(
-        k_lp_small, dim3(ceil_div(waves * kWave, threads)), dim3(threads), 0, e->stream,
-        e->d_bucket_s, e->d_counts, chunk_base, iseed, e->d_xadj, e->d_adjncy, e->d_vwgt,
-        e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, props, e->d_prop_count
-    );
-  }
-  // M: 1 vertex/wave
-  {
-    const u32 waves = span; // upper bound
     const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
     hipLaunchKernelGGL(
-        k_lp_mid, dim3(ceil_div(waves * kWave, threads)), dim3(threads), lds, e->stream,
-        e->d_bucket_m, e->d_counts, chunk_base, iseed, e->k, e->d_xadj, e->d_adjncy, e->d_vwgt,
-        e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, props, e->d_prop_count
+        k_phase_m, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
+        e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
+        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
+        e->d_slots, e->d_l_list, e->d_l_count
     );
   }
-  // L: 1 vertex/workgroup. Grid must cover the worst case; blocks early-exit.
+  // L: grid-stride workgroups over the (rare) high-degree list
   {
     const size_t lds = (static_cast<size_t>((e->k + 1) & ~1u)) * sizeof(i32) + 16 * sizeof(i64);
     hipLaunchKernelGGL(
-        k_lp_large, dim3(4096), dim3(256), lds, e->stream, e->d_bucket_l, e->d_counts, chunk_base,
-        iseed, e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_maxw, props, e->d_prop_count
+        k_phase_l, dim3(2048), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k,
+        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
+        e->d_l_list, e->d_l_count, e->d_slots
     );
+  }
+  // compact valid slots in position order (stable select)
+  {
+    size_t tb = e->select_temp_bytes;
+    HIP_CHECK(rocprim::select(
+        e->d_select_temp, tb, e->d_slots, out, e->d_prop_count, span, PropValid(), e->stream
+    ));
   }
   HIP_CHECK(hipEventRecord(ev1, e->stream));
 
-  HIP_CHECK(hipMemcpyAsync(e->h_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(
+      hipMemcpyAsync(e->h_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream)
+  );
   HIP_CHECK(hipStreamSynchronize(e->stream));
   float ms = 0;
   HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
@@ -1011,103 +1049,101 @@ i64 kmp_lp_phase_a(
 }
 
 i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 count) {
-  (void)iter;
-  (void)chunk;
+  const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 threads = 256;
-  if (count == 0) {
-    // no proposals: still clear the active flags of processed vertices
-    hipLaunchKernelGGL(
-        k_clear_active, dim3(ceil_div(e->C, threads)), dim3(threads), 0, e->stream, e->d_processed,
-        e->d_counts, e->d_active
-    );
-    return 0;
-  }
+  const u32 chunk_lo = chunk * e->C;
+  const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
+
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
+
   const Prop *props = static_cast<const Prop *>(d_props);
-  const u32 grid = ceil_div(count, threads);
+  u32 *order = nullptr;
+  u32 *sto = nullptr;
 
-  // build sort keys (to<<32 | rank) + index values
-  hipLaunchKernelGGL(
-      k_make_keys, dim3(grid), dim3(threads), 0, e->stream, props, count, e->d_sort_keys[0],
-      e->d_sort_vals[0]
-  );
-  rocprim::double_buffer<u64> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
-  rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
-  size_t tb = e->sort_temp_bytes;
-  HIP_CHECK(rocprim::radix_sort_pairs(e->d_sort_temp, tb, keys, vals, count, 0, 64, e->stream));
-  u32 *order = vals.current();
+  if (count > 0) {
+    const u32 grid = ceil_div(count, threads);
+    hipLaunchKernelGGL(
+        k_make_keys, dim3(grid), dim3(threads), 0, e->stream, props, count, e->d_sort_keys[0],
+        e->d_sort_vals[0]
+    );
+    rocprim::double_buffer<u32> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
+    rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
+    size_t tb = e->sort_temp_bytes;
+    HIP_CHECK(rocprim::radix_sort_pairs(e->d_sort_temp, tb, keys, vals, count, 0, 32, e->stream));
+    order = vals.current();
+    sto = keys.current(); // sorted target clusters
 
-  hipLaunchKernelGGL(
-      k_extract_sorted, dim3(grid), dim3(threads), 0, e->stream, order, props, count, e->d_sto,
-      e->d_sw
-  );
-  size_t sb = e->scan_temp_bytes;
-  HIP_CHECK(rocprim::inclusive_scan_by_key(
-      e->d_scan_temp, sb, e->d_sto, e->d_sw, e->d_pw, count, rocprim::plus<i64>(),
-      rocprim::equal_to<u32>(), e->stream
-  ));
-  hipLaunchKernelGGL(
-      k_seg_bounds, dim3(grid), dim3(threads), 0, e->stream, e->d_sto, count, e->d_seg_begin,
-      e->d_seg_end, e->d_prefix_len
-  );
-  const u32 kgrid = ceil_div(e->k, threads);
-  hipLaunchKernelGGL(
-      k_seg_len, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
-      e->d_prefix_len
-  );
+    hipLaunchKernelGGL(
+        k_extract_w, dim3(grid), dim3(threads), 0, e->stream, order, props, count, e->d_sw
+    );
+    size_t sb = e->scan_temp_bytes;
+    HIP_CHECK(rocprim::inclusive_scan_by_key(
+        e->d_scan_temp, sb, sto, e->d_sw, e->d_pw, count, rocprim::plus<i64>(),
+        rocprim::equal_to<u32>(), e->stream
+    ));
+    hipLaunchKernelGGL(
+        k_seg_bounds, dim3(grid), dim3(threads), 0, e->stream, sto, count, e->d_seg_begin,
+        e->d_seg_end
+    );
+    const u32 kgrid = ceil_div(e->k, threads);
+    hipLaunchKernelGGL(
+        k_seg_len, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
+        e->d_prefix_len
+    );
 
-  // fixpoint rounds
-  while (true) {
-    hipLaunchKernelGGL(
-        k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep
-    );
-    hipLaunchKernelGGL(
-        k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, e->d_sto, count,
-        e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
-    );
-    HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
-    hipLaunchKernelGGL(
-        k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
-        e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
-    );
-    HIP_CHECK(
-        hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
-    );
-    HIP_CHECK(hipStreamSynchronize(e->stream));
-    if (!*e->h_changed) {
-      break;
+    // greatest-fixpoint rollback (kaminpar-dist lp_refiner.cc:296-333)
+    while (true) {
+      hipLaunchKernelGGL(k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep);
+      hipLaunchKernelGGL(
+          k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
+          e->d_prefix_len, e->d_labels, e->d_dep
+      );
+      HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
+      hipLaunchKernelGGL(
+          k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
+          e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
+      );
+      HIP_CHECK(
+          hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
+      );
+      HIP_CHECK(hipStreamSynchronize(e->stream));
+      if (!*e->h_changed) {
+        break;
+      }
     }
+
+    hipLaunchKernelGGL(
+        k_weights_update, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
+        e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
+    );
+    hipLaunchKernelGGL(
+        k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
+        e->d_prefix_len, e->d_labels, e->d_admitted_flags, e->d_moves
+    );
   }
 
+  // clear active for the WHOLE chunk's processed set (identical on all
+  // ranks) and tally scanned arcs
   hipLaunchKernelGGL(
-      k_weights_update, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
-      e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
+      k_clear_active, dim3(ceil_div(chunk_hi - chunk_lo, threads)), dim3(threads), 0, e->stream,
+      chunk_lo, chunk_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_arcs
   );
-  HIP_CHECK(hipMemsetAsync(e->d_admitted_count, 0, sizeof(u32), e->stream));
-  hipLaunchKernelGGL(
-      k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, e->d_sto, count,
-      e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_admitted_idx, e->d_admitted_count
-  );
-  hipLaunchKernelGGL(
-      k_clear_active, dim3(ceil_div(e->C, threads)), dim3(threads), 0, e->stream, e->d_processed,
-      e->d_counts, e->d_active
-  );
-  hipLaunchKernelGGL(
-      k_activate, dim3(ceil_div(count * kWave, threads)), dim3(threads), 0, e->stream,
-      e->d_admitted_idx, e->d_admitted_count, props, e->d_xadj, e->d_adjncy, e->d_active
-  );
-  // reset per-cluster segment state for the next chunk (touched entries only)
-  hipLaunchKernelGGL(
-      k_reset_segs, dim3(grid), dim3(threads), 0, e->stream, e->d_sto, count, e->d_seg_begin,
-      e->d_seg_end, e->d_prefix_len, e->d_dep
-  );
+  if (count > 0) {
+    hipLaunchKernelGGL(
+        k_activate, dim3(ceil_div(static_cast<u64>(count) * kWave, threads)), dim3(threads), 0,
+        e->stream, order, e->d_admitted_flags, props, count, e->d_xadj, e->d_adjncy, e->d_active
+    );
+    hipLaunchKernelGGL(
+        k_reset_segs, dim3(ceil_div(count, threads)), dim3(threads), 0, e->stream, props, count,
+        e->d_seg_begin, e->d_seg_end, e->d_prefix_len
+    );
+  }
 
-  HIP_CHECK(
-      hipMemcpyAsync(e->h_count + 1, e->d_admitted_count, sizeof(u32), hipMemcpyDeviceToHost,
-                     e->stream)
-  );
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
   HIP_CHECK(hipStreamSynchronize(e->stream));
-  e->moves += e->h_count[1];
-  return static_cast<i64>(e->h_count[1]);
+  return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
 }
 
 i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
@@ -1116,16 +1152,16 @@ i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
       k_edge_cut, dim3(ceil_div(static_cast<u64>(e->n) * kWave, 256)), dim3(256), 0, e->stream,
       e->n, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_labels, e->d_cut
   );
-  unsigned long long cut2 = 0;
+  unsigned long long cut2 = 0, arcs = 0, moves = 0;
   HIP_CHECK(hipMemcpyAsync(&cut2, e->d_cut, sizeof(cut2), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpyAsync(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpyAsync(&moves, e->d_moves, sizeof(moves), hipMemcpyDeviceToHost, e->stream));
   HIP_CHECK(hipMemcpy(partition, e->d_labels, sizeof(u32) * e->n, hipMemcpyDeviceToHost));
-  unsigned long long arcs = 0;
-  HIP_CHECK(hipMemcpy(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost));
   HIP_CHECK(hipStreamSynchronize(e->stream));
 
   if (stats) {
     stats->arcs_scanned = arcs;
-    stats->moves = e->moves;
+    stats->moves = moves;
     stats->phase_a_ns = static_cast<u64>(e->phase_a_ms * 1e6);
     stats->total_ns = 0; // caller times the region
     stats->num_clusters = 0;
@@ -1146,12 +1182,11 @@ i64 kmp_lp_refine(
   if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
     return -1;
   }
-  const u32 num_chunks = kmp::kNumChunks;
   for (int iter = 0; iter < iters; ++iter) {
     u64 sweep_moves = 0;
-    for (u32 chunk = 0; chunk < num_chunks; ++chunk) {
+    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
       const u32 pos_lo = chunk * e->C;
-      const u32 pos_hi = pos_lo + e->C > e->n ? e->n : pos_lo + e->C;
+      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
       if (pos_lo >= pos_hi) {
         continue;
       }

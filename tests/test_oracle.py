@@ -121,12 +121,15 @@ def test_perm_cross_impl(oracle):
     prod.kmp_perm.argtypes = [ctypes.c_uint32, ctypes.c_uint64, ctypes.c_int,
                               ctypes.POINTER(ctypes.c_uint32)]
     for n in [5, 64, 1000, 65536, 1 << 20]:
+        P = ((n + 63) // 64) * 64  # position space (pos_count)
         for seed, it in [(1, 0), (42, 3)]:
-            a = np.zeros(n, dtype=np.uint32)
-            b = np.zeros(n, dtype=np.uint32)
+            a = np.zeros(P, dtype=np.uint32)
+            b = np.zeros(P, dtype=np.uint32)
             oracle.kmp_oracle_perm(ctypes.c_uint32(n), ctypes.c_uint64(seed),
                                    ctypes.c_int(it), u32p(a))
             prod.kmp_perm(n, seed, it, u32p(b))
             assert (a == b).all(), (n, seed, it)
-            # is a permutation
-            assert len(np.unique(a)) == n and a.max() == n - 1
+            # valid entries form a permutation of [0, n)
+            valid = a[a < n]
+            assert len(valid) == n
+            assert len(np.unique(valid)) == n and valid.max() == n - 1
