@@ -40,6 +40,18 @@ using kmp::BlockPerm;
 using kmp::iter_seed_of;
 using kmp::tie_hash;
 
+// Check the sticky error right after a kernel launch (configuration errors
+// surface there, not at the launch statement).
+#define LAUNCH_CHECK()                                                                             \
+  do {                                                                                             \
+    hipError_t lerr_ = hipGetLastError();                                                          \
+    if (lerr_ != hipSuccess) {                                                                     \
+      fprintf(stderr, "kaminpar_amd: launch error %s at %s:%d\n", hipGetErrorString(lerr_),       \
+              __FILE__, __LINE__);                                                                 \
+      abort();                                                                                     \
+    }                                                                                              \
+  } while (0)
+
 #define HIP_CHECK(cmd)                                                                             \
   do {                                                                                             \
     hipError_t err_ = (cmd);                                                                       \
@@ -776,6 +788,7 @@ struct kmp_lp_t {
 
   // run bookkeeping
   double phase_a_ms = 0.0;
+  double commit_ms = 0.0;
   std::vector<hipEvent_t> ev_pool;
   size_t ev_used = 0;
 
@@ -951,6 +964,7 @@ int kmp_lp_refine_begin(
   e->k = k;
   e->seed = seed;
   e->phase_a_ms = 0.0;
+  e->commit_ms = 0.0;
   e->ev_used = 0;
 
   engine_free_k_buffers(e);
@@ -977,6 +991,7 @@ int kmp_lp_refine_begin(
         k_init_weights, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
         e->d_labels, e->d_vwgt, reinterpret_cast<unsigned long long *>(e->d_weights)
     );
+  LAUNCH_CHECK();
   }
   HIP_CHECK(hipStreamSynchronize(e->stream));
   return 0;
@@ -1005,6 +1020,7 @@ i64 kmp_lp_phase_a(
       e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
       e->d_slots
   );
+  LAUNCH_CHECK();
   // M: 1 position/wave
   {
     const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
@@ -1014,6 +1030,7 @@ i64 kmp_lp_phase_a(
         e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
         e->d_slots, e->d_l_list, e->d_l_count
     );
+  LAUNCH_CHECK();
   }
   // L: grid-stride workgroups over the (rare) high-degree list
   {
@@ -1023,6 +1040,7 @@ i64 kmp_lp_phase_a(
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
         e->d_l_list, e->d_l_count, e->d_slots
     );
+  LAUNCH_CHECK();
   }
   // compact valid slots in position order (stable select)
   {
@@ -1051,6 +1069,9 @@ i64 kmp_lp_phase_a(
 i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 count) {
   const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 threads = 256;
+  hipEvent_t cev0, cev1;
+  e->ev_pair(cev0, cev1);
+  HIP_CHECK(hipEventRecord(cev0, e->stream));
   const u32 chunk_lo = chunk * e->C;
   const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
 
@@ -1067,6 +1088,7 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_make_keys, dim3(grid), dim3(threads), 0, e->stream, props, count, e->d_sort_keys[0],
         e->d_sort_vals[0]
     );
+  LAUNCH_CHECK();
     rocprim::double_buffer<u32> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
     rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
     size_t tb = e->sort_temp_bytes;
@@ -1077,6 +1099,7 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
     hipLaunchKernelGGL(
         k_extract_w, dim3(grid), dim3(threads), 0, e->stream, order, props, count, e->d_sw
     );
+  LAUNCH_CHECK();
     size_t sb = e->scan_temp_bytes;
     HIP_CHECK(rocprim::inclusive_scan_by_key(
         e->d_scan_temp, sb, sto, e->d_sw, e->d_pw, count, rocprim::plus<i64>(),
@@ -1086,24 +1109,29 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_seg_bounds, dim3(grid), dim3(threads), 0, e->stream, sto, count, e->d_seg_begin,
         e->d_seg_end
     );
+  LAUNCH_CHECK();
     const u32 kgrid = ceil_div(e->k, threads);
     hipLaunchKernelGGL(
         k_seg_len, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
         e->d_prefix_len
     );
+  LAUNCH_CHECK();
 
     // greatest-fixpoint rollback (kaminpar-dist lp_refiner.cc:296-333)
     while (true) {
       hipLaunchKernelGGL(k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep);
+  LAUNCH_CHECK();
       hipLaunchKernelGGL(
           k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
           e->d_prefix_len, e->d_labels, e->d_dep
       );
+  LAUNCH_CHECK();
       HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
       hipLaunchKernelGGL(
           k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
           e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
       );
+  LAUNCH_CHECK();
       HIP_CHECK(
           hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
       );
@@ -1117,10 +1145,12 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_weights_update, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
         e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
     );
+  LAUNCH_CHECK();
     hipLaunchKernelGGL(
         k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
         e->d_prefix_len, e->d_labels, e->d_admitted_flags, e->d_moves
     );
+  LAUNCH_CHECK();
   }
 
   // clear active for the WHOLE chunk's processed set (identical on all
@@ -1129,20 +1159,27 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
       k_clear_active, dim3(ceil_div(chunk_hi - chunk_lo, threads)), dim3(threads), 0, e->stream,
       chunk_lo, chunk_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_arcs
   );
+  LAUNCH_CHECK();
   if (count > 0) {
     hipLaunchKernelGGL(
         k_activate, dim3(ceil_div(static_cast<u64>(count) * kWave, threads)), dim3(threads), 0,
         e->stream, order, e->d_admitted_flags, props, count, e->d_xadj, e->d_adjncy, e->d_active
     );
+  LAUNCH_CHECK();
     hipLaunchKernelGGL(
         k_reset_segs, dim3(ceil_div(count, threads)), dim3(threads), 0, e->stream, props, count,
         e->d_seg_begin, e->d_seg_end, e->d_prefix_len
     );
+  LAUNCH_CHECK();
   }
 
+  HIP_CHECK(hipEventRecord(cev1, e->stream));
   HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
                            hipMemcpyDeviceToHost, e->stream));
   HIP_CHECK(hipStreamSynchronize(e->stream));
+  float cms = 0;
+  HIP_CHECK(hipEventElapsedTime(&cms, cev0, cev1));
+  e->commit_ms += cms;
   return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
 }
 
@@ -1152,6 +1189,7 @@ i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
       k_edge_cut, dim3(ceil_div(static_cast<u64>(e->n) * kWave, 256)), dim3(256), 0, e->stream,
       e->n, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_labels, e->d_cut
   );
+  LAUNCH_CHECK();
   unsigned long long cut2 = 0, arcs = 0, moves = 0;
   HIP_CHECK(hipMemcpyAsync(&cut2, e->d_cut, sizeof(cut2), hipMemcpyDeviceToHost, e->stream));
   HIP_CHECK(hipMemcpyAsync(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost, e->stream));
@@ -1163,7 +1201,7 @@ i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
     stats->arcs_scanned = arcs;
     stats->moves = moves;
     stats->phase_a_ns = static_cast<u64>(e->phase_a_ms * 1e6);
-    stats->total_ns = 0; // caller times the region
+    stats->total_ns = static_cast<u64>(e->commit_ms * 1e6); // commit-region time
     stats->num_clusters = 0;
     stats->edge_cut = static_cast<i64>(cut2 / 2);
   }
