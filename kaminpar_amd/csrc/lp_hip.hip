@@ -102,6 +102,31 @@ __device__ inline bool key_better(i32 g, u64 h, u32 c, const BestState &b) {
   return c < b.c;
 }
 
+// Wave-cooperative gain accumulation for one 64-edge batch: resolve the
+// distinct clusters via ballot waterfall and issue ONE LDS atomic per
+// distinct cluster (instead of one per edge -- same-address LDS atomics
+// serialize). Unit-weight fast path uses popcount as the weight sum.
+__device__ inline void accumulate_batch(
+    u32 c, i32 w, bool valid, u32 lane, i32 *gains
+) {
+  unsigned long long unresolved = __ballot(valid);
+  while (unresolved) {
+    const u32 leader = __ffsll(static_cast<unsigned long long>(unresolved)) - 1;
+    const u32 cl = __shfl(c, leader, kWave);
+    const unsigned long long m = __ballot(valid && c == cl) & unresolved;
+    // masked weight sum (full-wave shuffle reduction)
+    i32 wsum = ((m >> lane) & 1ull) ? w : 0;
+    for (int off = 32; off > 0; off >>= 1) {
+      wsum += __shfl_down(wsum, off, kWave);
+    }
+    wsum = __shfl(wsum, 0, kWave);
+    if (lane == leader) {
+      atomicAdd(&gains[cl], wsum);
+    }
+    unresolved &= ~m;
+  }
+}
+
 // Weight-acceptance predicate (refiner variant, lp_refiner.cc:185-230).
 __device__ inline bool accept_refine(
     u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw, i64 cur_w, i64 cur_maxw
@@ -129,7 +154,11 @@ __global__ void k_phase_s(
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const uint8_t *__restrict__ active,
-    Prop *__restrict__ slots
+    Prop *__restrict__ slots,
+    u64 *__restrict__ m_list,
+    u32 *__restrict__ m_count,
+    u64 *__restrict__ l_list,
+    u32 *__restrict__ l_count
 ) {
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 sub = lane >> 4;  // subgroup 0..3
@@ -153,9 +182,42 @@ __global__ void k_phase_s(
     deg = xadj[u + 1] - row;
     if (!active[u] || deg > max_degree) {
       emit_invalid = true;
-    } else if (deg > kSmallDeg) {
-      return; // M/L owns this slot
     }
+  }
+
+  // append active mid/high-degree positions to the M/L work lists
+  // (wave-aggregated: one atomic per wave per list)
+  {
+    const bool is_m = !emit_invalid && u < n && deg > kSmallDeg && deg <= kMidDeg && slot == 0;
+    const bool is_l = !emit_invalid && u < n && deg > kMidDeg && slot == 0;
+    const u64 rec = (static_cast<u64>(p) << 32) | u;
+    unsigned long long mm = __ballot(is_m);
+    if (mm) {
+      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
+      u32 bbase = 0;
+      if (lane == leader) {
+        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
+      }
+      bbase = __shfl(bbase, leader, kWave);
+      if (is_m) {
+        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] = rec;
+      }
+    }
+    unsigned long long ll = __ballot(is_l);
+    if (ll) {
+      const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
+      u32 bbase = 0;
+      if (lane == leader) {
+        bbase = atomicAdd(l_count, static_cast<u32>(__popcll(ll)));
+      }
+      bbase = __shfl(bbase, leader, kWave);
+      if (is_l) {
+        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] = rec;
+      }
+    }
+  }
+  if (!emit_invalid && deg > kSmallDeg) {
+    return; // M/L owns this slot
   }
   if (emit_invalid) {
     if (slot == 0) {
@@ -223,17 +285,14 @@ __global__ void k_phase_s(
 }
 
 // ------------------------------------------------------------ M path
-// One wavefront per position; handles active vertices with
-// kSmallDeg < deg <= kMidDeg (dense per-wave LDS gains, k <= kMaxDenseK).
-// Active deg > kMidDeg vertices are appended to the L list (rare).
+// Grid-stride over the M list (built by k_phase_s): one wavefront per
+// vertex, dense per-wave LDS gains (k <= kMaxDenseK), ballot-waterfall
+// accumulation (one LDS atomic per distinct cluster per 64-edge batch).
 // blockDim.x = 256 (4 waves); dynamic LDS = 4 * k * sizeof(i32).
 __global__ void k_phase_m(
     u32 pos_lo,
-    u32 pos_hi,
     u32 chunk_base,
-    u32 n,
     u64 iter_seed,
-    u32 max_degree,
     u32 k,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -242,88 +301,82 @@ __global__ void k_phase_m(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    const uint8_t *__restrict__ active,
-    Prop *__restrict__ slots,
-    u64 *__restrict__ l_list, // (position << 32) | u
-    u32 *__restrict__ l_count
+    const u64 *__restrict__ m_list,
+    const u32 *__restrict__ m_count,
+    Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wave_in_wg = threadIdx.x >> 6;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const u32 p = pos_lo + wave_id;
-  if (p >= pos_hi) {
-    return;
-  }
-
-  const BlockPerm perm(n, iter_seed);
-  const u32 u = perm(p);
-  if (u >= n) {
-    return; // S wrote the slot
-  }
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
-  if (deg <= kSmallDeg || deg > max_degree || !active[u]) {
-    return; // S wrote the slot
-  }
-  const u32 sidx = p - pos_lo;
-  if (deg > kMidDeg) {
-    if (lane == 0) {
-      l_list[atomicAdd(l_count, 1u)] = (static_cast<u64>(p) << 32) | u;
-    }
-    return; // L kernel writes the slot
-  }
-
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
   i32 *gains = lds + wave_in_wg * k;
-  for (u32 c = lane; c < k; c += kWave) {
-    gains[c] = 0;
-  }
-  __threadfence_block(); // LDS ordering; gains slice is private to this wave
 
-  for (u32 e = lane; e < deg; e += kWave) {
-    const u32 v = adjncy[row + e];
-    const i32 w = adjwgt ? adjwgt[row + e] : 1;
-    atomicAdd(&gains[labels[v]], w);
-  }
-  __threadfence_block();
-
-  const u32 cur = labels[u];
-  const i32 u_w = vwgt ? vwgt[u] : 1;
-  const i64 cur_w = weights[cur];
-  const i64 cur_maxw = maxw[cur];
-
-  BestState best{0, 0, 0, false};
-  for (u32 c = lane; c < k; c += kWave) {
-    const i32 g = gains[c];
-    if (g <= 0) {
-      continue;
+  const u32 count = *m_count;
+  for (u32 vid = wave_id; vid < count; vid += num_waves) {
+    for (u32 c = lane; c < k; c += kWave) {
+      gains[c] = 0;
     }
-    const i64 cw = weights[c];
-    const i64 mw = maxw[c];
-    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
-      continue;
-    }
-    const u64 h = tie_hash(iter_seed, u, c);
-    if (key_better(g, h, c, best)) {
-      best = BestState{g, h, c, true};
-    }
-  }
+    __threadfence_block(); // LDS ordering; gains slice is private to this wave
 
-  for (int off = 32; off > 0; off >>= 1) {
-    const i32 og = __shfl_down(best.gain, off, kWave);
-    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
-    const u32 oc = __shfl_down(best.c, off, kWave);
-    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
-    if (ohave && key_better(og, oh, oc, best)) {
-      best = BestState{og, oh, oc, true};
-    }
-  }
+    const u64 rec = m_list[vid];
+    const u32 p = static_cast<u32>(rec >> 32);
+    const u32 u = static_cast<u32>(rec);
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
 
-  if (lane == 0) {
-    if (best.have && best.c != cur) {
-      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-    } else {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    for (u32 base_e = 0; base_e < deg; base_e += kWave) {
+      const u32 e = base_e + lane;
+      const bool valid = e < deg;
+      u32 c = 0;
+      i32 w = 0;
+      if (valid) {
+        c = labels[adjncy[row + e]];
+        w = adjwgt ? adjwgt[row + e] : 1;
+      }
+      accumulate_batch(c, w, valid, lane, gains);
+    }
+    __threadfence_block();
+
+    const u32 cur = labels[u];
+    const i32 u_w = vwgt ? vwgt[u] : 1;
+    const i64 cur_w = weights[cur];
+    const i64 cur_maxw = maxw[cur];
+
+    BestState best{0, 0, 0, false};
+    for (u32 c = lane; c < k; c += kWave) {
+      const i32 g = gains[c];
+      if (g <= 0) {
+        continue;
+      }
+      const i64 cw = weights[c];
+      const i64 mw = maxw[c];
+      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+        continue;
+      }
+      const u64 h = tie_hash(iter_seed, u, c);
+      if (key_better(g, h, c, best)) {
+        best = BestState{g, h, c, true};
+      }
+    }
+
+    for (int off = 32; off > 0; off >>= 1) {
+      const i32 og = __shfl_down(best.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+      const u32 oc = __shfl_down(best.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+      if (ohave && key_better(og, oh, oc, best)) {
+        best = BestState{og, oh, oc, true};
+      }
+    }
+
+    if (lane == 0) {
+      const u32 sidx = p - pos_lo;
+      if (best.have && best.c != cur) {
+        slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[sidx] = Prop{0, kInvalid, 0, 0};
+      }
     }
   }
 }
@@ -364,10 +417,17 @@ __global__ void k_phase_l(
     const u32 row = xadj[u];
     const u32 deg = xadj[u + 1] - row;
 
-    for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
-      const u32 v = adjncy[row + e];
-      const i32 w = adjwgt ? adjwgt[row + e] : 1;
-      atomicAdd(&gains[labels[v]], w);
+    const u32 lane_l = threadIdx.x & (kWave - 1);
+    for (u32 base_e = threadIdx.x - lane_l; base_e < deg; base_e += blockDim.x) {
+      const u32 e = base_e + lane_l;
+      const bool valid = e < deg;
+      u32 c = 0;
+      i32 w = 0;
+      if (valid) {
+        c = labels[adjncy[row + e]];
+        w = adjwgt ? adjwgt[row + e] : 1;
+      }
+      accumulate_batch(c, w, valid, lane_l, gains);
     }
     __syncthreads();
 
@@ -686,18 +746,35 @@ __global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep
   }
 }
 
+// Grid-stride with a per-workgroup LDS histogram (one global atomic per
+// cluster per WG instead of one per vertex -- same-address global atomics
+// serialize at ~11ns).
 __global__ void k_init_weights(
     u32 n,
+    u32 k,
     const u32 *__restrict__ labels,
     const i32 *__restrict__ vwgt,
     unsigned long long *__restrict__ weights
 ) {
-  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
-  if (u < n) {
-    atomicAdd(&weights[labels[u]], static_cast<unsigned long long>(vwgt ? vwgt[u] : 1));
+  extern __shared__ unsigned long long hist[];
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    hist[c] = 0;
+  }
+  __syncthreads();
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 u = blockIdx.x * blockDim.x + threadIdx.x; u < n; u += stride) {
+    atomicAdd(&hist[labels[u]], static_cast<unsigned long long>(vwgt ? vwgt[u] : 1));
+  }
+  __syncthreads();
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    if (hist[c]) {
+      atomicAdd(&weights[c], hist[c]);
+    }
   }
 }
 
+// Grid-stride (total thread count must stay < 2^32): one wave per vertex,
+// strided over the vertex set, per-wave partial sums, one atomic per wave.
 __global__ void k_edge_cut(
     u32 n,
     const u32 *__restrict__ xadj,
@@ -707,18 +784,17 @@ __global__ void k_edge_cut(
     unsigned long long *__restrict__ cut
 ) {
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
   const u32 lane = threadIdx.x & (kWave - 1);
-  if (wave_id >= n) {
-    return;
-  }
-  const u32 u = wave_id;
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
-  const u32 lu = labels[u];
   u64 local = 0;
-  for (u32 e = lane; e < deg; e += kWave) {
-    if (labels[adjncy[row + e]] != lu) {
-      local += adjwgt ? adjwgt[row + e] : 1;
+  for (u32 u = wave_id; u < n; u += num_waves) {
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+    const u32 lu = labels[u];
+    for (u32 e = lane; e < deg; e += kWave) {
+      if (labels[adjncy[row + e]] != lu) {
+        local += adjwgt ? adjwgt[row + e] : 1;
+      }
     }
   }
   for (int off = 32; off > 0; off >>= 1) {
@@ -756,6 +832,8 @@ struct kmp_lp_t {
   // phase buffers
   Prop *d_slots = nullptr; // C
   Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
+  u64 *d_m_list = nullptr; // C
+  u32 *d_m_count = nullptr;
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr;
   u32 *d_prop_count = nullptr;
@@ -881,6 +959,8 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   const u32 C = e->C;
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
+  HIP_CHECK(hipMalloc(&e->d_m_list, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
   HIP_CHECK(hipMalloc(&e->d_l_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
@@ -928,8 +1008,8 @@ void kmp_lp_free(kmp_lp_t *e) {
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
-                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
-                  (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_m_list,
+                  (void *)e->d_m_count, (void *)e->d_l_list, (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
@@ -987,11 +1067,12 @@ int kmp_lp_refine_begin(
   HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * k, e->stream));
   {
     const u32 threads = 256;
+    const size_t lds = static_cast<size_t>(k) * sizeof(unsigned long long);
     hipLaunchKernelGGL(
-        k_init_weights, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
-        e->d_labels, e->d_vwgt, reinterpret_cast<unsigned long long *>(e->d_weights)
+        k_init_weights, dim3(2048), dim3(threads), lds, e->stream, e->n, k, e->d_labels, e->d_vwgt,
+        reinterpret_cast<unsigned long long *>(e->d_weights)
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
   }
   HIP_CHECK(hipStreamSynchronize(e->stream));
   return 0;
@@ -1007,6 +1088,7 @@ i64 kmp_lp_phase_a(
   const u32 threads = 256;
   const u32 max_degree = 0xFFFFFFFFu;
 
+  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_l_count, 0, sizeof(u32), e->stream));
 
   hipEvent_t ev0, ev1;
@@ -1018,19 +1100,18 @@ i64 kmp_lp_phase_a(
       k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->d_xadj,
       e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-      e->d_slots
+      e->d_slots, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
   );
   LAUNCH_CHECK();
-  // M: 1 position/wave
+  // M: grid-stride waves over the M list
   {
     const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
     hipLaunchKernelGGL(
-        k_phase_m, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
-        e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
-        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-        e->d_slots, e->d_l_list, e->d_l_count
+        k_phase_m, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, iseed, e->k,
+        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
+        e->d_m_list, e->d_m_count, e->d_slots
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
   }
   // L: grid-stride workgroups over the (rare) high-degree list
   {
@@ -1117,21 +1198,27 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
     );
   LAUNCH_CHECK();
 
-    // greatest-fixpoint rollback (kaminpar-dist lp_refiner.cc:296-333)
+    // greatest-fixpoint rollback (kaminpar-dist lp_refiner.cc:296-333).
+    // Two rounds are enqueued per host sync (the fixpoint almost always
+    // converges within two); the changed flag of the SECOND round decides.
     while (true) {
-      hipLaunchKernelGGL(k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep);
-  LAUNCH_CHECK();
-      hipLaunchKernelGGL(
-          k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
-          e->d_prefix_len, e->d_labels, e->d_dep
-      );
-  LAUNCH_CHECK();
-      HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
-      hipLaunchKernelGGL(
-          k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
-          e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
-      );
-  LAUNCH_CHECK();
+      for (int half = 0; half < 2; ++half) {
+        hipLaunchKernelGGL(
+            k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep
+        );
+        LAUNCH_CHECK();
+        hipLaunchKernelGGL(
+            k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count,
+            e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
+        );
+        LAUNCH_CHECK();
+        HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
+        hipLaunchKernelGGL(
+            k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
+            e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
+        );
+        LAUNCH_CHECK();
+      }
       HIP_CHECK(
           hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
       );
@@ -1186,8 +1273,8 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
 i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
   HIP_CHECK(hipMemsetAsync(e->d_cut, 0, sizeof(unsigned long long), e->stream));
   hipLaunchKernelGGL(
-      k_edge_cut, dim3(ceil_div(static_cast<u64>(e->n) * kWave, 256)), dim3(256), 0, e->stream,
-      e->n, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_labels, e->d_cut
+      k_edge_cut, dim3(32768), dim3(256), 0, e->stream, e->n, e->d_xadj, e->d_adjncy, e->d_adjwgt,
+      e->d_labels, e->d_cut
   );
   LAUNCH_CHECK();
   unsigned long long cut2 = 0, arcs = 0, moves = 0;
