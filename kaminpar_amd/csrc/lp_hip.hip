@@ -106,6 +106,7 @@ __device__ inline bool key_better(i32 g, u64 h, u32 c, const BestState &b) {
 // distinct clusters via ballot waterfall and issue ONE LDS atomic per
 // distinct cluster (instead of one per edge -- same-address LDS atomics
 // serialize). Unit-weight fast path uses popcount as the weight sum.
+template <bool kUnitWeights>
 __device__ inline void accumulate_batch(
     u32 c, i32 w, bool valid, u32 lane, i32 *gains
 ) {
@@ -114,14 +115,19 @@ __device__ inline void accumulate_batch(
     const u32 leader = __ffsll(static_cast<unsigned long long>(unresolved)) - 1;
     const u32 cl = __shfl(c, leader, kWave);
     const unsigned long long m = __ballot(valid && c == cl) & unresolved;
-    // masked weight sum (full-wave shuffle reduction)
-    i32 wsum = ((m >> lane) & 1ull) ? w : 0;
-    for (int off = 32; off > 0; off >>= 1) {
-      wsum += __shfl_down(wsum, off, kWave);
-    }
-    wsum = __shfl(wsum, 0, kWave);
-    if (lane == leader) {
-      atomicAdd(&gains[cl], wsum);
+    if constexpr (kUnitWeights) {
+      if (lane == leader) {
+        atomicAdd(&gains[cl], static_cast<i32>(__popcll(m)));
+      }
+    } else {
+      // masked weight sum (full-wave shuffle reduction)
+      i32 wsum = ((m >> lane) & 1ull) ? w : 0;
+      for (int off = 32; off > 0; off >>= 1) {
+        wsum += __shfl_down(wsum, off, kWave);
+      }
+      if (lane == leader) {
+        atomicAdd(&gains[cl], __shfl(wsum, 0, kWave));
+      }
     }
     unresolved &= ~m;
   }
@@ -155,8 +161,6 @@ __global__ void k_phase_s(
     const i64 *__restrict__ maxw,
     const uint8_t *__restrict__ active,
     Prop *__restrict__ slots,
-    u64 *__restrict__ m_list,
-    u32 *__restrict__ m_count,
     u64 *__restrict__ l_list,
     u32 *__restrict__ l_count
 ) {
@@ -185,25 +189,11 @@ __global__ void k_phase_s(
     }
   }
 
-  // append active mid/high-degree positions to the M/L work lists
-  // (wave-aggregated: one atomic per wave per list)
+  // append active high-degree positions to the L work list (rare;
+  // wave-aggregated: one atomic per wave that holds an L candidate)
   {
-    const bool is_m = !emit_invalid && u < n && deg > kSmallDeg && deg <= kMidDeg && slot == 0;
     const bool is_l = !emit_invalid && u < n && deg > kMidDeg && slot == 0;
-    const u64 rec = (static_cast<u64>(p) << 32) | u;
-    unsigned long long mm = __ballot(is_m);
-    if (mm) {
-      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
-      u32 bbase = 0;
-      if (lane == leader) {
-        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
-      }
-      bbase = __shfl(bbase, leader, kWave);
-      if (is_m) {
-        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] = rec;
-      }
-    }
-    unsigned long long ll = __ballot(is_l);
+    const unsigned long long ll = __ballot(is_l);
     if (ll) {
       const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
       u32 bbase = 0;
@@ -212,7 +202,8 @@ __global__ void k_phase_s(
       }
       bbase = __shfl(bbase, leader, kWave);
       if (is_l) {
-        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] = rec;
+        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] =
+            (static_cast<u64>(p) << 32) | u;
       }
     }
   }
@@ -285,14 +276,20 @@ __global__ void k_phase_s(
 }
 
 // ------------------------------------------------------------ M path
-// Grid-stride over the M list (built by k_phase_s): one wavefront per
-// vertex, dense per-wave LDS gains (k <= kMaxDenseK), ballot-waterfall
-// accumulation (one LDS atomic per distinct cluster per 64-edge batch).
+// One wavefront per position (dead waves for non-M positions retire in a
+// few cycles; list building would need a serializing append counter).
+// Handles active vertices with kSmallDeg < deg <= kMidDeg: dense per-wave
+// LDS gains (k <= kMaxDenseK), ballot-waterfall accumulation. Active
+// deg > kMidDeg vertices were appended to the L list by k_phase_s.
 // blockDim.x = 256 (4 waves); dynamic LDS = 4 * k * sizeof(i32).
+template <bool kUnitWeights>
 __global__ void k_phase_m(
     u32 pos_lo,
+    u32 pos_hi,
     u32 chunk_base,
+    u32 n,
     u64 iter_seed,
+    u32 max_degree,
     u32 k,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -301,82 +298,88 @@ __global__ void k_phase_m(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    const u64 *__restrict__ m_list,
-    const u32 *__restrict__ m_count,
+    const uint8_t *__restrict__ active,
     Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wave_in_wg = threadIdx.x >> 6;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
+  const u32 p = pos_lo + wave_id;
+  if (p >= pos_hi) {
+    return;
+  }
+
+  const BlockPerm perm(n, iter_seed);
+  const u32 u = perm(p);
+  if (u >= n) {
+    return;
+  }
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+  if (deg <= kSmallDeg || deg > kMidDeg || deg > max_degree || !active[u]) {
+    return; // S or L owns this slot
+  }
+
   i32 *gains = lds + wave_in_wg * k;
+  for (u32 c = lane; c < k; c += kWave) {
+    gains[c] = 0;
+  }
+  __threadfence_block(); // LDS ordering; gains slice is private to this wave
 
-  const u32 count = *m_count;
-  for (u32 vid = wave_id; vid < count; vid += num_waves) {
-    for (u32 c = lane; c < k; c += kWave) {
-      gains[c] = 0;
-    }
-    __threadfence_block(); // LDS ordering; gains slice is private to this wave
-
-    const u64 rec = m_list[vid];
-    const u32 p = static_cast<u32>(rec >> 32);
-    const u32 u = static_cast<u32>(rec);
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
-
-    for (u32 base_e = 0; base_e < deg; base_e += kWave) {
-      const u32 e = base_e + lane;
-      const bool valid = e < deg;
-      u32 c = 0;
-      i32 w = 0;
-      if (valid) {
-        c = labels[adjncy[row + e]];
-        w = adjwgt ? adjwgt[row + e] : 1;
-      }
-      accumulate_batch(c, w, valid, lane, gains);
-    }
-    __threadfence_block();
-
-    const u32 cur = labels[u];
-    const i32 u_w = vwgt ? vwgt[u] : 1;
-    const i64 cur_w = weights[cur];
-    const i64 cur_maxw = maxw[cur];
-
-    BestState best{0, 0, 0, false};
-    for (u32 c = lane; c < k; c += kWave) {
-      const i32 g = gains[c];
-      if (g <= 0) {
-        continue;
-      }
-      const i64 cw = weights[c];
-      const i64 mw = maxw[c];
-      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
-        continue;
-      }
-      const u64 h = tie_hash(iter_seed, u, c);
-      if (key_better(g, h, c, best)) {
-        best = BestState{g, h, c, true};
+  for (u32 base_e = 0; base_e < deg; base_e += kWave) {
+    const u32 e = base_e + lane;
+    const bool valid = e < deg;
+    u32 c = 0;
+    i32 w = 0;
+    if (valid) {
+      c = labels[adjncy[row + e]];
+      if constexpr (!kUnitWeights) {
+        w = adjwgt[row + e];
       }
     }
+    accumulate_batch<kUnitWeights>(c, w, valid, lane, gains);
+  }
+  __threadfence_block();
 
-    for (int off = 32; off > 0; off >>= 1) {
-      const i32 og = __shfl_down(best.gain, off, kWave);
-      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
-      const u32 oc = __shfl_down(best.c, off, kWave);
-      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
-      if (ohave && key_better(og, oh, oc, best)) {
-        best = BestState{og, oh, oc, true};
-      }
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+  const i64 cur_maxw = maxw[cur];
+
+  BestState best{0, 0, 0, false};
+  for (u32 c = lane; c < k; c += kWave) {
+    const i32 g = gains[c];
+    if (g <= 0) {
+      continue;
     }
+    const i64 cw = weights[c];
+    const i64 mw = maxw[c];
+    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      continue;
+    }
+    const u64 h = tie_hash(iter_seed, u, c);
+    if (key_better(g, h, c, best)) {
+      best = BestState{g, h, c, true};
+    }
+  }
 
-    if (lane == 0) {
-      const u32 sidx = p - pos_lo;
-      if (best.have && best.c != cur) {
-        slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-      } else {
-        slots[sidx] = Prop{0, kInvalid, 0, 0};
-      }
+  for (int off = 32; off > 0; off >>= 1) {
+    const i32 og = __shfl_down(best.gain, off, kWave);
+    const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+    const u32 oc = __shfl_down(best.c, off, kWave);
+    const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+    if (ohave && key_better(og, oh, oc, best)) {
+      best = BestState{og, oh, oc, true};
+    }
+  }
+
+  if (lane == 0) {
+    const u32 sidx = p - pos_lo;
+    if (best.have && best.c != cur) {
+      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
     }
   }
 }
@@ -384,6 +387,7 @@ __global__ void k_phase_m(
 // ------------------------------------------------------------ L path
 // Grid-stride over the L list; one 256-thread workgroup per vertex, shared
 // dense LDS gains (k <= kMaxDenseK).
+template <bool kUnitWeights>
 __global__ void k_phase_l(
     u32 pos_lo,
     u32 chunk_base,
@@ -425,9 +429,11 @@ __global__ void k_phase_l(
       i32 w = 0;
       if (valid) {
         c = labels[adjncy[row + e]];
-        w = adjwgt ? adjwgt[row + e] : 1;
+        if constexpr (!kUnitWeights) {
+          w = adjwgt[row + e];
+        }
       }
-      accumulate_batch(c, w, valid, lane_l, gains);
+      accumulate_batch<kUnitWeights>(c, w, valid, lane_l, gains);
     }
     __syncthreads();
 
@@ -832,8 +838,6 @@ struct kmp_lp_t {
   // phase buffers
   Prop *d_slots = nullptr; // C
   Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
-  u64 *d_m_list = nullptr; // C
-  u32 *d_m_count = nullptr;
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr;
   u32 *d_prop_count = nullptr;
@@ -959,8 +963,6 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   const u32 C = e->C;
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
-  HIP_CHECK(hipMalloc(&e->d_m_list, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
   HIP_CHECK(hipMalloc(&e->d_l_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
@@ -1008,8 +1010,8 @@ void kmp_lp_free(kmp_lp_t *e) {
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
-                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_m_list,
-                  (void *)e->d_m_count, (void *)e->d_l_list, (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
+                  (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
@@ -1088,7 +1090,6 @@ i64 kmp_lp_phase_a(
   const u32 threads = 256;
   const u32 max_degree = 0xFFFFFFFFu;
 
-  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_l_count, 0, sizeof(u32), e->stream));
 
   hipEvent_t ev0, ev1;
@@ -1100,28 +1101,31 @@ i64 kmp_lp_phase_a(
       k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->d_xadj,
       e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-      e->d_slots, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+      e->d_slots, e->d_l_list, e->d_l_count
   );
   LAUNCH_CHECK();
-  // M: grid-stride waves over the M list
+  // M: one wave per position
   {
     const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
+    auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
     hipLaunchKernelGGL(
-        k_phase_m, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, iseed, e->k,
-        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_m_list, e->d_m_count, e->d_slots
+        kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
+        e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
+        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
+        e->d_slots
     );
     LAUNCH_CHECK();
   }
   // L: grid-stride workgroups over the (rare) high-degree list
   {
     const size_t lds = (static_cast<size_t>((e->k + 1) & ~1u)) * sizeof(i32) + 16 * sizeof(i64);
+    auto *kern = e->has_adjwgt ? k_phase_l<false> : k_phase_l<true>;
     hipLaunchKernelGGL(
-        k_phase_l, dim3(2048), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k,
-        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_l_list, e->d_l_count, e->d_slots
+        kern, dim3(2048), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
+        e->d_l_count, e->d_slots
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
   }
   // compact valid slots in position order (stable select)
   {
