@@ -102,35 +102,18 @@ __device__ inline bool key_better(i32 g, u64 h, u32 c, const BestState &b) {
   return c < b.c;
 }
 
-// Wave-cooperative gain accumulation for one 64-edge batch: resolve the
-// distinct clusters via ballot waterfall and issue ONE LDS atomic per
-// distinct cluster (instead of one per edge -- same-address LDS atomics
-// serialize). Unit-weight fast path uses popcount as the weight sum.
-template <bool kUnitWeights>
-__device__ inline void accumulate_batch(
-    u32 c, i32 w, bool valid, u32 lane, i32 *gains
-) {
-  unsigned long long unresolved = __ballot(valid);
-  while (unresolved) {
-    const u32 leader = __ffsll(static_cast<unsigned long long>(unresolved)) - 1;
-    const u32 cl = __shfl(c, leader, kWave);
-    const unsigned long long m = __ballot(valid && c == cl) & unresolved;
-    if constexpr (kUnitWeights) {
-      if (lane == leader) {
-        atomicAdd(&gains[cl], static_cast<i32>(__popcll(m)));
-      }
-    } else {
-      // masked weight sum (full-wave shuffle reduction)
-      i32 wsum = ((m >> lane) & 1ull) ? w : 0;
-      for (int off = 32; off > 0; off >>= 1) {
-        wsum += __shfl_down(wsum, off, kWave);
-      }
-      if (lane == leader) {
-        atomicAdd(&gains[cl], __shfl(wsum, 0, kWave));
-      }
-    }
-    unresolved &= ~m;
+// Gain accumulation uses REPLICATED per-cluster LDS counters: lane l adds
+// into replica l % R, so same-address LDS atomic serialization (the dominant
+// cost with a single counter per cluster) is cut by ~R. The replicas are
+// merged once per vertex. R is chosen so k * R stays within the LDS budget.
+__host__ __device__ inline u32 gain_replicas(u32 k) {
+  if (k <= 64) {
+    return 16;
   }
+  if (k <= 256) {
+    return 4;
+  }
+  return 1;
 }
 
 // Weight-acceptance predicate (refiner variant, lp_refiner.cc:185-230).
@@ -321,24 +304,18 @@ __global__ void k_phase_m(
     return; // S or L owns this slot
   }
 
-  i32 *gains = lds + wave_in_wg * k;
-  for (u32 c = lane; c < k; c += kWave) {
+  const u32 R = gain_replicas(k);
+  i32 *gains = lds + wave_in_wg * k * R; // R replicas of k counters
+  for (u32 c = lane; c < k * R; c += kWave) {
     gains[c] = 0;
   }
   __threadfence_block(); // LDS ordering; gains slice is private to this wave
 
-  for (u32 base_e = 0; base_e < deg; base_e += kWave) {
-    const u32 e = base_e + lane;
-    const bool valid = e < deg;
-    u32 c = 0;
-    i32 w = 0;
-    if (valid) {
-      c = labels[adjncy[row + e]];
-      if constexpr (!kUnitWeights) {
-        w = adjwgt[row + e];
-      }
-    }
-    accumulate_batch<kUnitWeights>(c, w, valid, lane, gains);
+  const u32 rep_off = (lane % R) * k;
+  for (u32 e = lane; e < deg; e += kWave) {
+    const u32 v = adjncy[row + e];
+    const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+    atomicAdd(&gains[rep_off + labels[v]], w);
   }
   __threadfence_block();
 
@@ -349,7 +326,10 @@ __global__ void k_phase_m(
 
   BestState best{0, 0, 0, false};
   for (u32 c = lane; c < k; c += kWave) {
-    const i32 g = gains[c];
+    i32 g = gains[c];
+    for (u32 r = 1; r < R; ++r) {
+      g += gains[r * k + c];
+    }
     if (g <= 0) {
       continue;
     }
@@ -405,12 +385,13 @@ __global__ void k_phase_l(
     Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
-  i32 *gains = lds;                                          // k i32 (padded)
-  i64 *red = reinterpret_cast<i64 *>(lds + ((k + 1) & ~1u)); // cross-wave scratch
+  const u32 R = gain_replicas(k);
+  i32 *gains = lds; // k * R i32 (replicated counters, whole workgroup)
+  i64 *red = reinterpret_cast<i64 *>(lds + ((k * R + 1) & ~1u)); // scratch
 
   const u32 count = *l_count;
   for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
-    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    for (u32 c = threadIdx.x; c < k * R; c += blockDim.x) {
       gains[c] = 0;
     }
     __syncthreads();
@@ -421,19 +402,11 @@ __global__ void k_phase_l(
     const u32 row = xadj[u];
     const u32 deg = xadj[u + 1] - row;
 
-    const u32 lane_l = threadIdx.x & (kWave - 1);
-    for (u32 base_e = threadIdx.x - lane_l; base_e < deg; base_e += blockDim.x) {
-      const u32 e = base_e + lane_l;
-      const bool valid = e < deg;
-      u32 c = 0;
-      i32 w = 0;
-      if (valid) {
-        c = labels[adjncy[row + e]];
-        if constexpr (!kUnitWeights) {
-          w = adjwgt[row + e];
-        }
-      }
-      accumulate_batch<kUnitWeights>(c, w, valid, lane_l, gains);
+    const u32 rep_off = (threadIdx.x % R) * k;
+    for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
+      const u32 v = adjncy[row + e];
+      const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+      atomicAdd(&gains[rep_off + labels[v]], w);
     }
     __syncthreads();
 
@@ -444,7 +417,10 @@ __global__ void k_phase_l(
 
     BestState best{0, 0, 0, false};
     for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
-      const i32 g = gains[c];
+      i32 g = gains[c];
+      for (u32 r = 1; r < R; ++r) {
+        g += gains[r * k + c];
+      }
       if (g <= 0) {
         continue;
       }
@@ -547,24 +523,38 @@ __global__ void k_seg_len(
   }
 }
 
+// Per-WG LDS histogram of departures (one global atomic per cluster per WG;
+// a per-proposal global atomic on k addresses serializes badly). Dynamic
+// LDS: k x u64. Grid-stride so the WG count stays bounded.
 __global__ void k_dep(
     const u32 *__restrict__ order,
     const Prop *__restrict__ props,
     const u32 *__restrict__ sto,
     u32 count,
+    u32 k,
     const u32 *__restrict__ seg_begin,
     const u32 *__restrict__ prefix_len,
     const u32 *__restrict__ labels,
     unsigned long long *__restrict__ dep
 ) {
-  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= count) {
-    return;
+  extern __shared__ unsigned long long hist[];
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    hist[c] = 0;
   }
-  const u32 to = sto[i];
-  if (i - seg_begin[to] < prefix_len[to]) {
-    const Prop pr = props[order[i]];
-    atomicAdd(&dep[labels[pr.u]], static_cast<unsigned long long>(pr.w));
+  __syncthreads();
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < count; i += stride) {
+    const u32 to = sto[i];
+    if (i - seg_begin[to] < prefix_len[to]) {
+      const Prop pr = props[order[i]];
+      atomicAdd(&hist[labels[pr.u]], static_cast<unsigned long long>(pr.w));
+    }
+  }
+  __syncthreads();
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    if (hist[c]) {
+      atomicAdd(&dep[c], hist[c]);
+    }
   }
 }
 
@@ -779,8 +769,8 @@ __global__ void k_init_weights(
   }
 }
 
-// Grid-stride (total thread count must stay < 2^32): one wave per vertex,
-// strided over the vertex set, per-wave partial sums, one atomic per wave.
+// Grid-stride, 16-lane subgroups (4 vertices per wave) so low-degree rows
+// keep lanes busy; per-wave partial sums, one atomic per wave.
 __global__ void k_edge_cut(
     u32 n,
     const u32 *__restrict__ xadj,
@@ -789,15 +779,21 @@ __global__ void k_edge_cut(
     const u32 *__restrict__ labels,
     unsigned long long *__restrict__ cut
 ) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 sub = lane >> 4;
+  const u32 slot = lane & 15;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
-  const u32 lane = threadIdx.x & (kWave - 1);
   u64 local = 0;
-  for (u32 u = wave_id; u < n; u += num_waves) {
+  for (u32 base = wave_id * 4; base < n; base += num_waves * 4) {
+    const u32 u = base + sub;
+    if (u >= n) {
+      continue;
+    }
     const u32 row = xadj[u];
     const u32 deg = xadj[u + 1] - row;
     const u32 lu = labels[u];
-    for (u32 e = lane; e < deg; e += kWave) {
+    for (u32 e = slot; e < deg; e += 16) {
       if (labels[adjncy[row + e]] != lu) {
         local += adjwgt ? adjwgt[row + e] : 1;
       }
@@ -1106,7 +1102,8 @@ i64 kmp_lp_phase_a(
   LAUNCH_CHECK();
   // M: one wave per position
   {
-    const size_t lds = static_cast<size_t>(threads / kWave) * e->k * sizeof(i32);
+    const size_t lds =
+        static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
     auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
     hipLaunchKernelGGL(
         kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
@@ -1118,7 +1115,9 @@ i64 kmp_lp_phase_a(
   }
   // L: grid-stride workgroups over the (rare) high-degree list
   {
-    const size_t lds = (static_cast<size_t>((e->k + 1) & ~1u)) * sizeof(i32) + 16 * sizeof(i64);
+    const size_t lds =
+        ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
+        16 * sizeof(i64);
     auto *kern = e->has_adjwgt ? k_phase_l<false> : k_phase_l<true>;
     hipLaunchKernelGGL(
         kern, dim3(2048), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
@@ -1212,8 +1211,9 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         );
         LAUNCH_CHECK();
         hipLaunchKernelGGL(
-            k_dep, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count,
-            e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
+            k_dep, dim3(grid > 2048 ? 2048 : grid), dim3(threads),
+            static_cast<size_t>(e->k) * sizeof(unsigned long long), e->stream, order, props, sto,
+            count, e->k, e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
         );
         LAUNCH_CHECK();
         HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
