@@ -160,6 +160,14 @@ int64_t kmp_lp_commit(
 /* Finish a sharded run: download the partition, report stats. */
 int64_t kmp_lp_refine_end(kmp_lp_t *e, uint32_t *partition, kmp_lp_stats_t *stats);
 
+/* ------------------- device-resident stepping (benchmark region) ---------
+ * The timed region excludes host transfers and the final edge-cut kernel,
+ * matching the reference's LP-only timing
+ * (shm_label_propagation_benchmark.cc:121-123). */
+int kmp_lp_reset(kmp_lp_t *e);                     /* restore initial state (D2D) */
+int64_t kmp_lp_run_sweeps(kmp_lp_t *e, int iters); /* the timed LP region */
+int kmp_lp_get_stats(kmp_lp_t *e, kmp_lp_stats_t *stats); /* no cut/download */
+
 #ifdef __cplusplus
 } /* extern "C" */
 #endif

@@ -73,6 +73,12 @@ def _load():
     lib.kmp_lp_commit.argtypes = [vp, ctypes.c_int, u32, vp, u32]
     lib.kmp_lp_refine_end.restype = i64
     lib.kmp_lp_refine_end.argtypes = [vp, p(u32), vp]
+    lib.kmp_lp_reset.restype = ctypes.c_int
+    lib.kmp_lp_reset.argtypes = [vp]
+    lib.kmp_lp_run_sweeps.restype = i64
+    lib.kmp_lp_run_sweeps.argtypes = [vp, ctypes.c_int]
+    lib.kmp_lp_get_stats.restype = ctypes.c_int
+    lib.kmp_lp_get_stats.argtypes = [vp, vp]
     return lib
 
 
@@ -239,6 +245,21 @@ class LpEngine:
         if mv < 0:
             raise RuntimeError("kmp_lp_commit failed")
         return int(mv)
+
+    def reset(self):
+        if _lib.kmp_lp_reset(self._h) != 0:
+            raise RuntimeError("kmp_lp_reset failed")
+
+    def run_sweeps(self, iters=5):
+        mv = _lib.kmp_lp_run_sweeps(self._h, iters)
+        if mv < 0:
+            raise RuntimeError("kmp_lp_run_sweeps failed")
+        return int(mv)
+
+    def get_stats(self):
+        stats = Stats()
+        _lib.kmp_lp_get_stats(self._h, ctypes.byref(stats))
+        return stats
 
     def refine_end(self):
         part = np.zeros(self._graph.n, dtype=np.uint32)

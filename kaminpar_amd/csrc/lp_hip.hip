@@ -365,10 +365,190 @@ __global__ void k_phase_m(
 }
 
 // ------------------------------------------------------------ L path
-// Grid-stride over the L list; one 256-thread workgroup per vertex, shared
-// dense LDS gains (k <= kMaxDenseK).
+// High-degree vertices are processed slice-parallel: rows are cut into
+// kLSlice-edge slices, each handled by one workgroup accumulating into a
+// per-vertex global gains row (via a per-WG replicated LDS histogram), then
+// a selection kernel reduces each row. This keeps mega-hubs (100K+ edges)
+// from serializing on a single workgroup.
+constexpr u32 kLSlice = 8192;
+
+// Single tiny kernel: per-vertex slice counts -> exclusive prefix (l_off),
+// total in l_off[count]. One wave; l_count is small (hubs are rare).
+__global__ void k_l_prep(
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    const u32 *__restrict__ xadj,
+    u32 l_cap,
+    u32 *__restrict__ l_off
+) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    const u32 count = *l_count < l_cap ? *l_count : l_cap;
+    u32 acc = 0;
+    for (u32 i = 0; i < count; ++i) {
+      l_off[i] = acc;
+      const u32 u = static_cast<u32>(l_list[i]);
+      const u32 deg = xadj[u + 1] - xadj[u];
+      acc += (deg + kLSlice - 1) / kLSlice;
+    }
+    l_off[count] = acc;
+  }
+}
+
+// Accumulate one slice per workgroup into the vertex's global gains row.
 template <bool kUnitWeights>
-__global__ void k_phase_l(
+__global__ void k_phase_l_acc(
+    u32 k,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    u32 l_cap,
+    const u32 *__restrict__ l_off,
+    i32 *__restrict__ l_gains // l_cap x k
+) {
+  extern __shared__ i32 lds[];
+  const u32 R = gain_replicas(k);
+  i32 *hist = lds;
+
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  const u32 total = l_off[count];
+  for (u32 s = blockIdx.x; s < total; s += gridDim.x) {
+    // binary search the vertex owning slice s
+    u32 lo = 0, hi = count - 1;
+    while (lo < hi) {
+      const u32 mid = (lo + hi + 1) >> 1;
+      if (l_off[mid] <= s) {
+        lo = mid;
+      } else {
+        hi = mid - 1;
+      }
+    }
+    const u32 vid = lo;
+    const u32 u = static_cast<u32>(l_list[vid]);
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+    const u32 e_lo = (s - l_off[vid]) * kLSlice;
+    const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
+
+    for (u32 c = threadIdx.x; c < k * R; c += blockDim.x) {
+      hist[c] = 0;
+    }
+    __syncthreads();
+    const u32 rep_off = (threadIdx.x % R) * k;
+    for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
+      const u32 v = adjncy[row + e];
+      const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+      atomicAdd(&hist[rep_off + labels[v]], w);
+    }
+    __syncthreads();
+    i32 *grow = l_gains + static_cast<size_t>(vid) * k;
+    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+      i32 g = hist[c];
+      for (u32 r = 1; r < R; ++r) {
+        g += hist[r * k + c];
+      }
+      if (g) {
+        atomicAdd(&grow[c], g);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// Select per L vertex (one workgroup each, grid-stride), write the slot,
+// and reset the gains row for the next chunk.
+__global__ void k_phase_l_sel(
+    u32 pos_lo,
+    u32 chunk_base,
+    u64 iter_seed,
+    u32 k,
+    const u32 *__restrict__ xadj,
+    const i32 *__restrict__ vwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    u32 l_cap,
+    i32 *__restrict__ l_gains,
+    Prop *__restrict__ slots
+) {
+  __shared__ i64 red[16];
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+    const u64 rec = l_list[vid];
+    const u32 p = static_cast<u32>(rec >> 32);
+    const u32 u = static_cast<u32>(rec);
+    const u32 cur = labels[u];
+    const i32 u_w = vwgt ? vwgt[u] : 1;
+    const i64 cur_w = weights[cur];
+    const i64 cur_maxw = maxw[cur];
+    i32 *grow = l_gains + static_cast<size_t>(vid) * k;
+
+    BestState best{0, 0, 0, false};
+    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+      const i32 g = grow[c];
+      grow[c] = 0; // reset for the next chunk
+      if (g <= 0) {
+        continue;
+      }
+      const i64 cw = weights[c];
+      const i64 mw = maxw[c];
+      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+        continue;
+      }
+      const u64 h = tie_hash(iter_seed, u, c);
+      if (key_better(g, h, c, best)) {
+        best = BestState{g, h, c, true};
+      }
+    }
+    const u32 lane = threadIdx.x & (kWave - 1);
+    for (int off = 32; off > 0; off >>= 1) {
+      const i32 og = __shfl_down(best.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+      const u32 oc = __shfl_down(best.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+      if (ohave && key_better(og, oh, oc, best)) {
+        best = BestState{og, oh, oc, true};
+      }
+    }
+    const u32 wave_in_wg = threadIdx.x >> 6;
+    if (lane == 0) {
+      red[wave_in_wg * 2] = (static_cast<i64>(best.gain) << 1) | (best.have ? 1 : 0);
+      red[wave_in_wg * 2 + 1] = static_cast<i64>(best.h);
+      reinterpret_cast<u32 *>(red + 8)[wave_in_wg] = best.c;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      BestState total{0, 0, 0, false};
+      const u32 waves = blockDim.x >> 6;
+      for (u32 wv = 0; wv < waves; ++wv) {
+        const i64 packed = red[wv * 2];
+        if (packed & 1) {
+          const i32 g = static_cast<i32>(packed >> 1);
+          const u64 h = static_cast<u64>(red[wv * 2 + 1]);
+          const u32 c = reinterpret_cast<u32 *>(red + 8)[wv];
+          if (key_better(g, h, c, total)) {
+            total = BestState{g, h, c, true};
+          }
+        }
+      }
+      if (total.have && total.c != cur) {
+        slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// Fallback for L entries beyond l_cap (pathological): one workgroup per
+// vertex, whole row, replicated LDS histogram.
+template <bool kUnitWeights>
+__global__ void k_phase_l_direct(
     u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
@@ -382,15 +562,16 @@ __global__ void k_phase_l(
     const i64 *__restrict__ maxw,
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
+    u32 l_cap,
     Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
   const u32 R = gain_replicas(k);
-  i32 *gains = lds; // k * R i32 (replicated counters, whole workgroup)
-  i64 *red = reinterpret_cast<i64 *>(lds + ((k * R + 1) & ~1u)); // scratch
+  i32 *gains = lds;
+  i64 *red = reinterpret_cast<i64 *>(lds + ((k * R + 1) & ~1u));
 
   const u32 count = *l_count;
-  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+  for (u32 vid = l_cap + blockIdx.x; vid < count; vid += gridDim.x) {
     for (u32 c = threadIdx.x; c < k * R; c += blockDim.x) {
       gains[c] = 0;
     }
@@ -471,7 +652,7 @@ __global__ void k_phase_l(
         slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
       }
     }
-    __syncthreads(); // gains/red reuse across grid-stride iterations
+    __syncthreads();
   }
 }
 
@@ -827,6 +1008,7 @@ struct kmp_lp_t {
 
   // device LP state
   u32 *d_labels = nullptr;
+  u32 *d_labels0 = nullptr; // initial labels (for kmp_lp_reset)
   i64 *d_weights = nullptr;
   i64 *d_maxw = nullptr;
   uint8_t *d_active = nullptr;
@@ -836,6 +1018,9 @@ struct kmp_lp_t {
   Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr;
+  u32 *d_l_off = nullptr;  // l_cap + 1 (slice prefix)
+  i32 *d_l_gains = nullptr; // l_cap x k (allocated at refine_begin)
+  u32 l_cap = 0;
   u32 *d_prop_count = nullptr;
   unsigned long long *d_arcs = nullptr;
   unsigned long long *d_moves = nullptr;
@@ -954,6 +1139,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   }
 
   HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
+  HIP_CHECK(hipMalloc(&e->d_labels0, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_active, e->n));
 
   const u32 C = e->C;
@@ -1005,9 +1191,9 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
                   (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
-                  (void *)e->d_l_count, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
@@ -1056,8 +1242,27 @@ int kmp_lp_refine_begin(
   }
   HIP_CHECK(hipMalloc(&e->d_weights, sizeof(i64) * k));
   HIP_CHECK(hipMalloc(&e->d_maxw, sizeof(i64) * k));
+  if (e->d_l_off) {
+    HIP_CHECK(hipFree(e->d_l_off));
+  }
+  if (e->d_l_gains) {
+    HIP_CHECK(hipFree(e->d_l_gains));
+  }
+  e->l_cap = (1u << 22) / k;
+  if (e->l_cap > e->C) {
+    e->l_cap = e->C;
+  }
+  if (e->l_cap < 1024) {
+    e->l_cap = 1024;
+  }
+  HIP_CHECK(hipMalloc(&e->d_l_off, sizeof(u32) * (e->l_cap + 1)));
+  HIP_CHECK(hipMalloc(&e->d_l_gains, sizeof(i32) * e->l_cap * k));
+  HIP_CHECK(hipMemsetAsync(e->d_l_gains, 0, sizeof(i32) * e->l_cap * k, e->stream));
   HIP_CHECK(hipMemcpy(e->d_maxw, max_block_weights, sizeof(i64) * k, hipMemcpyHostToDevice));
   HIP_CHECK(hipMemcpy(e->d_labels, partition, sizeof(u32) * e->n, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpyAsync(
+      e->d_labels0, e->d_labels, sizeof(u32) * e->n, hipMemcpyDeviceToDevice, e->stream
+  ));
   HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
@@ -1113,18 +1318,41 @@ i64 kmp_lp_phase_a(
     );
     LAUNCH_CHECK();
   }
-  // L: grid-stride workgroups over the (rare) high-degree list
+  // L: slice-parallel accumulation over the (rare) high-degree list
   {
-    const size_t lds =
-        ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
-        16 * sizeof(i64);
-    auto *kern = e->has_adjwgt ? k_phase_l<false> : k_phase_l<true>;
     hipLaunchKernelGGL(
-        kern, dim3(2048), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
-        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
-        e->d_l_count, e->d_slots
+        k_l_prep, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj, e->l_cap,
+        e->d_l_off
     );
     LAUNCH_CHECK();
+    const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
+    {
+      auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+      hipLaunchKernelGGL(
+          kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
+          e->d_adjwgt, e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+      );
+      LAUNCH_CHECK();
+    }
+    hipLaunchKernelGGL(
+        k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->k,
+        e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
+        e->l_cap, e->d_l_gains, e->d_slots
+    );
+    LAUNCH_CHECK();
+    // pathological overflow beyond l_cap: direct per-vertex workgroups
+    {
+      const size_t lds =
+          ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
+          16 * sizeof(i64);
+      auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
+      hipLaunchKernelGGL(
+          kern, dim3(512), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+          e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
+          e->d_l_count, e->l_cap, e->d_slots
+      );
+      LAUNCH_CHECK();
+    }
   }
   // compact valid slots in position order (stable select)
   {
@@ -1297,6 +1525,77 @@ i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {
     stats->edge_cut = static_cast<i64>(cut2 / 2);
   }
   return static_cast<i64>(cut2 / 2);
+}
+
+// Reset the engine to the initial partition of the last refine_begin, fully
+// on-device (no host transfers): labels, weights, active flags, counters.
+int kmp_lp_reset(kmp_lp_t *e) {
+  HIP_CHECK(hipMemcpyAsync(
+      e->d_labels, e->d_labels0, sizeof(u32) * e->n, hipMemcpyDeviceToDevice, e->stream
+  ));
+  HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * e->k, e->stream));
+  e->phase_a_ms = 0.0;
+  e->commit_ms = 0.0;
+  e->ev_used = 0;
+  {
+    const u32 threads = 256;
+    const size_t lds = static_cast<size_t>(e->k) * sizeof(unsigned long long);
+    hipLaunchKernelGGL(
+        k_init_weights, dim3(2048), dim3(threads), lds, e->stream, e->n, e->k, e->d_labels,
+        e->d_vwgt, reinterpret_cast<unsigned long long *>(e->d_weights)
+    );
+    LAUNCH_CHECK();
+  }
+  return 0;
+}
+
+// Run the LP sweeps on the current device state (the timed region: all data
+// resident in HBM, no host transfers besides the per-chunk control syncs).
+// Returns total committed moves.
+i64 kmp_lp_run_sweeps(kmp_lp_t *e, int iters) {
+  u64 total_moves = 0;
+  for (int iter = 0; iter < iters; ++iter) {
+    u64 sweep_moves = 0;
+    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
+      const u32 pos_lo = chunk * e->C;
+      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
+      if (pos_lo >= pos_hi) {
+        continue;
+      }
+      const i64 cnt = kmp_lp_phase_a(e, iter, chunk, pos_lo, pos_hi, e->d_props, e->C);
+      if (cnt < 0) {
+        return -1;
+      }
+      const i64 mv = kmp_lp_commit(e, iter, chunk, e->d_props, static_cast<u32>(cnt));
+      if (mv < 0) {
+        return -1;
+      }
+      sweep_moves += mv;
+    }
+    total_moves += sweep_moves;
+    if (sweep_moves == 0) {
+      break;
+    }
+  }
+  return static_cast<i64>(total_moves);
+}
+
+// Light stats (no cut kernel, no label download).
+int kmp_lp_get_stats(kmp_lp_t *e, kmp_lp_stats_t *stats) {
+  unsigned long long arcs = 0, moves = 0;
+  HIP_CHECK(hipMemcpyAsync(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpyAsync(&moves, e->d_moves, sizeof(moves), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  stats->arcs_scanned = arcs;
+  stats->moves = moves;
+  stats->phase_a_ns = static_cast<u64>(e->phase_a_ms * 1e6);
+  stats->total_ns = static_cast<u64>(e->commit_ms * 1e6);
+  stats->num_clusters = 0;
+  stats->edge_cut = -1;
+  return 0;
 }
 
 i64 kmp_lp_refine(
