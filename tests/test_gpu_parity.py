@@ -145,3 +145,26 @@ def test_refine_fullsize_properties():
     assert cut < cut0
     assert np.bincount(part, minlength=k).max() <= mbw[0]
     assert stats.arcs_scanned > 0 and stats.moves > 0
+
+
+def test_refine_parity_fullsize_golden():
+    """Bit-parity with the oracle at R-MAT scale-18/20 (committed oracle
+    outputs; the oracle itself is too slow to run in the GPU test)."""
+    import json
+    import os
+
+    golden = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                         "oracle_fullsize.json")))
+    for name, case in golden.items():
+        g = ka.Graph.rmat(case["scale"], 8, seed=case["gseed"])
+        part0 = ka.random_partition(g.n, case["k"], seed=case["pseed"])
+        mbw = np.full(case["k"], case["mbw"], dtype=np.int64)
+        eng = ka.LpEngine(g)
+        cut, part, stats = eng.refine(case["k"], mbw, part0, seed=case["seed"], iters=5)
+        assert cut == case["cut"], (name, cut, case["cut"])
+        assert stats.arcs_scanned == case["arcs"]
+        assert stats.moves == case["moves"]
+        assert int(part.astype(np.int64).sum()) == case["label_sum"]
+        xor = int(np.bitwise_xor.reduce(part.astype(np.uint64)
+                                        * (np.arange(g.n, dtype=np.uint64) + 1)))
+        assert xor == case["label_xor"], name
