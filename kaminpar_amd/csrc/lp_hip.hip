@@ -123,6 +123,15 @@ __device__ inline bool accept_refine(
   return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw)) || (c == cur);
 }
 
+// Weight-acceptance predicate (clusterer variant, lp_clusterer.cc:199-204).
+__device__ inline bool accept_cluster(u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw_uniform) {
+  return (cw + u_w <= maxw_uniform) || (c == cur);
+}
+
+__device__ inline u32 hash_u32(u32 c) {
+  return static_cast<u32>(kmp::splitmix64(c));
+}
+
 // ------------------------------------------------------------ S path
 // Covers EVERY position of the slice: 4 positions per wave, 16 lanes each.
 // Owns the slot for: tail positions (u >= n), inactive or degree-filtered
@@ -656,6 +665,822 @@ __global__ void k_phase_l_direct(
   }
 }
 
+// ==================== clustering (clusters = vertices) ====================
+// LP clusterer instantiation (lp_clusterer.cc:23-28): ClusterID space = n,
+// clusters start as singletons, uniform weight cap, favored-cluster tracking
+// for two-hop merging. Gain maps are hash-based (cluster ids are unbounded):
+// deg <= 16 register waterfall, deg <= 512 per-wave LDS hash, larger rows
+// slice-parallel into a pooled global hash.
+
+constexpr u32 kClusterMidDeg = 512; // LDS-hash path bound
+constexpr u32 kHashSlots = 1024;    // per-wave LDS hash (8 KB per wave)
+
+// S path for clustering: identical structure to k_phase_s, clusterer accept
+// + favored-cluster argmax over all candidates.
+__global__ void k_phase_s_c(
+    u32 pos_lo,
+    u32 pos_hi,
+    u32 chunk_base,
+    u32 n,
+    u64 iter_seed,
+    u32 max_degree,
+    i64 maxw_uniform,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ vwgt,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const uint8_t *__restrict__ active,
+    u32 *__restrict__ favored,
+    Prop *__restrict__ slots,
+    u64 *__restrict__ l_list, // deg > kClusterMidDeg
+    u32 *__restrict__ l_count
+) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 sub = lane >> 4;
+  const u32 slot = lane & 15;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 p = pos_lo + wave_id * 4 + sub;
+  if (p >= pos_hi) {
+    return;
+  }
+
+  const BlockPerm perm(n, iter_seed);
+  const u32 u = perm(p);
+  const u32 sidx = p - pos_lo;
+
+  bool emit_invalid = false;
+  u32 row = 0, deg = 0;
+  if (u >= n) {
+    emit_invalid = true;
+  } else {
+    row = xadj[u];
+    deg = xadj[u + 1] - row;
+    if (!active[u] || deg > max_degree) {
+      emit_invalid = true;
+    }
+  }
+
+  // L list append (deg > kClusterMidDeg), wave-aggregated
+  {
+    const bool is_l = !emit_invalid && u < n && deg > kClusterMidDeg && slot == 0;
+    const unsigned long long ll = __ballot(is_l);
+    if (ll) {
+      const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
+      u32 bbase = 0;
+      if (lane == leader) {
+        bbase = atomicAdd(l_count, static_cast<u32>(__popcll(ll)));
+      }
+      bbase = __shfl(bbase, leader, kWave);
+      if (is_l) {
+        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] = (static_cast<u64>(p) << 32) | u;
+      }
+    }
+  }
+  if (!emit_invalid && deg > kSmallDeg) {
+    return; // M/L owns this slot
+  }
+  if (emit_invalid) {
+    if (slot == 0) {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
+    return;
+  }
+
+  u32 c = kInvalid;
+  i32 w = 0;
+  if (slot < deg) {
+    const u32 v = adjncy[row + slot];
+    c = labels[v];
+    w = adjwgt ? adjwgt[row + slot] : 1;
+  }
+
+  i32 gain = w;
+  bool owner = (slot < deg);
+  const u32 base = sub * 16;
+  for (u32 j = 0; j < 16; ++j) {
+    const u32 cj = __shfl(c, base + j, kWave);
+    const i32 wj = __shfl(w, base + j, kWave);
+    if (j != slot && c != kInvalid && cj == c) {
+      gain += wj;
+      if (j < slot) {
+        owner = false;
+      }
+    }
+  }
+
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+
+  BestState best{0, 0, 0, false};
+  BestState fav{0, 0, 0, false};
+  if (owner && c != kInvalid) {
+    const u64 h = tie_hash(iter_seed, u, c);
+    fav = BestState{gain, h, c, true};
+    if (accept_cluster(c, cur, u_w, weights[c], maxw_uniform)) {
+      best = BestState{gain, h, c, true};
+    }
+  }
+
+  for (int off = 8; off > 0; off >>= 1) {
+    {
+      const i32 og = __shfl_down(best.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+      const u32 oc = __shfl_down(best.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+      if ((slot + off) < 16 && ohave && key_better(og, oh, oc, best)) {
+        best = BestState{og, oh, oc, true};
+      }
+    }
+    {
+      const i32 og = __shfl_down(fav.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(fav.h), off, kWave);
+      const u32 oc = __shfl_down(fav.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(fav.have), off, kWave);
+      if ((slot + off) < 16 && ohave && key_better(og, oh, oc, fav)) {
+        fav = BestState{og, oh, oc, true};
+      }
+    }
+  }
+
+  if (slot == 0) {
+    // favored cluster for two-hop (label_propagation.h:517-535): stored only
+    // while the vertex still sits in a light singleton cluster
+    const bool store_favored = (u_w == cur_w) && (cur_w <= maxw_uniform / 2);
+    if (store_favored) {
+      favored[u] = fav.have ? fav.c : cur;
+    }
+    if (best.have && best.c != cur) {
+      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
+  }
+}
+
+// M path for clustering: one wave per position, per-wave LDS hash
+// (kHashSlots key/value pairs), deg in (kSmallDeg, kClusterMidDeg].
+template <bool kUnitWeights>
+__global__ void k_phase_m_c(
+    u32 pos_lo,
+    u32 pos_hi,
+    u32 chunk_base,
+    u32 n,
+    u64 iter_seed,
+    u32 max_degree,
+    i64 maxw_uniform,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ vwgt,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const uint8_t *__restrict__ active,
+    u32 *__restrict__ favored,
+    Prop *__restrict__ slots
+) {
+  extern __shared__ u32 ldsu[];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wave_in_wg = threadIdx.x >> 6;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 p = pos_lo + wave_id;
+  if (p >= pos_hi) {
+    return;
+  }
+
+  const BlockPerm perm(n, iter_seed);
+  const u32 u = perm(p);
+  if (u >= n) {
+    return;
+  }
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+  if (deg <= kSmallDeg || deg > kClusterMidDeg || deg > max_degree || !active[u]) {
+    return;
+  }
+
+  u32 *hkeys = ldsu + wave_in_wg * 2 * kHashSlots;
+  i32 *hvals = reinterpret_cast<i32 *>(hkeys + kHashSlots);
+  for (u32 s = lane; s < kHashSlots; s += kWave) {
+    hkeys[s] = kInvalid;
+    hvals[s] = 0;
+  }
+  __threadfence_block();
+
+  for (u32 e = lane; e < deg; e += kWave) {
+    const u32 v = adjncy[row + e];
+    const u32 c = labels[v];
+    const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+    u32 s = hash_u32(c) & (kHashSlots - 1);
+    while (true) {
+      const u32 kcur = hkeys[s];
+      if (kcur == c) {
+        atomicAdd(&hvals[s], w);
+        break;
+      }
+      if (kcur == kInvalid) {
+        const u32 old = atomicCAS(&hkeys[s], kInvalid, c);
+        if (old == kInvalid || old == c) {
+          atomicAdd(&hvals[s], w);
+          break;
+        }
+      }
+      s = (s + 1) & (kHashSlots - 1);
+    }
+  }
+  __threadfence_block();
+
+  const u32 cur = labels[u];
+  const i32 u_w = vwgt ? vwgt[u] : 1;
+  const i64 cur_w = weights[cur];
+
+  BestState best{0, 0, 0, false};
+  BestState fav{0, 0, 0, false};
+  for (u32 s = lane; s < kHashSlots; s += kWave) {
+    const u32 c = hkeys[s];
+    if (c == kInvalid) {
+      continue;
+    }
+    const i32 g = hvals[s];
+    if (g <= 0) {
+      continue;
+    }
+    const u64 h = tie_hash(iter_seed, u, c);
+    if (key_better(g, h, c, fav)) {
+      fav = BestState{g, h, c, true};
+    }
+    if (accept_cluster(c, cur, u_w, weights[c], maxw_uniform) && key_better(g, h, c, best)) {
+      best = BestState{g, h, c, true};
+    }
+  }
+
+  for (int off = 32; off > 0; off >>= 1) {
+    {
+      const i32 og = __shfl_down(best.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+      const u32 oc = __shfl_down(best.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+      if (ohave && key_better(og, oh, oc, best)) {
+        best = BestState{og, oh, oc, true};
+      }
+    }
+    {
+      const i32 og = __shfl_down(fav.gain, off, kWave);
+      const u64 oh = __shfl_down(static_cast<unsigned long long>(fav.h), off, kWave);
+      const u32 oc = __shfl_down(fav.c, off, kWave);
+      const int ohave = __shfl_down(static_cast<int>(fav.have), off, kWave);
+      if (ohave && key_better(og, oh, oc, fav)) {
+        fav = BestState{og, oh, oc, true};
+      }
+    }
+  }
+
+  if (lane == 0) {
+    const bool store_favored = (u_w == cur_w) && (cur_w <= maxw_uniform / 2);
+    if (store_favored) {
+      favored[u] = fav.have ? fav.c : cur;
+    }
+    const u32 sidx = p - pos_lo;
+    if (best.have && best.c != cur) {
+      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0, kInvalid, 0, 0};
+    }
+  }
+}
+
+// L prep for clustering: per-vertex slice prefix AND pooled-hash region
+// prefix (region size = next power of two >= 2*deg). Sets *pool_overflow if
+// the pool cannot hold all regions (caller aborts with a clear message).
+__global__ void k_l_prep_c(
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    const u32 *__restrict__ xadj,
+    u32 l_cap,
+    u64 pool_slots,
+    u32 *__restrict__ l_off,  // slice prefix
+    u64 *__restrict__ l_hoff, // hash-region prefix
+    u32 *__restrict__ l_hbits, // log2(region size)
+    int *__restrict__ pool_overflow
+) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    const u32 count = *l_count < l_cap ? *l_count : l_cap;
+    u32 sacc = 0;
+    u64 hacc = 0;
+    for (u32 i = 0; i < count; ++i) {
+      l_off[i] = sacc;
+      l_hoff[i] = hacc;
+      const u32 u = static_cast<u32>(l_list[i]);
+      const u32 deg = xadj[u + 1] - xadj[u];
+      sacc += (deg + kLSlice - 1) / kLSlice;
+      u32 bits = 11; // >= 2048 slots
+      while ((1u << bits) < 2 * deg) {
+        ++bits;
+      }
+      l_hbits[i] = bits;
+      hacc += 1ull << bits;
+    }
+    l_off[count] = sacc;
+    l_hoff[count] = hacc;
+    if (hacc > pool_slots) {
+      *pool_overflow = 1;
+    }
+  }
+}
+
+// Slice-parallel accumulation into the pooled per-vertex global hash.
+template <bool kUnitWeights>
+__global__ void k_phase_l_acc_c(
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ labels,
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    u32 l_cap,
+    const u32 *__restrict__ l_off,
+    const u64 *__restrict__ l_hoff,
+    const u32 *__restrict__ l_hbits,
+    u32 *__restrict__ pool_keys,
+    i32 *__restrict__ pool_vals
+) {
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  if (count == 0) {
+    return;
+  }
+  const u32 total = l_off[count];
+  for (u32 s = blockIdx.x; s < total; s += gridDim.x) {
+    u32 lo = 0, hi = count - 1;
+    while (lo < hi) {
+      const u32 mid = (lo + hi + 1) >> 1;
+      if (l_off[mid] <= s) {
+        lo = mid;
+      } else {
+        hi = mid - 1;
+      }
+    }
+    const u32 vid = lo;
+    const u32 u = static_cast<u32>(l_list[vid]);
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+    const u32 e_lo = (s - l_off[vid]) * kLSlice;
+    const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
+    u32 *hk = pool_keys + l_hoff[vid];
+    i32 *hv = pool_vals + l_hoff[vid];
+    const u32 mask = (1u << l_hbits[vid]) - 1;
+
+    for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
+      const u32 v = adjncy[row + e];
+      const u32 c = labels[v];
+      const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+      u32 slot = hash_u32(c) & mask;
+      while (true) {
+        const u32 kcur = hk[slot];
+        if (kcur == c) {
+          atomicAdd(&hv[slot], w);
+          break;
+        }
+        if (kcur == kInvalid) {
+          const u32 old = atomicCAS(&hk[slot], kInvalid, c);
+          if (old == kInvalid || old == c) {
+            atomicAdd(&hv[slot], w);
+            break;
+          }
+        }
+        slot = (slot + 1) & mask;
+      }
+    }
+  }
+}
+
+// Selection per clustering L vertex: scan its hash region, clear it for the
+// next chunk, write slot + favored.
+__global__ void k_phase_l_sel_c(
+    u32 pos_lo,
+    u32 chunk_base,
+    u64 iter_seed,
+    u32 n,
+    i64 maxw_uniform,
+    const i32 *__restrict__ vwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    u32 l_cap,
+    const u64 *__restrict__ l_hoff,
+    const u32 *__restrict__ l_hbits,
+    u32 *__restrict__ pool_keys,
+    i32 *__restrict__ pool_vals,
+    u32 *__restrict__ favored,
+    Prop *__restrict__ slots
+) {
+  __shared__ i64 red[24];
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+    const u64 rec = l_list[vid];
+    const u32 p = static_cast<u32>(rec >> 32);
+    const u32 u = static_cast<u32>(rec);
+    const u32 cur = labels[u];
+    const i32 u_w = vwgt ? vwgt[u] : 1;
+    const i64 cur_w = weights[cur];
+    u32 *hk = pool_keys + l_hoff[vid];
+    i32 *hv = pool_vals + l_hoff[vid];
+    const u32 slots_n = 1u << l_hbits[vid];
+
+    BestState best{0, 0, 0, false};
+    BestState fav{0, 0, 0, false};
+    for (u32 s = threadIdx.x; s < slots_n; s += blockDim.x) {
+      const u32 c = hk[s];
+      if (c == kInvalid) {
+        continue;
+      }
+      const i32 g = hv[s];
+      hk[s] = kInvalid; // clear for the next chunk
+      hv[s] = 0;
+      if (g <= 0) {
+        continue;
+      }
+      const u64 h = tie_hash(iter_seed, u, c);
+      if (key_better(g, h, c, fav)) {
+        fav = BestState{g, h, c, true};
+      }
+      if (accept_cluster(c, cur, u_w, weights[c], maxw_uniform) && key_better(g, h, c, best)) {
+        best = BestState{g, h, c, true};
+      }
+    }
+    const u32 lane = threadIdx.x & (kWave - 1);
+    for (int off = 32; off > 0; off >>= 1) {
+      {
+        const i32 og = __shfl_down(best.gain, off, kWave);
+        const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+        const u32 oc = __shfl_down(best.c, off, kWave);
+        const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+        if (ohave && key_better(og, oh, oc, best)) {
+          best = BestState{og, oh, oc, true};
+        }
+      }
+      {
+        const i32 og = __shfl_down(fav.gain, off, kWave);
+        const u64 oh = __shfl_down(static_cast<unsigned long long>(fav.h), off, kWave);
+        const u32 oc = __shfl_down(fav.c, off, kWave);
+        const int ohave = __shfl_down(static_cast<int>(fav.have), off, kWave);
+        if (ohave && key_better(og, oh, oc, fav)) {
+          fav = BestState{og, oh, oc, true};
+        }
+      }
+    }
+    const u32 wave_in_wg = threadIdx.x >> 6;
+    if (lane == 0) {
+      red[wave_in_wg * 3] = (static_cast<i64>(best.gain) << 2) | (best.have ? 1 : 0) |
+                            (fav.have ? 2 : 0);
+      red[wave_in_wg * 3 + 1] = static_cast<i64>(best.h);
+      red[wave_in_wg * 3 + 2] = static_cast<i64>(fav.h);
+      reinterpret_cast<u32 *>(red + 12)[wave_in_wg * 2] = best.c;
+      reinterpret_cast<u32 *>(red + 12)[wave_in_wg * 2 + 1] = fav.c;
+      reinterpret_cast<i32 *>(red + 16)[wave_in_wg] = fav.gain;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      BestState tb{0, 0, 0, false};
+      BestState tf{0, 0, 0, false};
+      const u32 waves = blockDim.x >> 6;
+      for (u32 wv = 0; wv < waves; ++wv) {
+        const i64 packed = red[wv * 3];
+        const i32 bg = static_cast<i32>(packed >> 2);
+        if (packed & 1) {
+          const u64 h = static_cast<u64>(red[wv * 3 + 1]);
+          const u32 c = reinterpret_cast<u32 *>(red + 12)[wv * 2];
+          if (key_better(bg, h, c, tb)) {
+            tb = BestState{bg, h, c, true};
+          }
+        }
+        if (packed & 2) {
+          const i32 fg = reinterpret_cast<i32 *>(red + 16)[wv];
+          const u64 h = static_cast<u64>(red[wv * 3 + 2]);
+          const u32 c = reinterpret_cast<u32 *>(red + 12)[wv * 2 + 1];
+          if (key_better(fg, h, c, tf)) {
+            tf = BestState{fg, h, c, true};
+          }
+        }
+      }
+      const bool store_favored = (u_w == cur_w) && (cur_w <= maxw_uniform / 2);
+      if (store_favored) {
+        favored[u] = tf.have ? tf.c : cur;
+      }
+      if (tb.have && tb.c != cur) {
+        slots[p - pos_lo] = Prop{u, tb.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// -------- big-k commit variants (cluster space = n) --------
+
+// Reset dep entries touched by this chunk's proposals (idempotent).
+__global__ void k_dep_reset_touched(
+    const Prop *__restrict__ props,
+    u32 count,
+    const u32 *__restrict__ labels,
+    unsigned long long *__restrict__ dep
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    dep[labels[props[i].u]] = 0;
+  }
+}
+
+// Departures via direct global atomics (sources spread over many clusters).
+__global__ void k_dep_direct(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ labels,
+    unsigned long long *__restrict__ dep
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  if (i - seg_begin[to] < prefix_len[to]) {
+    const Prop pr = props[order[i]];
+    atomicAdd(&dep[labels[pr.u]], static_cast<unsigned long long>(pr.w));
+  }
+}
+
+// Head-driven cutoff: one thread per SEGMENT HEAD in the sorted proposals
+// (avoids an O(num_clusters) sweep per fixpoint round).
+__global__ void k_cutoff_heads(
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len,
+    const i64 *__restrict__ pw,
+    const i64 *__restrict__ weights,
+    i64 maxw_uniform,
+    const unsigned long long *__restrict__ dep,
+    int *__restrict__ changed
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 c = sto[i];
+  const u32 b = seg_begin[c];
+  if (i != b) {
+    return; // not the segment head
+  }
+  const i64 capacity = maxw_uniform - weights[c] + static_cast<i64>(dep[c]);
+  const u32 old_len = prefix_len[c];
+  u32 lo = 0, hi = old_len;
+  while (lo < hi) {
+    const u32 mid = (lo + hi + 1) >> 1;
+    if (pw[b + mid - 1] <= capacity) {
+      lo = mid;
+    } else {
+      hi = mid - 1;
+    }
+  }
+  if (lo < old_len) {
+    prefix_len[c] = lo;
+    atomicExch(changed, 1);
+  }
+}
+
+__global__ void k_seg_len_heads(
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ seg_end,
+    u32 *__restrict__ prefix_len
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 c = sto[i];
+  if (i == seg_begin[c]) {
+    prefix_len[c] = seg_end[c] - seg_begin[c];
+  }
+}
+
+// Weight updates for big-k: arrivals head-driven, departures via a
+// once-per-cluster atomic exchange of the final dep value.
+__global__ void k_weights_update_big(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ prefix_len,
+    const i64 *__restrict__ pw,
+    const u32 *__restrict__ labels,
+    unsigned long long *__restrict__ dep,
+    i64 *__restrict__ weights
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  const u32 b = seg_begin[to];
+  if (i == b && prefix_len[to] > 0) {
+    atomicAdd(reinterpret_cast<unsigned long long *>(&weights[to]),
+              static_cast<unsigned long long>(pw[b + prefix_len[to] - 1]));
+  }
+  // departures: first toucher per source cluster applies the whole dep
+  const u32 from = labels[props[order[i]].u];
+  const unsigned long long d = atomicExch(&dep[from], 0ull);
+  if (d) {
+    atomicAdd(reinterpret_cast<unsigned long long *>(&weights[from]),
+              static_cast<unsigned long long>(-static_cast<i64>(d)));
+  }
+}
+
+// Count clusters emptied by this chunk (distinct sources with final weight
+// zero); eflag entries are reset by k_emptied_reset.
+__global__ void k_emptied_count(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ sto,
+    u32 count,
+    const u32 *__restrict__ seg_begin,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    uint8_t *__restrict__ eflag,
+    unsigned long long *__restrict__ emptied
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= count) {
+    return;
+  }
+  const u32 to = sto[i];
+  if (i - seg_begin[to] >= prefix_len[to]) {
+    return;
+  }
+  const u32 from = labels[props[order[i]].u];
+  if (weights[from] == 0) {
+    // no 8-bit atomicExch: use 32-bit CAS on the aligned word
+    u32 *word = reinterpret_cast<u32 *>(reinterpret_cast<uintptr_t>(&eflag[from]) & ~3ull);
+    const u32 shift = (from & 3u) * 8;
+    const u32 bit = 1u << shift;
+    const u32 old = atomicOr(word, bit);
+    if ((old & bit) == 0) {
+      atomicAdd(emptied, 1ull);
+    }
+  }
+}
+
+__global__ void k_emptied_reset(
+    const u32 *__restrict__ order,
+    const Prop *__restrict__ props,
+    u32 count,
+    const u32 *__restrict__ labels,
+    uint8_t *__restrict__ eflag
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    eflag[labels[props[i].u]] = 0;
+  }
+}
+
+// -------- post passes (clusterer) --------
+
+__global__ void k_iota(u32 n, u32 *__restrict__ a) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    a[i] = i;
+  }
+}
+
+__global__ void k_init_cluster_weights(
+    u32 n, const i32 *__restrict__ vwgt, i64 *__restrict__ weights
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    weights[i] = vwgt ? vwgt[i] : 1;
+  }
+}
+
+// Two-hop candidate collection (label_propagation.h:939-975 threadwise
+// considered-set): deg > 0, still in its own singleton cluster at its
+// initial weight, light. Slot-write + stable select keeps ascending-u order.
+__global__ void k_twohop_cand(
+    u32 n,
+    i64 maxw_uniform,
+    const u32 *__restrict__ xadj,
+    const i32 *__restrict__ vwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const u32 *__restrict__ favored,
+    u64 *__restrict__ cand_slots // (favored << 32) | u, or ~0 invalid
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u >= n) {
+    return;
+  }
+  u64 out = ~0ull;
+  const u32 deg = xadj[u + 1] - xadj[u];
+  if (deg > 0 && labels[u] == u) {
+    const i64 w = weights[u];
+    const i64 vw = vwgt ? vwgt[u] : 1;
+    if (w == vw && w <= maxw_uniform / 2) {
+      out = (static_cast<u64>(favored[u]) << 32) | u;
+    }
+  }
+  cand_slots[u] = out;
+}
+
+struct CandValid {
+  __host__ __device__ bool operator()(const u64 &v) const { return v != ~0ull; }
+};
+
+// Pair consecutive candidates with equal favored cluster: even rank joins
+// the preceding candidate's (singleton) cluster. rank = 1-based rank within
+// the favored segment (scan_by_key output). Always fits: both weights
+// <= max/2 (label_propagation.h:977-1002 match semantics).
+__global__ void k_twohop_pair(
+    const u64 *__restrict__ cand_sorted,
+    const u32 *__restrict__ rank, // 1-based within favored segment
+    u32 count,
+    u32 *__restrict__ labels,
+    i64 *__restrict__ weights,
+    unsigned long long *__restrict__ merged
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  u64 local = 0;
+  if (i < count && (rank[i] & 1u) == 0) {
+    const u32 u = static_cast<u32>(cand_sorted[i]);
+    const u32 rep = static_cast<u32>(cand_sorted[i - 1]);
+    labels[u] = rep;
+    // no atomics needed: each candidate appears in exactly one pair
+    weights[rep] += weights[u];
+    weights[u] = 0;
+    local = 1;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    local += __shfl_down(static_cast<unsigned long long>(local), off, kWave);
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0 && local) {
+    atomicAdd(merged, static_cast<unsigned long long>(local));
+  }
+}
+
+__global__ void k_extract_fav(const u64 *__restrict__ cand, u32 count, u32 *__restrict__ fav) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    fav[i] = static_cast<u32>(cand[i] >> 32);
+  }
+}
+
+__global__ void k_ones(u32 count, u32 *__restrict__ a) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    a[i] = 1;
+  }
+}
+
+// Apply host-computed isolated-node matches: pairs (u -> rep).
+__global__ void k_apply_pairs(
+    const u64 *__restrict__ pairs, // (rep << 32) | u
+    u32 count,
+    u32 *__restrict__ labels,
+    i64 *__restrict__ weights
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
+    const u32 u = static_cast<u32>(pairs[i]);
+    const u32 rep = static_cast<u32>(pairs[i] >> 32);
+    labels[u] = rep;
+    weights[rep] += weights[u];
+    weights[u] = 0;
+  }
+}
+
+// Count non-empty clusters.
+__global__ void k_count_nonempty(
+    u32 n, const i64 *__restrict__ weights, unsigned long long *__restrict__ out
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  u64 local = (i < n && weights[i] != 0) ? 1 : 0;
+  for (int off = 32; off > 0; off >>= 1) {
+    local += __shfl_down(static_cast<unsigned long long>(local), off, kWave);
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0 && local) {
+    atomicAdd(out, static_cast<unsigned long long>(local));
+  }
+}
+
 // -------------------------------------------------------------- commit
 __global__ void k_make_keys(
     const Prop *__restrict__ props, u32 count, u32 *__restrict__ keys, u32 *__restrict__ vals
@@ -1042,10 +1867,37 @@ struct kmp_lp_t {
   u32 *d_admitted_flags = nullptr;
   unsigned long long *d_cut = nullptr;
 
+  // clustering state
+  u32 *d_favored = nullptr;      // n (two-hop favored clusters)
+  uint8_t *d_eflag = nullptr;    // n (emptied-cluster flags)
+  u32 *d_pool_keys = nullptr;    // pooled L hash
+  i32 *d_pool_vals = nullptr;
+  u64 pool_slots = 0;
+  u64 *d_l_hoff = nullptr;       // l_cap+1
+  u32 *d_l_hbits = nullptr;      // l_cap
+  int *d_pool_overflow = nullptr;
+  u64 *d_cand = nullptr;         // n (two-hop candidate slots)
+  u64 *d_cand2 = nullptr;        // n (sorted)
+  u32 *d_cfav = nullptr;         // n
+  u32 *d_crank = nullptr;        // n
+  u32 *d_cones = nullptr;        // n
+  void *d_cand_select_temp = nullptr;
+  size_t cand_select_temp_bytes = 0;
+  void *d_cand_sort_temp = nullptr;
+  size_t cand_sort_temp_bytes = 0;
+  void *d_cand_scan_temp = nullptr;
+  size_t cand_scan_temp_bytes = 0;
+  unsigned long long *d_emptied = nullptr;
+  std::vector<u32> isolated;   // host: vertices with degree 0
+  std::vector<i32> iso_weights; // their node weights
+  bool clusterer = false;
+  i64 maxw_uniform = 0;
+
   // pinned host mirrors
   u32 *h_count = nullptr;
   int *h_changed = nullptr;
-  unsigned long long *h_moves = nullptr; // [0]=before [1]=after
+  unsigned long long *h_moves = nullptr; // [0..1]=moves before/after [2..3]=emptied
+  u64 last_emptied = 0;
 
   hipStream_t stream = nullptr;
 
@@ -1175,9 +2027,19 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   ));
   HIP_CHECK(hipMalloc(&e->d_select_temp, e->select_temp_bytes));
 
+  {
+    const u32 *xadj = kmp_graph_xadj(g);
+    const i32 *vwgt = kmp_graph_vwgt(g);
+    for (u32 u = 0; u < e->n; ++u) {
+      if (xadj[u + 1] == xadj[u]) {
+        e->isolated.push_back(u);
+        e->iso_weights.push_back(vwgt ? vwgt[u] : 1);
+      }
+    }
+  }
   HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
   HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
-  HIP_CHECK(hipHostMalloc(&e->h_moves, sizeof(unsigned long long) * 2));
+  HIP_CHECK(hipHostMalloc(&e->h_moves, sizeof(unsigned long long) * 4));
   return e;
 }
 
@@ -1199,6 +2061,16 @@ void kmp_lp_free(kmp_lp_t *e) {
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
                   (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_flags,
                   (void *)e->d_cut}) {
+    if (p) {
+      (void)hipFree(p);
+    }
+  }
+  for (void *p : {(void *)e->d_favored, (void *)e->d_eflag, (void *)e->d_pool_keys,
+                  (void *)e->d_pool_vals, (void *)e->d_l_hoff, (void *)e->d_l_hbits,
+                  (void *)e->d_pool_overflow, (void *)e->d_cand, (void *)e->d_cand2,
+                  (void *)e->d_cfav, (void *)e->d_crank, (void *)e->d_cones,
+                  (void *)e->d_cand_select_temp, (void *)e->d_cand_sort_temp,
+                  (void *)e->d_cand_scan_temp, (void *)e->d_emptied}) {
     if (p) {
       (void)hipFree(p);
     }
@@ -1227,6 +2099,7 @@ int kmp_lp_refine_begin(
   }
   e->k = k;
   e->seed = seed;
+  e->clusterer = false;
   e->phase_a_ms = 0.0;
   e->commit_ms = 0.0;
   e->ev_used = 0;
@@ -1297,59 +2170,103 @@ i64 kmp_lp_phase_a(
   e->ev_pair(ev0, ev1);
   HIP_CHECK(hipEventRecord(ev0, e->stream));
 
-  // S: 4 positions/wave (default owner of every position's slot)
-  hipLaunchKernelGGL(
-      k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
-      dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->d_xadj,
-      e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-      e->d_slots, e->d_l_list, e->d_l_count
-  );
-  LAUNCH_CHECK();
-  // M: one wave per position
-  {
-    const size_t lds =
-        static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
-    auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
+  if (!e->clusterer) {
+    // S: 4 positions/wave (default owner of every position's slot)
     hipLaunchKernelGGL(
-        kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
-        e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
-        e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-        e->d_slots
+        k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+        dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
+        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
+        e->d_active, e->d_slots, e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
-  }
-  // L: slice-parallel accumulation over the (rare) high-degree list
-  {
-    hipLaunchKernelGGL(
-        k_l_prep, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj, e->l_cap,
-        e->d_l_off
-    );
-    LAUNCH_CHECK();
-    const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
+    // M: one wave per position
     {
-      auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+      const size_t lds =
+          static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
+      auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
       hipLaunchKernelGGL(
-          kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
-          e->d_adjwgt, e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+          kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
+          e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
+          e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
+          e->d_slots
       );
       LAUNCH_CHECK();
     }
+    // L: slice-parallel accumulation over the (rare) high-degree list
+    {
+      hipLaunchKernelGGL(
+          k_l_prep, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
+          e->l_cap, e->d_l_off
+      );
+      LAUNCH_CHECK();
+      const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
+      {
+        auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+        hipLaunchKernelGGL(
+            kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
+            e->d_adjwgt, e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+        );
+        LAUNCH_CHECK();
+      }
+      hipLaunchKernelGGL(
+          k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->k,
+          e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
+          e->l_cap, e->d_l_gains, e->d_slots
+      );
+      LAUNCH_CHECK();
+      {
+        const size_t lds =
+            ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
+            16 * sizeof(i64);
+        auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
+        hipLaunchKernelGGL(
+            kern, dim3(512), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+            e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
+            e->d_l_count, e->l_cap, e->d_slots
+        );
+        LAUNCH_CHECK();
+      }
+    }
+  } else {
+    // clustering: hash-based gain maps, favored-cluster tracking
     hipLaunchKernelGGL(
-        k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->k,
-        e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
-        e->l_cap, e->d_l_gains, e->d_slots
+        k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+        dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
+        e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
+        e->d_active, e->d_favored, e->d_slots, e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
-    // pathological overflow beyond l_cap: direct per-vertex workgroups
     {
-      const size_t lds =
-          ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
-          16 * sizeof(i64);
-      auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
+      const size_t lds = static_cast<size_t>(threads / kWave) * 2 * kHashSlots * sizeof(u32);
+      auto *kern = e->has_adjwgt ? k_phase_m_c<false> : k_phase_m_c<true>;
       hipLaunchKernelGGL(
-          kern, dim3(512), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
-          e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
-          e->d_l_count, e->l_cap, e->d_slots
+          kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
+          e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->maxw_uniform,
+          e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_active,
+          e->d_favored, e->d_slots
+      );
+      LAUNCH_CHECK();
+    }
+    {
+      hipLaunchKernelGGL(
+          k_l_prep_c, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
+          e->l_cap, e->pool_slots, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_pool_overflow
+      );
+      LAUNCH_CHECK();
+      {
+        auto *kern = e->has_adjwgt ? k_phase_l_acc_c<false> : k_phase_l_acc_c<true>;
+        hipLaunchKernelGGL(
+            kern, dim3(2048), dim3(256), 0, e->stream, e->d_xadj, e->d_adjncy, e->d_adjwgt,
+            e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_hoff,
+            e->d_l_hbits, e->d_pool_keys, e->d_pool_vals
+        );
+        LAUNCH_CHECK();
+      }
+      hipLaunchKernelGGL(
+          k_phase_l_sel_c, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->n,
+          e->maxw_uniform, e->d_vwgt, e->d_labels, e->d_weights, e->d_l_list, e->d_l_count,
+          e->l_cap, e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_favored,
+          e->d_slots
       );
       LAUNCH_CHECK();
     }
@@ -1366,11 +2283,20 @@ i64 kmp_lp_phase_a(
   HIP_CHECK(
       hipMemcpyAsync(e->h_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream)
   );
+  if (e->clusterer) {
+    HIP_CHECK(hipMemcpyAsync(
+        e->h_changed, e->d_pool_overflow, sizeof(int), hipMemcpyDeviceToHost, e->stream
+    ));
+  }
   HIP_CHECK(hipStreamSynchronize(e->stream));
   float ms = 0;
   HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
   e->phase_a_ms += ms;
 
+  if (e->clusterer && *e->h_changed) {
+    fprintf(stderr, "kaminpar_amd: clustering L hash pool overflow (raise pool size)\n");
+    return -1;
+  }
   if (*e->h_count > cap) {
     fprintf(stderr, "kaminpar_amd: proposal buffer overflow (%u > %u)\n", *e->h_count, cap);
     return -1;
@@ -1381,14 +2307,19 @@ i64 kmp_lp_phase_a(
 i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 count) {
   const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 threads = 256;
+  const u32 chunk_lo = chunk * e->C;
+  const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
+  const bool big_k = e->clusterer; // cluster space = n
+
   hipEvent_t cev0, cev1;
   e->ev_pair(cev0, cev1);
   HIP_CHECK(hipEventRecord(cev0, e->stream));
-  const u32 chunk_lo = chunk * e->C;
-  const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
-
   HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
                            hipMemcpyDeviceToHost, e->stream));
+  if (big_k) {
+    HIP_CHECK(hipMemcpyAsync(&e->h_moves[2], e->d_emptied, sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, e->stream));
+  }
 
   const Prop *props = static_cast<const Prop *>(d_props);
   u32 *order = nullptr;
@@ -1400,18 +2331,18 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_make_keys, dim3(grid), dim3(threads), 0, e->stream, props, count, e->d_sort_keys[0],
         e->d_sort_vals[0]
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
     rocprim::double_buffer<u32> keys(e->d_sort_keys[0], e->d_sort_keys[1]);
     rocprim::double_buffer<u32> vals(e->d_sort_vals[0], e->d_sort_vals[1]);
     size_t tb = e->sort_temp_bytes;
     HIP_CHECK(rocprim::radix_sort_pairs(e->d_sort_temp, tb, keys, vals, count, 0, 32, e->stream));
     order = vals.current();
-    sto = keys.current(); // sorted target clusters
+    sto = keys.current();
 
     hipLaunchKernelGGL(
         k_extract_w, dim3(grid), dim3(threads), 0, e->stream, order, props, count, e->d_sw
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
     size_t sb = e->scan_temp_bytes;
     HIP_CHECK(rocprim::inclusive_scan_by_key(
         e->d_scan_temp, sb, sto, e->d_sw, e->d_pw, count, rocprim::plus<i64>(),
@@ -1421,35 +2352,55 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_seg_bounds, dim3(grid), dim3(threads), 0, e->stream, sto, count, e->d_seg_begin,
         e->d_seg_end
     );
-  LAUNCH_CHECK();
-    const u32 kgrid = ceil_div(e->k, threads);
+    LAUNCH_CHECK();
     hipLaunchKernelGGL(
-        k_seg_len, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
-        e->d_prefix_len
+        k_seg_len_heads, dim3(grid), dim3(threads), 0, e->stream, sto, count, e->d_seg_begin,
+        e->d_seg_end, e->d_prefix_len
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
 
     // greatest-fixpoint rollback (kaminpar-dist lp_refiner.cc:296-333).
-    // Two rounds are enqueued per host sync (the fixpoint almost always
-    // converges within two); the changed flag of the SECOND round decides.
+    // Two rounds per host sync; the second round's changed flag decides.
+    const u32 kgrid = big_k ? 0 : ceil_div(e->k, threads);
     while (true) {
       for (int half = 0; half < 2; ++half) {
-        hipLaunchKernelGGL(
-            k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep
-        );
-        LAUNCH_CHECK();
-        hipLaunchKernelGGL(
-            k_dep, dim3(grid > 2048 ? 2048 : grid), dim3(threads),
-            static_cast<size_t>(e->k) * sizeof(unsigned long long), e->stream, order, props, sto,
-            count, e->k, e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
-        );
-        LAUNCH_CHECK();
-        HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
-        hipLaunchKernelGGL(
-            k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin, e->d_seg_end,
-            e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep, e->d_changed
-        );
-        LAUNCH_CHECK();
+        if (big_k) {
+          hipLaunchKernelGGL(
+              k_dep_reset_touched, dim3(grid), dim3(threads), 0, e->stream, props, count,
+              e->d_labels, e->d_dep
+          );
+          LAUNCH_CHECK();
+          hipLaunchKernelGGL(
+              k_dep_direct, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count,
+              e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
+          );
+          LAUNCH_CHECK();
+          HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
+          hipLaunchKernelGGL(
+              k_cutoff_heads, dim3(grid), dim3(threads), 0, e->stream, sto, count, e->d_seg_begin,
+              e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_weights, e->maxw_uniform, e->d_dep,
+              e->d_changed
+          );
+          LAUNCH_CHECK();
+        } else {
+          hipLaunchKernelGGL(
+              k_dep_reset_all, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_dep
+          );
+          LAUNCH_CHECK();
+          hipLaunchKernelGGL(
+              k_dep, dim3(grid > 2048 ? 2048 : grid), dim3(threads),
+              static_cast<size_t>(e->k) * sizeof(unsigned long long), e->stream, order, props, sto,
+              count, e->k, e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_dep
+          );
+          LAUNCH_CHECK();
+          HIP_CHECK(hipMemsetAsync(e->d_changed, 0, sizeof(int), e->stream));
+          hipLaunchKernelGGL(
+              k_cutoff, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
+              e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_weights, e->d_maxw, e->d_dep,
+              e->d_changed
+          );
+          LAUNCH_CHECK();
+        }
       }
       HIP_CHECK(
           hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
@@ -1460,16 +2411,34 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
       }
     }
 
-    hipLaunchKernelGGL(
-        k_weights_update, dim3(kgrid), dim3(threads), 0, e->stream, e->k, e->d_seg_begin,
-        e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
-    );
-  LAUNCH_CHECK();
+    if (big_k) {
+      hipLaunchKernelGGL(
+          k_weights_update_big, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count,
+          e->d_seg_begin, e->d_prefix_len, e->d_pw, e->d_labels, e->d_dep, e->d_weights
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_emptied_count, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count,
+          e->d_seg_begin, e->d_prefix_len, e->d_labels, e->d_weights, e->d_eflag, e->d_emptied
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_emptied_reset, dim3(grid), dim3(threads), 0, e->stream, order, props, count,
+          e->d_labels, e->d_eflag
+      );
+      LAUNCH_CHECK();
+    } else {
+      hipLaunchKernelGGL(
+          k_weights_update, dim3(ceil_div(e->k, threads)), dim3(threads), 0, e->stream, e->k,
+          e->d_seg_begin, e->d_seg_end, e->d_prefix_len, e->d_pw, e->d_dep, e->d_weights
+      );
+      LAUNCH_CHECK();
+    }
     hipLaunchKernelGGL(
         k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
         e->d_prefix_len, e->d_labels, e->d_admitted_flags, e->d_moves
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
   }
 
   // clear active for the WHOLE chunk's processed set (identical on all
@@ -1484,21 +2453,26 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
         k_activate, dim3(ceil_div(static_cast<u64>(count) * kWave, threads)), dim3(threads), 0,
         e->stream, order, e->d_admitted_flags, props, count, e->d_xadj, e->d_adjncy, e->d_active
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
     hipLaunchKernelGGL(
         k_reset_segs, dim3(ceil_div(count, threads)), dim3(threads), 0, e->stream, props, count,
         e->d_seg_begin, e->d_seg_end, e->d_prefix_len
     );
-  LAUNCH_CHECK();
+    LAUNCH_CHECK();
   }
 
   HIP_CHECK(hipEventRecord(cev1, e->stream));
   HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
                            hipMemcpyDeviceToHost, e->stream));
+  if (big_k) {
+    HIP_CHECK(hipMemcpyAsync(&e->h_moves[3], e->d_emptied, sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, e->stream));
+  }
   HIP_CHECK(hipStreamSynchronize(e->stream));
   float cms = 0;
   HIP_CHECK(hipEventElapsedTime(&cms, cev0, cev1));
   e->commit_ms += cms;
+  e->last_emptied = big_k ? (e->h_moves[3] - e->h_moves[2]) : 0;
   return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
 }
 
@@ -1639,15 +2613,223 @@ i64 kmp_lp_cluster(
     kmp_lp_t *e, i64 max_cluster_weight, u32 desired_clusters, u32 *clustering, u64 seed,
     int iters, kmp_lp_stats_t *stats
 ) {
-  (void)e;
-  (void)max_cluster_weight;
-  (void)desired_clusters;
-  (void)clustering;
-  (void)seed;
-  (void)iters;
-  (void)stats;
-  fprintf(stderr, "kaminpar_amd: kmp_lp_cluster GPU path not implemented yet\n");
-  return -1;
+  const u32 n = e->n;
+  const u32 threads = 256;
+  e->k = n;
+  e->seed = seed;
+  e->clusterer = true;
+  e->maxw_uniform = max_cluster_weight;
+  e->phase_a_ms = 0.0;
+  e->commit_ms = 0.0;
+  e->ev_used = 0;
+
+  engine_free_k_buffers(e);
+  engine_alloc_k_buffers(e, n);
+  if (e->d_weights) {
+    HIP_CHECK(hipFree(e->d_weights));
+    e->d_weights = nullptr;
+  }
+  HIP_CHECK(hipMalloc(&e->d_weights, sizeof(i64) * n));
+  if (!e->d_favored) {
+    HIP_CHECK(hipMalloc(&e->d_favored, sizeof(u32) * n));
+    HIP_CHECK(hipMalloc(&e->d_eflag, n));
+    HIP_CHECK(hipMemset(e->d_eflag, 0, n));
+    HIP_CHECK(hipMalloc(&e->d_emptied, sizeof(unsigned long long)));
+    HIP_CHECK(hipMalloc(&e->d_pool_overflow, sizeof(int)));
+    HIP_CHECK(hipMalloc(&e->d_l_hoff, sizeof(u64) * (e->C + 1)));
+    HIP_CHECK(hipMalloc(&e->d_l_hbits, sizeof(u32) * e->C));
+    // pooled hash for high-degree rows: ~8 slots per average chunk arc
+    u64 slots = (e->m / kmp::kNumChunks) * 8;
+    if (slots < (1ull << 22)) {
+      slots = 1ull << 22;
+    }
+    if (slots > (1ull << 27)) {
+      slots = 1ull << 27;
+    }
+    e->pool_slots = slots;
+    HIP_CHECK(hipMalloc(&e->d_pool_keys, sizeof(u32) * slots));
+    HIP_CHECK(hipMalloc(&e->d_pool_vals, sizeof(i32) * slots));
+    HIP_CHECK(hipMemset(e->d_pool_keys, 0xFF, sizeof(u32) * slots));
+    HIP_CHECK(hipMemset(e->d_pool_vals, 0, sizeof(i32) * slots));
+    // two-hop buffers + temps
+    HIP_CHECK(hipMalloc(&e->d_cand, sizeof(u64) * n));
+    HIP_CHECK(hipMalloc(&e->d_cand2, sizeof(u64) * n));
+    HIP_CHECK(hipMalloc(&e->d_cfav, sizeof(u32) * n));
+    HIP_CHECK(hipMalloc(&e->d_crank, sizeof(u32) * n));
+    HIP_CHECK(hipMalloc(&e->d_cones, sizeof(u32) * n));
+    HIP_CHECK(rocprim::select(
+        nullptr, e->cand_select_temp_bytes, e->d_cand, e->d_cand2, e->d_prop_count, n, CandValid()
+    ));
+    HIP_CHECK(hipMalloc(&e->d_cand_select_temp, e->cand_select_temp_bytes));
+    rocprim::double_buffer<u64> ckeys(e->d_cand2, e->d_cand);
+    HIP_CHECK(rocprim::radix_sort_keys(nullptr, e->cand_sort_temp_bytes, ckeys, n));
+    HIP_CHECK(hipMalloc(&e->d_cand_sort_temp, e->cand_sort_temp_bytes));
+    HIP_CHECK(rocprim::inclusive_scan_by_key(
+        nullptr, e->cand_scan_temp_bytes, e->d_cfav, e->d_cones, e->d_crank, n,
+        rocprim::plus<u32>(), rocprim::equal_to<u32>()
+    ));
+    HIP_CHECK(hipMalloc(&e->d_cand_scan_temp, e->cand_scan_temp_bytes));
+  }
+  // the L list cap is the full chunk span for clustering (a refine run may
+  // have shrunk l_cap for its dense-gains buffer)
+  e->l_cap = e->C;
+  if (e->d_l_off) {
+    HIP_CHECK(hipFree(e->d_l_off));
+  }
+  HIP_CHECK(hipMalloc(&e->d_l_off, sizeof(u32) * (e->l_cap + 1)));
+  HIP_CHECK(hipMemsetAsync(e->d_pool_overflow, 0, sizeof(int), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_emptied, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_active, 1, n, e->stream));
+
+  const u32 ngrid = ceil_div(n, threads);
+  hipLaunchKernelGGL(k_iota, dim3(ngrid), dim3(threads), 0, e->stream, n, e->d_labels);
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(k_iota, dim3(ngrid), dim3(threads), 0, e->stream, n, e->d_favored);
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_init_cluster_weights, dim3(ngrid), dim3(threads), 0, e->stream, n, e->d_vwgt, e->d_weights
+  );
+  LAUNCH_CHECK();
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+
+  // ---- LP sweeps (driver: lp_clusterer.cc:89-109 + should_stop) ----
+  u64 live = n;
+  for (int iter = 0; iter < iters; ++iter) {
+    u64 sweep_moves = 0;
+    bool stopped = false;
+    for (u32 chunk = 0; chunk < kmp::kNumChunks && !stopped; ++chunk) {
+      const u32 pos_lo = chunk * e->C;
+      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
+      if (pos_lo >= pos_hi) {
+        continue;
+      }
+      const i64 cnt = kmp_lp_phase_a(e, iter, chunk, pos_lo, pos_hi, e->d_props, e->C);
+      if (cnt < 0) {
+        return -1;
+      }
+      const i64 mv = kmp_lp_commit(e, iter, chunk, e->d_props, static_cast<u32>(cnt));
+      if (mv < 0) {
+        return -1;
+      }
+      sweep_moves += mv;
+      live -= e->last_emptied;
+      if (desired_clusters > 0 && live <= desired_clusters) {
+        stopped = true;
+      }
+    }
+    if (sweep_moves == 0) {
+      break;
+    }
+  }
+
+  // ---- isolated nodes + two-hop (default preset strategies;
+  //      lp_clusterer.cc:112-162, oracle parity contract) ----
+  const bool handle_two_hop = (1.0 - 1.0 * live / n) <= 0.5;
+  if (handle_two_hop) {
+    // isolated nodes, MATCH semantics: deterministic host chain (isolated
+    // clusters stay singletons through LP, so their weights are the node
+    // weights captured at engine creation)
+    std::vector<u64> pairs;
+    u32 pending = 0xFFFFFFFFu;
+    i64 pending_w = 0;
+    for (size_t i = 0; i < e->isolated.size(); ++i) {
+      const u32 cu = e->isolated[i];
+      const i64 w = e->iso_weights[i];
+      if (pending != 0xFFFFFFFFu && pending_w + w <= max_cluster_weight) {
+        pairs.push_back((static_cast<u64>(pending) << 32) | cu);
+        pending = 0xFFFFFFFFu;
+      } else {
+        pending = cu;
+        pending_w = w;
+      }
+    }
+    if (!pairs.empty()) {
+      HIP_CHECK(hipMemcpyAsync(
+          e->d_cand, pairs.data(), sizeof(u64) * pairs.size(), hipMemcpyHostToDevice, e->stream
+      ));
+      hipLaunchKernelGGL(
+          k_apply_pairs, dim3(ceil_div(pairs.size(), threads)), dim3(threads), 0, e->stream,
+          e->d_cand, static_cast<u32>(pairs.size()), e->d_labels, e->d_weights
+      );
+      LAUNCH_CHECK();
+      live -= pairs.size();
+    }
+
+    // two-hop matching (MATCH_THREADWISE single-chain semantics): collect
+    // singleton candidates, sort by (favored, u), pair consecutive equals
+    hipLaunchKernelGGL(
+        k_twohop_cand, dim3(ngrid), dim3(threads), 0, e->stream, n, max_cluster_weight, e->d_xadj,
+        e->d_vwgt, e->d_labels, e->d_weights, e->d_favored, e->d_cand
+    );
+    LAUNCH_CHECK();
+    size_t stb = e->cand_select_temp_bytes;
+    HIP_CHECK(rocprim::select(
+        e->d_cand_select_temp, stb, e->d_cand, e->d_cand2, e->d_prop_count, n, CandValid(),
+        e->stream
+    ));
+    u32 cand_count = 0;
+    HIP_CHECK(hipMemcpyAsync(
+        &cand_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream
+    ));
+    HIP_CHECK(hipStreamSynchronize(e->stream));
+    if (cand_count > 1) {
+      rocprim::double_buffer<u64> ckeys(e->d_cand2, e->d_cand);
+      size_t ktb = e->cand_sort_temp_bytes;
+      HIP_CHECK(rocprim::radix_sort_keys(e->d_cand_sort_temp, ktb, ckeys, cand_count, 0, 64,
+                                         e->stream));
+      const u64 *sorted = ckeys.current();
+      const u32 cgrid = ceil_div(cand_count, threads);
+      hipLaunchKernelGGL(
+          k_extract_fav, dim3(cgrid), dim3(threads), 0, e->stream, sorted, cand_count, e->d_cfav
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_ones, dim3(cgrid), dim3(threads), 0, e->stream, cand_count, e->d_cones
+      );
+      LAUNCH_CHECK();
+      size_t sctb = e->cand_scan_temp_bytes;
+      HIP_CHECK(rocprim::inclusive_scan_by_key(
+          e->d_cand_scan_temp, sctb, e->d_cfav, e->d_cones, e->d_crank, cand_count,
+          rocprim::plus<u32>(), rocprim::equal_to<u32>(), e->stream
+      ));
+      HIP_CHECK(hipMemsetAsync(e->d_emptied, 0, sizeof(unsigned long long), e->stream));
+      hipLaunchKernelGGL(
+          k_twohop_pair, dim3(cgrid), dim3(threads), 0, e->stream, sorted, e->d_crank, cand_count,
+          e->d_labels, e->d_weights, e->d_emptied
+      );
+      LAUNCH_CHECK();
+      unsigned long long merged = 0;
+      HIP_CHECK(hipMemcpyAsync(&merged, e->d_emptied, sizeof(merged), hipMemcpyDeviceToHost,
+                               e->stream));
+      HIP_CHECK(hipStreamSynchronize(e->stream));
+      live -= merged;
+    }
+  }
+
+  // exact non-empty cluster count + download
+  HIP_CHECK(hipMemsetAsync(e->d_cut, 0, sizeof(unsigned long long), e->stream));
+  hipLaunchKernelGGL(
+      k_count_nonempty, dim3(ngrid), dim3(threads), 0, e->stream, n, e->d_weights, e->d_cut
+  );
+  LAUNCH_CHECK();
+  unsigned long long nonempty = 0, arcs = 0, moves = 0;
+  HIP_CHECK(hipMemcpyAsync(&nonempty, e->d_cut, sizeof(nonempty), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpyAsync(&arcs, e->d_arcs, sizeof(arcs), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpyAsync(&moves, e->d_moves, sizeof(moves), hipMemcpyDeviceToHost, e->stream));
+  HIP_CHECK(hipMemcpy(clustering, e->d_labels, sizeof(u32) * n, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+
+  if (stats) {
+    stats->arcs_scanned = arcs;
+    stats->moves = moves;
+    stats->phase_a_ns = static_cast<u64>(e->phase_a_ms * 1e6);
+    stats->total_ns = static_cast<u64>(e->commit_ms * 1e6);
+    stats->num_clusters = nonempty;
+    stats->edge_cut = 0;
+  }
+  return static_cast<i64>(nonempty);
 }
 
 } // extern "C"

@@ -168,3 +168,83 @@ def test_refine_parity_fullsize_golden():
         xor = int(np.bitwise_xor.reduce(part.astype(np.uint64)
                                         * (np.arange(g.n, dtype=np.uint64) + 1)))
         assert xor == case["label_xor"], name
+
+
+@pytest.mark.parametrize("scale,max_w,seed", [(10, 16, 1), (12, 32, 2), (14, 64, 3)])
+def test_cluster_parity_rmat(oracle, scale, max_w, seed):
+    from helpers import oracle_cluster
+
+    g = ka.Graph.rmat(scale, 8, seed=7)
+    eng = ka.LpEngine(g)
+    nc, clus, stats = eng.cluster(max_w, seed=seed, iters=5)
+    onc, oclus, ostats = oracle_cluster(oracle, g, max_w, seed=seed, iters=5)
+    assert nc == onc, f"cluster count gpu={nc} oracle={onc}"
+    assert (clus == oclus).all(), f"{(clus != oclus).sum()} labels differ"
+    assert stats.arcs_scanned == ostats[0]
+    assert stats.moves == ostats[1]
+    sizes = np.bincount(clus, minlength=g.n)
+    assert sizes.max() <= max_w
+
+
+def test_cluster_parity_star(oracle):
+    """Hub graph exercises the pooled-hash L path for clustering."""
+    from helpers import oracle_cluster
+
+    n = 50_000
+    arcs = set()
+    for v in range(1, n):
+        arcs.add((0, v)); arcs.add((v, 0))
+        w = v % (n - 1) + 1
+        if w != v:
+            arcs.add((v, w)); arcs.add((w, v))
+    arcs = sorted(arcs)
+    xadj = np.zeros(n + 1, np.uint32)
+    for u, v in arcs:
+        xadj[u + 1] += 1
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = np.array([v for _, v in arcs], dtype=np.uint32)
+    g = ka.Graph.from_csr(xadj, adjncy)
+    eng = ka.LpEngine(g)
+    nc, clus, _ = eng.cluster(64, seed=1, iters=5)
+    onc, oclus, _ = oracle_cluster(oracle, g, 64, seed=1, iters=5)
+    assert nc == onc and (clus == oclus).all()
+
+
+def test_cluster_parity_isolated(oracle):
+    """Graph with isolated vertices: isolated matching + two-hop pass."""
+    from helpers import oracle_cluster
+
+    rng = np.random.default_rng(4)
+    n = 4096
+    src = rng.integers(0, n // 2, 8000)  # upper half isolated
+    dst = rng.integers(0, n // 2, 8000)
+    mask = src != dst
+    pairs = np.unique(np.stack([np.concatenate([src[mask], dst[mask]]),
+                                np.concatenate([dst[mask], src[mask]])], 1), axis=0)
+    pairs = pairs[np.lexsort((pairs[:, 1], pairs[:, 0]))]
+    xadj = np.zeros(n + 1, np.uint32)
+    np.add.at(xadj, pairs[:, 0] + 1, 1)
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = pairs[:, 1].astype(np.uint32)
+    g = ka.Graph.from_csr(xadj, adjncy)
+    eng = ka.LpEngine(g)
+    nc, clus, _ = eng.cluster(32, seed=2, iters=5)
+    onc, oclus, _ = oracle_cluster(oracle, g, 32, seed=2, iters=5)
+    assert nc == onc and (clus == oclus).all()
+
+
+def test_cluster_then_refine_same_engine(oracle):
+    """Engine state is reusable across modes."""
+    from helpers import oracle_cluster
+
+    g = ka.Graph.rmat(12, 8, seed=7)
+    eng = ka.LpEngine(g)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    cut1, p1, _ = eng.refine(k, mbw, part0, seed=1)
+    nc, clus, _ = eng.cluster(32, seed=2)
+    cut2, p2, _ = eng.refine(k, mbw, part0, seed=1)
+    assert cut1 == cut2 and (p1 == p2).all()
+    onc, oclus, _ = oracle_cluster(oracle, g, 32, seed=2)
+    assert nc == onc and (clus == oclus).all()
