@@ -130,27 +130,23 @@ def main():
         last_cut, _part, _stats = eng.refine_end()
 
     elapsed = t_end - t_start
-    # per-rank arcs: in sharded mode each rank scans its slice; whole-job arcs
-    # = sum over ranks (phase A split); reduce.
+    # arcs_scanned counts the WHOLE chunk's processed set and is identical on
+    # every rank (labels are replicated), so no reduction is needed; the
+    # whole-job rate is arcs / max-over-ranks elapsed.
     if world > 1:
         import torch
         import torch.distributed as dist
 
-        t = torch.tensor([total_arcs, phase_a_ns], dtype=torch.int64, device=device)
-        dist.all_reduce(t)
-        total_arcs = int(t[0].item())
-        phase_a_ns = int(t[1].item())  # summed over ranks
         tmax = torch.tensor([elapsed], dtype=torch.float64, device=device)
         dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
         elapsed = float(tmax.item())
 
     value = total_arcs / elapsed
 
-    # roofline of the dominant kernel group (phase A: bin+gain/select),
-    # algorithmic bytes 8 B/arc over HIP-event time of those launches
-    achieved_gbs = (total_arcs * BYTES_PER_ARC) / max(phase_a_ns, 1)  # B/ns = GB/s
-    if world > 1:
-        achieved_gbs *= world  # phase_a_ns was summed across ranks; arcs too
+    # roofline of the dominant kernel group (phase A: gain/select+compact):
+    # algorithmic bytes 8 B/arc over HIP-event time of those launches. Each
+    # rank's phase A scans arcs/world of the work.
+    achieved_gbs = (total_arcs / world * BYTES_PER_ARC) / max(phase_a_ns, 1)  # B/ns = GB/s
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved_gbs, 1),
@@ -205,11 +201,11 @@ def run_cpu_baseline(g, k, mbw, part0, args):
     if res is None:
         return None
     arcs, dt = res
-    # the oracle's LP sweep is currently single-threaded; report that honestly
+    cores = int(os.environ.get("OMP_NUM_THREADS", os.cpu_count() or 1))
     return {
         "value": round(arcs / dt, 1),
         "unit": "arcs/s",
-        "cores": 1,
+        "cores": cores,
         "kind": "port",
         "sample": f"1 LP sweep over the full workload graph ({arcs} arcs, {dt:.1f}s)",
     }

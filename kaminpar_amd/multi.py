@@ -22,10 +22,20 @@ logic; compute still requires a GPU.
 import numpy as np
 
 
+def pos_count(n):
+    """Position space of the 64-vertex-unit permutation (lp_common.h)."""
+    return ((n + 63) // 64) * 64
+
+
+def chunk_size(n, num_chunks=64):
+    nu = (n + 63) // 64
+    return ((nu + num_chunks - 1) // num_chunks) * 64
+
+
 def chunk_ranges(n, num_chunks, chunk):
-    C = (n + num_chunks - 1) // num_chunks
+    C = chunk_size(n, num_chunks)
     lo = chunk * C
-    hi = min(lo + C, n)
+    hi = min(lo + C, pos_count(n))
     return lo, hi
 
 
@@ -45,7 +55,7 @@ def refine_dist(eng, k, max_block_weights, partition, seed, iters, comm):
     """
     n = len(partition)
     num_chunks = eng.num_chunks()
-    C = (n + num_chunks - 1) // num_chunks
+    C = chunk_size(n, num_chunks)
     cap = C // comm.world + 2
 
     eng.refine_begin(k, max_block_weights, partition, seed)

@@ -253,7 +253,7 @@ struct Proposal {
   u32 rank; // position within chunk (admission order)
   u32 u;
   u32 from;
-  u32 to;
+  u32 to; // 0xFFFFFFFF marks an empty slot (parallel phase A)
   i32 w;
 };
 
@@ -296,9 +296,8 @@ void lp_run(
   const u32 num_chunks = (P + C - 1) / C;
 
   std::vector<Proposal> proposals;
+  std::vector<Proposal> slots; // per-position results (parallel phase A)
   std::vector<u32> processed;
-  RatingMapOracle map;
-  map.reserve(1024);
 
   u32 live_clusters = live_clusters_io ? *live_clusters_io : 0;
 
@@ -314,9 +313,17 @@ void lp_run(
 
       proposals.clear();
       processed.clear();
+      slots.assign(pos_end - pos_begin, Proposal{0, 0, 0, 0xFFFFFFFFu, 0});
 
       // ---- phase A: gains + selection against the chunk-start snapshot ----
-      for (u32 p = pos_begin; p < pos_end; ++p) {
+      // (parallel over positions; results land in per-position slots so the
+      // compacted order is deterministic, mirroring the GPU design)
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 256)
+#endif
+      for (long long pp = pos_begin; pp < static_cast<long long>(pos_end); ++pp) {
+        const u32 p = static_cast<u32>(pp);
+        static thread_local RatingMapOracle map;
         const u32 u = perm(p);
         if (u >= n) {
           continue; // tail of the last unit
@@ -328,8 +335,6 @@ void lp_run(
         if (!active[u]) {
           continue;
         }
-        processed.push_back(u);
-        stats.arcs_scanned += deg;
 
         const u32 u_cluster = labels[u];
         const i32 u_weight = g.node_weight(u);
@@ -420,10 +425,26 @@ void lp_run(
         }
 
         if (have && best != u_cluster) {
-          proposals.push_back(
-              Proposal{p - pos_begin, u, u_cluster, best, u_weight}
-          );
+          slots[p - pos_begin] = Proposal{p - pos_begin, u, u_cluster, best, u_weight};
         }
+      }
+      // compact valid slots (position order) + arcs/processed bookkeeping
+      for (u32 p = pos_begin; p < pos_end; ++p) {
+        if (slots[p - pos_begin].to != 0xFFFFFFFFu) {
+          proposals.push_back(slots[p - pos_begin]);
+        }
+      }
+      for (u32 p = pos_begin; p < pos_end; ++p) {
+        const u32 u = perm(p);
+        if (u >= n) {
+          continue;
+        }
+        const u32 deg = g.degree(u);
+        if (deg > par.max_degree || !active[u]) {
+          continue;
+        }
+        processed.push_back(u);
+        stats.arcs_scanned += deg;
       }
 
       // ---- phase B: deterministic commit (greatest-fixpoint rollback) ----

@@ -248,3 +248,50 @@ def test_cluster_then_refine_same_engine(oracle):
     assert cut1 == cut2 and (p1 == p2).all()
     onc, oclus, _ = oracle_cluster(oracle, g, 32, seed=2)
     assert nc == onc and (clus == oclus).all()
+
+
+def test_sharded_phase_api_matches_monolithic(oracle):
+    """Simulate 2-rank sharding on one GPU: per chunk, run phase A for each
+    position slice separately, concatenate the proposal lists in rank order,
+    and commit the union -- must be bit-identical to the monolithic run
+    (this is exactly what kaminpar_amd.multi does across real ranks)."""
+    import ctypes
+
+    g = ka.Graph.rmat(12, 8, seed=7)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=5)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+
+    eng = ka.LpEngine(g)
+    cut_ref, part_ref, _ = eng.refine(k, mbw, part0, seed=3, iters=5)
+
+    import torch
+
+    from kaminpar_amd.multi import chunk_ranges, rank_slice
+
+    eng.refine_begin(k, mbw, part0, seed=3)
+    num_chunks = eng.num_chunks()
+    n = g.n
+    C = (((n + 63) // 64 + num_chunks - 1) // num_chunks) * 64
+    P = ((n + 63) // 64) * 64
+    cap = C
+    world = 2
+    bufs = [torch.zeros((cap, 4), dtype=torch.int32, device="cuda:0") for _ in range(world)]
+    for it in range(5):
+        sweep_moves = 0
+        for chunk in range(num_chunks):
+            lo = chunk * C
+            hi = min(lo + C, P)
+            if lo >= hi:
+                continue
+            counts = []
+            for r in range(world):
+                slo, shi = rank_slice(lo, hi, r, world)
+                counts.append(eng.phase_a(it, chunk, slo, shi, bufs[r].data_ptr(), cap))
+            cat = torch.cat([bufs[r][: counts[r]] for r in range(world)]).contiguous()
+            sweep_moves += eng.commit(it, chunk, cat.data_ptr(), sum(counts))
+        if sweep_moves == 0:
+            break
+    cut2, part2, _ = eng.refine_end()
+    assert cut2 == cut_ref
+    assert (part2 == part_ref).all()
