@@ -147,13 +147,26 @@ def main():
     # algorithmic bytes 8 B/arc over HIP-event time of those launches. Each
     # rank's phase A scans arcs/world of the work.
     achieved_gbs = (total_arcs / world * BYTES_PER_ARC) / max(phase_a_ns, 1)  # B/ns = GB/s
+    # measured HBM traffic per launch of the dominant kernel, from the
+    # committed rocprofv3 PMC run (profiles/pmc_traffic.json); null when the
+    # workload has no committed measurement
+    traffic = None
+    try:
+        with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "profiles", "pmc_traffic.json")) as fh:
+            tbl = json.load(fh)
+        entry = tbl.get(f"rmat{args.scale}_k{args.k}_lp_refine")
+        if entry:
+            traffic = entry["fetch_bytes_per_launch"] + entry["write_bytes_per_launch"]
+    except OSError:
+        pass
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved_gbs, 1),
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-        "traffic": None,  # filled from rocprofv3 PMC runs (profiles/)
+        "traffic": traffic,
     }
 
     result = None
@@ -201,7 +214,8 @@ def run_cpu_baseline(g, k, mbw, part0, args):
     if res is None:
         return None
     arcs, dt = res
-    cores = int(os.environ.get("OMP_NUM_THREADS", os.cpu_count() or 1))
+    cores = int(os.environ.get("KMP_ORACLE_THREADS",
+                               min(32, os.cpu_count() or 1)))
     return {
         "value": round(arcs / dt, 1),
         "unit": "arcs/s",
