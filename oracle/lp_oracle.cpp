@@ -339,7 +339,10 @@ void lp_run(
       // (parallel over positions; results land in per-position slots so the
       // compacted order is deterministic, mirroring the GPU design)
 #ifdef _OPENMP
-#pragma omp parallel for schedule(dynamic, 256) num_threads(oracle_threads())
+// parallelism only pays for large chunks; fork/join + spin-wait overhead
+// dominates on small ones (measured 300x slowdown on 4K-vertex graphs)
+#pragma omp parallel for schedule(dynamic, 256) num_threads(oracle_threads()) \
+    if (pos_end - pos_begin >= 65536)
 #endif
       for (long long pp = pos_begin; pp < static_cast<long long>(pos_end); ++pp) {
         const u32 p = static_cast<u32>(pp);
