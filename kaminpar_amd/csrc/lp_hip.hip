@@ -152,6 +152,7 @@ __global__ void k_phase_s(
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const uint8_t *__restrict__ active,
+    const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots,
     u64 *__restrict__ l_list,
     u32 *__restrict__ l_count
@@ -166,25 +167,31 @@ __global__ void k_phase_s(
   }
 
   const BlockPerm perm(n, iter_seed);
-  const u32 u = perm(p);
+  // all 4 positions of this wave share one 64-vertex unit; one byte decides
+  // whether anything in it is active (slots are pre-marked invalid)
+  const u32 vb = perm.fp(p / kmp::kUnit);
+  if (!unit_active[vb]) {
+    return;
+  }
+  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
   const u32 sidx = p - pos_lo;
 
-  bool emit_invalid = false;
+  bool skip = false;
   u32 row = 0, deg = 0;
   if (u >= n) {
-    emit_invalid = true;
+    skip = true;
   } else {
     row = xadj[u];
     deg = xadj[u + 1] - row;
     if (!active[u] || deg > max_degree) {
-      emit_invalid = true;
+      skip = true;
     }
   }
 
-  // append active high-degree positions to the L work list (rare;
-  // wave-aggregated: one atomic per wave that holds an L candidate)
+  // append active (processed) high-degree positions to the L work list
+  // (rare; wave-aggregated: one atomic per wave that holds an L candidate)
   {
-    const bool is_l = !emit_invalid && u < n && deg > kMidDeg && slot == 0;
+    const bool is_l = !skip && deg > kMidDeg && slot == 0;
     const unsigned long long ll = __ballot(is_l);
     if (ll) {
       const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
@@ -199,14 +206,8 @@ __global__ void k_phase_s(
       }
     }
   }
-  if (!emit_invalid && deg > kSmallDeg) {
-    return; // M/L owns this slot
-  }
-  if (emit_invalid) {
-    if (slot == 0) {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
-    }
-    return;
+  if (skip || deg > kSmallDeg) {
+    return; // nothing to do here (M/L own larger degrees)
   }
 
   // candidate load: lane handles one edge
@@ -258,12 +259,8 @@ __global__ void k_phase_s(
     }
   }
 
-  if (slot == 0) {
-    if (best.have && best.c != cur) {
-      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-    } else {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
-    }
+  if (slot == 0 && best.have && best.c != cur) {
+    slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
   }
 }
 
@@ -291,6 +288,7 @@ __global__ void k_phase_m(
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const uint8_t *__restrict__ active,
+    const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
@@ -303,7 +301,11 @@ __global__ void k_phase_m(
   }
 
   const BlockPerm perm(n, iter_seed);
-  const u32 u = perm(p);
+  const u32 vb = perm.fp(p / kmp::kUnit);
+  if (!unit_active[vb]) {
+    return;
+  }
+  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
   if (u >= n) {
     return;
   }
@@ -363,13 +365,8 @@ __global__ void k_phase_m(
     }
   }
 
-  if (lane == 0) {
-    const u32 sidx = p - pos_lo;
-    if (best.have && best.c != cur) {
-      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-    } else {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
-    }
+  if (lane == 0 && best.have && best.c != cur) {
+    slots[p - pos_lo] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
   }
 }
 
@@ -381,8 +378,9 @@ __global__ void k_phase_m(
 // from serializing on a single workgroup.
 constexpr u32 kLSlice = 8192;
 
-// Single tiny kernel: per-vertex slice counts -> exclusive prefix (l_off),
-// total in l_off[count]. One wave; l_count is small (hubs are rare).
+// L prep: per-vertex slice counts gathered in parallel (the scattered
+// xadj reads are the expensive part), then a serial prefix over the small
+// L2-hot counts array. One 1024-thread workgroup.
 __global__ void k_l_prep(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
@@ -390,14 +388,19 @@ __global__ void k_l_prep(
     u32 l_cap,
     u32 *__restrict__ l_off
 ) {
-  if (threadIdx.x == 0 && blockIdx.x == 0) {
-    const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  for (u32 i = threadIdx.x; i < count; i += blockDim.x) {
+    const u32 u = static_cast<u32>(l_list[i]);
+    const u32 deg = xadj[u + 1] - xadj[u];
+    l_off[i] = (deg + kLSlice - 1) / kLSlice;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
     u32 acc = 0;
     for (u32 i = 0; i < count; ++i) {
+      const u32 c = l_off[i];
       l_off[i] = acc;
-      const u32 u = static_cast<u32>(l_list[i]);
-      const u32 deg = xadj[u + 1] - xadj[u];
-      acc += (deg + kLSlice - 1) / kLSlice;
+      acc += c;
     }
     l_off[count] = acc;
   }
@@ -546,8 +549,6 @@ __global__ void k_phase_l_sel(
       }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
-      } else {
-        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
       }
     }
     __syncthreads();
@@ -657,8 +658,6 @@ __global__ void k_phase_l_direct(
       }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
-      } else {
-        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
       }
     }
     __syncthreads();
@@ -692,6 +691,7 @@ __global__ void k_phase_s_c(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const uint8_t *__restrict__ active,
+    const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots,
     u64 *__restrict__ l_list, // deg > kClusterMidDeg
@@ -707,24 +707,29 @@ __global__ void k_phase_s_c(
   }
 
   const BlockPerm perm(n, iter_seed);
-  const u32 u = perm(p);
+  // all 4 positions of this wave share one 64-vertex unit (slots pre-invalid)
+  const u32 vb = perm.fp(p / kmp::kUnit);
+  if (!unit_active[vb]) {
+    return;
+  }
+  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
   const u32 sidx = p - pos_lo;
 
-  bool emit_invalid = false;
+  bool skip = false;
   u32 row = 0, deg = 0;
   if (u >= n) {
-    emit_invalid = true;
+    skip = true;
   } else {
     row = xadj[u];
     deg = xadj[u + 1] - row;
     if (!active[u] || deg > max_degree) {
-      emit_invalid = true;
+      skip = true;
     }
   }
 
   // L list append (deg > kClusterMidDeg), wave-aggregated
   {
-    const bool is_l = !emit_invalid && u < n && deg > kClusterMidDeg && slot == 0;
+    const bool is_l = !skip && deg > kClusterMidDeg && slot == 0;
     const unsigned long long ll = __ballot(is_l);
     if (ll) {
       const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
@@ -738,14 +743,8 @@ __global__ void k_phase_s_c(
       }
     }
   }
-  if (!emit_invalid && deg > kSmallDeg) {
-    return; // M/L owns this slot
-  }
-  if (emit_invalid) {
-    if (slot == 0) {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
-    }
-    return;
+  if (skip || deg > kSmallDeg) {
+    return; // M/L own larger degrees
   }
 
   u32 c = kInvalid;
@@ -814,8 +813,6 @@ __global__ void k_phase_s_c(
     }
     if (best.have && best.c != cur) {
       slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-    } else {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
     }
   }
 }
@@ -838,6 +835,7 @@ __global__ void k_phase_m_c(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const uint8_t *__restrict__ active,
+    const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots
 ) {
@@ -851,7 +849,11 @@ __global__ void k_phase_m_c(
   }
 
   const BlockPerm perm(n, iter_seed);
-  const u32 u = perm(p);
+  const u32 vb = perm.fp(p / kmp::kUnit);
+  if (!unit_active[vb]) {
+    return;
+  }
+  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
   if (u >= n) {
     return;
   }
@@ -942,11 +944,8 @@ __global__ void k_phase_m_c(
     if (store_favored) {
       favored[u] = fav.have ? fav.c : cur;
     }
-    const u32 sidx = p - pos_lo;
     if (best.have && best.c != cur) {
-      slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
-    } else {
-      slots[sidx] = Prop{0, kInvalid, 0, 0};
+      slots[p - pos_lo] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
     }
   }
 }
@@ -965,22 +964,27 @@ __global__ void k_l_prep_c(
     u32 *__restrict__ l_hbits, // log2(region size)
     int *__restrict__ pool_overflow
 ) {
-  if (threadIdx.x == 0 && blockIdx.x == 0) {
-    const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  for (u32 i = threadIdx.x; i < count; i += blockDim.x) {
+    const u32 u = static_cast<u32>(l_list[i]);
+    const u32 deg = xadj[u + 1] - xadj[u];
+    l_off[i] = (deg + kLSlice - 1) / kLSlice;
+    u32 bits = 11; // >= 2048 slots
+    while ((1u << bits) < 2 * deg) {
+      ++bits;
+    }
+    l_hbits[i] = bits;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
     u32 sacc = 0;
     u64 hacc = 0;
     for (u32 i = 0; i < count; ++i) {
+      const u32 c = l_off[i];
       l_off[i] = sacc;
+      sacc += c;
       l_hoff[i] = hacc;
-      const u32 u = static_cast<u32>(l_list[i]);
-      const u32 deg = xadj[u + 1] - xadj[u];
-      sacc += (deg + kLSlice - 1) / kLSlice;
-      u32 bits = 11; // >= 2048 slots
-      while ((1u << bits) < 2 * deg) {
-        ++bits;
-      }
-      l_hbits[i] = bits;
-      hacc += 1ull << bits;
+      hacc += 1ull << l_hbits[i];
     }
     l_off[count] = sacc;
     l_hoff[count] = hacc;
@@ -1171,8 +1175,6 @@ __global__ void k_phase_l_sel_c(
       }
       if (tb.have && tb.c != cur) {
         slots[p - pos_lo] = Prop{u, tb.c, p - chunk_base, static_cast<u32>(u_w)};
-      } else {
-        slots[p - pos_lo] = Prop{0, kInvalid, 0, 0};
       }
     }
     __syncthreads();
@@ -1665,22 +1667,37 @@ __global__ void k_clear_active(
     u32 max_degree,
     const u32 *__restrict__ xadj,
     uint8_t *__restrict__ active,
+    uint8_t *__restrict__ unit_active,
     unsigned long long *__restrict__ arcs
 ) {
   __shared__ unsigned long long wg_sum[4];
   const u32 tid = blockIdx.x * blockDim.x + threadIdx.x;
   const u32 p = chunk_lo + tid;
   const BlockPerm perm(n, iter_seed);
+  const u32 lane0 = threadIdx.x & (kWave - 1);
+
+  // one wave covers exactly one 64-vertex unit (chunks are 64-aligned):
+  // skip fully-inactive units, clear the unit bit after processing (the
+  // activation kernel re-sets it for units that received movers)
+  u32 vb = 0;
+  bool unit_on = false;
+  if (p < chunk_hi) {
+    vb = perm.fp(p / kmp::kUnit);
+    unit_on = unit_active[vb] != 0;
+  }
 
   u64 my_deg = 0;
-  if (p < chunk_hi) {
-    const u32 u = perm(p);
+  if (p < chunk_hi && unit_on) {
+    const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
     if (u < n) {
       const u32 deg = xadj[u + 1] - xadj[u];
       if (deg <= max_degree && active[u]) {
         my_deg = deg;
         active[u] = 0;
       }
+    }
+    if (lane0 == 0) {
+      unit_active[vb] = 0;
     }
   }
   for (int off = 32; off > 0; off >>= 1) {
@@ -1707,7 +1724,8 @@ __global__ void k_activate(
     u32 count,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
-    uint8_t *__restrict__ active
+    uint8_t *__restrict__ active,
+    uint8_t *__restrict__ unit_active
 ) {
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const u32 lane = threadIdx.x & (kWave - 1);
@@ -1718,7 +1736,10 @@ __global__ void k_activate(
   const u32 row = xadj[u];
   const u32 deg = xadj[u + 1] - row;
   for (u32 e = lane; e < deg; e += kWave) {
-    active[adjncy[row + e]] = 1;
+    const u32 v = adjncy[row + e];
+    active[v] = 1;
+    unit_active[v >> 6]; // (read avoids compiler warning removal)
+    unit_active[v >> 6] = 1;
   }
 }
 
@@ -1837,6 +1858,7 @@ struct kmp_lp_t {
   i64 *d_weights = nullptr;
   i64 *d_maxw = nullptr;
   uint8_t *d_active = nullptr;
+  uint8_t *d_unit_active = nullptr; // one byte per 64-vertex unit
 
   // phase buffers
   Prop *d_slots = nullptr; // C
@@ -1993,6 +2015,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels0, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_active, e->n));
+  HIP_CHECK(hipMalloc(&e->d_unit_active, kmp::num_units(e->n)));
 
   const u32 C = e->C;
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
@@ -2053,7 +2076,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
                   (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
@@ -2137,6 +2160,7 @@ int kmp_lp_refine_begin(
       e->d_labels0, e->d_labels, sizeof(u32) * e->n, hipMemcpyDeviceToDevice, e->stream
   ));
   HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_unit_active, 1, kmp::num_units(e->n), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
 
@@ -2165,18 +2189,21 @@ i64 kmp_lp_phase_a(
   const u32 max_degree = 0xFFFFFFFFu;
 
   HIP_CHECK(hipMemsetAsync(e->d_l_count, 0, sizeof(u32), e->stream));
+  // pre-mark every slot invalid (kernels only write actual proposals; a
+  // skipped unit costs one byte read instead of 64 slot writes)
+  HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * span, e->stream));
 
   hipEvent_t ev0, ev1;
   e->ev_pair(ev0, ev1);
   HIP_CHECK(hipEventRecord(ev0, e->stream));
 
   if (!e->clusterer) {
-    // S: 4 positions/wave (default owner of every position's slot)
+    // S: 4 positions/wave (unit-gated; slots pre-marked invalid)
     hipLaunchKernelGGL(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_active, e->d_slots, e->d_l_list, e->d_l_count
+        e->d_active, e->d_unit_active, e->d_slots, e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
     // M: one wave per position
@@ -2188,14 +2215,14 @@ i64 kmp_lp_phase_a(
           kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
           e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
           e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-          e->d_slots
+          e->d_unit_active, e->d_slots
       );
       LAUNCH_CHECK();
     }
     // L: slice-parallel accumulation over the (rare) high-degree list
     {
       hipLaunchKernelGGL(
-          k_l_prep, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
+          k_l_prep, dim3(1), dim3(1024), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
           e->l_cap, e->d_l_off
       );
       LAUNCH_CHECK();
@@ -2214,6 +2241,7 @@ i64 kmp_lp_phase_a(
           e->l_cap, e->d_l_gains, e->d_slots
       );
       LAUNCH_CHECK();
+      // pathological overflow beyond l_cap: direct per-vertex workgroups
       {
         const size_t lds =
             ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
@@ -2233,7 +2261,7 @@ i64 kmp_lp_phase_a(
         k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_active, e->d_favored, e->d_slots, e->d_l_list, e->d_l_count
+        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
     {
@@ -2243,13 +2271,13 @@ i64 kmp_lp_phase_a(
           kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
           e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->maxw_uniform,
           e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_active,
-          e->d_favored, e->d_slots
+          e->d_unit_active, e->d_favored, e->d_slots
       );
       LAUNCH_CHECK();
     }
     {
       hipLaunchKernelGGL(
-          k_l_prep_c, dim3(1), dim3(64), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
+          k_l_prep_c, dim3(1), dim3(1024), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
           e->l_cap, e->pool_slots, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_pool_overflow
       );
       LAUNCH_CHECK();
@@ -2445,13 +2473,15 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
   // ranks) and tally scanned arcs
   hipLaunchKernelGGL(
       k_clear_active, dim3(ceil_div(chunk_hi - chunk_lo, threads)), dim3(threads), 0, e->stream,
-      chunk_lo, chunk_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_arcs
+      chunk_lo, chunk_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active,
+      e->d_arcs
   );
   LAUNCH_CHECK();
   if (count > 0) {
     hipLaunchKernelGGL(
         k_activate, dim3(ceil_div(static_cast<u64>(count) * kWave, threads)), dim3(threads), 0,
-        e->stream, order, e->d_admitted_flags, props, count, e->d_xadj, e->d_adjncy, e->d_active
+        e->stream, order, e->d_admitted_flags, props, count, e->d_xadj, e->d_adjncy, e->d_active,
+        e->d_unit_active
     );
     LAUNCH_CHECK();
     hipLaunchKernelGGL(
@@ -2508,6 +2538,7 @@ int kmp_lp_reset(kmp_lp_t *e) {
       e->d_labels, e->d_labels0, sizeof(u32) * e->n, hipMemcpyDeviceToDevice, e->stream
   ));
   HIP_CHECK(hipMemsetAsync(e->d_active, 1, e->n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_unit_active, 1, kmp::num_units(e->n), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * e->k, e->stream));
@@ -2682,6 +2713,7 @@ i64 kmp_lp_cluster(
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_active, 1, n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_unit_active, 1, kmp::num_units(e->n), e->stream));
 
   const u32 ngrid = ceil_div(n, threads);
   hipLaunchKernelGGL(k_iota, dim3(ngrid), dim3(threads), 0, e->stream, n, e->d_labels);
