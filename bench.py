@@ -63,6 +63,9 @@ def main():
 
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local_rank)
+        # force torch's HIP runtime to initialize before the engine's
+        # (torch bundles its own libamdhip64; engine-first breaks torch)
+        torch.zeros(1, device=f"cuda:{local_rank}")
         dist.init_process_group("nccl")
         device = f"cuda:{local_rank}"
     else:

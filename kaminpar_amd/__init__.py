@@ -5,6 +5,11 @@ Host-side graph handling works everywhere; the LP engine requires a GPU and
 fails loudly if the HIP device or the native library is missing -- there is
 no CPU fallback on the product path (the CPU oracle under oracle/ is test
 infrastructure only).
+
+Interop note: PyTorch wheels bundle their own HIP runtime. When combining
+this library with torch GPU tensors in one process (multi-GPU sharding),
+initialize torch.cuda FIRST -- the engine then binds torch's already-loaded
+runtime (same SONAME). Engine-first leaves torch unable to see the GPU.
 """
 
 import ctypes
@@ -171,7 +176,7 @@ class Graph:
 
     def __del__(self):
         h = getattr(self, "_h", None)
-        if h:
+        if h and _lib is not None:
             _lib.kmp_graph_free(h)
             self._h = None
 
@@ -269,6 +274,6 @@ class LpEngine:
 
     def __del__(self):
         h = getattr(self, "_h", None)
-        if h:
+        if h and _lib is not None:
             _lib.kmp_lp_free(h)
             self._h = None

@@ -255,7 +255,12 @@ def test_sharded_phase_api_matches_monolithic(oracle):
     position slice separately, concatenate the proposal lists in rank order,
     and commit the union -- must be bit-identical to the monolithic run
     (this is exactly what kaminpar_amd.multi does across real ranks)."""
-    import ctypes
+    # torch bundles its own HIP runtime; it must initialize BEFORE the
+    # engine creates a context with the system runtime (same SONAME: the
+    # engine then binds torch's already-loaded runtime)
+    import torch
+
+    torch.zeros(1, device="cuda:0")
 
     g = ka.Graph.rmat(12, 8, seed=7)
     k = 16
@@ -264,8 +269,6 @@ def test_sharded_phase_api_matches_monolithic(oracle):
 
     eng = ka.LpEngine(g)
     cut_ref, part_ref, _ = eng.refine(k, mbw, part0, seed=3, iters=5)
-
-    import torch
 
     from kaminpar_amd.multi import chunk_ranges, rank_slice
 
