@@ -2185,6 +2185,9 @@ i64 kmp_lp_phase_a(
   const u32 chunk_base = chunk * e->C;
   Prop *out = static_cast<Prop *>(d_out);
   const u32 span = pos_hi - pos_lo;
+  if (span == 0) {
+    return 0; // empty rank slice
+  }
   const u32 threads = 256;
   const u32 max_degree = 0xFFFFFFFFu;
 
