@@ -73,7 +73,23 @@ def main():
 
     log(f"[bench] generating R-MAT scale-{args.scale} ef={args.edgefactor} ...")
     t0 = time.time()
-    g = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+    if world > 1:
+        # one generation per node (8 concurrent generator scratches would
+        # exhaust host RAM): rank 0 generates + saves, the rest load
+        import torch.distributed as dist
+
+        cache = f"/tmp/kmp_rmat{args.scale}_{args.edgefactor}"
+        if rank == 0:
+            g0 = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+            np.save(cache + "_xadj.npy", np.asarray(g0.xadj))
+            np.save(cache + "_adjncy.npy", np.asarray(g0.adjncy))
+            del g0
+        dist.barrier()
+        xadj = np.load(cache + "_xadj.npy", mmap_mode="r")
+        adjncy = np.load(cache + "_adjncy.npy", mmap_mode="r")
+        g = ka.Graph.from_csr(np.asarray(xadj), np.asarray(adjncy))
+    else:
+        g = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
     log(f"[bench] n={g.n} m={g.m} ({time.time()-t0:.1f}s); uploading ...")
 
     k = args.k

@@ -378,31 +378,23 @@ __global__ void k_phase_m(
 // from serializing on a single workgroup.
 constexpr u32 kLSlice = 8192;
 
-// L prep: per-vertex slice counts gathered in parallel (the scattered
-// xadj reads are the expensive part), then a serial prefix over the small
-// L2-hot counts array. One 1024-thread workgroup.
-__global__ void k_l_prep(
+// L prep (refine): write per-vertex slice counts in parallel into a
+// zeroed l_cap+1 array; the exclusive prefix runs as a rocprim scan on the
+// host side (entries beyond the live count are zero, so l_off[count] holds
+// the total).
+__global__ void k_l_sizes(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     const u32 *__restrict__ xadj,
     u32 l_cap,
-    u32 *__restrict__ l_off
+    u32 *__restrict__ sizes
 ) {
   const u32 count = *l_count < l_cap ? *l_count : l_cap;
-  for (u32 i = threadIdx.x; i < count; i += blockDim.x) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < count) {
     const u32 u = static_cast<u32>(l_list[i]);
     const u32 deg = xadj[u + 1] - xadj[u];
-    l_off[i] = (deg + kLSlice - 1) / kLSlice;
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    u32 acc = 0;
-    for (u32 i = 0; i < count; ++i) {
-      const u32 c = l_off[i];
-      l_off[i] = acc;
-      acc += c;
-    }
-    l_off[count] = acc;
+    sizes[i] = (deg + kLSlice - 1) / kLSlice;
   }
 }
 
@@ -1865,7 +1857,10 @@ struct kmp_lp_t {
   Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr;
-  u32 *d_l_off = nullptr;  // l_cap + 1 (slice prefix)
+  u32 *d_l_off = nullptr;   // l_cap + 1 (slice prefix)
+  u32 *d_l_sizes = nullptr; // l_cap + 1 (slice counts, scan input)
+  void *d_lscan_temp = nullptr;
+  size_t lscan_temp_bytes = 0;
   i32 *d_l_gains = nullptr; // l_cap x k (allocated at refine_begin)
   u32 l_cap = 0;
   u32 *d_prop_count = nullptr;
@@ -2078,7 +2073,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
-                  (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
@@ -2152,6 +2147,17 @@ int kmp_lp_refine_begin(
     e->l_cap = 1024;
   }
   HIP_CHECK(hipMalloc(&e->d_l_off, sizeof(u32) * (e->l_cap + 1)));
+  if (e->d_l_sizes) {
+    HIP_CHECK(hipFree(e->d_l_sizes));
+  }
+  if (e->d_lscan_temp) {
+    HIP_CHECK(hipFree(e->d_lscan_temp));
+  }
+  HIP_CHECK(hipMalloc(&e->d_l_sizes, sizeof(u32) * (e->l_cap + 1)));
+  HIP_CHECK(rocprim::exclusive_scan(
+      nullptr, e->lscan_temp_bytes, e->d_l_sizes, e->d_l_off, 0u, e->l_cap + 1
+  ));
+  HIP_CHECK(hipMalloc(&e->d_lscan_temp, e->lscan_temp_bytes));
   HIP_CHECK(hipMalloc(&e->d_l_gains, sizeof(i32) * e->l_cap * k));
   HIP_CHECK(hipMemsetAsync(e->d_l_gains, 0, sizeof(i32) * e->l_cap * k, e->stream));
   HIP_CHECK(hipMemcpy(e->d_maxw, max_block_weights, sizeof(i64) * k, hipMemcpyHostToDevice));
@@ -2224,11 +2230,17 @@ i64 kmp_lp_phase_a(
     }
     // L: slice-parallel accumulation over the (rare) high-degree list
     {
+      HIP_CHECK(hipMemsetAsync(e->d_l_sizes, 0, sizeof(u32) * (e->l_cap + 1), e->stream));
       hipLaunchKernelGGL(
-          k_l_prep, dim3(1), dim3(1024), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
-          e->l_cap, e->d_l_off
+          k_l_sizes, dim3(ceil_div(e->l_cap, 256)), dim3(256), 0, e->stream, e->d_l_list,
+          e->d_l_count, e->d_xadj, e->l_cap, e->d_l_sizes
       );
       LAUNCH_CHECK();
+      size_t ltb = e->lscan_temp_bytes;
+      HIP_CHECK(rocprim::exclusive_scan(
+          e->d_lscan_temp, ltb, e->d_l_sizes, e->d_l_off, 0u, e->l_cap + 1,
+          rocprim::plus<u32>(), e->stream
+      ));
       const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
       {
         auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;

@@ -6,6 +6,15 @@ weighted graphs, several k, and full-size property checks."""
 import numpy as np
 import pytest
 
+# torch bundles its own HIP runtime; it must initialize BEFORE any engine in
+# this process creates a context with the system runtime (same SONAME: the
+# engine then binds torch's already-loaded runtime). Engine-first leaves
+# torch.cuda unable to see the GPU.
+import torch as _torch
+
+if _torch.cuda.is_available():
+    _torch.zeros(1, device="cuda:0")
+
 import kaminpar_amd as ka
 from helpers import oracle_refine
 
@@ -255,12 +264,7 @@ def test_sharded_phase_api_matches_monolithic(oracle):
     position slice separately, concatenate the proposal lists in rank order,
     and commit the union -- must be bit-identical to the monolithic run
     (this is exactly what kaminpar_amd.multi does across real ranks)."""
-    # torch bundles its own HIP runtime; it must initialize BEFORE the
-    # engine creates a context with the system runtime (same SONAME: the
-    # engine then binds torch's already-loaded runtime)
     import torch
-
-    torch.zeros(1, device="cuda:0")
 
     g = ka.Graph.rmat(12, 8, seed=7)
     k = 16
