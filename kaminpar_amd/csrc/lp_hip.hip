@@ -154,6 +154,7 @@ __global__ void k_phase_s(
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots,
+    u64 *__restrict__ m_slots,
     u64 *__restrict__ l_list,
     u32 *__restrict__ l_count
 ) {
@@ -188,8 +189,12 @@ __global__ void k_phase_s(
     }
   }
 
-  // append active (processed) high-degree positions to the L work list
-  // (rare; wave-aggregated: one atomic per wave that holds an L candidate)
+  // M-class positions go to per-position slots (compacted into the M work
+  // list by a stable select; no append atomics); L-class (rare) appends to
+  // the L list wave-aggregated.
+  if (!skip && slot == 0 && deg > kSmallDeg && deg <= kMidDeg) {
+    m_slots[sidx] = (static_cast<u64>(p) << 32) | u;
+  }
   {
     const bool is_l = !skip && deg > kMidDeg && slot == 0;
     const unsigned long long ll = __ballot(is_l);
@@ -274,49 +279,36 @@ __global__ void k_phase_s(
 template <bool kUnitWeights>
 __global__ void k_phase_m(
     u32 pos_lo,
-    u32 pos_hi,
     u32 chunk_base,
-    u32 n,
-    u64 iter_seed,
-    u32 max_degree,
     u32 k,
+    u64 iter_seed,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
-    const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
+    const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    const uint8_t *__restrict__ active,
-    const uint8_t *__restrict__ unit_active,
+    const u64 *__restrict__ m_list,
+    const u32 *__restrict__ m_count,
     Prop *__restrict__ slots
 ) {
   extern __shared__ i32 lds[];
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wave_in_wg = threadIdx.x >> 6;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const u32 p = pos_lo + wave_id;
-  if (p >= pos_hi) {
-    return;
-  }
-
-  const BlockPerm perm(n, iter_seed);
-  const u32 vb = perm.fp(p / kmp::kUnit);
-  if (!unit_active[vb]) {
-    return;
-  }
-  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
-  if (u >= n) {
-    return;
-  }
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
-  if (deg <= kSmallDeg || deg > kMidDeg || deg > max_degree || !active[u]) {
-    return; // S or L owns this slot
-  }
-
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
   const u32 R = gain_replicas(k);
   i32 *gains = lds + wave_in_wg * k * R; // R replicas of k counters
+
+  const u32 count = *m_count;
+  for (u32 vid = wave_id; vid < count; vid += num_waves) {
+  const u64 rec = m_list[vid];
+  const u32 p = static_cast<u32>(rec >> 32);
+  const u32 u = static_cast<u32>(rec);
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+
   for (u32 c = lane; c < k * R; c += kWave) {
     gains[c] = 0;
   }
@@ -367,6 +359,8 @@ __global__ void k_phase_m(
 
   if (lane == 0 && best.have && best.c != cur) {
     slots[p - pos_lo] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  }
+  __threadfence_block(); // gains reuse across grid-stride iterations
   }
 }
 
@@ -686,6 +680,7 @@ __global__ void k_phase_s_c(
     const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots,
+    u64 *__restrict__ m_slots,
     u64 *__restrict__ l_list, // deg > kClusterMidDeg
     u32 *__restrict__ l_count
 ) {
@@ -719,6 +714,9 @@ __global__ void k_phase_s_c(
     }
   }
 
+  if (!skip && slot == 0 && deg > kSmallDeg && deg <= kClusterMidDeg) {
+    m_slots[sidx] = (static_cast<u64>(p) << 32) | u;
+  }
   // L list append (deg > kClusterMidDeg), wave-aggregated
   {
     const bool is_l = !skip && deg > kClusterMidDeg && slot == 0;
@@ -814,20 +812,17 @@ __global__ void k_phase_s_c(
 template <bool kUnitWeights>
 __global__ void k_phase_m_c(
     u32 pos_lo,
-    u32 pos_hi,
     u32 chunk_base,
-    u32 n,
     u64 iter_seed,
-    u32 max_degree,
     i64 maxw_uniform,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
-    const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
+    const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
-    const uint8_t *__restrict__ active,
-    const uint8_t *__restrict__ unit_active,
+    const u64 *__restrict__ m_list,
+    const u32 *__restrict__ m_count,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots
 ) {
@@ -835,28 +830,18 @@ __global__ void k_phase_m_c(
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wave_in_wg = threadIdx.x >> 6;
   const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const u32 p = pos_lo + wave_id;
-  if (p >= pos_hi) {
-    return;
-  }
-
-  const BlockPerm perm(n, iter_seed);
-  const u32 vb = perm.fp(p / kmp::kUnit);
-  if (!unit_active[vb]) {
-    return;
-  }
-  const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
-  if (u >= n) {
-    return;
-  }
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
-  if (deg <= kSmallDeg || deg > kClusterMidDeg || deg > max_degree || !active[u]) {
-    return;
-  }
-
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
   u32 *hkeys = ldsu + wave_in_wg * 2 * kHashSlots;
   i32 *hvals = reinterpret_cast<i32 *>(hkeys + kHashSlots);
+
+  const u32 count = *m_count;
+  for (u32 vid = wave_id; vid < count; vid += num_waves) {
+  const u64 rec = m_list[vid];
+  const u32 p = static_cast<u32>(rec >> 32);
+  const u32 u = static_cast<u32>(rec);
+  const u32 row = xadj[u];
+  const u32 deg = xadj[u + 1] - row;
+
   for (u32 s = lane; s < kHashSlots; s += kWave) {
     hkeys[s] = kInvalid;
     hvals[s] = 0;
@@ -939,6 +924,8 @@ __global__ void k_phase_m_c(
     if (best.have && best.c != cur) {
       slots[p - pos_lo] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
     }
+  }
+  __threadfence_block(); // hash reuse across grid-stride iterations
   }
 }
 
@@ -1855,6 +1842,11 @@ struct kmp_lp_t {
   // phase buffers
   Prop *d_slots = nullptr; // C
   Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
+  u64 *d_m_slots = nullptr; // C (per-position M-candidate records)
+  u64 *d_m_list = nullptr;  // C (compacted)
+  u32 *d_m_count = nullptr;
+  void *d_m_select_temp = nullptr;
+  size_t m_select_temp_bytes = 0;
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr;
   u32 *d_l_off = nullptr;   // l_cap + 1 (slice prefix)
@@ -2015,6 +2007,9 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   const u32 C = e->C;
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
+  HIP_CHECK(hipMalloc(&e->d_m_slots, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_m_list, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
   HIP_CHECK(hipMalloc(&e->d_l_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
@@ -2044,6 +2039,10 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
       nullptr, e->select_temp_bytes, e->d_slots, e->d_props, e->d_prop_count, C, PropValid()
   ));
   HIP_CHECK(hipMalloc(&e->d_select_temp, e->select_temp_bytes));
+  HIP_CHECK(rocprim::select(
+      nullptr, e->m_select_temp_bytes, e->d_m_slots, e->d_m_list, e->d_m_count, C, CandValid()
+  ));
+  HIP_CHECK(hipMalloc(&e->d_m_select_temp, e->m_select_temp_bytes));
 
   {
     const u32 *xadj = kmp_graph_xadj(g);
@@ -2072,8 +2071,9 @@ void kmp_lp_free(kmp_lp_t *e) {
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
-                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_l_list,
-                  (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_m_slots,
+                  (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m_select_temp,
+                  (void *)e->d_l_list, (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
@@ -2201,6 +2201,7 @@ i64 kmp_lp_phase_a(
   // pre-mark every slot invalid (kernels only write actual proposals; a
   // skipped unit costs one byte read instead of 64 slot writes)
   HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * span, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_m_slots, 0xFF, sizeof(u64) * span, e->stream));
 
   hipEvent_t ev0, ev1;
   e->ev_pair(ev0, ev1);
@@ -2212,19 +2213,24 @@ i64 kmp_lp_phase_a(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_active, e->d_unit_active, e->d_slots, e->d_l_list, e->d_l_count
+        e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
-    // M: one wave per position
+    // compact the M work list (stable select; no append atomics), then one
+    // wave per listed vertex (grid-stride)
     {
+      size_t mtb = e->m_select_temp_bytes;
+      HIP_CHECK(rocprim::select(
+          e->d_m_select_temp, mtb, e->d_m_slots, e->d_m_list, e->d_m_count, span, CandValid(),
+          e->stream
+      ));
       const size_t lds =
           static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
       auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
       hipLaunchKernelGGL(
-          kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
-          e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->k, e->d_xadj,
-          e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_active,
-          e->d_unit_active, e->d_slots
+          kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
+          e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw,
+          e->d_m_list, e->d_m_count, e->d_slots
       );
       LAUNCH_CHECK();
     }
@@ -2276,17 +2282,22 @@ i64 kmp_lp_phase_a(
         k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_l_list, e->d_l_count
+        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_m_slots, e->d_l_list,
+        e->d_l_count
     );
     LAUNCH_CHECK();
     {
+      size_t mtb = e->m_select_temp_bytes;
+      HIP_CHECK(rocprim::select(
+          e->d_m_select_temp, mtb, e->d_m_slots, e->d_m_list, e->d_m_count, span, CandValid(),
+          e->stream
+      ));
       const size_t lds = static_cast<size_t>(threads / kWave) * 2 * kHashSlots * sizeof(u32);
       auto *kern = e->has_adjwgt ? k_phase_m_c<false> : k_phase_m_c<true>;
       hipLaunchKernelGGL(
-          kern, dim3(ceil_div(static_cast<u64>(span) * kWave, threads)), dim3(threads), lds,
-          e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree, e->maxw_uniform,
-          e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_active,
-          e->d_unit_active, e->d_favored, e->d_slots
+          kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, iseed,
+          e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels,
+          e->d_weights, e->d_m_list, e->d_m_count, e->d_favored, e->d_slots
       );
       LAUNCH_CHECK();
     }
