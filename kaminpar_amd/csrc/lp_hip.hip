@@ -1916,6 +1916,8 @@ struct kmp_lp_t {
   std::vector<hipEvent_t> ev_pool;
   size_t ev_used = 0;
 
+  hipEvent_t sync_ev = nullptr;
+
   void ev_pair(hipEvent_t &a, hipEvent_t &b) {
     if (ev_used + 2 > ev_pool.size()) {
       hipEvent_t x, y;
@@ -1933,6 +1935,11 @@ struct kmp_lp_t {
 namespace {
 
 u32 ceil_div(u64 a, u64 b) { return static_cast<u32>((a + b - 1) / b); }
+
+// Blocking hipStreamSynchronize waits cost ~100+ us each on a busy queue
+// (yield/interrupt based); the per-chunk control syncs are latency-critical,
+// so spin on an event instead (~5 us).
+void sync_spin(kmp_lp_t *e);
 
 void engine_alloc_k_buffers(kmp_lp_t *e, u32 k_or_n) {
   HIP_CHECK(hipMalloc(&e->d_seg_begin, sizeof(u32) * k_or_n));
@@ -1964,6 +1971,19 @@ void engine_free_k_buffers(kmp_lp_t *e) {
   }
 }
 
+void sync_spin(kmp_lp_t *e) {
+  HIP_CHECK(hipEventRecord(e->sync_ev, e->stream));
+  while (true) {
+    const hipError_t st = hipEventQuery(e->sync_ev);
+    if (st == hipSuccess) {
+      return;
+    }
+    if (st != hipErrorNotReady) {
+      HIP_CHECK(st);
+    }
+  }
+}
+
 } // namespace
 
 extern "C" {
@@ -1976,6 +1996,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   }
 
   auto *e = new kmp_lp_t();
+  (void)hipSetDeviceFlags(hipDeviceScheduleSpin); // ignore if context exists
   e->n = kmp_graph_n(g);
   e->m = kmp_graph_m(g);
   e->C = kmp::chunk_size_for(e->n);
@@ -1983,6 +2004,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   e->has_vwgt = kmp_graph_vwgt(g) != nullptr;
   e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
   HIP_CHECK(hipStreamCreate(&e->stream));
+  HIP_CHECK(hipEventCreateWithFlags(&e->sync_ev, hipEventDisableTiming));
 
   HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
   HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
@@ -2067,6 +2089,9 @@ void kmp_lp_free(kmp_lp_t *e) {
   HIP_CHECK(hipDeviceSynchronize());
   for (hipEvent_t ev : e->ev_pool) {
     (void)hipEventDestroy(ev);
+  }
+  if (e->sync_ev) {
+    (void)hipEventDestroy(e->sync_ev);
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
@@ -2342,7 +2367,7 @@ i64 kmp_lp_phase_a(
         e->h_changed, e->d_pool_overflow, sizeof(int), hipMemcpyDeviceToHost, e->stream
     ));
   }
-  HIP_CHECK(hipStreamSynchronize(e->stream));
+  sync_spin(e);
   float ms = 0;
   HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
   e->phase_a_ms += ms;
@@ -2459,7 +2484,7 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
       HIP_CHECK(
           hipMemcpyAsync(e->h_changed, e->d_changed, sizeof(int), hipMemcpyDeviceToHost, e->stream)
       );
-      HIP_CHECK(hipStreamSynchronize(e->stream));
+      sync_spin(e);
       if (!*e->h_changed) {
         break;
       }
@@ -2524,7 +2549,7 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
     HIP_CHECK(hipMemcpyAsync(&e->h_moves[3], e->d_emptied, sizeof(unsigned long long),
                              hipMemcpyDeviceToHost, e->stream));
   }
-  HIP_CHECK(hipStreamSynchronize(e->stream));
+  sync_spin(e);
   float cms = 0;
   HIP_CHECK(hipEventElapsedTime(&cms, cev0, cev1));
   e->commit_ms += cms;
