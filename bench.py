@@ -48,6 +48,9 @@ def main():
     ap.add_argument("--k", type=int, default=16)
     ap.add_argument("--iters", type=int, default=5)
     ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--workload", choices=["refine", "cluster"], default="refine",
+                    help="cluster mirrors the reference's own LP benchmark "
+                         "(shm_label_propagation_benchmark.cc: LP clustering)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -95,9 +98,16 @@ def main():
     k = args.k
     part0 = ka.random_partition(g.n, k, seed=5)
     mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
-    cut0 = g.edge_cut(part0)
+    cut0 = g.edge_cut(part0) if args.workload == "refine" else 0
 
     eng = ka.LpEngine(g)
+
+    if args.workload == "cluster":
+        # max cluster weight per the coarsening formula
+        # (max_cluster_weights.h:18-46, EPSILON_BLOCK_WEIGHT, contraction
+        # limit 2000, eps=0.03)
+        import math
+        mcw = int(0.03 * g.total_node_weight / min(max(g.n // 2000, 2), k))
 
     from kaminpar_amd.multi import TorchComm, refine_dist
 
@@ -108,6 +118,9 @@ def main():
         eng.refine_begin(k, mbw, part0, seed=args.seed)
 
     def one_step():
+        if args.workload == "cluster":
+            nc, _clus, stats = eng.cluster(mcw, seed=args.seed, iters=args.iters)
+            return nc, stats
         if comm is None:
             # timed region: reset (D2D) + sweeps; no host transfers, no cut
             eng.reset()
@@ -144,7 +157,7 @@ def main():
         last_cut = cut
     barrier_sync()
     t_end = time.time()
-    if world == 1:
+    if world == 1 and args.workload == "refine":
         # cut + label download happen once, outside the timed region
         last_cut, _part, _stats = eng.refine_end()
 
@@ -174,7 +187,7 @@ def main():
         with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
                                "profiles", "pmc_traffic.json")) as fh:
             tbl = json.load(fh)
-        entry = tbl.get(f"rmat{args.scale}_k{args.k}_lp_refine")
+        entry = tbl.get(f"rmat{args.scale}_k{args.k}_lp_{args.workload}")
         if entry:
             traffic = entry["fetch_bytes_per_launch"] + entry["write_bytes_per_launch"]
     except OSError:
@@ -208,7 +221,7 @@ def main():
             "dtype": "int32",
             "data": "synthetic",
             "config": {
-                "workload": f"rmat{args.scale}_k{args.k}_lp_refine",
+                "workload": f"rmat{args.scale}_k{args.k}_lp_{args.workload}",
                 "n": int(g.n),
                 "arcs": int(g.m),
                 "k": k,
