@@ -151,6 +151,7 @@ __global__ void k_phase_s(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
+    const uint16_t *__restrict__ labels16,
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots,
@@ -220,7 +221,7 @@ __global__ void k_phase_s(
   i32 w = 0;
   if (slot < deg) {
     const u32 v = adjncy[row + slot];
-    c = labels[v];
+    c = labels16[v]; // u16 shadow: k <= 2048, half the gather footprint
     w = adjwgt ? adjwgt[row + slot] : 1;
   }
 
@@ -287,6 +288,7 @@ __global__ void k_phase_m(
     const i32 *__restrict__ adjwgt,
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
+    const uint16_t *__restrict__ labels16,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const u64 *__restrict__ m_list,
@@ -318,7 +320,7 @@ __global__ void k_phase_m(
   for (u32 e = lane; e < deg; e += kWave) {
     const u32 v = adjncy[row + e];
     const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-    atomicAdd(&gains[rep_off + labels[v]], w);
+    atomicAdd(&gains[rep_off + labels16[v]], w);
   }
   __threadfence_block();
 
@@ -399,7 +401,7 @@ __global__ void k_phase_l_acc(
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
-    const u32 *__restrict__ labels,
+    const uint16_t *__restrict__ labels16,
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     u32 l_cap,
@@ -438,7 +440,7 @@ __global__ void k_phase_l_acc(
     for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
       const u32 v = adjncy[row + e];
       const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-      atomicAdd(&hist[rep_off + labels[v]], w);
+      atomicAdd(&hist[rep_off + labels16[v]], w);
     }
     __syncthreads();
     i32 *grow = l_gains + static_cast<size_t>(vid) * k;
@@ -554,6 +556,7 @@ __global__ void k_phase_l_direct(
     const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
+    const uint16_t *__restrict__ labels16,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const u64 *__restrict__ l_list,
@@ -583,7 +586,7 @@ __global__ void k_phase_l_direct(
     for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
       const u32 v = adjncy[row + e];
       const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-      atomicAdd(&gains[rep_off + labels[v]], w);
+      atomicAdd(&gains[rep_off + labels16[v]], w);
     }
     __syncthreads();
 
@@ -1613,6 +1616,7 @@ __global__ void k_apply(
     const u32 *__restrict__ seg_begin,
     const u32 *__restrict__ prefix_len,
     u32 *__restrict__ labels,
+    uint16_t *__restrict__ labels16, // refine shadow; null for clustering
     u32 *__restrict__ admitted_flags, // per sorted index
     unsigned long long *__restrict__ moves
 ) {
@@ -1623,7 +1627,11 @@ __global__ void k_apply(
     const bool admitted = (i - seg_begin[to]) < prefix_len[to];
     admitted_flags[i] = admitted ? 1u : 0u;
     if (admitted) {
-      labels[props[order[i]].u] = to;
+      const u32 u = props[order[i]].u;
+      labels[u] = to;
+      if (labels16 != nullptr) {
+        labels16[u] = static_cast<uint16_t>(to);
+      }
       local = 1;
     }
   }
@@ -1751,6 +1759,15 @@ __global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep
 // Grid-stride with a per-workgroup LDS histogram (one global atomic per
 // cluster per WG instead of one per vertex -- same-address global atomics
 // serialize at ~11ns).
+__global__ void k_sync_labels16(
+    u32 n, const u32 *__restrict__ labels, uint16_t *__restrict__ labels16
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u < n) {
+    labels16[u] = static_cast<uint16_t>(labels[u]);
+  }
+}
+
 __global__ void k_init_weights(
     u32 n,
     u32 k,
@@ -1833,7 +1850,10 @@ struct kmp_lp_t {
 
   // device LP state
   u32 *d_labels = nullptr;
-  u32 *d_labels0 = nullptr; // initial labels (for kmp_lp_reset)
+  u32 *d_labels0 = nullptr;   // initial labels (for kmp_lp_reset)
+  uint16_t *d_labels16 = nullptr; // u16 shadow for refine gathers (k <= 2048
+                                  // fits; halves the gather cache footprint:
+                                  // scale-26 labels 128 MB -> L3-resident)
   i64 *d_weights = nullptr;
   i64 *d_maxw = nullptr;
   uint8_t *d_active = nullptr;
@@ -2023,6 +2043,7 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
 
   HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels0, sizeof(u32) * e->n));
+  HIP_CHECK(hipMalloc(&e->d_labels16, sizeof(uint16_t) * e->n));
   HIP_CHECK(hipMalloc(&e->d_active, e->n));
   HIP_CHECK(hipMalloc(&e->d_unit_active, kmp::num_units(e->n)));
 
@@ -2095,7 +2116,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props, (void *)e->d_m_slots,
                   (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m_select_temp,
                   (void *)e->d_l_list, (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
@@ -2198,6 +2219,11 @@ int kmp_lp_refine_begin(
   HIP_CHECK(hipMemsetAsync(e->d_weights, 0, sizeof(i64) * k, e->stream));
   {
     const u32 threads = 256;
+    hipLaunchKernelGGL(
+        k_sync_labels16, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
+        e->d_labels, e->d_labels16
+    );
+    LAUNCH_CHECK();
     const size_t lds = static_cast<size_t>(k) * sizeof(unsigned long long);
     hipLaunchKernelGGL(
         k_init_weights, dim3(2048), dim3(threads), lds, e->stream, e->n, k, e->d_labels, e->d_vwgt,
@@ -2238,7 +2264,8 @@ i64 kmp_lp_phase_a(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list, e->d_l_count
+        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list,
+        e->d_l_count
     );
     LAUNCH_CHECK();
     // compact the M work list (stable select; no append atomics), then one
@@ -2254,8 +2281,8 @@ i64 kmp_lp_phase_a(
       auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
       hipLaunchKernelGGL(
           kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
-          e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw,
-          e->d_m_list, e->d_m_count, e->d_slots
+          e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
+          e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
       );
       LAUNCH_CHECK();
     }
@@ -2277,7 +2304,8 @@ i64 kmp_lp_phase_a(
         auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
         hipLaunchKernelGGL(
             kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
-            e->d_adjwgt, e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+            e->d_adjwgt, e->d_labels16, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off,
+            e->d_l_gains
         );
         LAUNCH_CHECK();
       }
@@ -2295,8 +2323,8 @@ i64 kmp_lp_phase_a(
         auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
         hipLaunchKernelGGL(
             kern, dim3(512), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
-            e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
-            e->d_l_count, e->l_cap, e->d_slots
+            e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
+            e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
         );
         LAUNCH_CHECK();
       }
@@ -2515,7 +2543,8 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
     }
     hipLaunchKernelGGL(
         k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
-        e->d_prefix_len, e->d_labels, e->d_admitted_flags, e->d_moves
+        e->d_prefix_len, e->d_labels, big_k ? nullptr : e->d_labels16, e->d_admitted_flags,
+        e->d_moves
     );
     LAUNCH_CHECK();
   }
@@ -2598,6 +2627,11 @@ int kmp_lp_reset(kmp_lp_t *e) {
   e->ev_used = 0;
   {
     const u32 threads = 256;
+    hipLaunchKernelGGL(
+        k_sync_labels16, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
+        e->d_labels, e->d_labels16
+    );
+    LAUNCH_CHECK();
     const size_t lds = static_cast<size_t>(e->k) * sizeof(unsigned long long);
     hipLaunchKernelGGL(
         k_init_weights, dim3(2048), dim3(threads), lds, e->stream, e->n, e->k, e->d_labels,
