@@ -6,10 +6,10 @@ fails loudly if the HIP device or the native library is missing -- there is
 no CPU fallback on the product path (the CPU oracle under oracle/ is test
 infrastructure only).
 
-Interop note: PyTorch wheels bundle their own HIP runtime. When combining
-this library with torch GPU tensors in one process (multi-GPU sharding),
-initialize torch.cuda FIRST -- the engine then binds torch's already-loaded
-runtime (same SONAME). Engine-first leaves torch unable to see the GPU.
+Interop note: PyTorch wheels bundle their own HIP runtime; this module
+preloads it (when torch is installed) before loading the native library so
+the whole process shares one runtime regardless of import/initialization
+order.
 """
 
 import ctypes
@@ -21,12 +21,35 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 _LIB_PATH = os.path.join(_HERE, "libkaminpar_lp.so")
 
 
+def _preload_torch_hip_runtime():
+    """Bind everything to ONE HIP runtime.
+
+    PyTorch wheels bundle their own libamdhip64 and load it via absolute
+    RPATH paths, ignoring an already-loaded system copy -- two HSA clients in
+    one process leave whichever initializes second unable to see the GPU.
+    Preloading torch's copy under its SONAME makes this library resolve to
+    it, and torch's later absolute-path load maps the same file. Without
+    torch installed this is a no-op and the system runtime is used.
+    """
+    try:
+        import importlib.util
+
+        spec = importlib.util.find_spec("torch")
+        if spec and spec.origin:
+            cand = os.path.join(os.path.dirname(spec.origin), "lib", "libamdhip64.so")
+            if os.path.exists(cand):
+                ctypes.CDLL(cand, mode=ctypes.RTLD_GLOBAL)
+    except Exception:
+        pass  # fall back to the system runtime
+
+
 def _load():
     if not os.path.exists(_LIB_PATH):
         raise ImportError(
             f"kaminpar_amd: native library not found at {_LIB_PATH}; "
             "build it with __graft_entry__.build() or kaminpar_amd/csrc/build.sh"
         )
+    _preload_torch_hip_runtime()
     lib = ctypes.CDLL(_LIB_PATH)
 
     u32, u64, i32, i64 = ctypes.c_uint32, ctypes.c_uint64, ctypes.c_int32, ctypes.c_int64
