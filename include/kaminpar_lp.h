@@ -173,3 +173,22 @@ int kmp_lp_get_stats(kmp_lp_t *e, kmp_lp_stats_t *stats); /* no cut/download */
 #endif
 
 #endif /* KAMINPAR_LP_H */
+/* ---------------------------------------------------------------- v1.1 ---
+ * Cluster contraction on the GPU (restates
+ * kaminpar-shm/coarsening/contraction/ semantics: coarse id = prefix rank of
+ * occupied cluster ids, intra-cluster edges dropped, parallel edges merged
+ * with summed weights, coarse adjacency sorted by (cu, cv)). Returns the
+ * coarse node count; *coarse_out receives a new host graph handle and
+ * mapping_out (n entries) the fine->coarse projection. -1 on error. */
+#ifdef __cplusplus
+extern "C" {
+#endif
+int64_t kmp_contract(
+    kmp_lp_t *e,
+    const uint32_t *clustering,
+    uint32_t *mapping_out,
+    kmp_graph_t **coarse_out
+);
+#ifdef __cplusplus
+}
+#endif

@@ -107,6 +107,8 @@ def _load():
     lib.kmp_lp_run_sweeps.argtypes = [vp, ctypes.c_int]
     lib.kmp_lp_get_stats.restype = ctypes.c_int
     lib.kmp_lp_get_stats.argtypes = [vp, vp]
+    lib.kmp_contract.restype = i64
+    lib.kmp_contract.argtypes = [vp, p(u32), p(u32), ctypes.POINTER(vp)]
     return lib
 
 
@@ -288,6 +290,18 @@ class LpEngine:
         stats = Stats()
         _lib.kmp_lp_get_stats(self._h, ctypes.byref(stats))
         return stats
+
+    def contract(self, clustering):
+        """Contract a clustering into the coarse graph (GPU).
+
+        Returns (coarse_graph: Graph, mapping: np.ndarray[u32])."""
+        clus = np.ascontiguousarray(clustering, dtype=np.uint32)
+        mapping = np.zeros(self._graph.n, dtype=np.uint32)
+        out = ctypes.c_void_p()
+        c_n = _lib.kmp_contract(self._h, _u32p(clus), _u32p(mapping), ctypes.byref(out))
+        if c_n < 0:
+            raise RuntimeError("kmp_contract failed")
+        return Graph(out.value), mapping
 
     def refine_end(self):
         part = np.zeros(self._graph.n, dtype=np.uint32)

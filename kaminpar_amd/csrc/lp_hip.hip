@@ -2950,3 +2950,228 @@ i64 kmp_lp_cluster(
 }
 
 } // extern "C"
+
+// ==================== cluster contraction (GPU) ====================
+// Restates kaminpar-shm/coarsening/contraction/ semantics (see
+// kmp_contract in include/kaminpar_lp.h). All steps are order-free integer
+// sums / stable sorts, so the result is deterministic and bit-identical to
+// the CPU oracle and (canonically sorted) to the reference implementation.
+
+namespace {
+
+__global__ void k_mark_clusters(
+    u32 n, const u32 *__restrict__ clus, u32 *__restrict__ rank
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u < n) {
+    rank[clus[u]] = 1;
+  }
+}
+
+__global__ void k_map_ranks(
+    u32 n, const u32 *__restrict__ clus, const u32 *__restrict__ rank, u32 *__restrict__ map
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u < n) {
+    map[u] = rank[clus[u]] - 1;
+  }
+}
+
+__global__ void k_coarse_vwgt(
+    u32 n, const u32 *__restrict__ map, const i32 *__restrict__ vwgt, i32 *__restrict__ cvw
+) {
+  const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u < n) {
+    atomicAdd(&cvw[map[u]], vwgt ? vwgt[u] : 1);
+  }
+}
+
+// One 16-lane subgroup per vertex (grid-stride): emit (cu<<32|cv) keys per
+// arc; intra-cluster arcs get the ~0 sentinel (sorted to the end, dropped).
+__global__ void k_arc_keys(
+    u32 n,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const u32 *__restrict__ map,
+    u64 *__restrict__ keys,
+    i32 *__restrict__ vals
+) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 sub = lane >> 4;
+  const u32 slot = lane & 15;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
+  for (u32 base = wave_id * 4; base < n; base += num_waves * 4) {
+    const u32 u = base + sub;
+    if (u >= n) {
+      continue;
+    }
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+    const u32 cu = map[u];
+    for (u32 e = slot; e < deg; e += 16) {
+      const u32 cv = map[adjncy[row + e]];
+      keys[row + e] = (cu == cv) ? ~0ull : ((static_cast<u64>(cu) << 32) | cv);
+      vals[row + e] = adjwgt ? adjwgt[row + e] : 1;
+    }
+  }
+}
+
+__global__ void k_coarse_hist(
+    u32 c_m, const u64 *__restrict__ ukeys, u32 *__restrict__ cxadj
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < c_m) {
+    atomicAdd(&cxadj[(ukeys[i] >> 32) + 1], 1u);
+  }
+}
+
+__global__ void k_coarse_adj(
+    u32 c_m, const u64 *__restrict__ ukeys, u32 *__restrict__ cadj
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < c_m) {
+    cadj[i] = static_cast<u32>(ukeys[i] & 0xFFFFFFFFu);
+  }
+}
+
+} // namespace
+
+extern "C" {
+
+i64 kmp_contract(
+    kmp_lp_t *e, const u32 *clustering, u32 *mapping_out, kmp_graph_t **coarse_out
+) {
+  const u32 n = e->n;
+  const u64 m = e->m;
+  if (m > 0xFFFFFFFFull) {
+    fprintf(stderr, "kaminpar_amd: kmp_contract supports m < 2^32 arcs\n");
+    return -1;
+  }
+  const u32 threads = 256;
+  const u32 ngrid = ceil_div(n, threads);
+  hipStream_t s = e->stream;
+
+  u32 *d_clus = nullptr, *d_rank = nullptr, *d_map = nullptr, *d_cxadj = nullptr,
+      *d_cadj = nullptr;
+  i32 *d_cvw = nullptr, *d_vals[2] = {nullptr, nullptr}, *d_usums = nullptr;
+  u64 *d_keys[2] = {nullptr, nullptr}, *d_ukeys = nullptr;
+  u32 *d_uniq = nullptr;
+  HIP_CHECK(hipMalloc(&d_clus, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_rank, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_map, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_cvw, sizeof(i32) * n));
+  HIP_CHECK(hipMalloc(&d_keys[0], sizeof(u64) * m));
+  HIP_CHECK(hipMalloc(&d_keys[1], sizeof(u64) * m));
+  HIP_CHECK(hipMalloc(&d_vals[0], sizeof(i32) * m));
+  HIP_CHECK(hipMalloc(&d_vals[1], sizeof(i32) * m));
+  HIP_CHECK(hipMalloc(&d_ukeys, sizeof(u64) * m));
+  HIP_CHECK(hipMalloc(&d_usums, sizeof(i32) * m));
+  HIP_CHECK(hipMalloc(&d_uniq, sizeof(u32)));
+
+  HIP_CHECK(hipMemcpyAsync(d_clus, clustering, sizeof(u32) * n, hipMemcpyHostToDevice, s));
+  HIP_CHECK(hipMemsetAsync(d_rank, 0, sizeof(u32) * n, s));
+  hipLaunchKernelGGL(k_mark_clusters, dim3(ngrid), dim3(threads), 0, s, n, d_clus, d_rank);
+  LAUNCH_CHECK();
+
+  void *tmp = nullptr;
+  size_t tmp_bytes = 0;
+  HIP_CHECK(rocprim::inclusive_scan(nullptr, tmp_bytes, d_rank, d_rank, n, rocprim::plus<u32>()));
+  HIP_CHECK(hipMalloc(&tmp, tmp_bytes));
+  HIP_CHECK(rocprim::inclusive_scan(tmp, tmp_bytes, d_rank, d_rank, n, rocprim::plus<u32>(), s));
+  u32 c_n = 0;
+  HIP_CHECK(hipMemcpyAsync(&c_n, d_rank + n - 1, sizeof(u32), hipMemcpyDeviceToHost, s));
+  hipLaunchKernelGGL(k_map_ranks, dim3(ngrid), dim3(threads), 0, s, n, d_clus, d_rank, d_map);
+  LAUNCH_CHECK();
+  HIP_CHECK(hipMemsetAsync(d_cvw, 0, sizeof(i32) * n, s));
+  hipLaunchKernelGGL(k_coarse_vwgt, dim3(ngrid), dim3(threads), 0, s, n, d_map, e->d_vwgt, d_cvw);
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_arc_keys, dim3(16384), dim3(threads), 0, s, n, e->d_xadj, e->d_adjncy, e->d_adjwgt, d_map,
+      d_keys[0], d_vals[0]
+  );
+  LAUNCH_CHECK();
+
+  rocprim::double_buffer<u64> kb(d_keys[0], d_keys[1]);
+  rocprim::double_buffer<i32> vb(d_vals[0], d_vals[1]);
+  void *tmp2 = nullptr;
+  size_t tmp2_bytes = 0;
+  HIP_CHECK(rocprim::radix_sort_pairs(nullptr, tmp2_bytes, kb, vb, m, 0, 64));
+  HIP_CHECK(hipMalloc(&tmp2, tmp2_bytes));
+  HIP_CHECK(rocprim::radix_sort_pairs(tmp2, tmp2_bytes, kb, vb, m, 0, 64, s));
+
+  void *tmp3 = nullptr;
+  size_t tmp3_bytes = 0;
+  HIP_CHECK(rocprim::reduce_by_key(
+      nullptr, tmp3_bytes, kb.current(), vb.current(), m, d_ukeys, d_usums, d_uniq,
+      rocprim::plus<i32>(), rocprim::equal_to<u64>()
+  ));
+  HIP_CHECK(hipMalloc(&tmp3, tmp3_bytes));
+  HIP_CHECK(rocprim::reduce_by_key(
+      tmp3, tmp3_bytes, kb.current(), vb.current(), m, d_ukeys, d_usums, d_uniq,
+      rocprim::plus<i32>(), rocprim::equal_to<u64>(), s
+  ));
+  u32 uniq = 0;
+  HIP_CHECK(hipMemcpyAsync(&uniq, d_uniq, sizeof(u32), hipMemcpyDeviceToHost, s));
+  u64 last_key = 0;
+  HIP_CHECK(hipStreamSynchronize(s));
+  if (uniq > 0) {
+    HIP_CHECK(hipMemcpy(&last_key, d_ukeys + uniq - 1, sizeof(u64), hipMemcpyDeviceToHost));
+  }
+  const u32 c_m = (uniq > 0 && last_key == ~0ull) ? uniq - 1 : uniq;
+
+  HIP_CHECK(hipMalloc(&d_cxadj, sizeof(u32) * (c_n + 1)));
+  HIP_CHECK(hipMalloc(&d_cadj, sizeof(u32) * (c_m > 0 ? c_m : 1)));
+  HIP_CHECK(hipMemsetAsync(d_cxadj, 0, sizeof(u32) * (c_n + 1), s));
+  if (c_m > 0) {
+    hipLaunchKernelGGL(
+        k_coarse_hist, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, d_cxadj
+    );
+    LAUNCH_CHECK();
+    hipLaunchKernelGGL(
+        k_coarse_adj, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, d_cadj
+    );
+    LAUNCH_CHECK();
+  }
+  void *tmp4 = nullptr;
+  size_t tmp4_bytes = 0;
+  HIP_CHECK(rocprim::inclusive_scan(
+      nullptr, tmp4_bytes, d_cxadj, d_cxadj, c_n + 1, rocprim::plus<u32>()
+  ));
+  HIP_CHECK(hipMalloc(&tmp4, tmp4_bytes));
+  HIP_CHECK(rocprim::inclusive_scan(
+      tmp4, tmp4_bytes, d_cxadj, d_cxadj, c_n + 1, rocprim::plus<u32>(), s
+  ));
+
+  // download
+  std::vector<u32> h_cxadj(c_n + 1), h_cadj(c_m);
+  std::vector<i32> h_cvw(c_n), h_cwgt(c_m);
+  HIP_CHECK(hipMemcpyAsync(mapping_out, d_map, sizeof(u32) * n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), d_cxadj, sizeof(u32) * (c_n + 1),
+                           hipMemcpyDeviceToHost, s));
+  if (c_m > 0) {
+    HIP_CHECK(hipMemcpyAsync(h_cadj.data(), d_cadj, sizeof(u32) * c_m, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(
+        hipMemcpyAsync(h_cwgt.data(), d_usums, sizeof(i32) * c_m, hipMemcpyDeviceToHost, s)
+    );
+  }
+  HIP_CHECK(hipMemcpyAsync(h_cvw.data(), d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+
+  *coarse_out = kmp_graph_from_csr(
+      c_n, c_m, h_cxadj.data(), h_cadj.data(), h_cvw.data(), c_m ? h_cwgt.data() : nullptr
+  );
+
+  for (void *p : {(void *)d_clus, (void *)d_rank, (void *)d_map, (void *)d_cvw,
+                  (void *)d_keys[0], (void *)d_keys[1], (void *)d_vals[0], (void *)d_vals[1],
+                  (void *)d_ukeys, (void *)d_usums, (void *)d_uniq, (void *)d_cxadj,
+                  (void *)d_cadj, tmp, tmp2, tmp3, tmp4}) {
+    if (p) {
+      (void)hipFree(p);
+    }
+  }
+  return static_cast<i64>(c_n);
+}
+
+} // extern "C"

@@ -191,3 +191,84 @@ int64_t kref_edge_cut(
 }
 
 } // extern "C"
+
+#include "kaminpar-shm/coarsening/contraction/cluster_contraction.h"
+
+extern "C" {
+
+// Contract a clustering with the reference implementation; returns the
+// coarse node count and fills mapping + the coarse CSR with each coarse
+// adjacency list SORTED by target (the reference's own in-list order is
+// scheduling-dependent; sorting canonicalizes for comparison).
+int64_t kref_contract(
+    const uint32_t n,
+    const uint64_t m,
+    const uint32_t *xadj,
+    const uint32_t *adjncy,
+    const int32_t *vwgt,
+    const int32_t *adjwgt,
+    const uint32_t *clustering,
+    uint32_t *mapping_out,
+    uint32_t *c_xadj,
+    uint32_t *c_adjncy,
+    int32_t *c_vwgt,
+    int32_t *c_adjwgt,
+    uint64_t *c_m_out
+) {
+  Graph graph(std::make_unique<CSRGraph>(make_csr_graph(n, m, xadj, adjncy, vwgt, adjwgt)));
+
+  StaticArray<NodeID> clus(n);
+  for (uint32_t u = 0; u < n; ++u) {
+    clus[u] = clustering[u];
+  }
+
+  Context ctx = create_default_context();
+  auto coarse = contract_clustering(graph, std::move(clus), ctx.coarsening.contraction);
+
+  const auto &cg = coarse->get();
+  const CSRGraph &csr = *dynamic_cast<const CSRGraph *>(cg.underlying_graph());
+  const uint32_t c_n = csr.n();
+  const uint64_t c_m = csr.m();
+
+  // mapping: project a fine identity labelling upward is not exposed;
+  // recover it by projecting coarse ids down (project_up maps coarse->fine)
+  {
+    std::vector<BlockID> coarse_ids(c_n);
+    for (uint32_t c = 0; c < c_n; ++c) {
+      coarse_ids[c] = c;
+    }
+    std::vector<BlockID> fine_ids(n);
+    coarse->project_up(
+        std::span<const BlockID>(coarse_ids.data(), c_n),
+        std::span<BlockID>(fine_ids.data(), n)
+    );
+    for (uint32_t u = 0; u < n; ++u) {
+      mapping_out[u] = fine_ids[u];
+    }
+  }
+
+  for (uint32_t c = 0; c <= c_n; ++c) {
+    c_xadj[c] = csr.raw_nodes()[c];
+  }
+  for (uint32_t c = 0; c < c_n; ++c) {
+    c_vwgt[c] = csr.node_weight(c);
+  }
+  // sort each adjacency list by target for canonical comparison
+  for (uint32_t c = 0; c < c_n; ++c) {
+    std::vector<std::pair<uint32_t, int32_t>> row;
+    for (uint64_t e = csr.raw_nodes()[c]; e < csr.raw_nodes()[c + 1]; ++e) {
+      row.emplace_back(csr.raw_edges()[e], csr.edge_weight(e));
+    }
+    std::sort(row.begin(), row.end());
+    uint64_t e = csr.raw_nodes()[c];
+    for (const auto &[v, w] : row) {
+      c_adjncy[e] = v;
+      c_adjwgt[e] = w;
+      ++e;
+    }
+  }
+  *c_m_out = c_m;
+  return static_cast<int64_t>(c_n);
+}
+
+} // extern "C"

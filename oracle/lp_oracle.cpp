@@ -816,3 +816,98 @@ i64 kmp_oracle_lp_cluster(
 }
 
 } // extern "C"
+
+// ======================================================================
+// Cluster contraction (restates kaminpar-shm/coarsening/contraction/
+// cluster_contraction_preprocessing.cc:17-52 mapping semantics -- coarse id
+// = prefix rank of occupied cluster ids -- and the coarse-graph
+// construction of cluster_contraction.cc: intra-cluster edges dropped,
+// parallel edges merged with summed weights, node weights summed).
+// Coarse adjacency is emitted sorted by (coarse_u, coarse_v): a canonical
+// order shared with the GPU implementation (the reference's own adjacency
+// order is scheduling-dependent; sorted comparison is used for pinning).
+// ======================================================================
+extern "C" {
+
+// Returns coarse node count; fills mapping[n], c_xadj[c_n+1], c_adjncy,
+// c_vwgt, c_adjwgt (buffers sized for the fine graph are always enough).
+// c_m_out receives the coarse arc count.
+i64 kmp_oracle_contract(
+    u32 n,
+    u64 m,
+    const u32 *xadj,
+    const u32 *adjncy,
+    const i32 *vwgt,
+    const i32 *adjwgt,
+    const u32 *clustering,
+    u32 *mapping,
+    u32 *c_xadj,
+    u32 *c_adjncy,
+    i32 *c_vwgt,
+    i32 *c_adjwgt,
+    u64 *c_m_out
+) {
+  // coarse ids: prefix rank of occupied cluster ids
+  std::vector<u32> rank(n, 0);
+  for (u32 u = 0; u < n; ++u) {
+    rank[clustering[u]] = 1;
+  }
+  u32 acc = 0;
+  for (u32 c = 0; c < n; ++c) {
+    const u32 occ = rank[c];
+    rank[c] = acc;
+    acc += occ;
+  }
+  const u32 c_n = acc;
+  for (u32 u = 0; u < n; ++u) {
+    mapping[u] = rank[clustering[u]];
+  }
+
+  // coarse node weights
+  for (u32 c = 0; c < c_n; ++c) {
+    c_vwgt[c] = 0;
+  }
+  for (u32 u = 0; u < n; ++u) {
+    c_vwgt[mapping[u]] += vwgt ? vwgt[u] : 1;
+  }
+
+  // inter-cluster arcs, merged and sorted by (cu, cv)
+  std::vector<std::pair<u64, i64>> arcs;
+  arcs.reserve(m);
+  for (u32 u = 0; u < n; ++u) {
+    const u32 cu = mapping[u];
+    for (u64 e = xadj[u]; e < xadj[u + 1]; ++e) {
+      const u32 cv = mapping[adjncy[e]];
+      if (cu != cv) {
+        arcs.emplace_back((static_cast<u64>(cu) << 32) | cv, adjwgt ? adjwgt[e] : 1);
+      }
+    }
+  }
+  std::sort(arcs.begin(), arcs.end(), [](const auto &a, const auto &b) {
+    return a.first < b.first;
+  });
+
+  u64 c_m = 0;
+  for (u32 c = 0; c <= c_n; ++c) {
+    c_xadj[c] = 0;
+  }
+  for (size_t i = 0; i < arcs.size();) {
+    const u64 key = arcs[i].first;
+    i64 w = 0;
+    while (i < arcs.size() && arcs[i].first == key) {
+      w += arcs[i].second;
+      ++i;
+    }
+    c_adjncy[c_m] = static_cast<u32>(key & 0xFFFFFFFFu);
+    c_adjwgt[c_m] = static_cast<i32>(w);
+    ++c_xadj[(key >> 32) + 1];
+    ++c_m;
+  }
+  for (u32 c = 0; c < c_n; ++c) {
+    c_xadj[c + 1] += c_xadj[c];
+  }
+  *c_m_out = c_m;
+  return static_cast<i64>(c_n);
+}
+
+} // extern "C"

@@ -302,3 +302,67 @@ def test_sharded_phase_api_matches_monolithic(oracle):
     cut2, part2, _ = eng.refine_end()
     assert cut2 == cut_ref
     assert (part2 == part_ref).all()
+
+
+def test_contract_parity(oracle):
+    """GPU contraction bit-identical to the oracle restatement (which is
+    itself bit-identical to the reference's contract_clustering after
+    canonical sorting -- tests/test_contraction.py)."""
+    import ctypes
+
+    from helpers import i32p, oracle_cluster, u32p, u64p
+
+    oracle.kmp_oracle_contract.restype = ctypes.c_int64
+    g = ka.Graph.rmat(13, 8, seed=7)
+    n, m = g.n, g.m
+    nc, clus, _ = oracle_cluster(oracle, g, 64, seed=2)
+
+    eng = ka.LpEngine(g)
+    cg, mapping = eng.contract(clus)
+
+    xadj = np.ascontiguousarray(g.xadj)
+    adjncy = np.ascontiguousarray(g.adjncy)
+    o_map = np.zeros(n, np.uint32)
+    o_xadj = np.zeros(n + 1, np.uint32)
+    o_adj = np.zeros(m, np.uint32)
+    o_vw = np.zeros(n, np.int32)
+    o_wg = np.zeros(m, np.int32)
+    o_cm = np.zeros(1, np.uint64)
+    o_cn = oracle.kmp_oracle_contract(
+        ctypes.c_uint32(n), ctypes.c_uint64(m), u32p(xadj), u32p(adjncy), None, None,
+        u32p(clus), u32p(o_map), u32p(o_xadj), u32p(o_adj), i32p(o_vw), i32p(o_wg),
+        u64p(o_cm))
+    cm = int(o_cm[0])
+
+    assert cg.n == o_cn and cg.m == cm
+    assert (mapping == o_map).all()
+    assert (np.asarray(cg.xadj) == o_xadj[: cg.n + 1]).all()
+    assert (np.asarray(cg.adjncy) == o_adj[:cm]).all()
+
+
+def test_multilevel_coarsening_chain(oracle):
+    """Full GPU coarsening loop: LP clustering -> contraction -> repeat until
+    the contraction limit, as the reference coarsener drives it
+    (abstract_cluster_coarsener.cc:98-228); weights conserved per level."""
+    g = ka.Graph.rmat(14, 8, seed=7)
+    total_w = g.total_node_weight
+    levels = []
+    cur = g
+    eng = ka.LpEngine(cur)
+    for level in range(6):
+        if cur.n <= 2000:
+            break
+        # coarsening cap per max_cluster_weights.h:18-46 (eps block weight)
+        k = 16
+        mcw = max(2, int(0.03 * total_w / min(max(cur.n // 2000, 2), k)))
+        nc, clus, _ = eng.cluster(mcw, seed=level + 1, iters=5)
+        coarse, mapping = eng.contract(clus)
+        assert coarse.n == nc
+        assert coarse.total_node_weight == total_w  # node weight conserved
+        # inter-cluster fine weight == total coarse edge weight
+        levels.append((cur.n, coarse.n))
+        if coarse.n >= cur.n:  # no shrink -> stop
+            break
+        cur = coarse
+        eng = ka.LpEngine(cur)
+    assert len(levels) >= 2 and levels[-1][1] < g.n // 4, levels
