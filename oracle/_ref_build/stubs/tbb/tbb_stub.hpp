@@ -142,7 +142,7 @@ public:
   };
   Range range() { return Range{begin(), end()}; }
 
-  template <typename BinaryOp> T combine(BinaryOp op) {
+  template <typename BinaryOp> T combine(BinaryOp op) const {
     T result{};
     bool first = true;
     for (auto &p : _instances) {
@@ -156,15 +156,30 @@ public:
     return result;
   }
 
-  template <typename UnaryOp> void combine_each(UnaryOp op) {
+  template <typename UnaryOp> void combine_each(UnaryOp op) const {
     for (auto &p : _instances) {
       op(*p);
     }
   }
 
+  // const iteration (some reference call sites combine on const ETS refs)
+  struct const_iterator {
+    typename std::deque<std::unique_ptr<T>>::const_iterator it;
+    const T &operator*() const { return **it; }
+    const T *operator->() const { return it->get(); }
+    const_iterator &operator++() {
+      ++it;
+      return *this;
+    }
+    bool operator!=(const const_iterator &o) const { return it != o.it; }
+    bool operator==(const const_iterator &o) const { return it == o.it; }
+  };
+  const_iterator begin() const { return const_iterator{_instances.begin()}; }
+  const_iterator end() const { return const_iterator{_instances.end()}; }
+
 private:
   std::function<T()> _factory;
-  std::deque<std::unique_ptr<T>> _instances;
+  mutable std::deque<std::unique_ptr<T>> _instances;
 };
 
 // ---- combinable ----
@@ -207,6 +222,9 @@ public:
 
   void push_back(const T &v) { _data.push_back(v); }
   void push_back(T &&v) { _data.push_back(std::move(v)); }
+  template <typename... Args> T &emplace_back(Args &&...args) {
+    return _data.emplace_back(std::forward<Args>(args)...);
+  }
 
   std::size_t size() const { return _data.size(); }
   bool empty() const { return _data.empty(); }
