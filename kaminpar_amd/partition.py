@@ -1,20 +1,27 @@
 """Multilevel k-way partitioning pipeline (BASELINE config 3).
 
-Mirrors the reference's basic multilevel scheme
-(kaminpar-shm/partitioning/deep/deep_multilevel.cc:55-66:
-uncoarsen(initial_partition(coarsen())) with the default preset's knobs:
-LP clustering with the EPSILON_BLOCK_WEIGHT cap
-(coarsening/max_cluster_weights.h:18-46, contraction limit 2000,
-presets.cc:185), cluster contraction, and LP refinement at every level):
+Mirrors the reference's multilevel scheme (coarsen -> initial partition ->
+uncoarsen+refine, kaminpar-shm/partitioning/): GPU LP clustering + GPU
+contraction per level, CPU initial partitioning on the coarsest graph, GPU
+LP refinement at every level.
 
-  - coarsening: GPU LP clustering + GPU contraction per level;
-  - initial partitioning: recursive greedy graph-growing bisection on the
-    coarsest graph (CPU; a simplified stand-in for the reference's
-    sequential initial-partitioner pool -- quality is validated against the
-    compiled reference's full pipeline, not claimed bit-parity);
-  - uncoarsening: project through the contraction mapping and run the GPU
-    LP refiner at each level (the deterministic schedule; balancers are not
-    implemented yet, so the initial bisection respects the caps strictly).
+Initial partitioning restates the reference's recipe in simplified form
+(kaminpar-shm/initial_partitioning/): recursive bisection where each
+bisection is greedy graph growing (initial_ggg_bipartitioner.cc) polished by
+two-way FM (initial_two_way_fm_refiner.cc: best-prefix rollback, 100
+fruitless moves, 5 passes), followed by a gain-aware overload balancer
+(refinement/balancer/overload_balancer.cc in spirit). It is NOT claimed
+bit-parity with the reference's initial-partitioner pool (which races many
+repetitions); quality is validated end-to-end against the compiled
+reference's full pipeline within a band (tests/golden/ref_golden_partition.json).
+
+Cluster-weight rule per level: the reference's EPSILON_BLOCK_WEIGHT formula
+(coarsening/max_cluster_weights.h:18-46) capped by its BLOCK_WEIGHT rule
+with multiplier 1/12 (presets.cc: initial_partitioning.coarsening
+.cluster_weight_limit = BLOCK_WEIGHT, multiplier = 1/12) so coarse vertices
+stay small enough relative to the block capacity for a feasible initial
+assignment -- without the cap, deep coarsening produces vertices weighing
+~25% of a block, which makes balanced bin-packing infeasible.
 """
 
 import numpy as np
@@ -22,84 +29,209 @@ import numpy as np
 from . import LpEngine
 
 
-def _greedy_bisect(xadj, adjncy, adjwgt, vwgt, n_total, nodes, target1, cap1):
-    """Split `nodes` (array of vertex ids) into (part1, part2): grow part1
-    from a max-degree seed by repeatedly absorbing the frontier vertex with
-    the highest connection into the region, until its weight reaches
-    target1. Deterministic (ties: smaller vertex id). Restates the idea of
-    the reference's GreedyGraphGrowingBipartitioner
-    (kaminpar-shm/initial_partitioning/initial_ggg_bipartitioner.cc), not
-    its exact queue schedule."""
-    nodes = np.asarray(nodes)
-    loc = np.full(n_total, -1, dtype=np.int64)
-    loc[nodes] = np.arange(len(nodes))
+def _subgraph_csr(xadj, adjncy, adjwgt, nodes, loc):
+    """Extract local CSR of the induced subgraph (local vertex ids)."""
+    rows_j = []
+    rows_w = []
+    sub_xadj = np.zeros(len(nodes) + 1, dtype=np.int64)
+    for i, u in enumerate(nodes):
+        e0, e1 = int(xadj[u]), int(xadj[u + 1])
+        j = loc[adjncy[e0:e1]]
+        sel = j >= 0
+        rows_j.append(j[sel])
+        if adjwgt is not None:
+            rows_w.append(adjwgt[e0:e1][sel].astype(np.int64))
+        sub_xadj[i + 1] = sub_xadj[i] + int(sel.sum())
+    sub_adj = np.concatenate(rows_j) if rows_j else np.zeros(0, np.int64)
+    if adjwgt is not None:
+        sub_w = np.concatenate(rows_w) if rows_w else np.zeros(0, np.int64)
+    else:
+        sub_w = np.ones(len(sub_adj), dtype=np.int64)
+    return sub_xadj, sub_adj, sub_w
 
-    # degree within the subgraph, for seed choice
-    mask = (loc[adjncy] >= 0).astype(np.int64)
-    cs = np.concatenate([[0], np.cumsum(mask)])
-    deg_in = cs[xadj[nodes + 1]] - cs[xadj[nodes]]
-    seed_best = deg_in == deg_in.max()
-    seed_idx = int(np.where(seed_best)[0][np.argmin(nodes[seed_best])])
 
-    in_region = np.zeros(len(nodes), dtype=bool)
-    gain = np.full(len(nodes), -1, dtype=np.int64)  # -1 = not frontier
+def _fm_refine_bisection(sub_xadj, sub_adj, sub_w, vw, side, cap1, cap2,
+                         max_passes=5, max_fruitless=100):
+    """Two-way FM with best-prefix rollback on a bisection (side=True is
+    part 1). Restates initial_two_way_fm_refiner.cc's schedule in simplified
+    form: repeated passes, each greedily moving the best-gain movable vertex
+    (ties: smallest id), locking it, tracking the best prefix."""
+    n = len(side)
+    w1 = int(vw[side].sum())
+    w2 = int(vw.sum()) - w1
+    for _ in range(max_passes):
+        # gains from scratch: external - internal connection
+        gains = np.zeros(n, dtype=np.int64)
+        for i in range(n):
+            e0, e1 = int(sub_xadj[i]), int(sub_xadj[i + 1])
+            nb = sub_adj[e0:e1]
+            w = sub_w[e0:e1]
+            cross = side[nb] != side[i]
+            gains[i] = int(w[cross].sum()) - int(w[~cross].sum())
+        locked = np.zeros(n, dtype=bool)
+        moves = []
+        cum = best = 0
+        best_len = 0
+        fruitless = 0
+        wa, wb = w1, w2
+        while fruitless < max_fruitless:
+            feas = (~locked) & (
+                (side & (wb + vw <= cap2)) | (~side & (wa + vw <= cap1))
+            )
+            if not feas.any():
+                break
+            masked = np.where(feas, gains, np.iinfo(np.int64).min)
+            i = int(np.argmax(masked))  # first occurrence = smallest id
+            cum += int(gains[i])
+            old = bool(side[i])
+            e0, e1 = int(sub_xadj[i]), int(sub_xadj[i + 1])
+            nb = sub_adj[e0:e1]
+            w = sub_w[e0:e1]
+            upd = ~locked[nb]
+            same = side[nb[upd]] == old
+            delta = np.where(same, 2 * w[upd], -2 * w[upd])
+            np.add.at(gains, nb[upd], delta)
+            gains[i] = -gains[i]
+            side[i] = not old
+            locked[i] = True
+            if old:
+                wa -= int(vw[i]); wb += int(vw[i])
+            else:
+                wa += int(vw[i]); wb -= int(vw[i])
+            moves.append(i)
+            if cum > best:
+                best = cum
+                best_len = len(moves)
+                fruitless = 0
+            else:
+                fruitless += 1
+        # rollback past the best prefix
+        for i in moves[best_len:]:
+            side[i] = not side[i]
+        w1 = int(vw[side].sum())
+        w2 = int(vw.sum()) - w1
+        if best <= 0:
+            break
+    return side
+
+
+def _greedy_grow(sub_xadj, sub_adj, sub_w, vw, target1, cap1, seed_rank=0):
+    """Greedy graph growing: grow part 1 from a high-degree seed (the
+    seed_rank-th vertex in descending-degree order -- repetitions use
+    different seeds, like the reference's bipartitioner pool), absorbing
+    the frontier vertex with the highest connection (ties: smallest id)."""
+    n = len(vw)
+    deg = np.diff(sub_xadj)
+    order = np.lexsort((np.arange(n), -deg))
+    seed_idx = int(order[seed_rank % n])
+
+    side = np.zeros(n, dtype=bool)
+    gain = np.full(n, -1, dtype=np.int64)  # -1 not frontier, -2 blocked/in
     w1 = 0
 
     def add(i):
         nonlocal w1
-        in_region[i] = True
-        w1 += int(vwgt[nodes[i]])
-        gain[i] = -2  # consumed / blocked
-        u = int(nodes[i])
-        e0, e1 = int(xadj[u]), int(xadj[u + 1])
-        j = loc[adjncy[e0:e1]]
-        sel = j >= 0
-        j = j[sel]
-        keep = ~in_region[j] & (gain[j] != -2)
-        j = j[keep]
-        if len(j) == 0:
+        side[i] = True
+        w1 += int(vw[i])
+        gain[i] = -2
+        e0, e1 = int(sub_xadj[i]), int(sub_xadj[i + 1])
+        j = sub_adj[e0:e1]
+        keep = ~side[j] & (gain[j] != -2)
+        j2 = j[keep]
+        if len(j2) == 0:
             return
-        gain[j[gain[j] < 0]] = 0
-        w = adjwgt[e0:e1][sel][keep] if adjwgt is not None else 1
-        np.add.at(gain, j, w)
+        gain[j2[gain[j2] < 0]] = 0
+        np.add.at(gain, j2, sub_w[e0:e1][keep])
 
     add(seed_idx)
     while w1 < target1:
-        cand = np.where(gain >= 0)[0]
+        cand = np.flatnonzero(gain >= 0)
         if len(cand) == 0:
-            # disconnected: seed a new component (smallest id outside)
-            rest = np.where(~in_region & (gain != -2))[0]
+            rest = np.flatnonzero(~side & (gain != -2))
             if len(rest) == 0:
                 break
-            nxt = int(rest[np.argmin(nodes[rest])])
+            nxt = int(rest[0])  # disconnected: smallest id outside
         else:
             best = cand[gain[cand] == gain[cand].max()]
-            nxt = int(best[np.argmin(nodes[best])])
-        if w1 + int(vwgt[nodes[nxt]]) > cap1:
-            gain[nxt] = -2  # cannot take it; block and continue
+            nxt = int(best[0])
+        if w1 + int(vw[nxt]) > cap1:
+            gain[nxt] = -2
             continue
         add(nxt)
-
-    part1 = nodes[in_region]
-    part2 = nodes[~in_region]
-    return part1, part2
+    return side
 
 
-def initial_partition(g, k, max_block_weight, seed=1):
-    """Recursive bisection into k blocks on the (small) coarsest graph."""
+def _balance(xadj, adjncy, adjwgt, vwgt, part, k, cap):
+    """Gain-aware overload balancer: while a block exceeds cap, move the
+    best vertex out of the most overloaded block -- preferring the
+    max-gain feasible move, else any move that strictly lowers the
+    overloaded block below the target's new weight (monotone, terminates)."""
+    n = len(part)
+    vw = vwgt if vwgt is not None else np.ones(n, dtype=np.int64)
+    bw = np.zeros(k, dtype=np.int64)
+    np.add.at(bw, part, vw)
+    guard = 8 * (k + 16)
+    while bw.max() > cap and guard > 0:
+        guard -= 1
+        b = int(np.argmax(bw))
+        vs = np.flatnonzero(part == b)
+        best_feas = None   # (gain, -weight, v, t)
+        best_force = None  # (new_target_weight, v, t)
+        for v in vs:
+            e0, e1 = int(xadj[v]), int(xadj[v + 1])
+            nb = part[adjncy[e0:e1]]
+            w = adjwgt[e0:e1] if adjwgt is not None else np.ones(e1 - e0, np.int64)
+            conn = np.zeros(k, dtype=np.int64)
+            np.add.at(conn, nb, w)
+            internal = int(conn[b])
+            for t in range(k):
+                if t == b:
+                    continue
+                nw = int(bw[t]) + int(vw[v])
+                g = int(conn[t]) - internal
+                if nw <= cap:
+                    key = (g, -int(vw[v]), -v, t)
+                    if best_feas is None or key > best_feas[:4]:
+                        best_feas = (g, -int(vw[v]), -v, t, v)
+                elif nw < int(bw[b]):
+                    key = (-nw, g, -v)
+                    if best_force is None or key > best_force[:3]:
+                        best_force = (-nw, g, -v, t, v)
+        if best_feas is not None:
+            t, v = best_feas[3], best_feas[4]
+        elif best_force is not None:
+            t, v = best_force[3], best_force[4]
+        else:
+            break
+        part[v] = t
+        bw[b] -= int(vw[v])
+        bw[t] += int(vw[v])
+    return part
+
+
+def initial_partition(g, k, max_block_weight, seed=1, reps=4):
+    """Recursive bisection into k blocks on the (small) coarsest graph:
+    per bisection, `reps` greedy-graph-growing attempts from different
+    high-degree seeds, each polished by two-way FM, best cut kept
+    (restates the reference's bipartitioner-pool repetition idea,
+    initial_partitioning with min_num_non_adaptive_repetitions=5); then a
+    k-way overload balancer."""
     xadj = np.asarray(g.xadj).astype(np.int64)
-    adjncy = np.asarray(g.adjncy)
-    vwgt = np.ones(g.n, dtype=np.int64)
-    # host graph may carry weights
+    adjncy = np.asarray(g.adjncy).astype(np.int64)
     from . import _lib
 
-    vw = _lib.kmp_graph_vwgt(g._h)
-    if vw:
-        vwgt = np.ctypeslib.as_array(vw, shape=(g.n,)).astype(np.int64)
-    aw = _lib.kmp_graph_adjwgt(g._h)
-    adjwgt = np.ctypeslib.as_array(aw, shape=(g.m,)) if aw else None
+    vwgt = None
+    vw_p = _lib.kmp_graph_vwgt(g._h)
+    if vw_p:
+        vwgt = np.ctypeslib.as_array(vw_p, shape=(g.n,)).astype(np.int64)
+    adjwgt = None
+    aw_p = _lib.kmp_graph_adjwgt(g._h)
+    if aw_p:
+        adjwgt = np.ctypeslib.as_array(aw_p, shape=(g.m,)).astype(np.int64)
+    vwgt_all = vwgt if vwgt is not None else np.ones(g.n, dtype=np.int64)
 
     part = np.zeros(g.n, dtype=np.uint32)
+    loc = np.full(g.n, -1, dtype=np.int64)
 
     def rec(nodes, k_lo, k_hi):
         if len(nodes) == 0:
@@ -109,21 +241,54 @@ def initial_partition(g, k, max_block_weight, seed=1):
             return
         k1 = (k_hi - k_lo + 1) // 2
         k2 = (k_hi - k_lo) - k1
-        total = int(vwgt[nodes].sum())
+        total = int(vwgt_all[nodes].sum())
         target1 = total * k1 // (k1 + k2)
-        p1, p2 = _greedy_bisect(
-            xadj, adjncy, adjwgt, vwgt, g.n, nodes,
-            target1, k1 * max_block_weight,
-        )
-        rec(p1, k_lo, k_lo + k1)
-        rec(p2, k_lo + k1, k_hi)
+        cap1 = k1 * max_block_weight
+        cap2 = k2 * max_block_weight
+
+        loc[nodes] = np.arange(len(nodes))
+        sub_xadj, sub_adj, sub_w = _subgraph_csr(xadj, adjncy, adjwgt, nodes, loc)
+        loc[nodes] = -1
+        vw = vwgt_all[nodes]
+
+        def bisection_cut(s):
+            u = np.repeat(np.arange(len(nodes)), np.diff(sub_xadj))
+            return int(sub_w[s[u] != s[sub_adj]].sum())
+
+        best_side = None
+        best_cut = None
+        for rep in range(reps):
+            side = _greedy_grow(sub_xadj, sub_adj, sub_w, vw, target1, cap1,
+                                seed_rank=rep)
+            side = _fm_refine_bisection(sub_xadj, sub_adj, sub_w, vw, side,
+                                        cap1, cap2)
+            c = bisection_cut(side)
+            if best_cut is None or c < best_cut:
+                best_cut = c
+                best_side = side
+        rec(nodes[best_side], k_lo, k_lo + k1)
+        rec(nodes[~best_side], k_lo + k1, k_hi)
 
     rec(np.arange(g.n), 0, k)
+    part = _balance(xadj, adjncy, adjwgt, vwgt, part, k, max_block_weight)
     return part
 
 
-def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000):
-    """Full multilevel partition. Returns (cut, partition, levels_info)."""
+def level_cluster_weight(total_w, n, k, eps, contraction_limit=2000):
+    """Per-level max cluster weight: the reference's EPSILON_BLOCK_WEIGHT
+    formula (max_cluster_weights.h:18-46) capped by the BLOCK_WEIGHT rule
+    with the reference's IP multiplier 1/12 (presets.cc)."""
+    shrink = min(max(n // contraction_limit, 2), k)
+    eps_rule = int(eps * total_w / shrink)
+    block_rule = total_w // (12 * k)
+    return max(1, min(eps_rule, block_rule) if block_rule > 0 else eps_rule)
+
+
+def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
+              stop_n=512):
+    """Full multilevel partition on the GPU engine.
+
+    Returns (cut, partition, level_sizes)."""
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
     mbw = np.full(k, mbw_val, dtype=np.int64)
@@ -132,10 +297,9 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000):
     graphs = [g]
     mappings = []
     engines = [LpEngine(g)]
-    while graphs[-1].n > max(2 * contraction_limit, 2 * k):
+    while graphs[-1].n > max(stop_n, 2 * k):
         cur = graphs[-1]
-        shrink = min(max(cur.n // contraction_limit, 2), k)
-        mcw = max(1, int(eps * total_w / shrink))
+        mcw = level_cluster_weight(total_w, cur.n, k, eps, contraction_limit)
         nc, clus, _ = engines[-1].cluster(mcw, seed=seed + len(mappings), iters=iters)
         coarse, mapping = engines[-1].contract(clus)
         if coarse.n > 0.95 * cur.n:
@@ -153,5 +317,4 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000):
         cut, part, _ = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
         if level > 0:
             part = part[mappings[level - 1]]
-    levels = [gr.n for gr in graphs]
-    return cut, part, levels
+    return cut, part, [gr.n for gr in graphs]

@@ -366,3 +366,63 @@ def test_multilevel_coarsening_chain(oracle):
         cur = coarse
         eng = ka.LpEngine(cur)
     assert len(levels) >= 2 and levels[-1][1] < g.n // 4, levels
+
+
+def _pipeline_case(name):
+    import json
+    import os
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    exp = json.load(open(os.path.join(here, "golden", "pipeline_expected.json")))[name]
+    band = json.load(open(os.path.join(here, "golden", "ref_golden_partition.json")))[name]
+    if name.startswith("walshaw"):
+        d = json.load(open(os.path.join(here, "golden", "walshaw_data.json")))
+        g = ka.Graph.from_csr(np.array(d["xadj"], np.uint32), np.array(d["adjncy"], np.uint32))
+    elif name.startswith("rgg2d"):
+        g = ka.Graph.read_metis(os.path.join(here, "golden", "rgg2d.metis"))
+    else:  # rmat{scale}_s{seed}_k{k}
+        scale = int(name.split("_")[0][4:])
+        g = ka.Graph.rmat(scale, 8, 42)
+    return g, exp, band
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", [
+    "walshaw_k2", "walshaw_k16", "rgg2d_k4",
+    "rmat14_s42_k16", "rmat16_s42_k16", "rmat18_s42_k16", "rmat18_s42_k64",
+])
+def test_partition_pipeline(name):
+    """Full multilevel pipeline on the GPU engine: bit-identical to the
+    oracle-mirrored pipeline (tests/golden/pipeline_expected.json -- every
+    stage is bit-reproducible), balanced within the reference's cap, and
+    within a documented quality band of the compiled reference's own full
+    deep-multilevel partitioner (golden cuts at 3 seeds). The band (<=1.75x
+    the reference's best seed) reflects that our pipeline is basic
+    multilevel with LP-only refinement while the reference runs deep
+    multilevel with FM-refined bisection extensions; see DESIGN.md."""
+    _require_gpu()
+    from kaminpar_amd.partition import partition
+
+    g, exp, band = _pipeline_case(name)
+    k = exp["k"]
+    cut, part, levels = partition(g, k, seed=1)
+
+    # bit-exact vs the oracle pipeline
+    assert cut == exp["cut"], (cut, exp["cut"])
+    assert levels == exp["levels"], (levels, exp["levels"])
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]
+
+    # balanced
+    from kaminpar_amd import _lib
+    vwp = _lib.kmp_graph_vwgt(g._h)
+    vw = np.ctypeslib.as_array(vwp, shape=(g.n,)).astype(np.int64) if vwp \
+        else np.ones(g.n, np.int64)
+    bw = np.zeros(k, np.int64)
+    np.add.at(bw, part, vw)
+    assert bw.max() <= band["cap"], (bw.max(), band["cap"])
+
+    # quality band vs the compiled reference full pipeline
+    ref_best = min(band[f"seed{s}"]["cut"] for s in (1, 2, 3))
+    assert cut <= 1.75 * ref_best, (cut, ref_best)

@@ -1,0 +1,73 @@
+"""CPU-side guards for the multilevel pipeline:
+
+- the oracle-mirrored pipeline (tests/oracle_pipeline.py, bit-identical to
+  the GPU pipeline stage by stage) reproduces the committed expected cuts --
+  catches drift between kaminpar_amd/partition.py's schedule and the
+  goldens before any GPU time is spent;
+- the initial partitioner alone is balanced and deterministic.
+"""
+
+import ctypes
+import json
+import os
+
+import numpy as np
+import pytest
+
+import kaminpar_amd as ka
+from kaminpar_amd.partition import initial_partition
+from oracle_pipeline import oracle_partition
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+def _load(name):
+    return json.load(open(os.path.join(HERE, "golden", name)))
+
+
+def _graph(name):
+    if name.startswith("walshaw"):
+        d = _load("walshaw_data.json")
+        return ka.Graph.from_csr(np.array(d["xadj"], np.uint32),
+                                 np.array(d["adjncy"], np.uint32))
+    if name.startswith("rgg2d"):
+        return ka.Graph.read_metis(os.path.join(HERE, "golden", "rgg2d.metis"))
+    scale = int(name.split("_")[0][4:])
+    return ka.Graph.rmat(scale, 8, 42)
+
+
+@pytest.mark.parametrize("name", [
+    "walshaw_k2", "walshaw_k16", "rgg2d_k4", "rmat14_s42_k16",
+    "rmat16_s42_k16",
+])
+def test_oracle_pipeline_matches_expected(oracle, name):
+    exp = _load("pipeline_expected.json")[name]
+    g = _graph(name)
+    cut, part, levels = oracle_partition(oracle, g, exp["k"], seed=1)
+    assert cut == exp["cut"], (cut, exp["cut"])
+    assert levels == exp["levels"]
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]
+
+
+@pytest.mark.parametrize("k", [2, 4, 16, 5])
+def test_initial_partition_balanced_deterministic(k):
+    g = ka.Graph.read_metis(os.path.join(HERE, "golden", "rgg2d.metis"))
+    cap = g.max_block_weight(k, 0.03)
+    p1 = initial_partition(g, k, cap, seed=1)
+    p2 = initial_partition(g, k, cap, seed=1)
+    assert np.array_equal(p1, p2)
+    assert p1.max() < k
+    counts = np.bincount(p1, minlength=k)
+    assert counts.max() <= cap
+
+
+def test_pipeline_band_vs_reference_goldens(oracle):
+    """Our full-pipeline cut stays within the documented band of the
+    compiled reference's deep-multilevel cut (best of 3 seeds)."""
+    exp = _load("pipeline_expected.json")
+    band = _load("ref_golden_partition.json")
+    for name in ("walshaw_k16", "rmat14_s42_k16"):
+        ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
+        assert exp[name]["cut"] <= 1.75 * ref_best, name
