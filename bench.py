@@ -48,9 +48,13 @@ def main():
     ap.add_argument("--k", type=int, default=16)
     ap.add_argument("--iters", type=int, default=5)
     ap.add_argument("--seed", type=int, default=1)
-    ap.add_argument("--workload", choices=["refine", "cluster"], default="refine",
+    ap.add_argument("--workload", choices=["refine", "cluster", "partition"],
+                    default="refine",
                     help="cluster mirrors the reference's own LP benchmark "
-                         "(shm_label_propagation_benchmark.cc: LP clustering)")
+                         "(shm_label_propagation_benchmark.cc: LP clustering); "
+                         "partition runs the full multilevel pipeline "
+                         "(BASELINE config 3: coarsen+IP+uncoarsen, "
+                         "kaminpar_amd/partition.py) -- single GPU only")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -59,6 +63,8 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     n_gpus = max(args.gpus, world)
+    if args.workload == "partition" and world > 1:
+        sys.exit("--workload partition is single-GPU only")
 
     if world > 1:
         import torch
@@ -111,13 +117,26 @@ def main():
 
     from kaminpar_amd.multi import TorchComm, refine_dist
 
+    level_sizes = []
     if world > 1:
         comm = TorchComm(device)
     else:
         comm = None  # fast path: device-resident stepping inside C++
-        eng.refine_begin(k, mbw, part0, seed=args.seed)
+        if args.workload == "refine":
+            eng.refine_begin(k, mbw, part0, seed=args.seed)
 
     def one_step():
+        if args.workload == "partition":
+            from types import SimpleNamespace
+
+            from kaminpar_amd.partition import partition as ml_partition
+
+            cut, _part, levels, arcs, ns = ml_partition(
+                g, k, seed=args.seed, iters=args.iters, engine=eng,
+                return_arcs=True)
+            level_sizes[:] = levels
+            return cut, SimpleNamespace(arcs_scanned=arcs, phase_a_ns=ns,
+                                        moves=0)
         if args.workload == "cluster":
             nc, _clus, stats = eng.cluster(mcw, seed=args.seed, iters=args.iters)
             return nc, stats
@@ -221,7 +240,9 @@ def main():
             "dtype": "int32",
             "data": "synthetic",
             "config": {
-                "workload": f"rmat{args.scale}_k{args.k}_lp_{args.workload}",
+                "workload": (f"rmat{args.scale}_k{args.k}_multilevel"
+                             if args.workload == "partition" else
+                             f"rmat{args.scale}_k{args.k}_lp_{args.workload}"),
                 "n": int(g.n),
                 "arcs": int(g.m),
                 "k": k,
@@ -230,6 +251,7 @@ def main():
                 "edge_cut_after": int(last_cut),
                 "moves": int(moves // max(args.steps, 1)),
                 "parallelism": f"shard{world}" if world > 1 else "single",
+                **({"levels": level_sizes} if args.workload == "partition" else {}),
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
@@ -241,7 +263,14 @@ def main():
 def run_cpu_baseline(g, k, mbw, part0, args):
     """Time the CPU oracle (the restated reference LP semantics, oracle/) on a
     bounded sample of the same workload: ONE sweep (iters=1) over the same
-    graph and initial partition (~10-30 s of CPU work at scale 26)."""
+    graph and initial partition (~10-30 s of CPU work at scale 26).
+
+    For the partition workload: time the compiled reference's own serial
+    full partitioner (oracle/_ref/libkaminpar_ref_full.so, kind
+    "reference") on an R-MAT scale-18 sample -- the full serial pipeline on
+    the scale-26 graph would take tens of minutes."""
+    if args.workload == "partition":
+        return _cpu_baseline_partition(k, args)
     res = _cpu_baseline_impl(g, k, mbw, part0, iters=1, seed=args.seed)
     if res is None:
         return None
@@ -254,6 +283,41 @@ def run_cpu_baseline(g, k, mbw, part0, args):
         "cores": cores,
         "kind": "port",
         "sample": f"1 LP sweep over the full workload graph ({arcs} arcs, {dt:.1f}s)",
+    }
+
+
+def _cpu_baseline_partition(k, args):
+    import ctypes
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    path = os.path.join(here, "oracle", "_ref", "libkaminpar_ref_full.so")
+    if not os.path.exists(path):
+        return None
+    import kaminpar_amd as ka
+
+    sample_scale = 18
+    gs = ka.Graph.rmat(sample_scale, args.edgefactor, seed=42)
+    lib = ctypes.CDLL(path)
+    u32p = ctypes.POINTER(ctypes.c_uint32)
+    lib.kref_compute_partition.restype = ctypes.c_int64
+    xadj = np.ascontiguousarray(gs.xadj, dtype=np.uint32)
+    adjncy = np.ascontiguousarray(gs.adjncy, dtype=np.uint32)
+    part = np.zeros(gs.n, dtype=np.uint32)
+    t0 = time.time()
+    cut = lib.kref_compute_partition(
+        ctypes.c_uint32(gs.n), ctypes.c_uint64(gs.m),
+        xadj.ctypes.data_as(u32p), adjncy.ctypes.data_as(u32p),
+        None, None, ctypes.c_uint32(k), ctypes.c_double(0.03),
+        ctypes.c_int(args.seed), part.ctypes.data_as(u32p))
+    dt = time.time() - t0
+    return {
+        "value": round(gs.m / dt, 1),
+        "unit": "fine arcs partitioned/s",
+        "cores": 1,
+        "kind": "reference",
+        "sample": (f"serial reference KaMinPar::compute_partition on R-MAT "
+                   f"scale-{sample_scale} k={k} ({gs.m} arcs, {dt:.1f}s, "
+                   f"cut={cut})"),
     }
 
 

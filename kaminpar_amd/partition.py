@@ -285,10 +285,13 @@ def level_cluster_weight(total_w, n, k, eps, contraction_limit=2000):
 
 
 def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
-              stop_n=512):
+              stop_n=512, engine=None, return_arcs=False):
     """Full multilevel partition on the GPU engine.
 
-    Returns (cut, partition, level_sizes)."""
+    `engine` reuses an existing LpEngine for the fine graph (keeps the
+    fine CSR resident in HBM across repeated runs, e.g. in bench.py).
+    Returns (cut, partition, level_sizes); with return_arcs also the LP
+    arcs scanned and phase-A kernel nanoseconds summed over all levels."""
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
     mbw = np.full(k, mbw_val, dtype=np.int64)
@@ -296,7 +299,9 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     # ---- coarsen (GPU) ----
     graphs = [g]
     mappings = []
-    engines = [LpEngine(g)]
+    engines = [engine if engine is not None else LpEngine(g)]
+    s0 = engines[0].get_stats()
+    arcs_prev, ns_prev = s0.arcs_scanned, s0.phase_a_ns
     while graphs[-1].n > max(stop_n, 2 * k):
         cur = graphs[-1]
         mcw = level_cluster_weight(total_w, cur.n, k, eps, contraction_limit)
@@ -317,4 +322,13 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
         cut, part, _ = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
         if level > 0:
             part = part[mappings[level - 1]]
-    return cut, part, [gr.n for gr in graphs]
+    levels = [gr.n for gr in graphs]
+    if return_arcs:
+        arcs = -arcs_prev
+        ns = -ns_prev
+        for e in engines:
+            s = e.get_stats()
+            arcs += s.arcs_scanned
+            ns += s.phase_a_ns
+        return cut, part, levels, int(arcs), int(ns)
+    return cut, part, levels
