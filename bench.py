@@ -267,8 +267,8 @@ def run_cpu_baseline(g, k, mbw, part0, args):
 
     For the partition workload: time the compiled reference's own serial
     full partitioner (oracle/_ref/libkaminpar_ref_full.so, kind
-    "reference") on an R-MAT scale-18 sample -- the full serial pipeline on
-    the scale-26 graph would take tens of minutes."""
+    "reference") on an R-MAT scale-21 sample (~10 s serial; the full
+    serial pipeline on the scale-26 graph extrapolates to ~3-4 min)."""
     if args.workload == "partition":
         return _cpu_baseline_partition(k, args)
     res = _cpu_baseline_impl(g, k, mbw, part0, iters=1, seed=args.seed)
@@ -295,7 +295,7 @@ def _cpu_baseline_partition(k, args):
         return None
     import kaminpar_amd as ka
 
-    sample_scale = 18
+    sample_scale = 21
     gs = ka.Graph.rmat(sample_scale, args.edgefactor, seed=42)
     lib = ctypes.CDLL(path)
     u32p = ctypes.POINTER(ctypes.c_uint32)

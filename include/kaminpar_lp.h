@@ -72,6 +72,13 @@ void kmp_graph_free(kmp_graph_t *g);
 /* Sequential host edge cut of a labelling (metrics.cc:73-84). */
 int64_t kmp_edge_cut_host(const kmp_graph_t *g, const uint32_t *labels);
 
+/* Degree-bucket rearrangement (the reference's default preprocessing,
+ * NodeOrdering::DEGREE_BUCKETS: graphutils/permutator.h:30-128, stable
+ * counting sort by floor_log2(deg)+1 with isolated vertices last).
+ * perm_out[u_old] = u_new, n entries; returns the permuted graph. */
+kmp_graph_t *kmp_rearrange_degree_buckets(const kmp_graph_t *g,
+                                          uint32_t *perm_out);
+
 /* Max block weight as the reference's PartitionContext computes it for
  * uniform epsilon (context.cc:27-39): (1+eps) * ceil(total_weight / k). */
 int64_t kmp_max_block_weight(const kmp_graph_t *g, uint32_t k, double eps);

@@ -83,6 +83,8 @@ def _load():
     lib.kmp_edge_cut_host.argtypes = [vp, p(u32)]
     lib.kmp_max_block_weight.restype = i64
     lib.kmp_max_block_weight.argtypes = [vp, u32, ctypes.c_double]
+    lib.kmp_rearrange_degree_buckets.restype = vp
+    lib.kmp_rearrange_degree_buckets.argtypes = [vp, p(u32)]
 
     lib.kmp_lp_create.restype = vp
     lib.kmp_lp_create.argtypes = [vp]
@@ -198,6 +200,16 @@ class Graph:
 
     def max_block_weight(self, k, eps=0.03):
         return _lib.kmp_max_block_weight(self._h, k, eps)
+
+    def rearrange_degree_buckets(self):
+        """Degree-bucket rearrangement (the reference's default
+        NodeOrdering::DEGREE_BUCKETS preprocessing).
+
+        Returns (permuted_graph, perm) with perm[u_old] = u_new; a labelling
+        l_new on the permuted graph maps back as l_old = l_new[perm]."""
+        perm = np.zeros(self.n, dtype=np.uint32)
+        h = _lib.kmp_rearrange_degree_buckets(self._h, _u32p(perm))
+        return Graph(h), perm
 
     def __del__(self):
         h = getattr(self, "_h", None)

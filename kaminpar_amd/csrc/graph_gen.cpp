@@ -453,4 +453,80 @@ i64 kmp_max_block_weight(const kmp_graph_t *g, u32 k, double eps) {
   return static_cast<i64>((1.0 + eps) * pbw);
 }
 
+// Degree-bucket rearrangement: stable counting sort of the vertices by
+// exponentially spaced degree buckets, isolated vertices moved to the back
+// -- the reference's default NodeOrdering::DEGREE_BUCKETS preprocessing
+// (graphutils/permutator.h:30-128 compute_node_permutation_by_degree_buckets
+// + build_permuted_graph; bucket(deg) = floor_log2(deg)+1, deg==0 -> last
+// bucket, kaminpar-common/degree_buckets.h:20-26). Adjacency-row order is
+// preserved; targets are remapped. perm_out[u_old] = u_new (n entries).
+// Beyond parity with the reference's preprocessing, this is a locality
+// lever for the LP gather path: hub labels become contiguous and stay
+// cache-resident.
+kmp_graph_t *kmp_rearrange_degree_buckets(const kmp_graph_t *g, u32 *perm_out) {
+  const u32 n = g->n;
+  constexpr int kBuckets = 33; // 32-bit degrees + deg-0 bucket at the end
+  auto bucket_of = [&](u32 u) -> int {
+    const u32 deg = g->xadj[u + 1] - g->xadj[u];
+    if (deg == 0) {
+      return kBuckets - 1;
+    }
+    return 31 - __builtin_clz(deg) + 1; // floor_log2(deg) + 1
+  };
+
+  std::vector<u64> counts(kBuckets + 1, 0);
+  for (u32 u = 0; u < n; ++u) {
+    ++counts[bucket_of(u) + 1];
+  }
+  for (int b = 1; b <= kBuckets; ++b) {
+    counts[b] += counts[b - 1];
+  }
+  std::vector<u32> inv(n); // inv[u_new] = u_old
+  {
+    std::vector<u64> cursor(counts.begin(), counts.end() - 1);
+    for (u32 u = 0; u < n; ++u) { // stable within bucket
+      const u64 pos = cursor[bucket_of(u)]++;
+      perm_out[u] = static_cast<u32>(pos);
+      inv[pos] = u;
+    }
+  }
+
+  auto *out = new kmp_graph_t();
+  out->n = n;
+  out->m = g->m;
+  out->total_node_weight = g->total_node_weight;
+  out->xadj.resize(n + 1);
+  out->adjncy.resize(g->m);
+  if (!g->vwgt.empty()) {
+    out->vwgt.resize(n);
+  }
+  if (!g->adjwgt.empty()) {
+    out->adjwgt.resize(g->m);
+  }
+  out->xadj[0] = 0;
+  for (u32 v = 0; v < n; ++v) {
+    const u32 u = inv[v];
+    out->xadj[v + 1] = out->xadj[v] + (g->xadj[u + 1] - g->xadj[u]);
+  }
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 4096)
+#endif
+  for (long long v = 0; v < static_cast<long long>(n); ++v) {
+    const u32 u = inv[v];
+    const u64 src = g->xadj[u];
+    const u64 dst = out->xadj[v];
+    const u32 deg = g->xadj[u + 1] - g->xadj[u];
+    for (u32 i = 0; i < deg; ++i) {
+      out->adjncy[dst + i] = perm_out[g->adjncy[src + i]];
+      if (!g->adjwgt.empty()) {
+        out->adjwgt[dst + i] = g->adjwgt[src + i];
+      }
+    }
+    if (!g->vwgt.empty()) {
+      out->vwgt[v] = g->vwgt[u];
+    }
+  }
+  return out;
+}
+
 } // extern "C"
