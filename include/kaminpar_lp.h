@@ -196,6 +196,25 @@ int64_t kmp_contract(
     uint32_t *mapping_out,
     kmp_graph_t **coarse_out
 );
+
+/* Contract and hand the coarse graph DIRECTLY to a new engine without the
+ * host round-trip (device-resident multilevel chain; the coarse CSR stays
+ * in HBM). mapping_out (n entries, host) is still produced. The new engine
+ * owns the coarse graph; use kmp_lp_download_graph when a host copy is
+ * needed (e.g. the coarsest level for initial partitioning). Identical
+ * results to kmp_contract + kmp_lp_create on the downloaded graph. */
+int64_t kmp_contract_engine(
+    kmp_lp_t *e,
+    const uint32_t *clustering,
+    uint32_t *mapping_out,
+    kmp_lp_t **coarse_eng_out
+);
+
+/* Download an engine's device-resident CSR into a new host graph handle. */
+kmp_graph_t *kmp_lp_download_graph(const kmp_lp_t *e);
+
+uint32_t kmp_lp_n(const kmp_lp_t *e);
+uint64_t kmp_lp_m(const kmp_lp_t *e);
 #ifdef __cplusplus
 }
 #endif

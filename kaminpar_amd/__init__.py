@@ -111,6 +111,14 @@ def _load():
     lib.kmp_lp_get_stats.argtypes = [vp, vp]
     lib.kmp_contract.restype = i64
     lib.kmp_contract.argtypes = [vp, p(u32), p(u32), ctypes.POINTER(vp)]
+    lib.kmp_contract_engine.restype = i64
+    lib.kmp_contract_engine.argtypes = [vp, p(u32), p(u32), ctypes.POINTER(vp)]
+    lib.kmp_lp_download_graph.restype = vp
+    lib.kmp_lp_download_graph.argtypes = [vp]
+    lib.kmp_lp_n.restype = u32
+    lib.kmp_lp_n.argtypes = [vp]
+    lib.kmp_lp_m.restype = ctypes.c_uint64
+    lib.kmp_lp_m.argtypes = [vp]
     return lib
 
 
@@ -232,13 +240,18 @@ def random_partition(n, k, seed=42):
 class LpEngine:
     """Device-resident LP engine (requires a GPU; fails loudly otherwise)."""
 
-    def __init__(self, graph):
-        self._graph = graph  # keep alive
-        self._h = _lib.kmp_lp_create(graph._h)
+    def __init__(self, graph=None, _handle=None):
+        self._graph = graph  # keep alive (None for device-resident engines)
+        if _handle is not None:
+            self._h = _handle
+        else:
+            self._h = _lib.kmp_lp_create(graph._h)
         if not self._h:
             raise RuntimeError(
                 "kaminpar_amd: LP engine creation failed (no HIP GPU available?)"
             )
+        self.n = int(_lib.kmp_lp_n(self._h))
+        self.m = int(_lib.kmp_lp_m(self._h))
 
     def refine(self, k, max_block_weights, partition, seed=1, iters=5):
         """Deterministic LP refinement; returns (cut, partition, Stats)."""
@@ -254,8 +267,7 @@ class LpEngine:
 
     def cluster(self, max_cluster_weight, clustering=None, desired=0, seed=1, iters=5):
         """Deterministic LP clustering; returns (n_clusters, clustering, Stats)."""
-        n = self._graph.n
-        clus = np.zeros(n, dtype=np.uint32)
+        clus = np.zeros(self.n, dtype=np.uint32)
         stats = Stats()
         nc = _lib.kmp_lp_cluster(
             self._h, max_cluster_weight, desired, _u32p(clus), seed, iters,
@@ -308,15 +320,35 @@ class LpEngine:
 
         Returns (coarse_graph: Graph, mapping: np.ndarray[u32])."""
         clus = np.ascontiguousarray(clustering, dtype=np.uint32)
-        mapping = np.zeros(self._graph.n, dtype=np.uint32)
+        mapping = np.zeros(self.n, dtype=np.uint32)
         out = ctypes.c_void_p()
         c_n = _lib.kmp_contract(self._h, _u32p(clus), _u32p(mapping), ctypes.byref(out))
         if c_n < 0:
             raise RuntimeError("kmp_contract failed")
         return Graph(out.value), mapping
 
+    def contract_engine(self, clustering):
+        """Contract a clustering and hand the coarse graph directly to a NEW
+        engine without a host round-trip (the coarse CSR stays in HBM).
+
+        Returns (coarse_engine: LpEngine, mapping: np.ndarray[u32]). Results
+        are identical to contract() + LpEngine(coarse_graph)."""
+        clus = np.ascontiguousarray(clustering, dtype=np.uint32)
+        mapping = np.zeros(self.n, dtype=np.uint32)
+        out = ctypes.c_void_p()
+        c_n = _lib.kmp_contract_engine(self._h, _u32p(clus), _u32p(mapping),
+                                       ctypes.byref(out))
+        if c_n < 0:
+            raise RuntimeError("kmp_contract_engine failed")
+        return LpEngine(_handle=out.value), mapping
+
+    def download_graph(self):
+        """Download the engine's device-resident CSR into a host Graph."""
+        h = _lib.kmp_lp_download_graph(self._h)
+        return Graph(h)
+
     def refine_end(self):
-        part = np.zeros(self._graph.n, dtype=np.uint32)
+        part = np.zeros(self.n, dtype=np.uint32)
         stats = Stats()
         cut = _lib.kmp_lp_refine_end(self._h, _u32p(part), ctypes.byref(stats))
         return cut, part, stats

@@ -2004,43 +2004,10 @@ void sync_spin(kmp_lp_t *e) {
   }
 }
 
-} // namespace
-
-extern "C" {
-
-kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
-  int ndev = 0;
-  if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) {
-    fprintf(stderr, "kaminpar_amd: no HIP device available -- the LP engine requires a GPU\n");
-    return nullptr;
-  }
-
-  auto *e = new kmp_lp_t();
-  (void)hipSetDeviceFlags(hipDeviceScheduleSpin); // ignore if context exists
-  e->n = kmp_graph_n(g);
-  e->m = kmp_graph_m(g);
-  e->C = kmp::chunk_size_for(e->n);
-  e->P = kmp::pos_count(e->n);
-  e->has_vwgt = kmp_graph_vwgt(g) != nullptr;
-  e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
-  HIP_CHECK(hipStreamCreate(&e->stream));
-  HIP_CHECK(hipEventCreateWithFlags(&e->sync_ev, hipEventDisableTiming));
-
-  HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
-  HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
-  HIP_CHECK(hipMemcpy(e->d_xadj, kmp_graph_xadj(g), sizeof(u32) * (e->n + 1), hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(e->d_adjncy, kmp_graph_adjncy(g), sizeof(u32) * e->m, hipMemcpyHostToDevice));
-  if (e->has_vwgt) {
-    HIP_CHECK(hipMalloc(&e->d_vwgt, sizeof(i32) * e->n));
-    HIP_CHECK(hipMemcpy(e->d_vwgt, kmp_graph_vwgt(g), sizeof(i32) * e->n, hipMemcpyHostToDevice));
-  }
-  if (e->has_adjwgt) {
-    HIP_CHECK(hipMalloc(&e->d_adjwgt, sizeof(i32) * e->m));
-    HIP_CHECK(
-        hipMemcpy(e->d_adjwgt, kmp_graph_adjwgt(g), sizeof(i32) * e->m, hipMemcpyHostToDevice)
-    );
-  }
-
+// Everything an engine allocates beyond the graph arrays themselves
+// (LP state, phase/commit buffers, rocprim temps, pinned mirrors).
+// Requires e->n, e->m, e->C, e->stream to be set.
+void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels0, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels16, sizeof(uint16_t) * e->n));
@@ -2087,19 +2054,61 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   ));
   HIP_CHECK(hipMalloc(&e->d_m_select_temp, e->m_select_temp_bytes));
 
-  {
-    const u32 *xadj = kmp_graph_xadj(g);
-    const i32 *vwgt = kmp_graph_vwgt(g);
-    for (u32 u = 0; u < e->n; ++u) {
-      if (xadj[u + 1] == xadj[u]) {
-        e->isolated.push_back(u);
-        e->iso_weights.push_back(vwgt ? vwgt[u] : 1);
-      }
-    }
-  }
   HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
   HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
   HIP_CHECK(hipHostMalloc(&e->h_moves, sizeof(unsigned long long) * 4));
+}
+
+// Isolated-vertex scan used by the clusterer's isolated-node handling
+// (lp_clusterer.cc cluster_isolated_nodes semantics).
+void engine_scan_isolated(kmp_lp_t *e, const u32 *xadj, const i32 *vwgt) {
+  for (u32 u = 0; u < e->n; ++u) {
+    if (xadj[u + 1] == xadj[u]) {
+      e->isolated.push_back(u);
+      e->iso_weights.push_back(vwgt ? vwgt[u] : 1);
+    }
+  }
+}
+
+} // namespace
+
+extern "C" {
+
+kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
+  int ndev = 0;
+  if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) {
+    fprintf(stderr, "kaminpar_amd: no HIP device available -- the LP engine requires a GPU\n");
+    return nullptr;
+  }
+
+  auto *e = new kmp_lp_t();
+  (void)hipSetDeviceFlags(hipDeviceScheduleSpin); // ignore if context exists
+  e->n = kmp_graph_n(g);
+  e->m = kmp_graph_m(g);
+  e->C = kmp::chunk_size_for(e->n);
+  e->P = kmp::pos_count(e->n);
+  e->has_vwgt = kmp_graph_vwgt(g) != nullptr;
+  e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
+  HIP_CHECK(hipStreamCreate(&e->stream));
+  HIP_CHECK(hipEventCreateWithFlags(&e->sync_ev, hipEventDisableTiming));
+
+  HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
+  HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
+  HIP_CHECK(hipMemcpy(e->d_xadj, kmp_graph_xadj(g), sizeof(u32) * (e->n + 1), hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(e->d_adjncy, kmp_graph_adjncy(g), sizeof(u32) * e->m, hipMemcpyHostToDevice));
+  if (e->has_vwgt) {
+    HIP_CHECK(hipMalloc(&e->d_vwgt, sizeof(i32) * e->n));
+    HIP_CHECK(hipMemcpy(e->d_vwgt, kmp_graph_vwgt(g), sizeof(i32) * e->n, hipMemcpyHostToDevice));
+  }
+  if (e->has_adjwgt) {
+    HIP_CHECK(hipMalloc(&e->d_adjwgt, sizeof(i32) * e->m));
+    HIP_CHECK(
+        hipMemcpy(e->d_adjwgt, kmp_graph_adjwgt(g), sizeof(i32) * e->m, hipMemcpyHostToDevice)
+    );
+  }
+
+  engine_alloc_common(e);
+  engine_scan_isolated(e, kmp_graph_xadj(g), kmp_graph_vwgt(g));
   return e;
 }
 
@@ -3040,9 +3049,21 @@ __global__ void k_coarse_adj(
 
 extern "C" {
 
-i64 kmp_contract(
-    kmp_lp_t *e, const u32 *clustering, u32 *mapping_out, kmp_graph_t **coarse_out
-) {
+namespace {
+
+// Output of contract_core: right-sized device arrays the caller either
+// downloads (kmp_contract) or adopts into a new engine (kmp_contract_engine).
+struct ContractOut {
+  u32 c_n = 0;
+  u32 c_m = 0;
+  u32 *d_map = nullptr;   // n
+  u32 *d_cxadj = nullptr; // c_n + 1 (prefix-summed)
+  u32 *d_cadj = nullptr;  // max(c_m, 1)
+  i32 *d_cvw = nullptr;   // c_n
+  i32 *d_cwgt = nullptr;  // max(c_m, 1)
+};
+
+int contract_core(kmp_lp_t *e, const u32 *clustering, ContractOut &o) {
   const u32 n = e->n;
   const u64 m = e->m;
   if (m > 0xFFFFFFFFull) {
@@ -3053,15 +3074,14 @@ i64 kmp_contract(
   const u32 ngrid = ceil_div(n, threads);
   hipStream_t s = e->stream;
 
-  u32 *d_clus = nullptr, *d_rank = nullptr, *d_map = nullptr, *d_cxadj = nullptr,
-      *d_cadj = nullptr;
-  i32 *d_cvw = nullptr, *d_vals[2] = {nullptr, nullptr}, *d_usums = nullptr;
+  u32 *d_clus = nullptr, *d_rank = nullptr;
+  i32 *d_cvw_full = nullptr, *d_vals[2] = {nullptr, nullptr}, *d_usums = nullptr;
   u64 *d_keys[2] = {nullptr, nullptr}, *d_ukeys = nullptr;
   u32 *d_uniq = nullptr;
   HIP_CHECK(hipMalloc(&d_clus, sizeof(u32) * n));
   HIP_CHECK(hipMalloc(&d_rank, sizeof(u32) * n));
-  HIP_CHECK(hipMalloc(&d_map, sizeof(u32) * n));
-  HIP_CHECK(hipMalloc(&d_cvw, sizeof(i32) * n));
+  HIP_CHECK(hipMalloc(&o.d_map, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_cvw_full, sizeof(i32) * n));
   HIP_CHECK(hipMalloc(&d_keys[0], sizeof(u64) * m));
   HIP_CHECK(hipMalloc(&d_keys[1], sizeof(u64) * m));
   HIP_CHECK(hipMalloc(&d_vals[0], sizeof(i32) * m));
@@ -3082,14 +3102,16 @@ i64 kmp_contract(
   HIP_CHECK(rocprim::inclusive_scan(tmp, tmp_bytes, d_rank, d_rank, n, rocprim::plus<u32>(), s));
   u32 c_n = 0;
   HIP_CHECK(hipMemcpyAsync(&c_n, d_rank + n - 1, sizeof(u32), hipMemcpyDeviceToHost, s));
-  hipLaunchKernelGGL(k_map_ranks, dim3(ngrid), dim3(threads), 0, s, n, d_clus, d_rank, d_map);
+  hipLaunchKernelGGL(k_map_ranks, dim3(ngrid), dim3(threads), 0, s, n, d_clus, d_rank, o.d_map);
   LAUNCH_CHECK();
-  HIP_CHECK(hipMemsetAsync(d_cvw, 0, sizeof(i32) * n, s));
-  hipLaunchKernelGGL(k_coarse_vwgt, dim3(ngrid), dim3(threads), 0, s, n, d_map, e->d_vwgt, d_cvw);
+  HIP_CHECK(hipMemsetAsync(d_cvw_full, 0, sizeof(i32) * n, s));
+  hipLaunchKernelGGL(
+      k_coarse_vwgt, dim3(ngrid), dim3(threads), 0, s, n, o.d_map, e->d_vwgt, d_cvw_full
+  );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
-      k_arc_keys, dim3(16384), dim3(threads), 0, s, n, e->d_xadj, e->d_adjncy, e->d_adjwgt, d_map,
-      d_keys[0], d_vals[0]
+      k_arc_keys, dim3(16384), dim3(threads), 0, s, n, e->d_xadj, e->d_adjncy, e->d_adjwgt,
+      o.d_map, d_keys[0], d_vals[0]
   );
   LAUNCH_CHECK();
 
@@ -3121,57 +3143,157 @@ i64 kmp_contract(
   }
   const u32 c_m = (uniq > 0 && last_key == ~0ull) ? uniq - 1 : uniq;
 
-  HIP_CHECK(hipMalloc(&d_cxadj, sizeof(u32) * (c_n + 1)));
-  HIP_CHECK(hipMalloc(&d_cadj, sizeof(u32) * (c_m > 0 ? c_m : 1)));
-  HIP_CHECK(hipMemsetAsync(d_cxadj, 0, sizeof(u32) * (c_n + 1), s));
+  HIP_CHECK(hipMalloc(&o.d_cxadj, sizeof(u32) * (c_n + 1)));
+  HIP_CHECK(hipMalloc(&o.d_cadj, sizeof(u32) * (c_m > 0 ? c_m : 1)));
+  HIP_CHECK(hipMemsetAsync(o.d_cxadj, 0, sizeof(u32) * (c_n + 1), s));
   if (c_m > 0) {
     hipLaunchKernelGGL(
-        k_coarse_hist, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, d_cxadj
+        k_coarse_hist, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, o.d_cxadj
     );
     LAUNCH_CHECK();
     hipLaunchKernelGGL(
-        k_coarse_adj, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, d_cadj
+        k_coarse_adj, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, o.d_cadj
     );
     LAUNCH_CHECK();
   }
   void *tmp4 = nullptr;
   size_t tmp4_bytes = 0;
   HIP_CHECK(rocprim::inclusive_scan(
-      nullptr, tmp4_bytes, d_cxadj, d_cxadj, c_n + 1, rocprim::plus<u32>()
+      nullptr, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u32>()
   ));
   HIP_CHECK(hipMalloc(&tmp4, tmp4_bytes));
   HIP_CHECK(rocprim::inclusive_scan(
-      tmp4, tmp4_bytes, d_cxadj, d_cxadj, c_n + 1, rocprim::plus<u32>(), s
+      tmp4, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u32>(), s
   ));
 
-  // download
+  // right-size the coarse node/edge weights
+  HIP_CHECK(hipMalloc(&o.d_cvw, sizeof(i32) * (c_n > 0 ? c_n : 1)));
+  HIP_CHECK(hipMalloc(&o.d_cwgt, sizeof(i32) * (c_m > 0 ? c_m : 1)));
+  HIP_CHECK(
+      hipMemcpyAsync(o.d_cvw, d_cvw_full, sizeof(i32) * c_n, hipMemcpyDeviceToDevice, s)
+  );
+  if (c_m > 0) {
+    HIP_CHECK(hipMemcpyAsync(o.d_cwgt, d_usums, sizeof(i32) * c_m, hipMemcpyDeviceToDevice, s));
+  }
+  HIP_CHECK(hipStreamSynchronize(s));
+
+  for (void *p : {(void *)d_clus, (void *)d_rank, (void *)d_cvw_full, (void *)d_keys[0],
+                  (void *)d_keys[1], (void *)d_vals[0], (void *)d_vals[1], (void *)d_ukeys,
+                  (void *)d_usums, (void *)d_uniq, tmp, tmp2, tmp3, tmp4}) {
+    if (p) {
+      (void)hipFree(p);
+    }
+  }
+  o.c_n = c_n;
+  o.c_m = c_m;
+  return 0;
+}
+
+void contract_out_free(ContractOut &o) {
+  for (void *p : {(void *)o.d_map, (void *)o.d_cxadj, (void *)o.d_cadj, (void *)o.d_cvw,
+                  (void *)o.d_cwgt}) {
+    if (p) {
+      (void)hipFree(p);
+    }
+  }
+}
+
+} // namespace
+
+i64 kmp_contract(
+    kmp_lp_t *e, const u32 *clustering, u32 *mapping_out, kmp_graph_t **coarse_out
+) {
+  ContractOut o;
+  if (contract_core(e, clustering, o) != 0) {
+    return -1;
+  }
+  hipStream_t s = e->stream;
+  const u32 c_n = o.c_n, c_m = o.c_m;
+
   std::vector<u32> h_cxadj(c_n + 1), h_cadj(c_m);
   std::vector<i32> h_cvw(c_n), h_cwgt(c_m);
-  HIP_CHECK(hipMemcpyAsync(mapping_out, d_map, sizeof(u32) * n, hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), d_cxadj, sizeof(u32) * (c_n + 1),
+  HIP_CHECK(hipMemcpyAsync(mapping_out, o.d_map, sizeof(u32) * e->n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), o.d_cxadj, sizeof(u32) * (c_n + 1),
                            hipMemcpyDeviceToHost, s));
   if (c_m > 0) {
-    HIP_CHECK(hipMemcpyAsync(h_cadj.data(), d_cadj, sizeof(u32) * c_m, hipMemcpyDeviceToHost, s));
     HIP_CHECK(
-        hipMemcpyAsync(h_cwgt.data(), d_usums, sizeof(i32) * c_m, hipMemcpyDeviceToHost, s)
+        hipMemcpyAsync(h_cadj.data(), o.d_cadj, sizeof(u32) * c_m, hipMemcpyDeviceToHost, s)
+    );
+    HIP_CHECK(
+        hipMemcpyAsync(h_cwgt.data(), o.d_cwgt, sizeof(i32) * c_m, hipMemcpyDeviceToHost, s)
     );
   }
-  HIP_CHECK(hipMemcpyAsync(h_cvw.data(), d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(h_cvw.data(), o.d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipStreamSynchronize(s));
 
   *coarse_out = kmp_graph_from_csr(
       c_n, c_m, h_cxadj.data(), h_cadj.data(), h_cvw.data(), c_m ? h_cwgt.data() : nullptr
   );
-
-  for (void *p : {(void *)d_clus, (void *)d_rank, (void *)d_map, (void *)d_cvw,
-                  (void *)d_keys[0], (void *)d_keys[1], (void *)d_vals[0], (void *)d_vals[1],
-                  (void *)d_ukeys, (void *)d_usums, (void *)d_uniq, (void *)d_cxadj,
-                  (void *)d_cadj, tmp, tmp2, tmp3, tmp4}) {
-    if (p) {
-      (void)hipFree(p);
-    }
-  }
+  contract_out_free(o);
   return static_cast<i64>(c_n);
 }
+
+i64 kmp_contract_engine(
+    kmp_lp_t *e, const u32 *clustering, u32 *mapping_out, kmp_lp_t **coarse_eng_out
+) {
+  ContractOut o;
+  if (contract_core(e, clustering, o) != 0) {
+    return -1;
+  }
+  hipStream_t s = e->stream;
+  const u32 c_n = o.c_n, c_m = o.c_m;
+
+  // host copies needed regardless: the mapping (projection happens on the
+  // host) and xadj/vwgt for the isolated-vertex scan of the new engine
+  std::vector<u32> h_cxadj(c_n + 1);
+  std::vector<i32> h_cvw(c_n);
+  HIP_CHECK(hipMemcpyAsync(mapping_out, o.d_map, sizeof(u32) * e->n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), o.d_cxadj, sizeof(u32) * (c_n + 1),
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(h_cvw.data(), o.d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+  HIP_CHECK(hipFree(o.d_map));
+  o.d_map = nullptr;
+
+  auto *e2 = new kmp_lp_t();
+  e2->n = c_n;
+  e2->m = c_m;
+  e2->C = kmp::chunk_size_for(c_n);
+  e2->P = kmp::pos_count(c_n);
+  e2->has_vwgt = true;
+  e2->has_adjwgt = true;
+  HIP_CHECK(hipStreamCreate(&e2->stream));
+  HIP_CHECK(hipEventCreateWithFlags(&e2->sync_ev, hipEventDisableTiming));
+  e2->d_xadj = o.d_cxadj;
+  e2->d_adjncy = o.d_cadj;
+  e2->d_vwgt = o.d_cvw;
+  e2->d_adjwgt = o.d_cwgt;
+  engine_alloc_common(e2);
+  engine_scan_isolated(e2, h_cxadj.data(), h_cvw.data());
+  *coarse_eng_out = e2;
+  return static_cast<i64>(c_n);
+}
+
+kmp_graph_t *kmp_lp_download_graph(const kmp_lp_t *e) {
+  std::vector<u32> h_xadj(e->n + 1), h_adj(e->m);
+  std::vector<i32> h_vw, h_wg;
+  HIP_CHECK(hipMemcpy(h_xadj.data(), e->d_xadj, sizeof(u32) * (e->n + 1), hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(h_adj.data(), e->d_adjncy, sizeof(u32) * e->m, hipMemcpyDeviceToHost));
+  if (e->has_vwgt) {
+    h_vw.resize(e->n);
+    HIP_CHECK(hipMemcpy(h_vw.data(), e->d_vwgt, sizeof(i32) * e->n, hipMemcpyDeviceToHost));
+  }
+  if (e->has_adjwgt) {
+    h_wg.resize(e->m);
+    HIP_CHECK(hipMemcpy(h_wg.data(), e->d_adjwgt, sizeof(i32) * e->m, hipMemcpyDeviceToHost));
+  }
+  return kmp_graph_from_csr(
+      e->n, e->m, h_xadj.data(), h_adj.data(), e->has_vwgt ? h_vw.data() : nullptr,
+      e->has_adjwgt ? h_wg.data() : nullptr
+  );
+}
+
+u32 kmp_lp_n(const kmp_lp_t *e) { return e->n; }
+u64 kmp_lp_m(const kmp_lp_t *e) { return e->m; }
 
 } // extern "C"

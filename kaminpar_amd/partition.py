@@ -296,33 +296,36 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     mbw_val = g.max_block_weight(k, eps)
     mbw = np.full(k, mbw_val, dtype=np.int64)
 
-    # ---- coarsen (GPU) ----
-    graphs = [g]
+    # ---- coarsen (GPU, device-resident chain: the coarse CSR is handed
+    # engine-to-engine in HBM; only the mapping comes back to the host) ----
+    sizes = [g.n]
     mappings = []
     engines = [engine if engine is not None else LpEngine(g)]
     s0 = engines[0].get_stats()
     arcs_prev, ns_prev = s0.arcs_scanned, s0.phase_a_ns
-    while graphs[-1].n > max(stop_n, 2 * k):
-        cur = graphs[-1]
-        mcw = level_cluster_weight(total_w, cur.n, k, eps, contraction_limit)
+    while sizes[-1] > max(stop_n, 2 * k):
+        cur_n = sizes[-1]
+        mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit)
         nc, clus, _ = engines[-1].cluster(mcw, seed=seed + len(mappings), iters=iters)
-        coarse, mapping = engines[-1].contract(clus)
-        if coarse.n > 0.95 * cur.n:
+        coarse_eng, mapping = engines[-1].contract_engine(clus)
+        if coarse_eng.n > 0.95 * cur_n:
+            del coarse_eng
             break
-        graphs.append(coarse)
+        engines.append(coarse_eng)
         mappings.append(mapping)
-        engines.append(LpEngine(coarse))
+        sizes.append(coarse_eng.n)
 
-    # ---- initial partition (CPU, coarsest) ----
-    part = initial_partition(graphs[-1], k, mbw_val, seed=seed)
+    # ---- initial partition (CPU, coarsest downloaded from HBM) ----
+    coarsest = engines[-1].download_graph() if len(engines) > 1 else g
+    part = initial_partition(coarsest, k, mbw_val, seed=seed)
 
     # ---- uncoarsen: refine at every level (GPU) ----
     cut = None
-    for level in range(len(graphs) - 1, -1, -1):
+    for level in range(len(engines) - 1, -1, -1):
         cut, part, _ = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
         if level > 0:
             part = part[mappings[level - 1]]
-    levels = [gr.n for gr in graphs]
+    levels = sizes
     if return_arcs:
         arcs = -arcs_prev
         ns = -ns_prev

@@ -426,3 +426,49 @@ def test_partition_pipeline(name):
     # quality band vs the compiled reference full pipeline
     ref_best = min(band[f"seed{s}"]["cut"] for s in (1, 2, 3))
     assert cut <= 1.75 * ref_best, (cut, ref_best)
+
+
+@pytest.mark.gpu
+def test_contract_engine_matches_contract(oracle):
+    """Device-resident contraction handoff (kmp_contract_engine) produces a
+    coarse engine whose graph and refinement results are bit-identical to
+    the host round-trip path (kmp_contract + kmp_lp_create)."""
+    _require_gpu()
+    g = ka.Graph.rmat(13, 8, seed=7)
+    eng = ka.LpEngine(g)
+    mcw = 64
+    nc, clus, _ = eng.cluster(mcw, seed=1, iters=5)
+
+    coarse_host, map_a = eng.contract(clus)
+    coarse_eng, map_b = eng.contract_engine(clus)
+    assert np.array_equal(map_a, map_b)
+    assert coarse_eng.n == coarse_host.n and coarse_eng.m == coarse_host.m
+
+    dl = coarse_eng.download_graph()
+    assert np.array_equal(np.asarray(dl.xadj), np.asarray(coarse_host.xadj))
+    assert np.array_equal(np.asarray(dl.adjncy), np.asarray(coarse_host.adjncy))
+    from kaminpar_amd import _lib
+    for getter, count in ((_lib.kmp_graph_vwgt, coarse_host.n),
+                          (_lib.kmp_graph_adjwgt, coarse_host.m)):
+        pa = getter(dl._h)
+        pb = getter(coarse_host._h)
+        assert bool(pa) == bool(pb)
+        if pa:
+            assert np.array_equal(np.ctypeslib.as_array(pa, shape=(count,)),
+                                  np.ctypeslib.as_array(pb, shape=(count,)))
+
+    # refinement on both engines is bit-identical
+    k = 16
+    mbw = np.full(k, coarse_host.max_block_weight(k, 0.03), dtype=np.int64)
+    part0 = ka.random_partition(coarse_host.n, k, seed=5)
+    ref_eng = ka.LpEngine(coarse_host)
+    cut_a, part_a, _ = ref_eng.refine(k, mbw, part0, seed=1, iters=5)
+    cut_b, part_b, _ = coarse_eng.refine(k, mbw, part0, seed=1, iters=5)
+    assert cut_a == cut_b
+    assert np.array_equal(part_a, part_b)
+
+    # clustering on the coarse engine too (isolated handling needs vwgt)
+    nc_a, clus_a, _ = ref_eng.cluster(500, seed=2, iters=5)
+    nc_b, clus_b, _ = coarse_eng.cluster(500, seed=2, iters=5)
+    assert nc_a == nc_b
+    assert np.array_equal(clus_a, clus_b)
