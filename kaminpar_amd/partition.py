@@ -209,7 +209,7 @@ def _balance(xadj, adjncy, adjwgt, vwgt, part, k, cap):
     return part
 
 
-def initial_partition(g, k, max_block_weight, seed=1, reps=4):
+def initial_partition(g, k, max_block_weight, seed=1, reps=8):
     """Recursive bisection into k blocks on the (small) coarsest graph:
     per bisection, `reps` greedy-graph-growing attempts from different
     high-degree seeds, each polished by two-way FM, best cut kept
