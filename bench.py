@@ -94,9 +94,11 @@ def main():
         # exhaust host RAM): rank 0 generates + saves, the rest load
         import torch.distributed as dist
 
-        cache = f"/tmp/kmp_rmat{args.scale}_{args.edgefactor}"
+        cache = f"/tmp/kmp_rmat{args.scale}_{args.edgefactor}_{args.order}"
         if rank == 0:
             g0 = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+            if args.order == "deg-buckets":
+                g0, _perm = g0.rearrange_degree_buckets()
             np.save(cache + "_xadj.npy", np.asarray(g0.xadj))
             np.save(cache + "_adjncy.npy", np.asarray(g0.adjncy))
             del g0
@@ -106,10 +108,10 @@ def main():
         g = ka.Graph.from_csr(np.asarray(xadj), np.asarray(adjncy))
     else:
         g = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
-    if args.order == "deg-buckets":
-        t1 = time.time()
-        g, _perm = g.rearrange_degree_buckets()
-        log(f"[bench] deg-bucket rearrangement ({time.time()-t1:.1f}s)")
+        if args.order == "deg-buckets":
+            t1 = time.time()
+            g, _perm = g.rearrange_degree_buckets()
+            log(f"[bench] deg-bucket rearrangement ({time.time()-t1:.1f}s)")
     log(f"[bench] n={g.n} m={g.m} ({time.time()-t0:.1f}s); uploading ...")
 
     k = args.k
