@@ -933,18 +933,19 @@ __global__ void k_phase_m_c(
 }
 
 // L prep for clustering: per-vertex slice prefix AND pooled-hash region
-// prefix (region size = next power of two >= 2*deg). Sets *pool_overflow if
-// the pool cannot hold all regions (caller aborts with a clear message).
+// prefix (region size = next power of two >= 2*deg). Writes the total
+// region demand to *hacc_out: when it exceeds the pool, the host processes
+// the L list in pool-sized batches (k_phase_l_sel_c clears each region
+// after reading it, so the pool can be reused within a chunk).
 __global__ void k_l_prep_c(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     const u32 *__restrict__ xadj,
     u32 l_cap,
-    u64 pool_slots,
     u32 *__restrict__ l_off,  // slice prefix
     u64 *__restrict__ l_hoff, // hash-region prefix
     u32 *__restrict__ l_hbits, // log2(region size)
-    int *__restrict__ pool_overflow
+    unsigned long long *__restrict__ hacc_out
 ) {
   const u32 count = *l_count < l_cap ? *l_count : l_cap;
   for (u32 i = threadIdx.x; i < count; i += blockDim.x) {
@@ -970,13 +971,14 @@ __global__ void k_l_prep_c(
     }
     l_off[count] = sacc;
     l_hoff[count] = hacc;
-    if (hacc > pool_slots) {
-      *pool_overflow = 1;
-    }
+    *hacc_out = hacc;
   }
 }
 
 // Slice-parallel accumulation into the pooled per-vertex global hash.
+// Processes L vertices [vid_lo, vid_hi) with pool offsets rebased to
+// l_hoff[vid_lo] (batched when a chunk's total region demand exceeds the
+// pool; k_phase_l_sel_c clears each region so batches can reuse the pool).
 template <bool kUnitWeights>
 __global__ void k_phase_l_acc_c(
     const u32 *__restrict__ xadj,
@@ -984,21 +986,23 @@ __global__ void k_phase_l_acc_c(
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
     const u64 *__restrict__ l_list,
-    const u32 *__restrict__ l_count,
-    u32 l_cap,
+    u32 vid_lo,
+    u32 vid_hi,
     const u32 *__restrict__ l_off,
     const u64 *__restrict__ l_hoff,
     const u32 *__restrict__ l_hbits,
     u32 *__restrict__ pool_keys,
     i32 *__restrict__ pool_vals
 ) {
-  const u32 count = *l_count < l_cap ? *l_count : l_cap;
-  if (count == 0) {
+  const u32 count = vid_hi;
+  if (count <= vid_lo) {
     return;
   }
+  const u64 hbase = l_hoff[vid_lo];
+  const u32 s_lo = l_off[vid_lo];
   const u32 total = l_off[count];
-  for (u32 s = blockIdx.x; s < total; s += gridDim.x) {
-    u32 lo = 0, hi = count - 1;
+  for (u32 s = s_lo + blockIdx.x; s < total; s += gridDim.x) {
+    u32 lo = vid_lo, hi = count - 1;
     while (lo < hi) {
       const u32 mid = (lo + hi + 1) >> 1;
       if (l_off[mid] <= s) {
@@ -1013,8 +1017,8 @@ __global__ void k_phase_l_acc_c(
     const u32 deg = xadj[u + 1] - row;
     const u32 e_lo = (s - l_off[vid]) * kLSlice;
     const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
-    u32 *hk = pool_keys + l_hoff[vid];
-    i32 *hv = pool_vals + l_hoff[vid];
+    u32 *hk = pool_keys + (l_hoff[vid] - hbase);
+    i32 *hv = pool_vals + (l_hoff[vid] - hbase);
     const u32 mask = (1u << l_hbits[vid]) - 1;
 
     for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
@@ -1053,8 +1057,8 @@ __global__ void k_phase_l_sel_c(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const u64 *__restrict__ l_list,
-    const u32 *__restrict__ l_count,
-    u32 l_cap,
+    u32 vid_lo,
+    u32 vid_hi,
     const u64 *__restrict__ l_hoff,
     const u32 *__restrict__ l_hbits,
     u32 *__restrict__ pool_keys,
@@ -1063,16 +1067,16 @@ __global__ void k_phase_l_sel_c(
     Prop *__restrict__ slots
 ) {
   __shared__ i64 red[24];
-  const u32 count = *l_count < l_cap ? *l_count : l_cap;
-  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+  const u64 hbase = vid_hi > vid_lo ? l_hoff[vid_lo] : 0;
+  for (u32 vid = vid_lo + blockIdx.x; vid < vid_hi; vid += gridDim.x) {
     const u64 rec = l_list[vid];
     const u32 p = static_cast<u32>(rec >> 32);
     const u32 u = static_cast<u32>(rec);
     const u32 cur = labels[u];
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
-    u32 *hk = pool_keys + l_hoff[vid];
-    i32 *hv = pool_vals + l_hoff[vid];
+    u32 *hk = pool_keys + (l_hoff[vid] - hbase);
+    i32 *hv = pool_vals + (l_hoff[vid] - hbase);
     const u32 slots_n = 1u << l_hbits[vid];
 
     BestState best{0, 0, 0, false};
@@ -1904,7 +1908,7 @@ struct kmp_lp_t {
   u64 pool_slots = 0;
   u64 *d_l_hoff = nullptr;       // l_cap+1
   u32 *d_l_hbits = nullptr;      // l_cap
-  int *d_pool_overflow = nullptr;
+  unsigned long long *d_l_hacc = nullptr; // total region demand of the chunk
   u64 *d_cand = nullptr;         // n (two-hop candidate slots)
   u64 *d_cand2 = nullptr;        // n (sorted)
   u32 *d_cfav = nullptr;         // n
@@ -1925,6 +1929,7 @@ struct kmp_lp_t {
   // pinned host mirrors
   u32 *h_count = nullptr;
   int *h_changed = nullptr;
+  unsigned long long *h_hacc = nullptr;
   unsigned long long *h_moves = nullptr; // [0..1]=moves before/after [2..3]=emptied
   u64 last_emptied = 0;
 
@@ -2057,6 +2062,7 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
   HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
   HIP_CHECK(hipHostMalloc(&e->h_moves, sizeof(unsigned long long) * 4));
+  HIP_CHECK(hipHostMalloc(&e->h_hacc, sizeof(unsigned long long)));
 }
 
 // Isolated-vertex scan used by the clusterer's isolated-node handling
@@ -2140,7 +2146,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   for (void *p : {(void *)e->d_favored, (void *)e->d_eflag, (void *)e->d_pool_keys,
                   (void *)e->d_pool_vals, (void *)e->d_l_hoff, (void *)e->d_l_hbits,
-                  (void *)e->d_pool_overflow, (void *)e->d_cand, (void *)e->d_cand2,
+                  (void *)e->d_l_hacc, (void *)e->d_cand, (void *)e->d_cand2,
                   (void *)e->d_cfav, (void *)e->d_crank, (void *)e->d_cones,
                   (void *)e->d_cand_select_temp, (void *)e->d_cand_sort_temp,
                   (void *)e->d_cand_scan_temp, (void *)e->d_emptied}) {
@@ -2156,6 +2162,9 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   if (e->h_moves) {
     (void)hipHostFree(e->h_moves);
+  }
+  if (e->h_hacc) {
+    (void)hipHostFree(e->h_hacc);
   }
   (void)hipStreamDestroy(e->stream);
   delete e;
@@ -2366,25 +2375,70 @@ i64 kmp_lp_phase_a(
     {
       hipLaunchKernelGGL(
           k_l_prep_c, dim3(1), dim3(1024), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
-          e->l_cap, e->pool_slots, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_pool_overflow
+          e->l_cap, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_l_hacc
       );
       LAUNCH_CHECK();
-      {
-        auto *kern = e->has_adjwgt ? k_phase_l_acc_c<false> : k_phase_l_acc_c<true>;
+      // read back the L count and the chunk's total hash-region demand:
+      // when it exceeds the pool (hub-dense chunks, e.g. after degree-bucket
+      // rearrangement packs 64 hubs per permutation unit), process the L
+      // list in pool-sized batches -- k_phase_l_sel_c clears each region
+      // after reading it, so batches can reuse the pool safely.
+      HIP_CHECK(hipMemcpyAsync(
+          e->h_count + 1, e->d_l_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream
+      ));
+      HIP_CHECK(hipMemcpyAsync(
+          e->h_hacc, e->d_l_hacc, sizeof(unsigned long long), hipMemcpyDeviceToHost, e->stream
+      ));
+      sync_spin(e);
+      const u32 lcount = e->h_count[1] < e->l_cap ? e->h_count[1] : e->l_cap;
+      const u64 hacc = *e->h_hacc;
+      auto *kern = e->has_adjwgt ? k_phase_l_acc_c<false> : k_phase_l_acc_c<true>;
+      auto launch_batch = [&](u32 lo, u32 hi) {
         hipLaunchKernelGGL(
             kern, dim3(2048), dim3(256), 0, e->stream, e->d_xadj, e->d_adjncy, e->d_adjwgt,
-            e->d_labels, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_hoff,
-            e->d_l_hbits, e->d_pool_keys, e->d_pool_vals
+            e->d_labels, e->d_l_list, lo, hi, e->d_l_off, e->d_l_hoff, e->d_l_hbits,
+            e->d_pool_keys, e->d_pool_vals
         );
         LAUNCH_CHECK();
+        hipLaunchKernelGGL(
+            k_phase_l_sel_c, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed,
+            e->n, e->maxw_uniform, e->d_vwgt, e->d_labels, e->d_weights, e->d_l_list, lo, hi,
+            e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_favored, e->d_slots
+        );
+        LAUNCH_CHECK();
+      };
+      if (lcount > 0 && hacc <= e->pool_slots) {
+        launch_batch(0, lcount);
+      } else if (lcount > 0) {
+        std::vector<u64> hoff(lcount + 1);
+        HIP_CHECK(hipMemcpyAsync(
+            hoff.data(), e->d_l_hoff, sizeof(u64) * (lcount + 1), hipMemcpyDeviceToHost,
+            e->stream
+        ));
+        sync_spin(e);
+        u32 lo = 0;
+        while (lo < lcount) {
+          const u64 need_one = hoff[lo + 1] - hoff[lo];
+          if (need_one > e->pool_slots) {
+            // a single region larger than the pool: grow it
+            HIP_CHECK(hipFree(e->d_pool_keys));
+            HIP_CHECK(hipFree(e->d_pool_vals));
+            e->pool_slots = need_one;
+            HIP_CHECK(hipMalloc(&e->d_pool_keys, sizeof(u32) * e->pool_slots));
+            HIP_CHECK(hipMalloc(&e->d_pool_vals, sizeof(i32) * e->pool_slots));
+            HIP_CHECK(hipMemsetAsync(e->d_pool_keys, 0xFF, sizeof(u32) * e->pool_slots,
+                                     e->stream));
+            HIP_CHECK(hipMemsetAsync(e->d_pool_vals, 0, sizeof(i32) * e->pool_slots,
+                                     e->stream));
+          }
+          u32 hi = lo + 1;
+          while (hi < lcount && hoff[hi + 1] - hoff[lo] <= e->pool_slots) {
+            ++hi;
+          }
+          launch_batch(lo, hi);
+          lo = hi;
+        }
       }
-      hipLaunchKernelGGL(
-          k_phase_l_sel_c, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->n,
-          e->maxw_uniform, e->d_vwgt, e->d_labels, e->d_weights, e->d_l_list, e->d_l_count,
-          e->l_cap, e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_favored,
-          e->d_slots
-      );
-      LAUNCH_CHECK();
     }
   }
   // compact valid slots in position order (stable select)
@@ -2399,20 +2453,11 @@ i64 kmp_lp_phase_a(
   HIP_CHECK(
       hipMemcpyAsync(e->h_count, e->d_prop_count, sizeof(u32), hipMemcpyDeviceToHost, e->stream)
   );
-  if (e->clusterer) {
-    HIP_CHECK(hipMemcpyAsync(
-        e->h_changed, e->d_pool_overflow, sizeof(int), hipMemcpyDeviceToHost, e->stream
-    ));
-  }
   sync_spin(e);
   float ms = 0;
   HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
   e->phase_a_ms += ms;
 
-  if (e->clusterer && *e->h_changed) {
-    fprintf(stderr, "kaminpar_amd: clustering L hash pool overflow (raise pool size)\n");
-    return -1;
-  }
   if (*e->h_count > cap) {
     fprintf(stderr, "kaminpar_amd: proposal buffer overflow (%u > %u)\n", *e->h_count, cap);
     return -1;
@@ -2760,7 +2805,7 @@ i64 kmp_lp_cluster(
     HIP_CHECK(hipMalloc(&e->d_eflag, n));
     HIP_CHECK(hipMemset(e->d_eflag, 0, n));
     HIP_CHECK(hipMalloc(&e->d_emptied, sizeof(unsigned long long)));
-    HIP_CHECK(hipMalloc(&e->d_pool_overflow, sizeof(int)));
+    HIP_CHECK(hipMalloc(&e->d_l_hacc, sizeof(unsigned long long)));
     HIP_CHECK(hipMalloc(&e->d_l_hoff, sizeof(u64) * (e->C + 1)));
     HIP_CHECK(hipMalloc(&e->d_l_hbits, sizeof(u32) * e->C));
     // pooled hash for high-degree rows: ~8 slots per average chunk arc
@@ -2802,7 +2847,6 @@ i64 kmp_lp_cluster(
     HIP_CHECK(hipFree(e->d_l_off));
   }
   HIP_CHECK(hipMalloc(&e->d_l_off, sizeof(u32) * (e->l_cap + 1)));
-  HIP_CHECK(hipMemsetAsync(e->d_pool_overflow, 0, sizeof(int), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_emptied, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_arcs, 0, sizeof(unsigned long long), e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_moves, 0, sizeof(unsigned long long), e->stream));
