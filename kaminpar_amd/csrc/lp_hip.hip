@@ -2087,6 +2087,14 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
     return nullptr;
   }
 
+  if (kmp_graph_m(g) > 0xFFFFFFFFull) {
+    // R-MAT scale-28/ef-8 dedups to just under 2^32 directed arcs and fits;
+    // anything larger needs the 64-bit edge-offset kernel variant (planned,
+    // DESIGN.md section 6)
+    fprintf(stderr, "kaminpar_amd: graphs with >= 2^32 directed arcs need the "
+                    "64-bit EdgeID kernel variant (not built yet)\n");
+    return nullptr;
+  }
   auto *e = new kmp_lp_t();
   (void)hipSetDeviceFlags(hipDeviceScheduleSpin); // ignore if context exists
   e->n = kmp_graph_n(g);
