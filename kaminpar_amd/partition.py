@@ -301,12 +301,14 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     sizes = [g.n]
     mappings = []
     engines = [engine if engine is not None else LpEngine(g)]
-    s0 = engines[0].get_stats()
-    arcs_prev, ns_prev = s0.arcs_scanned, s0.phase_a_ns
+    arcs_total = 0
+    ns_total = 0
     while sizes[-1] > max(stop_n, 2 * k):
         cur_n = sizes[-1]
         mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit)
-        nc, clus, _ = engines[-1].cluster(mcw, seed=seed + len(mappings), iters=iters)
+        nc, clus, cst = engines[-1].cluster(mcw, seed=seed + len(mappings), iters=iters)
+        arcs_total += cst.arcs_scanned
+        ns_total += cst.phase_a_ns
         coarse_eng, mapping = engines[-1].contract_engine(clus)
         if coarse_eng.n > 0.95 * cur_n:
             del coarse_eng
@@ -322,16 +324,12 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     # ---- uncoarsen: refine at every level (GPU) ----
     cut = None
     for level in range(len(engines) - 1, -1, -1):
-        cut, part, _ = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
+        cut, part, rst = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
+        arcs_total += rst.arcs_scanned
+        ns_total += rst.phase_a_ns
         if level > 0:
             part = part[mappings[level - 1]]
     levels = sizes
     if return_arcs:
-        arcs = -arcs_prev
-        ns = -ns_prev
-        for e in engines:
-            s = e.get_stats()
-            arcs += s.arcs_scanned
-            ns += s.phase_a_ns
-        return cut, part, levels, int(arcs), int(ns)
+        return cut, part, levels, int(arcs_total), int(ns_total)
     return cut, part, levels
