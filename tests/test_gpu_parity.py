@@ -472,3 +472,76 @@ def test_contract_engine_matches_contract(oracle):
     nc_b, clus_b, _ = coarse_eng.cluster(500, seed=2, iters=5)
     assert nc_a == nc_b
     assert np.array_equal(clus_a, clus_b)
+
+
+def _multi_hub_graph(n_hubs, n_leaves):
+    """n_hubs hub vertices (ids 0..n_hubs-1, one 64-vertex permutation unit
+    when n_hubs <= 64, so they all land in ONE chunk) each adjacent to every
+    leaf. Builds the worst case for the clustering L hash pool: a single
+    chunk whose total region demand exceeds the pool."""
+    xadj = np.zeros(n_hubs + n_leaves + 1, np.uint32)
+    xadj[1:n_hubs + 1] = n_leaves
+    xadj[n_hubs + 1:] = n_hubs
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    leaves = np.arange(n_hubs, n_hubs + n_leaves, dtype=np.uint32)
+    hubs = np.arange(n_hubs, dtype=np.uint32)
+    adjncy = np.concatenate([np.tile(leaves, n_hubs), np.tile(hubs, n_leaves)])
+    return ka.Graph.from_csr(xadj, adjncy.astype(np.uint32))
+
+
+@pytest.mark.gpu
+def test_cluster_pool_batching_parity(oracle):
+    """64 contiguous hubs of degree 64k share one chunk: region demand
+    64 x 128Ki slots = 8.4M > the 4.2M minimum pool, forcing the batched
+    L-pool path (the configuration that crashed before the pool fix).
+    Results must stay bit-identical to the oracle."""
+    _require_gpu()
+    from helpers import oracle_cluster
+
+    g = _multi_hub_graph(64, 65536)
+    eng = ka.LpEngine(g)
+    nc, clus, _ = eng.cluster(1000, seed=3, iters=5)
+    onc, oclus, _ = oracle_cluster(oracle, g, 1000, seed=3, iters=5)
+    assert nc == onc
+    assert np.array_equal(clus, oclus)
+
+
+@pytest.mark.gpu
+def test_cluster_pool_grow_parity(oracle):
+    """A single hub whose hash region alone exceeds the pool exercises the
+    pool-grow path."""
+    _require_gpu()
+    from helpers import oracle_cluster
+
+    g = _multi_hub_graph(1, 4_500_000)
+    eng = ka.LpEngine(g)
+    nc, clus, _ = eng.cluster(500_000, seed=1, iters=3)
+    onc, oclus, _ = oracle_cluster(oracle, g, 500_000, seed=1, iters=3)
+    assert nc == onc
+    assert np.array_equal(clus, oclus)
+
+
+@pytest.mark.gpu
+def test_parity_on_deg_bucket_rearranged_graph(oracle):
+    """The bench path runs on a degree-bucket-rearranged graph (hubs
+    contiguous): refine and cluster on that graph must match the oracle on
+    the same graph bit-exactly."""
+    _require_gpu()
+    from helpers import oracle_cluster, oracle_refine
+
+    g0 = ka.Graph.rmat(14, 8, seed=11)
+    g, _perm = g0.rearrange_degree_buckets()
+    eng = ka.LpEngine(g)
+
+    k = 16
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+    part0 = ka.random_partition(g.n, k, seed=5)
+    cut, part, _ = eng.refine(k, mbw, part0, seed=1, iters=5)
+    ocut, opart, _ = oracle_refine(oracle, g, k, mbw, part0, seed=1, iters=5)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+
+    nc, clus, _ = eng.cluster(128, seed=2, iters=5)
+    onc, oclus, _ = oracle_cluster(oracle, g, 128, seed=2, iters=5)
+    assert nc == onc
+    assert np.array_equal(clus, oclus)
