@@ -60,6 +60,9 @@ def summarize_db(path, out):
             or pick(sc, "kernel", "name") or pick(sc, "name")
         if sname:
             name_expr = f"s.{sname}"
+            alt = pick(sc, "kernel", "name")
+            if alt and alt != sname:
+                name_expr = f"COALESCE(NULLIF(s.{sname}, ''), s.{alt})"
             join = f"JOIN {sym_t} s ON d.{kid} = s.{sid}"
     if name_expr is None:
         nm = pick(dc, "name")
@@ -72,7 +75,7 @@ def summarize_db(path, out):
     q = (f"SELECT {name_expr} AS name, COUNT(*), SUM(d.{end}-d.{start}), "
          f"AVG(d.{end}-d.{start}) FROM {dis_t} d {join} GROUP BY 1")
     for name, n, total, avg in con.execute(q):
-        short = str(name).split("(")[0].strip()
+        short = shorten(name)
         k = out.setdefault("kernels", {}).setdefault(
             short, {"launches": 0, "total_ns": 0})
         k["launches"] += n
@@ -98,13 +101,13 @@ def summarize_db(path, out):
                 cname_expr = f"i.{iname}"
                 cjoin = f"JOIN {pmc_i} i ON p.{pid} = i.{iid}"
         if val and link and cname_expr:
-            dl = pick(dc, "id") or dc[0]
+            dl = link if link in dc else (pick(dc, "id") or dc[0])
             q = (f"SELECT {name_expr}, {cname_expr}, SUM(p.{val}), COUNT(*) "
                  f"FROM {pmc_t} p JOIN {dis_t} d ON p.{link} = d.{dl} {join} "
                  f"GROUP BY 1, 2")
             try:
                 for name, counter, total, n in con.execute(q):
-                    short = str(name).split("(")[0].strip()
+                    short = shorten(name)
                     k = out.setdefault("kernels", {}).setdefault(
                         short, {"launches": 0, "total_ns": 0})
                     c = k.setdefault("counters", {})
@@ -117,6 +120,15 @@ def summarize_db(path, out):
         elif any(con.execute(f"SELECT 1 FROM {pmc_t} LIMIT 1")):
             out.setdefault("_schema_debug", {})[path + ":pmc"] = dbg
     con.close()
+
+
+def shorten(name):
+    """'void (anonymous namespace)::k_phase_s(args)' -> 'k_phase_s';
+    'void rocprim::...::trampoline_kernel<...>(args)' -> 'trampoline_kernel'."""
+    s = str(name).replace("(anonymous namespace)::", "")
+    head = s.split("(")[0].split("<")[0].strip()
+    parts = head.split()
+    return parts[-1] if parts else s[:60]
 
 
 def main():
