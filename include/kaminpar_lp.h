@@ -215,6 +215,66 @@ kmp_graph_t *kmp_lp_download_graph(const kmp_lp_t *e);
 
 uint32_t kmp_lp_n(const kmp_lp_t *e);
 uint64_t kmp_lp_m(const kmp_lp_t *e);
+
+/* ------------------------------------------------- multilevel pipeline */
+
+/* Recursive-bisection initial partitioning on a (small) host graph: greedy
+ * graph growing from `reps` distinct high-degree seeds per bisection, each
+ * polished by two-way FM with best-prefix rollback, then a gain-aware
+ * overload balancer. Restates the reference's initial-partitioning recipe
+ * in simplified form (kaminpar-shm/initial_partitioning/). CPU-only. */
+int kmp_initial_partition(
+    const kmp_graph_t *g,
+    uint32_t k,
+    int64_t max_block_weight,
+    int reps,
+    uint32_t *part_out
+);
+
+/* Full multilevel partition on the GPU (BASELINE config 3; the shape of
+ * KaMinPar::compute_partition, kaminpar.h:970-1025): GPU LP clustering +
+ * device-resident contraction per level, CPU initial partitioning on the
+ * coarsest graph, GPU LP refinement at every level. Deterministic;
+ * bit-identical to the Python driver kaminpar_amd.partition.partition.
+ * Returns the final edge cut, or -1 on error. */
+int64_t kmp_partition(
+    const kmp_graph_t *g,
+    uint32_t k,
+    double eps,
+    uint64_t seed,
+    int iters,
+    uint32_t contraction_limit, /* 0 -> default 2000 */
+    uint32_t stop_n,            /* 0 -> default 512 */
+    int ip_reps,                /* 0 -> default 8 */
+    uint32_t *part_out
+);
+
+/* --------------------------------------------- ckaminpar-shaped C shim
+ * Mirrors the reference's public C interface (include/kaminpar-shm/
+ * ckaminpar.h:61-132: kaminpar_create / kaminpar_copy_graph /
+ * kaminpar_set_k / kaminpar_set_uniform_max_block_weights /
+ * kaminpar_compute_partition / kaminpar_free) with the same call order,
+ * argument meaning and default types (NodeID/EdgeID u32, weights i32).
+ * num_threads is accepted for signature parity; the implementation runs on
+ * the GPU. */
+typedef struct kaminpar_amd_t kaminpar_amd_t;
+
+kaminpar_amd_t *kaminpar_amd_create(int num_threads);
+void kaminpar_amd_free(kaminpar_amd_t *shm);
+void kaminpar_amd_reseed(kaminpar_amd_t *shm, int seed);
+void kaminpar_amd_copy_graph(
+    kaminpar_amd_t *shm,
+    uint32_t n,
+    const uint32_t *xadj,
+    const uint32_t *adjncy,
+    const int32_t *vwgt,   /* NULL for unit weights */
+    const int32_t *adjwgt  /* NULL for unit weights */
+);
+void kaminpar_amd_set_k(kaminpar_amd_t *shm, uint32_t k);
+void kaminpar_amd_set_uniform_max_block_weights(kaminpar_amd_t *shm, double epsilon);
+/* Returns the final edge cut, or -1 on error (missing GPU, no graph). */
+int64_t kaminpar_amd_compute_partition(kaminpar_amd_t *shm, uint32_t *partition);
+
 #ifdef __cplusplus
 }
 #endif

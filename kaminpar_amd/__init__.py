@@ -85,6 +85,11 @@ def _load():
     lib.kmp_max_block_weight.argtypes = [vp, u32, ctypes.c_double]
     lib.kmp_rearrange_degree_buckets.restype = vp
     lib.kmp_rearrange_degree_buckets.argtypes = [vp, p(u32)]
+    lib.kmp_initial_partition.restype = ctypes.c_int
+    lib.kmp_initial_partition.argtypes = [vp, u32, i64, ctypes.c_int, p(u32)]
+    lib.kmp_partition.restype = i64
+    lib.kmp_partition.argtypes = [vp, u32, ctypes.c_double, u64, ctypes.c_int,
+                                  u32, u32, ctypes.c_int, p(u32)]
 
     lib.kmp_lp_create.restype = vp
     lib.kmp_lp_create.argtypes = [vp]
@@ -208,6 +213,29 @@ class Graph:
 
     def max_block_weight(self, k, eps=0.03):
         return _lib.kmp_max_block_weight(self._h, k, eps)
+
+    def partition_native(self, k, eps=0.03, seed=1, iters=5):
+        """Full multilevel partition driven entirely from the C ABI
+        (kmp_partition; the shape of KaMinPar::compute_partition).
+        Bit-identical to kaminpar_amd.partition.partition. Requires a GPU.
+
+        Returns (cut, partition)."""
+        part = np.zeros(self.n, dtype=np.uint32)
+        cut = _lib.kmp_partition(self._h, k, eps, seed, iters, 0, 0, 0,
+                                 _u32p(part))
+        if cut < 0:
+            raise RuntimeError("kmp_partition failed")
+        return cut, part
+
+    def initial_partition_native(self, k, max_block_weight, reps=8):
+        """C++ recursive-bisection initial partitioning (equivalent to
+        kaminpar_amd.partition.initial_partition; host-only, no GPU)."""
+        part = np.zeros(self.n, dtype=np.uint32)
+        rc = _lib.kmp_initial_partition(self._h, k, max_block_weight, reps,
+                                        _u32p(part))
+        if rc != 0:
+            raise RuntimeError("kmp_initial_partition failed")
+        return part
 
     def rearrange_degree_buckets(self):
         """Degree-bucket rearrangement (the reference's default

@@ -3349,3 +3349,178 @@ u32 kmp_lp_n(const kmp_lp_t *e) { return e->n; }
 u64 kmp_lp_m(const kmp_lp_t *e) { return e->m; }
 
 } // extern "C"
+
+// ---------------------------------------------------------------------------
+// Full multilevel pipeline (C-ABI twin of kaminpar_amd/partition.py --
+// keep the schedule in sync; bit-identical by construction and pinned by
+// tests/test_gpu_parity.py::test_c_abi_partition_matches_python).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+i64 level_cluster_weight(i64 total_w, u32 n, u32 k, double eps, u32 C) {
+  u64 shrink = n / C;
+  if (shrink < 2) {
+    shrink = 2;
+  }
+  if (shrink > k) {
+    shrink = k;
+  }
+  const i64 eps_rule = static_cast<i64>(eps * static_cast<double>(total_w) /
+                                        static_cast<double>(shrink));
+  const i64 block_rule = total_w / (12ll * k);
+  i64 mcw = block_rule > 0 ? std::min(eps_rule, block_rule) : eps_rule;
+  return mcw < 1 ? 1 : mcw;
+}
+
+} // namespace
+
+extern "C" {
+
+i64 kmp_partition(
+    const kmp_graph_t *g, u32 k, double eps, u64 seed, int iters,
+    u32 contraction_limit, u32 stop_n, int ip_reps, u32 *part_out
+) {
+  if (contraction_limit == 0) {
+    contraction_limit = 2000;
+  }
+  if (stop_n == 0) {
+    stop_n = 512;
+  }
+  if (ip_reps == 0) {
+    ip_reps = 8;
+  }
+  const i64 total_w = kmp_graph_total_node_weight(g);
+  const i64 mbw_val = kmp_max_block_weight(g, k, eps);
+  std::vector<i64> mbw(k, mbw_val);
+
+  std::vector<kmp_lp_t *> engines;
+  std::vector<u32> sizes;
+  std::vector<std::vector<u32>> mappings;
+  engines.push_back(kmp_lp_create(g));
+  if (!engines[0]) {
+    return -1;
+  }
+  sizes.push_back(kmp_graph_n(g));
+
+  i64 rc = -1;
+  kmp_graph_t *coarsest = nullptr;
+  std::vector<u32> clus, part;
+  kmp_lp_stats_t st;
+
+  const u32 stop = std::max(stop_n, 2 * k);
+  while (sizes.back() > stop) {
+    const u32 cur_n = sizes.back();
+    const i64 mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit);
+    clus.resize(cur_n);
+    if (kmp_lp_cluster(engines.back(), mcw, 0, clus.data(),
+                       seed + mappings.size(), iters, &st) < 0) {
+      goto done;
+    }
+    {
+      std::vector<u32> mapping(cur_n);
+      kmp_lp_t *coarse_eng = nullptr;
+      if (kmp_contract_engine(engines.back(), clus.data(), mapping.data(),
+                              &coarse_eng) < 0) {
+        goto done;
+      }
+      const u32 c_n = kmp_lp_n(coarse_eng);
+      if (static_cast<double>(c_n) > 0.95 * static_cast<double>(cur_n)) {
+        kmp_lp_free(coarse_eng);
+        break;
+      }
+      engines.push_back(coarse_eng);
+      mappings.push_back(std::move(mapping));
+      sizes.push_back(c_n);
+    }
+  }
+
+  // initial partition on the coarsest graph (CPU)
+  coarsest = engines.size() > 1 ? kmp_lp_download_graph(engines.back()) : nullptr;
+  part.resize(sizes.back());
+  if (kmp_initial_partition(coarsest ? coarsest : g, k, mbw_val, ip_reps,
+                            part.data()) != 0) {
+    goto done;
+  }
+
+  // uncoarsen: refine at every level, projecting through the mappings
+  for (size_t level = engines.size(); level-- > 0;) {
+    rc = kmp_lp_refine(engines[level], k, mbw.data(), part.data(), seed,
+                       iters, &st);
+    if (rc < 0) {
+      goto done;
+    }
+    if (level > 0) {
+      const std::vector<u32> &map = mappings[level - 1];
+      std::vector<u32> fine(map.size());
+      for (size_t u = 0; u < map.size(); ++u) {
+        fine[u] = part[map[u]];
+      }
+      part = std::move(fine);
+    }
+  }
+  for (u32 u = 0; u < kmp_graph_n(g); ++u) {
+    part_out[u] = part[u];
+  }
+
+done:
+  if (coarsest) {
+    kmp_graph_free(coarsest);
+  }
+  for (kmp_lp_t *e : engines) {
+    kmp_lp_free(e);
+  }
+  return rc;
+}
+
+// --------------------------- ckaminpar-shaped shim (see kaminpar_lp.h) ----
+
+struct kaminpar_amd_t {
+  kmp_graph_t *g = nullptr;
+  u32 k = 2;
+  double eps = 0.03;
+  u64 seed = 1;
+};
+
+kaminpar_amd_t *kaminpar_amd_create(int /*num_threads*/) {
+  return new kaminpar_amd_t();
+}
+
+void kaminpar_amd_free(kaminpar_amd_t *shm) {
+  if (shm) {
+    if (shm->g) {
+      kmp_graph_free(shm->g);
+    }
+    delete shm;
+  }
+}
+
+void kaminpar_amd_reseed(kaminpar_amd_t *shm, int seed) {
+  shm->seed = static_cast<u64>(seed);
+}
+
+void kaminpar_amd_copy_graph(
+    kaminpar_amd_t *shm, u32 n, const u32 *xadj, const u32 *adjncy,
+    const i32 *vwgt, const i32 *adjwgt
+) {
+  if (shm->g) {
+    kmp_graph_free(shm->g);
+  }
+  shm->g = kmp_graph_from_csr(n, xadj[n], xadj, adjncy, vwgt, adjwgt);
+}
+
+void kaminpar_amd_set_k(kaminpar_amd_t *shm, u32 k) { shm->k = k; }
+
+void kaminpar_amd_set_uniform_max_block_weights(kaminpar_amd_t *shm, double epsilon) {
+  shm->eps = epsilon;
+}
+
+i64 kaminpar_amd_compute_partition(kaminpar_amd_t *shm, u32 *partition) {
+  if (!shm->g) {
+    fprintf(stderr, "kaminpar_amd: no graph set (call kaminpar_amd_copy_graph)\n");
+    return -1;
+  }
+  return kmp_partition(shm->g, shm->k, shm->eps, shm->seed, 5, 0, 0, 0, partition);
+}
+
+} // extern "C"

@@ -1,0 +1,426 @@
+// Host-side initial partitioning for the multilevel pipeline: recursive
+// bisection (greedy graph growing from distinct high-degree seeds + two-way
+// FM with best-prefix rollback, best-of-`reps` per bisection) followed by a
+// gain-aware k-way overload balancer.
+//
+// Restates the reference's initial-partitioning recipe in simplified form
+// (kaminpar-shm/initial_partitioning/: initial_ggg_bipartitioner.cc,
+// initial_two_way_fm_refiner.cc with num_fruitless_moves=100 and 5 passes,
+// refinement/balancer/overload_balancer.cc in spirit). Deterministic and
+// EXACTLY equivalent to kaminpar_amd/partition.py's numpy implementation --
+// tests/test_pipeline_cpu.py pins the equivalence, and the committed
+// pipeline goldens (tests/golden/pipeline_expected.json) pin the results.
+// CPU-only: no HIP; runs in the dev container.
+
+#include <algorithm>
+#include <cstdint>
+#include <cstdio>
+#include <vector>
+
+using u32 = uint32_t;
+using u64 = uint64_t;
+using i32 = int32_t;
+using i64 = int64_t;
+
+extern "C" {
+// from graph_gen.cpp
+typedef struct kmp_graph_t kmp_graph_t;
+u32 kmp_graph_n(const kmp_graph_t *);
+u64 kmp_graph_m(const kmp_graph_t *);
+const u32 *kmp_graph_xadj(const kmp_graph_t *);
+const u32 *kmp_graph_adjncy(const kmp_graph_t *);
+const i32 *kmp_graph_vwgt(const kmp_graph_t *);
+const i32 *kmp_graph_adjwgt(const kmp_graph_t *);
+}
+
+namespace {
+
+struct SubCsr {
+  std::vector<i64> xadj;
+  std::vector<i64> adj; // local ids
+  std::vector<i64> w;
+};
+
+// Induced-subgraph CSR in local ids (nodes ascending; local id = position).
+SubCsr subgraph_csr(
+    const u32 *xadj, const u32 *adjncy, const i32 *adjwgt,
+    const std::vector<i64> &nodes, std::vector<i64> &loc
+) {
+  SubCsr s;
+  s.xadj.assign(nodes.size() + 1, 0);
+  for (size_t i = 0; i < nodes.size(); ++i) {
+    const u32 u = static_cast<u32>(nodes[i]);
+    for (u64 e = xadj[u]; e < xadj[u + 1]; ++e) {
+      const i64 j = loc[adjncy[e]];
+      if (j >= 0) {
+        s.adj.push_back(j);
+        s.w.push_back(adjwgt ? adjwgt[e] : 1);
+      }
+    }
+    s.xadj[i + 1] = static_cast<i64>(s.adj.size());
+  }
+  return s;
+}
+
+// Greedy graph growing (partition.py _greedy_grow): grow part 1 from the
+// seed_rank-th vertex in descending-degree order (ties: smaller local id).
+std::vector<uint8_t> greedy_grow(
+    const SubCsr &s, const std::vector<i64> &vw, i64 target1, i64 cap1,
+    int seed_rank
+) {
+  const size_t n = vw.size();
+  std::vector<i64> deg(n);
+  for (size_t i = 0; i < n; ++i) {
+    deg[i] = s.xadj[i + 1] - s.xadj[i];
+  }
+  std::vector<u32> order(n);
+  for (size_t i = 0; i < n; ++i) {
+    order[i] = static_cast<u32>(i);
+  }
+  std::stable_sort(order.begin(), order.end(),
+                   [&](u32 a, u32 b) { return deg[a] > deg[b]; });
+  const u32 seed_idx = order[static_cast<size_t>(seed_rank) % n];
+
+  std::vector<uint8_t> side(n, 0);
+  std::vector<i64> gain(n, -1); // -1 not frontier, -2 blocked/in
+  i64 w1 = 0;
+
+  auto add = [&](u32 i) {
+    side[i] = 1;
+    w1 += vw[i];
+    gain[i] = -2;
+    for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+      const i64 j = s.adj[e];
+      if (!side[j] && gain[j] != -2) {
+        if (gain[j] < 0) {
+          gain[j] = 0;
+        }
+        gain[j] += s.w[e];
+      }
+    }
+  };
+
+  add(seed_idx);
+  while (w1 < target1) {
+    // frontier argmax (ties: smallest id), else smallest unblocked outside
+    i64 best_g = -1;
+    i64 nxt = -1;
+    for (size_t i = 0; i < n; ++i) {
+      if (gain[i] > best_g) {
+        best_g = gain[i];
+        nxt = static_cast<i64>(i);
+      }
+    }
+    if (best_g < 0) {
+      nxt = -1;
+      for (size_t i = 0; i < n; ++i) {
+        if (!side[i] && gain[i] != -2) {
+          nxt = static_cast<i64>(i);
+          break;
+        }
+      }
+      if (nxt < 0) {
+        break;
+      }
+    }
+    if (w1 + vw[nxt] > cap1) {
+      gain[nxt] = -2;
+      continue;
+    }
+    add(static_cast<u32>(nxt));
+  }
+  return side;
+}
+
+// Two-way FM with best-prefix rollback (partition.py _fm_refine_bisection).
+void fm_refine(
+    const SubCsr &s, const std::vector<i64> &vw, std::vector<uint8_t> &side,
+    i64 cap1, i64 cap2, int max_passes = 5, int max_fruitless = 100
+) {
+  const size_t n = side.size();
+  i64 total = 0;
+  for (i64 w : vw) {
+    total += w;
+  }
+  i64 w1 = 0;
+  for (size_t i = 0; i < n; ++i) {
+    if (side[i]) {
+      w1 += vw[i];
+    }
+  }
+  std::vector<i64> gains(n);
+  std::vector<uint8_t> locked(n);
+  std::vector<u32> moves;
+  for (int pass = 0; pass < max_passes; ++pass) {
+    for (size_t i = 0; i < n; ++i) {
+      i64 cross = 0, intern = 0;
+      for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+        if (side[s.adj[e]] != side[i]) {
+          cross += s.w[e];
+        } else {
+          intern += s.w[e];
+        }
+      }
+      gains[i] = cross - intern;
+    }
+    std::fill(locked.begin(), locked.end(), 0);
+    moves.clear();
+    i64 cum = 0, best = 0;
+    size_t best_len = 0;
+    int fruitless = 0;
+    i64 wa = w1, wb = total - w1;
+    while (fruitless < max_fruitless) {
+      i64 best_g = 0;
+      i64 pick = -1;
+      bool found = false;
+      for (size_t i = 0; i < n; ++i) {
+        if (locked[i]) {
+          continue;
+        }
+        const bool feas =
+            side[i] ? (wb + vw[i] <= cap2) : (wa + vw[i] <= cap1);
+        if (!feas) {
+          continue;
+        }
+        if (!found || gains[i] > best_g) {
+          best_g = gains[i];
+          pick = static_cast<i64>(i);
+          found = true;
+        }
+      }
+      if (!found) {
+        break;
+      }
+      const u32 i = static_cast<u32>(pick);
+      cum += gains[i];
+      const uint8_t old = side[i];
+      for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+        const i64 j = s.adj[e];
+        if (!locked[j]) {
+          gains[j] += (side[j] == old) ? 2 * s.w[e] : -2 * s.w[e];
+        }
+      }
+      gains[i] = -gains[i];
+      side[i] = !old;
+      locked[i] = 1;
+      if (old) {
+        wa -= vw[i];
+        wb += vw[i];
+      } else {
+        wa += vw[i];
+        wb -= vw[i];
+      }
+      moves.push_back(i);
+      if (cum > best) {
+        best = cum;
+        best_len = moves.size();
+        fruitless = 0;
+      } else {
+        ++fruitless;
+      }
+    }
+    for (size_t m = best_len; m < moves.size(); ++m) {
+      side[moves[m]] = !side[moves[m]];
+    }
+    w1 = 0;
+    for (size_t i = 0; i < n; ++i) {
+      if (side[i]) {
+        w1 += vw[i];
+      }
+    }
+    if (best <= 0) {
+      break;
+    }
+  }
+}
+
+// 2x the bisection cut (both arc directions), for best-of-reps selection.
+i64 bisection_cut2(const SubCsr &s, const std::vector<uint8_t> &side) {
+  i64 c = 0;
+  for (size_t i = 0; i < side.size(); ++i) {
+    for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+      if (side[s.adj[e]] != side[i]) {
+        c += s.w[e];
+      }
+    }
+  }
+  return c;
+}
+
+struct IpCtx {
+  const u32 *xadj;
+  const u32 *adjncy;
+  const i32 *adjwgt;
+  std::vector<i64> vwgt;
+  std::vector<i64> loc;
+  i64 mbw;
+  int reps;
+  u32 *part;
+};
+
+void rec(IpCtx &c, std::vector<i64> &nodes, u32 k_lo, u32 k_hi) {
+  if (nodes.empty()) {
+    return;
+  }
+  if (k_hi - k_lo == 1) {
+    for (i64 u : nodes) {
+      c.part[u] = k_lo;
+    }
+    return;
+  }
+  const u32 k1 = (k_hi - k_lo + 1) / 2;
+  const u32 k2 = (k_hi - k_lo) - k1;
+  i64 total = 0;
+  for (i64 u : nodes) {
+    total += c.vwgt[u];
+  }
+  const i64 target1 = total * k1 / (k1 + k2);
+  const i64 cap1 = static_cast<i64>(k1) * c.mbw;
+  const i64 cap2 = static_cast<i64>(k2) * c.mbw;
+
+  for (size_t i = 0; i < nodes.size(); ++i) {
+    c.loc[nodes[i]] = static_cast<i64>(i);
+  }
+  SubCsr s = subgraph_csr(c.xadj, c.adjncy, c.adjwgt, nodes, c.loc);
+  for (i64 u : nodes) {
+    c.loc[u] = -1;
+  }
+  std::vector<i64> vw(nodes.size());
+  for (size_t i = 0; i < nodes.size(); ++i) {
+    vw[i] = c.vwgt[nodes[i]];
+  }
+
+  std::vector<uint8_t> best_side;
+  i64 best_cut = -1;
+  for (int rep = 0; rep < c.reps; ++rep) {
+    std::vector<uint8_t> side = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, side, cap1, cap2);
+    const i64 cut = bisection_cut2(s, side);
+    if (best_cut < 0 || cut < best_cut) {
+      best_cut = cut;
+      best_side = std::move(side);
+    }
+  }
+
+  std::vector<i64> p1, p2;
+  for (size_t i = 0; i < nodes.size(); ++i) {
+    (best_side[i] ? p1 : p2).push_back(nodes[i]);
+  }
+  rec(c, p1, k_lo, k_lo + k1);
+  rec(c, p2, k_lo + k1, k_hi);
+}
+
+// Gain-aware overload balancer (partition.py _balance).
+void balance(
+    const u32 *xadj, const u32 *adjncy, const i32 *adjwgt,
+    const std::vector<i64> &vw, u32 *part, u32 n, u32 k, i64 cap
+) {
+  std::vector<i64> bw(k, 0);
+  for (u32 u = 0; u < n; ++u) {
+    bw[part[u]] += vw[u];
+  }
+  i64 guard = 8 * (static_cast<i64>(k) + 16);
+  std::vector<i64> conn(k);
+  while (guard-- > 0) {
+    u32 b = 0;
+    for (u32 t = 1; t < k; ++t) {
+      if (bw[t] > bw[b]) {
+        b = t;
+      }
+    }
+    if (bw[b] <= cap) {
+      break;
+    }
+    // best feasible move: max (gain, -weight, -v, t); else the move that
+    // most lowers the overloaded block: max (-new_target_weight, gain, -v)
+    bool have_f = false, have_o = false;
+    i64 f0 = 0, f1 = 0, f2 = 0;
+    i64 o0 = 0, o1 = 0, o2 = 0;
+    u32 f3 = 0, fv = 0, ot = 0, ov = 0;
+    for (u32 v = 0; v < n; ++v) {
+      if (part[v] != b) {
+        continue;
+      }
+      std::fill(conn.begin(), conn.end(), 0);
+      for (u64 e = xadj[v]; e < xadj[v + 1]; ++e) {
+        conn[part[adjncy[e]]] += adjwgt ? adjwgt[e] : 1;
+      }
+      const i64 internal = conn[b];
+      for (u32 t = 0; t < k; ++t) {
+        if (t == b) {
+          continue;
+        }
+        const i64 nw = bw[t] + vw[v];
+        const i64 g = conn[t] - internal;
+        if (nw <= cap) {
+          const i64 k0 = g, k1v = -vw[v], k2v = -static_cast<i64>(v);
+          if (!have_f || k0 > f0 ||
+              (k0 == f0 && (k1v > f1 ||
+               (k1v == f1 && (k2v > f2 ||
+                (k2v == f2 && t > f3)))))) {
+            have_f = true;
+            f0 = k0; f1 = k1v; f2 = k2v; f3 = t; fv = v;
+          }
+        } else if (nw < bw[b]) {
+          const i64 k0 = -nw, k1v = g, k2v = -static_cast<i64>(v);
+          if (!have_o || k0 > o0 ||
+              (k0 == o0 && (k1v > o1 || (k1v == o1 && k2v > o2)))) {
+            have_o = true;
+            o0 = k0; o1 = k1v; o2 = k2v; ot = t; ov = v;
+          }
+        }
+      }
+    }
+    u32 t, v;
+    if (have_f) {
+      t = f3; v = fv;
+    } else if (have_o) {
+      t = ot; v = ov;
+    } else {
+      break;
+    }
+    part[v] = t;
+    bw[b] -= vw[v];
+    bw[t] += vw[v];
+  }
+}
+
+} // namespace
+
+extern "C" {
+
+// Recursive-bisection initial partitioning on a (small) host graph.
+// Equivalent to kaminpar_amd.partition.initial_partition (numpy); the
+// equivalence is pinned by tests/test_pipeline_cpu.py.
+int kmp_initial_partition(
+    const kmp_graph_t *g, u32 k, i64 max_block_weight, int reps, u32 *part_out
+) {
+  const u32 n = kmp_graph_n(g);
+  const u32 *xadj = kmp_graph_xadj(g);
+  const u32 *adjncy = kmp_graph_adjncy(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  const i32 *adjwgt = kmp_graph_adjwgt(g);
+
+  IpCtx c;
+  c.xadj = xadj;
+  c.adjncy = adjncy;
+  c.adjwgt = adjwgt;
+  c.vwgt.resize(n);
+  for (u32 u = 0; u < n; ++u) {
+    c.vwgt[u] = vwgt ? vwgt[u] : 1;
+  }
+  c.loc.assign(n, -1);
+  c.mbw = max_block_weight;
+  c.reps = reps;
+  c.part = part_out;
+
+  std::vector<i64> all(n);
+  for (u32 u = 0; u < n; ++u) {
+    all[u] = u;
+  }
+  rec(c, all, 0, k);
+  balance(xadj, adjncy, adjwgt, c.vwgt, part_out, n, k, max_block_weight);
+  return 0;
+}
+
+} // extern "C"

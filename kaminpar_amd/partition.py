@@ -317,9 +317,10 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
         mappings.append(mapping)
         sizes.append(coarse_eng.n)
 
-    # ---- initial partition (CPU, coarsest downloaded from HBM) ----
+    # ---- initial partition (CPU, coarsest downloaded from HBM; the C++
+    # implementation, bit-identical to initial_partition() below) ----
     coarsest = engines[-1].download_graph() if len(engines) > 1 else g
-    part = initial_partition(coarsest, k, mbw_val, seed=seed)
+    part = coarsest.initial_partition_native(k, mbw_val)
 
     # ---- uncoarsen: refine at every level (GPU) ----
     cut = None

@@ -545,3 +545,18 @@ def test_parity_on_deg_bucket_rearranged_graph(oracle):
     onc, oclus, _ = oracle_cluster(oracle, g, 128, seed=2, iters=5)
     assert nc == onc
     assert np.array_equal(clus, oclus)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", ["walshaw_k16", "rmat14_s42_k16", "rmat18_s42_k64"])
+def test_c_abi_partition_matches_python(name):
+    """kmp_partition (the all-C-ABI multilevel driver behind the
+    ckaminpar-shaped shim) is bit-identical to the Python pipeline and
+    therefore to the committed goldens."""
+    _require_gpu()
+    g, exp, band = _pipeline_case(name)
+    cut, part = g.partition_native(exp["k"], seed=1)
+    assert cut == exp["cut"], (cut, exp["cut"])
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]

@@ -71,3 +71,21 @@ def test_pipeline_band_vs_reference_goldens(oracle):
     for name in ("walshaw_k16", "rmat14_s42_k16"):
         ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
         assert exp[name]["cut"] <= 1.75 * ref_best, name
+
+
+@pytest.mark.parametrize("case,k", [("walshaw", 2), ("walshaw", 16),
+                                    ("rgg2d", 5), ("rmat12", 16)])
+def test_native_initial_partition_matches_python(case, k):
+    """The C++ initial partitioner (kmp_initial_partition, used by the
+    product pipeline and the C-ABI driver) is bit-identical to the numpy
+    reference implementation."""
+    if case == "walshaw":
+        g = _graph("walshaw_k2")
+    elif case == "rgg2d":
+        g = _graph("rgg2d_k4")
+    else:
+        g = ka.Graph.rmat(12, 8, 42)
+    mbw = g.max_block_weight(k, 0.03)
+    py = initial_partition(g, k, mbw, seed=1, reps=8)
+    cc = g.initial_partition_native(k, mbw, reps=8)
+    assert np.array_equal(py, cc)
