@@ -106,7 +106,8 @@ def summarize_db(path, out):
                  f"FROM {pmc_t} p JOIN {dis_t} d ON p.{link} = d.{dl} {join} "
                  f"GROUP BY 1, 2")
             try:
-                for name, counter, total, n in con.execute(q):
+                rows = list(con.execute(q))
+                for name, counter, total, n in rows:
                     short = shorten(name)
                     k = out.setdefault("kernels", {}).setdefault(
                         short, {"launches": 0, "total_ns": 0})
@@ -114,6 +115,20 @@ def summarize_db(path, out):
                     c[str(counter)] = c.get(str(counter), 0) + float(total or 0)
                     c[f"{counter}_dispatches"] = \
                         c.get(f"{counter}_dispatches", 0) + n
+                if not rows:
+                    npmc = next(con.execute(f"SELECT COUNT(*) FROM {pmc_t}"))[0]
+                    if npmc:
+                        out.setdefault("_schema_debug", {})[path + ":pmc"] = {
+                            "note": f"join matched 0 of {npmc} pmc rows",
+                            "query": q,
+                            "pmc_sample": [list(r) for r in con.execute(
+                                f"SELECT * FROM {pmc_t} LIMIT 3")],
+                            "dispatch_sample": [list(r) for r in con.execute(
+                                f"SELECT {', '.join(dc[:8])} FROM {dis_t} "
+                                f"LIMIT 3")],
+                            "dispatch_cols": dc,
+                            "pmc_cols": pc,
+                        }
             except sqlite3.Error as exc:
                 out.setdefault("_schema_debug", {})[path + ":pmc"] = \
                     {"error": str(exc), **dbg}
