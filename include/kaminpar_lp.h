@@ -60,6 +60,14 @@ kmp_graph_t *kmp_gen_rgg2d(uint32_t n, double avg_deg, uint64_t seed);
 /* METIS ASCII reader (kaminpar-io/metis_parser.h:17-25 format). */
 kmp_graph_t *kmp_read_metis(const char *path);
 
+/* ParHIP binary reader/writer (docs/graph_file_format.md "ParHIP Graph
+ * File Format"; kaminpar-io/parhip_parser.cc:42-136): 24-byte header
+ * (version bit-field, n, m), byte offsets, adjacency, optional weights.
+ * Reader accepts 32- and 64-bit stored ids/weights that fit u32/i32;
+ * writer emits 32-bit ids (64-bit offsets when the file is large). */
+kmp_graph_t *kmp_read_parhip(const char *path);
+int kmp_write_parhip(const kmp_graph_t *g, const char *path);
+
 uint32_t kmp_graph_n(const kmp_graph_t *g);
 uint64_t kmp_graph_m(const kmp_graph_t *g); /* number of directed arcs */
 const uint32_t *kmp_graph_xadj(const kmp_graph_t *g);

@@ -16,6 +16,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <type_traits>
 #include <vector>
 
 #ifdef _OPENMP
@@ -451,6 +452,158 @@ i64 kmp_max_block_weight(const kmp_graph_t *g, u32 k, double eps) {
   // context.cc:27-39: (1+eps) * ceil(total_node_weight / k), truncated to int
   const double pbw = std::ceil(1.0 * g->total_node_weight / k);
   return static_cast<i64>((1.0 + eps) * pbw);
+}
+
+// ParHIP binary format (docs/graph_file_format.md "ParHIP Graph File
+// Format"; kaminpar-io/parhip_parser.cc:42-136): 24-byte header (version
+// bit-field, n, m as u64), then BYTE offsets ((n+1) x EdgeID width,
+// relative to the file start), adjacency (m x NodeID width), optional node
+// weights, optional edge weights. Version bits (0 = present / 64-bit):
+// b0 edge weights absent, b1 node weights absent, b2 edge ids 32-bit,
+// b3 node ids 32-bit, b4 node weights 32-bit, b5 edge weights 32-bit.
+kmp_graph_t *kmp_read_parhip(const char *path) {
+  FILE *f = std::fopen(path, "rb");
+  if (!f) {
+    std::fprintf(stderr, "kaminpar_amd: cannot open %s\n", path);
+    return nullptr;
+  }
+  u64 header[3];
+  if (std::fread(header, 8, 3, f) != 3) {
+    std::fclose(f);
+    return nullptr;
+  }
+  const u64 version = header[0];
+  const u64 n = header[1];
+  const u64 m = header[2];
+  const bool has_ewgt = (version & 1) == 0;
+  const bool has_vwgt = (version & 2) == 0;
+  const int eid_w = (version & 4) == 0 ? 8 : 4;
+  const int nid_w = (version & 8) == 0 ? 8 : 4;
+  const int vw_w = (version & 16) == 0 ? 8 : 4;
+  const int ew_w = (version & 32) == 0 ? 8 : 4;
+  if (n > 0xFFFFFFFFull || m > 0xFFFFFFFFull) {
+    std::fprintf(stderr, "kaminpar_amd: parhip graph too large for u32 ids\n");
+    std::fclose(f);
+    return nullptr;
+  }
+
+  auto read_ints = [&](int width, u64 count, auto &out) -> bool {
+    out.resize(count);
+    if (width == 8) {
+      std::vector<u64> tmp(count);
+      if (std::fread(tmp.data(), 8, count, f) != count) {
+        return false;
+      }
+      for (u64 i = 0; i < count; ++i) {
+        out[i] = static_cast<typename std::decay_t<decltype(out)>::value_type>(tmp[i]);
+      }
+    } else {
+      std::vector<u32> tmp(count);
+      if (std::fread(tmp.data(), 4, count, f) != count) {
+        return false;
+      }
+      for (u64 i = 0; i < count; ++i) {
+        out[i] = static_cast<typename std::decay_t<decltype(out)>::value_type>(tmp[i]);
+      }
+    }
+    return true;
+  };
+
+  // offsets are byte addresses; map to edge indices
+  const u64 nodes_base = 24 + (n + 1) * static_cast<u64>(eid_w);
+  std::vector<u64> off;
+  if (!read_ints(eid_w, n + 1, off)) {
+    std::fclose(f);
+    return nullptr;
+  }
+  auto *g = new kmp_graph_t();
+  g->n = static_cast<u32>(n);
+  g->m = m;
+  g->xadj.resize(n + 1);
+  for (u64 i = 0; i <= n; ++i) {
+    g->xadj[i] = static_cast<u32>((off[i] - nodes_base) / nid_w);
+  }
+  std::vector<u64> adj;
+  if (!read_ints(nid_w, m, adj)) {
+    delete g;
+    std::fclose(f);
+    return nullptr;
+  }
+  g->adjncy.resize(m);
+  for (u64 e = 0; e < m; ++e) {
+    g->adjncy[e] = static_cast<u32>(adj[e]);
+  }
+  g->total_node_weight = static_cast<i64>(n);
+  if (has_vwgt) {
+    std::vector<i64> vw;
+    if (!read_ints(vw_w, n, vw)) {
+      delete g;
+      std::fclose(f);
+      return nullptr;
+    }
+    g->vwgt.resize(n);
+    i64 tot = 0;
+    for (u64 i = 0; i < n; ++i) {
+      g->vwgt[i] = static_cast<i32>(vw[i]);
+      tot += vw[i];
+    }
+    g->total_node_weight = tot;
+  }
+  if (has_ewgt) {
+    std::vector<i64> ew;
+    if (!read_ints(ew_w, m, ew)) {
+      delete g;
+      std::fclose(f);
+      return nullptr;
+    }
+    g->adjwgt.resize(m);
+    for (u64 e = 0; e < m; ++e) {
+      g->adjwgt[e] = static_cast<i32>(ew[e]);
+    }
+  }
+  std::fclose(f);
+  return g;
+}
+
+// Writer (32-bit ids and weights), for fixtures and round-trip tests.
+int kmp_write_parhip(const kmp_graph_t *g, const char *path) {
+  FILE *f = std::fopen(path, "wb");
+  if (!f) {
+    return -1;
+  }
+  const bool has_vwgt = !g->vwgt.empty();
+  const bool has_ewgt = !g->adjwgt.empty();
+  // byte offsets can exceed u32 for large graphs: pick the EdgeID width
+  const bool wide_eid = 24 + (static_cast<u64>(g->n) + 1) * 8 + g->m * 4 > 0xFFFFFFFFull;
+  // bits set = absent / 32-bit (see reader above)
+  const u64 version = (has_ewgt ? 0 : 1) | (has_vwgt ? 0 : 2) |
+                      (wide_eid ? 0 : 4) | 8 | 16 | 32;
+  u64 header[3] = {version, g->n, g->m};
+  std::fwrite(header, 8, 3, f);
+  const u64 eid_w = wide_eid ? 8 : 4;
+  const u64 nodes_base = 24 + (static_cast<u64>(g->n) + 1) * eid_w;
+  if (wide_eid) {
+    std::vector<u64> off(g->n + 1);
+    for (u32 i = 0; i <= g->n; ++i) {
+      off[i] = nodes_base + static_cast<u64>(g->xadj[i]) * 4;
+    }
+    std::fwrite(off.data(), 8, off.size(), f);
+  } else {
+    std::vector<u32> off(g->n + 1);
+    for (u32 i = 0; i <= g->n; ++i) {
+      off[i] = static_cast<u32>(nodes_base + static_cast<u64>(g->xadj[i]) * 4);
+    }
+    std::fwrite(off.data(), 4, off.size(), f);
+  }
+  std::fwrite(g->adjncy.data(), 4, g->adjncy.size(), f);
+  if (has_vwgt) {
+    std::fwrite(g->vwgt.data(), 4, g->vwgt.size(), f);
+  }
+  if (has_ewgt) {
+    std::fwrite(g->adjwgt.data(), 4, g->adjwgt.size(), f);
+  }
+  std::fclose(f);
+  return 0;
 }
 
 // Degree-bucket rearrangement: stable counting sort of the vertices by
