@@ -104,7 +104,7 @@ def summarize_db(path, out):
             dl = link if link in dc else (pick(dc, "id") or dc[0])
             q = (f"SELECT {name_expr}, {cname_expr}, SUM(p.{val}), COUNT(*) "
                  f"FROM {pmc_t} p JOIN {dis_t} d ON p.{link} = d.{dl} {join} "
-                 f"GROUP BY 1, 2")
+                 f"{cjoin} GROUP BY 1, 2")
             try:
                 rows = list(con.execute(q))
                 for name, counter, total, n in rows:
