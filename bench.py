@@ -45,6 +45,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--scale", type=int, default=26, help="R-MAT scale")
     ap.add_argument("--edgefactor", type=int, default=8)
+    ap.add_argument("--graph", choices=["rmat", "rgg2d"], default="rmat",
+                    help="rgg2d generates the BASELINE config-4 graph "
+                         "(n=2^scale, avg degree 16, high locality)")
     ap.add_argument("--k", type=int, default=16)
     ap.add_argument("--iters", type=int, default=5)
     ap.add_argument("--seed", type=int, default=1)
@@ -87,16 +90,21 @@ def main():
     else:
         device = "cuda:0"
 
-    log(f"[bench] generating R-MAT scale-{args.scale} ef={args.edgefactor} ...")
+    def generate():
+        if args.graph == "rgg2d":
+            return ka.Graph.rgg2d(1 << args.scale, avg_deg=16.0, seed=42)
+        return ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+
+    log(f"[bench] generating {args.graph} scale-{args.scale} ...")
     t0 = time.time()
     if world > 1:
         # one generation per node (8 concurrent generator scratches would
         # exhaust host RAM): rank 0 generates + saves, the rest load
         import torch.distributed as dist
 
-        cache = f"/tmp/kmp_rmat{args.scale}_{args.edgefactor}_{args.order}"
+        cache = f"/tmp/kmp_{args.graph}{args.scale}_{args.edgefactor}_{args.order}"
         if rank == 0:
-            g0 = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+            g0 = generate()
             if args.order == "deg-buckets":
                 g0, _perm = g0.rearrange_degree_buckets()
             np.save(cache + "_xadj.npy", np.asarray(g0.xadj))
@@ -107,7 +115,7 @@ def main():
         adjncy = np.load(cache + "_adjncy.npy", mmap_mode="r")
         g = ka.Graph.from_csr(np.asarray(xadj), np.asarray(adjncy))
     else:
-        g = ka.Graph.rmat(args.scale, args.edgefactor, seed=42)
+        g = generate()
         if args.order == "deg-buckets":
             t1 = time.time()
             g, _perm = g.rearrange_degree_buckets()
@@ -253,9 +261,9 @@ def main():
             "dtype": "int32",
             "data": "synthetic",
             "config": {
-                "workload": (f"rmat{args.scale}_k{args.k}_multilevel"
+                "workload": (f"{args.graph}{args.scale}_k{args.k}_multilevel"
                              if args.workload == "partition" else
-                             f"rmat{args.scale}_k{args.k}_lp_{args.workload}"),
+                             f"{args.graph}{args.scale}_k{args.k}_lp_{args.workload}"),
                 "n": int(g.n),
                 "arcs": int(g.m),
                 "k": k,

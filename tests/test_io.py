@@ -55,3 +55,35 @@ def test_parhip_reads_64bit_stored_ids(tmp_path):
     assert h.n == n and h.m == 4
     assert list(h.xadj) == xadj
     assert list(h.adjncy) == adjncy
+
+
+def test_c_shim_fails_loudly_without_gpu():
+    """kaminpar_amd_compute_partition returns -1 (no silent CPU fallback)
+    when no HIP device is available; skipped if a GPU is present."""
+    import ctypes
+    import torch
+    if torch.cuda.is_available():
+        import pytest
+        pytest.skip("GPU present")
+    from kaminpar_amd import _lib, _u32p
+    lib = ctypes.CDLL(None)  # symbols are in the already-loaded library
+    h = _lib_shim = _lib  # reuse loaded lib handle for the shim symbols
+    _lib.kaminpar_amd_create.restype = ctypes.c_void_p
+    _lib.kaminpar_amd_compute_partition.restype = ctypes.c_int64
+    _lib.kaminpar_amd_compute_partition.argtypes = [ctypes.c_void_p,
+                                                    ctypes.POINTER(ctypes.c_uint32)]
+    shm = _lib.kaminpar_amd_create(1)
+    g = ka.Graph.rmat(10, 8, 1)
+    _lib.kaminpar_amd_copy_graph.argtypes = [
+        ctypes.c_void_p, ctypes.c_uint32, ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.c_void_p, ctypes.c_void_p]
+    xadj = np.ascontiguousarray(g.xadj)
+    adjncy = np.ascontiguousarray(g.adjncy)
+    _lib.kaminpar_amd_copy_graph(shm, g.n, _u32p(xadj), _u32p(adjncy), None, None)
+    _lib.kaminpar_amd_set_k.argtypes = [ctypes.c_void_p, ctypes.c_uint32]
+    _lib.kaminpar_amd_set_k(shm, 4)
+    part = np.zeros(g.n, np.uint32)
+    cut = _lib.kaminpar_amd_compute_partition(shm, _u32p(part))
+    assert cut == -1
+    _lib.kaminpar_amd_free.argtypes = [ctypes.c_void_p]
+    _lib.kaminpar_amd_free(shm)
