@@ -116,6 +116,10 @@ class TorchComm:
         cat = torch.cat([self._gather_bufs[r][: counts[r]] for r in range(self.world)])
         cat = cat.contiguous()
         self._cat_keepalive = cat  # keep device memory alive through commit
+        # the engine commits on its own HIP stream: make sure the collective
+        # + cat (torch's stream) are complete before handing over the pointer
+        if cat.is_cuda:
+            torch.cuda.current_stream().synchronize()
         return cat.data_ptr(), total
 
 
