@@ -91,6 +91,9 @@ def _load():
     lib.kmp_rearrange_degree_buckets.argtypes = [vp, p(u32)]
     lib.kmp_initial_partition.restype = ctypes.c_int
     lib.kmp_initial_partition.argtypes = [vp, u32, i64, ctypes.c_int, p(u32)]
+    lib.kmp_bisect_subset.restype = ctypes.c_int
+    lib.kmp_bisect_subset.argtypes = [vp, p(u32), u32, i64, i64, i64,
+                                      ctypes.c_int, p(ctypes.c_uint8)]
     lib.kmp_partition.restype = i64
     lib.kmp_partition.argtypes = [vp, u32, ctypes.c_double, u64, ctypes.c_int,
                                   u32, u32, ctypes.c_int, p(u32)]
@@ -238,6 +241,18 @@ class Graph:
         if cut < 0:
             raise RuntimeError("kmp_partition failed")
         return cut, part
+
+    def bisect_subset(self, nodes, target1, cap1, cap2, reps=8):
+        """Bisect an arbitrary vertex subset (greedy grow + FM, best of
+        reps). Returns a boolean side array aligned with `nodes`."""
+        nodes = np.ascontiguousarray(nodes, dtype=np.uint32)
+        side = np.zeros(len(nodes), dtype=np.uint8)
+        rc = _lib.kmp_bisect_subset(
+            self._h, _u32p(nodes), len(nodes), int(target1), int(cap1),
+            int(cap2), reps, side.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+        if rc != 0:
+            raise RuntimeError("kmp_bisect_subset failed")
+        return side.astype(bool)
 
     def initial_partition_native(self, k, max_block_weight, reps=8):
         """C++ recursive-bisection initial partitioning (equivalent to

@@ -389,6 +389,51 @@ void balance(
 
 extern "C" {
 
+// Bisect an arbitrary vertex subset of a host graph: `reps` greedy-grow
+// attempts from distinct high-degree seeds, each FM-polished, best
+// (2x directed) bisection cut kept. side_out[i] = 1 puts nodes[i] in part 1
+// (weight target target1, cap cap1). Used by the progressive-k partition
+// extension (the shape of the reference's deep-multilevel bisection
+// extension, kaminpar-shm/partitioning/deep/deep_multilevel.cc) and
+// equivalent to one recursion step of kmp_initial_partition.
+int kmp_bisect_subset(
+    const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
+    i64 cap1, i64 cap2, int reps, uint8_t *side_out
+) {
+  const u32 n = kmp_graph_n(g);
+  const u32 *xadj = kmp_graph_xadj(g);
+  const u32 *adjncy = kmp_graph_adjncy(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  const i32 *adjwgt = kmp_graph_adjwgt(g);
+
+  std::vector<i64> nodes(nodes_in, nodes_in + n_sub);
+  std::vector<i64> loc(n, -1);
+  for (u32 i = 0; i < n_sub; ++i) {
+    loc[nodes[i]] = i;
+  }
+  SubCsr s = subgraph_csr(xadj, adjncy, adjwgt, nodes, loc);
+  std::vector<i64> vw(n_sub);
+  for (u32 i = 0; i < n_sub; ++i) {
+    vw[i] = vwgt ? vwgt[nodes[i]] : 1;
+  }
+
+  std::vector<uint8_t> best_side;
+  i64 best_cut = -1;
+  for (int rep = 0; rep < reps; ++rep) {
+    std::vector<uint8_t> side = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, side, cap1, cap2);
+    const i64 cut = bisection_cut2(s, side);
+    if (best_cut < 0 || cut < best_cut) {
+      best_cut = cut;
+      best_side = std::move(side);
+    }
+  }
+  for (u32 i = 0; i < n_sub; ++i) {
+    side_out[i] = best_side[i];
+  }
+  return 0;
+}
+
 // Recursive-bisection initial partitioning on a (small) host graph.
 // Equivalent to kaminpar_amd.partition.initial_partition (numpy); the
 // equivalence is pinned by tests/test_pipeline_cpu.py.
