@@ -150,7 +150,7 @@ def main():
         if args.workload == "partition":
             from types import SimpleNamespace
 
-            from kaminpar_amd.partition import partition as ml_partition
+            from kaminpar_amd.partition import partition_deep as ml_partition
 
             cut, _part, levels, arcs, ns = ml_partition(
                 g, k, seed=args.seed, iters=args.iters, engine=eng,

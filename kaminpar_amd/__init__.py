@@ -91,9 +91,14 @@ def _load():
     lib.kmp_rearrange_degree_buckets.argtypes = [vp, p(u32)]
     lib.kmp_initial_partition.restype = ctypes.c_int
     lib.kmp_initial_partition.argtypes = [vp, u32, i64, ctypes.c_int, p(u32)]
+    lib.kmp_balance_partition.restype = ctypes.c_int
+    lib.kmp_balance_partition.argtypes = [vp, u32, i64, p(u32)]
     lib.kmp_bisect_subset.restype = ctypes.c_int
     lib.kmp_bisect_subset.argtypes = [vp, p(u32), u32, i64, i64, i64,
                                       ctypes.c_int, p(ctypes.c_uint8)]
+    lib.kmp_bisect_subset_ml.restype = ctypes.c_int
+    lib.kmp_bisect_subset_ml.argtypes = [vp, p(u32), u32, i64, i64, i64,
+                                         ctypes.c_int, p(ctypes.c_uint8)]
     lib.kmp_partition.restype = i64
     lib.kmp_partition.argtypes = [vp, u32, ctypes.c_double, u64, ctypes.c_int,
                                   u32, u32, ctypes.c_int, p(u32)]
@@ -242,6 +247,12 @@ class Graph:
             raise RuntimeError("kmp_partition failed")
         return cut, part
 
+    def balance_partition(self, k, cap, part):
+        """Gain-aware overload balancer (uniform cap), in place."""
+        part = np.ascontiguousarray(part, dtype=np.uint32)
+        _lib.kmp_balance_partition(self._h, k, int(cap), _u32p(part))
+        return part
+
     def bisect_subset(self, nodes, target1, cap1, cap2, reps=8):
         """Bisect an arbitrary vertex subset (greedy grow + FM, best of
         reps). Returns a boolean side array aligned with `nodes`."""
@@ -252,6 +263,18 @@ class Graph:
             int(cap2), reps, side.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
         if rc != 0:
             raise RuntimeError("kmp_bisect_subset failed")
+        return side.astype(bool)
+
+    def bisect_subset_ml(self, nodes, target1, cap1, cap2, reps=8):
+        """Multilevel bisection of a vertex subset (heavy-edge matching
+        + FM at every level; best of reps)."""
+        nodes = np.ascontiguousarray(nodes, dtype=np.uint32)
+        side = np.zeros(len(nodes), dtype=np.uint8)
+        rc = _lib.kmp_bisect_subset_ml(
+            self._h, _u32p(nodes), len(nodes), int(target1), int(cap1),
+            int(cap2), reps, side.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+        if rc != 0:
+            raise RuntimeError("kmp_bisect_subset_ml failed")
         return side.astype(bool)
 
     def initial_partition_native(self, k, max_block_weight, reps=8):

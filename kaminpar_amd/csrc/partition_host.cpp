@@ -234,6 +234,128 @@ void fm_refine(
   }
 }
 
+u64 mix64(u64 x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+// Multilevel bisection: heavy-edge matching -> contract -> recurse -> FM at
+// every level on the way back up (the shape of the reference's
+// initial-partitioner pool, which runs its own mini-multilevel per
+// bisection with TWOWAY_SIMPLE_FM refinement at every level,
+// presets.cc: initial_partitioning.coarsening contraction_limit 20 +
+// pool.refinement TWOWAY_SIMPLE_FM). rep varies the matching visit order
+// and the base-case grow seed.
+void ml_bisect_rec(
+    const SubCsr &s, const std::vector<i64> &vw, i64 target1, i64 cap1,
+    i64 cap2, int rep, std::vector<uint8_t> &side
+) {
+  const size_t n = vw.size();
+  i64 total = 0;
+  for (i64 w : vw) {
+    total += w;
+  }
+  if (n <= 128) {
+    side = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, side, cap1, cap2);
+    return;
+  }
+
+  // heavy-edge matching (visit order: hash of (id, rep); ties smaller id)
+  std::vector<u32> order(n);
+  for (size_t i = 0; i < n; ++i) {
+    order[i] = static_cast<u32>(i);
+  }
+  std::stable_sort(order.begin(), order.end(), [&](u32 a, u32 b) {
+    return mix64(a * 2654435761u + rep) < mix64(b * 2654435761u + rep);
+  });
+  const i64 wlimit = std::max<i64>(1, total / 32);
+  std::vector<i64> match(n, -1);
+  for (u32 u : order) {
+    if (match[u] >= 0) {
+      continue;
+    }
+    i64 best = -1, bw = -1;
+    for (i64 e = s.xadj[u]; e < s.xadj[u + 1]; ++e) {
+      const i64 v = s.adj[e];
+      if (v == (i64)u || match[v] >= 0 || vw[u] + vw[v] > wlimit) {
+        continue;
+      }
+      if (s.w[e] > bw || (s.w[e] == bw && v < best)) {
+        bw = s.w[e];
+        best = v;
+      }
+    }
+    match[u] = best >= 0 ? best : (i64)u;
+    if (best >= 0) {
+      match[best] = u;
+    }
+  }
+
+  // coarse ids in fine-index order
+  std::vector<i64> cid(n, -1);
+  u32 cn = 0;
+  for (size_t u = 0; u < n; ++u) {
+    if (cid[u] < 0) {
+      cid[u] = cn;
+      if (match[u] != (i64)u) {
+        cid[match[u]] = cn;
+      }
+      ++cn;
+    }
+  }
+  if (cn > 0.95 * n) { // matching stalled: fall back to flat
+    side = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, side, cap1, cap2);
+    return;
+  }
+
+  // contract
+  std::vector<i64> cvw(cn, 0);
+  for (size_t u = 0; u < n; ++u) {
+    cvw[cid[u]] += vw[u];
+  }
+  std::vector<std::pair<u64, i64>> arcs;
+  arcs.reserve(s.adj.size());
+  for (size_t u = 0; u < n; ++u) {
+    for (i64 e = s.xadj[u]; e < s.xadj[u + 1]; ++e) {
+      const i64 cu = cid[u], cv = cid[s.adj[e]];
+      if (cu != cv) {
+        arcs.emplace_back((static_cast<u64>(cu) << 32) | static_cast<u32>(cv),
+                          s.w[e]);
+      }
+    }
+  }
+  std::sort(arcs.begin(), arcs.end());
+  SubCsr cs;
+  cs.xadj.assign(cn + 1, 0);
+  for (size_t i = 0; i < arcs.size();) {
+    size_t j = i;
+    i64 wsum = 0;
+    while (j < arcs.size() && arcs[j].first == arcs[i].first) {
+      wsum += arcs[j].second;
+      ++j;
+    }
+    cs.adj.push_back(static_cast<i64>(arcs[i].first & 0xFFFFFFFFu));
+    cs.w.push_back(wsum);
+    ++cs.xadj[(arcs[i].first >> 32) + 1];
+    i = j;
+  }
+  for (u32 c = 0; c < cn; ++c) {
+    cs.xadj[c + 1] += cs.xadj[c];
+  }
+
+  std::vector<uint8_t> cside;
+  ml_bisect_rec(cs, cvw, target1, cap1, cap2, rep, cside);
+  side.resize(n);
+  for (size_t u = 0; u < n; ++u) {
+    side[u] = cside[cid[u]];
+  }
+  fm_refine(s, vw, side, cap1, cap2);
+}
+
 // 2x the bisection cut (both arc directions), for best-of-reps selection.
 i64 bisection_cut2(const SubCsr &s, const std::vector<uint8_t> &side) {
   i64 c = 0;
@@ -422,6 +544,64 @@ int kmp_bisect_subset(
   for (int rep = 0; rep < reps; ++rep) {
     std::vector<uint8_t> side = greedy_grow(s, vw, target1, cap1, rep);
     fm_refine(s, vw, side, cap1, cap2);
+    const i64 cut = bisection_cut2(s, side);
+    if (best_cut < 0 || cut < best_cut) {
+      best_cut = cut;
+      best_side = std::move(side);
+    }
+  }
+  for (u32 i = 0; i < n_sub; ++i) {
+    side_out[i] = best_side[i];
+  }
+  return 0;
+}
+
+// Gain-aware overload balancer on a host graph (uniform cap), exposed for
+// the progressive-k driver.
+int kmp_balance_partition(
+    const kmp_graph_t *g, u32 k, i64 cap, u32 *part
+) {
+  const u32 n = kmp_graph_n(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  std::vector<i64> vw(n);
+  for (u32 u = 0; u < n; ++u) {
+    vw[u] = vwgt ? vwgt[u] : 1;
+  }
+  balance(kmp_graph_xadj(g), kmp_graph_adjncy(g), kmp_graph_adjwgt(g), vw,
+          part, n, k, cap);
+  return 0;
+}
+
+// Multilevel bisection of a vertex subset (heavy-edge matching coarsening
+// with FM at every level; best of `reps` matching orders). Higher quality
+// than kmp_bisect_subset on subgraphs beyond a few hundred vertices; used
+// by the progressive-k partition extension.
+int kmp_bisect_subset_ml(
+    const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
+    i64 cap1, i64 cap2, int reps, uint8_t *side_out
+) {
+  const u32 n = kmp_graph_n(g);
+  const u32 *xadj = kmp_graph_xadj(g);
+  const u32 *adjncy = kmp_graph_adjncy(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  const i32 *adjwgt = kmp_graph_adjwgt(g);
+
+  std::vector<i64> nodes(nodes_in, nodes_in + n_sub);
+  std::vector<i64> loc(n, -1);
+  for (u32 i = 0; i < n_sub; ++i) {
+    loc[nodes[i]] = i;
+  }
+  SubCsr s = subgraph_csr(xadj, adjncy, adjwgt, nodes, loc);
+  std::vector<i64> vw(n_sub);
+  for (u32 i = 0; i < n_sub; ++i) {
+    vw[i] = vwgt ? vwgt[nodes[i]] : 1;
+  }
+
+  std::vector<uint8_t> best_side;
+  i64 best_cut = -1;
+  for (int rep = 0; rep < reps; ++rep) {
+    std::vector<uint8_t> side;
+    ml_bisect_rec(s, vw, target1, cap1, cap2, rep, side);
     const i64 cut = bisection_cut2(s, side);
     if (best_cut < 0 || cut < best_cut) {
       best_cut = cut;

@@ -334,3 +334,112 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     if return_arcs:
         return cut, part, levels, int(arcs_total), int(ns_total)
     return cut, part, levels
+
+
+def _group_caps(groups, k, mbw_val):
+    caps = np.zeros(k, dtype=np.int64)
+    for b, w in groups:
+        caps[b] = w * mbw_val
+    return caps
+
+
+def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
+                      force=False):
+    """Split every splittable block group in half via FM-polished bisection
+    of its induced subgraph (the shape of the reference's deep-multilevel
+    partition extension, kaminpar-shm/partitioning/deep/deep_multilevel.cc:
+    bipartition blocks while uncoarsening instead of full-k at the coarsest).
+    groups is a list of (first_block_id, width); repeats until every block
+    would drop below split_c vertices (or all widths are 1)."""
+    from . import _lib
+
+    vwp = _lib.kmp_graph_vwgt(hg._h)
+    vw = (np.ctypeslib.as_array(vwp, shape=(hg.n,)).astype(np.int64)
+          if vwp else np.ones(hg.n, np.int64))
+    while True:
+        num = len(groups)
+        if num >= k:
+            break
+        if not force and hg.n < 2 * split_c * num:
+            break
+        new_groups = []
+        for b, w in groups:
+            if w < 2:
+                new_groups.append((b, w))
+                continue
+            k1 = (w + 1) // 2
+            k2 = w - k1
+            nodes = np.flatnonzero(part == b).astype(np.uint32)
+            if len(nodes) == 0:
+                new_groups += [(b, k1), (b + k1, k2)]
+                continue
+            total = int(vw[nodes].sum())
+            t1 = total * k1 // w
+            side = hg.bisect_subset(nodes, t1, k1 * mbw_val, k2 * mbw_val,
+                                    reps=reps)
+            part[nodes[~side]] = b + k1
+            new_groups += [(b, k1), (b + k1, k2)]
+        groups = new_groups
+    return part, groups
+
+
+def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
+                   stop_n=512, split_c=2000, reps=8, engine=None,
+                   return_arcs=False):
+    """Progressive-k multilevel partition: coarsen as in partition(), then
+    instead of full-k initial partitioning at the coarsest level, grow k by
+    FM-polished block bisections DURING uncoarsening whenever every block
+    still holds >= split_c vertices -- the shape of the reference's deep
+    multilevel mode. LP refinement runs at every level with per-group block
+    caps (width x uniform cap; unopened block ids get cap 0).
+
+    Returns (cut, partition, level_sizes)."""
+    total_w = g.total_node_weight
+    mbw_val = g.max_block_weight(k, eps)
+
+    sizes = [g.n]
+    mappings = []
+    engines = [engine if engine is not None else LpEngine(g)]
+    arcs_total = 0
+    ns_total = 0
+    while sizes[-1] > max(stop_n, 2 * k):
+        cur_n = sizes[-1]
+        mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit)
+        nc, clus, cst = engines[-1].cluster(mcw, seed=seed + len(mappings),
+                                           iters=iters)
+        arcs_total += cst.arcs_scanned
+        ns_total += cst.phase_a_ns
+        coarse_eng, mapping = engines[-1].contract_engine(clus)
+        if coarse_eng.n > 0.95 * cur_n:
+            del coarse_eng
+            break
+        engines.append(coarse_eng)
+        mappings.append(mapping)
+        sizes.append(coarse_eng.n)
+
+    part = np.zeros(sizes[-1], dtype=np.uint32)
+    groups = [(0, k)]
+    cut = None
+    coarsest = len(engines) - 1
+    for level in range(coarsest, -1, -1):
+        # at the coarsest level split eagerly (down to ~32-vertex blocks,
+        # like the reference's initial bipartition of the coarsest graph);
+        # afterwards extend only when every block keeps >= split_c vertices
+        sc = min(split_c, 48) if level == coarsest else split_c
+        if len(groups) < k and (sizes[level] >= 2 * sc * len(groups)
+                                or level == 0):
+            hg = g if level == 0 else engines[level].download_graph()
+            part, groups = _extend_partition(hg, part, groups, mbw_val, k,
+                                             sc, reps,
+                                             force=(level == 0))
+            if len(groups) == k:
+                hg.balance_partition(k, mbw_val, part)
+        cut, part, rst = engines[level].refine(
+            k, _group_caps(groups, k, mbw_val), part, seed=seed, iters=iters)
+        arcs_total += rst.arcs_scanned
+        ns_total += rst.phase_a_ns
+        if level > 0:
+            part = part[mappings[level - 1]]
+    if return_arcs:
+        return cut, part, sizes, int(arcs_total), int(ns_total)
+    return cut, part, sizes

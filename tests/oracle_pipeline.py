@@ -77,3 +77,61 @@ def oracle_partition(oracle, g, k, eps=0.03, seed=1, iters=5,
         if level > 0:
             part = part[mappings[level - 1]]
     return cut, part, [gr.n for gr in graphs]
+
+
+def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
+                          contraction_limit=2000, stop_n=512, split_c=2000,
+                          reps=8):
+    """CPU mirror of kaminpar_amd.partition.partition_deep (keep in sync):
+    progressive-k extension by FM-polished block bisections during
+    uncoarsening."""
+    from kaminpar_amd.partition import (_extend_partition, _group_caps,
+                                        level_cluster_weight)
+
+    total_w = g.total_node_weight
+    mbw_val = g.max_block_weight(k, eps)
+
+    def weights(gr):
+        from kaminpar_amd import _lib
+        vw = _lib.kmp_graph_vwgt(gr._h)
+        aw = _lib.kmp_graph_adjwgt(gr._h)
+        v = np.ctypeslib.as_array(vw, shape=(gr.n,)) if vw else None
+        a = np.ctypeslib.as_array(aw, shape=(gr.m,)) if aw else None
+        return v, a
+
+    graphs = [g]
+    mappings = []
+    while graphs[-1].n > max(stop_n, 2 * k):
+        cur = graphs[-1]
+        mcw = level_cluster_weight(total_w, cur.n, k, eps, contraction_limit)
+        vw, aw = weights(cur)
+        nc, clus, _ = oracle_cluster(oracle, cur, mcw, seed=seed + len(mappings),
+                                     iters=iters, vwgt=vw, adjwgt=aw)
+        coarse, mapping = oracle_contract(oracle, cur, clus, vwgt=vw, adjwgt=aw)
+        if coarse.n > 0.95 * cur.n:
+            break
+        graphs.append(coarse)
+        mappings.append(mapping)
+
+    part = np.zeros(graphs[-1].n, dtype=np.uint32)
+    groups = [(0, k)]
+    cut = None
+    coarsest = len(graphs) - 1
+    for level in range(coarsest, -1, -1):
+        hg = graphs[level]
+        sc = min(split_c, 48) if level == coarsest else split_c
+        if len(groups) < k and (hg.n >= 2 * sc * len(groups)
+                                or level == 0):
+            part, groups = _extend_partition(hg, part, groups, mbw_val, k,
+                                             sc, reps,
+                                             force=(level == 0))
+            if len(groups) == k:
+                hg.balance_partition(k, mbw_val, part)
+        vw, aw = weights(hg)
+        cut, part, _ = oracle_refine(oracle, hg, k,
+                                     _group_caps(groups, k, mbw_val), part,
+                                     seed=seed, iters=iters, vwgt=vw,
+                                     adjwgt=aw)
+        if level > 0:
+            part = part[mappings[level - 1]]
+    return cut, part, [gr.n for gr in graphs]

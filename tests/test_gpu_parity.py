@@ -560,3 +560,35 @@ def test_c_abi_partition_matches_python(name):
     checksum = int(np.bitwise_xor.reduce(
         np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
     assert checksum == exp["part_checksum"]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", [
+    "walshaw_k16", "rgg2d_k4", "rmat14_s42_k16", "rmat18_s42_k64",
+])
+def test_partition_deep_pipeline(name):
+    """Progressive-k (deep) pipeline on the GPU engine: bit-identical to the
+    oracle mirror's committed goldens (pipeline_deep_expected.json)."""
+    _require_gpu()
+    from kaminpar_amd.partition import partition_deep
+
+    import json
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    exp = json.load(open(os.path.join(here, "golden",
+                                      "pipeline_deep_expected.json")))[name]
+    g, _old_exp, band = _pipeline_case(name)
+    k = exp["k"]
+    cut, part, levels = partition_deep(g, k, seed=1)
+    assert cut == exp["cut"], (cut, exp["cut"])
+    assert levels == exp["levels"]
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]
+    from kaminpar_amd import _lib
+    vwp = _lib.kmp_graph_vwgt(g._h)
+    vw = np.ctypeslib.as_array(vwp, shape=(g.n,)).astype(np.int64) if vwp \
+        else np.ones(g.n, np.int64)
+    bw = np.zeros(k, np.int64)
+    np.add.at(bw, part, vw)
+    assert bw.max() <= band["cap"]

@@ -89,3 +89,33 @@ def test_native_initial_partition_matches_python(case, k):
     py = initial_partition(g, k, mbw, seed=1, reps=8)
     cc = g.initial_partition_native(k, mbw, reps=8)
     assert np.array_equal(py, cc)
+
+
+@pytest.mark.parametrize("name", [
+    "walshaw_k2", "walshaw_k16", "rgg2d_k4", "rmat14_s42_k16",
+])
+def test_oracle_deep_pipeline_matches_expected(oracle, name):
+    """The progressive-k (deep) pipeline mirror reproduces the committed
+    expected cuts."""
+    from oracle_pipeline import oracle_partition_deep
+
+    exp = _load("pipeline_deep_expected.json")[name]
+    g = _graph(name)
+    cut, part, levels = oracle_partition_deep(oracle, g, exp["k"], seed=1)
+    assert cut == exp["cut"], (cut, exp["cut"])
+    assert levels == exp["levels"]
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]
+
+
+def test_deep_pipeline_band_vs_reference_goldens():
+    """The deep pipeline's cuts stay within a tighter band of the compiled
+    reference's deep-multilevel cuts than the basic pipeline's (<=1.5x the
+    best seed on these cases)."""
+    exp = _load("pipeline_deep_expected.json")
+    band = _load("ref_golden_partition.json")
+    for name in ("walshaw_k2", "walshaw_k16", "rgg2d_k4", "rmat14_s42_k16",
+                 "rmat16_s42_k16", "rmat18_s42_k64"):
+        ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
+        assert exp[name]["cut"] <= 1.5 * ref_best, (name, exp[name]["cut"], ref_best)
