@@ -257,6 +257,50 @@ int64_t kmp_partition(
     uint32_t *part_out
 );
 
+/* Progressive-k extension helper (C twin of partition.py
+ * _extend_partition): split block groups in half via FM-polished subset
+ * bisections while every block keeps >= split_c vertices (or force != 0).
+ * group_lo/group_w are k-sized arrays holding *num_groups entries. */
+int kmp_extend_partition(
+    const kmp_graph_t *g, uint32_t *part, uint32_t k, int64_t mbw_val,
+    uint32_t split_c, int reps, int force, uint32_t *group_lo,
+    uint32_t *group_w, uint32_t *num_groups
+);
+
+/* Overload balancer (uniform cap) on a host graph, in place. */
+int kmp_balance_partition(
+    const kmp_graph_t *g, uint32_t k, int64_t cap, uint32_t *part
+);
+
+/* Flat / multilevel FM bisection of a vertex subset (see partition_host.cpp). */
+int kmp_bisect_subset(
+    const kmp_graph_t *g, const uint32_t *nodes, uint32_t n_sub,
+    int64_t target1, int64_t cap1, int64_t cap2, int reps, uint8_t *side_out
+);
+int kmp_bisect_subset_ml(
+    const kmp_graph_t *g, const uint32_t *nodes, uint32_t n_sub,
+    int64_t target1, int64_t cap1, int64_t cap2, int reps, uint8_t *side_out
+);
+
+/* Progressive-k (deep) multilevel partition: grows k by block bisections
+ * during uncoarsening (the shape of the reference's deep multilevel mode,
+ * kaminpar-shm/partitioning/deep/deep_multilevel.cc). Bit-identical to the
+ * Python driver kaminpar_amd.partition.partition_deep. Better cuts than
+ * kmp_partition on every golden case. 0 defaults:
+ * contraction_limit 2000, stop_n 512, split_c 2000, ip_reps 8. */
+int64_t kmp_partition_deep(
+    const kmp_graph_t *g,
+    uint32_t k,
+    double eps,
+    uint64_t seed,
+    int iters,
+    uint32_t contraction_limit,
+    uint32_t stop_n,
+    uint32_t split_c,
+    int ip_reps,
+    uint32_t *part_out
+);
+
 /* --------------------------------------------- ckaminpar-shaped C shim
  * Mirrors the reference's public C interface (include/kaminpar-shm/
  * ckaminpar.h:61-132: kaminpar_create / kaminpar_copy_graph /

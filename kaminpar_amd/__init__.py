@@ -102,6 +102,10 @@ def _load():
     lib.kmp_partition.restype = i64
     lib.kmp_partition.argtypes = [vp, u32, ctypes.c_double, u64, ctypes.c_int,
                                   u32, u32, ctypes.c_int, p(u32)]
+    lib.kmp_partition_deep.restype = i64
+    lib.kmp_partition_deep.argtypes = [vp, u32, ctypes.c_double, u64,
+                                       ctypes.c_int, u32, u32, u32,
+                                       ctypes.c_int, p(u32)]
 
     lib.kmp_lp_create.restype = vp
     lib.kmp_lp_create.argtypes = [vp]
@@ -276,6 +280,17 @@ class Graph:
         if rc != 0:
             raise RuntimeError("kmp_bisect_subset_ml failed")
         return side.astype(bool)
+
+    def partition_deep_native(self, k, eps=0.03, seed=1, iters=5):
+        """Progressive-k multilevel partition driven entirely from the C
+        ABI (kmp_partition_deep). Bit-identical to
+        kaminpar_amd.partition.partition_deep. Requires a GPU."""
+        part = np.zeros(self.n, dtype=np.uint32)
+        cut = _lib.kmp_partition_deep(self._h, k, eps, seed, iters, 0, 0, 0,
+                                      0, _u32p(part))
+        if cut < 0:
+            raise RuntimeError("kmp_partition_deep failed")
+        return cut, part
 
     def initial_partition_native(self, k, max_block_weight, reps=8):
         """C++ recursive-bisection initial partitioning (equivalent to

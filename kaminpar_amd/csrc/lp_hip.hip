@@ -3520,7 +3520,126 @@ i64 kaminpar_amd_compute_partition(kaminpar_amd_t *shm, u32 *partition) {
     fprintf(stderr, "kaminpar_amd: no graph set (call kaminpar_amd_copy_graph)\n");
     return -1;
   }
-  return kmp_partition(shm->g, shm->k, shm->eps, shm->seed, 5, 0, 0, 0, partition);
+  // progressive-k (deep) pipeline: best measured cuts
+  return kmp_partition_deep(shm->g, shm->k, shm->eps, shm->seed, 5, 0, 0, 0, 0,
+                            partition);
+}
+
+} // extern "C"
+
+extern "C" {
+
+i64 kmp_partition_deep(
+    const kmp_graph_t *g, u32 k, double eps, u64 seed, int iters,
+    u32 contraction_limit, u32 stop_n, u32 split_c, int ip_reps, u32 *part_out
+) {
+  if (contraction_limit == 0) {
+    contraction_limit = 2000;
+  }
+  if (stop_n == 0) {
+    stop_n = 512;
+  }
+  if (split_c == 0) {
+    split_c = 2000;
+  }
+  if (ip_reps == 0) {
+    ip_reps = 8;
+  }
+  const i64 total_w = kmp_graph_total_node_weight(g);
+  const i64 mbw_val = kmp_max_block_weight(g, k, eps);
+
+  std::vector<kmp_lp_t *> engines;
+  std::vector<u32> sizes;
+  std::vector<std::vector<u32>> mappings;
+  engines.push_back(kmp_lp_create(g));
+  if (!engines[0]) {
+    return -1;
+  }
+  sizes.push_back(kmp_graph_n(g));
+
+  i64 rc = -1;
+  std::vector<u32> clus, part;
+  std::vector<u32> group_lo(k, 0), group_w(k, 0);
+  u32 num_groups = 1;
+  group_w[0] = k;
+  std::vector<i64> caps(k, 0);
+  kmp_lp_stats_t st;
+  size_t coarsest = 0;
+
+  const u32 stop = std::max(stop_n, 2 * k);
+  while (sizes.back() > stop) {
+    const u32 cur_n = sizes.back();
+    const i64 mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit);
+    clus.resize(cur_n);
+    if (kmp_lp_cluster(engines.back(), mcw, 0, clus.data(),
+                       seed + mappings.size(), iters, &st) < 0) {
+      goto done;
+    }
+    {
+      std::vector<u32> mapping(cur_n);
+      kmp_lp_t *coarse_eng = nullptr;
+      if (kmp_contract_engine(engines.back(), clus.data(), mapping.data(),
+                              &coarse_eng) < 0) {
+        goto done;
+      }
+      const u32 c_n = kmp_lp_n(coarse_eng);
+      if (static_cast<double>(c_n) > 0.95 * static_cast<double>(cur_n)) {
+        kmp_lp_free(coarse_eng);
+        break;
+      }
+      engines.push_back(coarse_eng);
+      mappings.push_back(std::move(mapping));
+      sizes.push_back(c_n);
+    }
+  }
+
+  part.assign(sizes.back(), 0);
+  coarsest = engines.size() - 1;
+  for (size_t level = engines.size(); level-- > 0;) {
+    const u32 sc = level == coarsest ? std::min(split_c, 48u) : split_c;
+    if (num_groups < k &&
+        (static_cast<u64>(sizes[level]) >= 2ull * sc * num_groups ||
+         level == 0)) {
+      kmp_graph_t *hg_owned =
+          level == 0 ? nullptr : kmp_lp_download_graph(engines[level]);
+      const kmp_graph_t *hg = level == 0 ? g : hg_owned;
+      kmp_extend_partition(hg, part.data(), k, mbw_val, sc, ip_reps,
+                           level == 0 ? 1 : 0, group_lo.data(),
+                           group_w.data(), &num_groups);
+      if (num_groups == k) {
+        kmp_balance_partition(hg, k, mbw_val, part.data());
+      }
+      if (hg_owned) {
+        kmp_graph_free(hg_owned);
+      }
+    }
+    std::fill(caps.begin(), caps.end(), 0);
+    for (u32 i = 0; i < num_groups; ++i) {
+      caps[group_lo[i]] = static_cast<i64>(group_w[i]) * mbw_val;
+    }
+    rc = kmp_lp_refine(engines[level], k, caps.data(), part.data(), seed,
+                       iters, &st);
+    if (rc < 0) {
+      goto done;
+    }
+    if (level > 0) {
+      const std::vector<u32> &map = mappings[level - 1];
+      std::vector<u32> fine(map.size());
+      for (size_t u = 0; u < map.size(); ++u) {
+        fine[u] = part[map[u]];
+      }
+      part = std::move(fine);
+    }
+  }
+  for (u32 u = 0; u < kmp_graph_n(g); ++u) {
+    part_out[u] = part[u];
+  }
+
+done:
+  for (kmp_lp_t *e : engines) {
+    kmp_lp_free(e);
+  }
+  return rc;
 }
 
 } // extern "C"

@@ -556,6 +556,72 @@ int kmp_bisect_subset(
   return 0;
 }
 
+// Progressive-k partition extension on a host graph (C twin of
+// kaminpar_amd.partition._extend_partition -- keep in sync): split every
+// block group with >= 2 target blocks in half via FM-polished bisection,
+// repeating while every block keeps >= split_c vertices (or force).
+// groups arrays (size k): group_lo[i], group_w[i] for i < *num_groups,
+// updated in place. Returns 0.
+int kmp_extend_partition(
+    const kmp_graph_t *g, u32 *part, u32 k, i64 mbw_val, u32 split_c,
+    int reps, int force, u32 *group_lo, u32 *group_w, u32 *num_groups
+) {
+  const u32 n = kmp_graph_n(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+
+  while (true) {
+    const u32 num = *num_groups;
+    if (num >= k) {
+      break;
+    }
+    if (!force && n < 2ull * split_c * num) {
+      break;
+    }
+    std::vector<u32> nlo, nw;
+    for (u32 i = 0; i < num; ++i) {
+      const u32 b = group_lo[i], w = group_w[i];
+      if (w < 2) {
+        nlo.push_back(b);
+        nw.push_back(w);
+        continue;
+      }
+      const u32 k1 = (w + 1) / 2, k2 = w - k1;
+      std::vector<u32> nodes;
+      for (u32 u = 0; u < n; ++u) {
+        if (part[u] == b) {
+          nodes.push_back(u);
+        }
+      }
+      if (!nodes.empty()) {
+        i64 total = 0;
+        for (u32 u : nodes) {
+          total += vwgt ? vwgt[u] : 1;
+        }
+        const i64 t1 = total * k1 / w;
+        std::vector<uint8_t> side(nodes.size());
+        kmp_bisect_subset(g, nodes.data(), nodes.size(), t1,
+                          static_cast<i64>(k1) * mbw_val,
+                          static_cast<i64>(k2) * mbw_val, reps, side.data());
+        for (size_t i2 = 0; i2 < nodes.size(); ++i2) {
+          if (!side[i2]) {
+            part[nodes[i2]] = b + k1;
+          }
+        }
+      }
+      nlo.push_back(b);
+      nw.push_back(k1);
+      nlo.push_back(b + k1);
+      nw.push_back(k2);
+    }
+    *num_groups = static_cast<u32>(nlo.size());
+    for (u32 i = 0; i < *num_groups; ++i) {
+      group_lo[i] = nlo[i];
+      group_w[i] = nw[i];
+    }
+  }
+  return 0;
+}
+
 // Gain-aware overload balancer on a host graph (uniform cap), exposed for
 // the progressive-k driver.
 int kmp_balance_partition(

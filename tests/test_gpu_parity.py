@@ -592,3 +592,22 @@ def test_partition_deep_pipeline(name):
     bw = np.zeros(k, np.int64)
     np.add.at(bw, part, vw)
     assert bw.max() <= band["cap"]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", ["walshaw_k16", "rmat14_s42_k16"])
+def test_c_abi_partition_deep_matches_python(name):
+    """kmp_partition_deep (behind the ckaminpar-shaped shim) is
+    bit-identical to the Python deep pipeline's committed goldens."""
+    _require_gpu()
+    import json
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    exp = json.load(open(os.path.join(here, "golden",
+                                      "pipeline_deep_expected.json")))[name]
+    g, _e, _b = _pipeline_case(name)
+    cut, part = g.partition_deep_native(exp["k"], seed=1)
+    assert cut == exp["cut"], (cut, exp["cut"])
+    checksum = int(np.bitwise_xor.reduce(
+        np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
+    assert checksum == exp["part_checksum"]

@@ -119,3 +119,36 @@ def test_deep_pipeline_band_vs_reference_goldens():
                  "rmat16_s42_k16", "rmat18_s42_k64"):
         ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
         assert exp[name]["cut"] <= 1.5 * ref_best, (name, exp[name]["cut"], ref_best)
+
+
+def test_native_extend_partition_matches_python():
+    """kmp_extend_partition (used by the C deep driver) is bit-identical to
+    the Python _extend_partition."""
+    import ctypes
+    from kaminpar_amd import _lib, _u32p
+    from kaminpar_amd.partition import _extend_partition
+
+    _lib.kmp_extend_partition.restype = ctypes.c_int
+    _lib.kmp_extend_partition.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_uint32,
+        ctypes.c_int64, ctypes.c_uint32, ctypes.c_int, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32)]
+
+    for scale, k, split_c, force in ((12, 16, 48, 1), (12, 8, 200, 0)):
+        g = ka.Graph.rmat(scale, 8, 42)
+        mbw = g.max_block_weight(k, 0.03)
+        part_py = np.zeros(g.n, np.uint32)
+        part_py, groups_py = _extend_partition(
+            g, part_py, [(0, k)], mbw, k, split_c=split_c, reps=8,
+            force=bool(force))
+        part_c = np.zeros(g.n, np.uint32)
+        lo = np.zeros(k, np.uint32)
+        w = np.zeros(k, np.uint32)
+        w[0] = k
+        num = ctypes.c_uint32(1)
+        _lib.kmp_extend_partition(g._h, _u32p(part_c), k, mbw, split_c, 8,
+                                  force, _u32p(lo), _u32p(w),
+                                  ctypes.byref(num))
+        assert np.array_equal(part_py, part_c)
+        assert groups_py == [(int(lo[i]), int(w[i])) for i in range(num.value)]
