@@ -287,7 +287,7 @@ int kmp_bisect_subset_ml(
  * kaminpar-shm/partitioning/deep/deep_multilevel.cc). Bit-identical to the
  * Python driver kaminpar_amd.partition.partition_deep. Better cuts than
  * kmp_partition on every golden case. 0 defaults:
- * contraction_limit 2000, stop_n 512, split_c 2000, ip_reps 8. */
+ * contraction_limit 2000, stop_n 512, split_c 262144, ip_reps 8. */
 int64_t kmp_partition_deep(
     const kmp_graph_t *g,
     uint32_t k,

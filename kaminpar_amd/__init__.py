@@ -99,6 +99,9 @@ def _load():
     lib.kmp_bisect_subset_ml.restype = ctypes.c_int
     lib.kmp_bisect_subset_ml.argtypes = [vp, p(u32), u32, i64, i64, i64,
                                          ctypes.c_int, p(ctypes.c_uint8)]
+    lib.kmp_bisect_subset_fast.restype = ctypes.c_int
+    lib.kmp_bisect_subset_fast.argtypes = [vp, p(u32), u32, i64, i64, i64,
+                                           ctypes.c_int, p(ctypes.c_uint8)]
     lib.kmp_partition.restype = i64
     lib.kmp_partition.argtypes = [vp, u32, ctypes.c_double, u64, ctypes.c_int,
                                   u32, u32, ctypes.c_int, p(u32)]
@@ -267,6 +270,18 @@ class Graph:
             int(cap2), reps, side.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
         if rc != 0:
             raise RuntimeError("kmp_bisect_subset failed")
+        return side.astype(bool)
+
+    def bisect_subset_fast(self, nodes, target1, cap1, cap2, reps=8):
+        """O(m log n) bisection (lazy-PQ greedy grow + FM); for subgraphs
+        beyond a few thousand vertices."""
+        nodes = np.ascontiguousarray(nodes, dtype=np.uint32)
+        side = np.zeros(len(nodes), dtype=np.uint8)
+        rc = _lib.kmp_bisect_subset_fast(
+            self._h, _u32p(nodes), len(nodes), int(target1), int(cap1),
+            int(cap2), reps, side.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+        if rc != 0:
+            raise RuntimeError("kmp_bisect_subset_fast failed")
         return side.astype(bool)
 
     def bisect_subset_ml(self, nodes, target1, cap1, cap2, reps=8):

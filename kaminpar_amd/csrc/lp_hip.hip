@@ -3540,7 +3540,7 @@ i64 kmp_partition_deep(
     stop_n = 512;
   }
   if (split_c == 0) {
-    split_c = 2000;
+    split_c = 262144;
   }
   if (ip_reps == 0) {
     ip_reps = 8;

@@ -234,6 +234,212 @@ void fm_refine(
   }
 }
 
+// ---- O(m log n) bisection (lazy max-gain priority queues) -------------
+// Same greedy-grow + FM recipe as above, but selection uses lazy PQs with
+// (gain, smaller-id-wins) ordering instead of O(n) scans, so bisections of
+// 10^5..10^6-vertex subgraphs are affordable -- which lets the progressive-k
+// driver split blocks at FINE levels, where the cut structure still exists
+// (one clustering level destroys R-MAT cut structure).
+// Deterministic: pure sequential code, total order (gain, -id).
+
+struct LazyPQ {
+  // max-heap of (gain, vertex); ties -> smaller vertex id first
+  std::vector<std::pair<i64, i64>> h; // (gain, -id)
+  void push(i64 gain, u32 v) {
+    h.emplace_back(gain, -static_cast<i64>(v));
+    std::push_heap(h.begin(), h.end());
+  }
+  bool empty() const { return h.empty(); }
+  std::pair<i64, u32> pop() {
+    std::pop_heap(h.begin(), h.end());
+    auto t = h.back();
+    h.pop_back();
+    return {t.first, static_cast<u32>(-t.second)};
+  }
+};
+
+std::vector<uint8_t> greedy_grow_fast(
+    const SubCsr &s, const std::vector<i64> &vw, i64 target1, i64 cap1,
+    int seed_rank
+) {
+  const size_t n = vw.size();
+  std::vector<i64> deg(n);
+  for (size_t i = 0; i < n; ++i) {
+    deg[i] = s.xadj[i + 1] - s.xadj[i];
+  }
+  std::vector<u32> order(n);
+  for (size_t i = 0; i < n; ++i) {
+    order[i] = static_cast<u32>(i);
+  }
+  std::stable_sort(order.begin(), order.end(),
+                   [&](u32 a, u32 b) { return deg[a] > deg[b]; });
+  const u32 seed_idx = order[static_cast<size_t>(seed_rank) % n];
+
+  std::vector<uint8_t> side(n, 0), blocked(n, 0);
+  std::vector<i64> gain(n, -1);
+  LazyPQ pq;
+  i64 w1 = 0;
+  size_t scan_pos = 0; // for the disconnected fallback (monotone cursor)
+
+  auto add = [&](u32 i) {
+    side[i] = 1;
+    w1 += vw[i];
+    for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+      const i64 j = s.adj[e];
+      if (!side[j] && !blocked[j]) {
+        if (gain[j] < 0) {
+          gain[j] = 0;
+        }
+        gain[j] += s.w[e];
+        pq.push(gain[j], static_cast<u32>(j));
+      }
+    }
+  };
+
+  add(seed_idx);
+  while (w1 < target1) {
+    i64 v = -1;
+    while (!pq.empty()) {
+      auto [pg, pv] = pq.pop();
+      if (side[pv] || blocked[pv] || gain[pv] != pg) {
+        continue; // stale
+      }
+      v = pv;
+      break;
+    }
+    if (v < 0) {
+      // disconnected: first unassigned, unblocked vertex
+      while (scan_pos < n && (side[scan_pos] || blocked[scan_pos])) {
+        ++scan_pos;
+      }
+      if (scan_pos >= n) {
+        break;
+      }
+      v = static_cast<i64>(scan_pos);
+    }
+    if (w1 + vw[v] > cap1) {
+      blocked[v] = 1;
+      continue;
+    }
+    add(static_cast<u32>(v));
+  }
+  return side;
+}
+
+void fm_refine_fast(
+    const SubCsr &s, const std::vector<i64> &vw, std::vector<uint8_t> &side,
+    i64 cap1, i64 cap2, int max_passes = 5, int max_fruitless = 100
+) {
+  const size_t n = side.size();
+  i64 total = 0;
+  for (i64 w : vw) {
+    total += w;
+  }
+  i64 w1 = 0;
+  for (size_t i = 0; i < n; ++i) {
+    if (side[i]) {
+      w1 += vw[i];
+    }
+  }
+  std::vector<i64> gains(n);
+  std::vector<uint8_t> locked(n);
+  std::vector<u32> moves, waitA, waitB;
+  for (int pass = 0; pass < max_passes; ++pass) {
+    for (size_t i = 0; i < n; ++i) {
+      i64 cross = 0, intern = 0;
+      for (i64 e = s.xadj[i]; e < s.xadj[i + 1]; ++e) {
+        if (side[s.adj[e]] != side[i]) {
+          cross += s.w[e];
+        } else {
+          intern += s.w[e];
+        }
+      }
+      gains[i] = cross - intern;
+    }
+    std::fill(locked.begin(), locked.end(), 0);
+    moves.clear();
+    waitA.clear();
+    waitB.clear();
+    LazyPQ pq;
+    for (size_t i = 0; i < n; ++i) {
+      pq.push(gains[i], static_cast<u32>(i));
+    }
+    i64 cum = 0, best = 0;
+    size_t best_len = 0;
+    int fruitless = 0;
+    i64 wa = w1, wb = total - w1;
+    const size_t pop_budget = 64 * n + 4096; // stale/wait safety bound
+    size_t pops = 0;
+    while (fruitless < max_fruitless && pops < pop_budget) {
+      if (pq.empty()) {
+        break;
+      }
+      ++pops;
+      auto [pg, v] = pq.pop();
+      if (locked[v] || gains[v] != pg) {
+        continue;
+      }
+      const uint8_t old = side[v];
+      const bool feas = old ? (wb + vw[v] <= cap2) : (wa + vw[v] <= cap1);
+      if (!feas) {
+        (old ? waitA : waitB).push_back(v);
+        continue;
+      }
+      cum += gains[v];
+      for (i64 e = s.xadj[v]; e < s.xadj[v + 1]; ++e) {
+        const i64 j = s.adj[e];
+        if (!locked[j]) {
+          gains[j] += (side[j] == old) ? 2 * s.w[e] : -2 * s.w[e];
+          pq.push(gains[j], static_cast<u32>(j));
+        }
+      }
+      gains[v] = -gains[v];
+      side[v] = !old;
+      locked[v] = 1;
+      if (old) {
+        wa -= vw[v];
+        wb += vw[v];
+        // side B lost headroom? no: A shrank -> B->A movers gain room
+        for (u32 u : waitB) {
+          if (!locked[u]) {
+            pq.push(gains[u], u);
+          }
+        }
+        waitB.clear();
+      } else {
+        wa += vw[v];
+        wb -= vw[v];
+        for (u32 u : waitA) {
+          if (!locked[u]) {
+            pq.push(gains[u], u);
+          }
+        }
+        waitA.clear();
+      }
+      moves.push_back(v);
+      if (cum > best) {
+        best = cum;
+        best_len = moves.size();
+        fruitless = 0;
+      } else {
+        ++fruitless;
+      }
+    }
+    for (size_t m = best_len; m < moves.size(); ++m) {
+      side[moves[m]] = !side[moves[m]];
+    }
+    w1 = 0;
+    for (size_t i = 0; i < n; ++i) {
+      if (side[i]) {
+        w1 += vw[i];
+      }
+    }
+    if (best <= 0) {
+      break;
+    }
+  }
+}
+
 u64 mix64(u64 x) {
   x += 0x9E3779B97F4A7C15ull;
   x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
@@ -511,6 +717,10 @@ void balance(
 
 extern "C" {
 
+int kmp_bisect_subset_fast(
+    const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
+    i64 cap1, i64 cap2, int reps, uint8_t *side_out);
+
 // Bisect an arbitrary vertex subset of a host graph: `reps` greedy-grow
 // attempts from distinct high-degree seeds, each FM-polished, best
 // (2x directed) bisection cut kept. side_out[i] = 1 puts nodes[i] in part 1
@@ -599,9 +809,20 @@ int kmp_extend_partition(
         }
         const i64 t1 = total * k1 / w;
         std::vector<uint8_t> side(nodes.size());
-        kmp_bisect_subset(g, nodes.data(), nodes.size(), t1,
-                          static_cast<i64>(k1) * mbw_val,
-                          static_cast<i64>(k2) * mbw_val, reps, side.data());
+        // deterministic size-based dispatch (keep in sync with
+        // partition.py _extend_partition): the lazy-PQ bisector beyond
+        // 4096 vertices, with fewer repetitions on large subgraphs
+        const size_t ns = nodes.size();
+        int reps_eff = reps;
+        if (ns > 131072) {
+          reps_eff = std::min(reps, 2);
+        } else if (ns > 16384) {
+          reps_eff = std::min(reps, 4);
+        }
+        auto *bisect = ns > 4096 ? kmp_bisect_subset_fast : kmp_bisect_subset;
+        bisect(g, nodes.data(), ns, t1,
+               static_cast<i64>(k1) * mbw_val,
+               static_cast<i64>(k2) * mbw_val, reps_eff, side.data());
         for (size_t i2 = 0; i2 < nodes.size(); ++i2) {
           if (!side[i2]) {
             part[nodes[i2]] = b + k1;
@@ -635,6 +856,47 @@ int kmp_balance_partition(
   }
   balance(kmp_graph_xadj(g), kmp_graph_adjncy(g), kmp_graph_adjwgt(g), vw,
           part, n, k, cap);
+  return 0;
+}
+
+// O(m log n) bisection of a vertex subset (lazy-PQ greedy grow + FM);
+// used for subgraphs beyond a few thousand vertices where the O(n^2)
+// selection of kmp_bisect_subset would dominate. Deterministic.
+int kmp_bisect_subset_fast(
+    const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
+    i64 cap1, i64 cap2, int reps, uint8_t *side_out
+) {
+  const u32 n = kmp_graph_n(g);
+  const u32 *xadj = kmp_graph_xadj(g);
+  const u32 *adjncy = kmp_graph_adjncy(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  const i32 *adjwgt = kmp_graph_adjwgt(g);
+
+  std::vector<i64> nodes(nodes_in, nodes_in + n_sub);
+  std::vector<i64> loc(n, -1);
+  for (u32 i = 0; i < n_sub; ++i) {
+    loc[nodes[i]] = i;
+  }
+  SubCsr s = subgraph_csr(xadj, adjncy, adjwgt, nodes, loc);
+  std::vector<i64> vw(n_sub);
+  for (u32 i = 0; i < n_sub; ++i) {
+    vw[i] = vwgt ? vwgt[nodes[i]] : 1;
+  }
+
+  std::vector<uint8_t> best_side;
+  i64 best_cut = -1;
+  for (int rep = 0; rep < reps; ++rep) {
+    std::vector<uint8_t> side = greedy_grow_fast(s, vw, target1, cap1, rep);
+    fm_refine_fast(s, vw, side, cap1, cap2);
+    const i64 cut = bisection_cut2(s, side);
+    if (best_cut < 0 || cut < best_cut) {
+      best_cut = cut;
+      best_side = std::move(side);
+    }
+  }
+  for (u32 i = 0; i < n_sub; ++i) {
+    side_out[i] = best_side[i];
+  }
   return 0;
 }
 

@@ -375,8 +375,15 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
                 continue
             total = int(vw[nodes].sum())
             t1 = total * k1 // w
-            side = hg.bisect_subset(nodes, t1, k1 * mbw_val, k2 * mbw_val,
-                                    reps=reps)
+            # deterministic size-based dispatch (keep in sync with the C
+            # twin kmp_extend_partition): lazy-PQ bisector beyond 4096
+            # vertices, fewer repetitions on large subgraphs
+            ns = len(nodes)
+            reps_eff = reps if ns <= 16384 else (4 if ns <= 131072 else 2)
+            reps_eff = min(reps, reps_eff)
+            bisect = hg.bisect_subset if ns <= 4096 else hg.bisect_subset_fast
+            side = bisect(nodes, t1, k1 * mbw_val, k2 * mbw_val,
+                          reps=reps_eff)
             part[nodes[~side]] = b + k1
             new_groups += [(b, k1), (b + k1, k2)]
         groups = new_groups
@@ -384,7 +391,7 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
 
 
 def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
-                   stop_n=512, split_c=2000, reps=8, engine=None,
+                   stop_n=512, split_c=262144, reps=8, engine=None,
                    return_arcs=False):
     """Progressive-k multilevel partition: coarsen as in partition(), then
     instead of full-k initial partitioning at the coarsest level, grow k by

@@ -80,7 +80,7 @@ def oracle_partition(oracle, g, k, eps=0.03, seed=1, iters=5,
 
 
 def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
-                          contraction_limit=2000, stop_n=512, split_c=2000,
+                          contraction_limit=2000, stop_n=512, split_c=262144,
                           reps=8):
     """CPU mirror of kaminpar_amd.partition.partition_deep (keep in sync):
     progressive-k extension by FM-polished block bisections during
