@@ -3540,7 +3540,9 @@ i64 kmp_partition_deep(
     stop_n = 512;
   }
   if (split_c == 0) {
-    split_c = 262144;
+    // auto: fine-level splits up to ~2M vertices, reference-like block
+    // sizes beyond (see kaminpar_amd/partition.py partition_deep)
+    split_c = kmp_graph_n(g) <= (1u << 21) ? 262144 : 2000;
   }
   if (ip_reps == 0) {
     ip_reps = 8;

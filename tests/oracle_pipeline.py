@@ -80,13 +80,16 @@ def oracle_partition(oracle, g, k, eps=0.03, seed=1, iters=5,
 
 
 def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
-                          contraction_limit=2000, stop_n=512, split_c=262144,
+                          contraction_limit=2000, stop_n=512, split_c=None,
                           reps=8):
     """CPU mirror of kaminpar_amd.partition.partition_deep (keep in sync):
     progressive-k extension by FM-polished block bisections during
     uncoarsening."""
     from kaminpar_amd.partition import (_extend_partition, _group_caps,
                                         level_cluster_weight)
+
+    if split_c is None:
+        split_c = 262144 if g.n <= (1 << 21) else 2000
 
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
