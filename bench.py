@@ -55,9 +55,10 @@ def main():
                     default="refine",
                     help="cluster mirrors the reference's own LP benchmark "
                          "(shm_label_propagation_benchmark.cc: LP clustering); "
-                         "partition runs the full multilevel pipeline "
-                         "(BASELINE config 3: coarsen+IP+uncoarsen, "
-                         "kaminpar_amd/partition.py) -- single GPU only")
+                         "partition runs the progressive-k multilevel "
+                         "pipeline (BASELINE config 3; "
+                         "kaminpar_amd.partition.partition_deep) -- "
+                         "single GPU only")
     ap.add_argument("--order", choices=["natural", "deg-buckets"],
                     default="deg-buckets",
                     help="deg-buckets (default, matching the reference's "
