@@ -59,3 +59,17 @@ def test_degree_bucket_cut_invariance():
     assert g.edge_cut(labels_old) == gp.edge_cut(labels_new)
     # and mapping back as documented: l_old = l_new[perm]
     assert np.array_equal(labels_new[perm], labels_old)
+
+
+def test_rgg2d_generator_properties():
+    """Config-4 generator: expected density, symmetric, no self loops."""
+    g = ka.Graph.rgg2d(1 << 14, avg_deg=16.0, seed=42)
+    assert g.n == 1 << 14
+    avg = g.m / g.n
+    assert 10.0 < avg < 24.0, avg  # m counts directed arcs; avg_deg ~16
+    xadj = np.asarray(g.xadj)
+    adjncy = np.asarray(g.adjncy)
+    u = np.repeat(np.arange(g.n, dtype=np.uint64), np.diff(xadj))
+    assert not (u == adjncy).any()  # no self loops
+    fwd = set(zip(u.tolist(), adjncy.tolist()))
+    assert all((v, w) in fwd for (w, v) in list(fwd)[:2000])  # symmetric
