@@ -124,3 +124,43 @@ def test_refine_dist_gloo_world2():
     for p in procs:
         p.join(120)
         assert p.exitcode == 0
+
+
+def _torchcomm_worker(rank, world, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from kaminpar_amd.multi import TorchComm
+
+        comm = TorchComm("cpu")
+        cap = 8
+        buf, _ptr = comm.alloc_prop_buffer(cap)
+        count = rank + 1
+        for i in range(count):
+            buf[i] = rank * 100 + i
+        ptr, total = comm.all_gather_props(buf, count)
+        assert total == world * (world + 1) // 2
+        cat = comm._cat_keepalive
+        # rank order preserved, counts respected
+        row = 0
+        for r in range(world):
+            for i in range(r + 1):
+                assert int(cat[row, 0]) == r * 100 + i
+                row += 1
+        # zero-count round (all ranks) exercises the early-out
+        ptr, total = comm.all_gather_props(buf, 0)
+        assert total == 0
+    finally:
+        dist.destroy_process_group()
+
+
+def test_torchcomm_fused_count_gloo():
+    """The production TorchComm collective (single all_gather with the
+    count riding in the trailing buffer row) over gloo, world 2."""
+    import torch.multiprocessing as mp
+
+    port = 29612
+    mp.spawn(_torchcomm_worker, args=(2, port), nprocs=2, join=True)
