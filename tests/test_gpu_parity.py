@@ -611,3 +611,71 @@ def test_c_abi_partition_deep_matches_python(name):
     checksum = int(np.bitwise_xor.reduce(
         np.asarray(part, np.uint64) * np.arange(1, g.n + 1, dtype=np.uint64)))
     assert checksum == exp["part_checksum"]
+
+
+def _random_graph(rng, n, avg_deg, weighted):
+    """Random simple symmetric graph with optional weights (may include
+    isolated vertices)."""
+    m_half = max(1, int(n * avg_deg / 2))
+    u = rng.integers(0, n, m_half)
+    v = rng.integers(0, n, m_half)
+    keep = u != v
+    pairs = np.unique(
+        np.stack([np.minimum(u[keep], v[keep]),
+                  np.maximum(u[keep], v[keep])], axis=1), axis=0)
+    arcs = np.concatenate([pairs, pairs[:, ::-1]])
+    order = np.lexsort((arcs[:, 1], arcs[:, 0]))
+    arcs = arcs[order]
+    xadj = np.zeros(n + 1, np.uint32)
+    np.add.at(xadj, arcs[:, 0] + 1, 1)
+    xadj = np.cumsum(xadj).astype(np.uint32)
+    adjncy = arcs[:, 1].astype(np.uint32)
+    vwgt = adjwgt = None
+    if weighted:
+        vwgt = rng.integers(1, 20, n).astype(np.int32)
+        wmap = {}
+        w = np.zeros(len(arcs), np.int32)
+        for i, (a, b) in enumerate(map(tuple, arcs)):
+            key = (min(a, b), max(a, b))
+            if key not in wmap:
+                wmap[key] = int(rng.integers(1, 10))
+            w[i] = wmap[key]
+        adjwgt = w
+    return ka.Graph.from_csr(xadj, adjncy, vwgt=vwgt, adjwgt=adjwgt), vwgt, adjwgt
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("trial", range(10))
+def test_fuzz_refine_and_cluster_parity(oracle, trial):
+    """Seeded fuzz: random graphs (random density, optional vertex/edge
+    weights, isolated vertices), random k and caps -- GPU refine and
+    cluster must stay bit-identical to the oracle."""
+    _require_gpu()
+    from helpers import oracle_cluster, oracle_refine
+
+    rng = np.random.default_rng(1234 + trial)
+    n = int(rng.integers(50, 20_000))
+    avg_deg = float(rng.uniform(1.0, 24.0))
+    weighted = bool(rng.integers(0, 2))
+    g, vwgt, adjwgt = _random_graph(rng, n, avg_deg, weighted)
+    eng = ka.LpEngine(g)
+
+    k = int(rng.integers(2, 200))
+    eps = float(rng.uniform(0.01, 0.3))
+    mbw = np.full(k, g.max_block_weight(k, eps), np.int64)
+    part0 = ka.random_partition(g.n, k, seed=trial)
+    seed = int(rng.integers(1, 1000))
+    iters = int(rng.integers(1, 6))
+
+    cut, part, _ = eng.refine(k, mbw, part0, seed=seed, iters=iters)
+    ocut, opart, _ = oracle_refine(oracle, g, k, mbw, part0, seed=seed,
+                                   iters=iters, vwgt=vwgt, adjwgt=adjwgt)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+
+    mcw = int(rng.integers(2, max(3, n // 4)))
+    nc, clus, _ = eng.cluster(mcw, seed=seed, iters=iters)
+    onc, oclus, _ = oracle_cluster(oracle, g, mcw, seed=seed, iters=iters,
+                                   vwgt=vwgt, adjwgt=adjwgt)
+    assert nc == onc
+    assert np.array_equal(clus, oclus)
