@@ -122,6 +122,23 @@ int64_t kmp_lp_refine(
     kmp_lp_stats_t *stats
 );
 
+/* Overload-balancer mode (the role of the reference's OVERLOAD_BALANCER
+ * bracketing LP in the default refiner chain, presets.cc refinement
+ * algorithms list): same deterministic schedule and commit as
+ * kmp_lp_refine, but a vertex whose block exceeds its cap loses "stay" as
+ * a candidate, so overloaded blocks shed boundary vertices to the best
+ * admissible targets even at negative gain. Use before kmp_lp_refine when
+ * the input partition may violate the caps. Returns the edge cut, or -1. */
+int64_t kmp_lp_balance(
+    kmp_lp_t *e,
+    uint32_t k,
+    const int64_t *max_block_weights,
+    uint32_t *partition,
+    uint64_t seed,
+    int iters,
+    kmp_lp_stats_t *stats
+);
+
 /* Deterministic LP clustering (coarsening instantiation; clusters start as
  * singletons, uniform cap, isolated-node + two-hop passes). clustering: out,
  * n entries. Returns the number of non-empty clusters, or -1 on error. */

@@ -115,6 +115,8 @@ def _load():
     lib.kmp_lp_free.argtypes = [vp]
     lib.kmp_lp_refine.restype = i64
     lib.kmp_lp_refine.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
+    lib.kmp_lp_balance.restype = i64
+    lib.kmp_lp_balance.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_cluster.restype = i64
     lib.kmp_lp_cluster.argtypes = [vp, i64, u32, p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_num_chunks.restype = u32
@@ -371,6 +373,20 @@ class LpEngine:
         )
         if cut < 0:
             raise RuntimeError("kmp_lp_refine failed")
+        return cut, part, stats
+
+    def balance(self, k, max_block_weights, partition, seed=1, iters=5):
+        """Overload-balancer mode: repair an infeasible partition (blocks
+        over their caps shed boundary vertices at best admissible gain).
+        Returns (cut, partition, Stats)."""
+        part = np.ascontiguousarray(partition, dtype=np.uint32).copy()
+        mbw = np.ascontiguousarray(max_block_weights, dtype=np.int64)
+        stats = Stats()
+        cut = _lib.kmp_lp_balance(
+            self._h, k, _i64p(mbw), _u32p(part), seed, iters, ctypes.byref(stats)
+        )
+        if cut < 0:
+            raise RuntimeError("kmp_lp_balance failed")
         return cut, part, stats
 
     def cluster(self, max_cluster_weight, clustering=None, desired=0, seed=1, iters=5):

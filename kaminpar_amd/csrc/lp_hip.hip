@@ -117,10 +117,19 @@ __host__ __device__ inline u32 gain_replicas(u32 k) {
 }
 
 // Weight-acceptance predicate (refiner variant, lp_refiner.cc:185-230).
+// excl_cur implements BALANCE mode (the overload balancer's forcing rule,
+// refinement/balancer/overload_balancer.cc in spirit): a vertex whose block
+// exceeds its cap loses "stay" as a candidate, so it proposes its best
+// admissible target even at negative gain; everything else (commit
+// admission, caps, determinism) is the normal refiner machinery.
 __device__ inline bool accept_refine(
-    u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw, i64 cur_w, i64 cur_maxw
+    u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw, i64 cur_w, i64 cur_maxw,
+    bool excl_cur = false
 ) {
-  return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw)) || (c == cur);
+  if (c == cur) {
+    return !excl_cur;
+  }
+  return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw));
 }
 
 // Weight-acceptance predicate (clusterer variant, lp_clusterer.cc:199-204).
@@ -143,6 +152,7 @@ __global__ void k_phase_s(
     u32 chunk_base,
     u32 n,
     u64 iter_seed,
+    u32 balance,
     u32 max_degree,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -244,12 +254,13 @@ __global__ void k_phase_s(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
+  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
 
   BestState best{0, 0, 0, false};
   if (owner && c != kInvalid) {
     const i64 cw = weights[c];
     const i64 mw = maxw[c];
-    if (accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+    if (accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
       best = BestState{gain, tie_hash(iter_seed, u, c), c, true};
     }
   }
@@ -283,6 +294,7 @@ __global__ void k_phase_m(
     u32 chunk_base,
     u32 k,
     u64 iter_seed,
+    u32 balance,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
@@ -328,6 +340,7 @@ __global__ void k_phase_m(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
+  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
 
   BestState best{0, 0, 0, false};
   for (u32 c = lane; c < k; c += kWave) {
@@ -340,7 +353,7 @@ __global__ void k_phase_m(
     }
     const i64 cw = weights[c];
     const i64 mw = maxw[c];
-    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
       continue;
     }
     const u64 h = tie_hash(iter_seed, u, c);
@@ -460,6 +473,7 @@ __global__ void k_phase_l_acc(
 // Select per L vertex (one workgroup each, grid-stride), write the slot,
 // and reset the gains row for the next chunk.
 __global__ void k_phase_l_sel(
+    u32 balance,
     u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
@@ -485,6 +499,7 @@ __global__ void k_phase_l_sel(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
+  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
     i32 *grow = l_gains + static_cast<size_t>(vid) * k;
 
     BestState best{0, 0, 0, false};
@@ -496,7 +511,7 @@ __global__ void k_phase_l_sel(
       }
       const i64 cw = weights[c];
       const i64 mw = maxw[c];
-      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -547,6 +562,7 @@ __global__ void k_phase_l_sel(
 // vertex, whole row, replicated LDS histogram.
 template <bool kUnitWeights>
 __global__ void k_phase_l_direct(
+    u32 balance,
     u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
@@ -594,6 +610,7 @@ __global__ void k_phase_l_direct(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
+  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
 
     BestState best{0, 0, 0, false};
     for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
@@ -606,7 +623,7 @@ __global__ void k_phase_l_direct(
       }
       const i64 cw = weights[c];
       const i64 mw = maxw[c];
-      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw)) {
+      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -1924,6 +1941,7 @@ struct kmp_lp_t {
   std::vector<u32> isolated;   // host: vertices with degree 0
   std::vector<i32> iso_weights; // their node weights
   bool clusterer = false;
+  int balance = 0; // balance mode: overloaded vertices lose "stay" (select)
   i64 maxw_uniform = 0;
 
   // pinned host mirrors
@@ -2288,7 +2306,8 @@ i64 kmp_lp_phase_a(
     // S: 4 positions/wave (unit-gated; slots pre-marked invalid)
     hipLaunchKernelGGL(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
-        dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
+        dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
+        static_cast<u32>(e->balance), max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
         e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list,
         e->d_l_count
@@ -2307,6 +2326,7 @@ i64 kmp_lp_phase_a(
       auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
       hipLaunchKernelGGL(
           kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
+          static_cast<u32>(e->balance),
           e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
           e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
       );
@@ -2336,7 +2356,8 @@ i64 kmp_lp_phase_a(
         LAUNCH_CHECK();
       }
       hipLaunchKernelGGL(
-          k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed, e->k,
+          k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream,
+          static_cast<u32>(e->balance), pos_lo, chunk_base, iseed, e->k,
           e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
           e->l_cap, e->d_l_gains, e->d_slots
       );
@@ -2348,7 +2369,8 @@ i64 kmp_lp_phase_a(
             16 * sizeof(i64);
         auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
         hipLaunchKernelGGL(
-            kern, dim3(512), dim3(256), lds, e->stream, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+            kern, dim3(512), dim3(256), lds, e->stream,
+            static_cast<u32>(e->balance), pos_lo, chunk_base, iseed, e->k, e->d_xadj,
             e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
             e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
         );
@@ -2785,6 +2807,29 @@ i64 kmp_lp_refine(
     }
   }
   return kmp_lp_refine_end(e, partition, stats);
+}
+
+// Overload-balancer mode (the role of the reference's OVERLOAD_BALANCER
+// bracketing LP in the default refiner chain, presets.cc:332-338): same
+// deterministic schedule and commit as kmp_lp_refine, but a vertex whose
+// block exceeds its cap loses "stay" as a candidate, so overloaded blocks
+// shed their boundary vertices to the best admissible targets even at
+// negative gain (interior vertices follow over subsequent sweeps as the
+// boundary peels). Feasible partitions are left untouched up to normal LP
+// moves. Returns the resulting edge cut, or -1.
+i64 kmp_lp_balance(
+    kmp_lp_t *e,
+    u32 k,
+    const i64 *max_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    kmp_lp_stats_t *stats
+) {
+  e->balance = 1;
+  const i64 cut = kmp_lp_refine(e, k, max_block_weights, partition, seed, iters, stats);
+  e->balance = 0;
+  return cut;
 }
 
 i64 kmp_lp_cluster(

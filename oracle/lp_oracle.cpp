@@ -287,6 +287,7 @@ struct LpParams {
   u32 max_degree = 0xFFFFFFFFu;
   u32 desired_clusters = 0; // clusterer stop threshold (0 = never)
   bool clusterer = false;   // select semantics variant
+  bool balance = false;     // overloaded vertices lose "stay" (balancer mode)
   u32 k = 0;                // number of clusters (refiner: k; clusterer: n)
 };
 
@@ -408,10 +409,14 @@ void lp_run(
           i64 over = 0;
           if (par.clusterer) {
             accept = (cw + u_weight <= maxw) || (c == u_cluster);
+          } else if (c == u_cluster) {
+            // balance mode: a vertex in an over-cap block loses "stay"
+            accept = !(par.balance &&
+                       init_weight > par.max_weights[u_cluster]);
           } else {
             over = cw - maxw;
             const i64 init_over = init_weight - par.max_weights[u_cluster];
-            accept = (cw + u_weight <= maxw) || (over < init_over) || (c == u_cluster);
+            accept = (cw + u_weight <= maxw) || (over < init_over);
           }
           if (!accept) {
             continue;
@@ -675,6 +680,50 @@ i64 kmp_oracle_lp_refine(
   par.seed = seed;
   par.iters = iters;
   par.clusterer = false;
+  par.k = k;
+
+  std::vector<i64> weights(k, 0);
+  for (u32 u = 0; u < n; ++u) {
+    weights[partition[u]] += g.node_weight(u);
+  }
+
+  std::vector<uint8_t> active(n, 1);
+  LpStats stats;
+  lp_run(g, par, partition, weights.data(), nullptr, active, stats, nullptr);
+
+  if (stats_out) {
+    stats_out[0] = stats.arcs_scanned;
+    stats_out[1] = stats.moves;
+    stats_out[2] = 0;
+  }
+  return kmp_oracle_edge_cut(n, m, xadj, adjncy, adjwgt, partition);
+}
+
+// Balancer-mode twin of kmp_oracle_lp_refine (see kmp_lp_balance in
+// include/kaminpar_lp.h): overloaded vertices lose "stay" in selection.
+i64 kmp_oracle_lp_balance(
+    u32 n,
+    u64 m,
+    const u32 *xadj,
+    const u32 *adjncy,
+    const i32 *vwgt,
+    const i32 *adjwgt,
+    u32 k,
+    const i64 *max_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    u64 *stats_out
+) {
+  Csr g{n, m, xadj, adjncy, vwgt, adjwgt};
+
+  LpParams par;
+  par.n = n;
+  par.max_weights = max_block_weights;
+  par.seed = seed;
+  par.iters = iters;
+  par.clusterer = false;
+  par.balance = true;
   par.k = k;
 
   std::vector<i64> weights(k, 0);

@@ -679,3 +679,56 @@ def test_fuzz_refine_and_cluster_parity(oracle, trial):
                                    vwgt=vwgt, adjwgt=adjwgt)
     assert nc == onc
     assert np.array_equal(clus, oclus)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("case", ["skewed", "all_in_one", "weighted"])
+def test_balance_mode_parity_and_repair(oracle, case):
+    """Overload-balancer mode (kmp_lp_balance): bit-identical to the oracle
+    twin, and overloaded blocks shed weight toward feasibility (the role of
+    the reference's OVERLOAD_BALANCER in the default refiner chain)."""
+    _require_gpu()
+    from helpers import oracle_balance
+
+    rng = np.random.default_rng(5)
+    g = ka.Graph.rmat(13, 8, seed=9)
+    k = 16
+    vwgt = adjwgt = None
+    if case == "weighted":
+        vwgt = rng.integers(1, 8, g.n).astype(np.int32)
+        g = ka.Graph.from_csr(np.asarray(g.xadj).copy(),
+                              np.asarray(g.adjncy).copy(), vwgt=vwgt)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), np.int64)
+    if case == "all_in_one":
+        part0 = np.zeros(g.n, np.uint32)
+    else:
+        # skewed: half the vertices in block 0, rest random
+        part0 = ka.random_partition(g.n, k, seed=2)
+        part0[: g.n // 2] = 0
+
+    def overload(part):
+        vw = vwgt.astype(np.int64) if vwgt is not None else np.ones(g.n, np.int64)
+        bw = np.zeros(k, np.int64)
+        np.add.at(bw, part, vw)
+        return int(np.maximum(bw - mbw, 0).sum())
+
+    over0 = overload(part0)
+    assert over0 > 0  # the input really is infeasible
+
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.balance(k, mbw, part0, seed=1, iters=5)
+    ocut, opart, _ = oracle_balance(oracle, g, k, mbw, part0, seed=1, iters=5,
+                                    vwgt=vwgt)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+
+    over1 = overload(part)
+    assert over1 < over0  # substantial shedding
+    assert over1 <= over0 // 4, (over0, over1)
+
+    # a feasible partition passes through essentially as normal LP refinement
+    featble = ka.random_partition(g.n, k, seed=7)
+    cut_b, part_b, _ = eng.balance(k, mbw, featble, seed=3, iters=2)
+    ocut_b, opart_b, _ = oracle_balance(oracle, g, k, mbw, featble, seed=3,
+                                        iters=2, vwgt=vwgt)
+    assert cut_b == ocut_b and np.array_equal(part_b, opart_b)
