@@ -124,10 +124,15 @@ __host__ __device__ inline u32 gain_replicas(u32 k) {
 // admission, caps, determinism) is the normal refiner machinery.
 __device__ inline bool accept_refine(
     u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw, i64 cur_w, i64 cur_maxw,
-    bool excl_cur = false
+    bool balance = false
 ) {
   if (c == cur) {
-    return !excl_cur;
+    return !(balance && cur_w > cur_maxw);
+  }
+  if (balance) {
+    // room-only in balance mode: the relative-overload clause would let the
+    // deterministic select livelock on an already-full favourite target
+    return cw + u_w <= maxw;
   }
   return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw));
 }
@@ -153,6 +158,7 @@ __global__ void k_phase_s(
     u32 n,
     u64 iter_seed,
     u32 balance,
+    u32 fallback,
     u32 max_degree,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -254,7 +260,7 @@ __global__ void k_phase_s(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
+  const bool excl_cur = balance != 0;
 
   BestState best{0, 0, 0, false};
   if (owner && c != kInvalid) {
@@ -276,8 +282,17 @@ __global__ void k_phase_s(
     }
   }
 
-  if (slot == 0 && best.have && best.c != cur) {
-    slots[sidx] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  if (slot == 0) {
+    BestState fin = best;
+    if (!fin.have && excl_cur && cur_w > cur_maxw && fallback != kInvalid &&
+        fallback != cur) {
+      // balance fallback: no admissible adjacent target (e.g. everything in
+      // one block) -- propose the lightest block with room at gain 0
+      fin = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
+    }
+    if (fin.have && fin.c != cur) {
+      slots[sidx] = Prop{u, fin.c, p - chunk_base, static_cast<u32>(u_w)};
+    }
   }
 }
 
@@ -295,6 +310,7 @@ __global__ void k_phase_m(
     u32 k,
     u64 iter_seed,
     u32 balance,
+    u32 fallback,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
@@ -340,7 +356,7 @@ __global__ void k_phase_m(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
+  const bool excl_cur = balance != 0;
 
   BestState best{0, 0, 0, false};
   for (u32 c = lane; c < k; c += kWave) {
@@ -372,8 +388,15 @@ __global__ void k_phase_m(
     }
   }
 
-  if (lane == 0 && best.have && best.c != cur) {
-    slots[p - pos_lo] = Prop{u, best.c, p - chunk_base, static_cast<u32>(u_w)};
+  if (lane == 0) {
+    BestState fin = best;
+    if (!fin.have && excl_cur && cur_w > cur_maxw && fallback != kInvalid &&
+        fallback != cur) {
+      fin = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
+    }
+    if (fin.have && fin.c != cur) {
+      slots[p - pos_lo] = Prop{u, fin.c, p - chunk_base, static_cast<u32>(u_w)};
+    }
   }
   __threadfence_block(); // gains reuse across grid-stride iterations
   }
@@ -474,6 +497,7 @@ __global__ void k_phase_l_acc(
 // and reset the gains row for the next chunk.
 __global__ void k_phase_l_sel(
     u32 balance,
+    u32 fallback,
     u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
@@ -499,7 +523,7 @@ __global__ void k_phase_l_sel(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
+  const bool excl_cur = balance != 0;
     i32 *grow = l_gains + static_cast<size_t>(vid) * k;
 
     BestState best{0, 0, 0, false};
@@ -550,6 +574,10 @@ __global__ void k_phase_l_sel(
           }
         }
       }
+      if (!total.have && excl_cur && cur_w > cur_maxw &&
+          fallback != kInvalid && fallback != cur) {
+        total = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
+      }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
       }
@@ -563,6 +591,7 @@ __global__ void k_phase_l_sel(
 template <bool kUnitWeights>
 __global__ void k_phase_l_direct(
     u32 balance,
+    u32 fallback,
     u32 pos_lo,
     u32 chunk_base,
     u64 iter_seed,
@@ -610,7 +639,7 @@ __global__ void k_phase_l_direct(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0 && cur_w > cur_maxw;
+  const bool excl_cur = balance != 0;
 
     BestState best{0, 0, 0, false};
     for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
@@ -661,6 +690,10 @@ __global__ void k_phase_l_direct(
             total = BestState{g, h, c, true};
           }
         }
+      }
+      if (!total.have && excl_cur && cur_w > cur_maxw &&
+          fallback != kInvalid && fallback != cur) {
+        total = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
       }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
@@ -1942,6 +1975,7 @@ struct kmp_lp_t {
   std::vector<i32> iso_weights; // their node weights
   bool clusterer = false;
   int balance = 0; // balance mode: overloaded vertices lose "stay" (select)
+  std::vector<i64> maxw_host; // refine caps (host copy, for the fallback)
   i64 maxw_uniform = 0;
 
   // pinned host mirrors
@@ -2208,6 +2242,7 @@ int kmp_lp_refine_begin(
   e->k = k;
   e->seed = seed;
   e->clusterer = false;
+  e->maxw_host.assign(max_block_weights, max_block_weights + k);
   e->phase_a_ms = 0.0;
   e->commit_ms = 0.0;
   e->ev_used = 0;
@@ -2303,11 +2338,27 @@ i64 kmp_lp_phase_a(
   HIP_CHECK(hipEventRecord(ev0, e->stream));
 
   if (!e->clusterer) {
+    // balance mode: per-chunk fallback target = lightest block with room
+    // (for overloaded vertices with no admissible adjacent candidate)
+    u32 fallback = kInvalid;
+    if (e->balance) {
+      std::vector<i64> w(e->k);
+      HIP_CHECK(hipMemcpyAsync(w.data(), e->d_weights, sizeof(i64) * e->k,
+                               hipMemcpyDeviceToHost, e->stream));
+      sync_spin(e);
+      i64 bw = -1;
+      for (u32 c = 0; c < e->k; ++c) {
+        if (w[c] < e->maxw_host[c] && (bw < 0 || w[c] < bw)) {
+          bw = w[c];
+          fallback = c;
+        }
+      }
+    }
     // S: 4 positions/wave (unit-gated; slots pre-marked invalid)
     hipLaunchKernelGGL(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
-        static_cast<u32>(e->balance), max_degree,
+        static_cast<u32>(e->balance), fallback, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
         e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list,
         e->d_l_count
@@ -2326,7 +2377,7 @@ i64 kmp_lp_phase_a(
       auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
       hipLaunchKernelGGL(
           kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
-          static_cast<u32>(e->balance),
+          static_cast<u32>(e->balance), fallback,
           e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
           e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
       );
@@ -2357,7 +2408,7 @@ i64 kmp_lp_phase_a(
       }
       hipLaunchKernelGGL(
           k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream,
-          static_cast<u32>(e->balance), pos_lo, chunk_base, iseed, e->k,
+          static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k,
           e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
           e->l_cap, e->d_l_gains, e->d_slots
       );
@@ -2370,7 +2421,7 @@ i64 kmp_lp_phase_a(
         auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
         hipLaunchKernelGGL(
             kern, dim3(512), dim3(256), lds, e->stream,
-            static_cast<u32>(e->balance), pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+            static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
             e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
             e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
         );
@@ -2826,6 +2877,40 @@ i64 kmp_lp_balance(
     int iters,
     kmp_lp_stats_t *stats
 ) {
+  // Isolated vertices have no LP candidates and can never move through the
+  // sweeps: assign those sitting in over-cap blocks to the lightest block
+  // with room first (deterministic; zero cut impact -- they have no edges).
+  if (!e->isolated.empty()) {
+    if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
+      return -1;
+    }
+    std::vector<i64> w(k);
+    HIP_CHECK(hipMemcpyAsync(w.data(), e->d_weights, sizeof(i64) * k,
+                             hipMemcpyDeviceToHost, e->stream));
+    sync_spin(e);
+    for (size_t i = 0; i < e->isolated.size(); ++i) {
+      const u32 u = e->isolated[i];
+      const i32 uw = e->iso_weights[i];
+      const u32 b = partition[u];
+      if (w[b] <= max_block_weights[b]) {
+        continue;
+      }
+      i64 best = -1;
+      u32 t = b;
+      for (u32 c = 0; c < k; ++c) {
+        if (c != b && w[c] + uw <= max_block_weights[c] &&
+            (best < 0 || w[c] < best)) {
+          best = w[c];
+          t = c;
+        }
+      }
+      if (t != b) {
+        w[b] -= uw;
+        w[t] += uw;
+        partition[u] = t;
+      }
+    }
+  }
   e->balance = 1;
   const i64 cut = kmp_lp_refine(e, k, max_block_weights, partition, seed, iters, stats);
   e->balance = 0;

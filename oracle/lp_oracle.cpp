@@ -336,6 +336,19 @@ void lp_run(
       processed.clear();
       slots.assign(pos_end - pos_begin, Proposal{0, 0, 0, 0xFFFFFFFFu, 0});
 
+      // balance mode: per-chunk fallback target = lightest block with room
+      // (keep in sync with kmp_lp_phase_a on the device)
+      u32 balance_fallback = 0xFFFFFFFFu;
+      if (par.balance) {
+        i64 bw = -1;
+        for (u32 c = 0; c < par.k; ++c) {
+          if (weights[c] < par.max_weights[c] && (bw < 0 || weights[c] < bw)) {
+            bw = weights[c];
+            balance_fallback = c;
+          }
+        }
+      }
+
       // ---- phase A: gains + selection against the chunk-start snapshot ----
       // (parallel over positions; results land in per-position slots so the
       // compacted order is deterministic, mirroring the GPU design)
@@ -413,6 +426,9 @@ void lp_run(
             // balance mode: a vertex in an over-cap block loses "stay"
             accept = !(par.balance &&
                        init_weight > par.max_weights[u_cluster]);
+          } else if (par.balance) {
+            // room-only in balance mode (matches accept_refine on device)
+            accept = cw + u_weight <= maxw;
           } else {
             over = cw - maxw;
             const i64 init_over = init_weight - par.max_weights[u_cluster];
@@ -452,6 +468,13 @@ void lp_run(
           favored[u] = fav;
         }
 
+        if (!have && par.balance &&
+            init_weight > par.max_weights[u_cluster] &&
+            balance_fallback != 0xFFFFFFFFu && balance_fallback != u_cluster) {
+          // no admissible adjacent target: shed to the chunk's fallback
+          best = balance_fallback;
+          have = true;
+        }
         if (have && best != u_cluster) {
           slots[p - pos_begin] = Proposal{p - pos_begin, u, u_cluster, best, u_weight};
         }
@@ -729,6 +752,33 @@ i64 kmp_oracle_lp_balance(
   std::vector<i64> weights(k, 0);
   for (u32 u = 0; u < n; ++u) {
     weights[partition[u]] += g.node_weight(u);
+  }
+
+  // isolated pre-pass (keep in sync with kmp_lp_balance): deg-0 vertices in
+  // over-cap blocks go to the lightest block with room, ascending id order
+  for (u32 u = 0; u < n; ++u) {
+    if (g.degree(u) != 0) {
+      continue;
+    }
+    const u32 b = partition[u];
+    if (weights[b] <= max_block_weights[b]) {
+      continue;
+    }
+    const i32 uw = g.node_weight(u);
+    i64 best = -1;
+    u32 t = b;
+    for (u32 c = 0; c < k; ++c) {
+      if (c != b && weights[c] + uw <= max_block_weights[c] &&
+          (best < 0 || weights[c] < best)) {
+        best = weights[c];
+        t = c;
+      }
+    }
+    if (t != b) {
+      weights[b] -= uw;
+      weights[t] += uw;
+      partition[u] = t;
+    }
   }
 
   std::vector<uint8_t> active(n, 1);
