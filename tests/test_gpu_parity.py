@@ -731,3 +731,41 @@ def test_balance_mode_parity_and_repair(oracle, case):
     ocut_b, opart_b, _ = oracle_balance(oracle, g, k, mbw, featble, seed=3,
                                         iters=2, vwgt=vwgt)
     assert cut_b == ocut_b and np.array_equal(part_b, opart_b)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("trial", range(5))
+def test_fuzz_balance_parity(oracle, trial):
+    """Seeded fuzz for balance mode: random graphs, random infeasible
+    partitions, non-uniform per-block caps -- GPU bit-identical to the
+    oracle twin, overload strictly reduced."""
+    _require_gpu()
+    from helpers import oracle_balance
+
+    rng = np.random.default_rng(777 + trial)
+    n = int(rng.integers(100, 12_000))
+    g, vwgt, adjwgt = _random_graph(rng, n, float(rng.uniform(2, 16)),
+                                    bool(rng.integers(0, 2)))
+    k = int(rng.integers(2, 64))
+    vw = vwgt.astype(np.int64) if vwgt is not None else np.ones(g.n, np.int64)
+    total = int(vw.sum())
+    # non-uniform caps with total headroom ~15%
+    caps = rng.uniform(0.8, 1.6, k)
+    caps = (caps / caps.sum() * total * 1.15).astype(np.int64) + 1
+    # skewed infeasible partition
+    part0 = ka.random_partition(g.n, k, seed=trial)
+    part0[: g.n // 3] = int(rng.integers(0, k))
+
+    def overload(part):
+        bw = np.zeros(k, np.int64)
+        np.add.at(bw, part, vw)
+        return int(np.maximum(bw - caps, 0).sum())
+
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.balance(k, caps, part0, seed=trial + 1, iters=5)
+    ocut, opart, _ = oracle_balance(oracle, g, k, caps, part0, seed=trial + 1,
+                                    iters=5, vwgt=vwgt, adjwgt=adjwgt)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+    if overload(part0) > 0:
+        assert overload(part) < overload(part0)
