@@ -60,6 +60,9 @@ kmp_graph_t *kmp_gen_rgg2d(uint32_t n, double avg_deg, uint64_t seed);
 /* METIS ASCII reader (kaminpar-io/metis_parser.h:17-25 format). */
 kmp_graph_t *kmp_read_metis(const char *path);
 
+/* METIS ASCII writer (counterpart of kmp_read_metis). */
+int kmp_write_metis(const kmp_graph_t *g, const char *path);
+
 /* ParHIP binary reader/writer (docs/graph_file_format.md "ParHIP Graph
  * File Format"; kaminpar-io/parhip_parser.cc:42-136): 24-byte header
  * (version bit-field, n, m), byte offsets, adjacency, optional weights.

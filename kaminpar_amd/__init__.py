@@ -64,6 +64,8 @@ def _load():
     lib.kmp_gen_rgg2d.argtypes = [u32, ctypes.c_double, u64]
     lib.kmp_read_metis.restype = vp
     lib.kmp_read_metis.argtypes = [ctypes.c_char_p]
+    lib.kmp_write_metis.restype = ctypes.c_int
+    lib.kmp_write_metis.argtypes = [vp, ctypes.c_char_p]
     lib.kmp_read_parhip.restype = vp
     lib.kmp_read_parhip.argtypes = [ctypes.c_char_p]
     lib.kmp_write_parhip.restype = ctypes.c_int
@@ -211,6 +213,10 @@ class Graph:
     @classmethod
     def read_parhip(cls, path):
         return cls(_lib.kmp_read_parhip(os.fsencode(path)))
+
+    def write_metis(self, path):
+        if _lib.kmp_write_metis(self._h, os.fsencode(path)) != 0:
+            raise IOError(f"cannot write {path}")
 
     def write_parhip(self, path):
         if _lib.kmp_write_parhip(self._h, os.fsencode(path)) != 0:

@@ -454,6 +454,40 @@ i64 kmp_max_block_weight(const kmp_graph_t *g, u32 k, double eps) {
   return static_cast<i64>((1.0 + eps) * pbw);
 }
 
+// METIS ASCII writer (kaminpar-io metis format; see the reader below and
+// docs/graph_file_format.md "METIS Graph File Format").
+int kmp_write_metis(const kmp_graph_t *g, const char *path) {
+  FILE *f = std::fopen(path, "w");
+  if (!f) {
+    return -1;
+  }
+  const bool has_vwgt = !g->vwgt.empty();
+  const bool has_ewgt = !g->adjwgt.empty();
+  std::fprintf(f, "%u %llu", g->n,
+               static_cast<unsigned long long>(g->m / 2));
+  if (has_vwgt || has_ewgt) {
+    std::fprintf(f, " %d%d", has_vwgt ? 1 : 0, has_ewgt ? 1 : 0);
+  }
+  std::fputc('\n', f);
+  for (u32 u = 0; u < g->n; ++u) {
+    bool first = true;
+    if (has_vwgt) {
+      std::fprintf(f, "%d", g->vwgt[u]);
+      first = false;
+    }
+    for (u32 e = g->xadj[u]; e < g->xadj[u + 1]; ++e) {
+      std::fprintf(f, first ? "%u" : " %u", g->adjncy[e] + 1);
+      first = false;
+      if (has_ewgt) {
+        std::fprintf(f, " %d", g->adjwgt[e]);
+      }
+    }
+    std::fputc('\n', f);
+  }
+  std::fclose(f);
+  return 0;
+}
+
 // ParHIP binary format (docs/graph_file_format.md "ParHIP Graph File
 // Format"; kaminpar-io/parhip_parser.cc:42-136): 24-byte header (version
 // bit-field, n, m as u64), then BYTE offsets ((n+1) x EdgeID width,

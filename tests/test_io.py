@@ -87,3 +87,26 @@ def test_c_shim_fails_loudly_without_gpu():
     assert cut == -1
     _lib.kaminpar_amd_free.argtypes = [ctypes.c_void_p]
     _lib.kaminpar_amd_free(shm)
+
+
+def test_metis_roundtrip(tmp_path):
+    g = ka.Graph.rmat(10, 8, 3)
+    p = str(tmp_path / "g.metis")
+    g.write_metis(p)
+    h = ka.Graph.read_metis(p)
+    assert h.n == g.n and h.m == g.m
+    assert np.array_equal(np.asarray(h.xadj), np.asarray(g.xadj))
+    assert np.array_equal(np.asarray(h.adjncy), np.asarray(g.adjncy))
+
+
+def test_metis_roundtrip_weighted(tmp_path):
+    rng = np.random.default_rng(4)
+    g0 = ka.Graph.rmat(9, 8, 5)
+    vw = rng.integers(1, 9, g0.n).astype(np.int32)
+    g = ka.Graph.from_csr(np.asarray(g0.xadj), np.asarray(g0.adjncy), vwgt=vw)
+    p = str(tmp_path / "gw.metis")
+    g.write_metis(p)
+    h = ka.Graph.read_metis(p)
+    assert h.total_node_weight == int(vw.sum())
+    labels = ka.random_partition(g.n, 3, seed=1)
+    assert g.edge_cut(labels) == h.edge_cut(labels)
