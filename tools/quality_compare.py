@@ -27,10 +27,14 @@ def main():
     scale = int(sys.argv[1]) if len(sys.argv) > 1 else 23
     k = int(sys.argv[2]) if len(sys.argv) > 2 else 16
     out_path = sys.argv[3] if len(sys.argv) > 3 else None
+    kind = sys.argv[4] if len(sys.argv) > 4 else "rmat"
 
-    g = ka.Graph.rmat(scale, 8, 42)
-    print(f"rmat{scale}: n={g.n} m={g.m} k={k}", flush=True)
-    out = {"graph": f"rmat{scale}_s42", "n": g.n, "m": g.m, "k": k,
+    if kind == "rgg2d":
+        g = ka.Graph.rgg2d(1 << scale, avg_deg=16.0, seed=42)
+    else:
+        g = ka.Graph.rmat(scale, 8, 42)
+    print(f"{kind}{scale}: n={g.n} m={g.m} k={k}", flush=True)
+    out = {"graph": f"{kind}{scale}_s42", "n": g.n, "m": g.m, "k": k,
            "eps": 0.03, "cap": int(g.max_block_weight(k, 0.03))}
 
     lib_path = os.path.join(REPO, "oracle", "_ref", "libkaminpar_ref_full.so")
