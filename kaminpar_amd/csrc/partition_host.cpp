@@ -720,6 +720,9 @@ extern "C" {
 int kmp_bisect_subset_fast(
     const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
     i64 cap1, i64 cap2, int reps, uint8_t *side_out);
+int kmp_bisect_subset_ml(
+    const kmp_graph_t *g, const u32 *nodes_in, u32 n_sub, i64 target1,
+    i64 cap1, i64 cap2, int reps, uint8_t *side_out);
 
 // Bisect an arbitrary vertex subset of a host graph: `reps` greedy-grow
 // attempts from distinct high-degree seeds, each FM-polished, best
@@ -809,9 +812,14 @@ int kmp_extend_partition(
         }
         const i64 t1 = total * k1 / w;
         std::vector<uint8_t> side(nodes.size());
-        // deterministic size-based dispatch (keep in sync with
-        // partition.py _extend_partition): the lazy-PQ bisector beyond
-        // 4096 vertices, with fewer repetitions on large subgraphs
+        // deterministic dispatch (keep in sync with partition.py
+        // _extend_partition): <=4096 vertices use the pinned O(n^2)
+        // bisector; larger subgraphs use the lazy-PQ flat bisector for
+        // heavy-tailed degree distributions (flat FM works, HEM collapses
+        // hubs) and the HEM multilevel bisector for low-variance degrees
+        // (geometric/mesh-like, where flat FM gets lost at scale) --
+        // split at squared coefficient of variation >= 1, i.e.
+        // n * sum(d^2) >= 2 * sum(d)^2.
         const size_t ns = nodes.size();
         int reps_eff = reps;
         if (ns > 131072) {
@@ -819,7 +827,19 @@ int kmp_extend_partition(
         } else if (ns > 16384) {
           reps_eff = std::min(reps, 4);
         }
-        auto *bisect = ns > 4096 ? kmp_bisect_subset_fast : kmp_bisect_subset;
+        auto *bisect = kmp_bisect_subset;
+        if (ns > 4096) {
+          const u32 *xadj = kmp_graph_xadj(g);
+          unsigned __int128 sum = 0, sq = 0;
+          for (u32 u : nodes) {
+            const u64 d = xadj[u + 1] - xadj[u];
+            sum += d;
+            sq += d * d;
+          }
+          const bool heavy_tail =
+              static_cast<unsigned __int128>(ns) * sq >= 2 * sum * sum;
+          bisect = heavy_tail ? kmp_bisect_subset_fast : kmp_bisect_subset_ml;
+        }
         bisect(g, nodes.data(), ns, t1,
                static_cast<i64>(k1) * mbw_val,
                static_cast<i64>(k2) * mbw_val, reps_eff, side.data());

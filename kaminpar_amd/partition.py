@@ -375,13 +375,25 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
                 continue
             total = int(vw[nodes].sum())
             t1 = total * k1 // w
-            # deterministic size-based dispatch (keep in sync with the C
-            # twin kmp_extend_partition): lazy-PQ bisector beyond 4096
-            # vertices, fewer repetitions on large subgraphs
+            # deterministic dispatch (keep in sync with the C twin
+            # kmp_extend_partition): pinned O(n^2) bisector <= 4096
+            # vertices; above that, flat lazy-PQ FM for heavy-tailed
+            # degrees (CV^2 >= 1), HEM multilevel for low-variance
+            # (geometric/mesh-like) subgraphs where flat FM gets lost
             ns = len(nodes)
             reps_eff = reps if ns <= 16384 else (4 if ns <= 131072 else 2)
             reps_eff = min(reps, reps_eff)
-            bisect = hg.bisect_subset if ns <= 4096 else hg.bisect_subset_fast
+            if ns <= 4096:
+                bisect = hg.bisect_subset
+            else:
+                xadj = np.asarray(hg.xadj)
+                d = (xadj[nodes.astype(np.int64) + 1]
+                     - xadj[nodes.astype(np.int64)]).astype(object)
+                s = int(np.sum(d))
+                sq = int(np.sum(d * d))
+                heavy_tail = ns * sq >= 2 * s * s
+                bisect = (hg.bisect_subset_fast if heavy_tail
+                          else hg.bisect_subset_ml)
             side = bisect(nodes, t1, k1 * mbw_val, k2 * mbw_val,
                           reps=reps_eff)
             part[nodes[~side]] = b + k1
