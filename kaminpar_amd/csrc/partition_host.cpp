@@ -15,6 +15,7 @@
 #include <algorithm>
 #include <cstdint>
 #include <cstdio>
+#include <omp.h>
 #include <vector>
 
 using u32 = uint32_t;
@@ -618,17 +619,21 @@ void rec(IpCtx &c, std::vector<i64> &nodes, u32 k_lo, u32 k_hi) {
     vw[i] = c.vwgt[nodes[i]];
   }
 
-  std::vector<uint8_t> best_side;
-  i64 best_cut = -1;
+  std::vector<std::vector<uint8_t>> sides(c.reps);
+  std::vector<i64> cuts(c.reps);
+#pragma omp parallel for schedule(dynamic, 1) if (!omp_in_parallel())
   for (int rep = 0; rep < c.reps; ++rep) {
-    std::vector<uint8_t> side = greedy_grow(s, vw, target1, cap1, rep);
-    fm_refine(s, vw, side, cap1, cap2);
-    const i64 cut = bisection_cut2(s, side);
-    if (best_cut < 0 || cut < best_cut) {
-      best_cut = cut;
-      best_side = std::move(side);
+    sides[rep] = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, sides[rep], cap1, cap2);
+    cuts[rep] = bisection_cut2(s, sides[rep]);
+  }
+  int bi = 0;
+  for (int rep = 1; rep < c.reps; ++rep) {
+    if (cuts[rep] < cuts[bi]) {
+      bi = rep;
     }
   }
+  const std::vector<uint8_t> &best_side = sides[bi];
 
   std::vector<i64> p1, p2;
   for (size_t i = 0; i < nodes.size(); ++i) {
@@ -752,19 +757,25 @@ int kmp_bisect_subset(
     vw[i] = vwgt ? vwgt[nodes[i]] : 1;
   }
 
-  std::vector<uint8_t> best_side;
-  i64 best_cut = -1;
+  // reps run in parallel when not already inside the per-group parallel
+  // loop; first-minimal-rep selection matches the sequential
+  // keep-if-strictly-better semantics bit-exactly.
+  std::vector<std::vector<uint8_t>> sides(reps);
+  std::vector<i64> cuts(reps);
+#pragma omp parallel for schedule(dynamic, 1) if (!omp_in_parallel())
   for (int rep = 0; rep < reps; ++rep) {
-    std::vector<uint8_t> side = greedy_grow(s, vw, target1, cap1, rep);
-    fm_refine(s, vw, side, cap1, cap2);
-    const i64 cut = bisection_cut2(s, side);
-    if (best_cut < 0 || cut < best_cut) {
-      best_cut = cut;
-      best_side = std::move(side);
+    sides[rep] = greedy_grow(s, vw, target1, cap1, rep);
+    fm_refine(s, vw, sides[rep], cap1, cap2);
+    cuts[rep] = bisection_cut2(s, sides[rep]);
+  }
+  int best = 0;
+  for (int rep = 1; rep < reps; ++rep) {
+    if (cuts[rep] < cuts[best]) {
+      best = rep;
     }
   }
   for (u32 i = 0; i < n_sub; ++i) {
-    side_out[i] = best_side[i];
+    side_out[i] = sides[best][i];
   }
   return 0;
 }
@@ -790,21 +801,41 @@ int kmp_extend_partition(
     if (!force && n < 2ull * split_c * num) {
       break;
     }
+    // Bin vertices by group in ONE pass (part values are exactly the
+    // group_lo ids during extension), replacing the O(n * num) per-group
+    // scans; ascending-u order within each bin matches the scan order.
+    std::vector<u32> gid_of_block(k, 0);
+    for (u32 i = 0; i < num; ++i) {
+      gid_of_block[group_lo[i]] = i;
+    }
+    std::vector<u32> off(num + 1, 0);
+    for (u32 u = 0; u < n; ++u) {
+      off[gid_of_block[part[u]] + 1]++;
+    }
+    for (u32 i = 0; i < num; ++i) {
+      off[i + 1] += off[i];
+    }
+    std::vector<u32> binned(n);
+    {
+      std::vector<u32> cur(off.begin(), off.end() - 1);
+      for (u32 u = 0; u < n; ++u) {
+        binned[cur[gid_of_block[part[u]]]++] = u;
+      }
+    }
     std::vector<u32> nlo, nw;
+    // Groups are independent subproblems (disjoint part[] writes, the
+    // extracted subgraph keeps intra-subset arcs only), so the bisections
+    // run in parallel over host cores; results are bit-identical to the
+    // serial loop.
+#pragma omp parallel for schedule(dynamic, 1)
     for (u32 i = 0; i < num; ++i) {
       const u32 b = group_lo[i], w = group_w[i];
       if (w < 2) {
-        nlo.push_back(b);
-        nw.push_back(w);
         continue;
       }
       const u32 k1 = (w + 1) / 2, k2 = w - k1;
-      std::vector<u32> nodes;
-      for (u32 u = 0; u < n; ++u) {
-        if (part[u] == b) {
-          nodes.push_back(u);
-        }
-      }
+      std::vector<u32> nodes(binned.begin() + off[i],
+                             binned.begin() + off[i + 1]);
       if (!nodes.empty()) {
         i64 total = 0;
         for (u32 u : nodes) {
@@ -849,10 +880,19 @@ int kmp_extend_partition(
           }
         }
       }
-      nlo.push_back(b);
-      nw.push_back(k1);
-      nlo.push_back(b + k1);
-      nw.push_back(k2);
+    }
+    for (u32 i = 0; i < num; ++i) {
+      const u32 b = group_lo[i], w = group_w[i];
+      if (w < 2) {
+        nlo.push_back(b);
+        nw.push_back(w);
+      } else {
+        const u32 k1 = (w + 1) / 2;
+        nlo.push_back(b);
+        nw.push_back(k1);
+        nlo.push_back(b + k1);
+        nw.push_back(w - k1);
+      }
     }
     *num_groups = static_cast<u32>(nlo.size());
     for (u32 i = 0; i < *num_groups; ++i) {
@@ -903,19 +943,22 @@ int kmp_bisect_subset_fast(
     vw[i] = vwgt ? vwgt[nodes[i]] : 1;
   }
 
-  std::vector<uint8_t> best_side;
-  i64 best_cut = -1;
+  std::vector<std::vector<uint8_t>> sides(reps);
+  std::vector<i64> cuts(reps);
+#pragma omp parallel for schedule(dynamic, 1) if (!omp_in_parallel())
   for (int rep = 0; rep < reps; ++rep) {
-    std::vector<uint8_t> side = greedy_grow_fast(s, vw, target1, cap1, rep);
-    fm_refine_fast(s, vw, side, cap1, cap2);
-    const i64 cut = bisection_cut2(s, side);
-    if (best_cut < 0 || cut < best_cut) {
-      best_cut = cut;
-      best_side = std::move(side);
+    sides[rep] = greedy_grow_fast(s, vw, target1, cap1, rep);
+    fm_refine_fast(s, vw, sides[rep], cap1, cap2);
+    cuts[rep] = bisection_cut2(s, sides[rep]);
+  }
+  int best = 0;
+  for (int rep = 1; rep < reps; ++rep) {
+    if (cuts[rep] < cuts[best]) {
+      best = rep;
     }
   }
   for (u32 i = 0; i < n_sub; ++i) {
-    side_out[i] = best_side[i];
+    side_out[i] = sides[best][i];
   }
   return 0;
 }
@@ -945,19 +988,21 @@ int kmp_bisect_subset_ml(
     vw[i] = vwgt ? vwgt[nodes[i]] : 1;
   }
 
-  std::vector<uint8_t> best_side;
-  i64 best_cut = -1;
+  std::vector<std::vector<uint8_t>> sides(reps);
+  std::vector<i64> cuts(reps);
+#pragma omp parallel for schedule(dynamic, 1) if (!omp_in_parallel())
   for (int rep = 0; rep < reps; ++rep) {
-    std::vector<uint8_t> side;
-    ml_bisect_rec(s, vw, target1, cap1, cap2, rep, side);
-    const i64 cut = bisection_cut2(s, side);
-    if (best_cut < 0 || cut < best_cut) {
-      best_cut = cut;
-      best_side = std::move(side);
+    ml_bisect_rec(s, vw, target1, cap1, cap2, rep, sides[rep]);
+    cuts[rep] = bisection_cut2(s, sides[rep]);
+  }
+  int best = 0;
+  for (int rep = 1; rep < reps; ++rep) {
+    if (cuts[rep] < cuts[best]) {
+      best = rep;
     }
   }
   for (u32 i = 0; i < n_sub; ++i) {
-    side_out[i] = best_side[i];
+    side_out[i] = sides[best][i];
   }
   return 0;
 }
