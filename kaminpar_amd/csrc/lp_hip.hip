@@ -3677,6 +3677,24 @@ i64 kmp_partition_deep(
   if (ip_reps == 0) {
     ip_reps = 8;
   }
+  // split-schedule dispatch by degree variance of the fine graph (keep in
+  // sync with partition_deep): heavy-tailed graphs defer all splits to the
+  // finest level (no eager coarsest split); CV^2 >= 1 as for the bisector
+  // dispatch.
+  bool late_splits = false;
+  {
+    const u32 fn = kmp_graph_n(g);
+    const u32 *fx = kmp_graph_xadj(g);
+    unsigned __int128 sum = 0, sq = 0;
+    for (u32 u = 0; u < fn; ++u) {
+      const u64 d = fx[u + 1] - fx[u];
+      sum += d;
+      sq += d * d;
+    }
+    const bool heavy =
+        static_cast<unsigned __int128>(fn) * sq >= 2 * sum * sum;
+    late_splits = heavy && fn <= (1u << 21);
+  }
   const i64 total_w = kmp_graph_total_node_weight(g);
   const i64 mbw_val = kmp_max_block_weight(g, k, eps);
 
@@ -3728,7 +3746,8 @@ i64 kmp_partition_deep(
   part.assign(sizes.back(), 0);
   coarsest = engines.size() - 1;
   for (size_t level = engines.size(); level-- > 0;) {
-    const u32 sc = level == coarsest ? std::min(split_c, 48u) : split_c;
+    const u32 sc = (level == coarsest && !late_splits)
+                       ? std::min(split_c, 48u) : split_c;
     if (num_groups < k &&
         (static_cast<u64>(sizes[level]) >= 2ull * sc * num_groups ||
          level == 0)) {

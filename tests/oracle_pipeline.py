@@ -90,6 +90,13 @@ def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
 
     if split_c is None:
         split_c = 262144 if g.n <= (1 << 21) else 2000
+    # split-schedule dispatch by degree variance (keep in sync with
+    # partition_deep): heavy-tailed fine graphs defer all splits to the
+    # finest level (no eager coarsest split)
+    xadj = np.asarray(g.xadj, dtype=np.int64)
+    d = xadj[1:] - xadj[:-1]
+    heavy = g.n * int((d.astype(object) * d).sum()) >= 2 * int(d.sum()) ** 2
+    late_splits = heavy and g.n <= (1 << 21)
 
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
@@ -122,7 +129,8 @@ def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
     coarsest = len(graphs) - 1
     for level in range(coarsest, -1, -1):
         hg = graphs[level]
-        sc = min(split_c, 48) if level == coarsest else split_c
+        sc = (min(split_c, 48) if level == coarsest and not late_splits
+              else split_c)
         if len(groups) < k and (hg.n >= 2 * sc * len(groups)
                                 or level == 0):
             part, groups = _extend_partition(hg, part, groups, mbw_val, k,

@@ -110,15 +110,22 @@ def test_oracle_deep_pipeline_matches_expected(oracle, name):
 
 
 def test_deep_pipeline_band_vs_reference_goldens():
-    """The deep pipeline's cuts stay within a tighter band of the compiled
-    reference's deep-multilevel cuts than the basic pipeline's (<=1.35x the
-    best seed on these cases; measured 1.04-1.29)."""
+    """The deep pipeline's cuts stay within a band of the compiled
+    reference's deep-multilevel cuts (<=1.30x the best of 3 seeds on these
+    cases; measured 0.74-1.25 with the split-schedule dispatch: heavy-tail
+    graphs defer splits to the finest level and BEAT the reference on all
+    four R-MAT cases)."""
     exp = _load("pipeline_deep_expected.json")
     band = _load("ref_golden_partition.json")
     for name in ("walshaw_k2", "walshaw_k16", "rgg2d_k4", "rmat14_s42_k16",
                  "rmat16_s42_k16", "rmat18_s42_k16", "rmat18_s42_k64"):
         ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
-        assert exp[name]["cut"] <= 1.35 * ref_best, (name, exp[name]["cut"], ref_best)
+        assert exp[name]["cut"] <= 1.30 * ref_best, (name, exp[name]["cut"], ref_best)
+    # heavy-tail cases must stay at or below the reference's best seed
+    for name in ("rmat14_s42_k16", "rmat16_s42_k16", "rmat18_s42_k16",
+                 "rmat18_s42_k64"):
+        ref_best = min(band[name][f"seed{s}"]["cut"] for s in (1, 2, 3))
+        assert exp[name]["cut"] <= ref_best, (name, exp[name]["cut"], ref_best)
 
 
 def test_native_extend_partition_matches_python():
