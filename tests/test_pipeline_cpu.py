@@ -142,7 +142,11 @@ def test_native_extend_partition_matches_python():
         ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
         ctypes.POINTER(ctypes.c_uint32)]
 
-    for scale, k, split_c, force in ((12, 16, 48, 1), (12, 8, 200, 0)):
+    # cases cross the bisector-dispatch thresholds (4096-vertex O(n^2)
+    # cutoff, heavy/light-tail CV^2 split) and the parallel per-group path
+    # (many groups at once, with w<2 pass-through groups at k=23)
+    for scale, k, split_c, force in ((12, 16, 48, 1), (12, 8, 200, 0),
+                                     (14, 23, 1, 1), (13, 64, 16, 0)):
         g = ka.Graph.rmat(scale, 8, 42)
         mbw = g.max_block_weight(k, 0.03)
         part_py = np.zeros(g.n, np.uint32)
