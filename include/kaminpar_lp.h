@@ -292,6 +292,17 @@ int kmp_balance_partition(
     const kmp_graph_t *g, uint32_t k, int64_t cap, uint32_t *part
 );
 
+/* Deterministic k-way boundary FM with pass-level best-prefix rollback
+ * (serial-deterministic restatement of the reference's k-way FM refiner
+ * shape, kaminpar-shm/refinement/fm/fm_refiner.cc). caps[k] are hard
+ * per-block weight caps (0 closes a block). 0 -> defaults: max_passes 3,
+ * max_fruitless 300. CPU-only; the multilevel drivers run it per level on
+ * graphs <= ~2M fine vertices. */
+int kmp_kway_fm(
+    const kmp_graph_t *g, uint32_t k, const int64_t *caps, uint32_t *part,
+    int max_passes, int max_fruitless
+);
+
 /* Flat / multilevel FM bisection of a vertex subset (see partition_host.cpp). */
 int kmp_bisect_subset(
     const kmp_graph_t *g, const uint32_t *nodes, uint32_t n_sub,

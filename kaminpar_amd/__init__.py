@@ -95,6 +95,9 @@ def _load():
     lib.kmp_initial_partition.argtypes = [vp, u32, i64, ctypes.c_int, p(u32)]
     lib.kmp_balance_partition.restype = ctypes.c_int
     lib.kmp_balance_partition.argtypes = [vp, u32, i64, p(u32)]
+    lib.kmp_kway_fm.restype = ctypes.c_int
+    lib.kmp_kway_fm.argtypes = [vp, u32, p(i64), p(u32),
+                                ctypes.c_int, ctypes.c_int]
     lib.kmp_bisect_subset.restype = ctypes.c_int
     lib.kmp_bisect_subset.argtypes = [vp, p(u32), u32, i64, i64, i64,
                                       ctypes.c_int, p(ctypes.c_uint8)]
@@ -266,6 +269,17 @@ class Graph:
         """Gain-aware overload balancer (uniform cap), in place."""
         part = np.ascontiguousarray(part, dtype=np.uint32)
         _lib.kmp_balance_partition(self._h, k, int(cap), _u32p(part))
+        return part
+
+    def kway_fm(self, k, caps, part, max_passes=0, max_fruitless=0):
+        """Deterministic k-way boundary FM (best-prefix rollback), in
+        place; caps is a k-length per-block hard cap array (0 closes a
+        block)."""
+        caps = np.ascontiguousarray(caps, dtype=np.int64)
+        part = np.ascontiguousarray(part, dtype=np.uint32)
+        _lib.kmp_kway_fm(self._h, k,
+                         caps.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                         _u32p(part), max_passes, max_fruitless)
         return part
 
     def bisect_subset(self, nodes, target1, cap1, cap2, reps=8):

@@ -3748,20 +3748,18 @@ i64 kmp_partition_deep(
   for (size_t level = engines.size(); level-- > 0;) {
     const u32 sc = (level == coarsest && !late_splits)
                        ? std::min(split_c, 48u) : split_c;
+    kmp_graph_t *hg_owned = nullptr;
+    const kmp_graph_t *hg = nullptr;
     if (num_groups < k &&
         (static_cast<u64>(sizes[level]) >= 2ull * sc * num_groups ||
          level == 0)) {
-      kmp_graph_t *hg_owned =
-          level == 0 ? nullptr : kmp_lp_download_graph(engines[level]);
-      const kmp_graph_t *hg = level == 0 ? g : hg_owned;
+      hg_owned = level == 0 ? nullptr : kmp_lp_download_graph(engines[level]);
+      hg = level == 0 ? g : hg_owned;
       kmp_extend_partition(hg, part.data(), k, mbw_val, sc, ip_reps,
                            level == 0 ? 1 : 0, group_lo.data(),
                            group_w.data(), &num_groups);
       if (num_groups == k) {
         kmp_balance_partition(hg, k, mbw_val, part.data());
-      }
-      if (hg_owned) {
-        kmp_graph_free(hg_owned);
       }
     }
     std::fill(caps.begin(), caps.end(), 0);
@@ -3771,7 +3769,26 @@ i64 kmp_partition_deep(
     rc = kmp_lp_refine(engines[level], k, caps.data(), part.data(), seed,
                        iters, &st);
     if (rc < 0) {
+      if (hg_owned) {
+        kmp_graph_free(hg_owned);
+      }
       goto done;
+    }
+    // per-level k-way boundary FM on small graphs (keep in sync with
+    // partition_deep / the oracle mirror)
+    if (kmp_graph_n(g) <= (1u << 21)) {
+      if (!hg) {
+        hg_owned =
+            level == 0 ? nullptr : kmp_lp_download_graph(engines[level]);
+        hg = level == 0 ? g : hg_owned;
+      }
+      kmp_kway_fm(hg, k, caps.data(), part.data(), 0, 0);
+      if (level == 0) {
+        rc = kmp_edge_cut_host(g, part.data());
+      }
+    }
+    if (hg_owned) {
+      kmp_graph_free(hg_owned);
     }
     if (level > 0) {
       const std::vector<u32> &map = mappings[level - 1];

@@ -903,6 +903,184 @@ int kmp_extend_partition(
   return 0;
 }
 
+// Deterministic k-way boundary FM with pass-level best-prefix rollback --
+// the serial-deterministic restatement of the shape of the reference's
+// k-way FM refiner (kaminpar-shm/refinement/fm/fm_refiner.cc: gain-PQ over
+// boundary nodes, moves to the best feasible adjacent block, bounded
+// negative-gain hill climbing, rollback to the best seen prefix). Used by
+// the multilevel drivers on fine graphs (<= ~2M vertices), where LP
+// refinement alone cannot recover bisection quality on mesh-like graphs.
+// caps[k]: per-block hard weight caps (0 closes a block). Deterministic:
+// serial, lazy max-heap ordered by (gain, smaller vertex id), target ties
+// broken by smaller block id.
+int kmp_kway_fm(
+    const kmp_graph_t *g, u32 k, const i64 *caps, u32 *part,
+    int max_passes, int max_fruitless
+) {
+  const u32 n = kmp_graph_n(g);
+  const u32 *xadj = kmp_graph_xadj(g);
+  const u32 *adjncy = kmp_graph_adjncy(g);
+  const i32 *vwgt = kmp_graph_vwgt(g);
+  const i32 *adjwgt = kmp_graph_adjwgt(g);
+  if (max_passes <= 0) {
+    max_passes = 3;
+  }
+  if (max_fruitless <= 0) {
+    max_fruitless = 300;
+  }
+
+  std::vector<i64> bw(k, 0);
+  for (u32 u = 0; u < n; ++u) {
+    bw[part[u]] += vwgt ? vwgt[u] : 1;
+  }
+
+  // scratch connectivity (k-sized, cleared via touched list)
+  std::vector<i64> conn(k, 0);
+  std::vector<u32> touched;
+  touched.reserve(64);
+
+  // best feasible move for v under the CURRENT state: (gain, target) with
+  // target among adjacent blocks, ties -> smaller block id. Returns false
+  // if v has no adjacent block other than its own or no feasible target.
+  auto best_move = [&](u32 v, i64 *gain_out, u32 *t_out) -> bool {
+    const u32 b = part[v];
+    const i64 wv = vwgt ? vwgt[v] : 1;
+    touched.clear();
+    for (u64 e = xadj[v]; e < xadj[v + 1]; ++e) {
+      const u32 t = part[adjncy[e]];
+      if (conn[t] == 0) {
+        touched.push_back(t);
+      }
+      conn[t] += adjwgt ? adjwgt[e] : 1;
+    }
+    const i64 internal = conn[b];
+    bool found = false;
+    i64 bg = 0;
+    u32 bt = 0;
+    for (u32 t : touched) {
+      if (t == b || bw[t] + wv > caps[t]) {
+        continue;
+      }
+      const i64 gn = conn[t] - internal;
+      if (!found || gn > bg || (gn == bg && t < bt)) {
+        found = true;
+        bg = gn;
+        bt = t;
+      }
+    }
+    for (u32 t : touched) {
+      conn[t] = 0;
+    }
+    *gain_out = bg;
+    *t_out = bt;
+    return found;
+  };
+
+  constexpr i64 kNone = INT64_MIN;
+  std::vector<i64> key(n);      // gain currently in the heap (kNone = out)
+  std::vector<uint8_t> locked(n);
+  std::vector<std::pair<i64, i64>> heap; // (gain, -v) max-heap
+  struct Move {
+    u32 v, from, to;
+    i64 w;
+  };
+  std::vector<Move> moves;
+
+  for (int pass = 0; pass < max_passes; ++pass) {
+    std::fill(key.begin(), key.end(), kNone);
+    std::fill(locked.begin(), locked.end(), 0);
+    heap.clear();
+    moves.clear();
+    for (u32 v = 0; v < n; ++v) {
+      bool boundary = false;
+      for (u64 e = xadj[v]; e < xadj[v + 1] && !boundary; ++e) {
+        boundary = part[adjncy[e]] != part[v];
+      }
+      if (!boundary) {
+        continue;
+      }
+      i64 gn;
+      u32 t;
+      if (best_move(v, &gn, &t)) {
+        key[v] = gn;
+        heap.emplace_back(gn, -static_cast<i64>(v));
+      }
+    }
+    std::make_heap(heap.begin(), heap.end());
+
+    i64 cum = 0, best = 0;
+    size_t best_len = 0;
+    int fruitless = 0;
+    while (!heap.empty() && fruitless < max_fruitless) {
+      std::pop_heap(heap.begin(), heap.end());
+      const i64 gn = heap.back().first;
+      const u32 v = static_cast<u32>(-heap.back().second);
+      heap.pop_back();
+      if (locked[v] || key[v] != gn) {
+        continue; // stale entry
+      }
+      i64 cur_g;
+      u32 t;
+      if (!best_move(v, &cur_g, &t)) {
+        key[v] = kNone;
+        continue;
+      }
+      if (cur_g != gn) {
+        key[v] = cur_g;
+        heap.emplace_back(cur_g, -static_cast<i64>(v));
+        std::push_heap(heap.begin(), heap.end());
+        continue;
+      }
+      // perform the move
+      const u32 b = part[v];
+      const i64 wv = vwgt ? vwgt[v] : 1;
+      part[v] = t;
+      bw[b] -= wv;
+      bw[t] += wv;
+      locked[v] = 1;
+      key[v] = kNone;
+      moves.push_back({v, b, t, wv});
+      cum += cur_g;
+      if (cum > best) {
+        best = cum;
+        best_len = moves.size();
+        fruitless = 0;
+      } else {
+        ++fruitless;
+      }
+      // refresh unlocked neighbours (lazy: push the new exact key)
+      for (u64 e = xadj[v]; e < xadj[v + 1]; ++e) {
+        const u32 u = adjncy[e];
+        if (locked[u]) {
+          continue;
+        }
+        i64 ug;
+        u32 ut;
+        if (best_move(u, &ug, &ut)) {
+          if (key[u] != ug) {
+            key[u] = ug;
+            heap.emplace_back(ug, -static_cast<i64>(u));
+            std::push_heap(heap.begin(), heap.end());
+          }
+        } else {
+          key[u] = kNone;
+        }
+      }
+    }
+    // roll back past the best prefix
+    for (size_t i = moves.size(); i-- > best_len;) {
+      const Move &m = moves[i];
+      part[m.v] = m.from;
+      bw[m.to] -= m.w;
+      bw[m.from] += m.w;
+    }
+    if (best <= 0) {
+      break;
+    }
+  }
+  return 0;
+}
+
 // Gain-aware overload balancer on a host graph (uniform cap), exposed for
 // the progressive-k driver.
 int kmp_balance_partition(

@@ -464,6 +464,7 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
         # afterwards extend only when every block keeps >= split_c vertices
         sc = (min(split_c, 48) if level == coarsest and not late_splits
               else split_c)
+        hg = None
         if len(groups) < k and (sizes[level] >= 2 * sc * len(groups)
                                 or level == 0):
             hg = g if level == 0 else engines[level].download_graph()
@@ -472,10 +473,20 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
                                              force=(level == 0))
             if len(groups) == k:
                 hg.balance_partition(k, mbw_val, part)
+        caps = _group_caps(groups, k, mbw_val)
         cut, part, rst = engines[level].refine(
-            k, _group_caps(groups, k, mbw_val), part, seed=seed, iters=iters)
+            k, caps, part, seed=seed, iters=iters)
         arcs_total += rst.arcs_scanned
         ns_total += rst.phase_a_ns
+        # per-level k-way boundary FM on small graphs (<= ~2M fine
+        # vertices): recovers the bisection quality LP refinement alone
+        # cannot on mesh-like graphs (classic multilevel FM recipe)
+        if g.n <= (1 << 21):
+            if hg is None:
+                hg = g if level == 0 else engines[level].download_graph()
+            part = hg.kway_fm(k, caps, part)
+            if level == 0:
+                cut = g.edge_cut(part)
         if level > 0:
             part = part[mappings[level - 1]]
     if return_arcs:

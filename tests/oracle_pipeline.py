@@ -139,10 +139,16 @@ def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
             if len(groups) == k:
                 hg.balance_partition(k, mbw_val, part)
         vw, aw = weights(hg)
-        cut, part, _ = oracle_refine(oracle, hg, k,
-                                     _group_caps(groups, k, mbw_val), part,
+        caps = _group_caps(groups, k, mbw_val)
+        cut, part, _ = oracle_refine(oracle, hg, k, caps, part,
                                      seed=seed, iters=iters, vwgt=vw,
                                      adjwgt=aw)
+        # per-level k-way boundary FM on small graphs (keep in sync with
+        # partition_deep)
+        if g.n <= (1 << 21):
+            part = hg.kway_fm(k, caps, part)
+            if level == 0:
+                cut = g.edge_cut(part)
         if level > 0:
             part = part[mappings[level - 1]]
     return cut, part, [gr.n for gr in graphs]

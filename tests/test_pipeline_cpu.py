@@ -163,3 +163,30 @@ def test_native_extend_partition_matches_python():
                                   ctypes.byref(num))
         assert np.array_equal(part_py, part_c)
         assert groups_py == [(int(lo[i]), int(w[i])) for i in range(num.value)]
+
+
+@pytest.mark.parametrize("k", [2, 8, 16])
+def test_kway_fm_improves_respects_caps_deterministic(k):
+    """kmp_kway_fm: lowers the cut from a random partition, never violates
+    the per-block caps (0-cap blocks never receive), and is deterministic."""
+    rng = np.random.default_rng(11)
+    g = ka.Graph.rgg2d(4096, 8, seed=9)
+    vw = np.ones(g.n, np.int64)
+    part0 = rng.integers(0, k, g.n).astype(np.uint32)
+    cut0 = g.edge_cut(part0)
+    caps = np.full(k, g.max_block_weight(k, 0.03), np.int64)
+    if k > 2:
+        caps[k - 1] = 0  # closed block: must only lose vertices
+    p1 = g.kway_fm(k, caps, part0.copy())
+    p2 = g.kway_fm(k, caps, part0.copy())
+    assert np.array_equal(p1, p2)
+    cut1 = g.edge_cut(p1)
+    assert cut1 < cut0
+    bw = np.zeros(k, np.int64)
+    np.add.at(bw, p1, vw)
+    bw0 = np.zeros(k, np.int64)
+    np.add.at(bw0, part0, vw)
+    for b in range(k):
+        assert bw[b] <= max(caps[b], bw0[b])
+    if k > 2:
+        assert bw[k - 1] <= bw0[k - 1]
