@@ -3573,12 +3573,26 @@ i64 kmp_partition(
     goto done;
   }
 
-  // uncoarsen: refine at every level, projecting through the mappings
+  // uncoarsen: refine at every level, projecting through the mappings;
+  // per-level k-way boundary FM on small graphs (keep in sync with
+  // partition() / the oracle mirror)
   for (size_t level = engines.size(); level-- > 0;) {
     rc = kmp_lp_refine(engines[level], k, mbw.data(), part.data(), seed,
                        iters, &st);
     if (rc < 0) {
       goto done;
+    }
+    if (kmp_graph_n(g) <= (1u << 21)) {
+      kmp_graph_t *hg_owned =
+          level == 0 ? nullptr : kmp_lp_download_graph(engines[level]);
+      const kmp_graph_t *hg = level == 0 ? g : hg_owned;
+      kmp_kway_fm(hg, k, mbw.data(), part.data(), 0, 0);
+      if (level == 0) {
+        rc = kmp_edge_cut_host(g, part.data());
+      }
+      if (hg_owned) {
+        kmp_graph_free(hg_owned);
+      }
     }
     if (level > 0) {
       const std::vector<u32> &map = mappings[level - 1];

@@ -322,12 +322,20 @@ def partition(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     coarsest = engines[-1].download_graph() if len(engines) > 1 else g
     part = coarsest.initial_partition_native(k, mbw_val)
 
-    # ---- uncoarsen: refine at every level (GPU) ----
+    # ---- uncoarsen: refine at every level (GPU), plus per-level k-way
+    # boundary FM on small graphs (<= ~2M fine vertices; same recipe as
+    # partition_deep) ----
     cut = None
+    fm_on = g.n <= (1 << 21)
     for level in range(len(engines) - 1, -1, -1):
         cut, part, rst = engines[level].refine(k, mbw, part, seed=seed, iters=iters)
         arcs_total += rst.arcs_scanned
         ns_total += rst.phase_a_ns
+        if fm_on:
+            hg = g if level == 0 else engines[level].download_graph()
+            part = hg.kway_fm(k, mbw, part)
+            if level == 0:
+                cut = g.edge_cut(part)
         if level > 0:
             part = part[mappings[level - 1]]
     levels = sizes

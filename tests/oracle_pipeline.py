@@ -69,11 +69,17 @@ def oracle_partition(oracle, g, k, eps=0.03, seed=1, iters=5,
     part = initial_partition(graphs[-1], k, mbw_val, seed=seed)
 
     cut = None
+    fm_on = g.n <= (1 << 21)
     for level in range(len(graphs) - 1, -1, -1):
         gr = graphs[level]
         vw, aw = weights(gr)
         cut, part, _ = oracle_refine(oracle, gr, k, mbw, part, seed=seed,
                                      iters=iters, vwgt=vw, adjwgt=aw)
+        # per-level k-way boundary FM (keep in sync with partition())
+        if fm_on:
+            part = gr.kway_fm(k, mbw, part)
+            if level == 0:
+                cut = g.edge_cut(part)
         if level > 0:
             part = part[mappings[level - 1]]
     return cut, part, [gr.n for gr in graphs]
