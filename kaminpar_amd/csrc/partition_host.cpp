@@ -844,13 +844,14 @@ int kmp_extend_partition(
         const i64 t1 = total * k1 / w;
         std::vector<uint8_t> side(nodes.size());
         // deterministic dispatch (keep in sync with partition.py
-        // _extend_partition): <=4096 vertices use the pinned O(n^2)
-        // bisector; larger subgraphs use the lazy-PQ flat bisector for
-        // heavy-tailed degree distributions (flat FM works, HEM collapses
-        // hubs) and the HEM multilevel bisector for low-variance degrees
-        // (geometric/mesh-like, where flat FM gets lost at scale) --
-        // split at squared coefficient of variation >= 1, i.e.
-        // n * sum(d^2) >= 2 * sum(d)^2.
+        // _extend_partition): <=256 vertices use the pinned O(n^2)
+        // bisector; above that by degree variance (squared coefficient
+        // of variation >= 1, i.e. n * sum(d^2) >= 2 * sum(d)^2):
+        // heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then the
+        // lazy-PQ bisector -- HEM collapses hubs), low-variance
+        // (geometric/mesh-like) subgraphs use the HEM multilevel
+        // bisector, where flat FM gets lost (measured: rgg2d k=2 at
+        // 2.5x the reference with flat vs 1.0x with HEM).
         const size_t ns = nodes.size();
         int reps_eff = reps;
         if (ns > 131072) {
@@ -859,7 +860,7 @@ int kmp_extend_partition(
           reps_eff = std::min(reps, 4);
         }
         auto *bisect = kmp_bisect_subset;
-        if (ns > 4096) {
+        if (ns > 256) {
           const u32 *xadj = kmp_graph_xadj(g);
           unsigned __int128 sum = 0, sq = 0;
           for (u32 u : nodes) {
@@ -869,7 +870,11 @@ int kmp_extend_partition(
           }
           const bool heavy_tail =
               static_cast<unsigned __int128>(ns) * sq >= 2 * sum * sum;
-          bisect = heavy_tail ? kmp_bisect_subset_fast : kmp_bisect_subset_ml;
+          if (heavy_tail) {
+            bisect = ns <= 4096 ? kmp_bisect_subset : kmp_bisect_subset_fast;
+          } else {
+            bisect = kmp_bisect_subset_ml;
+          }
         }
         bisect(g, nodes.data(), ns, t1,
                static_cast<i64>(k1) * mbw_val,

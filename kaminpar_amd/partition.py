@@ -384,14 +384,17 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
             total = int(vw[nodes].sum())
             t1 = total * k1 // w
             # deterministic dispatch (keep in sync with the C twin
-            # kmp_extend_partition): pinned O(n^2) bisector <= 4096
-            # vertices; above that, flat lazy-PQ FM for heavy-tailed
-            # degrees (CV^2 >= 1), HEM multilevel for low-variance
-            # (geometric/mesh-like) subgraphs where flat FM gets lost
+            # kmp_extend_partition): pinned O(n^2) bisector <= 256
+            # vertices; above that by degree variance (CV^2 >= 1):
+            # heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then
+            # lazy-PQ -- HEM collapses hubs), low-variance
+            # (geometric/mesh-like) subgraphs use the HEM multilevel
+            # bisector, where flat FM gets lost (measured: rgg2d k=2 at
+            # 2.5x the reference with flat vs 1.0x with HEM)
             ns = len(nodes)
             reps_eff = reps if ns <= 16384 else (4 if ns <= 131072 else 2)
             reps_eff = min(reps, reps_eff)
-            if ns <= 4096:
+            if ns <= 256:
                 bisect = hg.bisect_subset
             else:
                 xadj = np.asarray(hg.xadj)
@@ -400,8 +403,11 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
                 s = int(np.sum(d))
                 sq = int(np.sum(d * d))
                 heavy_tail = ns * sq >= 2 * s * s
-                bisect = (hg.bisect_subset_fast if heavy_tail
-                          else hg.bisect_subset_ml)
+                if heavy_tail:
+                    bisect = (hg.bisect_subset if ns <= 4096
+                              else hg.bisect_subset_fast)
+                else:
+                    bisect = hg.bisect_subset_ml
             side = bisect(nodes, t1, k1 * mbw_val, k2 * mbw_val,
                           reps=reps_eff)
             part[nodes[~side]] = b + k1
