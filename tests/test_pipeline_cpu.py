@@ -142,12 +142,18 @@ def test_native_extend_partition_matches_python():
         ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
         ctypes.POINTER(ctypes.c_uint32)]
 
-    # cases cross the bisector-dispatch thresholds (4096-vertex O(n^2)
-    # cutoff, heavy/light-tail CV^2 split) and the parallel per-group path
-    # (many groups at once, with w<2 pass-through groups at k=23)
-    for scale, k, split_c, force in ((12, 16, 48, 1), (12, 8, 200, 0),
-                                     (14, 23, 1, 1), (13, 64, 16, 0)):
-        g = ka.Graph.rmat(scale, 8, 42)
+    # cases cross the bisector-dispatch thresholds (256-vertex O(n^2)
+    # cutoff, heavy/light-tail CV^2 split incl. the HEM path on mesh
+    # subgraphs) and the parallel per-group path (many groups at once,
+    # with w<2 pass-through groups at k=23)
+    for graph, k, split_c, force in (("rmat12", 16, 48, 1),
+                                     ("rmat12", 8, 200, 0),
+                                     ("rmat14", 23, 1, 1),
+                                     ("rmat13", 64, 16, 0),
+                                     ("rgg16k", 16, 1, 1),
+                                     ("rgg16k", 7, 300, 0)):
+        g = (ka.Graph.rgg2d(16384, 8, seed=5) if graph == "rgg16k"
+             else ka.Graph.rmat(int(graph[4:]), 8, 42))
         mbw = g.max_block_weight(k, 0.03)
         part_py = np.zeros(g.n, np.uint32)
         part_py, groups_py = _extend_partition(
