@@ -24,6 +24,8 @@ assignment -- without the cap, deep coarsening produces vertices weighing
 ~25% of a block, which makes balanced bin-packing infeasible.
 """
 
+import os
+
 import numpy as np
 
 from . import LpEngine
@@ -364,6 +366,35 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
     vwp = _lib.kmp_graph_vwgt(hg._h)
     vw = (np.ctypeslib.as_array(vwp, shape=(hg.n,)).astype(np.int64)
           if vwp else np.ones(hg.n, np.int64))
+    from concurrent.futures import ThreadPoolExecutor
+
+    def _bisect_group(task):
+        _b, _k1, nodes, t1, reps_eff, cap1, cap2 = task
+        # deterministic dispatch (keep in sync with the C twin
+        # kmp_extend_partition): pinned O(n^2) bisector <= 256
+        # vertices; above that by degree variance (CV^2 >= 1):
+        # heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then
+        # lazy-PQ -- HEM collapses hubs), low-variance
+        # (geometric/mesh-like) subgraphs use the HEM multilevel
+        # bisector, where flat FM gets lost (measured: rgg2d k=2 at
+        # 2.5x the reference with flat vs 1.0x with HEM)
+        ns = len(nodes)
+        if ns <= 256:
+            bisect = hg.bisect_subset
+        else:
+            xadj = np.asarray(hg.xadj)
+            d = (xadj[nodes.astype(np.int64) + 1]
+                 - xadj[nodes.astype(np.int64)]).astype(object)
+            s = int(np.sum(d))
+            sq = int(np.sum(d * d))
+            heavy_tail = ns * sq >= 2 * s * s
+            if heavy_tail:
+                bisect = (hg.bisect_subset if ns <= 4096
+                          else hg.bisect_subset_fast)
+            else:
+                bisect = hg.bisect_subset_ml
+        return bisect(nodes, t1, cap1, cap2, reps=reps_eff)
+
     while True:
         num = len(groups)
         if num >= k:
@@ -371,6 +402,7 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
         if not force and hg.n < 2 * split_c * num:
             break
         new_groups = []
+        tasks = []
         for b, w in groups:
             if w < 2:
                 new_groups.append((b, w))
@@ -383,35 +415,23 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
                 continue
             total = int(vw[nodes].sum())
             t1 = total * k1 // w
-            # deterministic dispatch (keep in sync with the C twin
-            # kmp_extend_partition): pinned O(n^2) bisector <= 256
-            # vertices; above that by degree variance (CV^2 >= 1):
-            # heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then
-            # lazy-PQ -- HEM collapses hubs), low-variance
-            # (geometric/mesh-like) subgraphs use the HEM multilevel
-            # bisector, where flat FM gets lost (measured: rgg2d k=2 at
-            # 2.5x the reference with flat vs 1.0x with HEM)
             ns = len(nodes)
             reps_eff = reps if ns <= 16384 else (4 if ns <= 131072 else 2)
             reps_eff = min(reps, reps_eff)
-            if ns <= 256:
-                bisect = hg.bisect_subset
-            else:
-                xadj = np.asarray(hg.xadj)
-                d = (xadj[nodes.astype(np.int64) + 1]
-                     - xadj[nodes.astype(np.int64)]).astype(object)
-                s = int(np.sum(d))
-                sq = int(np.sum(d * d))
-                heavy_tail = ns * sq >= 2 * s * s
-                if heavy_tail:
-                    bisect = (hg.bisect_subset if ns <= 4096
-                              else hg.bisect_subset_fast)
-                else:
-                    bisect = hg.bisect_subset_ml
-            side = bisect(nodes, t1, k1 * mbw_val, k2 * mbw_val,
-                          reps=reps_eff)
-            part[nodes[~side]] = b + k1
+            tasks.append((b, k1, nodes, t1, reps_eff,
+                          k1 * mbw_val, k2 * mbw_val))
             new_groups += [(b, k1), (b + k1, k2)]
+        # groups are independent subproblems (disjoint part[] writes); the
+        # ctypes bisector calls release the GIL, so a thread pool gives the
+        # same per-group parallelism as the C twin's OpenMP loop, with
+        # bit-identical results (sides applied serially afterwards)
+        if len(tasks) > 1:
+            with ThreadPoolExecutor(max_workers=os.cpu_count()) as ex:
+                sides = list(ex.map(_bisect_group, tasks))
+        else:
+            sides = [_bisect_group(t) for t in tasks]
+        for (b, k1, nodes, _t1, _r, _c1, _c2), side in zip(tasks, sides):
+            part[nodes[~side]] = b + k1
         groups = new_groups
     return part, groups
 
