@@ -854,9 +854,7 @@ int kmp_extend_partition(
         // 2.5x the reference with flat vs 1.0x with HEM).
         const size_t ns = nodes.size();
         int reps_eff = reps;
-        if (ns > 131072) {
-          reps_eff = std::min(reps, 2);
-        } else if (ns > 16384) {
+        if (ns > 16384) {
           reps_eff = std::min(reps, 4);
         }
         auto *bisect = kmp_bisect_subset;

@@ -416,7 +416,7 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
             total = int(vw[nodes].sum())
             t1 = total * k1 // w
             ns = len(nodes)
-            reps_eff = reps if ns <= 16384 else (4 if ns <= 131072 else 2)
+            reps_eff = reps if ns <= 16384 else 4
             reps_eff = min(reps, reps_eff)
             tasks.append((b, k1, nodes, t1, reps_eff,
                           k1 * mbw_val, k2 * mbw_val))
