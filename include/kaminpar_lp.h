@@ -318,7 +318,11 @@ int kmp_bisect_subset_ml(
  * kaminpar-shm/partitioning/deep/deep_multilevel.cc). Bit-identical to the
  * Python driver kaminpar_amd.partition.partition_deep. Better cuts than
  * kmp_partition on every golden case. 0 defaults:
- * contraction_limit 2000, stop_n 512, split_c 262144, ip_reps 8. */
+ * contraction_limit 2000, stop_n 512, split_c auto (262144 up to 2M
+ * vertices, 2000 beyond), ip_reps 8. split_c >= n selects the full
+ * late-split quality mode on heavy-tailed graphs: every split at the
+ * finest level (measured at R-MAT scale 23: cut 0.44x the reference's
+ * best seed, at minutes of host-side bisection cost -- see DESIGN.md). */
 int64_t kmp_partition_deep(
     const kmp_graph_t *g,
     uint32_t k,

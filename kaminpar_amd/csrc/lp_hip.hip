@@ -3707,7 +3707,7 @@ i64 kmp_partition_deep(
     }
     const bool heavy =
         static_cast<unsigned __int128>(fn) * sq >= 2 * sum * sum;
-    late_splits = heavy && fn <= (1u << 21);
+    late_splits = heavy && (fn <= (1u << 21) || split_c >= fn);
   }
   const i64 total_w = kmp_graph_total_node_weight(g);
   const i64 mbw_val = kmp_max_block_weight(g, k, eps);

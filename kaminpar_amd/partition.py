@@ -464,7 +464,10 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     xadj = np.asarray(g.xadj, dtype=np.int64)
     d = xadj[1:] - xadj[:-1]
     heavy = g.n * int((d.astype(object) * d).sum()) >= 2 * int(d.sum()) ** 2
-    late_splits = heavy and g.n <= (1 << 21)
+    # split_c >= n is the explicit full-late quality mode: every split at
+    # the finest level regardless of size (measured at scale 23: cut 0.44x
+    # the reference's best seed, profiles/round1/quality_rmat23_k16_late_full.json)
+    late_splits = heavy and (g.n <= (1 << 21) or split_c >= g.n)
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
 

@@ -102,7 +102,7 @@ def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
     xadj = np.asarray(g.xadj, dtype=np.int64)
     d = xadj[1:] - xadj[:-1]
     heavy = g.n * int((d.astype(object) * d).sum()) >= 2 * int(d.sum()) ** 2
-    late_splits = heavy and g.n <= (1 << 21)
+    late_splits = heavy and (g.n <= (1 << 21) or split_c >= g.n)
 
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
