@@ -146,14 +146,34 @@ def test_native_extend_partition_matches_python():
     # cutoff, heavy/light-tail CV^2 split incl. the HEM path on mesh
     # subgraphs) and the parallel per-group path (many groups at once,
     # with w<2 pass-through groups at k=23)
+    def weighted_rmat(scale):
+        base = ka.Graph.rmat(scale, 8, 42)
+        rng = np.random.default_rng(17)
+        vwgt = rng.integers(1, 12, base.n).astype(np.int32)
+        # symmetric arc weights keyed on the endpoint pair
+        xadj = np.asarray(base.xadj).astype(np.int64)
+        adjncy = np.asarray(base.adjncy).astype(np.int64)
+        src = np.repeat(np.arange(base.n, dtype=np.int64), np.diff(xadj))
+        lo = np.minimum(src, adjncy)
+        hi = np.maximum(src, adjncy)
+        adjwgt = (1 + (lo * 31 + hi * 7) % 9).astype(np.int32)
+        return ka.Graph.from_csr(np.asarray(base.xadj).copy(),
+                                 np.asarray(base.adjncy).copy(),
+                                 vwgt=vwgt, adjwgt=adjwgt)
+
     for graph, k, split_c, force in (("rmat12", 16, 48, 1),
                                      ("rmat12", 8, 200, 0),
                                      ("rmat14", 23, 1, 1),
                                      ("rmat13", 64, 16, 0),
                                      ("rgg16k", 16, 1, 1),
-                                     ("rgg16k", 7, 300, 0)):
-        g = (ka.Graph.rgg2d(16384, 8, seed=5) if graph == "rgg16k"
-             else ka.Graph.rmat(int(graph[4:]), 8, 42))
+                                     ("rgg16k", 7, 300, 0),
+                                     ("wrmat12", 16, 1, 1)):
+        if graph == "rgg16k":
+            g = ka.Graph.rgg2d(16384, 8, seed=5)
+        elif graph == "wrmat12":
+            g = weighted_rmat(12)
+        else:
+            g = ka.Graph.rmat(int(graph[4:]), 8, 42)
         mbw = g.max_block_weight(k, 0.03)
         part_py = np.zeros(g.n, np.uint32)
         part_py, groups_py = _extend_partition(
