@@ -167,7 +167,8 @@ def test_native_extend_partition_matches_python():
                                      ("rmat13", 64, 16, 0),
                                      ("rgg16k", 16, 1, 1),
                                      ("rgg16k", 7, 300, 0),
-                                     ("wrmat12", 16, 1, 1)):
+                                     ("wrmat12", 16, 1, 1),
+                                     ("rmat18", 2, 1, 1)):  # ns > 131072
         if graph == "rgg16k":
             g = ka.Graph.rgg2d(16384, 8, seed=5)
         elif graph == "wrmat12":
