@@ -217,3 +217,64 @@ def test_kway_fm_improves_respects_caps_deterministic(k):
         assert bw[b] <= max(caps[b], bw0[b])
     if k > 2:
         assert bw[k - 1] <= bw0[k - 1]
+
+
+def test_extend_partition_fuzz_python_vs_native():
+    """Seeded fuzz: Python and C extension drivers stay bit-identical on
+    random graphs across random k / split_c / force combinations."""
+    import ctypes
+    from kaminpar_amd import _lib, _u32p
+    from kaminpar_amd.partition import _extend_partition
+
+    _lib.kmp_extend_partition.restype = ctypes.c_int
+    _lib.kmp_extend_partition.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_uint32,
+        ctypes.c_int64, ctypes.c_uint32, ctypes.c_int, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_uint32), ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32)]
+
+    rng = np.random.default_rng(99)
+    for trial in range(6):
+        if trial % 2:
+            g = ka.Graph.rmat(int(rng.integers(10, 13)), 8,
+                              seed=int(rng.integers(1, 1000)))
+        else:
+            g = ka.Graph.rgg2d(int(rng.integers(512, 6000)), 8,
+                               seed=int(rng.integers(1, 1000)))
+        k = int(rng.integers(2, 33))
+        split_c = int(rng.integers(1, 500))
+        force = int(rng.integers(0, 2))
+        mbw = g.max_block_weight(k, 0.03)
+        part_py = np.zeros(g.n, np.uint32)
+        part_py, groups_py = _extend_partition(
+            g, part_py, [(0, k)], mbw, k, split_c=split_c, reps=8,
+            force=bool(force))
+        part_c = np.zeros(g.n, np.uint32)
+        lo = np.zeros(k, np.uint32)
+        w = np.zeros(k, np.uint32)
+        w[0] = k
+        num = ctypes.c_uint32(1)
+        _lib.kmp_extend_partition(g._h, _u32p(part_c), k, mbw, split_c, 8,
+                                  force, _u32p(lo), _u32p(w),
+                                  ctypes.byref(num))
+        assert np.array_equal(part_py, part_c), (trial, k, split_c, force)
+        assert groups_py == [(int(lo[i]), int(w[i]))
+                             for i in range(num.value)], trial
+
+
+def test_kway_fm_never_worsens_fuzz():
+    """Seeded fuzz: the pass-level best-prefix rollback guarantees the cut
+    never increases, from arbitrary (also infeasible) starts."""
+    rng = np.random.default_rng(123)
+    for trial in range(8):
+        n = int(rng.integers(256, 4096))
+        g = (ka.Graph.rgg2d(n, 8, seed=int(rng.integers(1, 1000)))
+             if trial % 2 else
+             ka.Graph.rmat(int(rng.integers(9, 12)), 8,
+                           seed=int(rng.integers(1, 1000))))
+        k = int(rng.integers(2, 17))
+        part = rng.integers(0, k, g.n).astype(np.uint32)
+        caps = np.full(k, g.max_block_weight(k, 0.03), np.int64)
+        cut0 = g.edge_cut(part)
+        out = g.kway_fm(k, caps, part.copy())
+        assert g.edge_cut(out) <= cut0, (trial, k)
