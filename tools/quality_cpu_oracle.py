@@ -3,11 +3,14 @@ through the CPU oracle mirror (tests/oracle_pipeline.py), which is
 bit-identical to the GPU pipeline stage by stage, so the cuts it reports
 are exactly what the GPU pipeline produces on the same inputs.
 
-Usage: python tools/quality_cpu_oracle.py {rmat|rgg2d} SCALE K [out.json]
+Usage: python tools/quality_cpu_oracle.py {rmat|rgg2d} SCALE K [out.json] [late]
 Requires oracle/_ref/libkaminpar_ref_full.so (built by
 oracle/_ref_build/Makefile.full from /root/reference in the dev
 container). rgg2d uses avg_deg 16 / seed 42, rmat edgefactor 8 / seed 42
-(the graphs of profiles/round1/quality_*.json).
+(the graphs of profiles/round1/quality_*.json). A trailing "late"
+selects the full late-split quality mode (split_c = n; minutes of
+host-side bisection work at scale 22+ — the 0.44x-of-reference artifact
+profiles/round1/quality_rmat23_k16_late_full.json).
 """
 
 import ctypes
@@ -31,6 +34,7 @@ def main():
     scale = int(sys.argv[2])
     k = int(sys.argv[3])
     out_path = sys.argv[4] if len(sys.argv) > 4 else None
+    late = len(sys.argv) > 5 and sys.argv[5] == "late"
 
     oracle = ctypes.CDLL(os.path.join(REPO, "oracle", "liblp_oracle.so"))
     ref = ctypes.CDLL(os.path.join(REPO, "oracle", "_ref",
@@ -60,10 +64,11 @@ def main():
         print(f"reference seed{seed}: cut={cut}", flush=True)
 
     t0 = time.time()
-    cut, part, levels = oracle_partition_deep(oracle, g, k, seed=1)
+    cut, part, levels = oracle_partition_deep(
+        oracle, g, k, seed=1, split_c=g.n if late else None)
     dt = time.time() - t0
     out["ours_deep_oracle"] = {
-        "cut": int(cut), "seconds": round(dt, 1),
+        "cut": int(cut), "seconds": round(dt, 1), "late": late,
         "note": "CPU oracle mirror, bit-identical to the GPU pipeline"}
     best = min(v["cut"] for v in out["reference"].values())
     out["deep_vs_reference_best"] = round(cut / best, 4)
