@@ -844,7 +844,7 @@ int kmp_extend_partition(
         const i64 t1 = total * k1 / w;
         std::vector<uint8_t> side(nodes.size());
         // deterministic dispatch (keep in sync with partition.py
-        // _extend_partition): <=256 vertices use the pinned O(n^2)
+        // _extend_partition): <=128 vertices use the pinned O(n^2)
         // bisector; above that by degree variance (squared coefficient
         // of variation >= 1, i.e. n * sum(d^2) >= 2 * sum(d)^2):
         // heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then the
@@ -858,7 +858,7 @@ int kmp_extend_partition(
           reps_eff = std::min(reps, 4);
         }
         auto *bisect = kmp_bisect_subset;
-        if (ns > 256) {
+        if (ns > 128) {
           const u32 *xadj = kmp_graph_xadj(g);
           unsigned __int128 sum = 0, sq = 0;
           for (u32 u : nodes) {

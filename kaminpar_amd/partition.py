@@ -371,7 +371,7 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
     def _bisect_group(task):
         _b, _k1, nodes, t1, reps_eff, cap1, cap2 = task
         # deterministic dispatch (keep in sync with the C twin
-        # kmp_extend_partition): pinned O(n^2) bisector <= 256
+        # kmp_extend_partition): pinned O(n^2) bisector <= 128
         # vertices; above that by degree variance (CV^2 >= 1):
         # heavy-tailed subgraphs use flat FM (O(n^2) to 4096, then
         # lazy-PQ -- HEM collapses hubs), low-variance
@@ -379,7 +379,7 @@ def _extend_partition(hg, part, groups, mbw_val, k, split_c=256, reps=8,
         # bisector, where flat FM gets lost (measured: rgg2d k=2 at
         # 2.5x the reference with flat vs 1.0x with HEM)
         ns = len(nodes)
-        if ns <= 256:
+        if ns <= 128:
             bisect = hg.bisect_subset
         else:
             xadj = np.asarray(hg.xadj)
