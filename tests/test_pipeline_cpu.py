@@ -168,11 +168,26 @@ def test_native_extend_partition_matches_python():
                                      ("rgg16k", 16, 1, 1),
                                      ("rgg16k", 7, 300, 0),
                                      ("wrmat12", 16, 1, 1),
+                                     ("wrgg8k", 8, 1, 1),  # weighted HEM path
                                      ("rmat18", 2, 1, 1)):  # ns > 131072
         if graph == "rgg16k":
             g = ka.Graph.rgg2d(16384, 8, seed=5)
         elif graph == "wrmat12":
             g = weighted_rmat(12)
+        elif graph == "wrgg8k":
+            base = ka.Graph.rgg2d(8192, 8, seed=5)
+            rng = np.random.default_rng(23)
+            vwgt = rng.integers(1, 9, base.n).astype(np.int32)
+            xadj = np.asarray(base.xadj).astype(np.int64)
+            adjncy = np.asarray(base.adjncy).astype(np.int64)
+            src = np.repeat(np.arange(base.n, dtype=np.int64),
+                            np.diff(xadj))
+            lo = np.minimum(src, adjncy)
+            hi = np.maximum(src, adjncy)
+            adjwgt = (1 + (lo * 13 + hi * 5) % 7).astype(np.int32)
+            g = ka.Graph.from_csr(np.asarray(base.xadj).copy(),
+                                  np.asarray(base.adjncy).copy(),
+                                  vwgt=vwgt, adjwgt=adjwgt)
         else:
             g = ka.Graph.rmat(int(graph[4:]), 8, 42)
         mbw = g.max_block_weight(k, 0.03)
