@@ -1,7 +1,7 @@
 // Shared host/device primitives for the deterministic LP schedule.
 // The PRNG/permutation/tie-hash definitions here are the GPU-side half of the
-// parity contract; oracle/lp_oracle.cpp restates them independently and
-// tests/test_schedule.py asserts both sides agree bit-for-bit.
+// parity contract; oracle/lp_oracle.cpp restates them independently and the
+// bit-parity suites in tests/test_gpu_parity.py assert both sides agree.
 #pragma once
 
 #include <stdint.h>

@@ -171,7 +171,8 @@ __global__ void k_phase_s(
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots,
-    u64 *__restrict__ m_slots,
+    u64 *__restrict__ m_list,
+    u32 *__restrict__ m_count,
     u64 *__restrict__ l_list,
     u32 *__restrict__ l_count
 ) {
@@ -186,7 +187,7 @@ __global__ void k_phase_s(
 
   const BlockPerm perm(n, iter_seed);
   // all 4 positions of this wave share one 64-vertex unit; one byte decides
-  // whether anything in it is active (slots are pre-marked invalid)
+  // whether anything in it is active (inactive units are skipped downstream)
   const u32 vb = perm.fp(p / kmp::kUnit);
   if (!unit_active[vb]) {
     return;
@@ -206,11 +207,24 @@ __global__ void k_phase_s(
     }
   }
 
-  // M-class positions go to per-position slots (compacted into the M work
-  // list by a stable select; no append atomics); L-class (rare) appends to
-  // the L list wave-aggregated.
-  if (!skip && slot == 0 && deg > kSmallDeg && deg <= kMidDeg) {
-    m_slots[sidx] = (static_cast<u64>(p) << 32) | u;
+  // M-class positions append to the M work list wave-aggregated (list order
+  // is irrelevant for determinism: M writes land in per-position slots);
+  // L-class (rare) appends to the L list the same way.
+  {
+    const bool is_m = !skip && slot == 0 && deg > kSmallDeg && deg <= kMidDeg;
+    const unsigned long long mm = __ballot(is_m);
+    if (mm) {
+      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
+      u32 bbase = 0;
+      if (lane == leader) {
+        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
+      }
+      bbase = __shfl(bbase, leader, kWave);
+      if (is_m) {
+        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] =
+            (static_cast<u64>(p) << 32) | u;
+      }
+    }
   }
   {
     const bool is_l = !skip && deg > kMidDeg && slot == 0;
@@ -228,8 +242,17 @@ __global__ void k_phase_s(
       }
     }
   }
-  if (skip || deg > kSmallDeg) {
-    return; // nothing to do here (M/L own larger degrees)
+  if (skip) {
+    // always-write contract: every position of an ACTIVE unit gets a fresh
+    // slot each chunk (proposal or invalid), so the memset-free v2 commit
+    // can read raw slots gated only by unit_active
+    if (slot == 0) {
+      slots[sidx] = Prop{0u, kInvalid, 0u, 0u};
+    }
+    return;
+  }
+  if (deg > kSmallDeg) {
+    return; // M/L kernels own (and always write) these slots
   }
 
   // candidate load: lane handles one edge
@@ -292,6 +315,8 @@ __global__ void k_phase_s(
     }
     if (fin.have && fin.c != cur) {
       slots[sidx] = Prop{u, fin.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[sidx] = Prop{0u, kInvalid, 0u, 0u};
     }
   }
 }
@@ -396,6 +421,8 @@ __global__ void k_phase_m(
     }
     if (fin.have && fin.c != cur) {
       slots[p - pos_lo] = Prop{u, fin.c, p - chunk_base, static_cast<u32>(u_w)};
+    } else {
+      slots[p - pos_lo] = Prop{0u, kInvalid, 0u, 0u};
     }
   }
   __threadfence_block(); // gains reuse across grid-stride iterations
@@ -580,6 +607,8 @@ __global__ void k_phase_l_sel(
       }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[p - pos_lo] = Prop{0u, kInvalid, 0u, 0u};
       }
     }
     __syncthreads();
@@ -697,6 +726,8 @@ __global__ void k_phase_l_direct(
       }
       if (total.have && total.c != cur) {
         slots[p - pos_lo] = Prop{u, total.c, p - chunk_base, static_cast<u32>(u_w)};
+      } else {
+        slots[p - pos_lo] = Prop{0u, kInvalid, 0u, 0u};
       }
     }
     __syncthreads();
@@ -733,7 +764,8 @@ __global__ void k_phase_s_c(
     const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots,
-    u64 *__restrict__ m_slots,
+    u64 *__restrict__ m_list,
+    u32 *__restrict__ m_count,
     u64 *__restrict__ l_list, // deg > kClusterMidDeg
     u32 *__restrict__ l_count
 ) {
@@ -767,8 +799,21 @@ __global__ void k_phase_s_c(
     }
   }
 
-  if (!skip && slot == 0 && deg > kSmallDeg && deg <= kClusterMidDeg) {
-    m_slots[sidx] = (static_cast<u64>(p) << 32) | u;
+  {
+    const bool is_m = !skip && slot == 0 && deg > kSmallDeg && deg <= kClusterMidDeg;
+    const unsigned long long mm = __ballot(is_m);
+    if (mm) {
+      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
+      u32 bbase = 0;
+      if (lane == leader) {
+        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
+      }
+      bbase = __shfl(bbase, leader, kWave);
+      if (is_m) {
+        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] =
+            (static_cast<u64>(p) << 32) | u;
+      }
+    }
   }
   // L list append (deg > kClusterMidDeg), wave-aggregated
   {
@@ -1884,6 +1929,582 @@ __global__ void k_edge_cut(
   }
 }
 
+// ==================== commit v2 (refine, k <= 256) ====================
+// The round-1 commit issued ~30 launches + 3 rocprim calls + 2-3 host syncs
+// per chunk (~85 ms/step control floor). v2 replaces it
+// with 4 fixed-shape launches and ZERO host syncs: a stable counting sort
+// by target block (k <= 256 distinct keys -- radix sort is overkill), then
+// ONE kernel holding the whole admission fixpoint + weight update + label
+// apply + active-set maintenance, using a hand-rolled resident-grid barrier
+// (all blocks co-resident; hipLaunchCooperativeKernel semantics without the
+// API so the launch stays graph-capturable). Admission semantics unchanged:
+// per target, proposals are admitted in rank-prefix order under the
+// greatest-fixpoint rollback of kaminpar-dist lp_refiner.cc:296-333.
+
+// Refine L-path prep: per-vertex slice counts + exclusive prefix in one
+// single-workgroup kernel (replaces k_l_sizes + a rocprim scan over l_cap).
+__global__ void k_l_prep_r(
+    const u64 *__restrict__ l_list,
+    const u32 *__restrict__ l_count,
+    const u32 *__restrict__ xadj,
+    u32 l_cap,
+    u32 *__restrict__ l_off
+) {
+  __shared__ u32 red[17];
+  const u32 count = *l_count < l_cap ? *l_count : l_cap;
+  const u32 tid = threadIdx.x;
+  u32 carry = 0;
+  for (u32 base = 0; base < count; base += blockDim.x) {
+    const u32 i = base + tid;
+    u32 v = 0;
+    if (i < count) {
+      const u32 u = static_cast<u32>(l_list[i]);
+      const u32 deg = xadj[u + 1] - xadj[u];
+      v = (deg + kLSlice - 1) / kLSlice;
+    }
+    u32 inc = v;
+    for (int off = 1; off < 64; off <<= 1) {
+      const u32 o = __shfl_up(inc, off, kWave);
+      if ((tid & 63) >= static_cast<u32>(off)) {
+        inc += o;
+      }
+    }
+    __syncthreads();
+    if ((tid & 63) == 63) {
+      red[tid >> 6] = inc;
+    }
+    __syncthreads();
+    u32 wbase = 0;
+    for (u32 w = 0; w < (tid >> 6); ++w) {
+      wbase += red[w];
+    }
+    if (i < count) {
+      l_off[i] = carry + wbase + inc - v; // exclusive
+    }
+    u32 tsum = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      tsum += red[w];
+    }
+    carry += tsum;
+    __syncthreads();
+  }
+  if (tid == 0) {
+    l_off[count] = carry;
+  }
+}
+
+// Per-wave-row histogram of proposal targets over the raw slot array.
+// Each wave owns a contiguous run of 64-position tiles (= permutation
+// units); fully-inactive units are skipped without reading their (stale)
+// slots -- the phase-A kernels freshly write every slot of an ACTIVE unit.
+__global__ void k_hist_v2(
+    u32 span,
+    u32 pos_lo,
+    u32 n,
+    u64 iter_seed,
+    u32 k,
+    u32 rows,
+    u32 tpw,
+    const Prop *__restrict__ slots,
+    const uint8_t *__restrict__ unit_active,
+    u32 *__restrict__ histT // k x rows, column-major by target
+) {
+  extern __shared__ u32 cnt_lds[];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * k;
+  for (u32 c = lane; c < k; c += kWave) {
+    cnt[c] = 0;
+  }
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 T = span >> 6;
+  const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    u32 vb = 0;
+    if (lane == 0) {
+      vb = fp(pos_lo / kmp::kUnit + t);
+    }
+    vb = __shfl(vb, 0, kWave);
+    if (!unit_active[vb]) {
+      continue;
+    }
+    const Prop pr = slots[(t << 6) + lane];
+    if (pr.to != kInvalid) {
+      atomicAdd(&cnt[pr.to], 1u);
+    }
+  }
+  for (u32 c = lane; c < k; c += kWave) {
+    histT[c * rows + row] = cnt[c];
+  }
+}
+
+// Single-workgroup exclusive scan of the k x rows histogram (k*rows <=
+// 65536 entries) + segment offsets + prefix_len init (= full admission,
+// the greatest-fixpoint starting point).
+__global__ void k_matscan_v2(
+    u32 k,
+    u32 rows,
+    const u32 *__restrict__ histT,
+    u32 *__restrict__ offT,
+    u32 *__restrict__ seg_off, // k+1
+    u32 *__restrict__ prefix_len
+) {
+  __shared__ u32 red[17];
+  const u32 entries = k * rows;
+  const u32 tid = threadIdx.x;
+  u32 carry = 0;
+  for (u32 base = 0; base < entries; base += blockDim.x) {
+    const u32 i = base + tid;
+    u32 v = i < entries ? histT[i] : 0;
+    u32 inc = v;
+    for (int off = 1; off < 64; off <<= 1) {
+      const u32 o = __shfl_up(inc, off, kWave);
+      if ((tid & 63) >= static_cast<u32>(off)) {
+        inc += o;
+      }
+    }
+    __syncthreads();
+    if ((tid & 63) == 63) {
+      red[tid >> 6] = inc;
+    }
+    __syncthreads();
+    u32 wbase = 0;
+    for (u32 w = 0; w < (tid >> 6); ++w) {
+      wbase += red[w];
+    }
+    if (i < entries) {
+      offT[i] = carry + wbase + inc - v; // exclusive
+    }
+    u32 tsum = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      tsum += red[w];
+    }
+    carry += tsum;
+    __syncthreads();
+  }
+  __threadfence_block();
+  __syncthreads();
+  for (u32 c = tid; c < k; c += blockDim.x) {
+    const u32 b = offT[c * rows];
+    const u32 e2 = (c + 1 < k) ? offT[(c + 1) * rows] : carry;
+    seg_off[c] = b;
+    prefix_len[c] = e2 - b;
+  }
+  if (tid == 0) {
+    seg_off[k] = carry;
+  }
+}
+
+// Stable scatter into target-sorted order. Each wave replays its tile run
+// in position order; within a 64-slot tile, stable per-target ranks come
+// from a 64-step shuffle waterfall (no LDS read-after-write ordering
+// assumptions: the leader's base is broadcast by shuffle). Also freezes the
+// source block of every proposal (labels16 gather, coalesced within the
+// unit) so the fixpoint never gathers labels again.
+__global__ void k_scatter_v2(
+    u32 span,
+    u32 pos_lo,
+    u32 n,
+    u64 iter_seed,
+    u32 k,
+    u32 rows,
+    u32 tpw,
+    const Prop *__restrict__ slots,
+    const uint8_t *__restrict__ unit_active,
+    const u32 *__restrict__ offT,
+    const uint16_t *__restrict__ labels16,
+    u32 *__restrict__ s_u,
+    u32 *__restrict__ s_w,
+    uint16_t *__restrict__ s_to,
+    uint16_t *__restrict__ s_b
+) {
+  extern __shared__ u32 cnt_lds[];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * k;
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  for (u32 c = lane; c < k; c += kWave) {
+    cnt[c] = offT[c * rows + row];
+  }
+  const u32 T = span >> 6;
+  const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    u32 vb = 0;
+    if (lane == 0) {
+      vb = fp(pos_lo / kmp::kUnit + t);
+    }
+    vb = __shfl(vb, 0, kWave);
+    if (!unit_active[vb]) {
+      continue;
+    }
+    const Prop pr = slots[(t << 6) + lane];
+    const u32 c = pr.to;
+    const bool valid = c != kInvalid;
+    u32 rank = 0, total = 0, leader = 0;
+    bool seen = false;
+    for (u32 j = 0; j < kWave; ++j) {
+      const u32 cj = __shfl(c, j, kWave);
+      if (valid && cj == c) {
+        ++total;
+        if (j < lane) {
+          ++rank;
+        }
+        if (!seen) {
+          leader = j;
+          seen = true;
+        }
+      }
+    }
+    u32 base0 = 0;
+    if (valid && rank == 0) {
+      base0 = cnt[c];
+      cnt[c] = base0 + total;
+    }
+    const u32 base = __shfl(base0, leader, kWave);
+    if (valid) {
+      const u32 dst = base + rank;
+      s_u[dst] = pr.u;
+      s_w[dst] = pr.w;
+      s_to[dst] = static_cast<uint16_t>(c);
+      s_b[dst] = labels16[pr.u];
+    }
+  }
+}
+
+// Resident-grid barrier: two-level arrival (8 sub-counters keyed by
+// blockIdx & 7 -> root), generation-released. Requires nblk to be a
+// multiple of 8 and every block co-resident (grid sized from occupancy).
+__device__ inline void coop_bar(u32 *bar, u32 nblk) {
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    u32 *sub = bar;
+    u32 *root = bar + 8;
+    u32 *gen = bar + 9;
+    const u32 g = __hip_atomic_load(gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    const u32 grp = blockIdx.x & 7u;
+    const u32 gsz = nblk >> 3;
+    bool last = false;
+    if (__hip_atomic_fetch_add(&sub[grp], 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT) ==
+        gsz - 1) {
+      if (__hip_atomic_fetch_add(root, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT) == 7u) {
+        for (u32 i = 0; i < 8; ++i) {
+          __hip_atomic_store(&sub[i], 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+        __hip_atomic_store(root, 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        __hip_atomic_fetch_add(gen, 1u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        last = true;
+      }
+    }
+    if (!last) {
+      while (__hip_atomic_load(gen, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT) == g) {
+        __builtin_amdgcn_s_sleep(1);
+      }
+    }
+  }
+  __syncthreads();
+}
+
+// The whole deterministic admission in ONE launch: (optional) segmented
+// prefix weights, the greatest-fixpoint rollback loop with device-side
+// convergence (changed flag double-buffered by round parity), weight
+// update, label apply, chunk active-set clear + arcs tally, and neighbour
+// activation. Restates kaminpar-dist/refinement/lp/lp_refiner.cc:296-333
+// under the deterministic schedule; bit-identical to the legacy commit.
+__global__ void k_commit_coop(
+    u32 k,
+    u32 chunk_lo,
+    u32 chunk_hi,
+    u32 n,
+    u64 iter_seed,
+    u32 has_vwgt,
+    u32 nblk,
+    const u32 *__restrict__ seg_off,
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep,
+    i64 *__restrict__ pw,
+    i64 *__restrict__ blocksums,
+    const u32 *__restrict__ s_u,
+    const u32 *__restrict__ s_w,
+    const uint16_t *__restrict__ s_to,
+    const uint16_t *__restrict__ s_b,
+    i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    u32 *__restrict__ labels,
+    uint16_t *__restrict__ labels16,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    uint8_t *__restrict__ active,
+    uint8_t *__restrict__ unit_active,
+    unsigned long long *__restrict__ arcs,
+    unsigned long long *__restrict__ moves,
+    int *__restrict__ changed2, // int[2]
+    u32 *__restrict__ bar
+) {
+  extern __shared__ unsigned long long lds64[];
+  unsigned long long *udep = lds64;                  // k entries
+  i64 *red = reinterpret_cast<i64 *>(lds64 + k);     // 17 entries
+  const u32 tid = threadIdx.x;
+  const u32 gid = blockIdx.x * blockDim.x + tid;
+  const u32 gsz = nblk * blockDim.x;
+  const u32 count = seg_off[k];
+  const u32 lane = tid & (kWave - 1);
+
+  // ---- segmented prefix weights (weighted graphs only; unweighted uses
+  // pw[i] = i+1 implicitly) ----
+  if (has_vwgt && count) {
+    const u32 rlen = (count + nblk - 1) / nblk;
+    const u32 lo = blockIdx.x * rlen < count ? blockIdx.x * rlen : count;
+    const u32 hi = lo + rlen < count ? lo + rlen : count;
+    i64 acc = 0;
+    for (u32 i = lo + tid; i < hi; i += blockDim.x) {
+      acc += s_w[i];
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      acc += __shfl_down(acc, off, kWave);
+    }
+    if (lane == 0) {
+      red[tid >> 6] = acc;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      i64 t = 0;
+      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+        t += red[w];
+      }
+      blocksums[blockIdx.x] = t;
+    }
+    coop_bar(bar, nblk);
+    if (blockIdx.x == 0 && tid == 0) {
+      i64 runsum = 0;
+      for (u32 b = 0; b < nblk; ++b) {
+        const i64 v = blocksums[b];
+        blocksums[b] = runsum;
+        runsum += v;
+      }
+    }
+    coop_bar(bar, nblk);
+    i64 carry = blocksums[blockIdx.x];
+    for (u32 base = lo; base < hi; base += blockDim.x) {
+      const u32 i = base + tid;
+      i64 v = (i < hi) ? static_cast<i64>(s_w[i]) : 0;
+      i64 inc = v;
+      for (int off = 1; off < 64; off <<= 1) {
+        const i64 o = __shfl_up(inc, off, kWave);
+        if (lane >= static_cast<u32>(off)) {
+          inc += o;
+        }
+      }
+      __syncthreads();
+      if (lane == 63) {
+        red[tid >> 6] = inc;
+      }
+      __syncthreads();
+      i64 wbase = 0;
+      for (u32 w = 0; w < (tid >> 6); ++w) {
+        wbase += red[w];
+      }
+      if (i < hi) {
+        pw[i] = carry + wbase + inc;
+      }
+      i64 tsum = 0;
+      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+        tsum += red[w];
+      }
+      carry += tsum;
+      __syncthreads();
+    }
+    coop_bar(bar, nblk);
+  }
+
+  // ---- greatest-fixpoint rollback ----
+  u32 round = 0;
+  for (;;) {
+    const u32 cs = round & 1;
+    if (blockIdx.x == 0) {
+      for (u32 c = tid; c < k; c += blockDim.x) {
+        dep[c] = 0;
+      }
+      if (tid == 0) {
+        changed2[cs] = 0;
+      }
+    }
+    coop_bar(bar, nblk);
+    for (u32 c = tid; c < k; c += blockDim.x) {
+      udep[c] = 0;
+    }
+    __syncthreads();
+    for (u32 i = gid; i < count; i += gsz) {
+      const u32 c = s_to[i];
+      if (i - seg_off[c] < prefix_len[c]) {
+        atomicAdd(&udep[s_b[i]],
+                  has_vwgt ? static_cast<unsigned long long>(s_w[i]) : 1ull);
+      }
+    }
+    __syncthreads();
+    for (u32 c = tid; c < k; c += blockDim.x) {
+      if (udep[c]) {
+        atomicAdd(&dep[c], udep[c]);
+      }
+    }
+    coop_bar(bar, nblk);
+    if (blockIdx.x == 0) {
+      for (u32 c = tid; c < k; c += blockDim.x) {
+        const u32 old_len = prefix_len[c];
+        if (old_len == 0) {
+          continue;
+        }
+        const u32 b = seg_off[c];
+        const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(dep[c]);
+        u32 nl;
+        if (!has_vwgt) {
+          nl = capacity <= 0 ? 0u
+                             : (capacity >= static_cast<i64>(old_len) ? old_len
+                                                                      : static_cast<u32>(capacity));
+        } else {
+          const i64 base0 = b ? pw[b - 1] : 0;
+          u32 lo2 = 0, hi2 = old_len;
+          while (lo2 < hi2) {
+            const u32 mid = (lo2 + hi2 + 1) >> 1;
+            if (pw[b + mid - 1] - base0 <= capacity) {
+              lo2 = mid;
+            } else {
+              hi2 = mid - 1;
+            }
+          }
+          nl = lo2;
+        }
+        if (nl < old_len) {
+          prefix_len[c] = nl;
+          __hip_atomic_store(&changed2[cs], 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+      }
+    }
+    coop_bar(bar, nblk);
+    if (!__hip_atomic_load(&changed2[cs], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+      break;
+    }
+    ++round;
+  }
+
+  // ---- weights + moves (block 0) ----
+  if (blockIdx.x == 0) {
+    unsigned long long mv = 0;
+    for (u32 c = tid; c < k; c += blockDim.x) {
+      const u32 len = prefix_len[c];
+      mv += len;
+      i64 arr = 0;
+      if (len) {
+        if (!has_vwgt) {
+          arr = static_cast<i64>(len);
+        } else {
+          const u32 b = seg_off[c];
+          arr = pw[b + len - 1] - (b ? pw[b - 1] : 0);
+        }
+      }
+      const i64 delta = arr - static_cast<i64>(dep[c]);
+      if (delta) {
+        weights[c] += delta;
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      mv += __shfl_down(mv, off, kWave);
+    }
+    __syncthreads();
+    if (lane == 0) {
+      reinterpret_cast<unsigned long long *>(red)[tid >> 6] = mv;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      unsigned long long t = 0;
+      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+        t += reinterpret_cast<unsigned long long *>(red)[w];
+      }
+      if (t) {
+        atomicAdd(moves, t);
+      }
+    }
+  }
+
+  // ---- apply admitted labels (all blocks) ----
+  for (u32 i = gid; i < count; i += gsz) {
+    const u32 c = s_to[i];
+    if (i - seg_off[c] < prefix_len[c]) {
+      const u32 u = s_u[i];
+      labels[u] = c;
+      labels16[u] = static_cast<uint16_t>(c);
+    }
+  }
+
+  // ---- clear active for the chunk's processed set + arcs tally ----
+  {
+    unsigned long long my = 0;
+    const u32 waves_total = gsz >> 6;
+    const u32 wave_gid = gid >> 6;
+    const u32 units = (chunk_hi - chunk_lo) >> 6;
+    const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+    for (u32 uix = wave_gid; uix < units; uix += waves_total) {
+      u32 vb = 0;
+      if (lane == 0) {
+        vb = fp(chunk_lo / kmp::kUnit + uix);
+      }
+      vb = __shfl(vb, 0, kWave);
+      if (!unit_active[vb]) {
+        continue;
+      }
+      const u32 u = vb * kmp::kUnit + lane;
+      if (u < n) {
+        const u32 deg = xadj[u + 1] - xadj[u];
+        if (active[u]) {
+          my += deg;
+          active[u] = 0;
+        }
+      }
+      if (lane == 0) {
+        unit_active[vb] = 0;
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      my += __shfl_down(my, off, kWave);
+    }
+    __syncthreads(); // red reuse across sections
+    if (lane == 0) {
+      reinterpret_cast<unsigned long long *>(red)[tid >> 6] = my;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      unsigned long long t = 0;
+      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+        t += reinterpret_cast<unsigned long long *>(red)[w];
+      }
+      if (t) {
+        atomicAdd(arcs, t);
+      }
+    }
+  }
+  coop_bar(bar, nblk);
+
+  // ---- activate neighbours of admitted movers (wave per entry) ----
+  {
+    const u32 waves_total = gsz >> 6;
+    for (u32 i = gid >> 6; i < count; i += waves_total) {
+      const u32 c = s_to[i];
+      if (i - seg_off[c] >= prefix_len[c]) {
+        continue;
+      }
+      const u32 u = s_u[i];
+      const u32 row = xadj[u];
+      const u32 deg = xadj[u + 1] - row;
+      for (u32 e2 = lane; e2 < deg; e2 += kWave) {
+        const u32 v = adjncy[row + e2];
+        active[v] = 1;
+        unit_active[v >> 6] = 1;
+      }
+    }
+  }
+}
+
 } // namespace
 
 // ================================================================ engine
@@ -1915,14 +2536,11 @@ struct kmp_lp_t {
 
   // phase buffers
   Prop *d_slots = nullptr; // C
-  Prop *d_props = nullptr; // C (compacted; single-GPU commit input)
-  u64 *d_m_slots = nullptr; // C (per-position M-candidate records)
-  u64 *d_m_list = nullptr;  // C (compacted)
-  u32 *d_m_count = nullptr;
-  void *d_m_select_temp = nullptr;
-  size_t m_select_temp_bytes = 0;
+  Prop *d_props = nullptr; // C (compacted; legacy/sharded commit input)
+  u64 *d_m_list = nullptr;  // C (wave-aggregated append by the S kernels)
+  u32 *d_m_count = nullptr; // base of a 2-u32 alloc; [1] is d_l_count
   u64 *d_l_list = nullptr; // C
-  u32 *d_l_count = nullptr;
+  u32 *d_l_count = nullptr; // = d_m_count + 1
   u32 *d_l_off = nullptr;   // l_cap + 1 (slice prefix)
   u32 *d_l_sizes = nullptr; // l_cap + 1 (slice counts, scan input)
   void *d_lscan_temp = nullptr;
@@ -1949,6 +2567,22 @@ struct kmp_lp_t {
   int *d_changed = nullptr;
   u32 *d_admitted_flags = nullptr;
   unsigned long long *d_cut = nullptr;
+
+  // commit v2 (refine, k <= 256): counting sort + cooperative fixpoint.
+  // Replaces the per-chunk radix sort / scan-by-key / host fixpoint loop
+  // with 4 fixed-shape launches and ZERO host syncs per chunk.
+  u32 *d_s_u = nullptr;       // C (sorted-by-target source vertices)
+  u32 *d_s_w = nullptr;       // C (their node weights)
+  uint16_t *d_s_to = nullptr; // C (target block per sorted index)
+  uint16_t *d_s_b = nullptr;  // C (source block, frozen at scatter)
+  u32 *d_histT = nullptr;     // k x rows column-major histogram
+  u32 *d_offT = nullptr;      // its exclusive scan
+  u32 *d_seg_off = nullptr;   // k+1 segment offsets
+  u32 *d_bar = nullptr;       // grid barrier state (sub[8], root, gen)
+  i64 *d_blocksums = nullptr; // coop pw-scan block partials
+  u32 coop_nblk = 0;          // resident-guaranteed grid for k_commit_coop
+  u32 rows_v2 = 0;            // histogram rows for current k
+  u32 max_deg = 0;            // max degree (gates the L-path launches)
 
   // clustering state
   u32 *d_favored = nullptr;      // n (two-hop favored clusters)
@@ -1977,6 +2611,7 @@ struct kmp_lp_t {
   int balance = 0; // balance mode: overloaded vertices lose "stay" (select)
   std::vector<i64> maxw_host; // refine caps (host copy, for the fallback)
   i64 maxw_uniform = 0;
+  int mp_count = 0; // CUs (for the resident-grid commit kernel)
 
   // pinned host mirrors
   u32 *h_count = nullptr;
@@ -2074,11 +2709,22 @@ void engine_alloc_common(kmp_lp_t *e) {
   const u32 C = e->C;
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
-  HIP_CHECK(hipMalloc(&e->d_m_slots, sizeof(u64) * C));
   HIP_CHECK(hipMalloc(&e->d_m_list, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32)));
+  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32) * 2)); // [0]=m, [1]=l
+  e->d_l_count = e->d_m_count + 1;
   HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_l_count, sizeof(u32)));
+
+  // commit v2 buffers (sized for the largest supported k = 256 rows layout)
+  HIP_CHECK(hipMalloc(&e->d_s_u, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_s_w, sizeof(u32) * C));
+  HIP_CHECK(hipMalloc(&e->d_s_to, sizeof(uint16_t) * C));
+  HIP_CHECK(hipMalloc(&e->d_s_b, sizeof(uint16_t) * C));
+  HIP_CHECK(hipMalloc(&e->d_histT, sizeof(u32) * 65536));
+  HIP_CHECK(hipMalloc(&e->d_offT, sizeof(u32) * 65536));
+  HIP_CHECK(hipMalloc(&e->d_seg_off, sizeof(u32) * 257));
+  HIP_CHECK(hipMalloc(&e->d_bar, sizeof(u32) * 16));
+  HIP_CHECK(hipMemsetAsync(e->d_bar, 0, sizeof(u32) * 16, e->stream));
+  HIP_CHECK(hipMalloc(&e->d_blocksums, sizeof(i64) * 512));
   HIP_CHECK(hipMalloc(&e->d_prop_count, sizeof(u32)));
   HIP_CHECK(hipMalloc(&e->d_arcs, sizeof(unsigned long long)));
   HIP_CHECK(hipMalloc(&e->d_moves, sizeof(unsigned long long)));
@@ -2089,7 +2735,7 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_sort_vals[1], sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_sw, sizeof(i64) * C));
   HIP_CHECK(hipMalloc(&e->d_pw, sizeof(i64) * C));
-  HIP_CHECK(hipMalloc(&e->d_changed, sizeof(int)));
+  HIP_CHECK(hipMalloc(&e->d_changed, sizeof(int) * 2)); // [round-parity] for coop
   HIP_CHECK(hipMalloc(&e->d_admitted_flags, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_cut, sizeof(unsigned long long)));
 
@@ -2106,10 +2752,6 @@ void engine_alloc_common(kmp_lp_t *e) {
       nullptr, e->select_temp_bytes, e->d_slots, e->d_props, e->d_prop_count, C, PropValid()
   ));
   HIP_CHECK(hipMalloc(&e->d_select_temp, e->select_temp_bytes));
-  HIP_CHECK(rocprim::select(
-      nullptr, e->m_select_temp_bytes, e->d_m_slots, e->d_m_list, e->d_m_count, C, CandValid()
-  ));
-  HIP_CHECK(hipMalloc(&e->d_m_select_temp, e->m_select_temp_bytes));
 
   HIP_CHECK(hipHostMalloc(&e->h_count, sizeof(u32) * 2));
   HIP_CHECK(hipHostMalloc(&e->h_changed, sizeof(int)));
@@ -2120,12 +2762,176 @@ void engine_alloc_common(kmp_lp_t *e) {
 // Isolated-vertex scan used by the clusterer's isolated-node handling
 // (lp_clusterer.cc cluster_isolated_nodes semantics).
 void engine_scan_isolated(kmp_lp_t *e, const u32 *xadj, const i32 *vwgt) {
+  u32 maxd = 0;
   for (u32 u = 0; u < e->n; ++u) {
-    if (xadj[u + 1] == xadj[u]) {
+    const u32 deg = xadj[u + 1] - xadj[u];
+    if (deg > maxd) {
+      maxd = deg;
+    }
+    if (deg == 0) {
       e->isolated.push_back(u);
       e->iso_weights.push_back(vwgt ? vwgt[u] : 1);
     }
   }
+  e->max_deg = maxd;
+}
+
+hipEvent_t ev_one(kmp_lp_t *e) {
+  if (e->ev_used >= e->ev_pool.size()) {
+    hipEvent_t x;
+    HIP_CHECK(hipEventCreate(&x));
+    e->ev_pool.push_back(x);
+  }
+  return e->ev_pool[e->ev_used++];
+}
+
+// ---- commit v2 driver (refine, k <= 256; zero host syncs per chunk) ----
+
+bool v2_eligible(const kmp_lp_t *e) {
+  return !e->clusterer && !e->balance && e->k <= 256 && e->coop_nblk >= 8;
+}
+
+// Phase A without slot memsets: S/M/L freshly write every slot of an
+// active unit; inactive units are gated downstream by unit_active.
+void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
+  const u64 iseed = iter_seed_of(e->seed, iter);
+  const u32 chunk_base = pos_lo;
+  const u32 span = pos_hi - pos_lo;
+  const u32 threads = 256;
+  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 2, e->stream));
+  hipLaunchKernelGGL(
+      k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+      dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, 0u, kInvalid,
+      0xFFFFFFFFu, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
+      e->d_maxw, e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_list,
+      e->d_m_count, e->d_l_list, e->d_l_count
+  );
+  LAUNCH_CHECK();
+  {
+    const size_t lds =
+        static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
+    auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
+    hipLaunchKernelGGL(
+        kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed, 0u,
+        kInvalid, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
+        e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+    );
+    LAUNCH_CHECK();
+  }
+  if (e->max_deg > kMidDeg) {
+    hipLaunchKernelGGL(
+        k_l_prep_r, dim3(1), dim3(256), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
+        e->l_cap, e->d_l_off
+    );
+    LAUNCH_CHECK();
+    const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
+    {
+      auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+      hipLaunchKernelGGL(
+          kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
+          e->d_adjwgt, e->d_labels16, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+      );
+      LAUNCH_CHECK();
+    }
+    hipLaunchKernelGGL(
+        k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, 0u, kInvalid, pos_lo, chunk_base,
+        iseed, e->k, e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
+        e->d_l_count, e->l_cap, e->d_l_gains, e->d_slots
+    );
+    LAUNCH_CHECK();
+    {
+      const size_t lds =
+          ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
+          16 * sizeof(i64);
+      auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
+      hipLaunchKernelGGL(
+          kern, dim3(512), dim3(256), lds, e->stream, 0u, kInvalid, pos_lo, chunk_base, iseed,
+          e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16,
+          e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+      );
+      LAUNCH_CHECK();
+    }
+  }
+}
+
+void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
+  const u64 iseed = iter_seed_of(e->seed, iter);
+  const u32 span = pos_hi - pos_lo;
+  const u32 rows = e->rows_v2;
+  const u32 T = span >> 6;
+  const u32 tpw = (T + rows - 1) / rows;
+  const u32 threads = 256;
+  const size_t lds_h = static_cast<size_t>(threads / kWave) * e->k * sizeof(u32);
+  hipLaunchKernelGGL(
+      k_hist_v2, dim3(rows / 4), dim3(threads), lds_h, e->stream, span, pos_lo, e->n, iseed,
+      e->k, rows, tpw, e->d_slots, e->d_unit_active, e->d_histT
+  );
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_matscan_v2, dim3(1), dim3(1024), 0, e->stream, e->k, rows, e->d_histT, e->d_offT,
+      e->d_seg_off, e->d_prefix_len
+  );
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_scatter_v2, dim3(rows / 4), dim3(threads), lds_h, e->stream, span, pos_lo, e->n, iseed,
+      e->k, rows, tpw, e->d_slots, e->d_unit_active, e->d_offT, e->d_labels16, e->d_s_u,
+      e->d_s_w, e->d_s_to, e->d_s_b
+  );
+  LAUNCH_CHECK();
+  const size_t lds_c = static_cast<size_t>(e->k + 17) * sizeof(unsigned long long);
+  hipLaunchKernelGGL(
+      k_commit_coop, dim3(e->coop_nblk), dim3(threads), lds_c, e->stream, e->k, pos_lo, pos_hi,
+      e->n, iseed, static_cast<u32>(e->has_vwgt ? 1 : 0), e->coop_nblk, e->d_seg_off,
+      e->d_prefix_len, e->d_dep, e->d_pw, e->d_blocksums, e->d_s_u, e->d_s_w, e->d_s_to,
+      e->d_s_b, e->d_weights, e->d_maxw, e->d_labels, e->d_labels16, e->d_xadj, e->d_adjncy,
+      e->d_active, e->d_unit_active, e->d_arcs, e->d_moves, e->d_changed, e->d_bar
+  );
+  LAUNCH_CHECK();
+}
+
+// One sweep = 64 chunk trains enqueued back-to-back with no host sync; the
+// single sync per sweep reads the device move counter (early-exit check)
+// and settles the per-chunk timing events.
+i64 run_sweeps_v2(kmp_lp_t *e, int iters) {
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
+  sync_spin(e);
+  unsigned long long last = e->h_moves[0];
+  u64 total = 0;
+  for (int iter = 0; iter < iters; ++iter) {
+    e->ev_used = 0;
+    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
+      const u32 pos_lo = chunk * e->C;
+      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
+      if (pos_lo >= pos_hi) {
+        continue;
+      }
+      hipEvent_t a0 = ev_one(e), a1 = ev_one(e), c1 = ev_one(e);
+      HIP_CHECK(hipEventRecord(a0, e->stream));
+      phase_a_v2(e, iter, pos_lo, pos_hi);
+      HIP_CHECK(hipEventRecord(a1, e->stream));
+      commit_v2(e, iter, pos_lo, pos_hi);
+      HIP_CHECK(hipEventRecord(c1, e->stream));
+    }
+    HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, e->stream));
+    sync_spin(e);
+    for (size_t i = 0; i + 2 < e->ev_used; i += 3) {
+      float ms = 0;
+      HIP_CHECK(hipEventElapsedTime(&ms, e->ev_pool[i], e->ev_pool[i + 1]));
+      e->phase_a_ms += ms;
+      HIP_CHECK(hipEventElapsedTime(&ms, e->ev_pool[i + 1], e->ev_pool[i + 2]));
+      e->commit_ms += ms;
+    }
+    const unsigned long long cur = e->h_moves[1];
+    const u64 sweep_moves = cur - last;
+    last = cur;
+    total += sweep_moves;
+    if (sweep_moves == 0) {
+      break;
+    }
+  }
+  return static_cast<i64>(total);
 }
 
 } // namespace
@@ -2157,6 +2963,13 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   e->has_adjwgt = kmp_graph_adjwgt(g) != nullptr;
   HIP_CHECK(hipStreamCreate(&e->stream));
   HIP_CHECK(hipEventCreateWithFlags(&e->sync_ev, hipEventDisableTiming));
+  {
+    int dev = 0;
+    HIP_CHECK(hipGetDevice(&dev));
+    hipDeviceProp_t props;
+    HIP_CHECK(hipGetDeviceProperties(&props, dev));
+    e->mp_count = props.multiProcessorCount;
+  }
 
   HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
   HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
@@ -2192,14 +3005,16 @@ void kmp_lp_free(kmp_lp_t *e) {
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
-                  (void *)e->d_slots, (void *)e->d_props, (void *)e->d_m_slots,
-                  (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m_select_temp,
-                  (void *)e->d_l_list, (void *)e->d_l_count, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
+                  (void *)e->d_slots, (void *)e->d_props,
+                  (void *)e->d_m_list, (void *)e->d_m_count,
+                  (void *)e->d_l_list, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
                   (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_flags,
-                  (void *)e->d_cut}) {
+                  (void *)e->d_cut, (void *)e->d_s_u, (void *)e->d_s_w, (void *)e->d_s_to,
+                  (void *)e->d_s_b, (void *)e->d_histT, (void *)e->d_offT, (void *)e->d_seg_off,
+                  (void *)e->d_bar, (void *)e->d_blocksums}) {
     if (p) {
       (void)hipFree(p);
     }
@@ -2310,6 +3125,25 @@ int kmp_lp_refine_begin(
     );
     LAUNCH_CHECK();
   }
+  // commit v2 setup: histogram row count + resident-guaranteed grid for the
+  // single-launch commit kernel (falls back to the legacy commit if the
+  // occupancy query says the grid cannot be made co-resident)
+  e->rows_v2 = (k <= 64) ? 1024u : 256u;
+  e->coop_nblk = 0;
+  if (k <= 256) {
+    int occ = 0;
+    const size_t lds_c = static_cast<size_t>(k + 17) * sizeof(unsigned long long);
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ, k_commit_coop, 256, lds_c) ==
+            hipSuccess &&
+        occ > 0 && e->mp_count > 0) {
+      u64 nb = static_cast<u64>(occ) * static_cast<u64>(e->mp_count);
+      if (nb > 256) {
+        nb = 256;
+      }
+      nb &= ~7ull;
+      e->coop_nblk = static_cast<u32>(nb);
+    }
+  }
   HIP_CHECK(hipStreamSynchronize(e->stream));
   return 0;
 }
@@ -2327,11 +3161,10 @@ i64 kmp_lp_phase_a(
   const u32 threads = 256;
   const u32 max_degree = 0xFFFFFFFFu;
 
-  HIP_CHECK(hipMemsetAsync(e->d_l_count, 0, sizeof(u32), e->stream));
-  // pre-mark every slot invalid (kernels only write actual proposals; a
-  // skipped unit costs one byte read instead of 64 slot writes)
+  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 2, e->stream)); // m+l counts
+  // pre-mark every slot invalid (the legacy/sharded/clusterer commit reads
+  // compacted proposals; the clusterer S kernel does not always-write)
   HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * span, e->stream));
-  HIP_CHECK(hipMemsetAsync(e->d_m_slots, 0xFF, sizeof(u64) * span, e->stream));
 
   hipEvent_t ev0, ev1;
   e->ev_pair(ev0, ev1);
@@ -2360,18 +3193,12 @@ i64 kmp_lp_phase_a(
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
         static_cast<u32>(e->balance), fallback, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_slots, e->d_l_list,
-        e->d_l_count
+        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_list, e->d_m_count,
+        e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
-    // compact the M work list (stable select; no append atomics), then one
-    // wave per listed vertex (grid-stride)
+    // one wave per M-listed vertex (grid-stride; list appended by k_phase_s)
     {
-      size_t mtb = e->m_select_temp_bytes;
-      HIP_CHECK(rocprim::select(
-          e->d_m_select_temp, mtb, e->d_m_slots, e->d_m_list, e->d_m_count, span, CandValid(),
-          e->stream
-      ));
       const size_t lds =
           static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
       auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
@@ -2434,16 +3261,11 @@ i64 kmp_lp_phase_a(
         k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_m_slots, e->d_l_list,
-        e->d_l_count
+        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_m_list, e->d_m_count,
+        e->d_l_list, e->d_l_count
     );
     LAUNCH_CHECK();
     {
-      size_t mtb = e->m_select_temp_bytes;
-      HIP_CHECK(rocprim::select(
-          e->d_m_select_temp, mtb, e->d_m_slots, e->d_m_list, e->d_m_count, span, CandValid(),
-          e->stream
-      ));
       const size_t lds = static_cast<size_t>(threads / kWave) * 2 * kHashSlots * sizeof(u32);
       auto *kern = e->has_adjwgt ? k_phase_m_c<false> : k_phase_m_c<true>;
       hipLaunchKernelGGL(
@@ -2778,9 +3600,12 @@ int kmp_lp_reset(kmp_lp_t *e) {
 }
 
 // Run the LP sweeps on the current device state (the timed region: all data
-// resident in HBM, no host transfers besides the per-chunk control syncs).
-// Returns total committed moves.
+// resident in HBM; the v2 path does ONE host sync per sweep, the legacy
+// path syncs per chunk). Returns total committed moves.
 i64 kmp_lp_run_sweeps(kmp_lp_t *e, int iters) {
+  if (v2_eligible(e)) {
+    return run_sweeps_v2(e, iters);
+  }
   u64 total_moves = 0;
   for (int iter = 0; iter < iters; ++iter) {
     u64 sweep_moves = 0;
@@ -2835,27 +3660,8 @@ i64 kmp_lp_refine(
   if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
     return -1;
   }
-  for (int iter = 0; iter < iters; ++iter) {
-    u64 sweep_moves = 0;
-    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
-      const u32 pos_lo = chunk * e->C;
-      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
-      if (pos_lo >= pos_hi) {
-        continue;
-      }
-      const i64 cnt = kmp_lp_phase_a(e, iter, chunk, pos_lo, pos_hi, e->d_props, e->C);
-      if (cnt < 0) {
-        return -1;
-      }
-      const i64 mv = kmp_lp_commit(e, iter, chunk, e->d_props, static_cast<u32>(cnt));
-      if (mv < 0) {
-        return -1;
-      }
-      sweep_moves += mv;
-    }
-    if (sweep_moves == 0) {
-      break;
-    }
+  if (kmp_lp_run_sweeps(e, iters) < 0) {
+    return -1;
   }
   return kmp_lp_refine_end(e, partition, stats);
 }
@@ -3446,6 +4252,13 @@ i64 kmp_contract_engine(
   e2->has_adjwgt = true;
   HIP_CHECK(hipStreamCreate(&e2->stream));
   HIP_CHECK(hipEventCreateWithFlags(&e2->sync_ev, hipEventDisableTiming));
+  {
+    int dev = 0;
+    HIP_CHECK(hipGetDevice(&dev));
+    hipDeviceProp_t props;
+    HIP_CHECK(hipGetDeviceProperties(&props, dev));
+    e2->mp_count = props.multiProcessorCount;
+  }
   e2->d_xadj = o.d_cxadj;
   e2->d_adjncy = o.d_cadj;
   e2->d_vwgt = o.d_cvw;
