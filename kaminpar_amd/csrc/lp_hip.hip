@@ -2198,8 +2198,14 @@ __device__ inline void coop_bar(u32 *bar, u32 nblk) {
       }
     }
     if (!last) {
+      // deadman: if residency was ever violated this spin would hang the
+      // device -- trap instead so the failure is a visible abort
+      u32 spins = 0;
       while (__hip_atomic_load(gen, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT) == g) {
         __builtin_amdgcn_s_sleep(1);
+        if (++spins > 400000000u) {
+          __builtin_trap();
+        }
       }
     }
   }
