@@ -2048,7 +2048,9 @@ __global__ void k_matscan_v2(
     const u32 *__restrict__ histT,
     u32 *__restrict__ offT,
     u32 *__restrict__ seg_off, // k+1
-    u32 *__restrict__ prefix_len
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep, // 2*k (double-buffered, pre-zeroed)
+    int *__restrict__ changed2            // int[2]
 ) {
   __shared__ u32 red[17];
   const u32 entries = k * rows;
@@ -2090,9 +2092,13 @@ __global__ void k_matscan_v2(
     const u32 e2 = (c + 1 < k) ? offT[(c + 1) * rows] : carry;
     seg_off[c] = b;
     prefix_len[c] = e2 - b;
+    dep[c] = 0;
+    dep[k + c] = 0;
   }
   if (tid == 0) {
     seg_off[k] = carry;
+    changed2[0] = 0;
+    changed2[1] = 0;
   }
 }
 
@@ -2173,41 +2179,35 @@ __global__ void k_scatter_v2(
   }
 }
 
-// Resident-grid barrier: two-level arrival (8 sub-counters keyed by
-// blockIdx & 7 -> root), generation-released. Requires nblk to be a
-// multiple of 8 and every block co-resident (grid sized from occupancy).
+// Resident-grid barrier: flat arrival counter + generation release, relaxed
+// agent-scope atomics bracketed by __threadfence (measured the fastest
+// variant on gfx950: ~3 us at 64 blocks vs ~20 us for acq_rel two-level --
+// tools/bar_bench.hip). Requires every block co-resident (grid sized from
+// occupancy, capped small since the commit stages are latency- not
+// bandwidth-bound).
 __device__ inline void coop_bar(u32 *bar, u32 nblk) {
   __syncthreads();
   if (threadIdx.x == 0) {
-    u32 *sub = bar;
     u32 *root = bar + 8;
     u32 *gen = bar + 9;
+    __threadfence(); // publish this block's writes before arrival
     const u32 g = __hip_atomic_load(gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    const u32 grp = blockIdx.x & 7u;
-    const u32 gsz = nblk >> 3;
-    bool last = false;
-    if (__hip_atomic_fetch_add(&sub[grp], 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT) ==
-        gsz - 1) {
-      if (__hip_atomic_fetch_add(root, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT) == 7u) {
-        for (u32 i = 0; i < 8; ++i) {
-          __hip_atomic_store(&sub[i], 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        }
-        __hip_atomic_store(root, 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        __hip_atomic_fetch_add(gen, 1u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-        last = true;
-      }
-    }
-    if (!last) {
+    if (__hip_atomic_fetch_add(root, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) ==
+        nblk - 1) {
+      __hip_atomic_store(root, 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_fetch_add(gen, 1u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+    } else {
       // deadman: if residency was ever violated this spin would hang the
       // device -- trap instead so the failure is a visible abort
       u32 spins = 0;
-      while (__hip_atomic_load(gen, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT) == g) {
+      while (__hip_atomic_load(gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) == g) {
         __builtin_amdgcn_s_sleep(1);
         if (++spins > 400000000u) {
           __builtin_trap();
         }
       }
     }
+    __threadfence(); // see every other block's published writes
   }
   __syncthreads();
 }
@@ -2324,19 +2324,15 @@ __global__ void k_commit_coop(
     coop_bar(bar, nblk);
   }
 
-  // ---- greatest-fixpoint rollback ----
+  // ---- greatest-fixpoint rollback (2 barriers per round) ----
+  // dep is double-buffered by round parity: round r accumulates into slot
+  // r&1 (pre-zeroed by k_matscan_v2 / the previous round's cutoff stage),
+  // so the zero stage costs no extra barrier.
   u32 round = 0;
+  u32 fin = 0;
   for (;;) {
-    const u32 cs = round & 1;
-    if (blockIdx.x == 0) {
-      for (u32 c = tid; c < k; c += blockDim.x) {
-        dep[c] = 0;
-      }
-      if (tid == 0) {
-        changed2[cs] = 0;
-      }
-    }
-    coop_bar(bar, nblk);
+    const u32 A = round & 1;
+    unsigned long long *depA = dep + static_cast<size_t>(A) * k;
     for (u32 c = tid; c < k; c += blockDim.x) {
       udep[c] = 0;
     }
@@ -2351,18 +2347,20 @@ __global__ void k_commit_coop(
     __syncthreads();
     for (u32 c = tid; c < k; c += blockDim.x) {
       if (udep[c]) {
-        atomicAdd(&dep[c], udep[c]);
+        atomicAdd(&depA[c], udep[c]);
       }
     }
     coop_bar(bar, nblk);
     if (blockIdx.x == 0) {
+      unsigned long long *depB = dep + static_cast<size_t>(A ^ 1) * k;
       for (u32 c = tid; c < k; c += blockDim.x) {
+        depB[c] = 0; // pre-zero the next round's slot
         const u32 old_len = prefix_len[c];
         if (old_len == 0) {
           continue;
         }
         const u32 b = seg_off[c];
-        const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(dep[c]);
+        const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(depA[c]);
         u32 nl;
         if (!has_vwgt) {
           nl = capacity <= 0 ? 0u
@@ -2383,16 +2381,21 @@ __global__ void k_commit_coop(
         }
         if (nl < old_len) {
           prefix_len[c] = nl;
-          __hip_atomic_store(&changed2[cs], 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+          __hip_atomic_store(&changed2[A], 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         }
+      }
+      if (tid == 0) {
+        changed2[A ^ 1] = 0; // pre-zero the next round's flag
       }
     }
     coop_bar(bar, nblk);
-    if (!__hip_atomic_load(&changed2[cs], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+    fin = A;
+    if (!__hip_atomic_load(&changed2[A], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
       break;
     }
     ++round;
   }
+  unsigned long long *depF = dep + static_cast<size_t>(fin) * k;
 
   // ---- weights + moves (block 0) ----
   if (blockIdx.x == 0) {
@@ -2409,7 +2412,7 @@ __global__ void k_commit_coop(
           arr = pw[b + len - 1] - (b ? pw[b - 1] : 0);
         }
       }
-      const i64 delta = arr - static_cast<i64>(dep[c]);
+      const i64 delta = arr - static_cast<i64>(depF[c]);
       if (delta) {
         weights[c] += delta;
       }
@@ -2660,14 +2663,17 @@ u32 ceil_div(u64 a, u64 b) { return static_cast<u32>((a + b - 1) / b); }
 void sync_spin(kmp_lp_t *e);
 
 void engine_alloc_k_buffers(kmp_lp_t *e, u32 k_or_n) {
+  // dep is double-buffered by the v2 commit's fixpoint (round parity);
+  // legacy paths use the first k_or_n entries only
+  const u64 dep_n = static_cast<u64>(k_or_n) * (k_or_n <= 256 ? 2 : 1);
   HIP_CHECK(hipMalloc(&e->d_seg_begin, sizeof(u32) * k_or_n));
   HIP_CHECK(hipMalloc(&e->d_seg_end, sizeof(u32) * k_or_n));
   HIP_CHECK(hipMalloc(&e->d_prefix_len, sizeof(u32) * k_or_n));
-  HIP_CHECK(hipMalloc(&e->d_dep, sizeof(unsigned long long) * k_or_n));
+  HIP_CHECK(hipMalloc(&e->d_dep, sizeof(unsigned long long) * dep_n));
   HIP_CHECK(hipMemsetAsync(e->d_seg_begin, 0, sizeof(u32) * k_or_n, e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_seg_end, 0, sizeof(u32) * k_or_n, e->stream));
   HIP_CHECK(hipMemsetAsync(e->d_prefix_len, 0, sizeof(u32) * k_or_n, e->stream));
-  HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * k_or_n, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * dep_n, e->stream));
 }
 
 void engine_free_k_buffers(kmp_lp_t *e) {
@@ -2875,7 +2881,7 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
       k_matscan_v2, dim3(1), dim3(1024), 0, e->stream, e->k, rows, e->d_histT, e->d_offT,
-      e->d_seg_off, e->d_prefix_len
+      e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
@@ -3133,8 +3139,10 @@ int kmp_lp_refine_begin(
   }
   // commit v2 setup: histogram row count + resident-guaranteed grid for the
   // single-launch commit kernel (falls back to the legacy commit if the
-  // occupancy query says the grid cannot be made co-resident)
-  e->rows_v2 = (k <= 64) ? 1024u : 256u;
+  // occupancy query says the grid cannot be made co-resident). The grid is
+  // deliberately SMALL (64 blocks): the commit stages are latency-bound and
+  // the grid barrier cost grows with the block count (tools/bar_bench.hip).
+  e->rows_v2 = 256u;
   e->coop_nblk = 0;
   if (k <= 256) {
     int occ = 0;
@@ -3143,8 +3151,8 @@ int kmp_lp_refine_begin(
             hipSuccess &&
         occ > 0 && e->mp_count > 0) {
       u64 nb = static_cast<u64>(occ) * static_cast<u64>(e->mp_count);
-      if (nb > 256) {
-        nb = 256;
+      if (nb > 64) {
+        nb = 64;
       }
       nb &= ~7ull;
       e->coop_nblk = static_cast<u32>(nb);
