@@ -2220,10 +2220,6 @@ __device__ inline void coop_bar(u32 *bar, u32 nblk) {
 // under the deterministic schedule; bit-identical to the legacy commit.
 __global__ void k_commit_coop(
     u32 k,
-    u32 chunk_lo,
-    u32 chunk_hi,
-    u32 n,
-    u64 iter_seed,
     u32 has_vwgt,
     u32 nblk,
     const u32 *__restrict__ seg_off,
@@ -2231,19 +2227,11 @@ __global__ void k_commit_coop(
     unsigned long long *__restrict__ dep,
     i64 *__restrict__ pw,
     i64 *__restrict__ blocksums,
-    const u32 *__restrict__ s_u,
     const u32 *__restrict__ s_w,
     const uint16_t *__restrict__ s_to,
     const uint16_t *__restrict__ s_b,
     i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    u32 *__restrict__ labels,
-    uint16_t *__restrict__ labels16,
-    const u32 *__restrict__ xadj,
-    const u32 *__restrict__ adjncy,
-    uint8_t *__restrict__ active,
-    uint8_t *__restrict__ unit_active,
-    unsigned long long *__restrict__ arcs,
     unsigned long long *__restrict__ moves,
     int *__restrict__ changed2, // int[2]
     u32 *__restrict__ bar
@@ -2436,8 +2424,24 @@ __global__ void k_commit_coop(
     }
   }
 
-  // ---- apply admitted labels (all blocks) ----
-  for (u32 i = gid; i < count; i += gsz) {
+  // apply / clear-active / activation run as separate full-grid kernels
+  // (they are bandwidth-bound; a launch boundary is cheaper than starving
+  // them on the small resident grid -- measured, tools/bar_bench.hip).
+}
+
+// Apply admitted labels (full grid, count read from device).
+__global__ void k_apply_v2(
+    u32 k,
+    const u32 *__restrict__ seg_off,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ s_u,
+    const uint16_t *__restrict__ s_to,
+    u32 *__restrict__ labels,
+    uint16_t *__restrict__ labels16
+) {
+  const u32 count = seg_off[k];
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 i = blockIdx.x * blockDim.x + threadIdx.x; i < count; i += stride) {
     const u32 c = s_to[i];
     if (i - seg_off[c] < prefix_len[c]) {
       const u32 u = s_u[i];
@@ -2445,71 +2449,35 @@ __global__ void k_commit_coop(
       labels16[u] = static_cast<uint16_t>(c);
     }
   }
+}
 
-  // ---- clear active for the chunk's processed set + arcs tally ----
-  {
-    unsigned long long my = 0;
-    const u32 waves_total = gsz >> 6;
-    const u32 wave_gid = gid >> 6;
-    const u32 units = (chunk_hi - chunk_lo) >> 6;
-    const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
-    for (u32 uix = wave_gid; uix < units; uix += waves_total) {
-      u32 vb = 0;
-      if (lane == 0) {
-        vb = fp(chunk_lo / kmp::kUnit + uix);
-      }
-      vb = __shfl(vb, 0, kWave);
-      if (!unit_active[vb]) {
-        continue;
-      }
-      const u32 u = vb * kmp::kUnit + lane;
-      if (u < n) {
-        const u32 deg = xadj[u + 1] - xadj[u];
-        if (active[u]) {
-          my += deg;
-          active[u] = 0;
-        }
-      }
-      if (lane == 0) {
-        unit_active[vb] = 0;
-      }
+// Activate neighbours of admitted movers (wave per admitted entry).
+__global__ void k_activate_v2(
+    u32 k,
+    const u32 *__restrict__ seg_off,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ s_u,
+    const uint16_t *__restrict__ s_to,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    uint8_t *__restrict__ active,
+    uint8_t *__restrict__ unit_active
+) {
+  const u32 count = seg_off[k];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 waves_total = (gridDim.x * blockDim.x) >> 6;
+  for (u32 i = (blockIdx.x * blockDim.x + threadIdx.x) >> 6; i < count; i += waves_total) {
+    const u32 c = s_to[i];
+    if (i - seg_off[c] >= prefix_len[c]) {
+      continue;
     }
-    for (int off = 32; off > 0; off >>= 1) {
-      my += __shfl_down(my, off, kWave);
-    }
-    __syncthreads(); // red reuse across sections
-    if (lane == 0) {
-      reinterpret_cast<unsigned long long *>(red)[tid >> 6] = my;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      unsigned long long t = 0;
-      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
-        t += reinterpret_cast<unsigned long long *>(red)[w];
-      }
-      if (t) {
-        atomicAdd(arcs, t);
-      }
-    }
-  }
-  coop_bar(bar, nblk);
-
-  // ---- activate neighbours of admitted movers (wave per entry) ----
-  {
-    const u32 waves_total = gsz >> 6;
-    for (u32 i = gid >> 6; i < count; i += waves_total) {
-      const u32 c = s_to[i];
-      if (i - seg_off[c] >= prefix_len[c]) {
-        continue;
-      }
-      const u32 u = s_u[i];
-      const u32 row = xadj[u];
-      const u32 deg = xadj[u + 1] - row;
-      for (u32 e2 = lane; e2 < deg; e2 += kWave) {
-        const u32 v = adjncy[row + e2];
-        active[v] = 1;
-        unit_active[v >> 6] = 1;
-      }
+    const u32 u = s_u[i];
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+    for (u32 e2 = lane; e2 < deg; e2 += kWave) {
+      const u32 v = adjncy[row + e2];
+      active[v] = 1;
+      unit_active[v >> 6] = 1;
     }
   }
 }
@@ -2892,11 +2860,26 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   LAUNCH_CHECK();
   const size_t lds_c = static_cast<size_t>(e->k + 17) * sizeof(unsigned long long);
   hipLaunchKernelGGL(
-      k_commit_coop, dim3(e->coop_nblk), dim3(threads), lds_c, e->stream, e->k, pos_lo, pos_hi,
-      e->n, iseed, static_cast<u32>(e->has_vwgt ? 1 : 0), e->coop_nblk, e->d_seg_off,
-      e->d_prefix_len, e->d_dep, e->d_pw, e->d_blocksums, e->d_s_u, e->d_s_w, e->d_s_to,
-      e->d_s_b, e->d_weights, e->d_maxw, e->d_labels, e->d_labels16, e->d_xadj, e->d_adjncy,
-      e->d_active, e->d_unit_active, e->d_arcs, e->d_moves, e->d_changed, e->d_bar
+      k_commit_coop, dim3(e->coop_nblk), dim3(threads), lds_c, e->stream, e->k,
+      static_cast<u32>(e->has_vwgt ? 1 : 0), e->coop_nblk, e->d_seg_off, e->d_prefix_len,
+      e->d_dep, e->d_pw, e->d_blocksums, e->d_s_w, e->d_s_to, e->d_s_b, e->d_weights,
+      e->d_maxw, e->d_moves, e->d_changed, e->d_bar
+  );
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_apply_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
+      e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16
+  );
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_clear_active, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, pos_lo,
+      pos_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active, e->d_arcs
+  );
+  LAUNCH_CHECK();
+  hipLaunchKernelGGL(
+      k_activate_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, e->d_seg_off,
+      e->d_prefix_len, e->d_s_u, e->d_s_to, e->d_xadj, e->d_adjncy, e->d_active,
+      e->d_unit_active
   );
   LAUNCH_CHECK();
 }
