@@ -170,11 +170,7 @@ __global__ void k_phase_s(
     const uint16_t *__restrict__ labels16,
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
-    Prop *__restrict__ slots,
-    u64 *__restrict__ m_list,
-    u32 *__restrict__ m_count,
-    u64 *__restrict__ l_list,
-    u32 *__restrict__ l_count
+    Prop *__restrict__ slots
 ) {
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 sub = lane >> 4;  // subgroup 0..3
@@ -207,41 +203,9 @@ __global__ void k_phase_s(
     }
   }
 
-  // M-class positions append to the M work list wave-aggregated (list order
-  // is irrelevant for determinism: M writes land in per-position slots);
-  // L-class (rare) appends to the L list the same way.
-  {
-    const bool is_m = !skip && slot == 0 && deg > kSmallDeg && deg <= kMidDeg;
-    const unsigned long long mm = __ballot(is_m);
-    if (mm) {
-      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
-      u32 bbase = 0;
-      if (lane == leader) {
-        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
-      }
-      bbase = __shfl(bbase, leader, kWave);
-      if (is_m) {
-        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] =
-            (static_cast<u64>(p) << 32) | u;
-      }
-    }
-  }
-  {
-    const bool is_l = !skip && deg > kMidDeg && slot == 0;
-    const unsigned long long ll = __ballot(is_l);
-    if (ll) {
-      const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
-      u32 bbase = 0;
-      if (lane == leader) {
-        bbase = atomicAdd(l_count, static_cast<u32>(__popcll(ll)));
-      }
-      bbase = __shfl(bbase, leader, kWave);
-      if (is_l) {
-        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] =
-            (static_cast<u64>(p) << 32) | u;
-      }
-    }
-  }
+  // M/L work lists are built by k_build_lists (one reservation atomic per
+  // wave-row; appending from here serialized millions of waves on a single
+  // counter -- measured 2x on k_phase_s at scale 26).
   if (skip) {
     // always-write contract: every position of an ACTIVE unit gets a fresh
     // slot each chunk (proposal or invalid), so the memset-free v2 commit
@@ -763,11 +727,7 @@ __global__ void k_phase_s_c(
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
-    Prop *__restrict__ slots,
-    u64 *__restrict__ m_list,
-    u32 *__restrict__ m_count,
-    u64 *__restrict__ l_list, // deg > kClusterMidDeg
-    u32 *__restrict__ l_count
+    Prop *__restrict__ slots
 ) {
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 sub = lane >> 4;
@@ -799,38 +759,7 @@ __global__ void k_phase_s_c(
     }
   }
 
-  {
-    const bool is_m = !skip && slot == 0 && deg > kSmallDeg && deg <= kClusterMidDeg;
-    const unsigned long long mm = __ballot(is_m);
-    if (mm) {
-      const u32 leader = __ffsll(static_cast<unsigned long long>(mm)) - 1;
-      u32 bbase = 0;
-      if (lane == leader) {
-        bbase = atomicAdd(m_count, static_cast<u32>(__popcll(mm)));
-      }
-      bbase = __shfl(bbase, leader, kWave);
-      if (is_m) {
-        m_list[bbase + __popcll(mm & ((1ull << lane) - 1))] =
-            (static_cast<u64>(p) << 32) | u;
-      }
-    }
-  }
-  // L list append (deg > kClusterMidDeg), wave-aggregated
-  {
-    const bool is_l = !skip && deg > kClusterMidDeg && slot == 0;
-    const unsigned long long ll = __ballot(is_l);
-    if (ll) {
-      const u32 leader = __ffsll(static_cast<unsigned long long>(ll)) - 1;
-      u32 bbase = 0;
-      if (lane == leader) {
-        bbase = atomicAdd(l_count, static_cast<u32>(__popcll(ll)));
-      }
-      bbase = __shfl(bbase, leader, kWave);
-      if (is_l) {
-        l_list[bbase + __popcll(ll & ((1ull << lane) - 1))] = (static_cast<u64>(p) << 32) | u;
-      }
-    }
-  }
+  // M/L work lists are built by k_build_lists
   if (skip || deg > kSmallDeg) {
     return; // M/L own larger degrees
   }
@@ -1993,6 +1922,114 @@ __global__ void k_l_prep_r(
   }
 }
 
+// Build the M and L work lists for a chunk straight from xadj/active under
+// the permutation (no S-kernel involvement): each wave owns a run of
+// 64-position tiles; pass 1 counts its M/L entries, ONE reservation atomic
+// per wave per list, pass 2 writes the records. List order is wave-row
+// order -- irrelevant for determinism since the M/L kernels write
+// per-position slots.
+__global__ void k_build_lists(
+    u32 pos_lo,
+    u32 pos_hi, // may be an unaligned sub-range (sharded path)
+    u32 n,
+    u64 iter_seed,
+    u32 rows,
+    u32 tpw,
+    u32 mid_lo, // kSmallDeg
+    u32 mid_hi, // kMidDeg (refine) / kClusterMidDeg (cluster)
+    u32 max_degree,
+    const u32 *__restrict__ xadj,
+    const uint8_t *__restrict__ active,
+    const uint8_t *__restrict__ unit_active,
+    u64 *__restrict__ m_list,
+    u32 *__restrict__ m_count,
+    u64 *__restrict__ l_list,
+    u32 *__restrict__ l_count
+) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 ta0 = pos_lo >> 6;                 // absolute unit-tile range
+  const u32 T = ((pos_hi + 63) >> 6) - ta0;
+  const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+  const u32 t0 = row * tpw;
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  u32 m_cnt = 0, l_cnt = 0;
+  for (u32 pass = 0; pass < 2; ++pass) {
+    u32 m_base = 0, l_base = 0;
+    if (pass == 1) {
+      // reserve (lane 0; contention = one atomic per wave-row per list)
+      if (lane == 0) {
+        if (m_cnt) {
+          m_base = atomicAdd(m_count, m_cnt);
+        }
+        if (l_cnt) {
+          l_base = atomicAdd(l_count, l_cnt);
+        }
+      }
+      m_base = __shfl(m_base, 0, kWave);
+      l_base = __shfl(l_base, 0, kWave);
+      if (m_cnt == 0 && l_cnt == 0) {
+        return;
+      }
+    }
+    for (u32 t = t0; t < t1; ++t) {
+      u32 vb = 0;
+      if (lane == 0) {
+        vb = fp(ta0 + t);
+      }
+      vb = __shfl(vb, 0, kWave);
+      if (!unit_active[vb]) {
+        continue;
+      }
+      const u32 p = ((ta0 + t) << 6) + lane;
+      const u32 u = vb * kmp::kUnit + lane;
+      bool is_m = false, is_l = false;
+      if (p >= pos_lo && p < pos_hi && u < n && active[u]) {
+        const u32 deg = xadj[u + 1] - xadj[u];
+        if (deg <= max_degree) {
+          is_m = deg > mid_lo && deg <= mid_hi;
+          is_l = deg > mid_hi;
+        }
+      }
+      const unsigned long long mm = __ballot(is_m);
+      const unsigned long long ll = __ballot(is_l);
+      if (pass == 0) {
+        m_cnt += __popcll(mm);
+        l_cnt += __popcll(ll);
+      } else {
+        if (is_m) {
+          m_list[m_base + __popcll(mm & ((1ull << lane) - 1))] =
+              (static_cast<u64>(p) << 32) | u;
+        }
+        if (is_l) {
+          l_list[l_base + __popcll(ll & ((1ull << lane) - 1))] =
+              (static_cast<u64>(p) << 32) | u;
+        }
+        m_base += __popcll(mm);
+        l_base += __popcll(ll);
+      }
+    }
+  }
+}
+
+// Resident-grid exclusive scan of the k x rows histogram (entries up to
+// 65536; a single-workgroup scan is latency-bound at that size) + segment
+// offsets + prefix_len/dep/changed init. Uses the same coop_bar as the
+// fixpoint kernel.
+__global__ void k_scan_coop(
+    u32 k,
+    u32 rows,
+    u32 nblk,
+    const u32 *__restrict__ histT,
+    u32 *__restrict__ offT,
+    u32 *__restrict__ seg_off, // k+1
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep, // 2*k
+    int *__restrict__ changed2,           // int[2]
+    i64 *__restrict__ blocksums,
+    u32 *__restrict__ bar
+);
+
 // Per-wave-row histogram of proposal targets over the raw slot array.
 // Each wave owns a contiguous run of 64-position tiles (= permutation
 // units); fully-inactive units are skipped without reading their (stale)
@@ -2036,69 +2073,6 @@ __global__ void k_hist_v2(
   }
   for (u32 c = lane; c < k; c += kWave) {
     histT[c * rows + row] = cnt[c];
-  }
-}
-
-// Single-workgroup exclusive scan of the k x rows histogram (k*rows <=
-// 65536 entries) + segment offsets + prefix_len init (= full admission,
-// the greatest-fixpoint starting point).
-__global__ void k_matscan_v2(
-    u32 k,
-    u32 rows,
-    const u32 *__restrict__ histT,
-    u32 *__restrict__ offT,
-    u32 *__restrict__ seg_off, // k+1
-    u32 *__restrict__ prefix_len,
-    unsigned long long *__restrict__ dep, // 2*k (double-buffered, pre-zeroed)
-    int *__restrict__ changed2            // int[2]
-) {
-  __shared__ u32 red[17];
-  const u32 entries = k * rows;
-  const u32 tid = threadIdx.x;
-  u32 carry = 0;
-  for (u32 base = 0; base < entries; base += blockDim.x) {
-    const u32 i = base + tid;
-    u32 v = i < entries ? histT[i] : 0;
-    u32 inc = v;
-    for (int off = 1; off < 64; off <<= 1) {
-      const u32 o = __shfl_up(inc, off, kWave);
-      if ((tid & 63) >= static_cast<u32>(off)) {
-        inc += o;
-      }
-    }
-    __syncthreads();
-    if ((tid & 63) == 63) {
-      red[tid >> 6] = inc;
-    }
-    __syncthreads();
-    u32 wbase = 0;
-    for (u32 w = 0; w < (tid >> 6); ++w) {
-      wbase += red[w];
-    }
-    if (i < entries) {
-      offT[i] = carry + wbase + inc - v; // exclusive
-    }
-    u32 tsum = 0;
-    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
-      tsum += red[w];
-    }
-    carry += tsum;
-    __syncthreads();
-  }
-  __threadfence_block();
-  __syncthreads();
-  for (u32 c = tid; c < k; c += blockDim.x) {
-    const u32 b = offT[c * rows];
-    const u32 e2 = (c + 1 < k) ? offT[(c + 1) * rows] : carry;
-    seg_off[c] = b;
-    prefix_len[c] = e2 - b;
-    dep[c] = 0;
-    dep[k + c] = 0;
-  }
-  if (tid == 0) {
-    seg_off[k] = carry;
-    changed2[0] = 0;
-    changed2[1] = 0;
   }
 }
 
@@ -2210,6 +2184,103 @@ __device__ inline void coop_bar(u32 *bar, u32 nblk) {
     __threadfence(); // see every other block's published writes
   }
   __syncthreads();
+}
+
+__global__ void k_scan_coop(
+    u32 k,
+    u32 rows,
+    u32 nblk,
+    const u32 *__restrict__ histT,
+    u32 *__restrict__ offT,
+    u32 *__restrict__ seg_off,
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep,
+    int *__restrict__ changed2,
+    i64 *__restrict__ blocksums,
+    u32 *__restrict__ bar
+) {
+  __shared__ u32 red[5];
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & (kWave - 1);
+  const u32 entries = k * rows;
+  const u32 rlen = (entries + nblk - 1) / nblk;
+  const u32 lo = blockIdx.x * rlen < entries ? blockIdx.x * rlen : entries;
+  const u32 hi = lo + rlen < entries ? lo + rlen : entries;
+
+  // stage 1: block-local exclusive scan of [lo, hi) into offT + block total
+  u32 carry = 0;
+  for (u32 base = lo; base < hi; base += blockDim.x) {
+    const u32 i = base + tid;
+    u32 v = i < hi ? histT[i] : 0;
+    u32 inc = v;
+    for (int off = 1; off < 64; off <<= 1) {
+      const u32 o = __shfl_up(inc, off, kWave);
+      if (lane >= static_cast<u32>(off)) {
+        inc += o;
+      }
+    }
+    __syncthreads();
+    if (lane == 63) {
+      red[tid >> 6] = inc;
+    }
+    __syncthreads();
+    u32 wbase = 0;
+    for (u32 w = 0; w < (tid >> 6); ++w) {
+      wbase += red[w];
+    }
+    if (i < hi) {
+      offT[i] = carry + wbase + inc - v; // local exclusive
+    }
+    u32 tsum = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      tsum += red[w];
+    }
+    carry += tsum;
+    __syncthreads();
+  }
+  if (tid == 0) {
+    blocksums[blockIdx.x] = static_cast<i64>(carry);
+  }
+  coop_bar(bar, nblk);
+  // stage 2: block 0 exclusive-scans the block totals; stashes the grand
+  // total at blocksums[nblk]
+  if (blockIdx.x == 0 && tid == 0) {
+    i64 run = 0;
+    for (u32 b = 0; b < nblk; ++b) {
+      const i64 v = blocksums[b];
+      blocksums[b] = run;
+      run += v;
+    }
+    blocksums[nblk] = run;
+  }
+  coop_bar(bar, nblk);
+  // stage 3: add carries
+  {
+    const u32 bc = static_cast<u32>(blocksums[blockIdx.x]);
+    if (bc) {
+      for (u32 i = lo + tid; i < hi; i += blockDim.x) {
+        offT[i] += bc;
+      }
+    }
+  }
+  coop_bar(bar, nblk);
+  // stage 4: segment offsets + fixpoint init
+  const u32 total = static_cast<u32>(blocksums[nblk]);
+  const u32 gid = blockIdx.x * blockDim.x + tid;
+  const u32 gsz = nblk * blockDim.x;
+  for (u32 c = gid; c < k; c += gsz) {
+    const u32 b = offT[c * rows];
+    const u32 e2 = (c + 1 < k) ? offT[(c + 1) * rows] : total;
+    seg_off[c] = b;
+    prefix_len[c] = e2 - b;
+    dep[c] = 0;
+    dep[k + c] = 0;
+  }
+  if (gid == 0) {
+    seg_off[k] = total;
+    changed2[0] = 0;
+    changed2[1] = 0;
+  }
 }
 
 // The whole deterministic admission in ONE launch: (optional) segmented
@@ -2779,12 +2850,22 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   const u32 span = pos_hi - pos_lo;
   const u32 threads = 256;
   HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 2, e->stream));
+  {
+    const u32 rows_bl = 4096;
+    const u32 T_bl = ((pos_hi + 63) >> 6) - (pos_lo >> 6);
+    const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
+    hipLaunchKernelGGL(
+        k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
+        iseed, rows_bl, tpw, kSmallDeg, kMidDeg, 0xFFFFFFFFu, e->d_xadj, e->d_active,
+        e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+    );
+    LAUNCH_CHECK();
+  }
   hipLaunchKernelGGL(
       k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, 0u, kInvalid,
       0xFFFFFFFFu, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-      e->d_maxw, e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_list,
-      e->d_m_count, e->d_l_list, e->d_l_count
+      e->d_maxw, e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
   );
   LAUNCH_CHECK();
   {
@@ -2848,8 +2929,9 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
-      k_matscan_v2, dim3(1), dim3(1024), 0, e->stream, e->k, rows, e->d_histT, e->d_offT,
-      e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed
+      k_scan_coop, dim3(e->coop_nblk), dim3(threads), 0, e->stream, e->k, rows, e->coop_nblk,
+      e->d_histT, e->d_offT, e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed,
+      e->d_blocksums, e->d_bar
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
@@ -3125,14 +3207,26 @@ int kmp_lp_refine_begin(
   // occupancy query says the grid cannot be made co-resident). The grid is
   // deliberately SMALL (64 blocks): the commit stages are latency-bound and
   // the grid barrier cost grows with the block count (tools/bar_bench.hip).
-  e->rows_v2 = 256u;
+  {
+    u32 rows = (65536u / k) & ~3u;
+    if (rows < 256) {
+      rows = 256;
+    }
+    if (rows > 4096) {
+      rows = 4096;
+    }
+    e->rows_v2 = rows;
+  }
   e->coop_nblk = 0;
   if (k <= 256) {
-    int occ = 0;
+    int occ_fix = 0, occ_scan = 0;
     const size_t lds_c = static_cast<size_t>(k + 17) * sizeof(unsigned long long);
-    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ, k_commit_coop, 256, lds_c) ==
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ_fix, k_commit_coop, 256, lds_c) ==
             hipSuccess &&
-        occ > 0 && e->mp_count > 0) {
+        hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ_scan, k_scan_coop, 256, 0) ==
+            hipSuccess &&
+        occ_fix > 0 && occ_scan > 0 && e->mp_count > 0) {
+      const int occ = occ_fix < occ_scan ? occ_fix : occ_scan;
       u64 nb = static_cast<u64>(occ) * static_cast<u64>(e->mp_count);
       if (nb > 64) {
         nb = 64;
@@ -3184,17 +3278,27 @@ i64 kmp_lp_phase_a(
         }
       }
     }
+    {
+      const u32 rows_bl = 4096;
+      const u32 T_bl = ((pos_hi + 63) >> 6) - (pos_lo >> 6);
+      const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
+      hipLaunchKernelGGL(
+          k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
+          iseed, rows_bl, tpw, kSmallDeg, kMidDeg, max_degree, e->d_xadj, e->d_active,
+          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+      );
+      LAUNCH_CHECK();
+    }
     // S: 4 positions/wave (unit-gated; slots pre-marked invalid)
     hipLaunchKernelGGL(
         k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
         static_cast<u32>(e->balance), fallback, max_degree,
         e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots, e->d_m_list, e->d_m_count,
-        e->d_l_list, e->d_l_count
+        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
     );
     LAUNCH_CHECK();
-    // one wave per M-listed vertex (grid-stride; list appended by k_phase_s)
+    // one wave per M-listed vertex (grid-stride; list built by k_build_lists)
     {
       const size_t lds =
           static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
@@ -3254,12 +3358,22 @@ i64 kmp_lp_phase_a(
     }
   } else {
     // clustering: hash-based gain maps, favored-cluster tracking
+    {
+      const u32 rows_bl = 4096;
+      const u32 T_bl = ((pos_hi + 63) >> 6) - (pos_lo >> 6);
+      const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
+      hipLaunchKernelGGL(
+          k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
+          iseed, rows_bl, tpw, kSmallDeg, kClusterMidDeg, max_degree, e->d_xadj, e->d_active,
+          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+      );
+      LAUNCH_CHECK();
+    }
     hipLaunchKernelGGL(
         k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_active, e->d_unit_active, e->d_favored, e->d_slots, e->d_m_list, e->d_m_count,
-        e->d_l_list, e->d_l_count
+        e->d_active, e->d_unit_active, e->d_favored, e->d_slots
     );
     LAUNCH_CHECK();
     {
