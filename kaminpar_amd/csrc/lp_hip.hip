@@ -2097,9 +2097,17 @@ __global__ void k_scatter_v2(
     u32 *__restrict__ s_u,
     u32 *__restrict__ s_w,
     uint16_t *__restrict__ s_to,
-    uint16_t *__restrict__ s_b
+    uint16_t *__restrict__ s_b,
+    unsigned long long *__restrict__ dep // [0..k) departures, [k..2k) arrivals
 ) {
+  // dynamic LDS: per-wave cnt (k u32 each) then one per-WG pair of u64
+  // hists: full-admission departure weights by source block and incoming
+  // weights by target (the fixpoint kernel starts from these totals).
   extern __shared__ u32 cnt_lds[];
+  const u32 waves_per_wg = blockDim.x >> 6;
+  unsigned long long *h_out =
+      reinterpret_cast<unsigned long long *>(cnt_lds + waves_per_wg * k);
+  unsigned long long *h_in = h_out + k;
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wv = threadIdx.x >> 6;
   u32 *cnt = cnt_lds + wv * k;
@@ -2107,6 +2115,11 @@ __global__ void k_scatter_v2(
   for (u32 c = lane; c < k; c += kWave) {
     cnt[c] = offT[c * rows + row];
   }
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    h_out[c] = 0;
+    h_in[c] = 0;
+  }
+  __syncthreads();
   const u32 T = span >> 6;
   const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
   const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
@@ -2145,10 +2158,22 @@ __global__ void k_scatter_v2(
     const u32 base = __shfl(base0, leader, kWave);
     if (valid) {
       const u32 dst = base + rank;
+      const uint16_t src = labels16[pr.u];
       s_u[dst] = pr.u;
       s_w[dst] = pr.w;
       s_to[dst] = static_cast<uint16_t>(c);
-      s_b[dst] = labels16[pr.u];
+      s_b[dst] = src;
+      atomicAdd(&h_out[src], static_cast<unsigned long long>(pr.w));
+      atomicAdd(&h_in[c], static_cast<unsigned long long>(pr.w));
+    }
+  }
+  __syncthreads();
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    if (h_out[c]) {
+      atomicAdd(&dep[c], h_out[c]);
+    }
+    if (h_in[c]) {
+      atomicAdd(&dep[k + c], h_in[c]);
     }
   }
 }
@@ -2283,221 +2308,108 @@ __global__ void k_scan_coop(
   }
 }
 
-// The whole deterministic admission in ONE launch: (optional) segmented
-// prefix weights, the greatest-fixpoint rollback loop with device-side
-// convergence (changed flag double-buffered by round parity), weight
-// update, label apply, chunk active-set clear + arcs tally, and neighbour
-// activation. Restates kaminpar-dist/refinement/lp/lp_refiner.cc:296-333
-// under the deterministic schedule; bit-identical to the legacy commit.
-__global__ void k_commit_coop(
+// The admission fixpoint in ONE single-workgroup launch (no grid
+// barriers, trivially stream-ordered). Starts from the full-admission
+// state the scatter kernel accumulated (dep = departures by source block,
+// arr = incoming weight by target) and walks each target's rank-cutoff
+// BACKWARD per round under capacities frozen at round start -- exactly the
+// synchronous rollback rounds of kaminpar-dist lp_refiner.cc:296-333, with
+// O(de-admitted) work per round instead of a full rescan. Bit-identical to
+// the legacy commit (same per-round cutoffs, greatest fixpoint).
+__global__ void k_fixpoint_v2(
     u32 k,
-    u32 has_vwgt,
-    u32 nblk,
     const u32 *__restrict__ seg_off,
     u32 *__restrict__ prefix_len,
-    unsigned long long *__restrict__ dep,
-    i64 *__restrict__ pw,
-    i64 *__restrict__ blocksums,
+    const unsigned long long *__restrict__ dep, // [0..k) out, [k..2k) in
     const u32 *__restrict__ s_w,
-    const uint16_t *__restrict__ s_to,
     const uint16_t *__restrict__ s_b,
     i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    unsigned long long *__restrict__ moves,
-    int *__restrict__ changed2, // int[2]
-    u32 *__restrict__ bar
+    unsigned long long *__restrict__ moves
 ) {
-  extern __shared__ unsigned long long lds64[];
-  unsigned long long *udep = lds64;                  // k entries
-  i64 *red = reinterpret_cast<i64 *>(lds64 + k);     // 17 entries
+  __shared__ unsigned long long ddep[256];   // current departures
+  __shared__ unsigned long long ddelta[256]; // this round's de-admissions
+  __shared__ long long arr_s[256];           // admitted incoming weight
+  __shared__ u32 plen_s[256];
+  __shared__ u32 soff_s[256];
+  __shared__ unsigned long long red[4];
+  __shared__ int chg;
   const u32 tid = threadIdx.x;
-  const u32 gid = blockIdx.x * blockDim.x + tid;
-  const u32 gsz = nblk * blockDim.x;
-  const u32 count = seg_off[k];
-  const u32 lane = tid & (kWave - 1);
-
-  // ---- segmented prefix weights (weighted graphs only; unweighted uses
-  // pw[i] = i+1 implicitly) ----
-  if (has_vwgt && count) {
-    const u32 rlen = (count + nblk - 1) / nblk;
-    const u32 lo = blockIdx.x * rlen < count ? blockIdx.x * rlen : count;
-    const u32 hi = lo + rlen < count ? lo + rlen : count;
-    i64 acc = 0;
-    for (u32 i = lo + tid; i < hi; i += blockDim.x) {
-      acc += s_w[i];
-    }
-    for (int off = 32; off > 0; off >>= 1) {
-      acc += __shfl_down(acc, off, kWave);
-    }
-    if (lane == 0) {
-      red[tid >> 6] = acc;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      i64 t = 0;
-      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
-        t += red[w];
-      }
-      blocksums[blockIdx.x] = t;
-    }
-    coop_bar(bar, nblk);
-    if (blockIdx.x == 0 && tid == 0) {
-      i64 runsum = 0;
-      for (u32 b = 0; b < nblk; ++b) {
-        const i64 v = blocksums[b];
-        blocksums[b] = runsum;
-        runsum += v;
-      }
-    }
-    coop_bar(bar, nblk);
-    i64 carry = blocksums[blockIdx.x];
-    for (u32 base = lo; base < hi; base += blockDim.x) {
-      const u32 i = base + tid;
-      i64 v = (i < hi) ? static_cast<i64>(s_w[i]) : 0;
-      i64 inc = v;
-      for (int off = 1; off < 64; off <<= 1) {
-        const i64 o = __shfl_up(inc, off, kWave);
-        if (lane >= static_cast<u32>(off)) {
-          inc += o;
-        }
-      }
-      __syncthreads();
-      if (lane == 63) {
-        red[tid >> 6] = inc;
-      }
-      __syncthreads();
-      i64 wbase = 0;
-      for (u32 w = 0; w < (tid >> 6); ++w) {
-        wbase += red[w];
-      }
-      if (i < hi) {
-        pw[i] = carry + wbase + inc;
-      }
-      i64 tsum = 0;
-      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
-        tsum += red[w];
-      }
-      carry += tsum;
-      __syncthreads();
-    }
-    coop_bar(bar, nblk);
+  if (tid < k) {
+    ddep[tid] = dep[tid];
+    arr_s[tid] = static_cast<long long>(dep[k + tid]);
+    plen_s[tid] = prefix_len[tid];
+    soff_s[tid] = seg_off[tid];
+    ddelta[tid] = 0;
   }
-
-  // ---- greatest-fixpoint rollback (2 barriers per round) ----
-  // dep is double-buffered by round parity: round r accumulates into slot
-  // r&1 (pre-zeroed by k_matscan_v2 / the previous round's cutoff stage),
-  // so the zero stage costs no extra barrier.
-  u32 round = 0;
-  u32 fin = 0;
+  __syncthreads();
   for (;;) {
-    const u32 A = round & 1;
-    unsigned long long *depA = dep + static_cast<size_t>(A) * k;
-    for (u32 c = tid; c < k; c += blockDim.x) {
-      udep[c] = 0;
+    if (tid == 0) {
+      chg = 0;
+    }
+    i64 cap = 0;
+    if (tid < k) {
+      // capacity frozen at round start (synchronous rounds)
+      cap = maxw[tid] - weights[tid] + static_cast<i64>(ddep[tid]);
     }
     __syncthreads();
-    for (u32 i = gid; i < count; i += gsz) {
-      const u32 c = s_to[i];
-      if (i - seg_off[c] < prefix_len[c]) {
-        atomicAdd(&udep[s_b[i]],
-                  has_vwgt ? static_cast<unsigned long long>(s_w[i]) : 1ull);
+    if (tid < k) {
+      u32 pl = plen_s[tid];
+      long long a = arr_s[tid];
+      const u32 b = soff_s[tid];
+      bool any = false;
+      while (a > cap && pl > 0) {
+        --pl;
+        const u32 i = b + pl;
+        const unsigned long long w = s_w[i];
+        a -= static_cast<long long>(w);
+        atomicAdd(&ddelta[s_b[i]], w);
+        any = true;
+      }
+      if (any) {
+        plen_s[tid] = pl;
+        arr_s[tid] = a;
+        chg = 1; // benign same-value race
       }
     }
     __syncthreads();
-    for (u32 c = tid; c < k; c += blockDim.x) {
-      if (udep[c]) {
-        atomicAdd(&depA[c], udep[c]);
-      }
+    if (tid < k && ddelta[tid]) {
+      ddep[tid] -= ddelta[tid];
+      ddelta[tid] = 0;
     }
-    coop_bar(bar, nblk);
-    if (blockIdx.x == 0) {
-      unsigned long long *depB = dep + static_cast<size_t>(A ^ 1) * k;
-      for (u32 c = tid; c < k; c += blockDim.x) {
-        depB[c] = 0; // pre-zero the next round's slot
-        const u32 old_len = prefix_len[c];
-        if (old_len == 0) {
-          continue;
-        }
-        const u32 b = seg_off[c];
-        const i64 capacity = maxw[c] - weights[c] + static_cast<i64>(depA[c]);
-        u32 nl;
-        if (!has_vwgt) {
-          nl = capacity <= 0 ? 0u
-                             : (capacity >= static_cast<i64>(old_len) ? old_len
-                                                                      : static_cast<u32>(capacity));
-        } else {
-          const i64 base0 = b ? pw[b - 1] : 0;
-          u32 lo2 = 0, hi2 = old_len;
-          while (lo2 < hi2) {
-            const u32 mid = (lo2 + hi2 + 1) >> 1;
-            if (pw[b + mid - 1] - base0 <= capacity) {
-              lo2 = mid;
-            } else {
-              hi2 = mid - 1;
-            }
-          }
-          nl = lo2;
-        }
-        if (nl < old_len) {
-          prefix_len[c] = nl;
-          __hip_atomic_store(&changed2[A], 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        }
-      }
-      if (tid == 0) {
-        changed2[A ^ 1] = 0; // pre-zero the next round's flag
-      }
-    }
-    coop_bar(bar, nblk);
-    fin = A;
-    if (!__hip_atomic_load(&changed2[A], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+    __syncthreads();
+    if (!chg) {
       break;
     }
-    ++round;
-  }
-  unsigned long long *depF = dep + static_cast<size_t>(fin) * k;
-
-  // ---- weights + moves (block 0) ----
-  if (blockIdx.x == 0) {
-    unsigned long long mv = 0;
-    for (u32 c = tid; c < k; c += blockDim.x) {
-      const u32 len = prefix_len[c];
-      mv += len;
-      i64 arr = 0;
-      if (len) {
-        if (!has_vwgt) {
-          arr = static_cast<i64>(len);
-        } else {
-          const u32 b = seg_off[c];
-          arr = pw[b + len - 1] - (b ? pw[b - 1] : 0);
-        }
-      }
-      const i64 delta = arr - static_cast<i64>(depF[c]);
-      if (delta) {
-        weights[c] += delta;
-      }
-    }
-    for (int off = 32; off > 0; off >>= 1) {
-      mv += __shfl_down(mv, off, kWave);
-    }
-    __syncthreads();
-    if (lane == 0) {
-      reinterpret_cast<unsigned long long *>(red)[tid >> 6] = mv;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      unsigned long long t = 0;
-      for (u32 w = 0; w < blockDim.x / kWave; ++w) {
-        t += reinterpret_cast<unsigned long long *>(red)[w];
-      }
-      if (t) {
-        atomicAdd(moves, t);
-      }
-    }
+    __syncthreads(); // protect the chg reset of the next round
   }
 
-  // apply / clear-active / activation run as separate full-grid kernels
-  // (they are bandwidth-bound; a launch boundary is cheaper than starving
-  // them on the small resident grid -- measured, tools/bar_bench.hip).
+  // weights + writeback + move count
+  unsigned long long mv = 0;
+  if (tid < k) {
+    const long long delta = arr_s[tid] - static_cast<long long>(ddep[tid]);
+    if (delta) {
+      weights[tid] += delta;
+    }
+    prefix_len[tid] = plen_s[tid];
+    mv = plen_s[tid];
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    mv += __shfl_down(mv, off, kWave);
+  }
+  if ((tid & (kWave - 1)) == 0) {
+    red[tid >> 6] = mv;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    unsigned long long t = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      t += red[w];
+    }
+    if (t) {
+      atomicAdd(moves, t);
+    }
+  }
 }
 
 // Apply admitted labels (full grid, count read from device).
@@ -2934,18 +2846,16 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       e->d_blocksums, e->d_bar
   );
   LAUNCH_CHECK();
+  const size_t lds_sc = lds_h + static_cast<size_t>(2 * e->k) * sizeof(unsigned long long);
   hipLaunchKernelGGL(
-      k_scatter_v2, dim3(rows / 4), dim3(threads), lds_h, e->stream, span, pos_lo, e->n, iseed,
+      k_scatter_v2, dim3(rows / 4), dim3(threads), lds_sc, e->stream, span, pos_lo, e->n, iseed,
       e->k, rows, tpw, e->d_slots, e->d_unit_active, e->d_offT, e->d_labels16, e->d_s_u,
-      e->d_s_w, e->d_s_to, e->d_s_b
+      e->d_s_w, e->d_s_to, e->d_s_b, e->d_dep
   );
   LAUNCH_CHECK();
-  const size_t lds_c = static_cast<size_t>(e->k + 17) * sizeof(unsigned long long);
   hipLaunchKernelGGL(
-      k_commit_coop, dim3(e->coop_nblk), dim3(threads), lds_c, e->stream, e->k,
-      static_cast<u32>(e->has_vwgt ? 1 : 0), e->coop_nblk, e->d_seg_off, e->d_prefix_len,
-      e->d_dep, e->d_pw, e->d_blocksums, e->d_s_w, e->d_s_to, e->d_s_b, e->d_weights,
-      e->d_maxw, e->d_moves, e->d_changed, e->d_bar
+      k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
+      e->d_dep, e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
@@ -3219,15 +3129,11 @@ int kmp_lp_refine_begin(
   }
   e->coop_nblk = 0;
   if (k <= 256) {
-    int occ_fix = 0, occ_scan = 0;
-    const size_t lds_c = static_cast<size_t>(k + 17) * sizeof(unsigned long long);
-    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ_fix, k_commit_coop, 256, lds_c) ==
+    int occ_scan = 0;
+    if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ_scan, k_scan_coop, 256, 0) ==
             hipSuccess &&
-        hipOccupancyMaxActiveBlocksPerMultiprocessor(&occ_scan, k_scan_coop, 256, 0) ==
-            hipSuccess &&
-        occ_fix > 0 && occ_scan > 0 && e->mp_count > 0) {
-      const int occ = occ_fix < occ_scan ? occ_fix : occ_scan;
-      u64 nb = static_cast<u64>(occ) * static_cast<u64>(e->mp_count);
+        occ_scan > 0 && e->mp_count > 0) {
+      u64 nb = static_cast<u64>(occ_scan) * static_cast<u64>(e->mp_count);
       if (nb > 64) {
         nb = 64;
       }
