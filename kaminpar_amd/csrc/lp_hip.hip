@@ -2318,6 +2318,7 @@ __global__ void k_scan_coop(
 // the legacy commit (same per-round cutoffs, greatest fixpoint).
 __global__ void k_fixpoint_v2(
     u32 k,
+    u32 has_vwgt,
     const u32 *__restrict__ seg_off,
     u32 *__restrict__ prefix_len,
     const unsigned long long *__restrict__ dep, // [0..k) out, [k..2k) in
@@ -2332,6 +2333,7 @@ __global__ void k_fixpoint_v2(
   __shared__ long long arr_s[256];           // admitted incoming weight
   __shared__ u32 plen_s[256];
   __shared__ u32 soff_s[256];
+  __shared__ u32 rlo_s[256], rhi_s[256]; // this round's de-admitted ranges
   __shared__ unsigned long long red[4];
   __shared__ int chg;
   const u32 tid = threadIdx.x;
@@ -2341,6 +2343,8 @@ __global__ void k_fixpoint_v2(
     plen_s[tid] = prefix_len[tid];
     soff_s[tid] = seg_off[tid];
     ddelta[tid] = 0;
+    rlo_s[tid] = 0;
+    rhi_s[tid] = 0;
   }
   __syncthreads();
   for (;;) {
@@ -2353,35 +2357,51 @@ __global__ void k_fixpoint_v2(
       cap = maxw[tid] - weights[tid] + static_cast<i64>(ddep[tid]);
     }
     __syncthreads();
-    if (tid < k) {
-      u32 pl = plen_s[tid];
-      long long a = arr_s[tid];
+    if (tid < k && plen_s[tid] > 0 && arr_s[tid] > cap) {
+      const u32 old = plen_s[tid];
       const u32 b = soff_s[tid];
-      bool any = false;
-      while (a > cap && pl > 0) {
-        --pl;
-        const u32 i = b + pl;
-        const unsigned long long w = s_w[i];
-        a -= static_cast<long long>(w);
-        atomicAdd(&ddelta[s_b[i]], w);
-        any = true;
-      }
-      if (any) {
-        plen_s[tid] = pl;
+      u32 nl;
+      if (!has_vwgt) {
+        // unit weights: admitted weight of a prefix IS its length
+        nl = cap <= 0 ? 0u : (cap >= static_cast<i64>(old) ? old : static_cast<u32>(cap));
+        arr_s[tid] = static_cast<long long>(nl);
+      } else {
+        // weighted: serial backward walk (coarse graphs are small)
+        u32 pl = old;
+        long long a = arr_s[tid];
+        while (a > cap && pl > 0) {
+          --pl;
+          a -= static_cast<long long>(s_w[b + pl]);
+        }
+        nl = pl;
         arr_s[tid] = a;
-        chg = 1; // benign same-value race
       }
-    }
-    __syncthreads();
-    if (tid < k && ddelta[tid]) {
-      ddep[tid] -= ddelta[tid];
-      ddelta[tid] = 0;
+      plen_s[tid] = nl;
+      rlo_s[tid] = b + nl;
+      rhi_s[tid] = b + old;
+      chg = 1; // benign same-value race
     }
     __syncthreads();
     if (!chg) {
       break;
     }
-    __syncthreads(); // protect the chg reset of the next round
+    // block-parallel de-admission deltas (coalesced; O(total de-admitted))
+    for (u32 c = 0; c < k; ++c) {
+      const u32 lo2 = rlo_s[c], hi2 = rhi_s[c];
+      for (u32 i = lo2 + tid; i < hi2; i += blockDim.x) {
+        atomicAdd(&ddelta[s_b[i]], has_vwgt ? static_cast<unsigned long long>(s_w[i]) : 1ull);
+      }
+    }
+    __syncthreads();
+    if (tid < k) {
+      if (ddelta[tid]) {
+        ddep[tid] -= ddelta[tid];
+        ddelta[tid] = 0;
+      }
+      rlo_s[tid] = 0;
+      rhi_s[tid] = 0;
+    }
+    __syncthreads();
   }
 
   // weights + writeback + move count
@@ -2854,8 +2874,9 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
-      k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
-      e->d_dep, e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves
+      k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k,
+      static_cast<u32>(e->has_vwgt ? 1 : 0), e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_s_w,
+      e->d_s_b, e->d_weights, e->d_maxw, e->d_moves
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
