@@ -151,6 +151,7 @@ __device__ inline u32 hash_u32(u32 c) {
 // Owns the slot for: tail positions (u >= n), inactive or degree-filtered
 // vertices (invalid slot), and active deg <= 16 vertices (computed result).
 // Leaves deg in (16, inf) active slots for the M/L kernels.
+template <typename LT>
 __global__ void k_phase_s(
     u32 pos_lo,
     u32 pos_hi,
@@ -167,7 +168,7 @@ __global__ void k_phase_s(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    const uint16_t *__restrict__ labels16,
+    const LT *__restrict__ labels_s, // u8 shadow for k <= 256, else u16
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots
@@ -224,7 +225,8 @@ __global__ void k_phase_s(
   i32 w = 0;
   if (slot < deg) {
     const u32 v = adjncy[row + slot];
-    c = labels16[v]; // u16 shadow: k <= 2048, half the gather footprint
+    c = labels_s[v]; // narrow shadow: u8 for k <= 256 (64 MB at scale 26,
+                     // MALL-resident), u16 up to k <= 2048
     w = adjwgt ? adjwgt[row + slot] : 1;
   }
 
@@ -292,7 +294,7 @@ __global__ void k_phase_s(
 // LDS gains (k <= kMaxDenseK), ballot-waterfall accumulation. Active
 // deg > kMidDeg vertices were appended to the L list by k_phase_s.
 // blockDim.x = 256 (4 waves); dynamic LDS = 4 * k * sizeof(i32).
-template <bool kUnitWeights>
+template <bool kUnitWeights, typename LT>
 __global__ void k_phase_m(
     u32 pos_lo,
     u32 chunk_base,
@@ -305,7 +307,7 @@ __global__ void k_phase_m(
     const i32 *__restrict__ adjwgt,
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
-    const uint16_t *__restrict__ labels16,
+    const LT *__restrict__ labels_s,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const u64 *__restrict__ m_list,
@@ -337,7 +339,7 @@ __global__ void k_phase_m(
   for (u32 e = lane; e < deg; e += kWave) {
     const u32 v = adjncy[row + e];
     const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-    atomicAdd(&gains[rep_off + labels16[v]], w);
+    atomicAdd(&gains[rep_off + labels_s[v]], w);
   }
   __threadfence_block();
 
@@ -422,13 +424,13 @@ __global__ void k_l_sizes(
 }
 
 // Accumulate one slice per workgroup into the vertex's global gains row.
-template <bool kUnitWeights>
+template <bool kUnitWeights, typename LT>
 __global__ void k_phase_l_acc(
     u32 k,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
-    const uint16_t *__restrict__ labels16,
+    const LT *__restrict__ labels_s,
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     u32 l_cap,
@@ -467,7 +469,7 @@ __global__ void k_phase_l_acc(
     for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
       const u32 v = adjncy[row + e];
       const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-      atomicAdd(&hist[rep_off + labels16[v]], w);
+      atomicAdd(&hist[rep_off + labels_s[v]], w);
     }
     __syncthreads();
     i32 *grow = l_gains + static_cast<size_t>(vid) * k;
@@ -581,7 +583,7 @@ __global__ void k_phase_l_sel(
 
 // Fallback for L entries beyond l_cap (pathological): one workgroup per
 // vertex, whole row, replicated LDS histogram.
-template <bool kUnitWeights>
+template <bool kUnitWeights, typename LT>
 __global__ void k_phase_l_direct(
     u32 balance,
     u32 fallback,
@@ -594,7 +596,7 @@ __global__ void k_phase_l_direct(
     const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
-    const uint16_t *__restrict__ labels16,
+    const LT *__restrict__ labels_s,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const u64 *__restrict__ l_list,
@@ -624,7 +626,7 @@ __global__ void k_phase_l_direct(
     for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
       const u32 v = adjncy[row + e];
       const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
-      atomicAdd(&gains[rep_off + labels16[v]], w);
+      atomicAdd(&gains[rep_off + labels_s[v]], w);
     }
     __syncthreads();
 
@@ -1645,6 +1647,7 @@ __global__ void k_apply(
     const u32 *__restrict__ prefix_len,
     u32 *__restrict__ labels,
     uint16_t *__restrict__ labels16, // refine shadow; null for clustering
+    uint8_t *__restrict__ labels8,   // u8 shadow; null unless k <= 256
     u32 *__restrict__ admitted_flags, // per sorted index
     unsigned long long *__restrict__ moves
 ) {
@@ -1659,6 +1662,9 @@ __global__ void k_apply(
       labels[u] = to;
       if (labels16 != nullptr) {
         labels16[u] = static_cast<uint16_t>(to);
+      }
+      if (labels8 != nullptr) {
+        labels8[u] = static_cast<uint8_t>(to);
       }
       local = 1;
     }
@@ -1788,11 +1794,14 @@ __global__ void k_dep_reset_all(u32 k_or_n, unsigned long long *__restrict__ dep
 // cluster per WG instead of one per vertex -- same-address global atomics
 // serialize at ~11ns).
 __global__ void k_sync_labels16(
-    u32 n, const u32 *__restrict__ labels, uint16_t *__restrict__ labels16
+    u32 n, const u32 *__restrict__ labels, uint16_t *__restrict__ labels16,
+    uint8_t *__restrict__ labels8
 ) {
   const u32 u = blockIdx.x * blockDim.x + threadIdx.x;
   if (u < n) {
-    labels16[u] = static_cast<uint16_t>(labels[u]);
+    const u32 l = labels[u];
+    labels16[u] = static_cast<uint16_t>(l);
+    labels8[u] = static_cast<uint8_t>(l); // meaningful only for k <= 256
   }
 }
 
@@ -2440,7 +2449,8 @@ __global__ void k_apply_v2(
     const u32 *__restrict__ s_u,
     const uint16_t *__restrict__ s_to,
     u32 *__restrict__ labels,
-    uint16_t *__restrict__ labels16
+    uint16_t *__restrict__ labels16,
+    uint8_t *__restrict__ labels8 // null unless k <= 256
 ) {
   const u32 count = seg_off[k];
   const u32 stride = gridDim.x * blockDim.x;
@@ -2450,6 +2460,9 @@ __global__ void k_apply_v2(
       const u32 u = s_u[i];
       labels[u] = c;
       labels16[u] = static_cast<uint16_t>(c);
+      if (labels8 != nullptr) {
+        labels8[u] = static_cast<uint8_t>(c);
+      }
     }
   }
 }
@@ -2506,6 +2519,8 @@ struct kmp_lp_t {
   // device LP state
   u32 *d_labels = nullptr;
   u32 *d_labels0 = nullptr;   // initial labels (for kmp_lp_reset)
+  uint8_t *d_labels8 = nullptr;   // u8 shadow for refine gathers (k <= 256:
+                                  // 64 MB at scale 26, Infinity-Cache-resident)
   uint16_t *d_labels16 = nullptr; // u16 shadow for refine gathers (k <= 2048
                                   // fits; halves the gather cache footprint:
                                   // scale-26 labels 128 MB -> L3-resident)
@@ -2686,6 +2701,7 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_labels, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels0, sizeof(u32) * e->n));
   HIP_CHECK(hipMalloc(&e->d_labels16, sizeof(uint16_t) * e->n));
+  HIP_CHECK(hipMalloc(&e->d_labels8, e->n));
   HIP_CHECK(hipMalloc(&e->d_active, e->n));
   HIP_CHECK(hipMalloc(&e->d_unit_active, kmp::num_units(e->n)));
 
@@ -2794,19 +2810,19 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
     LAUNCH_CHECK();
   }
   hipLaunchKernelGGL(
-      k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+      k_phase_s<uint8_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, 0u, kInvalid,
       0xFFFFFFFFu, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-      e->d_maxw, e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
+      e->d_maxw, e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
   );
   LAUNCH_CHECK();
   {
     const size_t lds =
         static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
-    auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
+    auto *kern = e->has_adjwgt ? k_phase_m<false, uint8_t> : k_phase_m<true, uint8_t>;
     hipLaunchKernelGGL(
         kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed, 0u,
-        kInvalid, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
+        kInvalid, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels8,
         e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
     );
     LAUNCH_CHECK();
@@ -2819,10 +2835,10 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
     LAUNCH_CHECK();
     const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
     {
-      auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+      auto *kern = e->has_adjwgt ? k_phase_l_acc<false, uint8_t> : k_phase_l_acc<true, uint8_t>;
       hipLaunchKernelGGL(
           kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
-          e->d_adjwgt, e->d_labels16, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
+          e->d_adjwgt, e->d_labels8, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off, e->d_l_gains
       );
       LAUNCH_CHECK();
     }
@@ -2836,10 +2852,11 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       const size_t lds =
           ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
           16 * sizeof(i64);
-      auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
+      auto *kern =
+          e->has_adjwgt ? k_phase_l_direct<false, uint8_t> : k_phase_l_direct<true, uint8_t>;
       hipLaunchKernelGGL(
           kern, dim3(512), dim3(256), lds, e->stream, 0u, kInvalid, pos_lo, chunk_base, iseed,
-          e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16,
+          e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels8,
           e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
       );
       LAUNCH_CHECK();
@@ -2881,7 +2898,7 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
       k_apply_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
-      e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16
+      e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16, e->d_labels8
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
@@ -3012,7 +3029,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props,
                   (void *)e->d_m_list, (void *)e->d_m_count,
                   (void *)e->d_l_list, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
@@ -3123,7 +3140,7 @@ int kmp_lp_refine_begin(
     const u32 threads = 256;
     hipLaunchKernelGGL(
         k_sync_labels16, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
-        e->d_labels, e->d_labels16
+        e->d_labels, e->d_labels16, e->d_labels8
     );
     LAUNCH_CHECK();
     const size_t lds = static_cast<size_t>(k) * sizeof(unsigned long long);
@@ -3216,26 +3233,47 @@ i64 kmp_lp_phase_a(
       );
       LAUNCH_CHECK();
     }
+    const bool k8 = e->k <= 256;
     // S: 4 positions/wave (unit-gated; slots pre-marked invalid)
-    hipLaunchKernelGGL(
-        k_phase_s, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
-        dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
-        static_cast<u32>(e->balance), fallback, max_degree,
-        e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-        e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
-    );
+    if (k8) {
+      hipLaunchKernelGGL(
+          k_phase_s<uint8_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+          dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
+          static_cast<u32>(e->balance), fallback, max_degree,
+          e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
+          e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
+      );
+    } else {
+      hipLaunchKernelGGL(
+          k_phase_s<uint16_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
+          dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
+          static_cast<u32>(e->balance), fallback, max_degree,
+          e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
+          e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
+      );
+    }
     LAUNCH_CHECK();
     // one wave per M-listed vertex (grid-stride; list built by k_build_lists)
     {
       const size_t lds =
           static_cast<size_t>(threads / kWave) * e->k * gain_replicas(e->k) * sizeof(i32);
-      auto *kern = e->has_adjwgt ? k_phase_m<false> : k_phase_m<true>;
-      hipLaunchKernelGGL(
-          kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
-          static_cast<u32>(e->balance), fallback,
-          e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
-          e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
-      );
+      if (k8) {
+        auto *kern = e->has_adjwgt ? k_phase_m<false, uint8_t> : k_phase_m<true, uint8_t>;
+        hipLaunchKernelGGL(
+            kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
+            static_cast<u32>(e->balance), fallback,
+            e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels8,
+            e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+        );
+      } else {
+        auto *kern = e->has_adjwgt ? k_phase_m<false, uint16_t> : k_phase_m<true, uint16_t>;
+        hipLaunchKernelGGL(
+            kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
+            static_cast<u32>(e->balance), fallback,
+            e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
+            e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+        );
+      }
       LAUNCH_CHECK();
     }
     // L: slice-parallel accumulation over the (rare) high-degree list
@@ -3252,8 +3290,17 @@ i64 kmp_lp_phase_a(
           rocprim::plus<u32>(), e->stream
       ));
       const size_t hist_lds = static_cast<size_t>(e->k) * gain_replicas(e->k) * sizeof(i32);
-      {
-        auto *kern = e->has_adjwgt ? k_phase_l_acc<false> : k_phase_l_acc<true>;
+      if (k8) {
+        auto *kern = e->has_adjwgt ? k_phase_l_acc<false, uint8_t> : k_phase_l_acc<true, uint8_t>;
+        hipLaunchKernelGGL(
+            kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
+            e->d_adjwgt, e->d_labels8, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off,
+            e->d_l_gains
+        );
+        LAUNCH_CHECK();
+      } else {
+        auto *kern =
+            e->has_adjwgt ? k_phase_l_acc<false, uint16_t> : k_phase_l_acc<true, uint16_t>;
         hipLaunchKernelGGL(
             kern, dim3(2048), dim3(256), hist_lds, e->stream, e->k, e->d_xadj, e->d_adjncy,
             e->d_adjwgt, e->d_labels16, e->d_l_list, e->d_l_count, e->l_cap, e->d_l_off,
@@ -3273,13 +3320,25 @@ i64 kmp_lp_phase_a(
         const size_t lds =
             ((static_cast<size_t>(e->k) * gain_replicas(e->k) + 1) & ~1ull) * sizeof(i32) +
             16 * sizeof(i64);
-        auto *kern = e->has_adjwgt ? k_phase_l_direct<false> : k_phase_l_direct<true>;
-        hipLaunchKernelGGL(
-            kern, dim3(512), dim3(256), lds, e->stream,
-            static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
-            e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
-            e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
-        );
+        if (k8) {
+          auto *kern =
+              e->has_adjwgt ? k_phase_l_direct<false, uint8_t> : k_phase_l_direct<true, uint8_t>;
+          hipLaunchKernelGGL(
+              kern, dim3(512), dim3(256), lds, e->stream,
+              static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+              e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels8, e->d_weights,
+              e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+          );
+        } else {
+          auto *kern =
+              e->has_adjwgt ? k_phase_l_direct<false, uint16_t> : k_phase_l_direct<true, uint16_t>;
+          hipLaunchKernelGGL(
+              kern, dim3(512), dim3(256), lds, e->stream,
+              static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
+              e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
+              e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+          );
+        }
         LAUNCH_CHECK();
       }
     }
@@ -3538,7 +3597,8 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
     }
     hipLaunchKernelGGL(
         k_apply, dim3(grid), dim3(threads), 0, e->stream, order, props, sto, count, e->d_seg_begin,
-        e->d_prefix_len, e->d_labels, big_k ? nullptr : e->d_labels16, e->d_admitted_flags,
+        e->d_prefix_len, e->d_labels, big_k ? nullptr : e->d_labels16,
+        (big_k || e->k > 256) ? nullptr : e->d_labels8, e->d_admitted_flags,
         e->d_moves
     );
     LAUNCH_CHECK();
@@ -3624,7 +3684,7 @@ int kmp_lp_reset(kmp_lp_t *e) {
     const u32 threads = 256;
     hipLaunchKernelGGL(
         k_sync_labels16, dim3(ceil_div(e->n, threads)), dim3(threads), 0, e->stream, e->n,
-        e->d_labels, e->d_labels16
+        e->d_labels, e->d_labels16, e->d_labels8
     );
     LAUNCH_CHECK();
     const size_t lds = static_cast<size_t>(e->k) * sizeof(unsigned long long);
