@@ -161,6 +161,7 @@ __global__ void k_phase_s(
     u32 balance,
     u32 fallback,
     u32 max_degree,
+    u32 s_clear, // v2 path: S owns the chunk's active-flag clearing
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ vwgt,
@@ -169,7 +170,7 @@ __global__ void k_phase_s(
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     const LT *__restrict__ labels_s, // u8 shadow for k <= 256, else u16
-    const uint8_t *__restrict__ active,
+    uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     Prop *__restrict__ slots
 ) {
@@ -199,8 +200,14 @@ __global__ void k_phase_s(
   } else {
     row = xadj[u];
     deg = xadj[u + 1] - row;
-    if (!active[u] || deg > max_degree) {
+    const bool act = active[u] != 0;
+    if (!act || deg > max_degree) {
       skip = true;
+    }
+    // v2: clear the processed flag inline (replaces k_clear_active's pass;
+    // exactly the legacy condition: deg <= max_degree && active)
+    if (s_clear && slot == 0 && act && deg <= max_degree) {
+      active[u] = 0;
     }
   }
 
@@ -1953,12 +1960,15 @@ __global__ void k_build_lists(
     u64 *__restrict__ m_list,
     u32 *__restrict__ m_count,
     u64 *__restrict__ l_list,
-    u32 *__restrict__ l_count
+    u32 *__restrict__ l_count,
+    unsigned long long *__restrict__ arcs // null: no tally (legacy path)
 ) {
+  __shared__ unsigned long long arc_red[4];
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const u32 ta0 = pos_lo >> 6;                 // absolute unit-tile range
   const u32 T = ((pos_hi + 63) >> 6) - ta0;
+  unsigned long long my_arcs = 0;
   const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
   const u32 t0 = row * tpw;
   const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
@@ -1966,6 +1976,25 @@ __global__ void k_build_lists(
   for (u32 pass = 0; pass < 2; ++pass) {
     u32 m_base = 0, l_base = 0;
     if (pass == 1) {
+      // flush the arcs tally (one WG-aggregated atomic)
+      if (arcs != nullptr) {
+        for (int off = 32; off > 0; off >>= 1) {
+          my_arcs += __shfl_down(my_arcs, off, kWave);
+        }
+        __syncthreads();
+        if (lane == 0) {
+          arc_red[threadIdx.x >> 6] = my_arcs;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+          const unsigned long long t2 =
+              arc_red[0] + arc_red[1] + arc_red[2] + arc_red[3];
+          if (t2) {
+            atomicAdd(arcs, t2);
+          }
+        }
+        my_arcs = 0;
+      }
       // reserve (lane 0; contention = one atomic per wave-row per list)
       if (lane == 0) {
         if (m_cnt) {
@@ -1998,6 +2027,9 @@ __global__ void k_build_lists(
         if (deg <= max_degree) {
           is_m = deg > mid_lo && deg <= mid_hi;
           is_l = deg > mid_hi;
+          if (pass == 0 && arcs != nullptr) {
+            my_arcs += deg; // arcs-scanned tally (was k_clear_active's job)
+          }
         }
       }
       const unsigned long long mm = __ballot(is_m);
@@ -2328,6 +2360,10 @@ __global__ void k_scan_coop(
 __global__ void k_fixpoint_v2(
     u32 k,
     u32 has_vwgt,
+    u32 pos_lo,
+    u32 pos_hi,
+    u32 n,
+    u64 iter_seed,
     const u32 *__restrict__ seg_off,
     u32 *__restrict__ prefix_len,
     const unsigned long long *__restrict__ dep, // [0..k) out, [k..2k) in
@@ -2335,7 +2371,8 @@ __global__ void k_fixpoint_v2(
     const uint16_t *__restrict__ s_b,
     i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
-    unsigned long long *__restrict__ moves
+    unsigned long long *__restrict__ moves,
+    uint8_t *__restrict__ unit_active
 ) {
   __shared__ unsigned long long ddep[256];   // current departures
   __shared__ unsigned long long ddelta[256]; // this round's de-admissions
@@ -2438,6 +2475,82 @@ __global__ void k_fixpoint_v2(
     if (t) {
       atomicAdd(moves, t);
     }
+  }
+  // clear the chunk's unit-active bits (the per-vertex flags were cleared
+  // inline by k_phase_s; activation re-marks receiving units afterwards)
+  {
+    const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+    const u32 units = (pos_hi - pos_lo) >> 6;
+    const u32 u0 = pos_lo >> 6;
+    for (u32 uix = tid; uix < units; uix += blockDim.x) {
+      const u32 vb = fp(u0 + uix);
+      if (unit_active[vb]) {
+        unit_active[vb] = 0;
+      }
+    }
+  }
+}
+
+// Single-workgroup variant of the histogram scan for small chunks (entries
+// <= a few thousand; the resident-grid version's barriers dominate there).
+__global__ void k_scan_small(
+    u32 k,
+    u32 rows,
+    const u32 *__restrict__ histT,
+    u32 *__restrict__ offT,
+    u32 *__restrict__ seg_off,
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ dep,
+    int *__restrict__ changed2
+) {
+  __shared__ u32 red[5];
+  const u32 entries = k * rows;
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & (kWave - 1);
+  u32 carry = 0;
+  for (u32 base = 0; base < entries; base += blockDim.x) {
+    const u32 i = base + tid;
+    u32 v = i < entries ? histT[i] : 0;
+    u32 inc = v;
+    for (int off = 1; off < 64; off <<= 1) {
+      const u32 o = __shfl_up(inc, off, kWave);
+      if (lane >= static_cast<u32>(off)) {
+        inc += o;
+      }
+    }
+    __syncthreads();
+    if (lane == 63) {
+      red[tid >> 6] = inc;
+    }
+    __syncthreads();
+    u32 wbase = 0;
+    for (u32 w = 0; w < (tid >> 6); ++w) {
+      wbase += red[w];
+    }
+    if (i < entries) {
+      offT[i] = carry + wbase + inc - v; // exclusive
+    }
+    u32 tsum = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      tsum += red[w];
+    }
+    carry += tsum;
+    __syncthreads();
+  }
+  __threadfence_block();
+  __syncthreads();
+  for (u32 c = tid; c < k; c += blockDim.x) {
+    const u32 b = offT[c * rows];
+    const u32 e2 = (c + 1 < k) ? offT[(c + 1) * rows] : carry;
+    seg_off[c] = b;
+    prefix_len[c] = e2 - b;
+    dep[c] = 0;
+    dep[k + c] = 0;
+  }
+  if (tid == 0) {
+    seg_off[k] = carry;
+    changed2[0] = 0;
+    changed2[1] = 0;
   }
 }
 
@@ -2805,14 +2918,14 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
     hipLaunchKernelGGL(
         k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
         iseed, rows_bl, tpw, kSmallDeg, kMidDeg, 0xFFFFFFFFu, e->d_xadj, e->d_active,
-        e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+        e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count, e->d_arcs
     );
     LAUNCH_CHECK();
   }
   hipLaunchKernelGGL(
       k_phase_s<uint8_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, 0u, kInvalid,
-      0xFFFFFFFFu, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
+      0xFFFFFFFFu, 1u, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
       e->d_maxw, e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
   );
   LAUNCH_CHECK();
@@ -2867,8 +2980,16 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
 void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 span = pos_hi - pos_lo;
-  const u32 rows = e->rows_v2;
   const u32 T = span >> 6;
+  // clamp the histogram rows to the chunk's tile count (small chunks need
+  // neither the full table nor the resident-grid scan)
+  u32 rows = (T + 3) & ~3u;
+  if (rows < 4) {
+    rows = 4;
+  }
+  if (rows > e->rows_v2) {
+    rows = e->rows_v2;
+  }
   const u32 tpw = (T + rows - 1) / rows;
   const u32 threads = 256;
   const size_t lds_h = static_cast<size_t>(threads / kWave) * e->k * sizeof(u32);
@@ -2877,11 +2998,18 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       e->k, rows, tpw, e->d_slots, e->d_unit_active, e->d_histT
   );
   LAUNCH_CHECK();
-  hipLaunchKernelGGL(
-      k_scan_coop, dim3(e->coop_nblk), dim3(threads), 0, e->stream, e->k, rows, e->coop_nblk,
-      e->d_histT, e->d_offT, e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed,
-      e->d_blocksums, e->d_bar
-  );
+  if (e->k * rows <= 8192) {
+    hipLaunchKernelGGL(
+        k_scan_small, dim3(1), dim3(threads), 0, e->stream, e->k, rows, e->d_histT, e->d_offT,
+        e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed
+    );
+  } else {
+    hipLaunchKernelGGL(
+        k_scan_coop, dim3(e->coop_nblk), dim3(threads), 0, e->stream, e->k, rows, e->coop_nblk,
+        e->d_histT, e->d_offT, e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed,
+        e->d_blocksums, e->d_bar
+    );
+  }
   LAUNCH_CHECK();
   const size_t lds_sc = lds_h + static_cast<size_t>(2 * e->k) * sizeof(unsigned long long);
   hipLaunchKernelGGL(
@@ -2892,18 +3020,14 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
       k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k,
-      static_cast<u32>(e->has_vwgt ? 1 : 0), e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_s_w,
-      e->d_s_b, e->d_weights, e->d_maxw, e->d_moves
+      static_cast<u32>(e->has_vwgt ? 1 : 0), pos_lo, pos_hi, e->n, iseed, e->d_seg_off,
+      e->d_prefix_len, e->d_dep, e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves,
+      e->d_unit_active
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
       k_apply_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
       e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16, e->d_labels8
-  );
-  LAUNCH_CHECK();
-  hipLaunchKernelGGL(
-      k_clear_active, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, pos_lo,
-      pos_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active, e->d_arcs
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
@@ -3229,7 +3353,8 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
           iseed, rows_bl, tpw, kSmallDeg, kMidDeg, max_degree, e->d_xadj, e->d_active,
-          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count,
+          static_cast<unsigned long long *>(nullptr)
       );
       LAUNCH_CHECK();
     }
@@ -3239,7 +3364,7 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           k_phase_s<uint8_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
           dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
-          static_cast<u32>(e->balance), fallback, max_degree,
+          static_cast<u32>(e->balance), fallback, max_degree, 0u,
           e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
           e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
       );
@@ -3247,7 +3372,7 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           k_phase_s<uint16_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
           dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
-          static_cast<u32>(e->balance), fallback, max_degree,
+          static_cast<u32>(e->balance), fallback, max_degree, 0u,
           e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
           e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
       );
@@ -3351,7 +3476,8 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
           iseed, rows_bl, tpw, kSmallDeg, kClusterMidDeg, max_degree, e->d_xadj, e->d_active,
-          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count
+          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count,
+          static_cast<unsigned long long *>(nullptr)
       );
       LAUNCH_CHECK();
     }
