@@ -978,33 +978,66 @@ __global__ void k_l_prep_c(
     u32 *__restrict__ l_off,  // slice prefix
     u64 *__restrict__ l_hoff, // hash-region prefix
     u32 *__restrict__ l_hbits, // log2(region size)
+    u32 *__restrict__ l_ccnt, // per-vertex claimed-slot counts (zeroed here)
     unsigned long long *__restrict__ hacc_out
 ) {
+  __shared__ u64 red[17];
   const u32 count = *l_count < l_cap ? *l_count : l_cap;
-  for (u32 i = threadIdx.x; i < count; i += blockDim.x) {
-    const u32 u = static_cast<u32>(l_list[i]);
-    const u32 deg = xadj[u + 1] - xadj[u];
-    l_off[i] = (deg + kLSlice - 1) / kLSlice;
-    u32 bits = 11; // >= 2048 slots
-    while ((1u << bits) < 2 * deg) {
-      ++bits;
+  const u32 tid = threadIdx.x;
+  const u32 lane = tid & (kWave - 1);
+  u32 scarry = 0;
+  u64 hcarry = 0;
+  for (u32 base = 0; base < count; base += blockDim.x) {
+    const u32 i = base + tid;
+    u32 sc = 0;
+    u64 hs = 0;
+    if (i < count) {
+      const u32 u = static_cast<u32>(l_list[i]);
+      const u32 deg = xadj[u + 1] - xadj[u];
+      sc = (deg + kLSlice - 1) / kLSlice;
+      u32 bits = 11; // >= 2048 slots
+      while ((1u << bits) < 2 * deg) {
+        ++bits;
+      }
+      l_hbits[i] = bits;
+      l_ccnt[i] = 0;
+      hs = 1ull << bits;
     }
-    l_hbits[i] = bits;
+    // joint block exclusive scan: region sizes in the high 32 bits, slice
+    // counts in the low 32 (tile sums stay well under 2^32 each)
+    u64 inc = (hs << 32) | sc;
+    for (int off = 1; off < 64; off <<= 1) {
+      const u64 o = __shfl_up(static_cast<unsigned long long>(inc), off, kWave);
+      if (lane >= static_cast<u32>(off)) {
+        inc += o;
+      }
+    }
+    __syncthreads();
+    if (lane == 63) {
+      red[tid >> 6] = inc;
+    }
+    __syncthreads();
+    u64 wbase = 0;
+    for (u32 w = 0; w < (tid >> 6); ++w) {
+      wbase += red[w];
+    }
+    const u64 ex = wbase + inc - ((hs << 32) | sc);
+    if (i < count) {
+      l_off[i] = scarry + static_cast<u32>(ex & 0xFFFFFFFFu);
+      l_hoff[i] = hcarry + (ex >> 32);
+    }
+    u64 tsum = 0;
+    for (u32 w = 0; w < blockDim.x / kWave; ++w) {
+      tsum += red[w];
+    }
+    scarry += static_cast<u32>(tsum & 0xFFFFFFFFu);
+    hcarry += tsum >> 32;
+    __syncthreads();
   }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    u32 sacc = 0;
-    u64 hacc = 0;
-    for (u32 i = 0; i < count; ++i) {
-      const u32 c = l_off[i];
-      l_off[i] = sacc;
-      sacc += c;
-      l_hoff[i] = hacc;
-      hacc += 1ull << l_hbits[i];
-    }
-    l_off[count] = sacc;
-    l_hoff[count] = hacc;
-    *hacc_out = hacc;
+  if (tid == 0) {
+    l_off[count] = scarry;
+    l_hoff[count] = hcarry;
+    *hacc_out = hcarry;
   }
 }
 
@@ -1025,7 +1058,9 @@ __global__ void k_phase_l_acc_c(
     const u64 *__restrict__ l_hoff,
     const u32 *__restrict__ l_hbits,
     u32 *__restrict__ pool_keys,
-    i32 *__restrict__ pool_vals
+    i32 *__restrict__ pool_vals,
+    u32 *__restrict__ l_clist, // claimed-slot lists (region/2 per vertex)
+    u32 *__restrict__ l_ccnt   // per-vertex claimed counts
 ) {
   const u32 count = vid_hi;
   if (count <= vid_lo) {
@@ -1050,8 +1085,10 @@ __global__ void k_phase_l_acc_c(
     const u32 deg = xadj[u + 1] - row;
     const u32 e_lo = (s - l_off[vid]) * kLSlice;
     const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
-    u32 *hk = pool_keys + (l_hoff[vid] - hbase);
-    i32 *hv = pool_vals + (l_hoff[vid] - hbase);
+    const u64 roff = l_hoff[vid] - hbase;
+    u32 *hk = pool_keys + roff;
+    i32 *hv = pool_vals + roff;
+    u32 *cl = l_clist + (roff >> 1); // region sizes are even powers of two
     const u32 mask = (1u << l_hbits[vid]) - 1;
 
     for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
@@ -1068,6 +1105,11 @@ __global__ void k_phase_l_acc_c(
         if (kcur == kInvalid) {
           const u32 old = atomicCAS(&hk[slot], kInvalid, c);
           if (old == kInvalid || old == c) {
+            if (old == kInvalid) {
+              // record the claimed slot so selection/clearing is
+              // O(distinct) instead of O(region)
+              cl[atomicAdd(&l_ccnt[vid], 1u)] = slot;
+            }
             atomicAdd(&hv[slot], w);
             break;
           }
@@ -1096,6 +1138,8 @@ __global__ void k_phase_l_sel_c(
     const u32 *__restrict__ l_hbits,
     u32 *__restrict__ pool_keys,
     i32 *__restrict__ pool_vals,
+    const u32 *__restrict__ l_clist,
+    const u32 *__restrict__ l_ccnt,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots
 ) {
@@ -1108,17 +1152,17 @@ __global__ void k_phase_l_sel_c(
     const u32 cur = labels[u];
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
-    u32 *hk = pool_keys + (l_hoff[vid] - hbase);
-    i32 *hv = pool_vals + (l_hoff[vid] - hbase);
-    const u32 slots_n = 1u << l_hbits[vid];
+    const u64 roff = l_hoff[vid] - hbase;
+    u32 *hk = pool_keys + roff;
+    i32 *hv = pool_vals + roff;
+    const u32 *cl = l_clist + (roff >> 1);
+    const u32 claimed = l_ccnt[vid];
 
     BestState best{0, 0, 0, false};
     BestState fav{0, 0, 0, false};
-    for (u32 s = threadIdx.x; s < slots_n; s += blockDim.x) {
+    for (u32 j = threadIdx.x; j < claimed; j += blockDim.x) {
+      const u32 s = cl[j];
       const u32 c = hk[s];
-      if (c == kInvalid) {
-        continue;
-      }
       const i32 g = hv[s];
       hk[s] = kInvalid; // clear for the next chunk
       hv[s] = 0;
@@ -2697,6 +2741,8 @@ struct kmp_lp_t {
   uint8_t *d_eflag = nullptr;    // n (emptied-cluster flags)
   u32 *d_pool_keys = nullptr;    // pooled L hash
   i32 *d_pool_vals = nullptr;
+  u32 *d_l_clist = nullptr;      // claimed-slot lists (pool_slots/2)
+  u32 *d_l_ccnt = nullptr;       // C (per-L-vertex claimed counts)
   u64 pool_slots = 0;
   u64 *d_l_hoff = nullptr;       // l_cap+1
   u32 *d_l_hbits = nullptr;      // l_cap
@@ -3169,7 +3215,8 @@ void kmp_lp_free(kmp_lp_t *e) {
     }
   }
   for (void *p : {(void *)e->d_favored, (void *)e->d_eflag, (void *)e->d_pool_keys,
-                  (void *)e->d_pool_vals, (void *)e->d_l_hoff, (void *)e->d_l_hbits,
+                  (void *)e->d_pool_vals, (void *)e->d_l_clist, (void *)e->d_l_ccnt,
+                  (void *)e->d_l_hoff, (void *)e->d_l_hbits,
                   (void *)e->d_l_hacc, (void *)e->d_cand, (void *)e->d_cand2,
                   (void *)e->d_cfav, (void *)e->d_crank, (void *)e->d_cones,
                   (void *)e->d_cand_select_temp, (void *)e->d_cand_sort_temp,
@@ -3501,7 +3548,7 @@ i64 kmp_lp_phase_a(
     {
       hipLaunchKernelGGL(
           k_l_prep_c, dim3(1), dim3(1024), 0, e->stream, e->d_l_list, e->d_l_count, e->d_xadj,
-          e->l_cap, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_l_hacc
+          e->l_cap, e->d_l_off, e->d_l_hoff, e->d_l_hbits, e->d_l_ccnt, e->d_l_hacc
       );
       LAUNCH_CHECK();
       // read back the L count and the chunk's total hash-region demand:
@@ -3523,13 +3570,14 @@ i64 kmp_lp_phase_a(
         hipLaunchKernelGGL(
             kern, dim3(2048), dim3(256), 0, e->stream, e->d_xadj, e->d_adjncy, e->d_adjwgt,
             e->d_labels, e->d_l_list, lo, hi, e->d_l_off, e->d_l_hoff, e->d_l_hbits,
-            e->d_pool_keys, e->d_pool_vals
+            e->d_pool_keys, e->d_pool_vals, e->d_l_clist, e->d_l_ccnt
         );
         LAUNCH_CHECK();
         hipLaunchKernelGGL(
             k_phase_l_sel_c, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed,
             e->n, e->maxw_uniform, e->d_vwgt, e->d_labels, e->d_weights, e->d_l_list, lo, hi,
-            e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_favored, e->d_slots
+            e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_l_clist,
+            e->d_l_ccnt, e->d_favored, e->d_slots
         );
         LAUNCH_CHECK();
       };
@@ -3549,9 +3597,11 @@ i64 kmp_lp_phase_a(
             // a single region larger than the pool: grow it
             HIP_CHECK(hipFree(e->d_pool_keys));
             HIP_CHECK(hipFree(e->d_pool_vals));
+            HIP_CHECK(hipFree(e->d_l_clist));
             e->pool_slots = need_one;
             HIP_CHECK(hipMalloc(&e->d_pool_keys, sizeof(u32) * e->pool_slots));
             HIP_CHECK(hipMalloc(&e->d_pool_vals, sizeof(i32) * e->pool_slots));
+            HIP_CHECK(hipMalloc(&e->d_l_clist, sizeof(u32) * (e->pool_slots / 2)));
             HIP_CHECK(hipMemsetAsync(e->d_pool_keys, 0xFF, sizeof(u32) * e->pool_slots,
                                      e->stream));
             HIP_CHECK(hipMemsetAsync(e->d_pool_vals, 0, sizeof(i32) * e->pool_slots,
@@ -3989,6 +4039,8 @@ i64 kmp_lp_cluster(
     HIP_CHECK(hipMalloc(&e->d_pool_vals, sizeof(i32) * slots));
     HIP_CHECK(hipMemset(e->d_pool_keys, 0xFF, sizeof(u32) * slots));
     HIP_CHECK(hipMemset(e->d_pool_vals, 0, sizeof(i32) * slots));
+    HIP_CHECK(hipMalloc(&e->d_l_clist, sizeof(u32) * (slots / 2)));
+    HIP_CHECK(hipMalloc(&e->d_l_ccnt, sizeof(u32) * e->C));
     // two-hop buffers + temps
     HIP_CHECK(hipMalloc(&e->d_cand, sizeof(u64) * n));
     HIP_CHECK(hipMalloc(&e->d_cand2, sizeof(u64) * n));
