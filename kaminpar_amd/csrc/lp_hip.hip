@@ -2416,7 +2416,8 @@ __global__ void k_fixpoint_v2(
     i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     unsigned long long *__restrict__ moves,
-    uint8_t *__restrict__ unit_active
+    uint8_t *__restrict__ unit_active,
+    i64 *__restrict__ pw // scratch: segmented prefix weights (weighted only)
 ) {
   __shared__ unsigned long long ddep[256];   // current departures
   __shared__ unsigned long long ddelta[256]; // this round's de-admissions
@@ -2437,6 +2438,33 @@ __global__ void k_fixpoint_v2(
     rhi_s[tid] = 0;
   }
   __syncthreads();
+  if (has_vwgt) {
+    // build segmented prefix weights once (wave per target, tile scans) so
+    // per-round cutoffs are binary searches instead of serial walks
+    const u32 lane = tid & (kWave - 1);
+    const u32 wv = tid >> 6;
+    for (u32 c = wv; c < k; c += blockDim.x / kWave) {
+      const u32 b = soff_s[c];
+      const u32 len = plen_s[c];
+      i64 carry = 0;
+      for (u32 base = 0; base < len; base += kWave) {
+        const u32 i = base + lane;
+        i64 v = (i < len) ? static_cast<i64>(s_w[b + i]) : 0;
+        i64 inc = v;
+        for (int off = 1; off < 64; off <<= 1) {
+          const i64 o = __shfl_up(inc, off, kWave);
+          if (lane >= static_cast<u32>(off)) {
+            inc += o;
+          }
+        }
+        if (i < len) {
+          pw[b + i] = carry + inc;
+        }
+        carry += __shfl(inc, kWave - 1, kWave);
+      }
+    }
+    __syncthreads();
+  }
   for (;;) {
     if (tid == 0) {
       chg = 0;
@@ -2456,15 +2484,18 @@ __global__ void k_fixpoint_v2(
         nl = cap <= 0 ? 0u : (cap >= static_cast<i64>(old) ? old : static_cast<u32>(cap));
         arr_s[tid] = static_cast<long long>(nl);
       } else {
-        // weighted: serial backward walk (coarse graphs are small)
-        u32 pl = old;
-        long long a = arr_s[tid];
-        while (a > cap && pl > 0) {
-          --pl;
-          a -= static_cast<long long>(s_w[b + pl]);
+        // weighted: binary search the segmented prefix weights
+        u32 lo2 = 0, hi2 = old;
+        while (lo2 < hi2) {
+          const u32 mid = (lo2 + hi2 + 1) >> 1;
+          if (pw[b + mid - 1] <= cap) {
+            lo2 = mid;
+          } else {
+            hi2 = mid - 1;
+          }
         }
-        nl = pl;
-        arr_s[tid] = a;
+        nl = lo2;
+        arr_s[tid] = nl ? pw[b + nl - 1] : 0;
       }
       plen_s[tid] = nl;
       rlo_s[tid] = b + nl;
@@ -3068,7 +3099,7 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k,
       static_cast<u32>(e->has_vwgt ? 1 : 0), pos_lo, pos_hi, e->n, iseed, e->d_seg_off,
       e->d_prefix_len, e->d_dep, e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves,
-      e->d_unit_active
+      e->d_unit_active, e->d_pw
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
