@@ -471,6 +471,12 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     total_w = g.total_node_weight
     mbw_val = g.max_block_weight(k, eps)
 
+    import os as _os
+    import time as _time
+    _tdbg = _os.environ.get("KMP_TIME") == "1"
+    _tt = {"cluster": 0.0, "contract": 0.0, "extend": 0.0, "refine": 0.0,
+           "download": 0.0, "fm": 0.0, "other": 0.0}
+    _t0 = _time.perf_counter()
     sizes = [g.n]
     mappings = []
     engines = [engine if engine is not None else LpEngine(g)]
@@ -479,11 +485,15 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
     while sizes[-1] > max(stop_n, 2 * k):
         cur_n = sizes[-1]
         mcw = level_cluster_weight(total_w, cur_n, k, eps, contraction_limit)
+        _tc = _time.perf_counter()
         nc, clus, cst = engines[-1].cluster(mcw, seed=seed + len(mappings),
                                            iters=iters)
+        _tt["cluster"] += _time.perf_counter() - _tc
         arcs_total += cst.arcs_scanned
         ns_total += cst.phase_a_ns
+        _tc = _time.perf_counter()
         coarse_eng, mapping = engines[-1].contract_engine(clus)
+        _tt["contract"] += _time.perf_counter() - _tc
         if coarse_eng.n > 0.95 * cur_n:
             del coarse_eng
             break
@@ -504,28 +514,43 @@ def partition_deep(g, k, eps=0.03, seed=1, iters=5, contraction_limit=2000,
         hg = None
         if len(groups) < k and (sizes[level] >= 2 * sc * len(groups)
                                 or level == 0):
+            _tc = _time.perf_counter()
             hg = g if level == 0 else engines[level].download_graph()
+            _tt["download"] += _time.perf_counter() - _tc
+            _tc = _time.perf_counter()
             part, groups = _extend_partition(hg, part, groups, mbw_val, k,
                                              sc, reps,
                                              force=(level == 0))
             if len(groups) == k:
                 hg.balance_partition(k, mbw_val, part)
+            _tt["extend"] += _time.perf_counter() - _tc
         caps = _group_caps(groups, k, mbw_val)
+        _tc = _time.perf_counter()
         cut, part, rst = engines[level].refine(
             k, caps, part, seed=seed, iters=iters)
+        _tt["refine"] += _time.perf_counter() - _tc
         arcs_total += rst.arcs_scanned
         ns_total += rst.phase_a_ns
         # per-level k-way boundary FM on small graphs (<= ~2M fine
         # vertices): recovers the bisection quality LP refinement alone
         # cannot on mesh-like graphs (classic multilevel FM recipe)
         if g.n <= (1 << 21):
+            _tc = _time.perf_counter()
             if hg is None:
                 hg = g if level == 0 else engines[level].download_graph()
             part = hg.kway_fm(k, caps, part)
             if level == 0:
                 cut = g.edge_cut(part)
+            _tt["fm"] += _time.perf_counter() - _tc
         if level > 0:
             part = part[mappings[level - 1]]
+    if _tdbg:
+        _tt["other"] = (_time.perf_counter() - _t0) - sum(
+            v for q, v in _tt.items() if q != "other")
+        import sys as _sys
+        print("[deep-timing] " + " ".join(f"{q}={v:.2f}s"
+                                          for q, v in _tt.items()),
+              file=_sys.stderr)
     if return_arcs:
         return cut, part, sizes, int(arcs_total), int(ns_total)
     return cut, part, sizes
