@@ -714,8 +714,10 @@ __global__ void k_phase_l_direct(
 // deg <= 16 register waterfall, deg <= 512 per-wave LDS hash, larger rows
 // slice-parallel into a pooled global hash.
 
-constexpr u32 kClusterMidDeg = 512; // LDS-hash path bound
+constexpr u32 kClusterMidDeg = 512; // per-wave LDS-hash path bound
 constexpr u32 kHashSlots = 1024;    // per-wave LDS hash (8 KB per wave)
+constexpr u32 kClusterM2Deg = 1536; // per-WG LDS-hash path bound (load <= 0.375)
+constexpr u32 kM2HashSlots = 4096;  // per-WG LDS hash (32 KB per WG)
 
 // S path for clustering: identical structure to k_phase_s, clusterer accept
 // + favored-cluster argmax over all candidates.
@@ -962,6 +964,156 @@ __global__ void k_phase_m_c(
     }
   }
   __threadfence_block(); // hash reuse across grid-stride iterations
+  }
+}
+
+// M2 path for clustering: one 256-thread workgroup per vertex with a
+// 4096-slot LDS hash -- absorbs the dense coarse-level rows
+// (kClusterMidDeg < deg <= kClusterM2Deg) that otherwise take the pooled
+// GLOBAL hash path (measured dominant in the multilevel pipeline).
+template <bool kUnitWeights>
+__global__ void k_phase_m2_c(
+    u32 pos_lo,
+    u32 chunk_base,
+    u64 iter_seed,
+    i64 maxw_uniform,
+    const u32 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const i32 *__restrict__ vwgt,
+    const u32 *__restrict__ labels,
+    const i64 *__restrict__ weights,
+    const u64 *__restrict__ m2_list,
+    const u32 *__restrict__ m2_count,
+    u32 *__restrict__ favored,
+    Prop *__restrict__ slots
+) {
+  __shared__ u32 hkeys[kM2HashSlots];
+  __shared__ i32 hvals[kM2HashSlots];
+  __shared__ i64 red[24];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 count = *m2_count;
+  for (u32 vid = blockIdx.x; vid < count; vid += gridDim.x) {
+    const u64 rec = m2_list[vid];
+    const u32 p = static_cast<u32>(rec >> 32);
+    const u32 u = static_cast<u32>(rec);
+    const u32 row = xadj[u];
+    const u32 deg = xadj[u + 1] - row;
+
+    for (u32 s = threadIdx.x; s < kM2HashSlots; s += blockDim.x) {
+      hkeys[s] = kInvalid;
+      hvals[s] = 0;
+    }
+    __syncthreads();
+    for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
+      const u32 v = adjncy[row + e];
+      const u32 c = labels[v];
+      const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
+      u32 s = hash_u32(c) & (kM2HashSlots - 1);
+      while (true) {
+        const u32 kcur = hkeys[s];
+        if (kcur == c) {
+          atomicAdd(&hvals[s], w);
+          break;
+        }
+        if (kcur == kInvalid) {
+          const u32 old = atomicCAS(&hkeys[s], kInvalid, c);
+          if (old == kInvalid || old == c) {
+            atomicAdd(&hvals[s], w);
+            break;
+          }
+        }
+        s = (s + 1) & (kM2HashSlots - 1);
+      }
+    }
+    __syncthreads();
+
+    const u32 cur = labels[u];
+    const i32 u_w = vwgt ? vwgt[u] : 1;
+    const i64 cur_w = weights[cur];
+
+    BestState best{0, 0, 0, false};
+    BestState fav{0, 0, 0, false};
+    for (u32 s = threadIdx.x; s < kM2HashSlots; s += blockDim.x) {
+      const u32 c = hkeys[s];
+      if (c == kInvalid) {
+        continue;
+      }
+      const i32 g = hvals[s];
+      if (g <= 0) {
+        continue;
+      }
+      const u64 h = tie_hash(iter_seed, u, c);
+      if (key_better(g, h, c, fav)) {
+        fav = BestState{g, h, c, true};
+      }
+      if (accept_cluster(c, cur, u_w, weights[c], maxw_uniform) && key_better(g, h, c, best)) {
+        best = BestState{g, h, c, true};
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      {
+        const i32 og = __shfl_down(best.gain, off, kWave);
+        const u64 oh = __shfl_down(static_cast<unsigned long long>(best.h), off, kWave);
+        const u32 oc = __shfl_down(best.c, off, kWave);
+        const int ohave = __shfl_down(static_cast<int>(best.have), off, kWave);
+        if (ohave && key_better(og, oh, oc, best)) {
+          best = BestState{og, oh, oc, true};
+        }
+      }
+      {
+        const i32 og = __shfl_down(fav.gain, off, kWave);
+        const u64 oh = __shfl_down(static_cast<unsigned long long>(fav.h), off, kWave);
+        const u32 oc = __shfl_down(fav.c, off, kWave);
+        const int ohave = __shfl_down(static_cast<int>(fav.have), off, kWave);
+        if (ohave && key_better(og, oh, oc, fav)) {
+          fav = BestState{og, oh, oc, true};
+        }
+      }
+    }
+    const u32 wave_in_wg = threadIdx.x >> 6;
+    if (lane == 0) {
+      red[wave_in_wg * 3] = (static_cast<i64>(best.gain) << 2) | (best.have ? 1 : 0) |
+                            (fav.have ? 2 : 0);
+      red[wave_in_wg * 3 + 1] = static_cast<i64>(best.h);
+      red[wave_in_wg * 3 + 2] = static_cast<i64>(fav.h);
+      reinterpret_cast<u32 *>(red + 12)[wave_in_wg * 2] = best.c;
+      reinterpret_cast<u32 *>(red + 12)[wave_in_wg * 2 + 1] = fav.c;
+      reinterpret_cast<i32 *>(red + 16)[wave_in_wg] = fav.gain;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      BestState tb{0, 0, 0, false};
+      BestState tf{0, 0, 0, false};
+      const u32 waves = blockDim.x >> 6;
+      for (u32 wv = 0; wv < waves; ++wv) {
+        const i64 packed = red[wv * 3];
+        const i32 bg = static_cast<i32>(packed >> 2);
+        if (packed & 1) {
+          const u64 h = static_cast<u64>(red[wv * 3 + 1]);
+          const u32 c = reinterpret_cast<u32 *>(red + 12)[wv * 2];
+          if (key_better(bg, h, c, tb)) {
+            tb = BestState{bg, h, c, true};
+          }
+        }
+        if (packed & 2) {
+          const i32 fg = reinterpret_cast<i32 *>(red + 16)[wv];
+          const u64 h = static_cast<u64>(red[wv * 3 + 2]);
+          const u32 c = reinterpret_cast<u32 *>(red + 12)[wv * 2 + 1];
+          if (key_better(fg, h, c, tf)) {
+            tf = BestState{fg, h, c, true};
+          }
+        }
+      }
+      const bool store_favored = (u_w == cur_w) && (cur_w <= maxw_uniform / 2);
+      if (store_favored) {
+        favored[u] = tf.have ? tf.c : cur;
+      }
+      if (tb.have && tb.c != cur) {
+        slots[p - pos_lo] = Prop{u, tb.c, p - chunk_base, static_cast<u32>(u_w)};
+      }
+    }
+    __syncthreads();
   }
 }
 
@@ -1997,12 +2149,15 @@ __global__ void k_build_lists(
     u32 tpw,
     u32 mid_lo, // kSmallDeg
     u32 mid_hi, // kMidDeg (refine) / kClusterMidDeg (cluster)
+    u32 m2_hi,  // kClusterM2Deg (cluster; == mid_hi disables the M2 tier)
     u32 max_degree,
     const u32 *__restrict__ xadj,
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     u64 *__restrict__ m_list,
     u32 *__restrict__ m_count,
+    u64 *__restrict__ m2_list,
+    u32 *__restrict__ m2_count,
     u64 *__restrict__ l_list,
     u32 *__restrict__ l_count,
     unsigned long long *__restrict__ arcs // null: no tally (legacy path)
@@ -2016,9 +2171,9 @@ __global__ void k_build_lists(
   const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
   const u32 t0 = row * tpw;
   const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
-  u32 m_cnt = 0, l_cnt = 0;
+  u32 m_cnt = 0, m2_cnt = 0, l_cnt = 0;
   for (u32 pass = 0; pass < 2; ++pass) {
-    u32 m_base = 0, l_base = 0;
+    u32 m_base = 0, m2_base = 0, l_base = 0;
     if (pass == 1) {
       // flush the arcs tally (one WG-aggregated atomic)
       if (arcs != nullptr) {
@@ -2044,13 +2199,17 @@ __global__ void k_build_lists(
         if (m_cnt) {
           m_base = atomicAdd(m_count, m_cnt);
         }
+        if (m2_cnt) {
+          m2_base = atomicAdd(m2_count, m2_cnt);
+        }
         if (l_cnt) {
           l_base = atomicAdd(l_count, l_cnt);
         }
       }
       m_base = __shfl(m_base, 0, kWave);
+      m2_base = __shfl(m2_base, 0, kWave);
       l_base = __shfl(l_base, 0, kWave);
-      if (m_cnt == 0 && l_cnt == 0) {
+      if (m_cnt == 0 && m2_cnt == 0 && l_cnt == 0) {
         return;
       }
     }
@@ -2065,25 +2224,32 @@ __global__ void k_build_lists(
       }
       const u32 p = ((ta0 + t) << 6) + lane;
       const u32 u = vb * kmp::kUnit + lane;
-      bool is_m = false, is_l = false;
+      bool is_m = false, is_m2 = false, is_l = false;
       if (p >= pos_lo && p < pos_hi && u < n && active[u]) {
         const u32 deg = xadj[u + 1] - xadj[u];
         if (deg <= max_degree) {
           is_m = deg > mid_lo && deg <= mid_hi;
-          is_l = deg > mid_hi;
+          is_m2 = deg > mid_hi && deg <= m2_hi;
+          is_l = deg > m2_hi;
           if (pass == 0 && arcs != nullptr) {
             my_arcs += deg; // arcs-scanned tally (was k_clear_active's job)
           }
         }
       }
       const unsigned long long mm = __ballot(is_m);
+      const unsigned long long mm2 = __ballot(is_m2);
       const unsigned long long ll = __ballot(is_l);
       if (pass == 0) {
         m_cnt += __popcll(mm);
+        m2_cnt += __popcll(mm2);
         l_cnt += __popcll(ll);
       } else {
         if (is_m) {
           m_list[m_base + __popcll(mm & ((1ull << lane) - 1))] =
+              (static_cast<u64>(p) << 32) | u;
+        }
+        if (is_m2) {
+          m2_list[m2_base + __popcll(mm2 & ((1ull << lane) - 1))] =
               (static_cast<u64>(p) << 32) | u;
         }
         if (is_l) {
@@ -2091,6 +2257,7 @@ __global__ void k_build_lists(
               (static_cast<u64>(p) << 32) | u;
         }
         m_base += __popcll(mm);
+        m2_base += __popcll(mm2);
         l_base += __popcll(ll);
       }
     }
@@ -2720,10 +2887,12 @@ struct kmp_lp_t {
   // phase buffers
   Prop *d_slots = nullptr; // C
   Prop *d_props = nullptr; // C (compacted; legacy/sharded commit input)
-  u64 *d_m_list = nullptr;  // C (wave-aggregated append by the S kernels)
-  u32 *d_m_count = nullptr; // base of a 2-u32 alloc; [1] is d_l_count
+  u64 *d_m_list = nullptr;  // C (built by k_build_lists)
+  u32 *d_m_count = nullptr; // base of a 4-u32 alloc; [1]=l, [2]=m2
   u64 *d_l_list = nullptr; // C
   u32 *d_l_count = nullptr; // = d_m_count + 1
+  u64 *d_m2_list = nullptr; // C (clustering per-WG hash tier)
+  u32 *d_m2_count = nullptr; // = d_m_count + 2
   u32 *d_l_off = nullptr;   // l_cap + 1 (slice prefix)
   u32 *d_l_sizes = nullptr; // l_cap + 1 (slice counts, scan input)
   void *d_lscan_temp = nullptr;
@@ -2899,9 +3068,11 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_slots, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_props, sizeof(Prop) * C));
   HIP_CHECK(hipMalloc(&e->d_m_list, sizeof(u64) * C));
-  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32) * 2)); // [0]=m, [1]=l
+  HIP_CHECK(hipMalloc(&e->d_m_count, sizeof(u32) * 4)); // [0]=m, [1]=l, [2]=m2
   e->d_l_count = e->d_m_count + 1;
+  e->d_m2_count = e->d_m_count + 2;
   HIP_CHECK(hipMalloc(&e->d_l_list, sizeof(u64) * C));
+  HIP_CHECK(hipMalloc(&e->d_m2_list, sizeof(u64) * C));
 
   // commit v2 buffers (sized for the largest supported k = 256 rows layout)
   HIP_CHECK(hipMalloc(&e->d_s_u, sizeof(u32) * C));
@@ -2987,15 +3158,16 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   const u32 chunk_base = pos_lo;
   const u32 span = pos_hi - pos_lo;
   const u32 threads = 256;
-  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 2, e->stream));
+  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 3, e->stream));
   {
     const u32 rows_bl = 4096;
     const u32 T_bl = ((pos_hi + 63) >> 6) - (pos_lo >> 6);
     const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
     hipLaunchKernelGGL(
         k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
-        iseed, rows_bl, tpw, kSmallDeg, kMidDeg, 0xFFFFFFFFu, e->d_xadj, e->d_active,
-        e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count, e->d_arcs
+        iseed, rows_bl, tpw, kSmallDeg, kMidDeg, kMidDeg, 0xFFFFFFFFu, e->d_xadj, e->d_active,
+        e->d_unit_active, e->d_m_list, e->d_m_count, e->d_m2_list, e->d_m2_count, e->d_l_list,
+        e->d_l_count, e->d_arcs
     );
     LAUNCH_CHECK();
   }
@@ -3232,7 +3404,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
                   (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props,
-                  (void *)e->d_m_list, (void *)e->d_m_count,
+                  (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m2_list,
                   (void *)e->d_l_list, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
                   (void *)e->d_moves, (void *)e->d_sort_keys[0], (void *)e->d_sort_keys[1],
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
@@ -3398,7 +3570,7 @@ i64 kmp_lp_phase_a(
   const u32 threads = 256;
   const u32 max_degree = 0xFFFFFFFFu;
 
-  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 2, e->stream)); // m+l counts
+  HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 3, e->stream)); // m+l counts
   // pre-mark every slot invalid (the legacy/sharded/clusterer commit reads
   // compacted proposals; the clusterer S kernel does not always-write)
   HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * span, e->stream));
@@ -3430,9 +3602,9 @@ i64 kmp_lp_phase_a(
       const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
       hipLaunchKernelGGL(
           k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
-          iseed, rows_bl, tpw, kSmallDeg, kMidDeg, max_degree, e->d_xadj, e->d_active,
-          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count,
-          static_cast<unsigned long long *>(nullptr)
+          iseed, rows_bl, tpw, kSmallDeg, kMidDeg, kMidDeg, max_degree, e->d_xadj, e->d_active,
+          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_m2_list, e->d_m2_count,
+          e->d_l_list, e->d_l_count, static_cast<unsigned long long *>(nullptr)
       );
       LAUNCH_CHECK();
     }
@@ -3553,8 +3725,9 @@ i64 kmp_lp_phase_a(
       const u32 tpw = (T_bl + rows_bl - 1) / rows_bl;
       hipLaunchKernelGGL(
           k_build_lists, dim3(rows_bl / 4), dim3(threads), 0, e->stream, pos_lo, pos_hi, e->n,
-          iseed, rows_bl, tpw, kSmallDeg, kClusterMidDeg, max_degree, e->d_xadj, e->d_active,
-          e->d_unit_active, e->d_m_list, e->d_m_count, e->d_l_list, e->d_l_count,
+          iseed, rows_bl, tpw, kSmallDeg, kClusterMidDeg, kClusterM2Deg, max_degree, e->d_xadj,
+          e->d_active, e->d_unit_active, e->d_m_list, e->d_m_count, e->d_m2_list,
+          e->d_m2_count, e->d_l_list, e->d_l_count,
           static_cast<unsigned long long *>(nullptr)
       );
       LAUNCH_CHECK();
@@ -3573,6 +3746,13 @@ i64 kmp_lp_phase_a(
           kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, iseed,
           e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels,
           e->d_weights, e->d_m_list, e->d_m_count, e->d_favored, e->d_slots
+      );
+      LAUNCH_CHECK();
+      auto *kern2 = e->has_adjwgt ? k_phase_m2_c<false> : k_phase_m2_c<true>;
+      hipLaunchKernelGGL(
+          kern2, dim3(1024), dim3(threads), 0, e->stream, pos_lo, chunk_base, iseed,
+          e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels,
+          e->d_weights, e->d_m2_list, e->d_m2_count, e->d_favored, e->d_slots
       );
       LAUNCH_CHECK();
     }
