@@ -132,6 +132,22 @@ int64_t kmp_lp_refine(
  * a candidate, so overloaded blocks shed boundary vertices to the best
  * admissible targets even at negative gain. Use before kmp_lp_refine when
  * the input partition may violate the caps. Returns the edge cut, or -1. */
+/* Underload-balancer mode (presets.cc:332-338 UNDERLOAD_BALANCER;
+ * refinement/balancer/underload_balancer.cc semantics): fill blocks below
+ * min_block_weights[b] with best-gain admissible vertices, never dropping a
+ * source below its own minimum and never overshooting any maximum. A no-op
+ * when all minima are already satisfied (the reference's refine() gate). */
+int64_t kmp_lp_underload(
+    kmp_lp_t *e,
+    uint32_t k,
+    const int64_t *max_block_weights,
+    const int64_t *min_block_weights,
+    uint32_t *partition,
+    uint64_t seed,
+    int iters,
+    kmp_lp_stats_t *stats
+);
+
 int64_t kmp_lp_balance(
     kmp_lp_t *e,
     uint32_t k,

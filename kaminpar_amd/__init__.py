@@ -122,6 +122,9 @@ def _load():
     lib.kmp_lp_refine.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_balance.restype = i64
     lib.kmp_lp_balance.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
+    lib.kmp_lp_underload.restype = i64
+    lib.kmp_lp_underload.argtypes = [vp, u32, p(i64), p(i64), p(u32), u64,
+                                     ctypes.c_int, vp]
     lib.kmp_lp_cluster.restype = i64
     lib.kmp_lp_cluster.argtypes = [vp, i64, u32, p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_num_chunks.restype = u32
@@ -407,6 +410,24 @@ class LpEngine:
         )
         if cut < 0:
             raise RuntimeError("kmp_lp_balance failed")
+        return cut, part, stats
+
+    def underload(self, k, max_block_weights, min_block_weights, partition,
+                  seed=1, iters=5):
+        """Underload-balancer mode: fill blocks below their minimum weight
+        (presets.cc:332-338 UNDERLOAD_BALANCER role; semantics restated from
+        refinement/balancer/underload_balancer.cc). Returns
+        (cut, partition, Stats)."""
+        part = np.ascontiguousarray(partition, dtype=np.uint32).copy()
+        mbw = np.ascontiguousarray(max_block_weights, dtype=np.int64)
+        mnw = np.ascontiguousarray(min_block_weights, dtype=np.int64)
+        stats = Stats()
+        cut = _lib.kmp_lp_underload(
+            self._h, k, _i64p(mbw), _i64p(mnw), _u32p(part), seed, iters,
+            ctypes.byref(stats)
+        )
+        if cut < 0:
+            raise RuntimeError("kmp_lp_underload failed")
         return cut, part, stats
 
     def cluster(self, max_cluster_weight, clustering=None, desired=0, seed=1, iters=5):

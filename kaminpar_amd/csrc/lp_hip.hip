@@ -137,6 +137,12 @@ __device__ inline bool accept_refine(
   return (cw + u_w <= maxw) || ((cw - maxw) < (cur_w - cur_maxw));
 }
 
+// Underload-balancer acceptance (underload_balancer.cc is_movable_to): only
+// underloaded targets with room; the current block is never a candidate.
+__device__ inline bool accept_underload(u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw_c, i64 minw_c) {
+  return c != cur && cw < minw_c && cw + u_w <= maxw_c;
+}
+
 // Weight-acceptance predicate (clusterer variant, lp_clusterer.cc:199-204).
 __device__ inline bool accept_cluster(u32 c, u32 cur, i32 u_w, i64 cw, i64 maxw_uniform) {
   return (cw + u_w <= maxw_uniform) || (c == cur);
@@ -158,7 +164,7 @@ __global__ void k_phase_s(
     u32 chunk_base,
     u32 n,
     u64 iter_seed,
-    u32 balance,
+    u32 mode, // 0 = refine, 1 = overload balance, 2 = underload balance
     u32 fallback,
     u32 max_degree,
     u32 s_clear, // v2 path: S owns the chunk's active-flag clearing
@@ -169,6 +175,7 @@ __global__ void k_phase_s(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
+    const i64 *__restrict__ minw, // per-block minimums (mode 2 only)
     const LT *__restrict__ labels_s, // u8 shadow for k <= 256, else u16
     uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
@@ -256,13 +263,21 @@ __global__ void k_phase_s(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0;
+  const bool excl_cur = mode == 1;
+  bool vgate = true;
+  if (mode == 2) {
+    const i64 mnw = minw[cur];
+    vgate = cur_w >= mnw && cur_w - u_w >= mnw; // is_movable_from
+  }
 
   BestState best{0, 0, 0, false};
-  if (owner && c != kInvalid) {
+  if (owner && c != kInvalid && vgate) {
     const i64 cw = weights[c];
     const i64 mw = maxw[c];
-    if (accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
+    const bool ok = (mode == 2)
+                        ? accept_underload(c, cur, u_w, cw, mw, minw[c])
+                        : accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur);
+    if (ok) {
       best = BestState{gain, tie_hash(iter_seed, u, c), c, true};
     }
   }
@@ -280,7 +295,7 @@ __global__ void k_phase_s(
 
   if (slot == 0) {
     BestState fin = best;
-    if (!fin.have && excl_cur && cur_w > cur_maxw && fallback != kInvalid &&
+    if (!fin.have && mode == 1 && cur_w > cur_maxw && fallback != kInvalid &&
         fallback != cur) {
       // balance fallback: no admissible adjacent target (e.g. everything in
       // one block) -- propose the lightest block with room at gain 0
@@ -307,7 +322,7 @@ __global__ void k_phase_m(
     u32 chunk_base,
     u32 k,
     u64 iter_seed,
-    u32 balance,
+    u32 mode,
     u32 fallback,
     const u32 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
@@ -317,6 +332,7 @@ __global__ void k_phase_m(
     const LT *__restrict__ labels_s,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
+    const i64 *__restrict__ minw,
     const u64 *__restrict__ m_list,
     const u32 *__restrict__ m_count,
     Prop *__restrict__ slots
@@ -354,10 +370,15 @@ __global__ void k_phase_m(
   const i32 u_w = vwgt ? vwgt[u] : 1;
   const i64 cur_w = weights[cur];
   const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0;
+  const bool excl_cur = mode == 1;
+  bool vgate = true;
+  if (mode == 2) {
+    const i64 mnw = minw[cur];
+    vgate = cur_w >= mnw && cur_w - u_w >= mnw;
+  }
 
   BestState best{0, 0, 0, false};
-  for (u32 c = lane; c < k; c += kWave) {
+  for (u32 c = lane; vgate && c < k; c += kWave) {
     i32 g = gains[c];
     for (u32 r = 1; r < R; ++r) {
       g += gains[r * k + c];
@@ -367,7 +388,10 @@ __global__ void k_phase_m(
     }
     const i64 cw = weights[c];
     const i64 mw = maxw[c];
-    if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
+    const bool ok = (mode == 2)
+                        ? accept_underload(c, cur, u_w, cw, mw, minw[c])
+                        : accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur);
+    if (!ok) {
       continue;
     }
     const u64 h = tie_hash(iter_seed, u, c);
@@ -388,7 +412,7 @@ __global__ void k_phase_m(
 
   if (lane == 0) {
     BestState fin = best;
-    if (!fin.have && excl_cur && cur_w > cur_maxw && fallback != kInvalid &&
+    if (!fin.have && mode == 1 && cur_w > cur_maxw && fallback != kInvalid &&
         fallback != cur) {
       fin = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
     }
@@ -496,7 +520,7 @@ __global__ void k_phase_l_acc(
 // Select per L vertex (one workgroup each, grid-stride), write the slot,
 // and reset the gains row for the next chunk.
 __global__ void k_phase_l_sel(
-    u32 balance,
+    u32 mode,
     u32 fallback,
     u32 pos_lo,
     u32 chunk_base,
@@ -507,6 +531,7 @@ __global__ void k_phase_l_sel(
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
+    const i64 *__restrict__ minw,
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     u32 l_cap,
@@ -523,19 +548,27 @@ __global__ void k_phase_l_sel(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0;
+    const bool excl_cur = mode == 1;
+    bool vgate = true;
+    if (mode == 2) {
+      const i64 mnw = minw[cur];
+      vgate = cur_w >= mnw && cur_w - u_w >= mnw;
+    }
     i32 *grow = l_gains + static_cast<size_t>(vid) * k;
 
     BestState best{0, 0, 0, false};
     for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
       const i32 g = grow[c];
       grow[c] = 0; // reset for the next chunk
-      if (g <= 0) {
+      if (g <= 0 || !vgate) {
         continue;
       }
       const i64 cw = weights[c];
       const i64 mw = maxw[c];
-      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
+      const bool ok = (mode == 2)
+                          ? accept_underload(c, cur, u_w, cw, mw, minw[c])
+                          : accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur);
+      if (!ok) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -574,7 +607,7 @@ __global__ void k_phase_l_sel(
           }
         }
       }
-      if (!total.have && excl_cur && cur_w > cur_maxw &&
+      if (!total.have && mode == 1 && cur_w > cur_maxw &&
           fallback != kInvalid && fallback != cur) {
         total = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
       }
@@ -592,7 +625,7 @@ __global__ void k_phase_l_sel(
 // vertex, whole row, replicated LDS histogram.
 template <bool kUnitWeights, typename LT>
 __global__ void k_phase_l_direct(
-    u32 balance,
+    u32 mode,
     u32 fallback,
     u32 pos_lo,
     u32 chunk_base,
@@ -606,6 +639,7 @@ __global__ void k_phase_l_direct(
     const LT *__restrict__ labels_s,
     const i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
+    const i64 *__restrict__ minw,
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
     u32 l_cap,
@@ -641,10 +675,15 @@ __global__ void k_phase_l_direct(
     const i32 u_w = vwgt ? vwgt[u] : 1;
     const i64 cur_w = weights[cur];
     const i64 cur_maxw = maxw[cur];
-  const bool excl_cur = balance != 0;
+    const bool excl_cur = mode == 1;
+    bool vgate = true;
+    if (mode == 2) {
+      const i64 mnw = minw[cur];
+      vgate = cur_w >= mnw && cur_w - u_w >= mnw;
+    }
 
     BestState best{0, 0, 0, false};
-    for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    for (u32 c = threadIdx.x; vgate && c < k; c += blockDim.x) {
       i32 g = gains[c];
       for (u32 r = 1; r < R; ++r) {
         g += gains[r * k + c];
@@ -654,7 +693,10 @@ __global__ void k_phase_l_direct(
       }
       const i64 cw = weights[c];
       const i64 mw = maxw[c];
-      if (!accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur)) {
+      const bool ok = (mode == 2)
+                          ? accept_underload(c, cur, u_w, cw, mw, minw[c])
+                          : accept_refine(c, cur, u_w, cw, mw, cur_w, cur_maxw, excl_cur);
+      if (!ok) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -693,7 +735,7 @@ __global__ void k_phase_l_direct(
           }
         }
       }
-      if (!total.have && excl_cur && cur_w > cur_maxw &&
+      if (!total.have && mode == 1 && cur_w > cur_maxw &&
           fallback != kInvalid && fallback != cur) {
         total = BestState{0, tie_hash(iter_seed, u, fallback), fallback, true};
       }
@@ -1956,7 +1998,7 @@ __global__ void k_activate(
   if (wave_id >= count || !admitted_flags[wave_id]) {
     return;
   }
-  const u32 u = props[order[wave_id]].u;
+  const u32 u = props[order ? order[wave_id] : wave_id].u;
   const u32 row = xadj[u];
   const u32 deg = xadj[u + 1] - row;
   for (u32 e = lane; e < deg; e += kWave) {
@@ -2796,6 +2838,53 @@ __global__ void k_scan_small(
   }
 }
 
+// Serial rank-order admission for underload-balancer mode: both per-block
+// minima (source side) and maxima (target side) are re-checked against live
+// weights per move, in deterministic rank order -- the batch analogue of
+// underload_balancer.cc's locked per-move checks. Proposal counts in this
+// mode are small (only vertices adjacent to underloaded blocks propose).
+__global__ void k_commit_underload(
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ count_dev,
+    const i64 *__restrict__ maxw,
+    const i64 *__restrict__ minw,
+    i64 *__restrict__ weights,
+    u32 *__restrict__ labels,
+    uint16_t *__restrict__ labels16,
+    uint8_t *__restrict__ labels8, // null unless k <= 256
+    u32 *__restrict__ admitted_flags,
+    unsigned long long *__restrict__ moves
+) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) {
+    return;
+  }
+  const u32 count = *count_dev;
+  unsigned long long mv = 0;
+  for (u32 i = 0; i < count; ++i) {
+    const Prop pr = props[i];
+    const u32 from = labels[pr.u];
+    const i64 w = static_cast<i64>(static_cast<i32>(pr.w));
+    const i64 mn_from = minw[from];
+    const i64 mn_to = minw[pr.to];
+    const bool ok = weights[from] >= mn_from && weights[from] - w >= mn_from &&
+                    weights[pr.to] < mn_to && weights[pr.to] + w <= maxw[pr.to];
+    admitted_flags[i] = ok ? 1u : 0u;
+    if (ok) {
+      weights[from] -= w;
+      weights[pr.to] += w;
+      labels[pr.u] = pr.to;
+      labels16[pr.u] = static_cast<uint16_t>(pr.to);
+      if (labels8 != nullptr) {
+        labels8[pr.u] = static_cast<uint8_t>(pr.to);
+      }
+      ++mv;
+    }
+  }
+  if (mv) {
+    atomicAdd(moves, mv);
+  }
+}
+
 // Apply admitted labels (full grid, count read from device).
 __global__ void k_apply_v2(
     u32 k,
@@ -2881,6 +2970,7 @@ struct kmp_lp_t {
                                   // scale-26 labels 128 MB -> L3-resident)
   i64 *d_weights = nullptr;
   i64 *d_maxw = nullptr;
+  i64 *d_minw = nullptr; // per-block minimums (underload mode; else null)
   uint8_t *d_active = nullptr;
   uint8_t *d_unit_active = nullptr; // one byte per 64-vertex unit
 
@@ -3175,7 +3265,7 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       k_phase_s<uint8_t>, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
       dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, 0u, kInvalid,
       0xFFFFFFFFu, 1u, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-      e->d_maxw, e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
+      e->d_maxw, e->d_minw, e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
   );
   LAUNCH_CHECK();
   {
@@ -3185,7 +3275,7 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
     hipLaunchKernelGGL(
         kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed, 0u,
         kInvalid, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels8,
-        e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+        e->d_weights, e->d_maxw, e->d_minw, e->d_m_list, e->d_m_count, e->d_slots
     );
     LAUNCH_CHECK();
   }
@@ -3206,8 +3296,8 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
     }
     hipLaunchKernelGGL(
         k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream, 0u, kInvalid, pos_lo, chunk_base,
-        iseed, e->k, e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list,
-        e->d_l_count, e->l_cap, e->d_l_gains, e->d_slots
+        iseed, e->k, e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_minw,
+        e->d_l_list, e->d_l_count, e->l_cap, e->d_l_gains, e->d_slots
     );
     LAUNCH_CHECK();
     {
@@ -3219,7 +3309,7 @@ void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
       hipLaunchKernelGGL(
           kern, dim3(512), dim3(256), lds, e->stream, 0u, kInvalid, pos_lo, chunk_base, iseed,
           e->k, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels8,
-          e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+          e->d_weights, e->d_maxw, e->d_minw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
       );
       LAUNCH_CHECK();
     }
@@ -3402,7 +3492,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_active, (void *)e->d_unit_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_minw, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props,
                   (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m2_list,
                   (void *)e->d_l_list, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
@@ -3583,7 +3673,7 @@ i64 kmp_lp_phase_a(
     // balance mode: per-chunk fallback target = lightest block with room
     // (for overloaded vertices with no admissible adjacent candidate)
     u32 fallback = kInvalid;
-    if (e->balance) {
+    if (e->balance == 1) {
       std::vector<i64> w(e->k);
       HIP_CHECK(hipMemcpyAsync(w.data(), e->d_weights, sizeof(i64) * e->k,
                                hipMemcpyDeviceToHost, e->stream));
@@ -3616,7 +3706,7 @@ i64 kmp_lp_phase_a(
           dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
           static_cast<u32>(e->balance), fallback, max_degree, 0u,
           e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-          e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
+          e->d_minw, e->d_labels8, e->d_active, e->d_unit_active, e->d_slots
       );
     } else {
       hipLaunchKernelGGL(
@@ -3624,7 +3714,7 @@ i64 kmp_lp_phase_a(
           dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed,
           static_cast<u32>(e->balance), fallback, max_degree, 0u,
           e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights, e->d_maxw,
-          e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
+          e->d_minw, e->d_labels16, e->d_active, e->d_unit_active, e->d_slots
       );
     }
     LAUNCH_CHECK();
@@ -3638,7 +3728,7 @@ i64 kmp_lp_phase_a(
             kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
             static_cast<u32>(e->balance), fallback,
             e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels8,
-            e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+            e->d_weights, e->d_maxw, e->d_minw, e->d_m_list, e->d_m_count, e->d_slots
         );
       } else {
         auto *kern = e->has_adjwgt ? k_phase_m<false, uint16_t> : k_phase_m<true, uint16_t>;
@@ -3646,7 +3736,7 @@ i64 kmp_lp_phase_a(
             kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, e->k, iseed,
             static_cast<u32>(e->balance), fallback,
             e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels, e->d_labels16,
-            e->d_weights, e->d_maxw, e->d_m_list, e->d_m_count, e->d_slots
+            e->d_weights, e->d_maxw, e->d_minw, e->d_m_list, e->d_m_count, e->d_slots
         );
       }
       LAUNCH_CHECK();
@@ -3686,8 +3776,8 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           k_phase_l_sel, dim3(2048), dim3(256), 0, e->stream,
           static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k,
-          e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_l_list, e->d_l_count,
-          e->l_cap, e->d_l_gains, e->d_slots
+          e->d_xadj, e->d_vwgt, e->d_labels, e->d_weights, e->d_maxw, e->d_minw, e->d_l_list,
+          e->d_l_count, e->l_cap, e->d_l_gains, e->d_slots
       );
       LAUNCH_CHECK();
       // pathological overflow beyond l_cap: direct per-vertex workgroups
@@ -3702,7 +3792,7 @@ i64 kmp_lp_phase_a(
               kern, dim3(512), dim3(256), lds, e->stream,
               static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
               e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels8, e->d_weights,
-              e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+              e->d_maxw, e->d_minw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
           );
         } else {
           auto *kern =
@@ -3711,7 +3801,7 @@ i64 kmp_lp_phase_a(
               kern, dim3(512), dim3(256), lds, e->stream,
               static_cast<u32>(e->balance), fallback, pos_lo, chunk_base, iseed, e->k, e->d_xadj,
               e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_labels16, e->d_weights,
-              e->d_maxw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
+              e->d_maxw, e->d_minw, e->d_l_list, e->d_l_count, e->l_cap, e->d_slots
           );
         }
         LAUNCH_CHECK();
@@ -4206,6 +4296,86 @@ i64 kmp_lp_balance(
   const i64 cut = kmp_lp_refine(e, k, max_block_weights, partition, seed, iters, stats);
   e->balance = 0;
   return cut;
+}
+
+// Underload-balancer mode (the reference's UNDERLOAD_BALANCER closing the
+// default refiner chain, presets.cc:332-338; semantics restated from
+// refinement/balancer/underload_balancer.cc): fill blocks below their
+// minimum weight with best-gain admissible vertices under the deterministic
+// chunk schedule; admission is serial in rank order with BOTH per-block
+// minima and maxima re-checked per move (the batch analogue of the
+// reference's locked per-move checks). No-op when minima are satisfied.
+i64 kmp_lp_underload(
+    kmp_lp_t *e,
+    u32 k,
+    const i64 *max_block_weights,
+    const i64 *min_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    kmp_lp_stats_t *stats
+) {
+  if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
+    return -1;
+  }
+  if (e->d_minw) {
+    HIP_CHECK(hipFree(e->d_minw));
+    e->d_minw = nullptr;
+  }
+  HIP_CHECK(hipMalloc(&e->d_minw, sizeof(i64) * k));
+  HIP_CHECK(hipMemcpy(e->d_minw, min_block_weights, sizeof(i64) * k, hipMemcpyHostToDevice));
+  e->balance = 2;
+  const u32 threads = 256;
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
+  sync_spin(e);
+  unsigned long long last = e->h_moves[0];
+  for (int iter = 0; iter < iters; ++iter) {
+    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
+      const u32 pos_lo = chunk * e->C;
+      const u32 pos_hi = pos_lo + e->C > e->P ? e->P : pos_lo + e->C;
+      if (pos_lo >= pos_hi) {
+        continue;
+      }
+      const i64 cnt = kmp_lp_phase_a(e, iter, chunk, pos_lo, pos_hi, e->d_props, e->C);
+      if (cnt < 0) {
+        e->balance = 0;
+        return -1;
+      }
+      const u64 iseed = iter_seed_of(e->seed, iter);
+      hipLaunchKernelGGL(
+          k_commit_underload, dim3(1), dim3(64), 0, e->stream, e->d_props, e->d_prop_count,
+          e->d_maxw, e->d_minw, e->d_weights, e->d_labels, e->d_labels16,
+          k <= 256 ? e->d_labels8 : nullptr, e->d_admitted_flags, e->d_moves
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_clear_active, dim3(ceil_div(pos_hi - pos_lo, threads)), dim3(threads), 0, e->stream,
+          pos_lo, pos_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active,
+          e->d_arcs
+      );
+      LAUNCH_CHECK();
+      if (cnt > 0) {
+        hipLaunchKernelGGL(
+            k_activate, dim3(ceil_div(static_cast<u64>(cnt) * kWave, threads)), dim3(threads), 0,
+            e->stream, static_cast<const u32 *>(nullptr), e->d_admitted_flags, e->d_props,
+            static_cast<u32>(cnt), e->d_xadj, e->d_adjncy, e->d_active, e->d_unit_active
+        );
+        LAUNCH_CHECK();
+      }
+    }
+    HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, e->stream));
+    sync_spin(e);
+    const unsigned long long cur = e->h_moves[1];
+    const u64 sweep_moves = cur - last;
+    last = cur;
+    if (sweep_moves == 0) {
+      break;
+    }
+  }
+  e->balance = 0;
+  return kmp_lp_refine_end(e, partition, stats);
 }
 
 i64 kmp_lp_cluster(

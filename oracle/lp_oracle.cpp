@@ -288,6 +288,8 @@ struct LpParams {
   u32 desired_clusters = 0; // clusterer stop threshold (0 = never)
   bool clusterer = false;   // select semantics variant
   bool balance = false;     // overloaded vertices lose "stay" (balancer mode)
+  bool underload = false;   // underload-balancer mode (min weights set)
+  const i64 *min_weights = nullptr; // per-block minimums (underload mode)
   u32 k = 0;                // number of clusters (refiner: k; clusterer: n)
 };
 
@@ -377,9 +379,15 @@ void lp_run(
         const i32 u_weight = g.node_weight(u);
         const i64 init_weight = weights[u_cluster];
 
-        // Refiner min-weight early exit (lp_refiner.cc:160; min weights are
-        // zero in the default path, so this only fires for negative slack).
-        // Omitted: min weights are not configured in this tier's scope.
+        // Underload mode (underload_balancer.cc is_movable_from): only
+        // vertices whose block is NOT underloaded and stays above its
+        // minimum may move.
+        if (par.underload) {
+          const i64 mnw = par.min_weights[u_cluster];
+          if (init_weight < mnw || init_weight - u_weight < mnw) {
+            continue;
+          }
+        }
 
         map.reserve(std::min<u32>(deg, par.k) + 2);
         const u64 row_begin = g.xadj[u];
@@ -422,6 +430,11 @@ void lp_run(
           i64 over = 0;
           if (par.clusterer) {
             accept = (cw + u_weight <= maxw) || (c == u_cluster);
+          } else if (par.underload) {
+            // underload_balancer.cc is_movable_to: only underloaded targets
+            // with room; the current block is never a candidate
+            accept = c != u_cluster && cw < par.min_weights[c] &&
+                     cw + u_weight <= maxw;
           } else if (c == u_cluster) {
             // balance mode: a vertex in an over-cap block loses "stay"
             accept = !(par.balance &&
@@ -496,6 +509,45 @@ void lp_run(
         }
         processed.push_back(u);
         stats.arcs_scanned += deg;
+      }
+
+      // ---- phase B (underload mode): serial rank-order admission ----
+      // The underload balancer must respect BOTH per-block minima (source
+      // side) and maxima (target side); instead of a 2-dimensional
+      // fixpoint, admission is sequential in deterministic rank order with
+      // both caps re-checked against live weights (the batch analogue of
+      // underload_balancer.cc's locked per-move checks; proposal counts in
+      // this mode are small). Targets are re-checked to still be
+      // underloaded at admission time (is_movable_to semantics).
+      if (par.underload) {
+        std::vector<const Proposal *> admitted;
+        admitted.reserve(proposals.size());
+        for (const Proposal &pr : proposals) {
+          const i32 w = pr.w;
+          const i64 mn_from = par.min_weights[pr.from];
+          const i64 mn_to = par.min_weights[pr.to];
+          if (weights[pr.from] >= mn_from && weights[pr.from] - w >= mn_from &&
+              weights[pr.to] < mn_to &&
+              weights[pr.to] + w <= par.max_weights[pr.to]) {
+            weights[pr.from] -= w;
+            weights[pr.to] += w;
+            labels[pr.u] = pr.to;
+            admitted.push_back(&pr);
+          }
+        }
+        for (u32 u : processed) {
+          active[u] = 0;
+        }
+        for (const Proposal *pr : admitted) {
+          const u64 row_begin = g.xadj[pr->u];
+          const u64 row_end = g.xadj[pr->u + 1];
+          for (u64 e = row_begin; e < row_end; ++e) {
+            active[g.adjncy[e]] = 1;
+          }
+        }
+        sweep_moves += admitted.size();
+        stats.moves += admitted.size();
+        continue;
       }
 
       // ---- phase B: deterministic commit (greatest-fixpoint rollback) ----
@@ -779,6 +831,55 @@ i64 kmp_oracle_lp_balance(
       weights[t] += uw;
       partition[u] = t;
     }
+  }
+
+  std::vector<uint8_t> active(n, 1);
+  LpStats stats;
+  lp_run(g, par, partition, weights.data(), nullptr, active, stats, nullptr);
+
+  if (stats_out) {
+    stats_out[0] = stats.arcs_scanned;
+    stats_out[1] = stats.moves;
+    stats_out[2] = 0;
+  }
+  return kmp_oracle_edge_cut(n, m, xadj, adjncy, adjwgt, partition);
+}
+
+// Underload-balancer mode (the role of the reference's UNDERLOAD_BALANCER
+// closing the default refiner chain, presets.cc:332-338; semantics restated
+// from refinement/balancer/underload_balancer.cc): fill blocks below their
+// minimum weight with best-gain admissible vertices, never dropping any
+// source below its own minimum and never overshooting any maximum.
+i64 kmp_oracle_lp_underload(
+    u32 n,
+    u64 m,
+    const u32 *xadj,
+    const u32 *adjncy,
+    const i32 *vwgt,
+    const i32 *adjwgt,
+    u32 k,
+    const i64 *max_block_weights,
+    const i64 *min_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    u64 *stats_out
+) {
+  Csr g{n, m, xadj, adjncy, vwgt, adjwgt};
+
+  LpParams par;
+  par.n = n;
+  par.max_weights = max_block_weights;
+  par.min_weights = min_block_weights;
+  par.seed = seed;
+  par.iters = iters;
+  par.clusterer = false;
+  par.underload = true;
+  par.k = k;
+
+  std::vector<i64> weights(k, 0);
+  for (u32 u = 0; u < n; ++u) {
+    weights[partition[u]] += g.node_weight(u);
   }
 
   std::vector<uint8_t> active(n, 1);

@@ -77,6 +77,28 @@ def ref_cluster(ref, g, max_w, seed=1, iters=5, desired=0):
     return clus
 
 
+def oracle_underload(oracle, g, k, maxw, minw, part, seed=1, iters=5,
+                     vwgt=None, adjwgt=None):
+    """Run the CPU oracle's underload-balancer LP. Returns (cut, part, stats)."""
+    import ctypes
+    import numpy as np
+    part = np.ascontiguousarray(part, dtype=np.uint32).copy()
+    maxw = np.ascontiguousarray(maxw, dtype=np.int64)
+    minw = np.ascontiguousarray(minw, dtype=np.int64)
+    xadj = np.ascontiguousarray(g.xadj, dtype=np.uint32)
+    adjncy = np.ascontiguousarray(g.adjncy, dtype=np.uint32)
+    stats = np.zeros(3, dtype=np.uint64)
+    oracle.kmp_oracle_lp_underload.restype = ctypes.c_int64
+    cut = oracle.kmp_oracle_lp_underload(
+        ctypes.c_uint32(g.n), ctypes.c_uint64(g.m), u32p(xadj), u32p(adjncy),
+        i32p(vwgt) if vwgt is not None else None,
+        i32p(adjwgt) if adjwgt is not None else None,
+        ctypes.c_uint32(k), i64p(maxw), i64p(minw), u32p(part),
+        ctypes.c_uint64(seed), ctypes.c_int(iters), u64p(stats),
+    )
+    return cut, part, stats
+
+
 def oracle_balance(oracle, g, k, maxw, part, seed=1, iters=5, vwgt=None, adjwgt=None):
     """Run the CPU oracle's balancer-mode LP. Returns (cut, part, stats)."""
     part = np.ascontiguousarray(part, dtype=np.uint32).copy()

@@ -769,3 +769,106 @@ def test_fuzz_balance_parity(oracle, trial):
     assert np.array_equal(part, opart)
     if overload(part0) > 0:
         assert overload(part) < overload(part0)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("case", ["drained", "weighted"])
+def test_underload_mode_parity_and_fill(oracle, case):
+    """Underload-balancer mode (kmp_lp_underload): bit-identical to the
+    oracle twin, underloaded blocks are filled to their minimum weights,
+    and no block ever drops below its own minimum or exceeds its maximum
+    (the role of the reference's UNDERLOAD_BALANCER closing the default
+    refiner chain, presets.cc:332-338)."""
+    _require_gpu()
+    from helpers import oracle_underload
+
+    rng = np.random.default_rng(11)
+    g = ka.Graph.rmat(13, 8, seed=21)
+    k = 16
+    vwgt = None
+    if case == "weighted":
+        vwgt = rng.integers(1, 6, g.n).astype(np.int32)
+        g = ka.Graph.from_csr(np.asarray(g.xadj).copy(),
+                              np.asarray(g.adjncy).copy(), vwgt=vwgt)
+    vw = vwgt.astype(np.int64) if vwgt is not None else np.ones(g.n, np.int64)
+    total = int(vw.sum())
+    avg = total // k
+    mbw = np.full(k, int(avg * 1.25) + 1, np.int64)
+    # min weights a la setup_min_block_weights(min_epsilon=0.25)
+    mnw = np.full(k, int(avg * 0.75), np.int64)
+
+    # drain blocks 0..3 below their minimum: move most of their vertices out
+    part0 = ka.random_partition(g.n, k, seed=3)
+    drained = part0 < 4
+    part0[drained] = (4 + (np.arange(g.n)[drained] % (k - 4))).astype(np.uint32)
+    # keep a couple of seed vertices so the blocks are non-empty
+    for b in range(4):
+        part0[b] = b
+
+    def weights_of(part):
+        bw = np.zeros(k, np.int64)
+        np.add.at(bw, part, vw)
+        return bw
+
+    bw0 = weights_of(part0)
+    assert (bw0[:4] < mnw[:4]).all()  # really underloaded
+
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.underload(k, mbw, mnw, part0, seed=1, iters=5)
+    ocut, opart, _ = oracle_underload(oracle, g, k, mbw, mnw, part0, seed=1,
+                                      iters=5, vwgt=vwgt)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+
+    bw1 = weights_of(part)
+    # never overshoot max, never undershoot a block that was feasible
+    assert (bw1 <= mbw).all()
+    feasible0 = bw0 >= mnw
+    assert (bw1[feasible0] >= mnw[feasible0]).all()
+    # the underload deficit strictly shrinks
+    deficit0 = int(np.maximum(mnw - bw0, 0).sum())
+    deficit1 = int(np.maximum(mnw - bw1, 0).sum())
+    assert deficit1 < deficit0
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("trial", range(4))
+def test_fuzz_underload_parity(oracle, trial):
+    """Seeded fuzz for underload mode: random graphs, random drained
+    partitions, non-uniform minima -- GPU bit-identical to the oracle twin
+    and both weight invariants hold."""
+    _require_gpu()
+    from helpers import oracle_underload
+
+    rng = np.random.default_rng(4242 + trial)
+    n = int(rng.integers(200, 9_000))
+    g, vwgt, adjwgt = _random_graph(rng, n, float(rng.uniform(2, 12)),
+                                    bool(rng.integers(0, 2)))
+    k = int(rng.integers(2, 48))
+    vw = vwgt.astype(np.int64) if vwgt is not None else np.ones(g.n, np.int64)
+    total = int(vw.sum())
+    mbw = np.full(k, int(total / k * 1.4) + 4, np.int64)
+    frac = rng.uniform(0.3, 0.9, k)
+    mnw = (frac * total / k).astype(np.int64)
+    part0 = ka.random_partition(g.n, k, seed=trial * 3 + 1)
+    # drain a random subset of blocks
+    ndrain = int(rng.integers(1, max(2, k // 2)))
+    for b in rng.choice(k, ndrain, replace=False):
+        sel = part0 == b
+        part0[sel] = (int(b) + 1 + (np.arange(g.n)[sel] % max(1, k - 1))).astype(
+            np.uint32) % k
+
+    eng = ka.LpEngine(g)
+    cut, part, _ = eng.underload(k, mbw, mnw, part0, seed=7, iters=4)
+    ocut, opart, _ = oracle_underload(oracle, g, k, mbw, mnw, part0, seed=7,
+                                      iters=4, vwgt=vwgt, adjwgt=adjwgt)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+
+    bw = np.zeros(k, np.int64)
+    np.add.at(bw, part, vw)
+    assert (bw <= mbw).all()
+    bw0 = np.zeros(k, np.int64)
+    np.add.at(bw0, part0, vw)
+    feas0 = bw0 >= mnw
+    assert (bw[feas0] >= mnw[feas0]).all()
