@@ -71,6 +71,20 @@ int kmp_write_metis(const kmp_graph_t *g, const char *path);
 kmp_graph_t *kmp_read_parhip(const char *path);
 int kmp_write_parhip(const kmp_graph_t *g, const char *path);
 
+/* EdgeID-64 construction (ckaminpar.h:33-37 KAMINPAR_64BIT_EDGE_IDS
+ * analogue): CSR with 64-bit offsets for graphs of >= 2^32 directed arcs.
+ * Such graphs run on the LP engine (device offsets are always 64-bit);
+ * host-side utilities (IO, rearrangement, host edge cut) require 32-bit. */
+kmp_graph_t *kmp_graph_from_csr64(
+    uint32_t n,
+    uint64_t m,
+    const uint64_t *xadj,
+    const uint32_t *adjncy,
+    const int32_t *vwgt,
+    const int32_t *adjwgt
+);
+const uint64_t *kmp_graph_xadj64(const kmp_graph_t *g); /* null if 32-bit */
+
 uint32_t kmp_graph_n(const kmp_graph_t *g);
 uint64_t kmp_graph_m(const kmp_graph_t *g); /* number of directed arcs */
 const uint32_t *kmp_graph_xadj(const kmp_graph_t *g);

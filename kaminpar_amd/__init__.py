@@ -58,6 +58,8 @@ def _load():
 
     lib.kmp_graph_from_csr.restype = vp
     lib.kmp_graph_from_csr.argtypes = [u32, u64, p(u32), p(u32), p(i32), p(i32)]
+    lib.kmp_graph_from_csr64.restype = vp
+    lib.kmp_graph_from_csr64.argtypes = [u32, u64, p(u64), p(u32), p(i32), p(i32)]
     lib.kmp_gen_rmat.restype = vp
     lib.kmp_gen_rmat.argtypes = [ctypes.c_int, ctypes.c_int, u64]
     lib.kmp_gen_rgg2d.restype = vp
@@ -190,7 +192,6 @@ class Graph:
 
     @classmethod
     def from_csr(cls, xadj, adjncy, vwgt=None, adjwgt=None):
-        xadj = np.ascontiguousarray(xadj, dtype=np.uint32)
         adjncy = np.ascontiguousarray(adjncy, dtype=np.uint32)
         n = len(xadj) - 1
         m = len(adjncy)
@@ -202,6 +203,12 @@ class Graph:
         if adjwgt is not None:
             adjwgt = np.ascontiguousarray(adjwgt, dtype=np.int32)
             ap = adjwgt.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        if m >= (1 << 32):
+            # EdgeID-64 path (device offsets are 64-bit either way)
+            xadj = np.ascontiguousarray(xadj, dtype=np.uint64)
+            xp = xadj.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
+            return cls(_lib.kmp_graph_from_csr64(n, m, xp, _u32p(adjncy), vp, ap))
+        xadj = np.ascontiguousarray(xadj, dtype=np.uint32)
         return cls(_lib.kmp_graph_from_csr(n, m, _u32p(xadj), _u32p(adjncy), vp, ap))
 
     @classmethod

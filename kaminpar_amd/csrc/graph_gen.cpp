@@ -34,7 +34,8 @@ using kmp::u64;
 struct kmp_graph_t {
   u32 n = 0;
   u64 m = 0;
-  std::vector<u32> xadj;
+  std::vector<u32> xadj;   // empty when the graph uses 64-bit offsets
+  std::vector<u64> xadj64; // EdgeID-64 offsets (m >= 2^32); else empty
   std::vector<u32> adjncy;
   std::vector<i32> vwgt;   // empty -> unit
   std::vector<i32> adjwgt; // empty -> unit
@@ -140,6 +141,34 @@ kmp_graph_t *graph_from_pairs(u32 n, std::vector<u64> &arcs) {
 } // namespace
 
 extern "C" {
+
+/* EdgeID-64 construction path (ckaminpar.h:33-37 KAMINPAR_64BIT_EDGE_IDS
+ * analogue): graphs with >= 2^32 directed arcs carry 64-bit offsets. */
+kmp_graph_t *kmp_graph_from_csr64(
+    u32 n, u64 m, const u64 *xadj, const u32 *adjncy, const i32 *vwgt, const i32 *adjwgt
+) {
+  if (xadj == nullptr || (m > 0 && adjncy == nullptr) || xadj[n] != m) {
+    return nullptr;
+  }
+  auto *g = new kmp_graph_t();
+  g->n = n;
+  g->m = m;
+  g->xadj64.assign(xadj, xadj + n + 1);
+  g->adjncy.assign(adjncy, adjncy + m);
+  if (vwgt) {
+    g->vwgt.assign(vwgt, vwgt + n);
+    g->total_node_weight = 0;
+    for (u32 u = 0; u < n; ++u) {
+      g->total_node_weight += vwgt[u];
+    }
+  } else {
+    g->total_node_weight = n;
+  }
+  if (adjwgt) {
+    g->adjwgt.assign(adjwgt, adjwgt + m);
+  }
+  return g;
+}
 
 kmp_graph_t *kmp_graph_from_csr(
     u32 n, u64 m, const u32 *xadj, const u32 *adjncy, const i32 *vwgt, const i32 *adjwgt
@@ -405,7 +434,12 @@ kmp_graph_t *kmp_read_metis(const char *path) {
 
 u32 kmp_graph_n(const kmp_graph_t *g) { return g->n; }
 u64 kmp_graph_m(const kmp_graph_t *g) { return g->m; }
-const u32 *kmp_graph_xadj(const kmp_graph_t *g) { return g->xadj.data(); }
+const u32 *kmp_graph_xadj(const kmp_graph_t *g) {
+  return g->xadj64.empty() ? g->xadj.data() : nullptr;
+}
+const u64 *kmp_graph_xadj64(const kmp_graph_t *g) {
+  return g->xadj64.empty() ? nullptr : g->xadj64.data();
+}
 const u32 *kmp_graph_adjncy(const kmp_graph_t *g) { return g->adjncy.data(); }
 const i32 *kmp_graph_vwgt(const kmp_graph_t *g) {
   return g->vwgt.empty() ? nullptr : g->vwgt.data();

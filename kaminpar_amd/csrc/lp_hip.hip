@@ -168,7 +168,7 @@ __global__ void k_phase_s(
     u32 fallback,
     u32 max_degree,
     u32 s_clear, // v2 path: S owns the chunk's active-flag clearing
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
@@ -201,12 +201,13 @@ __global__ void k_phase_s(
   const u32 sidx = p - pos_lo;
 
   bool skip = false;
-  u32 row = 0, deg = 0;
+  u64 row = 0;
+  u32 deg = 0;
   if (u >= n) {
     skip = true;
   } else {
     row = xadj[u];
-    deg = xadj[u + 1] - row;
+    deg = static_cast<u32>(xadj[u + 1] - row);
     const bool act = active[u] != 0;
     if (!act || deg > max_degree) {
       skip = true;
@@ -324,7 +325,7 @@ __global__ void k_phase_m(
     u64 iter_seed,
     u32 mode,
     u32 fallback,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const i32 *__restrict__ vwgt,
@@ -350,8 +351,8 @@ __global__ void k_phase_m(
   const u64 rec = m_list[vid];
   const u32 p = static_cast<u32>(rec >> 32);
   const u32 u = static_cast<u32>(rec);
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
+  const u64 row = xadj[u];
+  const u32 deg = static_cast<u32>(xadj[u + 1] - row);
 
   for (u32 c = lane; c < k * R; c += kWave) {
     gains[c] = 0;
@@ -441,7 +442,7 @@ constexpr u32 kLSlice = 8192;
 __global__ void k_l_sizes(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     u32 l_cap,
     u32 *__restrict__ sizes
 ) {
@@ -449,7 +450,7 @@ __global__ void k_l_sizes(
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < count) {
     const u32 u = static_cast<u32>(l_list[i]);
-    const u32 deg = xadj[u + 1] - xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
     sizes[i] = (deg + kLSlice - 1) / kLSlice;
   }
 }
@@ -458,7 +459,7 @@ __global__ void k_l_sizes(
 template <bool kUnitWeights, typename LT>
 __global__ void k_phase_l_acc(
     u32 k,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const LT *__restrict__ labels_s,
@@ -487,8 +488,8 @@ __global__ void k_phase_l_acc(
     }
     const u32 vid = lo;
     const u32 u = static_cast<u32>(l_list[vid]);
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     const u32 e_lo = (s - l_off[vid]) * kLSlice;
     const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
 
@@ -526,7 +527,7 @@ __global__ void k_phase_l_sel(
     u32 chunk_base,
     u64 iter_seed,
     u32 k,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
@@ -631,7 +632,7 @@ __global__ void k_phase_l_direct(
     u32 chunk_base,
     u64 iter_seed,
     u32 k,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
@@ -660,8 +661,8 @@ __global__ void k_phase_l_direct(
     const u64 rec = l_list[vid];
     const u32 p = static_cast<u32>(rec >> 32);
     const u32 u = static_cast<u32>(rec);
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
 
     const u32 rep_off = (threadIdx.x % R) * k;
     for (u32 e = threadIdx.x; e < deg; e += blockDim.x) {
@@ -771,7 +772,7 @@ __global__ void k_phase_s_c(
     u64 iter_seed,
     u32 max_degree,
     i64 maxw_uniform,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ vwgt,
     const i32 *__restrict__ adjwgt,
@@ -801,12 +802,13 @@ __global__ void k_phase_s_c(
   const u32 sidx = p - pos_lo;
 
   bool skip = false;
-  u32 row = 0, deg = 0;
+  u64 row = 0;
+  u32 deg = 0;
   if (u >= n) {
     skip = true;
   } else {
     row = xadj[u];
-    deg = xadj[u + 1] - row;
+    deg = static_cast<u32>(xadj[u + 1] - row);
     if (!active[u] || deg > max_degree) {
       skip = true;
     }
@@ -895,7 +897,7 @@ __global__ void k_phase_m_c(
     u32 chunk_base,
     u64 iter_seed,
     i64 maxw_uniform,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const i32 *__restrict__ vwgt,
@@ -919,8 +921,8 @@ __global__ void k_phase_m_c(
   const u64 rec = m_list[vid];
   const u32 p = static_cast<u32>(rec >> 32);
   const u32 u = static_cast<u32>(rec);
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
+  const u64 row = xadj[u];
+  const u32 deg = static_cast<u32>(xadj[u + 1] - row);
 
   for (u32 s = lane; s < kHashSlots; s += kWave) {
     hkeys[s] = kInvalid;
@@ -1019,7 +1021,7 @@ __global__ void k_phase_m2_c(
     u32 chunk_base,
     u64 iter_seed,
     i64 maxw_uniform,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const i32 *__restrict__ vwgt,
@@ -1039,8 +1041,8 @@ __global__ void k_phase_m2_c(
     const u64 rec = m2_list[vid];
     const u32 p = static_cast<u32>(rec >> 32);
     const u32 u = static_cast<u32>(rec);
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
 
     for (u32 s = threadIdx.x; s < kM2HashSlots; s += blockDim.x) {
       hkeys[s] = kInvalid;
@@ -1167,7 +1169,7 @@ __global__ void k_phase_m2_c(
 __global__ void k_l_prep_c(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     u32 l_cap,
     u32 *__restrict__ l_off,  // slice prefix
     u64 *__restrict__ l_hoff, // hash-region prefix
@@ -1187,7 +1189,7 @@ __global__ void k_l_prep_c(
     u64 hs = 0;
     if (i < count) {
       const u32 u = static_cast<u32>(l_list[i]);
-      const u32 deg = xadj[u + 1] - xadj[u];
+      const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
       sc = (deg + kLSlice - 1) / kLSlice;
       u32 bits = 11; // >= 2048 slots
       while ((1u << bits) < 2 * deg) {
@@ -1241,7 +1243,7 @@ __global__ void k_l_prep_c(
 // pool; k_phase_l_sel_c clears each region so batches can reuse the pool).
 template <bool kUnitWeights>
 __global__ void k_phase_l_acc_c(
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
@@ -1275,8 +1277,8 @@ __global__ void k_phase_l_acc_c(
     }
     const u32 vid = lo;
     const u32 u = static_cast<u32>(l_list[vid]);
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     const u32 e_lo = (s - l_off[vid]) * kLSlice;
     const u32 e_hi = e_lo + kLSlice < deg ? e_lo + kLSlice : deg;
     const u64 roff = l_hoff[vid] - hbase;
@@ -1637,7 +1639,7 @@ __global__ void k_init_cluster_weights(
 __global__ void k_twohop_cand(
     u32 n,
     i64 maxw_uniform,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
@@ -1649,7 +1651,7 @@ __global__ void k_twohop_cand(
     return;
   }
   u64 out = ~0ull;
-  const u32 deg = xadj[u + 1] - xadj[u];
+  const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
   if (deg > 0 && labels[u] == u) {
     const i64 w = weights[u];
     const i64 vw = vwgt ? vwgt[u] : 1;
@@ -1931,7 +1933,7 @@ __global__ void k_clear_active(
     u32 n,
     u64 iter_seed,
     u32 max_degree,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     uint8_t *__restrict__ active,
     uint8_t *__restrict__ unit_active,
     unsigned long long *__restrict__ arcs
@@ -1956,7 +1958,7 @@ __global__ void k_clear_active(
   if (p < chunk_hi && unit_on) {
     const u32 u = vb * kmp::kUnit + (p % kmp::kUnit);
     if (u < n) {
-      const u32 deg = xadj[u + 1] - xadj[u];
+      const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
       if (deg <= max_degree && active[u]) {
         my_deg = deg;
         active[u] = 0;
@@ -1988,7 +1990,7 @@ __global__ void k_activate(
     const u32 *__restrict__ admitted_flags,
     const Prop *__restrict__ props,
     u32 count,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     uint8_t *__restrict__ active,
     uint8_t *__restrict__ unit_active
@@ -1999,8 +2001,8 @@ __global__ void k_activate(
     return;
   }
   const u32 u = props[order ? order[wave_id] : wave_id].u;
-  const u32 row = xadj[u];
-  const u32 deg = xadj[u + 1] - row;
+  const u64 row = xadj[u];
+  const u32 deg = static_cast<u32>(xadj[u + 1] - row);
   for (u32 e = lane; e < deg; e += kWave) {
     const u32 v = adjncy[row + e];
     active[v] = 1;
@@ -2078,7 +2080,7 @@ __global__ void k_init_weights(
 // keep lanes busy; per-wave partial sums, one atomic per wave.
 __global__ void k_edge_cut(
     u32 n,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
@@ -2095,8 +2097,8 @@ __global__ void k_edge_cut(
     if (u >= n) {
       continue;
     }
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     const u32 lu = labels[u];
     for (u32 e = slot; e < deg; e += 16) {
       if (labels[adjncy[row + e]] != lu) {
@@ -2129,7 +2131,7 @@ __global__ void k_edge_cut(
 __global__ void k_l_prep_r(
     const u64 *__restrict__ l_list,
     const u32 *__restrict__ l_count,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     u32 l_cap,
     u32 *__restrict__ l_off
 ) {
@@ -2142,7 +2144,7 @@ __global__ void k_l_prep_r(
     u32 v = 0;
     if (i < count) {
       const u32 u = static_cast<u32>(l_list[i]);
-      const u32 deg = xadj[u + 1] - xadj[u];
+      const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
       v = (deg + kLSlice - 1) / kLSlice;
     }
     u32 inc = v;
@@ -2193,7 +2195,7 @@ __global__ void k_build_lists(
     u32 mid_hi, // kMidDeg (refine) / kClusterMidDeg (cluster)
     u32 m2_hi,  // kClusterM2Deg (cluster; == mid_hi disables the M2 tier)
     u32 max_degree,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     u64 *__restrict__ m_list,
@@ -2268,7 +2270,7 @@ __global__ void k_build_lists(
       const u32 u = vb * kmp::kUnit + lane;
       bool is_m = false, is_m2 = false, is_l = false;
       if (p >= pos_lo && p < pos_hi && u < n && active[u]) {
-        const u32 deg = xadj[u + 1] - xadj[u];
+        const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
         if (deg <= max_degree) {
           is_m = deg > mid_lo && deg <= mid_hi;
           is_m2 = deg > mid_hi && deg <= m2_hi;
@@ -2918,7 +2920,7 @@ __global__ void k_activate_v2(
     const u32 *__restrict__ prefix_len,
     const u32 *__restrict__ s_u,
     const uint16_t *__restrict__ s_to,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     uint8_t *__restrict__ active,
     uint8_t *__restrict__ unit_active
@@ -2932,8 +2934,8 @@ __global__ void k_activate_v2(
       continue;
     }
     const u32 u = s_u[i];
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     for (u32 e2 = lane; e2 < deg; e2 += kWave) {
       const u32 v = adjncy[row + e2];
       active[v] = 1;
@@ -2954,8 +2956,9 @@ struct kmp_lp_t {
   u32 P = 0; // total positions (pos_count)
   bool has_vwgt = false, has_adjwgt = false;
 
-  // device graph
-  u32 *d_xadj = nullptr;
+  // device graph (xadj is u64 device-side: EdgeID-64 everywhere removes the
+  // 2^32-arc ceiling at ~zero cost -- xadj reads are per-vertex + coalesced)
+  u64 *d_xadj = nullptr;
   u32 *d_adjncy = nullptr;
   i32 *d_vwgt = nullptr;
   i32 *d_adjwgt = nullptr;
@@ -3211,10 +3214,11 @@ void engine_alloc_common(kmp_lp_t *e) {
 
 // Isolated-vertex scan used by the clusterer's isolated-node handling
 // (lp_clusterer.cc cluster_isolated_nodes semantics).
-void engine_scan_isolated(kmp_lp_t *e, const u32 *xadj, const i32 *vwgt) {
+template <typename XT>
+void engine_scan_isolated(kmp_lp_t *e, const XT *xadj, const i32 *vwgt) {
   u32 maxd = 0;
   for (u32 u = 0; u < e->n; ++u) {
-    const u32 deg = xadj[u + 1] - xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - xadj[u]);
     if (deg > maxd) {
       maxd = deg;
     }
@@ -3433,14 +3437,6 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
     return nullptr;
   }
 
-  if (kmp_graph_m(g) > 0xFFFFFFFFull) {
-    // R-MAT scale-28/ef-8 dedups to just under 2^32 directed arcs and fits;
-    // anything larger needs the 64-bit edge-offset kernel variant (planned,
-    // DESIGN.md section 6)
-    fprintf(stderr, "kaminpar_amd: graphs with >= 2^32 directed arcs need the "
-                    "64-bit EdgeID kernel variant (not built yet)\n");
-    return nullptr;
-  }
   auto *e = new kmp_lp_t();
   (void)hipSetDeviceFlags(hipDeviceScheduleSpin); // ignore if context exists
   e->n = kmp_graph_n(g);
@@ -3459,9 +3455,24 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
     e->mp_count = props.multiProcessorCount;
   }
 
-  HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u32) * (e->n + 1)));
+  HIP_CHECK(hipMalloc(&e->d_xadj, sizeof(u64) * (e->n + 1)));
   HIP_CHECK(hipMalloc(&e->d_adjncy, sizeof(u32) * e->m));
-  HIP_CHECK(hipMemcpy(e->d_xadj, kmp_graph_xadj(g), sizeof(u32) * (e->n + 1), hipMemcpyHostToDevice));
+  // device xadj is u64 (EdgeID-64); widen u32 host offsets on upload
+  const u64 *xadj64 = kmp_graph_xadj64(g);
+  if (xadj64 != nullptr) {
+    HIP_CHECK(
+        hipMemcpy(e->d_xadj, xadj64, sizeof(u64) * (e->n + 1), hipMemcpyHostToDevice)
+    );
+  } else {
+    const u32 *x32 = kmp_graph_xadj(g);
+    std::vector<u64> wide(static_cast<size_t>(e->n) + 1);
+    for (u32 i = 0; i <= e->n; ++i) {
+      wide[i] = x32[i];
+    }
+    HIP_CHECK(
+        hipMemcpy(e->d_xadj, wide.data(), sizeof(u64) * (e->n + 1), hipMemcpyHostToDevice)
+    );
+  }
   HIP_CHECK(hipMemcpy(e->d_adjncy, kmp_graph_adjncy(g), sizeof(u32) * e->m, hipMemcpyHostToDevice));
   if (e->has_vwgt) {
     HIP_CHECK(hipMalloc(&e->d_vwgt, sizeof(i32) * e->n));
@@ -3475,7 +3486,11 @@ kmp_lp_t *kmp_lp_create(const kmp_graph_t *g) {
   }
 
   engine_alloc_common(e);
-  engine_scan_isolated(e, kmp_graph_xadj(g), kmp_graph_vwgt(g));
+  if (kmp_graph_xadj64(g) != nullptr) {
+    engine_scan_isolated(e, kmp_graph_xadj64(g), kmp_graph_vwgt(g));
+  } else {
+    engine_scan_isolated(e, kmp_graph_xadj(g), kmp_graph_vwgt(g));
+  }
   return e;
 }
 
@@ -4644,7 +4659,7 @@ __global__ void k_coarse_vwgt(
 // arc; intra-cluster arcs get the ~0 sentinel (sorted to the end, dropped).
 __global__ void k_arc_keys(
     u32 n,
-    const u32 *__restrict__ xadj,
+    const u64 *__restrict__ xadj,
     const u32 *__restrict__ adjncy,
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ map,
@@ -4661,8 +4676,8 @@ __global__ void k_arc_keys(
     if (u >= n) {
       continue;
     }
-    const u32 row = xadj[u];
-    const u32 deg = xadj[u + 1] - row;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     const u32 cu = map[u];
     for (u32 e = slot; e < deg; e += 16) {
       const u32 cv = map[adjncy[row + e]];
@@ -4673,11 +4688,11 @@ __global__ void k_arc_keys(
 }
 
 __global__ void k_coarse_hist(
-    u32 c_m, const u64 *__restrict__ ukeys, u32 *__restrict__ cxadj
+    u32 c_m, const u64 *__restrict__ ukeys, unsigned long long *__restrict__ cxadj
 ) {
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < c_m) {
-    atomicAdd(&cxadj[(ukeys[i] >> 32) + 1], 1u);
+    atomicAdd(&cxadj[(ukeys[i] >> 32) + 1], 1ull);
   }
 }
 
@@ -4702,7 +4717,7 @@ struct ContractOut {
   u32 c_n = 0;
   u32 c_m = 0;
   u32 *d_map = nullptr;   // n
-  u32 *d_cxadj = nullptr; // c_n + 1 (prefix-summed)
+  u64 *d_cxadj = nullptr; // c_n + 1 (prefix-summed)
   u32 *d_cadj = nullptr;  // max(c_m, 1)
   i32 *d_cvw = nullptr;   // c_n
   i32 *d_cwgt = nullptr;  // max(c_m, 1)
@@ -4788,12 +4803,13 @@ int contract_core(kmp_lp_t *e, const u32 *clustering, ContractOut &o) {
   }
   const u32 c_m = (uniq > 0 && last_key == ~0ull) ? uniq - 1 : uniq;
 
-  HIP_CHECK(hipMalloc(&o.d_cxadj, sizeof(u32) * (c_n + 1)));
+  HIP_CHECK(hipMalloc(&o.d_cxadj, sizeof(u64) * (c_n + 1)));
   HIP_CHECK(hipMalloc(&o.d_cadj, sizeof(u32) * (c_m > 0 ? c_m : 1)));
-  HIP_CHECK(hipMemsetAsync(o.d_cxadj, 0, sizeof(u32) * (c_n + 1), s));
+  HIP_CHECK(hipMemsetAsync(o.d_cxadj, 0, sizeof(u64) * (c_n + 1), s));
   if (c_m > 0) {
     hipLaunchKernelGGL(
-        k_coarse_hist, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys, o.d_cxadj
+        k_coarse_hist, dim3(ceil_div(c_m, threads)), dim3(threads), 0, s, c_m, d_ukeys,
+        reinterpret_cast<unsigned long long *>(o.d_cxadj)
     );
     LAUNCH_CHECK();
     hipLaunchKernelGGL(
@@ -4804,11 +4820,11 @@ int contract_core(kmp_lp_t *e, const u32 *clustering, ContractOut &o) {
   void *tmp4 = nullptr;
   size_t tmp4_bytes = 0;
   HIP_CHECK(rocprim::inclusive_scan(
-      nullptr, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u32>()
+      nullptr, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u64>()
   ));
   HIP_CHECK(hipMalloc(&tmp4, tmp4_bytes));
   HIP_CHECK(rocprim::inclusive_scan(
-      tmp4, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u32>(), s
+      tmp4, tmp4_bytes, o.d_cxadj, o.d_cxadj, c_n + 1, rocprim::plus<u64>(), s
   ));
 
   // right-size the coarse node/edge weights
@@ -4855,10 +4871,11 @@ i64 kmp_contract(
   hipStream_t s = e->stream;
   const u32 c_n = o.c_n, c_m = o.c_m;
 
+  std::vector<u64> h_cxadj64(c_n + 1);
   std::vector<u32> h_cxadj(c_n + 1), h_cadj(c_m);
   std::vector<i32> h_cvw(c_n), h_cwgt(c_m);
   HIP_CHECK(hipMemcpyAsync(mapping_out, o.d_map, sizeof(u32) * e->n, hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), o.d_cxadj, sizeof(u32) * (c_n + 1),
+  HIP_CHECK(hipMemcpyAsync(h_cxadj64.data(), o.d_cxadj, sizeof(u64) * (c_n + 1),
                            hipMemcpyDeviceToHost, s));
   if (c_m > 0) {
     HIP_CHECK(
@@ -4870,6 +4887,9 @@ i64 kmp_contract(
   }
   HIP_CHECK(hipMemcpyAsync(h_cvw.data(), o.d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipStreamSynchronize(s));
+  for (u32 i = 0; i <= c_n; ++i) {
+    h_cxadj[i] = static_cast<u32>(h_cxadj64[i]); // c_m < 2^32 (entry guard)
+  }
 
   *coarse_out = kmp_graph_from_csr(
       c_n, c_m, h_cxadj.data(), h_cadj.data(), h_cvw.data(), c_m ? h_cwgt.data() : nullptr
@@ -4890,10 +4910,10 @@ i64 kmp_contract_engine(
 
   // host copies needed regardless: the mapping (projection happens on the
   // host) and xadj/vwgt for the isolated-vertex scan of the new engine
-  std::vector<u32> h_cxadj(c_n + 1);
+  std::vector<u64> h_cxadj(c_n + 1);
   std::vector<i32> h_cvw(c_n);
   HIP_CHECK(hipMemcpyAsync(mapping_out, o.d_map, sizeof(u32) * e->n, hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), o.d_cxadj, sizeof(u32) * (c_n + 1),
+  HIP_CHECK(hipMemcpyAsync(h_cxadj.data(), o.d_cxadj, sizeof(u64) * (c_n + 1),
                            hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipMemcpyAsync(h_cvw.data(), o.d_cvw, sizeof(i32) * c_n, hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipStreamSynchronize(s));
@@ -4927,9 +4947,19 @@ i64 kmp_contract_engine(
 }
 
 kmp_graph_t *kmp_lp_download_graph(const kmp_lp_t *e) {
+  if (e->m > 0xFFFFFFFFull) {
+    fprintf(stderr, "kaminpar_amd: kmp_lp_download_graph supports m < 2^32\n");
+    return nullptr;
+  }
+  std::vector<u64> h_xadj64(e->n + 1);
   std::vector<u32> h_xadj(e->n + 1), h_adj(e->m);
   std::vector<i32> h_vw, h_wg;
-  HIP_CHECK(hipMemcpy(h_xadj.data(), e->d_xadj, sizeof(u32) * (e->n + 1), hipMemcpyDeviceToHost));
+  HIP_CHECK(
+      hipMemcpy(h_xadj64.data(), e->d_xadj, sizeof(u64) * (e->n + 1), hipMemcpyDeviceToHost)
+  );
+  for (u32 i = 0; i <= e->n; ++i) {
+    h_xadj[i] = static_cast<u32>(h_xadj64[i]);
+  }
   HIP_CHECK(hipMemcpy(h_adj.data(), e->d_adjncy, sizeof(u32) * e->m, hipMemcpyDeviceToHost));
   if (e->has_vwgt) {
     h_vw.resize(e->n);
