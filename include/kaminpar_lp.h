@@ -151,6 +151,11 @@ int64_t kmp_lp_refine(
  * min_block_weights[b] with best-gain admissible vertices, never dropping a
  * source below its own minimum and never overshooting any maximum. A no-op
  * when all minima are already satisfied (the reference's refine() gate). */
+/* Clusterer::set_communities (coarsening/clusterer.h:35,
+ * lp_clusterer.cc:61-66,193-194): when set (len n), kmp_lp_cluster never
+ * merges vertices across community boundaries. NULL clears. */
+int kmp_lp_set_communities(kmp_lp_t *e, const uint32_t *communities);
+
 int64_t kmp_lp_underload(
     kmp_lp_t *e,
     uint32_t k,
@@ -390,6 +395,14 @@ void kaminpar_amd_copy_graph(
 void kaminpar_amd_set_k(kaminpar_amd_t *shm, uint32_t k);
 void kaminpar_amd_set_uniform_max_block_weights(kaminpar_amd_t *shm, double epsilon);
 /* Returns the final edge cut, or -1 on error (missing GPU, no graph). */
+void kaminpar_amd_set_absolute_max_block_weights(
+    kaminpar_amd_t *shm, const int64_t *weights, uint32_t count); /* kaminpar.h:961 */
+void kaminpar_amd_set_uniform_min_block_weights(
+    kaminpar_amd_t *shm, double min_epsilon); /* kaminpar.h:965; context.cc:72-80 */
+void kaminpar_amd_set_absolute_min_block_weights(
+    kaminpar_amd_t *shm, const int64_t *weights, uint32_t count); /* kaminpar.h:966 */
+void kaminpar_amd_clear_min_block_weights(kaminpar_amd_t *shm); /* kaminpar.h:968 */
+
 int64_t kaminpar_amd_compute_partition(kaminpar_amd_t *shm, uint32_t *partition);
 
 #ifdef __cplusplus

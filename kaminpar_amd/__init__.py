@@ -124,6 +124,8 @@ def _load():
     lib.kmp_lp_refine.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_balance.restype = i64
     lib.kmp_lp_balance.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
+    lib.kmp_lp_set_communities.restype = ctypes.c_int
+    lib.kmp_lp_set_communities.argtypes = [vp, p(u32)]
     lib.kmp_lp_underload.restype = i64
     lib.kmp_lp_underload.argtypes = [vp, u32, p(i64), p(i64), p(u32), u64,
                                      ctypes.c_int, vp]
@@ -436,6 +438,17 @@ class LpEngine:
         if cut < 0:
             raise RuntimeError("kmp_lp_underload failed")
         return cut, part, stats
+
+    def set_communities(self, communities):
+        """Restrict clustering merges to stay within communities
+        (coarsening/clusterer.h:35, lp_clusterer.cc:193-194). None clears."""
+        if communities is None:
+            _lib.kmp_lp_set_communities(self._h, None)
+            self._comm_keepalive = None
+            return
+        comm = np.ascontiguousarray(communities, dtype=np.uint32)
+        _lib.kmp_lp_set_communities(self._h, _u32p(comm))
+        self._comm_keepalive = comm
 
     def cluster(self, max_cluster_weight, clustering=None, desired=0, seed=1, iters=5):
         """Deterministic LP clustering; returns (n_clusters, clustering, Stats)."""

@@ -19,6 +19,7 @@
 // (stable), so the commit's stable 32-bit radix sort by target cluster
 // yields the deterministic (to, rank) admission order.
 
+#include <cmath>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -778,6 +779,7 @@ __global__ void k_phase_s_c(
     const i32 *__restrict__ adjwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
+    const u32 *__restrict__ comm, // null = unrestricted (clusterer.h:35)
     const uint8_t *__restrict__ active,
     const uint8_t *__restrict__ unit_active,
     u32 *__restrict__ favored,
@@ -847,7 +849,8 @@ __global__ void k_phase_s_c(
 
   BestState best{0, 0, 0, false};
   BestState fav{0, 0, 0, false};
-  if (owner && c != kInvalid) {
+  if (owner && c != kInvalid &&
+      (comm == nullptr || comm[c] == comm[cur])) { // lp_clusterer.cc:193-194
     const u64 h = tie_hash(iter_seed, u, c);
     fav = BestState{gain, h, c, true};
     if (accept_cluster(c, cur, u_w, weights[c], maxw_uniform)) {
@@ -903,6 +906,7 @@ __global__ void k_phase_m_c(
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
+    const u32 *__restrict__ comm,
     const u64 *__restrict__ m_list,
     const u32 *__restrict__ m_count,
     u32 *__restrict__ favored,
@@ -965,7 +969,7 @@ __global__ void k_phase_m_c(
       continue;
     }
     const i32 g = hvals[s];
-    if (g <= 0) {
+    if (g <= 0 || (comm != nullptr && comm[c] != comm[cur])) {
       continue;
     }
     const u64 h = tie_hash(iter_seed, u, c);
@@ -1027,6 +1031,7 @@ __global__ void k_phase_m2_c(
     const i32 *__restrict__ vwgt,
     const u32 *__restrict__ labels,
     const i64 *__restrict__ weights,
+    const u32 *__restrict__ comm,
     const u64 *__restrict__ m2_list,
     const u32 *__restrict__ m2_count,
     u32 *__restrict__ favored,
@@ -1084,7 +1089,7 @@ __global__ void k_phase_m2_c(
         continue;
       }
       const i32 g = hvals[s];
-      if (g <= 0) {
+      if (g <= 0 || (comm != nullptr && comm[c] != comm[cur])) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -1336,6 +1341,7 @@ __global__ void k_phase_l_sel_c(
     i32 *__restrict__ pool_vals,
     const u32 *__restrict__ l_clist,
     const u32 *__restrict__ l_ccnt,
+    const u32 *__restrict__ comm,
     u32 *__restrict__ favored,
     Prop *__restrict__ slots
 ) {
@@ -1362,7 +1368,7 @@ __global__ void k_phase_l_sel_c(
       const i32 g = hv[s];
       hk[s] = kInvalid; // clear for the next chunk
       hv[s] = 0;
-      if (g <= 0) {
+      if (g <= 0 || (comm != nullptr && comm[c] != comm[cur])) {
         continue;
       }
       const u64 h = tie_hash(iter_seed, u, c);
@@ -2974,6 +2980,7 @@ struct kmp_lp_t {
   i64 *d_weights = nullptr;
   i64 *d_maxw = nullptr;
   i64 *d_minw = nullptr; // per-block minimums (underload mode; else null)
+  u32 *d_comm = nullptr; // per-vertex communities (clusterer; else null)
   uint8_t *d_active = nullptr;
   uint8_t *d_unit_active = nullptr; // one byte per 64-vertex unit
 
@@ -3507,7 +3514,7 @@ void kmp_lp_free(kmp_lp_t *e) {
   }
   engine_free_k_buffers(e);
   for (void *p : {(void *)e->d_xadj, (void *)e->d_adjncy, (void *)e->d_vwgt, (void *)e->d_adjwgt,
-                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_minw, (void *)e->d_active, (void *)e->d_unit_active,
+                  (void *)e->d_labels, (void *)e->d_labels0, (void *)e->d_labels16, (void *)e->d_labels8, (void *)e->d_weights, (void *)e->d_maxw, (void *)e->d_minw, (void *)e->d_comm, (void *)e->d_active, (void *)e->d_unit_active,
                   (void *)e->d_slots, (void *)e->d_props,
                   (void *)e->d_m_list, (void *)e->d_m_count, (void *)e->d_m2_list,
                   (void *)e->d_l_list, (void *)e->d_l_off, (void *)e->d_l_sizes, (void *)e->d_lscan_temp, (void *)e->d_l_gains, (void *)e->d_prop_count, (void *)e->d_arcs,
@@ -3841,7 +3848,7 @@ i64 kmp_lp_phase_a(
         k_phase_s_c, dim3(ceil_div(static_cast<u64>(ceil_div(span, 4)) * kWave, threads)),
         dim3(threads), 0, e->stream, pos_lo, pos_hi, chunk_base, e->n, iseed, max_degree,
         e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_vwgt, e->d_adjwgt, e->d_labels, e->d_weights,
-        e->d_active, e->d_unit_active, e->d_favored, e->d_slots
+        e->d_comm, e->d_active, e->d_unit_active, e->d_favored, e->d_slots
     );
     LAUNCH_CHECK();
     {
@@ -3850,14 +3857,14 @@ i64 kmp_lp_phase_a(
       hipLaunchKernelGGL(
           kern, dim3(4096), dim3(threads), lds, e->stream, pos_lo, chunk_base, iseed,
           e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels,
-          e->d_weights, e->d_m_list, e->d_m_count, e->d_favored, e->d_slots
+          e->d_weights, e->d_comm, e->d_m_list, e->d_m_count, e->d_favored, e->d_slots
       );
       LAUNCH_CHECK();
       auto *kern2 = e->has_adjwgt ? k_phase_m2_c<false> : k_phase_m2_c<true>;
       hipLaunchKernelGGL(
           kern2, dim3(1024), dim3(threads), 0, e->stream, pos_lo, chunk_base, iseed,
           e->maxw_uniform, e->d_xadj, e->d_adjncy, e->d_adjwgt, e->d_vwgt, e->d_labels,
-          e->d_weights, e->d_m2_list, e->d_m2_count, e->d_favored, e->d_slots
+          e->d_weights, e->d_comm, e->d_m2_list, e->d_m2_count, e->d_favored, e->d_slots
       );
       LAUNCH_CHECK();
     }
@@ -3893,7 +3900,7 @@ i64 kmp_lp_phase_a(
             k_phase_l_sel_c, dim3(2048), dim3(256), 0, e->stream, pos_lo, chunk_base, iseed,
             e->n, e->maxw_uniform, e->d_vwgt, e->d_labels, e->d_weights, e->d_l_list, lo, hi,
             e->d_l_hoff, e->d_l_hbits, e->d_pool_keys, e->d_pool_vals, e->d_l_clist,
-            e->d_l_ccnt, e->d_favored, e->d_slots
+            e->d_l_ccnt, e->d_comm, e->d_favored, e->d_slots
         );
         LAUNCH_CHECK();
       };
@@ -4311,6 +4318,24 @@ i64 kmp_lp_balance(
   const i64 cut = kmp_lp_refine(e, k, max_block_weights, partition, seed, iters, stats);
   e->balance = 0;
   return cut;
+}
+
+// Clusterer::set_communities (coarsening/clusterer.h:35,
+// lp_clusterer.cc:61-66,193-194): when set, clustering never merges across
+// community boundaries (cluster ids are vertex ids, so the per-vertex array
+// indexes both sides of the check). Pass null to clear.
+int kmp_lp_set_communities(kmp_lp_t *e, const u32 *communities) {
+  if (e->d_comm) {
+    HIP_CHECK(hipFree(e->d_comm));
+    e->d_comm = nullptr;
+  }
+  if (communities != nullptr) {
+    HIP_CHECK(hipMalloc(&e->d_comm, sizeof(u32) * e->n));
+    HIP_CHECK(
+        hipMemcpy(e->d_comm, communities, sizeof(u32) * e->n, hipMemcpyHostToDevice)
+    );
+  }
+  return 0;
 }
 
 // Underload-balancer mode (the reference's UNDERLOAD_BALANCER closing the
@@ -5124,6 +5149,8 @@ struct kaminpar_amd_t {
   u32 k = 2;
   double eps = 0.03;
   u64 seed = 1;
+  std::vector<i64> abs_maxw; // per-block max weights (kaminpar.h:961)
+  std::vector<i64> minw;     // per-block min weights (kaminpar.h:965-968)
 };
 
 kaminpar_amd_t *kaminpar_amd_create(int /*num_threads*/) {
@@ -5157,6 +5184,36 @@ void kaminpar_amd_set_k(kaminpar_amd_t *shm, u32 k) { shm->k = k; }
 
 void kaminpar_amd_set_uniform_max_block_weights(kaminpar_amd_t *shm, double epsilon) {
   shm->eps = epsilon;
+  shm->abs_maxw.clear();
+}
+
+// kaminpar.h:961 set_absolute_max_block_weights: explicit per-block caps.
+void kaminpar_amd_set_absolute_max_block_weights(
+    kaminpar_amd_t *shm, const i64 *weights, u32 count
+) {
+  shm->abs_maxw.assign(weights, weights + count);
+}
+
+// kaminpar.h:965 set_uniform_min_block_weights: minimum block weights as a
+// fraction of the perfectly balanced weight (context.cc:72-80); triggers the
+// underload balancer after partitioning (the reference's refiner chain runs
+// it only when minima are configured, presets.cc:332-338).
+void kaminpar_amd_set_uniform_min_block_weights(kaminpar_amd_t *shm, double min_epsilon) {
+  // marker encoding {-1, min_epsilon in ppb}; materialized at compute time
+  // (the actual minima need k and the total node weight)
+  shm->minw.assign(2, -1);
+  shm->minw[1] = static_cast<i64>(min_epsilon * 1e9);
+}
+
+// kaminpar.h:966-967 absolute variant.
+void kaminpar_amd_set_absolute_min_block_weights(
+    kaminpar_amd_t *shm, const i64 *weights, u32 count
+) {
+  shm->minw.assign(weights, weights + count);
+}
+
+void kaminpar_amd_clear_min_block_weights(kaminpar_amd_t *shm) {
+  shm->minw.clear();
 }
 
 i64 kaminpar_amd_compute_partition(kaminpar_amd_t *shm, u32 *partition) {
@@ -5164,9 +5221,64 @@ i64 kaminpar_amd_compute_partition(kaminpar_amd_t *shm, u32 *partition) {
     fprintf(stderr, "kaminpar_amd: no graph set (call kaminpar_amd_copy_graph)\n");
     return -1;
   }
+  const u32 k = shm->k;
+  const i64 W = kmp_graph_total_node_weight(shm->g);
+  double eps = shm->eps;
+  std::vector<i64> caps;
+  if (!shm->abs_maxw.empty()) {
+    // pipeline under a uniform proxy derived from the average cap; exact
+    // per-block enforcement happens below via balance + refine under the
+    // true caps (both engine paths take per-block arrays)
+    caps = shm->abs_maxw;
+    caps.resize(k, caps.empty() ? 0 : caps.back());
+    i64 cap_sum = 0;
+    for (i64 c : caps) {
+      cap_sum += c;
+    }
+    const double avg_cap = static_cast<double>(cap_sum) / k;
+    eps = avg_cap * k / static_cast<double>(W) - 1.0;
+    if (eps < 0.001) {
+      eps = 0.001;
+    }
+  }
   // progressive-k (deep) pipeline: best measured cuts
-  return kmp_partition_deep(shm->g, shm->k, shm->eps, shm->seed, 5, 0, 0, 0, 0,
-                            partition);
+  i64 cut = kmp_partition_deep(shm->g, k, eps, shm->seed, 5, 0, 0, 0, 0, partition);
+  if (cut < 0) {
+    return cut;
+  }
+  const bool want_min = !shm->minw.empty();
+  if (!caps.empty() || want_min) {
+    kmp_lp_t *e = kmp_lp_create(shm->g);
+    if (!e) {
+      return -1;
+    }
+    if (caps.empty()) {
+      const i64 uni = kmp_max_block_weight(shm->g, k, shm->eps);
+      caps.assign(k, uni);
+    }
+    if (!caps.empty() && !shm->abs_maxw.empty()) {
+      // exact per-block caps: repair + refine under them
+      cut = kmp_lp_balance(e, k, caps.data(), partition, shm->seed, 5, nullptr);
+      if (cut >= 0) {
+        cut = kmp_lp_refine(e, k, caps.data(), partition, shm->seed, 5, nullptr);
+      }
+    }
+    if (cut >= 0 && want_min) {
+      std::vector<i64> minw = shm->minw;
+      if (minw.size() == 2 && minw[0] == -1) {
+        // uniform min epsilon (ppb-encoded): ceil((1-eps_min) * W/k)
+        const double me = static_cast<double>(minw[1]) / 1e9;
+        const i64 v = static_cast<i64>(
+            std::ceil((1.0 - me) * static_cast<double>(W) / k));
+        minw.assign(k, v);
+      } else {
+        minw.resize(k, 0);
+      }
+      cut = kmp_lp_underload(e, k, caps.data(), minw.data(), partition, shm->seed, 5, nullptr);
+    }
+    kmp_lp_free(e);
+  }
+  return cut;
 }
 
 } // extern "C"

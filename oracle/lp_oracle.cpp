@@ -290,6 +290,8 @@ struct LpParams {
   bool balance = false;     // overloaded vertices lose "stay" (balancer mode)
   bool underload = false;   // underload-balancer mode (min weights set)
   const i64 *min_weights = nullptr; // per-block minimums (underload mode)
+  const u32 *communities = nullptr; // clusterer: merges stay within community
+                                    // (clusterer.h:35, lp_clusterer.cc:193)
   u32 k = 0;                // number of clusters (refiner: k; clusterer: n)
 };
 
@@ -415,6 +417,13 @@ void lp_run(
           const u32 c = map.keys[slot];
           const i64 r = map.vals[slot];
           const u64 h = tie_hash(iter_seed, u, c);
+
+          // community filter (lp_clusterer.cc:193-194): cluster ids are
+          // vertex ids, so communities[] indexes both
+          if (par.communities != nullptr &&
+              par.communities[c] != par.communities[u_cluster]) {
+            continue;
+          }
 
           if (store_favored) {
             if (r > fav_gain || (r == fav_gain && (h > fav_h || (h == fav_h && c < fav)))) {
@@ -897,7 +906,7 @@ i64 kmp_oracle_lp_underload(
 // Deterministic LP clustering (coarsening instantiation;
 // lp_clusterer.cc:89-109 driver semantics + isolated/two-hop passes).
 // clustering: out, len n. Returns number of non-empty clusters.
-i64 kmp_oracle_lp_cluster(
+i64 kmp_oracle_lp_cluster_comm(
     u32 n,
     u64 m,
     const u32 *xadj,
@@ -906,6 +915,7 @@ i64 kmp_oracle_lp_cluster(
     const i32 *adjwgt,
     i64 max_cluster_weight,
     u32 desired_clusters,
+    const u32 *communities, // null = unrestricted (clusterer.h:35)
     u32 *clustering,
     u64 seed,
     int iters,
@@ -921,6 +931,7 @@ i64 kmp_oracle_lp_cluster(
   par.clusterer = true;
   par.k = n;
   par.desired_clusters = desired_clusters;
+  par.communities = communities;
 
   std::vector<i64> weights(n);
   std::vector<u32> favored(n);
@@ -1032,6 +1043,26 @@ extern "C" {
 // Returns coarse node count; fills mapping[n], c_xadj[c_n+1], c_adjncy,
 // c_vwgt, c_adjwgt (buffers sized for the fine graph are always enough).
 // c_m_out receives the coarse arc count.
+i64 kmp_oracle_lp_cluster(
+    u32 n,
+    u64 m,
+    const u32 *xadj,
+    const u32 *adjncy,
+    const i32 *vwgt,
+    const i32 *adjwgt,
+    i64 max_cluster_weight,
+    u32 desired_clusters,
+    u32 *clustering,
+    u64 seed,
+    int iters,
+    u64 *stats_out
+) {
+  return kmp_oracle_lp_cluster_comm(
+      n, m, xadj, adjncy, vwgt, adjwgt, max_cluster_weight, desired_clusters, nullptr,
+      clustering, seed, iters, stats_out
+  );
+}
+
 i64 kmp_oracle_contract(
     u32 n,
     u64 m,

@@ -99,6 +99,27 @@ def oracle_underload(oracle, g, k, maxw, minw, part, seed=1, iters=5,
     return cut, part, stats
 
 
+def oracle_cluster_comm(oracle, g, max_w, communities, seed=1, iters=5,
+                        vwgt=None, adjwgt=None, desired=0):
+    """Oracle clustering with Clusterer::set_communities semantics."""
+    import ctypes
+    import numpy as np
+    clus = np.zeros(g.n, dtype=np.uint32)
+    comm = np.ascontiguousarray(communities, dtype=np.uint32)
+    xadj = np.ascontiguousarray(g.xadj, dtype=np.uint32)
+    adjncy = np.ascontiguousarray(g.adjncy, dtype=np.uint32)
+    stats = np.zeros(3, dtype=np.uint64)
+    oracle.kmp_oracle_lp_cluster_comm.restype = ctypes.c_int64
+    nc = oracle.kmp_oracle_lp_cluster_comm(
+        ctypes.c_uint32(g.n), ctypes.c_uint64(g.m), u32p(xadj), u32p(adjncy),
+        i32p(vwgt) if vwgt is not None else None,
+        i32p(adjwgt) if adjwgt is not None else None,
+        ctypes.c_int64(int(max_w)), ctypes.c_uint32(desired), u32p(comm),
+        u32p(clus), ctypes.c_uint64(seed), ctypes.c_int(iters), u64p(stats),
+    )
+    return nc, clus, stats
+
+
 def oracle_balance(oracle, g, k, maxw, part, seed=1, iters=5, vwgt=None, adjwgt=None):
     """Run the CPU oracle's balancer-mode LP. Returns (cut, part, stats)."""
     part = np.ascontiguousarray(part, dtype=np.uint32).copy()

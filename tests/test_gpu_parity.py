@@ -941,3 +941,60 @@ def _concat_aranges(lengths):
     np.cumsum(lengths[:-1], out=starts[1:])
     out -= np.repeat(starts, lengths)
     return out
+
+
+@pytest.mark.gpu
+def test_cluster_communities_parity(oracle):
+    """Clusterer::set_communities (clusterer.h:35, lp_clusterer.cc:193-194):
+    clustering never merges across community boundaries, bit-identical to
+    the oracle twin."""
+    _require_gpu()
+    from helpers import oracle_cluster_comm
+
+    g = ka.Graph.rmat(13, 8, seed=7)
+    rng = np.random.default_rng(2)
+    comm = rng.integers(0, 5, g.n).astype(np.uint32)
+    max_w = 64
+
+    eng = ka.LpEngine(g)
+    eng.set_communities(comm)
+    nc, clus, stats = eng.cluster(max_w, seed=4, iters=5)
+    eng.set_communities(None)
+
+    onc, oclus, ostats = oracle_cluster_comm(oracle, g, max_w, comm, seed=4,
+                                             iters=5)
+    assert nc == onc
+    assert np.array_equal(clus, oclus)
+    assert stats.moves == ostats[1]
+    # hard constraint: no cluster spans two communities (cluster ids are
+    # vertex ids, so communities[] indexes both)
+    assert (comm[clus] == comm).all()
+
+    # clearing communities restores the unrestricted result
+    nc2, clus2, _ = eng.cluster(max_w, seed=4, iters=5)
+    from helpers import oracle_cluster
+    onc2, oclus2, _ = oracle_cluster(oracle, g, max_w, seed=4, iters=5)
+    assert nc2 == onc2 and np.array_equal(clus2, oclus2)
+
+
+@pytest.mark.gpu
+def test_c_api_caller(tmp_path):
+    """Compile and run the committed plain-C caller (tools/c_api_check.c)
+    against libkaminpar_lp.so: the drop-in boundary incl. per-block max
+    weights, min weights (underload chain) and set_communities works from
+    C11 with no C++/Python involved."""
+    _require_gpu()
+    import subprocess
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = os.path.join(repo, "tools", "c_api_check.c")
+    lib_dir = os.path.join(repo, "kaminpar_amd")
+    exe = str(tmp_path / "c_api_check")
+    subprocess.run(
+        ["gcc", "-std=c11", "-O1", src, "-o", exe,
+         f"-L{lib_dir}", "-lkaminpar_lp", f"-Wl,-rpath,{lib_dir}"],
+        check=True, capture_output=True)
+    r = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "ALL OK" in r.stdout
