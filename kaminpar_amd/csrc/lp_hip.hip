@@ -2981,6 +2981,7 @@ struct kmp_lp_t {
   i64 *d_maxw = nullptr;
   i64 *d_minw = nullptr; // per-block minimums (underload mode; else null)
   u32 *d_comm = nullptr; // per-vertex communities (clusterer; else null)
+  std::vector<u32> comm_host; // host copy (isolated-chain community breaks)
   uint8_t *d_active = nullptr;
   uint8_t *d_unit_active = nullptr; // one byte per 64-vertex unit
 
@@ -4329,11 +4330,13 @@ int kmp_lp_set_communities(kmp_lp_t *e, const u32 *communities) {
     HIP_CHECK(hipFree(e->d_comm));
     e->d_comm = nullptr;
   }
+  e->comm_host.clear();
   if (communities != nullptr) {
     HIP_CHECK(hipMalloc(&e->d_comm, sizeof(u32) * e->n));
     HIP_CHECK(
         hipMemcpy(e->d_comm, communities, sizeof(u32) * e->n, hipMemcpyHostToDevice)
     );
+    e->comm_host.assign(communities, communities + e->n);
   }
   return 0;
 }
@@ -4548,6 +4551,14 @@ i64 kmp_lp_cluster(
     for (size_t i = 0; i < e->isolated.size(); ++i) {
       const u32 cu = e->isolated[i];
       const i64 w = e->iso_weights[i];
+      // communities are a hard constraint: chains break at boundaries
+      // (keep in sync with the oracle twin)
+      if (pending != 0xFFFFFFFFu && !e->comm_host.empty() &&
+          e->comm_host[pending] != e->comm_host[cu]) {
+        pending = cu;
+        pending_w = w;
+        continue;
+      }
       if (pending != 0xFFFFFFFFu && pending_w + w <= max_cluster_weight) {
         pairs.push_back((static_cast<u64>(pending) << 32) | cu);
         pending = 0xFFFFFFFFu;

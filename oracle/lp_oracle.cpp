@@ -963,6 +963,14 @@ i64 kmp_oracle_lp_cluster_comm(
         continue;
       }
       const u32 cu = clustering[u];
+      // communities are a hard constraint: isolated chains break at
+      // community boundaries (strengthens the reference engine, whose
+      // isolated handler predates set_communities)
+      if (pending != 0xFFFFFFFFu && communities != nullptr &&
+          communities[pending] != communities[cu]) {
+        pending = cu;
+        continue;
+      }
       if (pending != 0xFFFFFFFFu && weights[pending] + weights[cu] <= max_cluster_weight) {
         weights[pending] += weights[cu];
         weights[cu] = 0;
