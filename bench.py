@@ -137,7 +137,7 @@ def main():
         import math
         mcw = int(0.03 * g.total_node_weight / min(max(g.n // 2000, 2), k))
 
-    from kaminpar_amd.multi import TorchComm, refine_dist
+    from kaminpar_amd.multi import TorchComm, refine_dist_sharded
 
     level_sizes = []
     if world > 1:
@@ -167,7 +167,7 @@ def main():
             eng.reset()
             eng.run_sweeps(args.iters)
             return None, eng.get_stats()
-        cut, part, stats = refine_dist(eng, k, mbw, part0, args.seed, args.iters, comm)
+        cut, part, stats = refine_dist_sharded(eng, k, mbw, part0, args.seed, args.iters, comm)
         return cut, stats
 
     def barrier_sync():

@@ -234,6 +234,32 @@ int64_t kmp_lp_refine_end(kmp_lp_t *e, uint32_t *partition, kmp_lp_stats_t *stat
  * The timed region excludes host transfers and the final edge-cut kernel,
  * matching the reference's LP-only timing
  * (shm_label_propagation_benchmark.cc:121-123). */
+/* Sharded deterministic commit (multi-GPU): rank owns targets [c_lo,c_hi)
+ * of the ALL-GATHERED proposal list (global rank order). Protocol per chunk
+ * (mirrors kaminpar-dist/refinement/lp/lp_refiner.cc:296-333):
+ *   shard_begin  -> local sort + full-admission departure contributions
+ *                   into d_dep_out (k+1 i64; caller allreduce-sums = global)
+ *   loop: shard_round(dep_global) -> d_delta_out (k+1; [k]=changed);
+ *         caller allreduces, subtracts delta from dep_global, repeats
+ *         until the summed changed flag is 0
+ *   shard_finish_meta -> own targets' rank cutoffs + admitted arrivals
+ *                   (caller allreduce-sums both = allgather)
+ *   shard_apply  -> every rank applies the identical admitted set + weight
+ *                   updates + active-set maintenance; returns moves.
+ * All d_* pointers are DEVICE pointers. Bit-identical to kmp_lp_commit. */
+int kmp_lp_shard_begin(kmp_lp_t *e, uint32_t c_lo, uint32_t c_hi,
+                       const void *d_props, uint32_t count, long long *d_dep_out);
+int kmp_lp_shard_round(kmp_lp_t *e, uint32_t c_lo, uint32_t c_hi,
+                       const long long *d_dep_global, long long *d_delta_out);
+int kmp_lp_shard_finish_meta(kmp_lp_t *e, uint32_t c_lo, uint32_t c_hi,
+                             unsigned long long *d_cutoff_out,
+                             long long *d_arr_out);
+int64_t kmp_lp_shard_apply(kmp_lp_t *e, int iter, uint32_t chunk,
+                           const void *d_props, uint32_t count,
+                           const unsigned long long *d_cutoff_all,
+                           const long long *d_arr_all,
+                           const long long *d_dep_global);
+
 int kmp_lp_reset(kmp_lp_t *e);                     /* restore initial state (D2D) */
 int64_t kmp_lp_run_sweeps(kmp_lp_t *e, int iters); /* the timed LP region */
 int kmp_lp_get_stats(kmp_lp_t *e, kmp_lp_stats_t *stats); /* no cut/download */

@@ -139,6 +139,14 @@ def _load():
     lib.kmp_lp_phase_a.argtypes = [vp, ctypes.c_int, u32, u32, u32, vp, u32]
     lib.kmp_lp_commit.restype = i64
     lib.kmp_lp_commit.argtypes = [vp, ctypes.c_int, u32, vp, u32]
+    lib.kmp_lp_shard_begin.restype = ctypes.c_int
+    lib.kmp_lp_shard_begin.argtypes = [vp, u32, u32, vp, u32, vp]
+    lib.kmp_lp_shard_round.restype = ctypes.c_int
+    lib.kmp_lp_shard_round.argtypes = [vp, u32, u32, vp, vp]
+    lib.kmp_lp_shard_finish_meta.restype = ctypes.c_int
+    lib.kmp_lp_shard_finish_meta.argtypes = [vp, u32, u32, vp, vp]
+    lib.kmp_lp_shard_apply.restype = i64
+    lib.kmp_lp_shard_apply.argtypes = [vp, ctypes.c_int, u32, vp, u32, vp, vp, vp]
     lib.kmp_lp_refine_end.restype = i64
     lib.kmp_lp_refine_end.argtypes = [vp, p(u32), vp]
     lib.kmp_lp_reset.restype = ctypes.c_int
@@ -483,6 +491,34 @@ class LpEngine:
         mv = _lib.kmp_lp_commit(self._h, it, chunk, d_props_ptr, count)
         if mv < 0:
             raise RuntimeError("kmp_lp_commit failed")
+        return int(mv)
+
+    def shard_begin(self, c_lo, c_hi, d_props_ptr, count, d_dep_out_ptr):
+        """Sharded commit step 1: sort own targets from the all-gathered
+        list + full-admission departure contributions (device ptrs)."""
+        rc = _lib.kmp_lp_shard_begin(self._h, c_lo, c_hi, d_props_ptr, count,
+                                     d_dep_out_ptr)
+        if rc != 0:
+            raise RuntimeError("kmp_lp_shard_begin failed")
+
+    def shard_round(self, c_lo, c_hi, d_dep_global_ptr, d_delta_out_ptr):
+        rc = _lib.kmp_lp_shard_round(self._h, c_lo, c_hi, d_dep_global_ptr,
+                                     d_delta_out_ptr)
+        if rc != 0:
+            raise RuntimeError("kmp_lp_shard_round failed")
+
+    def shard_finish_meta(self, c_lo, c_hi, d_cutoff_ptr, d_arr_ptr):
+        rc = _lib.kmp_lp_shard_finish_meta(self._h, c_lo, c_hi, d_cutoff_ptr,
+                                           d_arr_ptr)
+        if rc != 0:
+            raise RuntimeError("kmp_lp_shard_finish_meta failed")
+
+    def shard_apply(self, it, chunk, d_props_ptr, count, d_cutoff_ptr,
+                    d_arr_ptr, d_dep_ptr):
+        mv = _lib.kmp_lp_shard_apply(self._h, it, chunk, d_props_ptr, count,
+                                     d_cutoff_ptr, d_arr_ptr, d_dep_ptr)
+        if mv < 0:
+            raise RuntimeError("kmp_lp_shard_apply failed")
         return int(mv)
 
     def reset(self):

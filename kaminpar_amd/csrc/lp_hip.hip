@@ -2480,6 +2480,336 @@ __global__ void k_scatter_v2(
   }
 }
 
+// ==================== sharded commit (multi-GPU) ====================
+// Rank-sharded deterministic admission over the ALL-GATHERED proposal list
+// (dense Prop records in global rank order): each rank sorts and fixpoints
+// only its own target range [c_lo, c_hi); the k-sized de-admission delta is
+// allreduced per round (exactly the block-weight reconciliation of
+// kaminpar-dist/refinement/lp/lp_refiner.cc:296-333) and the final
+// per-target rank-cutoffs are exchanged so every rank applies the identical
+// admitted set. Bit-identical to the single-GPU commit: the per-round
+// global departure sums match the monolithic fixpoint's rounds.
+
+// Histogram over the dense proposal list, filtered to the rank's targets.
+__global__ void k_hist_props(
+    u32 count,
+    u32 k,
+    u32 rows,
+    u32 tpw,
+    u32 c_lo,
+    u32 c_hi,
+    const Prop *__restrict__ props,
+    u32 *__restrict__ histT
+) {
+  extern __shared__ u32 cnt_lds[];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * k;
+  for (u32 c = lane; c < k; c += kWave) {
+    cnt[c] = 0;
+  }
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 T = (count + 63) >> 6;
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    const u32 i = (t << 6) + lane;
+    if (i < count) {
+      const u32 to = props[i].to;
+      if (to >= c_lo && to < c_hi) {
+        atomicAdd(&cnt[to], 1u);
+      }
+    }
+  }
+  for (u32 c = lane; c < k; c += kWave) {
+    histT[c * rows + row] = cnt[c];
+  }
+}
+
+// Stable scatter of the rank's targets + full-admission dep/arr totals:
+// dep contributions (by SOURCE block, over the rank's own-target proposals)
+// go to the caller's allreduce buffer; arrival totals stay local.
+__global__ void k_scatter_props(
+    u32 count,
+    u32 k,
+    u32 rows,
+    u32 tpw,
+    u32 c_lo,
+    u32 c_hi,
+    const Prop *__restrict__ props,
+    const u32 *__restrict__ offT,
+    const u32 *__restrict__ labels,
+    u32 *__restrict__ s_u,
+    u32 *__restrict__ s_w,
+    u32 *__restrict__ s_r,
+    uint16_t *__restrict__ s_to,
+    uint16_t *__restrict__ s_b,
+    long long *__restrict__ dep_out,          // k+1 (caller buffer, zeroed)
+    unsigned long long *__restrict__ arr_loc  // e->d_dep + k
+) {
+  extern __shared__ u32 cnt_lds[];
+  const u32 waves_per_wg = blockDim.x >> 6;
+  unsigned long long *h_out =
+      reinterpret_cast<unsigned long long *>(cnt_lds + waves_per_wg * k);
+  unsigned long long *h_in = h_out + k;
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * k;
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  for (u32 c = lane; c < k; c += kWave) {
+    cnt[c] = offT[c * rows + row];
+  }
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    h_out[c] = 0;
+    h_in[c] = 0;
+  }
+  __syncthreads();
+  const u32 T = (count + 63) >> 6;
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    const u32 i = (t << 6) + lane;
+    Prop pr{0u, kInvalid, 0u, 0u};
+    if (i < count) {
+      pr = props[i];
+    }
+    const u32 c = pr.to;
+    const bool valid = c != kInvalid && c >= c_lo && c < c_hi;
+    u32 rank = 0, total = 0, leader = 0;
+    bool seen = false;
+    for (u32 j = 0; j < kWave; ++j) {
+      const u32 cj = __shfl(c, j, kWave);
+      const int vj = __shfl(static_cast<int>(valid), j, kWave);
+      if (valid && vj && cj == c) {
+        ++total;
+        if (j < lane) {
+          ++rank;
+        }
+        if (!seen) {
+          leader = j;
+          seen = true;
+        }
+      }
+    }
+    u32 base0 = 0;
+    if (valid && rank == 0) {
+      base0 = cnt[c];
+      cnt[c] = base0 + total;
+    }
+    const u32 base = __shfl(base0, leader, kWave);
+    if (valid) {
+      const u32 dst = base + rank;
+      const uint16_t srcb = static_cast<uint16_t>(labels[pr.u]);
+      s_u[dst] = pr.u;
+      s_w[dst] = pr.w;
+      s_r[dst] = pr.rank;
+      s_to[dst] = static_cast<uint16_t>(c);
+      s_b[dst] = srcb;
+      atomicAdd(&h_out[srcb], static_cast<unsigned long long>(pr.w));
+      atomicAdd(&h_in[c], static_cast<unsigned long long>(pr.w));
+    }
+  }
+  __syncthreads();
+  for (u32 c = threadIdx.x; c < k; c += blockDim.x) {
+    if (h_out[c]) {
+      atomicAdd(reinterpret_cast<unsigned long long *>(&dep_out[c]), h_out[c]);
+    }
+    if (h_in[c]) {
+      atomicAdd(&arr_loc[c], h_in[c]);
+    }
+  }
+}
+
+// Segmented prefix weights for the rank's targets (weighted graphs only;
+// wave per segment, tile scans).
+__global__ void k_shard_pw(
+    u32 k,
+    u32 c_lo,
+    u32 c_hi,
+    const u32 *__restrict__ seg_off,
+    const u32 *__restrict__ prefix_len,
+    const u32 *__restrict__ s_w,
+    i64 *__restrict__ pw
+) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 waves = (gridDim.x * blockDim.x) >> 6;
+  for (u32 c = c_lo + wv; c < c_hi; c += waves) {
+    const u32 b = seg_off[c];
+    const u32 len = prefix_len[c];
+    i64 carry = 0;
+    for (u32 base = 0; base < len; base += kWave) {
+      const u32 i = base + lane;
+      i64 v = (i < len) ? static_cast<i64>(s_w[b + i]) : 0;
+      i64 inc = v;
+      for (int off = 1; off < 64; off <<= 1) {
+        const i64 o = __shfl_up(inc, off, kWave);
+        if (lane >= static_cast<u32>(off)) {
+          inc += o;
+        }
+      }
+      if (i < len) {
+        pw[b + i] = carry + inc;
+      }
+      carry += __shfl(inc, kWave - 1, kWave);
+    }
+  }
+}
+
+// One synchronous fixpoint round over the rank's targets: capacities frozen
+// from the GLOBAL departure vector, de-admissions accumulated into the
+// caller's delta buffer ([k] = changed flag). plen/arr persist in
+// d_prefix_len / d_dep[k..2k).
+__global__ void k_shard_round(
+    u32 k,
+    u32 c_lo,
+    u32 c_hi,
+    u32 has_vwgt,
+    const u32 *__restrict__ seg_off,
+    u32 *__restrict__ prefix_len,
+    unsigned long long *__restrict__ arr_loc, // d_dep + k (live)
+    const long long *__restrict__ dep_global, // k (allreduced)
+    const u32 *__restrict__ s_w,
+    const uint16_t *__restrict__ s_b,
+    const i64 *__restrict__ pw, // segmented prefix weights (weighted only)
+    const i64 *__restrict__ weights,
+    const i64 *__restrict__ maxw,
+    long long *__restrict__ delta_out // k+1 (caller buffer, zeroed here)
+) {
+  __shared__ unsigned long long ddelta[256];
+  __shared__ u32 rlo_s[256], rhi_s[256];
+  __shared__ int chg;
+  const u32 tid = threadIdx.x;
+  if (tid < k) {
+    ddelta[tid] = 0;
+  }
+  if (tid == 0) {
+    chg = 0;
+  }
+  __syncthreads();
+  if (tid < k) {
+    const u32 c = tid;
+    u32 nlo = 0, nhi = 0;
+    if (c >= c_lo && c < c_hi && prefix_len[c] > 0) {
+      const i64 cap = maxw[c] - weights[c] + static_cast<i64>(dep_global[c]);
+      const long long a = static_cast<long long>(arr_loc[c]);
+      if (a > cap) {
+        const u32 old = prefix_len[c];
+        const u32 b = seg_off[c];
+        u32 nl;
+        if (!has_vwgt) {
+          nl = cap <= 0 ? 0u : (cap >= static_cast<i64>(old) ? old : static_cast<u32>(cap));
+          arr_loc[c] = nl;
+        } else {
+          u32 lo2 = 0, hi2 = old;
+          while (lo2 < hi2) {
+            const u32 mid = (lo2 + hi2 + 1) >> 1;
+            if (pw[b + mid - 1] <= cap) {
+              lo2 = mid;
+            } else {
+              hi2 = mid - 1;
+            }
+          }
+          nl = lo2;
+          arr_loc[c] = nl ? static_cast<unsigned long long>(pw[b + nl - 1]) : 0ull;
+        }
+        if (nl < old) {
+          prefix_len[c] = nl;
+          nlo = b + nl;
+          nhi = b + old;
+          chg = 1;
+        }
+      }
+    }
+    rlo_s[c] = nlo;
+    rhi_s[c] = nhi;
+  }
+  __syncthreads();
+  for (u32 c = 0; c < k; ++c) {
+    const u32 lo2 = rlo_s[c], hi2 = rhi_s[c];
+    for (u32 i = lo2 + tid; i < hi2; i += blockDim.x) {
+      atomicAdd(&ddelta[s_b[i]], has_vwgt ? static_cast<unsigned long long>(s_w[i]) : 1ull);
+    }
+  }
+  __syncthreads();
+  if (tid < k) {
+    delta_out[tid] = static_cast<long long>(ddelta[tid]);
+  }
+  if (tid == 0) {
+    delta_out[k] = chg;
+  }
+}
+
+// Export the rank's final per-target cutoff ranks + admitted arrival
+// weights into caller buffers (zero elsewhere; an allreduce-sum doubles as
+// the allgather).
+__global__ void k_shard_finish_meta(
+    u32 k,
+    u32 c_lo,
+    u32 c_hi,
+    const u32 *__restrict__ seg_off,
+    const u32 *__restrict__ prefix_len,
+    const unsigned long long *__restrict__ arr_loc,
+    const u32 *__restrict__ s_r,
+    unsigned long long *__restrict__ cutoff_out, // k i64 (zeroed by caller)
+    long long *__restrict__ arr_out              // k i64 (zeroed by caller)
+) {
+  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c < c_lo || c >= c_hi) {
+    return;
+  }
+  const u32 len = prefix_len[c];
+  const u32 seg_len = seg_off[c + 1] - seg_off[c];
+  // cutoff = rank of the first NON-admitted proposal (all ranks below it
+  // are admitted: the sorted order is rank order within the target); i64
+  // cells so a plain allreduce-sum doubles as the allgather
+  cutoff_out[c] = (len >= seg_len) ? ~0ull : s_r[seg_off[c] + len];
+  arr_out[c] = static_cast<long long>(arr_loc[c]);
+}
+
+// Apply the globally-agreed admission on the full proposal list + update
+// all block weights; every rank executes this identically.
+__global__ void k_shard_apply(
+    u32 count,
+    u32 k,
+    const Prop *__restrict__ props,
+    const unsigned long long *__restrict__ cutoff_all, // k (allreduced)
+    const long long *__restrict__ arr_all,   // k (allreduced)
+    const long long *__restrict__ dep_global, // k (final)
+    i64 *__restrict__ weights,
+    u32 *__restrict__ labels,
+    uint16_t *__restrict__ labels16,
+    uint8_t *__restrict__ labels8, // null unless k <= 256
+    u32 *__restrict__ admitted_flags,
+    unsigned long long *__restrict__ moves
+) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < k) {
+    const i64 delta = static_cast<i64>(arr_all[i]) - static_cast<i64>(dep_global[i]);
+    if (delta) {
+      weights[i] += delta;
+    }
+  }
+  unsigned long long local = 0;
+  if (i < count) {
+    const Prop pr = props[i];
+    const bool adm = pr.rank < cutoff_all[pr.to];
+    admitted_flags[i] = adm ? 1u : 0u;
+    if (adm) {
+      labels[pr.u] = pr.to;
+      labels16[pr.u] = static_cast<uint16_t>(pr.to);
+      if (labels8 != nullptr) {
+        labels8[pr.u] = static_cast<uint8_t>(pr.to);
+      }
+      local = 1;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    local += __shfl_down(static_cast<unsigned long long>(local), off, kWave);
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0 && local) {
+    atomicAdd(moves, local);
+  }
+}
+
 // Resident-grid barrier: flat arrival counter + generation release, relaxed
 // agent-scope atomics bracketed by __threadfence (measured the fastest
 // variant on gfx950: ~3 us at 64 blocks vs ~20 us for acq_rel two-level --
@@ -3026,6 +3356,7 @@ struct kmp_lp_t {
   // with 4 fixed-shape launches and ZERO host syncs per chunk.
   u32 *d_s_u = nullptr;       // C (sorted-by-target source vertices)
   u32 *d_s_w = nullptr;       // C (their node weights)
+  u32 *d_s_r = nullptr;       // C (global ranks; sharded-commit cutoffs)
   uint16_t *d_s_to = nullptr; // C (target block per sorted index)
   uint16_t *d_s_b = nullptr;  // C (source block, frozen at scatter)
   u32 *d_histT = nullptr;     // k x rows column-major histogram
@@ -3179,6 +3510,7 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_s_u, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_s_w, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_s_to, sizeof(uint16_t) * C));
+  HIP_CHECK(hipMalloc(&e->d_s_r, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_s_b, sizeof(uint16_t) * C));
   HIP_CHECK(hipMalloc(&e->d_histT, sizeof(u32) * 65536));
   HIP_CHECK(hipMalloc(&e->d_offT, sizeof(u32) * 65536));
@@ -3523,7 +3855,7 @@ void kmp_lp_free(kmp_lp_t *e) {
                   (void *)e->d_sort_vals[0], (void *)e->d_sort_vals[1], (void *)e->d_sort_temp,
                   (void *)e->d_select_temp, (void *)e->d_sw, (void *)e->d_pw,
                   (void *)e->d_scan_temp, (void *)e->d_changed, (void *)e->d_admitted_flags,
-                  (void *)e->d_cut, (void *)e->d_s_u, (void *)e->d_s_w, (void *)e->d_s_to,
+                  (void *)e->d_cut, (void *)e->d_s_u, (void *)e->d_s_w, (void *)e->d_s_r, (void *)e->d_s_to,
                   (void *)e->d_s_b, (void *)e->d_histT, (void *)e->d_offT, (void *)e->d_seg_off,
                   (void *)e->d_bar, (void *)e->d_blocksums}) {
     if (p) {
@@ -4138,6 +4470,128 @@ i64 kmp_lp_commit(kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 cou
   HIP_CHECK(hipEventElapsedTime(&cms, cev0, cev1));
   e->commit_ms += cms;
   e->last_emptied = big_k ? (e->h_moves[3] - e->h_moves[2]) : 0;
+  return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
+}
+
+// ---- sharded commit ABI (multi-GPU; see kernel block comment) ----
+// d_* arguments are DEVICE pointers (e.g. torch tensors' data_ptr()).
+
+int kmp_lp_shard_begin(
+    kmp_lp_t *e, u32 c_lo, u32 c_hi, const void *d_props, u32 count,
+    long long *d_dep_out /* k+1 i64, pre-zeroed by caller */
+) {
+  const u32 threads = 256;
+  const u32 rows = 1024;
+  const u32 T = (count + 63) >> 6;
+  const u32 tpw = (T + rows - 1) / rows;
+  const Prop *props = static_cast<const Prop *>(d_props);
+  const size_t lds_h = static_cast<size_t>(threads / kWave) * e->k * sizeof(u32);
+  // reset local arr slot
+  HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * 2 * e->k, e->stream));
+  if (count > 0) {
+    hipLaunchKernelGGL(
+        k_hist_props, dim3(rows / 4), dim3(threads), lds_h, e->stream, count, e->k, rows, tpw,
+        c_lo, c_hi, props, e->d_histT
+    );
+    LAUNCH_CHECK();
+  } else {
+    HIP_CHECK(hipMemsetAsync(e->d_histT, 0, sizeof(u32) * e->k * rows, e->stream));
+  }
+  hipLaunchKernelGGL(
+      k_scan_small, dim3(1), dim3(threads), 0, e->stream, e->k, rows, e->d_histT, e->d_offT,
+      e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed
+  );
+  LAUNCH_CHECK();
+  if (count > 0) {
+    const size_t lds_sc = lds_h + static_cast<size_t>(2 * e->k) * sizeof(unsigned long long);
+    hipLaunchKernelGGL(
+        k_scatter_props, dim3(rows / 4), dim3(threads), lds_sc, e->stream, count, e->k, rows,
+        tpw, c_lo, c_hi, props, e->d_offT, e->d_labels, e->d_s_u, e->d_s_w, e->d_s_r,
+        e->d_s_to, e->d_s_b, d_dep_out,
+        reinterpret_cast<unsigned long long *>(e->d_dep) + e->k
+    );
+    LAUNCH_CHECK();
+    if (e->has_vwgt) {
+      hipLaunchKernelGGL(
+          k_shard_pw, dim3(64), dim3(threads), 0, e->stream, e->k, c_lo, c_hi, e->d_seg_off,
+          e->d_prefix_len, e->d_s_w, e->d_pw
+      );
+      LAUNCH_CHECK();
+    }
+  }
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  return 0;
+}
+
+int kmp_lp_shard_round(
+    kmp_lp_t *e, u32 c_lo, u32 c_hi, const long long *d_dep_global,
+    long long *d_delta_out /* k+1 */
+) {
+  hipLaunchKernelGGL(
+      k_shard_round, dim3(1), dim3(256), 0, e->stream, e->k, c_lo, c_hi,
+      static_cast<u32>(e->has_vwgt ? 1 : 0), e->d_seg_off, e->d_prefix_len,
+      reinterpret_cast<unsigned long long *>(e->d_dep) + e->k, d_dep_global, e->d_s_w,
+      e->d_s_b, e->d_pw, e->d_weights, e->d_maxw, d_delta_out
+  );
+  LAUNCH_CHECK();
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  return 0;
+}
+
+int kmp_lp_shard_finish_meta(
+    kmp_lp_t *e, u32 c_lo, u32 c_hi,
+    unsigned long long *d_cutoff_out /* k i64, pre-zeroed */,
+    long long *d_arr_out /* k i64, pre-zeroed */
+) {
+  hipLaunchKernelGGL(
+      k_shard_finish_meta, dim3(ceil_div(e->k, 256u)), dim3(256), 0, e->stream, e->k, c_lo,
+      c_hi, e->d_seg_off, e->d_prefix_len,
+      reinterpret_cast<unsigned long long *>(e->d_dep) + e->k, e->d_s_r, d_cutoff_out,
+      d_arr_out
+  );
+  LAUNCH_CHECK();
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  return 0;
+}
+
+i64 kmp_lp_shard_apply(
+    kmp_lp_t *e, int iter, u32 chunk, const void *d_props, u32 count,
+    const unsigned long long *d_cutoff_all, const long long *d_arr_all,
+    const long long *d_dep_global
+) {
+  const u32 threads = 256;
+  const u64 iseed = iter_seed_of(e->seed, iter);
+  const u32 chunk_lo = chunk * e->C;
+  const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
+  const Prop *props = static_cast<const Prop *>(d_props);
+  {
+    const u32 span = count > e->k ? count : e->k;
+    hipLaunchKernelGGL(
+        k_shard_apply, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, count, e->k,
+        props, d_cutoff_all, d_arr_all, d_dep_global, e->d_weights, e->d_labels, e->d_labels16,
+        e->k <= 256 ? e->d_labels8 : nullptr, e->d_admitted_flags, e->d_moves
+    );
+    LAUNCH_CHECK();
+  }
+  hipLaunchKernelGGL(
+      k_clear_active, dim3(ceil_div(chunk_hi - chunk_lo, threads)), dim3(threads), 0, e->stream,
+      chunk_lo, chunk_hi, e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active,
+      e->d_arcs
+  );
+  LAUNCH_CHECK();
+  if (count > 0) {
+    hipLaunchKernelGGL(
+        k_activate, dim3(ceil_div(static_cast<u64>(count) * kWave, threads)), dim3(threads), 0,
+        e->stream, static_cast<const u32 *>(nullptr), e->d_admitted_flags, props, count,
+        e->d_xadj, e->d_adjncy, e->d_active, e->d_unit_active
+    );
+    LAUNCH_CHECK();
+  }
+  HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
+                           hipMemcpyDeviceToHost, e->stream));
+  sync_spin(e);
   return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
 }
 

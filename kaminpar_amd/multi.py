@@ -77,6 +77,70 @@ def refine_dist(eng, k, max_block_weights, partition, seed, iters, comm):
     return eng.refine_end()
 
 
+def target_range(k, rank, world):
+    """Rank's owned block range for the sharded commit."""
+    return (rank * k) // world, ((rank + 1) * k) // world
+
+
+def refine_dist_sharded(eng, k, max_block_weights, partition, seed, iters,
+                        comm):
+    """Sharded deterministic LP refinement with a SHARDED commit: each rank
+    sorts and fixpoints only its own target-block range; per fixpoint round
+    a k-sized de-admission delta is allreduced (exactly the block-weight
+    reconciliation of kaminpar-dist/refinement/lp/lp_refiner.cc:296-333) and
+    the final per-target rank-cutoffs are exchanged, so every rank applies
+    the identical admitted set. Bit-identical to refine_dist / single-GPU.
+
+    comm additionally provides zeros(shape, dtype) -> tensor and
+    allreduce_(tensor) (in-place sum)."""
+    n = len(partition)
+    num_chunks = eng.num_chunks()
+    C = chunk_size(n, num_chunks)
+    cap = C // comm.world + 2
+    c_lo, c_hi = target_range(k, comm.rank, comm.world)
+
+    eng.refine_begin(k, max_block_weights, partition, seed)
+    buf, buf_ptr = comm.alloc_prop_buffer(cap)
+    dep = comm.zeros(k + 1, "i64")
+    delta = comm.zeros(k + 1, "i64")
+    cutoff = comm.zeros(k, "i64")  # i64 cells (sum-as-allgather)
+    arr = comm.zeros(k, "i64")
+
+    for it in range(iters):
+        sweep_moves = 0
+        for chunk in range(num_chunks):
+            lo, hi = chunk_ranges(n, num_chunks, chunk)
+            if lo >= hi:
+                continue
+            slo, shi = rank_slice(lo, hi, comm.rank, comm.world)
+            cnt = eng.phase_a(it, chunk, slo, shi, buf_ptr, cap)
+            cat_ptr, total = comm.all_gather_props(buf, cnt)
+
+            dep.zero_()
+            eng.shard_begin(c_lo, c_hi, cat_ptr, total, dep.data_ptr())
+            comm.allreduce_(dep)  # global full-admission departures
+            while True:
+                eng.shard_round(c_lo, c_hi, dep.data_ptr(), delta.data_ptr())
+                comm.allreduce_(delta)
+                if int(delta[k].item()) == 0:
+                    break
+                dep[:k] -= delta[:k]
+            cutoff.zero_()
+            arr.zero_()
+            eng.shard_finish_meta(c_lo, c_hi, cutoff.data_ptr(),
+                                  arr.data_ptr())
+            comm.allreduce_(cutoff)  # zeros elsewhere: sum == allgather
+            comm.allreduce_(arr)
+            comm.sync()
+            sweep_moves += eng.shard_apply(it, chunk, cat_ptr, total,
+                                           cutoff.data_ptr(), arr.data_ptr(),
+                                           dep.data_ptr())
+        if sweep_moves == 0:
+            break
+
+    return eng.refine_end()
+
+
 class TorchComm:
     """torch.distributed-backed collective layer (nccl=RCCL on GPU)."""
 
@@ -107,6 +171,18 @@ class TorchComm:
         self._cap = cap
         self._buf = t
         return t, t.data_ptr()
+
+    def zeros(self, size, dtype):
+        assert dtype == "i64"
+        return self.torch.zeros(size, dtype=self.torch.int64,
+                                device=self.device)
+
+    def allreduce_(self, t):
+        self.dist.all_reduce(t)
+
+    def sync(self):
+        if str(self.device).startswith("cuda"):
+            self.torch.cuda.current_stream().synchronize()
 
     def all_gather_props(self, buf, count):
         torch, dist = self.torch, self.dist
@@ -145,3 +221,15 @@ class LocalComm:
 
     def all_gather_props(self, buf, count):
         return buf.data_ptr(), count
+
+    def zeros(self, size, dtype):
+        assert dtype == "i64"
+        return self.torch.zeros(size, dtype=self.torch.int64,
+                                device=self.device)
+
+    def allreduce_(self, t):
+        pass
+
+    def sync(self):
+        if str(self.device).startswith("cuda"):
+            self.torch.cuda.current_stream().synchronize()

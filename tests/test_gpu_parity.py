@@ -998,3 +998,110 @@ def test_c_api_caller(tmp_path):
     r = subprocess.run([exe], capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "ALL OK" in r.stdout
+
+
+@pytest.mark.gpu
+def test_sharded_commit_two_rank_sim(oracle):
+    """Simulate the SHARDED commit with 2 ranks on one GPU: two engines
+    (one per simulated rank, each owning half the target blocks), collect
+    proposals per position slice, run the k-sized delta-allreduce fixpoint
+    rounds (simulated by summing the two ranks' tensors), exchange the
+    rank-cutoffs, and apply on both -- both engines must end bit-identical
+    to the monolithic single-GPU run (the bit-parity contract of
+    kaminpar_amd.multi.refine_dist_sharded)."""
+    import torch
+
+    from kaminpar_amd.multi import rank_slice, target_range
+
+    g = ka.Graph.rmat(12, 8, seed=9)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=4)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+
+    ref_eng = ka.LpEngine(g)
+    cut_ref, part_ref, _ = ref_eng.refine(k, mbw, part0, seed=6, iters=5)
+    del ref_eng
+
+    world = 2
+    engs = [ka.LpEngine(g) for _ in range(world)]
+    for e in engs:
+        e.refine_begin(k, mbw, part0, seed=6)
+    num_chunks = engs[0].num_chunks()
+    n = g.n
+    C = (((n + 63) // 64 + num_chunks - 1) // num_chunks) * 64
+    P = ((n + 63) // 64) * 64
+    cap = C
+    dev = "cuda:0"
+    bufs = [torch.zeros((cap, 4), dtype=torch.int32, device=dev)
+            for _ in range(world)]
+    deps = [torch.zeros(k + 1, dtype=torch.int64, device=dev)
+            for _ in range(world)]
+    deltas = [torch.zeros(k + 1, dtype=torch.int64, device=dev)
+              for _ in range(world)]
+    cutoffs = [torch.zeros(k, dtype=torch.int64, device=dev)
+               for _ in range(world)]
+    arrs = [torch.zeros(k, dtype=torch.int64, device=dev)
+            for _ in range(world)]
+    ranges = [target_range(k, r, world) for r in range(world)]
+
+    for it in range(5):
+        sweep_moves = 0
+        for chunk in range(num_chunks):
+            lo = chunk * C
+            hi = min(lo + C, P)
+            if lo >= hi:
+                continue
+            counts = []
+            for r in range(world):
+                slo, shi = rank_slice(lo, hi, r, world)
+                counts.append(engs[r].phase_a(it, chunk, slo, shi,
+                                              bufs[r].data_ptr(), cap))
+            cat = torch.cat([bufs[r][: counts[r]]
+                             for r in range(world)]).contiguous()
+            torch.cuda.synchronize()
+            total = sum(counts)
+
+            for r in range(world):
+                deps[r].zero_()
+                engs[r].shard_begin(ranges[r][0], ranges[r][1],
+                                    cat.data_ptr(), total,
+                                    deps[r].data_ptr())
+            dep_sum = deps[0] + deps[1]  # simulated allreduce
+            for r in range(world):
+                deps[r].copy_(dep_sum)
+            while True:
+                for r in range(world):
+                    engs[r].shard_round(ranges[r][0], ranges[r][1],
+                                        deps[r].data_ptr(),
+                                        deltas[r].data_ptr())
+                dsum = deltas[0] + deltas[1]
+                if int(dsum[k].item()) == 0:
+                    break
+                for r in range(world):
+                    deps[r][:k] -= dsum[:k]
+            for r in range(world):
+                cutoffs[r].zero_()
+                arrs[r].zero_()
+                engs[r].shard_finish_meta(ranges[r][0], ranges[r][1],
+                                          cutoffs[r].data_ptr(),
+                                          arrs[r].data_ptr())
+            csum = cutoffs[0] + cutoffs[1]
+            asum = arrs[0] + arrs[1]
+            for r in range(world):
+                cutoffs[r].copy_(csum)
+                arrs[r].copy_(asum)
+            torch.cuda.synchronize()
+            mv = [engs[r].shard_apply(it, chunk, cat.data_ptr(), total,
+                                      cutoffs[r].data_ptr(),
+                                      arrs[r].data_ptr(),
+                                      deps[r].data_ptr())
+                  for r in range(world)]
+            assert mv[0] == mv[1]
+            sweep_moves += mv[0]
+        if sweep_moves == 0:
+            break
+
+    results = [e.refine_end() for e in engs]
+    for cut_r, part_r, _ in results:
+        assert cut_r == cut_ref
+        assert np.array_equal(part_r, part_ref)

@@ -164,3 +164,187 @@ def test_torchcomm_fused_count_gloo():
 
     port = 29612
     mp.spawn(_torchcomm_worker, args=(2, port), nprocs=2, join=True)
+
+
+# ---------------- sharded-commit protocol (refine_dist_sharded) ----------------
+
+class _Registry:
+    """ptr -> numpy view registry so the fake engine can read the torch
+    buffers refine_dist_sharded hands over as raw device pointers."""
+
+    def __init__(self):
+        self.views = {}
+
+    def add(self, t):
+        self.views[t.data_ptr()] = t.numpy()
+        return t
+
+    def get(self, ptr):
+        return self.views[ptr]
+
+
+class _ShardFakeEngine:
+    """Numpy restatement of the sharded-commit engine semantics: every 3rd
+    position proposes to block u % k at unit weight; the shard_* calls run
+    the identical sort/fixpoint/apply math the GPU kernels implement, so a
+    world-2 gloo run must leave all ranks with identical labels/weights."""
+
+    def __init__(self, n, k, registry):
+        self.n = n
+        self.k = k
+        self.reg = registry
+        self.state = {}
+
+    def num_chunks(self):
+        return 64
+
+    def refine_begin(self, k, mbw, part, seed):
+        self.labels = np.asarray(part, dtype=np.uint32).copy()
+        self.mbw = np.asarray(mbw, dtype=np.int64)
+        self.weights = np.bincount(self.labels, minlength=k).astype(np.int64)
+        self.moves = 0
+
+    def phase_a(self, it, chunk, lo, hi, ptr, cap):
+        out = self.reg.get(ptr)
+        cnt = 0
+        for p in range(lo, hi):
+            if p % 3 == 0 and p < self.n:
+                u = p  # fake permutation = identity
+                to = (u * 7 + it) % self.k
+                if to != self.labels[u]:
+                    out[cnt] = (u, to, p - (chunk * (hi - lo + 0)) & 0xFFFFFFFF, 1)
+                    out[cnt][2] = p  # rank = global position (monotone)
+                    cnt += 1
+        return cnt
+
+    def shard_begin(self, c_lo, c_hi, cat_ptr, total, dep_ptr):
+        props = self.reg.get(cat_ptr)[:total] if total else np.zeros((0, 4), np.int32)
+        dep = self.reg.get(dep_ptr)
+        mine = props[(props[:, 1] >= c_lo) & (props[:, 1] < c_hi)]
+        # stable by list order (= global rank order)
+        self.state = {
+            "segs": {c: mine[mine[:, 1] == c] for c in range(c_lo, c_hi)},
+            "plen": {c: int((mine[:, 1] == c).sum()) for c in range(c_lo, c_hi)},
+            "props": props.copy(),
+        }
+        for row in mine:
+            dep[self.labels[row[0]]] += row[3]
+
+    def shard_round(self, c_lo, c_hi, dep_ptr, delta_ptr):
+        dep = self.reg.get(dep_ptr)
+        delta = self.reg.get(delta_ptr)
+        delta[:] = 0
+        for c in range(c_lo, c_hi):
+            pl = self.state["plen"][c]
+            if pl == 0:
+                continue
+            cap = self.mbw[c] - self.weights[c] + dep[c]
+            seg = self.state["segs"][c]
+            arr = int(seg[:pl, 3].sum())
+            if arr > cap:
+                nl = max(0, min(pl, int(cap)))  # unit weights
+                for i in range(nl, pl):
+                    delta[self.labels[seg[i, 0]]] += seg[i, 3]
+                self.state["plen"][c] = nl
+                delta[self.k] = 1
+
+    def shard_finish_meta(self, c_lo, c_hi, cutoff_ptr, arr_ptr):
+        cutoff = self.reg.get(cutoff_ptr)
+        arr = self.reg.get(arr_ptr)
+        for c in range(c_lo, c_hi):
+            seg = self.state["segs"][c]
+            pl = self.state["plen"][c]
+            cutoff[c] = (2**63 - 1) if pl >= len(seg) else int(seg[pl, 2])
+            arr[c] = int(seg[:pl, 3].sum())
+
+    def shard_apply(self, it, chunk, cat_ptr, total, cutoff_ptr, arr_ptr,
+                    dep_ptr):
+        props = self.state["props"]
+        cutoff = self.reg.get(cutoff_ptr)
+        arr = self.reg.get(arr_ptr)
+        dep = self.reg.get(dep_ptr)
+        self.weights += arr - dep[: self.k]
+        mv = 0
+        for u, to, r, w in props:
+            if r < cutoff[to]:
+                self.labels[u] = to
+                mv += 1
+        self.moves += mv
+        return mv
+
+    def refine_end(self):
+        return 0, self.labels, None
+
+
+class _GlooShardComm(_GlooComm):
+    def __init__(self, registry):
+        super().__init__()
+        self.reg = registry
+
+    def alloc_prop_buffer(self, cap):
+        t = self.reg.add(self.torch.zeros((cap, 4), dtype=self.torch.int32))
+        return t, t.data_ptr()
+
+    def zeros(self, size, dtype):
+        return self.reg.add(self.torch.zeros(size, dtype=self.torch.int64))
+
+    def allreduce_(self, t):
+        self.dist.all_reduce(t)
+
+    def sync(self):
+        pass
+
+    def all_gather_props(self, buf, count):
+        ptr, total = super().all_gather_props(buf, count)
+        if total:
+            self.reg.add(self._keep)
+        return ptr, total
+
+
+def _shard_worker(rank, world, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from kaminpar_amd.multi import refine_dist_sharded
+
+        n, k = 4096, 8
+        reg = _Registry()
+        eng = _ShardFakeEngine(n, k, reg)
+        comm = _GlooShardComm(reg)
+        mbw = np.full(k, (n // k) * 1.05, np.int64)
+        part0 = (np.arange(n) % k).astype(np.uint32)
+        cut, labels, _ = refine_dist_sharded(eng, k, mbw, part0, 1, 2, comm)
+
+        # all ranks end bit-identical (gather labels and compare on rank 0)
+        import torch
+        lt = torch.from_numpy(labels.astype(np.int64))
+        gathered = [torch.zeros_like(lt) for _ in range(world)]
+        dist.all_gather(gathered, lt)
+        for g in gathered:
+            assert torch.equal(g, gathered[0])
+        # weights agree + caps never overshot
+        bw = np.bincount(labels, minlength=k).astype(np.int64)
+        assert (bw == eng.weights).all()
+        assert (bw <= mbw).all()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_refine_dist_sharded_gloo_world2():
+    """World-2 gloo run of the SHARDED-commit protocol with the numpy
+    engine restatement: collective plumbing works, the fixpoint converges,
+    both ranks end with identical labels and cap-respecting weights."""
+    import multiprocessing as mp
+
+    port = 29631
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_shard_worker, args=(r, 2, port))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(180)
+    assert all(p.exitcode == 0 for p in procs)
