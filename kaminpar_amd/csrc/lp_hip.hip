@@ -240,7 +240,9 @@ __global__ void k_phase_s(
   u32 c = kInvalid;
   i32 w = 0;
   if (slot < deg) {
-    const u32 v = adjncy[row + slot];
+    // non-temporal: the adjncy stream is read once per sweep and must not
+    // evict the label shadow from L2/MALL (the gather working set)
+    const u32 v = __builtin_nontemporal_load(&adjncy[row + slot]);
     c = labels_s[v]; // narrow shadow: u8 for k <= 256 (64 MB at scale 26,
                      // MALL-resident), u16 up to k <= 2048
     w = adjwgt ? adjwgt[row + slot] : 1;
@@ -362,7 +364,7 @@ __global__ void k_phase_m(
 
   const u32 rep_off = (lane % R) * k;
   for (u32 e = lane; e < deg; e += kWave) {
-    const u32 v = adjncy[row + e];
+    const u32 v = __builtin_nontemporal_load(&adjncy[row + e]);
     const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
     atomicAdd(&gains[rep_off + labels_s[v]], w);
   }
@@ -500,7 +502,7 @@ __global__ void k_phase_l_acc(
     __syncthreads();
     const u32 rep_off = (threadIdx.x % R) * k;
     for (u32 e = e_lo + threadIdx.x; e < e_hi; e += blockDim.x) {
-      const u32 v = adjncy[row + e];
+      const u32 v = __builtin_nontemporal_load(&adjncy[row + e]);
       const i32 w = kUnitWeights ? 1 : adjwgt[row + e];
       atomicAdd(&hist[rep_off + labels_s[v]], w);
     }
@@ -3274,8 +3276,14 @@ __global__ void k_activate_v2(
     const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     for (u32 e2 = lane; e2 < deg; e2 += kWave) {
       const u32 v = adjncy[row + e2];
-      active[v] = 1;
-      unit_active[v >> 6] = 1;
+      // conditional stores: most neighbours are already active in early
+      // sweeps (skip the dirty-line traffic), most units already marked
+      if (!active[v]) {
+        active[v] = 1;
+      }
+      if (!unit_active[v >> 6]) {
+        unit_active[v >> 6] = 1;
+      }
     }
   }
 }
