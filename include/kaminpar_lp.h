@@ -156,6 +156,12 @@ int64_t kmp_lp_refine(
  * merges vertices across community boundaries. NULL clears. */
 int kmp_lp_set_communities(kmp_lp_t *e, const uint32_t *communities);
 
+/* On-GPU degree-bucket rearrangement (permutator.cc:36-110 semantics,
+ * bit-identical to kmp_rearrange_degree_buckets): rebuilds the engine's
+ * device CSR in log2-degree-bucket order, writes perm_out[u_old] = u_new
+ * (len n). Call before refine/cluster; map partitions through perm_out. */
+int kmp_lp_rearrange_degree_buckets(kmp_lp_t *e, uint32_t *perm_out);
+
 int64_t kmp_lp_underload(
     kmp_lp_t *e,
     uint32_t k,

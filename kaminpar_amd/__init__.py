@@ -126,6 +126,8 @@ def _load():
     lib.kmp_lp_balance.argtypes = [vp, u32, p(i64), p(u32), u64, ctypes.c_int, vp]
     lib.kmp_lp_set_communities.restype = ctypes.c_int
     lib.kmp_lp_set_communities.argtypes = [vp, p(u32)]
+    lib.kmp_lp_rearrange_degree_buckets.restype = ctypes.c_int
+    lib.kmp_lp_rearrange_degree_buckets.argtypes = [vp, p(u32)]
     lib.kmp_lp_underload.restype = i64
     lib.kmp_lp_underload.argtypes = [vp, u32, p(i64), p(i64), p(u32), u64,
                                      ctypes.c_int, vp]
@@ -446,6 +448,16 @@ class LpEngine:
         if cut < 0:
             raise RuntimeError("kmp_lp_underload failed")
         return cut, part, stats
+
+    def rearrange_degree_buckets(self):
+        """On-GPU degree-bucket rearrangement of the engine's CSR (the
+        reference's default preprocessing, permutator.cc:36-110).
+        Returns perm with perm[u_old] = u_new."""
+        perm = np.zeros(self.n, dtype=np.uint32)
+        rc = _lib.kmp_lp_rearrange_degree_buckets(self._h, _u32p(perm))
+        if rc != 0:
+            raise RuntimeError("kmp_lp_rearrange_degree_buckets failed")
+        return perm
 
     def set_communities(self, communities):
         """Restrict clustering merges to stay within communities

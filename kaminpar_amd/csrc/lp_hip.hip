@@ -2482,6 +2482,153 @@ __global__ void k_scatter_v2(
   }
 }
 
+// ==================== degree-bucket rearrangement (on-GPU) ====================
+// The reference's default NodeOrdering::DEGREE_BUCKETS preprocessing
+// (graphutils/permutator.cc:36-110): stable counting sort of vertices into
+// 33 log2-degree buckets (deg-0 last) + CSR re-gather with neighbour
+// relabeling. Device-resident so the drop-in path needs no host pass
+// (bit-identical to the host kmp_rearrange_degree_buckets).
+constexpr u32 kDbBuckets = 33;
+
+__device__ inline u32 db_bucket_of(u32 deg) {
+  return deg == 0 ? kDbBuckets - 1 : (31 - __clz(deg) + 1);
+}
+
+__global__ void k_db_hist(
+    u32 n,
+    u32 rows,
+    u32 tpw,
+    const u64 *__restrict__ xadj,
+    u32 *__restrict__ histT // kDbBuckets x rows
+) {
+  __shared__ u32 cnt_lds[4 * kDbBuckets];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * kDbBuckets;
+  for (u32 b = lane; b < kDbBuckets; b += kWave) {
+    cnt[b] = 0;
+  }
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 T = (n + 63) >> 6;
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    const u32 u = (t << 6) + lane;
+    if (u < n) {
+      atomicAdd(&cnt[db_bucket_of(static_cast<u32>(xadj[u + 1] - xadj[u]))], 1u);
+    }
+  }
+  for (u32 b = lane; b < kDbBuckets; b += kWave) {
+    histT[b * rows + row] = cnt[b];
+  }
+}
+
+__global__ void k_db_scatter(
+    u32 n,
+    u32 rows,
+    u32 tpw,
+    const u64 *__restrict__ xadj,
+    const u32 *__restrict__ offT,
+    u32 *__restrict__ perm, // perm[u_old] = u_new
+    u32 *__restrict__ inv   // inv[u_new] = u_old
+) {
+  __shared__ u32 cnt_lds[4 * kDbBuckets];
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 wv = threadIdx.x >> 6;
+  u32 *cnt = cnt_lds + wv * kDbBuckets;
+  const u32 row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  for (u32 b = lane; b < kDbBuckets; b += kWave) {
+    cnt[b] = offT[b * rows + row];
+  }
+  const u32 T = (n + 63) >> 6;
+  const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
+  for (u32 t = row * tpw; t < t1; ++t) {
+    const u32 u = (t << 6) + lane;
+    const bool valid = u < n;
+    const u32 b = valid ? db_bucket_of(static_cast<u32>(xadj[u + 1] - xadj[u]))
+                        : 0xFFFFFFFFu;
+    u32 rank = 0, total = 0, leader = 0;
+    bool seen = false;
+    for (u32 j = 0; j < kWave; ++j) {
+      const u32 bj = __shfl(b, j, kWave);
+      if (valid && bj == b) {
+        ++total;
+        if (j < lane) {
+          ++rank;
+        }
+        if (!seen) {
+          leader = j;
+          seen = true;
+        }
+      }
+    }
+    u32 base0 = 0;
+    if (valid && rank == 0) {
+      base0 = cnt[b];
+      cnt[b] = base0 + total;
+    }
+    const u32 base = __shfl(base0, leader, kWave);
+    if (valid) {
+      const u32 pos = base + rank;
+      perm[u] = pos;
+      inv[pos] = u;
+    }
+  }
+}
+
+__global__ void k_db_degrees(
+    u32 n,
+    const u64 *__restrict__ xadj,
+    const u32 *__restrict__ inv,
+    u64 *__restrict__ new_deg // n+1 entries ([0] = 0 written by host memset)
+) {
+  const u32 v = blockIdx.x * blockDim.x + threadIdx.x;
+  if (v < n) {
+    const u32 u = inv[v];
+    new_deg[v + 1] = xadj[u + 1] - xadj[u];
+  }
+}
+
+// 16-lane subgroups, 4 vertices/wave: copy each new row from its old
+// position, relabeling neighbours through perm.
+__global__ void k_db_gather(
+    u32 n,
+    const u64 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    const i32 *__restrict__ adjwgt,
+    const i32 *__restrict__ vwgt,
+    const u32 *__restrict__ perm,
+    const u32 *__restrict__ inv,
+    const u64 *__restrict__ new_xadj,
+    u32 *__restrict__ new_adjncy,
+    i32 *__restrict__ new_adjwgt,
+    i32 *__restrict__ new_vwgt
+) {
+  const u32 lane = threadIdx.x & (kWave - 1);
+  const u32 sub = lane >> 4;
+  const u32 slot = lane & 15;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 num_waves = (gridDim.x * blockDim.x) >> 6;
+  for (u32 base = wave_id * 4; base < n; base += num_waves * 4) {
+    const u32 v = base + sub;
+    if (v >= n) {
+      continue;
+    }
+    const u32 u = inv[v];
+    const u64 src = xadj[u];
+    const u64 dst = new_xadj[v];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - src);
+    for (u32 i = slot; i < deg; i += 16) {
+      new_adjncy[dst + i] = perm[adjncy[src + i]];
+      if (adjwgt != nullptr) {
+        new_adjwgt[dst + i] = adjwgt[src + i];
+      }
+    }
+    if (slot == 0 && vwgt != nullptr) {
+      new_vwgt[v] = vwgt[u];
+    }
+  }
+}
+
 // ==================== sharded commit (multi-GPU) ====================
 // Rank-sharded deterministic admission over the ALL-GATHERED proposal list
 // (dense Prop records in global rank order): each rank sorts and fixpoints
@@ -5471,6 +5618,107 @@ kmp_graph_t *kmp_lp_download_graph(const kmp_lp_t *e) {
       e->n, e->m, h_xadj.data(), h_adj.data(), e->has_vwgt ? h_vw.data() : nullptr,
       e->has_adjwgt ? h_wg.data() : nullptr
   );
+}
+
+// On-GPU degree-bucket rearrangement (the reference's default
+// NodeOrdering::DEGREE_BUCKETS preprocessing, graphutils/permutator.cc:36-110;
+// bit-identical to the host kmp_rearrange_degree_buckets): rebuilds the
+// engine's device CSR in bucket order and writes perm_out[u_old] = u_new.
+// Call before any refine/cluster; LP state is reset by the next *_begin.
+int kmp_lp_rearrange_degree_buckets(kmp_lp_t *e, u32 *perm_out) {
+  const u32 n = e->n;
+  const u64 m = e->m;
+  const u32 threads = 256;
+  hipStream_t s = e->stream;
+  const u32 rows = 1024;
+  const u32 T = (n + 63) >> 6;
+  const u32 tpw = (T + rows - 1) / rows;
+
+  u32 *d_perm = nullptr, *d_inv = nullptr, *d_hist = nullptr, *d_off = nullptr;
+  u64 *d_nx = nullptr;
+  u32 *d_nadj = nullptr;
+  i32 *d_nwgt = nullptr, *d_nvw = nullptr;
+  HIP_CHECK(hipMalloc(&d_perm, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_inv, sizeof(u32) * n));
+  HIP_CHECK(hipMalloc(&d_hist, sizeof(u32) * kDbBuckets * rows));
+  HIP_CHECK(hipMalloc(&d_off, sizeof(u32) * kDbBuckets * rows));
+  HIP_CHECK(hipMalloc(&d_nx, sizeof(u64) * (n + 1)));
+  HIP_CHECK(hipMalloc(&d_nadj, sizeof(u32) * m));
+  if (e->has_adjwgt) {
+    HIP_CHECK(hipMalloc(&d_nwgt, sizeof(i32) * m));
+  }
+  if (e->has_vwgt) {
+    HIP_CHECK(hipMalloc(&d_nvw, sizeof(i32) * n));
+  }
+
+  hipLaunchKernelGGL(
+      k_db_hist, dim3(rows / 4), dim3(threads), 0, s, n, rows, tpw, e->d_xadj, d_hist
+  );
+  LAUNCH_CHECK();
+  // exclusive scan of the 33 x rows matrix (single WG; one-time setup cost)
+  {
+    // reuse k_scan_small's machinery via a plain rocprim scan
+    void *tmp = nullptr;
+    size_t tb = 0;
+    HIP_CHECK(rocprim::exclusive_scan(
+        nullptr, tb, d_hist, d_off, 0u, kDbBuckets * rows, rocprim::plus<u32>()
+    ));
+    HIP_CHECK(hipMalloc(&tmp, tb));
+    HIP_CHECK(rocprim::exclusive_scan(
+        tmp, tb, d_hist, d_off, 0u, kDbBuckets * rows, rocprim::plus<u32>(), s
+    ));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFree(tmp));
+  }
+  hipLaunchKernelGGL(
+      k_db_scatter, dim3(rows / 4), dim3(threads), 0, s, n, rows, tpw, e->d_xadj, d_off,
+      d_perm, d_inv
+  );
+  LAUNCH_CHECK();
+  HIP_CHECK(hipMemsetAsync(d_nx, 0, sizeof(u64), s));
+  hipLaunchKernelGGL(
+      k_db_degrees, dim3(ceil_div(n, threads)), dim3(threads), 0, s, n, e->d_xadj, d_inv, d_nx
+  );
+  LAUNCH_CHECK();
+  {
+    void *tmp = nullptr;
+    size_t tb = 0;
+    HIP_CHECK(rocprim::inclusive_scan(nullptr, tb, d_nx, d_nx, n + 1, rocprim::plus<u64>()));
+    HIP_CHECK(hipMalloc(&tmp, tb));
+    HIP_CHECK(rocprim::inclusive_scan(tmp, tb, d_nx, d_nx, n + 1, rocprim::plus<u64>(), s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFree(tmp));
+  }
+  hipLaunchKernelGGL(
+      k_db_gather, dim3(16384), dim3(threads), 0, s, n, e->d_xadj, e->d_adjncy, e->d_adjwgt,
+      e->d_vwgt, d_perm, d_inv, d_nx, d_nadj, d_nwgt, d_nvw
+  );
+  LAUNCH_CHECK();
+  HIP_CHECK(hipMemcpyAsync(perm_out, d_perm, sizeof(u32) * n, hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+
+  // adopt the rearranged CSR
+  HIP_CHECK(hipFree(e->d_xadj));
+  HIP_CHECK(hipFree(e->d_adjncy));
+  e->d_xadj = d_nx;
+  e->d_adjncy = d_nadj;
+  if (e->has_adjwgt) {
+    HIP_CHECK(hipFree(e->d_adjwgt));
+    e->d_adjwgt = d_nwgt;
+  }
+  if (e->has_vwgt) {
+    HIP_CHECK(hipFree(e->d_vwgt));
+    e->d_vwgt = d_nvw;
+  }
+  // remap the isolated-vertex list (ids changed)
+  for (u32 &u : e->isolated) {
+    u = perm_out[u];
+  }
+  HIP_CHECK(hipFree(d_perm));
+  HIP_CHECK(hipFree(d_inv));
+  HIP_CHECK(hipFree(d_hist));
+  HIP_CHECK(hipFree(d_off));
+  return 0;
 }
 
 u32 kmp_lp_n(const kmp_lp_t *e) { return e->n; }
