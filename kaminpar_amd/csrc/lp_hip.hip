@@ -3100,10 +3100,6 @@ __global__ void k_scan_coop(
 __global__ void k_fixpoint_v2(
     u32 k,
     u32 has_vwgt,
-    u32 pos_lo,
-    u32 pos_hi,
-    u32 n,
-    u64 iter_seed,
     const u32 *__restrict__ seg_off,
     u32 *__restrict__ prefix_len,
     const unsigned long long *__restrict__ dep, // [0..k) out, [k..2k) in
@@ -3112,7 +3108,6 @@ __global__ void k_fixpoint_v2(
     i64 *__restrict__ weights,
     const i64 *__restrict__ maxw,
     unsigned long long *__restrict__ moves,
-    uint8_t *__restrict__ unit_active,
     i64 *__restrict__ pw // scratch: segmented prefix weights (weighted only)
 ) {
   __shared__ unsigned long long ddep[256];   // current departures
@@ -3247,19 +3242,6 @@ __global__ void k_fixpoint_v2(
       atomicAdd(moves, t);
     }
   }
-  // clear the chunk's unit-active bits (the per-vertex flags were cleared
-  // inline by k_phase_s; activation re-marks receiving units afterwards)
-  {
-    const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
-    const u32 units = (pos_hi - pos_lo) >> 6;
-    const u32 u0 = pos_lo >> 6;
-    for (u32 uix = tid; uix < units; uix += blockDim.x) {
-      const u32 vb = fp(u0 + uix);
-      if (unit_active[vb]) {
-        unit_active[vb] = 0;
-      }
-    }
-  }
 }
 
 // Single-workgroup variant of the histogram scan for small chunks (entries
@@ -3372,16 +3354,24 @@ __global__ void k_commit_underload(
   }
 }
 
-// Apply admitted labels (full grid, count read from device).
+// Apply admitted labels + clear the chunk's unit-active bits (full grid;
+// the per-vertex flags were cleared inline by k_phase_s, and activation --
+// which re-marks receiving units -- runs in the NEXT kernel, so stream
+// order preserves the clear-then-activate semantics).
 __global__ void k_apply_v2(
     u32 k,
+    u32 pos_lo,
+    u32 pos_hi,
+    u32 n,
+    u64 iter_seed,
     const u32 *__restrict__ seg_off,
     const u32 *__restrict__ prefix_len,
     const u32 *__restrict__ s_u,
     const uint16_t *__restrict__ s_to,
     u32 *__restrict__ labels,
     uint16_t *__restrict__ labels16,
-    uint8_t *__restrict__ labels8 // null unless k <= 256
+    uint8_t *__restrict__ labels8, // null unless k <= 256
+    uint8_t *__restrict__ unit_active
 ) {
   const u32 count = seg_off[k];
   const u32 stride = gridDim.x * blockDim.x;
@@ -3393,6 +3383,17 @@ __global__ void k_apply_v2(
       labels16[u] = static_cast<uint16_t>(c);
       if (labels8 != nullptr) {
         labels8[u] = static_cast<uint8_t>(c);
+      }
+    }
+  }
+  {
+    const kmp::FeistelPerm fp(kmp::num_units(n), iter_seed);
+    const u32 units = (pos_hi - pos_lo) >> 6;
+    const u32 u0 = pos_lo >> 6;
+    for (u32 uix = blockIdx.x * blockDim.x + threadIdx.x; uix < units; uix += stride) {
+      const u32 vb = fp(u0 + uix);
+      if (unit_active[vb]) {
+        unit_active[vb] = 0;
       }
     }
   }
@@ -3423,14 +3424,10 @@ __global__ void k_activate_v2(
     const u32 deg = static_cast<u32>(xadj[u + 1] - row);
     for (u32 e2 = lane; e2 < deg; e2 += kWave) {
       const u32 v = adjncy[row + e2];
-      // conditional stores: most neighbours are already active in early
-      // sweeps (skip the dirty-line traffic), most units already marked
-      if (!active[v]) {
-        active[v] = 1;
-      }
-      if (!unit_active[v >> 6]) {
-        unit_active[v >> 6] = 1;
-      }
+      // blind stores: measured FASTER than read-test-write (the reads add
+      // serialized gather latency; the dirty-line cost is smaller)
+      active[v] = 1;
+      unit_active[v >> 6] = 1;
     }
   }
 }
@@ -3858,14 +3855,14 @@ void commit_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
       k_fixpoint_v2, dim3(1), dim3(threads), 0, e->stream, e->k,
-      static_cast<u32>(e->has_vwgt ? 1 : 0), pos_lo, pos_hi, e->n, iseed, e->d_seg_off,
-      e->d_prefix_len, e->d_dep, e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves,
-      e->d_unit_active, e->d_pw
+      static_cast<u32>(e->has_vwgt ? 1 : 0), e->d_seg_off, e->d_prefix_len, e->d_dep,
+      e->d_s_w, e->d_s_b, e->d_weights, e->d_maxw, e->d_moves, e->d_pw
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
-      k_apply_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, e->d_seg_off, e->d_prefix_len,
-      e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16, e->d_labels8
+      k_apply_v2, dim3(2048), dim3(threads), 0, e->stream, e->k, pos_lo, pos_hi, e->n, iseed,
+      e->d_seg_off, e->d_prefix_len, e->d_s_u, e->d_s_to, e->d_labels, e->d_labels16,
+      e->d_labels8, e->d_unit_active
   );
   LAUNCH_CHECK();
   hipLaunchKernelGGL(
