@@ -3664,8 +3664,8 @@ void engine_alloc_common(kmp_lp_t *e) {
   HIP_CHECK(hipMalloc(&e->d_s_to, sizeof(uint16_t) * C));
   HIP_CHECK(hipMalloc(&e->d_s_r, sizeof(u32) * C));
   HIP_CHECK(hipMalloc(&e->d_s_b, sizeof(uint16_t) * C));
-  HIP_CHECK(hipMalloc(&e->d_histT, sizeof(u32) * 65536));
-  HIP_CHECK(hipMalloc(&e->d_offT, sizeof(u32) * 65536));
+  HIP_CHECK(hipMalloc(&e->d_histT, sizeof(u32) * 262144));
+  HIP_CHECK(hipMalloc(&e->d_offT, sizeof(u32) * 262144));
   HIP_CHECK(hipMalloc(&e->d_seg_off, sizeof(u32) * 257));
   HIP_CHECK(hipMalloc(&e->d_bar, sizeof(u32) * 16));
   HIP_CHECK(hipMemsetAsync(e->d_bar, 0, sizeof(u32) * 16, e->stream));
@@ -4127,7 +4127,10 @@ int kmp_lp_refine_begin(
   // deliberately SMALL (64 blocks): the commit stages are latency-bound and
   // the grid barrier cost grows with the block count (tools/bar_bench.hip).
   {
-    u32 rows = (65536u / k) & ~3u;
+    // histogram rows: enough waves to hide tile-load latency even at
+    // k = 256 (a 256-row table serialized 256 tiles per wave at scale 28 --
+    // measured regression); table <= 262144 entries = 1 MB
+    u32 rows = (262144u / k) & ~3u;
     if (rows < 256) {
       rows = 256;
     }
