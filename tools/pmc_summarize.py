@@ -23,6 +23,9 @@ def main():
             with open(path, newline="") as fh:
                 for row in csv.DictReader(fh):
                     name = row.get("Kernel_Name") or row.get("Kernel-Name")
+                    if name:
+                        name = name.replace("(anonymous namespace)::", "")
+                        name = name.split("(")[0].strip()
                     counter = row.get("Counter_Name") or row.get("Counter-Name")
                     value = float(row.get("Counter_Value")
                                   or row.get("Counter-Value") or 0)
