@@ -5575,11 +5575,15 @@ i64 kmp_contract_engine(
   HIP_CHECK(hipStreamCreate(&e2->stream));
   HIP_CHECK(hipEventCreateWithFlags(&e2->sync_ev, hipEventDisableTiming));
   {
-    int dev = 0;
-    HIP_CHECK(hipGetDevice(&dev));
-    hipDeviceProp_t props;
-    HIP_CHECK(hipGetDeviceProperties(&props, dev));
-    e2->mp_count = props.multiProcessorCount;
+    static int cached_mp = 0; // hipGetDeviceProperties costs ms; per-level calls add up
+    if (cached_mp == 0) {
+      int dev = 0;
+      HIP_CHECK(hipGetDevice(&dev));
+      hipDeviceProp_t props;
+      HIP_CHECK(hipGetDeviceProperties(&props, dev));
+      cached_mp = props.multiProcessorCount;
+    }
+    e2->mp_count = cached_mp;
   }
   e2->d_xadj = o.d_cxadj;
   e2->d_adjncy = o.d_cadj;
