@@ -2014,7 +2014,6 @@ __global__ void k_activate(
   for (u32 e = lane; e < deg; e += kWave) {
     const u32 v = adjncy[row + e];
     active[v] = 1;
-    unit_active[v >> 6]; // (read avoids compiler warning removal)
     unit_active[v >> 6] = 1;
   }
 }

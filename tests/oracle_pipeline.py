@@ -101,7 +101,14 @@ def oracle_partition_deep(oracle, g, k, eps=0.03, seed=1, iters=5,
     # finest level (no eager coarsest split)
     xadj = np.asarray(g.xadj, dtype=np.int64)
     d = xadj[1:] - xadj[:-1]
-    heavy = g.n * int((d.astype(object) * d).sum()) >= 2 * int(d.sum()) ** 2
+    # exact integer sum-of-squares without the object-dtype blowup (an
+    # object sum over 67M degrees measured ~1.3 s per partition): split the
+    # int64 products into high/low 32-bit halves and recombine as Python
+    # ints -- bit-exact, vectorized
+    sq = d * d  # per-element fits int64 (deg < 2^31)
+    ssum = (int((sq & 0xFFFFFFFF).sum(dtype=np.int64))
+            + (int((sq >> 32).sum(dtype=np.int64)) << 32))
+    heavy = g.n * ssum >= 2 * int(d.sum()) ** 2
     late_splits = heavy and (g.n <= (1 << 21) or split_c >= g.n)
 
     total_w = g.total_node_weight
