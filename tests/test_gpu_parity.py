@@ -1138,3 +1138,21 @@ def test_gpu_degree_bucket_rearrangement(oracle):
     cut2, p2, _ = eng2.refine(k, mbw, part0r, seed=2, iters=5)
     assert cut1 == cut2
     assert np.array_equal(p1, p2)
+
+
+@pytest.mark.gpu
+def test_refine_parity_large_k_legacy_path(oracle):
+    """k in (256, 2048] takes the legacy commit path (radix sort + u16
+    shadow instead of the v2 counting-sort + u8 shadow): bit-parity with
+    the oracle at k=1024."""
+    _require_gpu()
+    g = ka.Graph.rmat(13, 8, seed=3)
+    k = 1024
+    part0 = ka.random_partition(g.n, k, seed=8)
+    mbw = np.full(k, g.max_block_weight(k, 0.10), dtype=np.int64)
+    eng = ka.LpEngine(g)
+    cut, part, stats = eng.refine(k, mbw, part0, seed=5, iters=5)
+    ocut, opart, ostats = oracle_refine(oracle, g, k, mbw, part0, seed=5, iters=5)
+    assert cut == ocut
+    assert np.array_equal(part, opart)
+    assert stats.moves == ostats[1]
