@@ -1786,18 +1786,6 @@ __global__ void k_seg_bounds(
   }
 }
 
-__global__ void k_seg_len(
-    u32 k_or_n,
-    const u32 *__restrict__ seg_begin,
-    const u32 *__restrict__ seg_end,
-    u32 *__restrict__ prefix_len
-) {
-  const u32 c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c < k_or_n && seg_end[c] > seg_begin[c]) {
-    prefix_len[c] = seg_end[c] - seg_begin[c];
-  }
-}
-
 // Per-WG LDS histogram of departures (one global atomic per cluster per WG;
 // a per-proposal global atomic on k addresses serializes badly). Dynamic
 // LDS: k x u64. Grid-stride so the WG count stays bounded.
