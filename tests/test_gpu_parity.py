@@ -1156,3 +1156,32 @@ def test_refine_parity_large_k_legacy_path(oracle):
     assert cut == ocut
     assert np.array_equal(part, opart)
     assert stats.moves == ostats[1]
+
+
+@pytest.mark.gpu
+def test_reset_and_replay_deterministic(oracle):
+    """The bench contract path (refine_begin once, then reset + run_sweeps
+    per step) replays captured sweep graphs from the second step on: every
+    step must produce identical moves/arcs, and the final labels must equal
+    a fresh monolithic refine (bit-parity of the replay path)."""
+    _require_gpu()
+    g = ka.Graph.rmat(12, 8, seed=11)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=2)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+
+    eng = ka.LpEngine(g)
+    cut_ref, part_ref, sref = eng.refine(k, mbw, part0, seed=7, iters=5)
+
+    eng.refine_begin(k, mbw, part0, seed=7)
+    seen = []
+    for step in range(4):  # step 0 plain, steps 1+ captured-graph replays
+        eng.reset()
+        eng.run_sweeps(5)
+        st = eng.get_stats()
+        seen.append((st.arcs_scanned, st.moves))
+    cut2, part2, _ = eng.refine_end()
+    assert all(s == seen[0] for s in seen), seen
+    assert seen[0] == (sref.arcs_scanned, sref.moves)
+    assert cut2 == cut_ref
+    assert np.array_equal(part2, part_ref)
