@@ -140,7 +140,21 @@ def main():
     from kaminpar_amd.multi import TorchComm, refine_dist_sharded
 
     level_sizes = []
-    if world > 1:
+    # KMP_FORCE_DIST=1 exercises the full sharded-commit path (TorchComm +
+    # real collectives) at world=1 -- a pre-flight for multi-GPU runs
+    force_dist = os.environ.get("KMP_FORCE_DIST") == "1"
+    if world > 1 or (force_dist and args.workload == "refine"):
+        if world == 1:
+            import torch
+            import torch.distributed as dist
+
+            torch.zeros(1, device=device)  # torch HIP runtime first
+            if not dist.is_initialized():
+                os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+                os.environ.setdefault("MASTER_PORT", "29517")
+                os.environ.setdefault("RANK", "0")
+                os.environ.setdefault("WORLD_SIZE", "1")
+                dist.init_process_group("nccl")
         comm = TorchComm(device)
     else:
         comm = None  # fast path: device-resident stepping inside C++
