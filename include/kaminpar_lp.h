@@ -253,6 +253,11 @@ int64_t kmp_lp_refine_end(kmp_lp_t *e, uint32_t *partition, kmp_lp_stats_t *stat
  *   shard_apply  -> every rank applies the identical admitted set + weight
  *                   updates + active-set maintenance; returns moves.
  * All d_* pointers are DEVICE pointers. Bit-identical to kmp_lp_commit. */
+/* Adopt an external HIP stream (e.g. torch's current stream) so the
+ * multi-GPU collectives and the engine's kernels share one stream; pass
+ * NULL to restore the engine's own stream. Engine must be idle. */
+int kmp_lp_set_stream(kmp_lp_t *e, void *external_stream);
+
 int kmp_lp_shard_begin(kmp_lp_t *e, uint32_t c_lo, uint32_t c_hi,
                        const void *d_props, uint32_t count, long long *d_dep_out);
 int kmp_lp_shard_round(kmp_lp_t *e, uint32_t c_lo, uint32_t c_hi,

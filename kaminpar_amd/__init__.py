@@ -141,6 +141,8 @@ def _load():
     lib.kmp_lp_phase_a.argtypes = [vp, ctypes.c_int, u32, u32, u32, vp, u32]
     lib.kmp_lp_commit.restype = i64
     lib.kmp_lp_commit.argtypes = [vp, ctypes.c_int, u32, vp, u32]
+    lib.kmp_lp_set_stream.restype = ctypes.c_int
+    lib.kmp_lp_set_stream.argtypes = [vp, vp]
     lib.kmp_lp_shard_begin.restype = ctypes.c_int
     lib.kmp_lp_shard_begin.argtypes = [vp, u32, u32, vp, u32, vp]
     lib.kmp_lp_shard_round.restype = ctypes.c_int
@@ -504,6 +506,13 @@ class LpEngine:
         if mv < 0:
             raise RuntimeError("kmp_lp_commit failed")
         return int(mv)
+
+    def set_stream(self, stream_ptr):
+        """Adopt an external HIP stream (torch: cuda.current_stream().cuda_stream);
+        None restores the engine's own stream."""
+        rc = _lib.kmp_lp_set_stream(self._h, stream_ptr)
+        if rc != 0:
+            raise RuntimeError("kmp_lp_set_stream failed")
 
     def shard_begin(self, c_lo, c_hi, d_props_ptr, count, d_dep_out_ptr):
         """Sharded commit step 1: sort own targets from the all-gathered

@@ -3546,6 +3546,7 @@ struct kmp_lp_t {
   u64 last_emptied = 0;
 
   hipStream_t stream = nullptr;
+  hipStream_t own_stream = nullptr; // saved when an external stream is adopted
 
   // run bookkeeping
   double phase_a_ms = 0.0;
@@ -4096,11 +4097,30 @@ void kmp_lp_free(kmp_lp_t *e) {
   if (e->h_hacc) {
     (void)hipHostFree(e->h_hacc);
   }
-  (void)hipStreamDestroy(e->stream);
+  (void)hipStreamDestroy(e->own_stream != nullptr ? e->own_stream : e->stream);
   delete e;
 }
 
 u32 kmp_lp_num_chunks(const kmp_lp_t *) { return kmp::kNumChunks; }
+
+// Adopt an external HIP stream (e.g. torch's current stream) so the
+// multi-GPU driver's collectives and the engine's kernels share ONE stream
+// and the per-call cross-stream hipStreamSynchronize bridges disappear.
+// Pass 0 to restore the engine's own stream. The engine must be idle.
+int kmp_lp_set_stream(kmp_lp_t *e, void *external_stream) {
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  engine_destroy_sweep_graphs(e); // graphs are bound to the old stream
+  if (external_stream != nullptr) {
+    if (e->own_stream == nullptr) {
+      e->own_stream = e->stream;
+    }
+    e->stream = static_cast<hipStream_t>(external_stream);
+  } else if (e->own_stream != nullptr) {
+    e->stream = e->own_stream;
+    e->own_stream = nullptr;
+  }
+  return 0;
+}
 
 int kmp_lp_refine_begin(
     kmp_lp_t *e, u32 k, const i64 *max_block_weights, const u32 *partition, u64 seed
@@ -4734,7 +4754,9 @@ int kmp_lp_shard_begin(
       LAUNCH_CHECK();
     }
   }
-  HIP_CHECK(hipStreamSynchronize(e->stream));
+  if (e->own_stream == nullptr) {
+    HIP_CHECK(hipStreamSynchronize(e->stream)); // cross-stream caller
+  }
   return 0;
 }
 
@@ -4749,7 +4771,9 @@ int kmp_lp_shard_round(
       e->d_s_b, e->d_pw, e->d_weights, e->d_maxw, d_delta_out
   );
   LAUNCH_CHECK();
-  HIP_CHECK(hipStreamSynchronize(e->stream));
+  if (e->own_stream == nullptr) {
+    HIP_CHECK(hipStreamSynchronize(e->stream));
+  }
   return 0;
 }
 
@@ -4765,7 +4789,9 @@ int kmp_lp_shard_finish_meta(
       d_arr_out
   );
   LAUNCH_CHECK();
-  HIP_CHECK(hipStreamSynchronize(e->stream));
+  if (e->own_stream == nullptr) {
+    HIP_CHECK(hipStreamSynchronize(e->stream));
+  }
   return 0;
 }
 

@@ -100,6 +100,11 @@ def refine_dist_sharded(eng, k, max_block_weights, partition, seed, iters,
     c_lo, c_hi = target_range(k, comm.rank, comm.world)
 
     eng.refine_begin(k, max_block_weights, partition, seed)
+    # share torch's stream with the engine: the collectives and the kernels
+    # then order naturally and the per-call cross-stream syncs disappear
+    stream_ptr = comm.engine_stream_ptr()
+    if stream_ptr is not None:
+        eng.set_stream(stream_ptr)
     buf, buf_ptr = comm.alloc_prop_buffer(cap)
     dep = comm.zeros(k + 1, "i64")
     delta = comm.zeros(k + 1, "i64")
@@ -177,6 +182,11 @@ class TorchComm:
         return self.torch.zeros(size, dtype=self.torch.int64,
                                 device=self.device)
 
+    def engine_stream_ptr(self):
+        if str(self.device).startswith("cuda"):
+            return self.torch.cuda.current_stream().cuda_stream
+        return None
+
     def allreduce_(self, t):
         self.dist.all_reduce(t)
 
@@ -226,6 +236,11 @@ class LocalComm:
         assert dtype == "i64"
         return self.torch.zeros(size, dtype=self.torch.int64,
                                 device=self.device)
+
+    def engine_stream_ptr(self):
+        if str(self.device).startswith("cuda"):
+            return self.torch.cuda.current_stream().cuda_stream
+        return None
 
     def allreduce_(self, t):
         pass

@@ -288,6 +288,9 @@ class _GlooShardComm(_GlooComm):
     def zeros(self, size, dtype):
         return self.reg.add(self.torch.zeros(size, dtype=self.torch.int64))
 
+    def engine_stream_ptr(self):
+        return None  # CPU
+
     def allreduce_(self, t):
         self.dist.all_reduce(t)
 
