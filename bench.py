@@ -137,7 +137,8 @@ def main():
         import math
         mcw = int(0.03 * g.total_node_weight / min(max(g.n // 2000, 2), k))
 
-    from kaminpar_amd.multi import TorchComm, refine_dist_sharded
+    from kaminpar_amd.multi import (TorchComm, refine_dist,
+                                    refine_dist_sharded)
 
     level_sizes = []
     # KMP_FORCE_DIST=1 exercises the full sharded-commit path (TorchComm +
@@ -181,7 +182,12 @@ def main():
             eng.reset()
             eng.run_sweeps(args.iters)
             return None, eng.get_stats()
-        cut, part, stats = refine_dist_sharded(eng, k, mbw, part0, args.seed, args.iters, comm)
+        # KMP_DIST_MODE=replicated selects the round-1 replicated-commit
+        # path (fewer per-chunk collectives; the sharded path is default)
+        dist_fn = (refine_dist
+                   if os.environ.get("KMP_DIST_MODE") == "replicated"
+                   else refine_dist_sharded)
+        cut, part, stats = dist_fn(eng, k, mbw, part0, args.seed, args.iters, comm)
         return cut, stats
 
     def barrier_sync():
