@@ -4804,8 +4804,6 @@ i64 kmp_lp_shard_apply(
   const u64 iseed = iter_seed_of(e->seed, iter);
   const u32 chunk_lo = chunk * e->C;
   const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
-  HIP_CHECK(hipMemcpyAsync(&e->h_moves[0], e->d_moves, sizeof(unsigned long long),
-                           hipMemcpyDeviceToHost, e->stream));
   const Prop *props = static_cast<const Prop *>(d_props);
   {
     const u32 span = count > e->k ? count : e->k;
@@ -4830,10 +4828,9 @@ i64 kmp_lp_shard_apply(
     );
     LAUNCH_CHECK();
   }
-  HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
-                           hipMemcpyDeviceToHost, e->stream));
-  sync_spin(e);
-  return static_cast<i64>(e->h_moves[1] - e->h_moves[0]);
+  // no per-call moves readback: callers poll the cumulative counter per
+  // sweep via kmp_lp_get_stats (one sync per sweep instead of per chunk)
+  return 0;
 }
 
 i64 kmp_lp_refine_end(kmp_lp_t *e, u32 *partition, kmp_lp_stats_t *stats) {

@@ -110,9 +110,9 @@ def refine_dist_sharded(eng, k, max_block_weights, partition, seed, iters,
     delta = comm.zeros(k + 1, "i64")
     cutoff = comm.zeros(k, "i64")  # i64 cells (sum-as-allgather)
     arr = comm.zeros(k, "i64")
+    moves_prev = 0
 
     for it in range(iters):
-        sweep_moves = 0
         for chunk in range(num_chunks):
             lo, hi = chunk_ranges(n, num_chunks, chunk)
             if lo >= hi:
@@ -137,11 +137,14 @@ def refine_dist_sharded(eng, k, max_block_weights, partition, seed, iters,
             comm.allreduce_(cutoff)  # zeros elsewhere: sum == allgather
             comm.allreduce_(arr)
             comm.sync()
-            sweep_moves += eng.shard_apply(it, chunk, cat_ptr, total,
-                                           cutoff.data_ptr(), arr.data_ptr(),
-                                           dep.data_ptr())
-        if sweep_moves == 0:
+            eng.shard_apply(it, chunk, cat_ptr, total,
+                            cutoff.data_ptr(), arr.data_ptr(),
+                            dep.data_ptr())
+        # one move-counter readback per sweep (not per chunk)
+        moves_now = eng.get_stats().moves
+        if moves_now == moves_prev:
             break
+        moves_prev = moves_now
 
     return eng.refine_end()
 
@@ -198,7 +201,10 @@ class TorchComm:
         torch, dist = self.torch, self.dist
         buf[self._cap, 0] = count
         dist.all_gather(self._gather_bufs, buf)
-        counts = [int(b[self._cap, 0].item()) for b in self._gather_bufs]
+        # ONE host sync for all ranks' counts (per-buffer .item() calls are
+        # world separate syncs)
+        counts = torch.stack(
+            [b[self._cap, 0] for b in self._gather_bufs]).cpu().tolist()
         total = sum(counts)
         if total == 0:
             return buf.data_ptr(), 0

@@ -1044,8 +1044,8 @@ def test_sharded_commit_two_rank_sim(oracle):
             for _ in range(world)]
     ranges = [target_range(k, r, world) for r in range(world)]
 
+    sweep_moves = 0
     for it in range(5):
-        sweep_moves = 0
         for chunk in range(num_chunks):
             lo = chunk * C
             hi = min(lo + C, P)
@@ -1091,15 +1091,16 @@ def test_sharded_commit_two_rank_sim(oracle):
                 cutoffs[r].copy_(csum)
                 arrs[r].copy_(asum)
             torch.cuda.synchronize()
-            mv = [engs[r].shard_apply(it, chunk, cat.data_ptr(), total,
-                                      cutoffs[r].data_ptr(),
-                                      arrs[r].data_ptr(),
-                                      deps[r].data_ptr())
-                  for r in range(world)]
-            assert mv[0] == mv[1]
-            sweep_moves += mv[0]
-        if sweep_moves == 0:
-            break
+            for r in range(world):
+                engs[r].shard_apply(it, chunk, cat.data_ptr(), total,
+                                    cutoffs[r].data_ptr(),
+                                    arrs[r].data_ptr(),
+                                    deps[r].data_ptr())
+        mv = [engs[r].get_stats().moves for r in range(world)]
+        assert mv[0] == mv[1]
+        if mv[0] == sweep_moves:
+            break  # no moves this sweep
+        sweep_moves = mv[0]
 
     results = [e.refine_end() for e in engs]
     for cut_r, part_r, _ in results:
