@@ -272,6 +272,11 @@ class _ShardFakeEngine:
         self.moves += mv
         return mv
 
+    def get_stats(self):
+        from types import SimpleNamespace
+
+        return SimpleNamespace(moves=self.moves, arcs_scanned=0)
+
     def refine_end(self):
         return 0, self.labels, None
 
