@@ -182,12 +182,24 @@ def main():
             eng.reset()
             eng.run_sweeps(args.iters)
             return None, eng.get_stats()
-        # KMP_DIST_MODE=replicated selects the round-1 replicated-commit
-        # path (fewer per-chunk collectives; the sharded path is default)
-        dist_fn = (refine_dist
-                   if os.environ.get("KMP_DIST_MODE") == "replicated"
-                   else refine_dist_sharded)
-        cut, part, stats = dist_fn(eng, k, mbw, part0, args.seed, args.iters, comm)
+        # KMP_DIST_MODE: cpp (default; C++ RCCL chunk loop) | sharded
+        # (python-orchestrated sharded commit) | replicated (round-1 path)
+        mode = os.environ.get("KMP_DIST_MODE", "cpp")
+        if mode == "cpp":
+            from kaminpar_amd.multi import nccl_cpp_comm, refine_dist_cpp
+
+            nccl_comm = getattr(one_step, "_nccl_comm", None)
+            if nccl_comm is None and world > 1:
+                nccl_comm = nccl_cpp_comm(rank, world)
+                one_step._nccl_comm = nccl_comm
+            cut, part, stats = refine_dist_cpp(
+                eng, k, mbw, part0, args.seed, args.iters, rank, world,
+                nccl_comm)
+        else:
+            dist_fn = (refine_dist if mode == "replicated"
+                       else refine_dist_sharded)
+            cut, part, stats = dist_fn(eng, k, mbw, part0, args.seed,
+                                       args.iters, comm)
         return cut, stats
 
     def barrier_sync():

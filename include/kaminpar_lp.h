@@ -271,6 +271,29 @@ int64_t kmp_lp_shard_apply(kmp_lp_t *e, int iter, uint32_t chunk,
                            const long long *d_arr_all,
                            const long long *d_dep_global);
 
+/* C++ RCCL distributed driver: the whole sharded-commit chunk loop with
+ * collectives issued directly on the engine stream (the python/torch
+ * orchestration has a measured ~0.65 ms/chunk host floor; this driver's
+ * per-chunk host work is the fixpoint convergence readback only).
+ * Bootstrap: rank 0 calls kmp_nccl_unique_id, the 128-byte id is
+ * exchanged out of band (e.g. a torch.distributed broadcast), every rank
+ * calls kmp_nccl_comm_init. nccl_comm may be NULL at world 1. */
+int kmp_nccl_unique_id(void *out128);
+void *kmp_nccl_comm_init(int world, int rank, const void *id128);
+void kmp_nccl_comm_destroy(void *comm);
+int64_t kmp_lp_refine_dist(
+    kmp_lp_t *e,
+    uint32_t k,
+    const int64_t *max_block_weights,
+    uint32_t *partition,
+    uint64_t seed,
+    int iters,
+    void *nccl_comm,
+    int rank,
+    int world,
+    kmp_lp_stats_t *stats
+);
+
 int kmp_lp_reset(kmp_lp_t *e);                     /* restore initial state (D2D) */
 int64_t kmp_lp_run_sweeps(kmp_lp_t *e, int iters); /* the timed LP region */
 int kmp_lp_get_stats(kmp_lp_t *e, kmp_lp_stats_t *stats); /* no cut/download */

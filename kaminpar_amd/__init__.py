@@ -141,6 +141,16 @@ def _load():
     lib.kmp_lp_phase_a.argtypes = [vp, ctypes.c_int, u32, u32, u32, vp, u32]
     lib.kmp_lp_commit.restype = i64
     lib.kmp_lp_commit.argtypes = [vp, ctypes.c_int, u32, vp, u32]
+    lib.kmp_nccl_unique_id.restype = ctypes.c_int
+    lib.kmp_nccl_unique_id.argtypes = [vp]
+    lib.kmp_nccl_comm_init.restype = vp
+    lib.kmp_nccl_comm_init.argtypes = [ctypes.c_int, ctypes.c_int, vp]
+    lib.kmp_nccl_comm_destroy.restype = None
+    lib.kmp_nccl_comm_destroy.argtypes = [vp]
+    lib.kmp_lp_refine_dist.restype = i64
+    lib.kmp_lp_refine_dist.argtypes = [vp, u32, p(i64), p(u32), u64,
+                                       ctypes.c_int, vp, ctypes.c_int,
+                                       ctypes.c_int, vp]
     lib.kmp_lp_set_stream.restype = ctypes.c_int
     lib.kmp_lp_set_stream.argtypes = [vp, vp]
     lib.kmp_lp_shard_begin.restype = ctypes.c_int
@@ -506,6 +516,19 @@ class LpEngine:
         if mv < 0:
             raise RuntimeError("kmp_lp_commit failed")
         return int(mv)
+
+    def refine_dist_cpp(self, k, max_block_weights, partition, seed, iters,
+                        nccl_comm, rank, world):
+        """C++ RCCL-driven sharded refinement (see kmp_lp_refine_dist)."""
+        part = np.ascontiguousarray(partition, dtype=np.uint32).copy()
+        mbw = np.ascontiguousarray(max_block_weights, dtype=np.int64)
+        stats = Stats()
+        cut = _lib.kmp_lp_refine_dist(
+            self._h, k, _i64p(mbw), _u32p(part), seed, iters,
+            nccl_comm, rank, world, ctypes.byref(stats))
+        if cut < 0:
+            raise RuntimeError("kmp_lp_refine_dist failed")
+        return cut, part, stats
 
     def set_stream(self, stream_ptr):
         """Adopt an external HIP stream (torch: cuda.current_stream().cuda_stream);

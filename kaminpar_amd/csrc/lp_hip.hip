@@ -30,6 +30,8 @@
 
 #include <rocprim/rocprim.hpp>
 
+#include <rccl/rccl.h>
+
 #include "../../include/kaminpar_lp.h"
 #include "lp_common.h"
 
@@ -2628,7 +2630,7 @@ __global__ void k_db_gather(
 
 // Histogram over the dense proposal list, filtered to the rank's targets.
 __global__ void k_hist_props(
-    u32 count,
+    const u32 *__restrict__ count_dev,
     u32 k,
     u32 rows,
     u32 tpw,
@@ -2638,6 +2640,7 @@ __global__ void k_hist_props(
     u32 *__restrict__ histT
 ) {
   extern __shared__ u32 cnt_lds[];
+  const u32 count = *count_dev;
   const u32 lane = threadIdx.x & (kWave - 1);
   const u32 wv = threadIdx.x >> 6;
   u32 *cnt = cnt_lds + wv * k;
@@ -2665,7 +2668,7 @@ __global__ void k_hist_props(
 // dep contributions (by SOURCE block, over the rank's own-target proposals)
 // go to the caller's allreduce buffer; arrival totals stay local.
 __global__ void k_scatter_props(
-    u32 count,
+    const u32 *__restrict__ count_dev,
     u32 k,
     u32 rows,
     u32 tpw,
@@ -2699,6 +2702,7 @@ __global__ void k_scatter_props(
     h_in[c] = 0;
   }
   __syncthreads();
+  const u32 count = *count_dev;
   const u32 T = (count + 63) >> 6;
   const u32 t1 = (row + 1) * tpw < T ? (row + 1) * tpw : T;
   for (u32 t = row * tpw; t < t1; ++t) {
@@ -2750,6 +2754,96 @@ __global__ void k_scatter_props(
     }
     if (h_in[c]) {
       atomicAdd(&arr_loc[c], h_in[c]);
+    }
+  }
+}
+
+// Copy the device proposal count into the send buffer's trailing row so a
+// single fixed-size allgather moves payload + counts with no host sync.
+__global__ void k_set_u32(u32 *__restrict__ dst, u32 v) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    *dst = v;
+  }
+}
+
+__global__ void k_pack_count(const u32 *__restrict__ count_dev, Prop *__restrict__ trailing) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    trailing->u = *count_dev;
+  }
+}
+
+// Dense-pack the gathered per-rank segments (cap+1 rows each, trailing row
+// carries the rank's count) into e->d_props, in rank order (= global rank
+// order since rank position slices ascend); writes the total count.
+__global__ void k_compact_gathered(
+    u32 world,
+    u32 cap,
+    const Prop *__restrict__ recv, // world * (cap+1) rows
+    Prop *__restrict__ dense,
+    u32 *__restrict__ total_out
+) {
+  __shared__ u32 cnt_s[8], off_s[9];
+  const u32 tid = blockIdx.x * blockDim.x + threadIdx.x;
+  if (threadIdx.x < world) {
+    cnt_s[threadIdx.x] = recv[(static_cast<size_t>(threadIdx.x) * (cap + 1) + cap)].u;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    u32 acc = 0;
+    for (u32 r = 0; r < world; ++r) {
+      off_s[r] = acc;
+      acc += cnt_s[r];
+    }
+    off_s[world] = acc;
+    if (blockIdx.x == 0) {
+      *total_out = acc;
+    }
+  }
+  __syncthreads();
+  const u32 stride = gridDim.x * blockDim.x;
+  for (u32 r = 0; r < world; ++r) {
+    const u32 n_r = cnt_s[r];
+    const Prop *src_r = recv + static_cast<size_t>(r) * (cap + 1);
+    Prop *dst_r = dense + off_s[r];
+    for (u32 i = tid; i < n_r; i += stride) {
+      dst_r[i] = src_r[i];
+    }
+  }
+}
+
+__global__ void k_vec_sub_i64(u32 n, long long *__restrict__ a, const long long *__restrict__ b) {
+  const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    a[i] -= b[i];
+  }
+}
+
+// Activation over the dense proposal list with a device count (wave per
+// admitted entry; the C++ dist driver's analogue of k_activate).
+__global__ void k_activate_props_dev(
+    const u32 *__restrict__ count_dev,
+    const u32 *__restrict__ admitted_flags,
+    const Prop *__restrict__ props,
+    const u64 *__restrict__ xadj,
+    const u32 *__restrict__ adjncy,
+    uint8_t *__restrict__ active,
+    uint8_t *__restrict__ unit_active
+) {
+  const u32 count = *count_dev;
+  const u32 wave_id = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const u32 waves = (gridDim.x * blockDim.x) >> 6;
+  const u32 lane = threadIdx.x & (kWave - 1);
+  for (u32 i = wave_id; i < count; i += waves) {
+    if (!admitted_flags[i]) {
+      continue;
+    }
+    const u32 u = props[i].u;
+    const u64 row = xadj[u];
+    const u32 deg = static_cast<u32>(xadj[u + 1] - row);
+    for (u32 e2 = lane; e2 < deg; e2 += kWave) {
+      const u32 v = adjncy[row + e2];
+      active[v] = 1;
+      unit_active[v >> 6] = 1;
     }
   }
 }
@@ -2904,7 +2998,7 @@ __global__ void k_shard_finish_meta(
 // Apply the globally-agreed admission on the full proposal list + update
 // all block weights; every rank executes this identically.
 __global__ void k_shard_apply(
-    u32 count,
+    const u32 *__restrict__ count_dev,
     u32 k,
     const Prop *__restrict__ props,
     const unsigned long long *__restrict__ cutoff_all, // k (allreduced)
@@ -2917,6 +3011,7 @@ __global__ void k_shard_apply(
     u32 *__restrict__ admitted_flags,
     unsigned long long *__restrict__ moves
 ) {
+  const u32 count = *count_dev;
   const u32 i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < k) {
     const i64 delta = static_cast<i64>(arr_all[i]) - static_cast<i64>(dep_global[i]);
@@ -3746,9 +3841,8 @@ bool v2_eligible(const kmp_lp_t *e) {
 
 // Phase A without slot memsets: S/M/L freshly write every slot of an
 // active unit; inactive units are gated downstream by unit_active.
-void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi) {
+void phase_a_v2(kmp_lp_t *e, int iter, u32 pos_lo, u32 pos_hi, u32 chunk_base) {
   const u64 iseed = iter_seed_of(e->seed, iter);
-  const u32 chunk_base = pos_lo;
   const u32 span = pos_hi - pos_lo;
   const u32 threads = 256;
   HIP_CHECK(hipMemsetAsync(e->d_m_count, 0, sizeof(u32) * 3, e->stream));
@@ -3896,7 +3990,7 @@ void enqueue_sweep_v2(kmp_lp_t *e, int iter, size_t *ev_base_io) {
     }
     hipEvent_t a0 = ev_one(e), a1 = ev_one(e), c1 = ev_one(e);
     HIP_CHECK(hipEventRecord(a0, e->stream));
-    phase_a_v2(e, iter, pos_lo, pos_hi);
+    phase_a_v2(e, iter, pos_lo, pos_hi, pos_lo);
     HIP_CHECK(hipEventRecord(a1, e->stream));
     commit_v2(e, iter, pos_lo, pos_hi);
     HIP_CHECK(hipEventRecord(c1, e->stream));
@@ -4721,12 +4815,16 @@ int kmp_lp_shard_begin(
   const u32 tpw = (T + rows - 1) / rows;
   const Prop *props = static_cast<const Prop *>(d_props);
   const size_t lds_h = static_cast<size_t>(threads / kWave) * e->k * sizeof(u32);
+  // stage the (host-known) count into the device scalar the kernels read
+  // (kernel arg by value: no pinned-buffer reuse races)
+  hipLaunchKernelGGL(k_set_u32, dim3(1), dim3(1), 0, e->stream, e->d_prop_count, count);
+  LAUNCH_CHECK();
   // reset local arr slot
   HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * 2 * e->k, e->stream));
   if (count > 0) {
     hipLaunchKernelGGL(
-        k_hist_props, dim3(rows / 4), dim3(threads), lds_h, e->stream, count, e->k, rows, tpw,
-        c_lo, c_hi, props, e->d_histT
+        k_hist_props, dim3(rows / 4), dim3(threads), lds_h, e->stream, e->d_prop_count, e->k,
+        rows, tpw, c_lo, c_hi, props, e->d_histT
     );
     LAUNCH_CHECK();
   } else {
@@ -4740,9 +4838,9 @@ int kmp_lp_shard_begin(
   if (count > 0) {
     const size_t lds_sc = lds_h + static_cast<size_t>(2 * e->k) * sizeof(unsigned long long);
     hipLaunchKernelGGL(
-        k_scatter_props, dim3(rows / 4), dim3(threads), lds_sc, e->stream, count, e->k, rows,
-        tpw, c_lo, c_hi, props, e->d_offT, e->d_labels, e->d_s_u, e->d_s_w, e->d_s_r,
-        e->d_s_to, e->d_s_b, d_dep_out,
+        k_scatter_props, dim3(rows / 4), dim3(threads), lds_sc, e->stream, e->d_prop_count,
+        e->k, rows, tpw, c_lo, c_hi, props, e->d_offT, e->d_labels, e->d_s_u, e->d_s_w,
+        e->d_s_r, e->d_s_to, e->d_s_b, d_dep_out,
         reinterpret_cast<unsigned long long *>(e->d_dep) + e->k
     );
     LAUNCH_CHECK();
@@ -4805,12 +4903,15 @@ i64 kmp_lp_shard_apply(
   const u32 chunk_lo = chunk * e->C;
   const u32 chunk_hi = chunk_lo + e->C > e->P ? e->P : chunk_lo + e->C;
   const Prop *props = static_cast<const Prop *>(d_props);
+  hipLaunchKernelGGL(k_set_u32, dim3(1), dim3(1), 0, e->stream, e->d_prop_count, count);
+  LAUNCH_CHECK();
   {
     const u32 span = count > e->k ? count : e->k;
     hipLaunchKernelGGL(
-        k_shard_apply, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, count, e->k,
-        props, d_cutoff_all, d_arr_all, d_dep_global, e->d_weights, e->d_labels, e->d_labels16,
-        e->k <= 256 ? e->d_labels8 : nullptr, e->d_admitted_flags, e->d_moves
+        k_shard_apply, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream,
+        e->d_prop_count, e->k, props, d_cutoff_all, d_arr_all, d_dep_global, e->d_weights,
+        e->d_labels, e->d_labels16, e->k <= 256 ? e->d_labels8 : nullptr, e->d_admitted_flags,
+        e->d_moves
     );
     LAUNCH_CHECK();
   }
@@ -5806,6 +5907,237 @@ int kmp_lp_rearrange_degree_buckets(kmp_lp_t *e, u32 *perm_out) {
   HIP_CHECK(hipFree(d_hist));
   HIP_CHECK(hipFree(d_off));
   return 0;
+}
+
+// ==================== C++ RCCL distributed driver ====================
+// The per-chunk loop of the sharded multi-GPU refinement, driven from C++
+// with RCCL called directly on the engine stream (the python/torch
+// orchestration of kaminpar_amd.multi measured a ~0.65 ms/chunk host
+// floor; this driver's floor is the per-round convergence readback only).
+// Protocol identical to refine_dist_sharded / the 2-rank-sim-verified
+// kernels (kaminpar-dist/refinement/lp/lp_refiner.cc:296-333).
+
+#define NCCL_CHECK(cmd)                                                        \
+  do {                                                                         \
+    ncclResult_t r_ = (cmd);                                                   \
+    if (r_ != ncclSuccess) {                                                   \
+      fprintf(stderr, "kaminpar_amd: RCCL error %s at %s:%d\n",               \
+              ncclGetErrorString(r_), __FILE__, __LINE__);                     \
+      return -1;                                                               \
+    }                                                                          \
+  } while (0)
+
+int kmp_nccl_unique_id(void *out128) {
+  ncclUniqueId id;
+  if (ncclGetUniqueId(&id) != ncclSuccess) {
+    return -1;
+  }
+  memcpy(out128, &id, sizeof(id));
+  return 0;
+}
+
+void *kmp_nccl_comm_init(int world, int rank, const void *id128) {
+  ncclUniqueId id;
+  memcpy(&id, id128, sizeof(id));
+  ncclComm_t comm = nullptr;
+  if (ncclCommInitRank(&comm, world, id, rank) != ncclSuccess) {
+    return nullptr;
+  }
+  return comm;
+}
+
+void kmp_nccl_comm_destroy(void *comm) {
+  if (comm) {
+    (void)ncclCommDestroy(static_cast<ncclComm_t>(comm));
+  }
+}
+
+i64 kmp_lp_refine_dist(
+    kmp_lp_t *e,
+    u32 k,
+    const i64 *max_block_weights,
+    u32 *partition,
+    u64 seed,
+    int iters,
+    void *nccl_comm, // null at world 1
+    int rank,
+    int world,
+    kmp_lp_stats_t *stats
+) {
+  if (kmp_lp_refine_begin(e, k, max_block_weights, partition, seed) != 0) {
+    return -1;
+  }
+  if (e->coop_nblk < 8 || k > 256) {
+    fprintf(stderr, "kaminpar_amd: refine_dist requires the v2 path (k <= 256)\n");
+    return -1;
+  }
+  ncclComm_t comm = static_cast<ncclComm_t>(nccl_comm);
+  const u32 threads = 256;
+  const u32 cap = e->C / world + 2;      // per-rank proposal capacity
+  const u32 seg = cap + 1;               // + trailing count row
+  const u32 c_lo = (static_cast<u64>(rank) * k) / world;
+  const u32 c_hi = (static_cast<u64>(rank) + 1) * k / world;
+
+  Prop *d_send = nullptr, *d_recv = nullptr;
+  long long *d_coll = nullptr; // dep[k+1] | delta[k+1] | meta[2k]
+  HIP_CHECK(hipMalloc(&d_send, sizeof(Prop) * seg));
+  HIP_CHECK(hipMalloc(&d_recv, sizeof(Prop) * seg * world));
+  HIP_CHECK(hipMalloc(&d_coll, sizeof(long long) * (2 * (k + 1) + 2 * k)));
+  long long *d_dep_g = d_coll;
+  long long *d_delta = d_coll + (k + 1);
+  long long *d_meta = d_coll + 2 * (k + 1); // [0..k) cutoffs, [k..2k) arr
+  const u32 rows = 1024;
+  const u32 cap_total = seg * world; // bound for tpw sizing
+  const u32 tpw = (((cap_total + 63) >> 6) + rows - 1) / rows;
+  const size_t lds_h = static_cast<size_t>(threads / kWave) * k * sizeof(u32);
+  const size_t lds_sc = lds_h + static_cast<size_t>(2 * k) * sizeof(unsigned long long);
+
+  unsigned long long moves_prev = 0;
+  for (int iter = 0; iter < iters; ++iter) {
+    const u64 iseed = iter_seed_of(e->seed, iter);
+    for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
+      const u32 lo = chunk * e->C;
+      const u32 hi = lo + e->C > e->P ? e->P : lo + e->C;
+      if (lo >= hi) {
+        continue;
+      }
+      const u32 span = hi - lo;
+      const u32 slo = lo + (static_cast<u64>(rank) * span) / world;
+      const u32 shi = lo + ((static_cast<u64>(rank) + 1) * span) / world;
+
+      // phase A on the rank's slice; compact (stable, position order =
+      // rank order) into the send buffer; count rides the trailing row
+      phase_a_v2(e, iter, slo, shi, lo);
+      {
+        size_t tb = e->select_temp_bytes;
+        HIP_CHECK(rocprim::select(
+            e->d_select_temp, tb, e->d_slots, d_send, e->d_prop_count, shi - slo, PropValid(),
+            e->stream
+        ));
+      }
+      hipLaunchKernelGGL(k_pack_count, dim3(1), dim3(1), 0, e->stream, e->d_prop_count,
+                         d_send + cap);
+      LAUNCH_CHECK();
+
+      if (world > 1) {
+        NCCL_CHECK(ncclAllGather(d_send, d_recv, static_cast<size_t>(seg) * 4, ncclUint32,
+                                 comm, e->stream));
+      } else {
+        HIP_CHECK(hipMemcpyAsync(d_recv, d_send, sizeof(Prop) * seg, hipMemcpyDeviceToDevice,
+                                 e->stream));
+      }
+      hipLaunchKernelGGL(
+          k_compact_gathered, dim3(256), dim3(threads), 0, e->stream, world, cap, d_recv,
+          e->d_props, e->d_prop_count
+      );
+      LAUNCH_CHECK();
+
+      // sort own targets + full-admission dep/arr
+      HIP_CHECK(hipMemsetAsync(e->d_dep, 0, sizeof(unsigned long long) * 2 * k, e->stream));
+      HIP_CHECK(hipMemsetAsync(d_coll, 0, sizeof(long long) * (k + 1), e->stream));
+      hipLaunchKernelGGL(
+          k_hist_props, dim3(rows / 4), dim3(threads), lds_h, e->stream, e->d_prop_count, k,
+          rows, tpw, c_lo, c_hi, e->d_props, e->d_histT
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_scan_small, dim3(1), dim3(threads), 0, e->stream, k, rows, e->d_histT, e->d_offT,
+          e->d_seg_off, e->d_prefix_len, e->d_dep, e->d_changed
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_scatter_props, dim3(rows / 4), dim3(threads), lds_sc, e->stream, e->d_prop_count,
+          k, rows, tpw, c_lo, c_hi, e->d_props, e->d_offT, e->d_labels, e->d_s_u, e->d_s_w,
+          e->d_s_r, e->d_s_to, e->d_s_b, d_dep_g,
+          reinterpret_cast<unsigned long long *>(e->d_dep) + k
+      );
+      LAUNCH_CHECK();
+      if (e->has_vwgt) {
+        hipLaunchKernelGGL(
+            k_shard_pw, dim3(64), dim3(threads), 0, e->stream, k, c_lo, c_hi, e->d_seg_off,
+            e->d_prefix_len, e->d_s_w, e->d_pw
+        );
+        LAUNCH_CHECK();
+      }
+      if (world > 1) {
+        NCCL_CHECK(ncclAllReduce(d_dep_g, d_dep_g, k + 1, ncclInt64, ncclSum, comm,
+                                 e->stream));
+      }
+
+      // fixpoint rounds: local cutoffs + allreduced de-admission deltas
+      for (;;) {
+        hipLaunchKernelGGL(
+            k_shard_round, dim3(1), dim3(256), 0, e->stream, k, c_lo, c_hi,
+            static_cast<u32>(e->has_vwgt ? 1 : 0), e->d_seg_off, e->d_prefix_len,
+            reinterpret_cast<unsigned long long *>(e->d_dep) + k, d_dep_g, e->d_s_w, e->d_s_b,
+            e->d_pw, e->d_weights, e->d_maxw, d_delta
+        );
+        LAUNCH_CHECK();
+        if (world > 1) {
+          NCCL_CHECK(ncclAllReduce(d_delta, d_delta, k + 1, ncclInt64, ncclSum, comm,
+                                   e->stream));
+        }
+        HIP_CHECK(hipMemcpyAsync(e->h_moves, d_delta + k, sizeof(long long),
+                                 hipMemcpyDeviceToHost, e->stream));
+        sync_spin(e);
+        if (*reinterpret_cast<long long *>(e->h_moves) == 0) {
+          break;
+        }
+        hipLaunchKernelGGL(
+            k_vec_sub_i64, dim3(ceil_div(k, threads)), dim3(threads), 0, e->stream, k, d_dep_g,
+            d_delta
+        );
+        LAUNCH_CHECK();
+      }
+
+      // exchange cutoffs + admitted arrivals (one summed 2k vector), apply
+      HIP_CHECK(hipMemsetAsync(d_meta, 0, sizeof(long long) * 2 * k, e->stream));
+      hipLaunchKernelGGL(
+          k_shard_finish_meta, dim3(ceil_div(k, 256u)), dim3(256), 0, e->stream, k, c_lo, c_hi,
+          e->d_seg_off, e->d_prefix_len,
+          reinterpret_cast<unsigned long long *>(e->d_dep) + k, e->d_s_r,
+          reinterpret_cast<unsigned long long *>(d_meta), d_meta + k
+      );
+      LAUNCH_CHECK();
+      if (world > 1) {
+        NCCL_CHECK(ncclAllReduce(d_meta, d_meta, 2 * k, ncclInt64, ncclSum, comm, e->stream));
+      }
+      {
+        const u32 span2 = cap_total > k ? cap_total : k;
+        hipLaunchKernelGGL(
+            k_shard_apply, dim3(ceil_div(span2, threads)), dim3(threads), 0, e->stream,
+            e->d_prop_count, k, e->d_props,
+            reinterpret_cast<const unsigned long long *>(d_meta), d_meta + k, d_dep_g,
+            e->d_weights, e->d_labels, e->d_labels16, k <= 256 ? e->d_labels8 : nullptr,
+            e->d_admitted_flags, e->d_moves
+        );
+        LAUNCH_CHECK();
+      }
+      hipLaunchKernelGGL(
+          k_clear_active, dim3(ceil_div(span, threads)), dim3(threads), 0, e->stream, lo, hi,
+          e->n, iseed, 0xFFFFFFFFu, e->d_xadj, e->d_active, e->d_unit_active, e->d_arcs
+      );
+      LAUNCH_CHECK();
+      hipLaunchKernelGGL(
+          k_activate_props_dev, dim3(2048), dim3(threads), 0, e->stream, e->d_prop_count,
+          e->d_admitted_flags, e->d_props, e->d_xadj, e->d_adjncy, e->d_active,
+          e->d_unit_active
+      );
+      LAUNCH_CHECK();
+    }
+    // early exit: moves counter is bit-identical on every rank
+    HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
+                             hipMemcpyDeviceToHost, e->stream));
+    sync_spin(e);
+    if (e->h_moves[1] == moves_prev) {
+      break;
+    }
+    moves_prev = e->h_moves[1];
+  }
+  HIP_CHECK(hipFree(d_send));
+  HIP_CHECK(hipFree(d_recv));
+  HIP_CHECK(hipFree(d_coll));
+  return kmp_lp_refine_end(e, partition, stats);
 }
 
 u32 kmp_lp_n(const kmp_lp_t *e) { return e->n; }

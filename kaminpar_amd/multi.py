@@ -149,6 +149,42 @@ def refine_dist_sharded(eng, k, max_block_weights, partition, seed, iters,
     return eng.refine_end()
 
 
+def nccl_cpp_comm(rank, world):
+    """Bootstrap an RCCL communicator for the C++ dist driver: rank 0
+    generates the 128-byte unique id, torch.distributed broadcasts it,
+    every rank inits. Returns an opaque handle (destroy with
+    kaminpar_amd._lib.kmp_nccl_comm_destroy) -- None at world 1."""
+    if world == 1:
+        return None
+    import ctypes
+
+    import torch
+    import torch.distributed as dist
+
+    import kaminpar_amd as ka
+
+    buf = (ctypes.c_char * 128)()
+    if rank == 0:
+        if ka._lib.kmp_nccl_unique_id(buf) != 0:
+            raise RuntimeError("kmp_nccl_unique_id failed")
+    t = torch.frombuffer(bytearray(buf.raw), dtype=torch.uint8).clone()
+    dist.broadcast(t, src=0)
+    buf = (ctypes.c_char * 128).from_buffer_copy(t.numpy().tobytes())
+    comm = ka._lib.kmp_nccl_comm_init(world, rank, buf)
+    if not comm:
+        raise RuntimeError("kmp_nccl_comm_init failed")
+    return comm
+
+
+def refine_dist_cpp(eng, k, max_block_weights, partition, seed, iters,
+                    rank, world, nccl_comm):
+    """C++ RCCL-driven sharded refinement (kmp_lp_refine_dist): the whole
+    chunk loop and all collectives run below the C-ABI; python only
+    bootstraps the communicator."""
+    return eng.refine_dist_cpp(k, max_block_weights, partition, seed, iters,
+                               nccl_comm, rank, world)
+
+
 class TorchComm:
     """torch.distributed-backed collective layer (nccl=RCCL on GPU)."""
 

@@ -1186,3 +1186,24 @@ def test_reset_and_replay_deterministic(oracle):
     assert seen[0] == (sref.arcs_scanned, sref.moves)
     assert cut2 == cut_ref
     assert np.array_equal(part2, part_ref)
+
+
+@pytest.mark.gpu
+def test_refine_dist_cpp_world1_parity(oracle):
+    """The C++ RCCL distributed driver (kmp_lp_refine_dist) at world 1:
+    bit-identical to the monolithic refine (same kernels as the
+    2-rank-sim-verified sharded protocol; collectives skipped at world 1)."""
+    _require_gpu()
+    g = ka.Graph.rmat(13, 8, seed=4)
+    k = 16
+    part0 = ka.random_partition(g.n, k, seed=6)
+    mbw = np.full(k, g.max_block_weight(k, 0.03), dtype=np.int64)
+
+    eng = ka.LpEngine(g)
+    cut_ref, part_ref, sref = eng.refine(k, mbw, part0, seed=9, iters=5)
+
+    cut, part, stats = eng.refine_dist_cpp(k, mbw, part0, seed=9, iters=5,
+                                           nccl_comm=None, rank=0, world=1)
+    assert cut == cut_ref
+    assert np.array_equal(part, part_ref)
+    assert stats.moves == sref.moves
