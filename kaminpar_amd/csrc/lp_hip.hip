@@ -6006,7 +6006,11 @@ i64 kmp_lp_refine_dist(
       const u32 shi = lo + ((static_cast<u64>(rank) + 1) * span) / world;
 
       // phase A on the rank's slice; compact (stable, position order =
-      // rank order) into the send buffer; count rides the trailing row
+      // rank order) into the send buffer; count rides the trailing row.
+      // Slots MUST be pre-invalidated here: the memset-free phase A leaves
+      // stale proposals in inactive units' slots, which the v2 commit
+      // gates by unit_active but a flat compaction cannot.
+      HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * (shi - slo), e->stream));
       phase_a_v2(e, iter, slo, shi, lo);
       {
         size_t tb = e->select_temp_bytes;
