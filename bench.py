@@ -86,7 +86,13 @@ def main():
         # force torch's HIP runtime to initialize before the engine's
         # (torch bundles its own libamdhip64; engine-first breaks torch)
         torch.zeros(1, device=f"cuda:{local_rank}")
-        dist.init_process_group("nccl")
+        # with the C++ RCCL driver (default) the data-path collectives run
+        # through the engine's own librccl; torch only coordinates
+        # (barriers, the ncclUniqueId broadcast, the elapsed max) -- gloo
+        # avoids two RCCL instances contending in one process
+        backend = ("gloo" if os.environ.get("KMP_DIST_MODE", "cpp") == "cpp"
+                   else "nccl")
+        dist.init_process_group(backend)
         device = f"cuda:{local_rank}"
     else:
         device = "cuda:0"
@@ -242,7 +248,8 @@ def main():
         import torch
         import torch.distributed as dist
 
-        tmax = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dev = (device if dist.get_backend() == "nccl" else "cpu")
+        tmax = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
         elapsed = float(tmax.item())
 
