@@ -1207,3 +1207,50 @@ def test_refine_dist_cpp_world1_parity(oracle):
     assert cut == cut_ref
     assert np.array_equal(part, part_ref)
     assert stats.moves == sref.moves
+
+
+@pytest.mark.gpu
+def test_edge_cases_tiny_graphs(oracle):
+    """Edge cases the reference's end-to-end suite pins
+    (shm_endtoend_test.cc:28-140 spirit): tiny graphs, isolated-only
+    graphs, k=2, engine reuse -- all bit-identical to the oracle and
+    crash-free."""
+    _require_gpu()
+    # 2-node path graph
+    xadj = np.array([0, 1, 2], np.uint32)
+    adjncy = np.array([1, 0], np.uint32)
+    g = ka.Graph.from_csr(xadj, adjncy)
+    eng = ka.LpEngine(g)
+    mbw = np.full(2, 2, np.int64)
+    cut, part, _ = eng.refine(2, mbw, np.array([0, 1], np.uint32), seed=1,
+                              iters=3)
+    ocut, opart, _ = oracle_refine(oracle, g, 2, mbw,
+                                   np.array([0, 1], np.uint32), seed=1,
+                                   iters=3)
+    assert cut == ocut and np.array_equal(part, opart)
+
+    # isolated-only graph (no edges at all): clustering pairs isolated
+    # nodes deterministically; refinement is a no-op
+    n = 70
+    g2 = ka.Graph.from_csr(np.zeros(n + 1, np.uint32),
+                           np.zeros(0, np.uint32))
+    eng2 = ka.LpEngine(g2)
+    from helpers import oracle_cluster
+    nc, clus, _ = eng2.cluster(4, seed=2, iters=3)
+    onc, oclus, _ = oracle_cluster(oracle, g2, 4, seed=2, iters=3)
+    assert nc == onc and np.array_equal(clus, oclus)
+    part0 = (np.arange(n) % 2).astype(np.uint32)
+    cut2, part2, _ = eng2.refine(2, np.full(2, n, np.int64), part0, seed=3,
+                                 iters=2)
+    assert cut2 == 0 and np.array_equal(part2, part0)
+
+    # engine reuse across modes: refine then cluster then refine
+    g3 = ka.Graph.rmat(10, 8, seed=2)
+    eng3 = ka.LpEngine(g3)
+    k = 8
+    p0 = ka.random_partition(g3.n, k, seed=1)
+    w = np.full(k, g3.max_block_weight(k, 0.1), np.int64)
+    c1, r1, _ = eng3.refine(k, w, p0, seed=4)
+    eng3.cluster(16, seed=5)
+    c2, r2, _ = eng3.refine(k, w, p0, seed=4)
+    assert c1 == c2 and np.array_equal(r1, r2)
