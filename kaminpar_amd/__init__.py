@@ -78,6 +78,8 @@ def _load():
     lib.kmp_graph_m.argtypes = [vp]
     lib.kmp_graph_xadj.restype = p(u32)
     lib.kmp_graph_xadj.argtypes = [vp]
+    lib.kmp_graph_xadj64.restype = p(u64)
+    lib.kmp_graph_xadj64.argtypes = [vp]
     lib.kmp_graph_adjncy.restype = p(u32)
     lib.kmp_graph_adjncy.argtypes = [vp]
     lib.kmp_graph_vwgt.restype = p(i32)
@@ -269,10 +271,17 @@ class Graph:
 
     @property
     def xadj(self):
-        return np.ctypeslib.as_array(_lib.kmp_graph_xadj(self._h), shape=(self.n + 1,))
+        p = _lib.kmp_graph_xadj(self._h)
+        if not p:
+            # 64-bit-offset graph (m >= 2^32): expose the wide offsets
+            p64 = _lib.kmp_graph_xadj64(self._h)
+            return np.ctypeslib.as_array(p64, shape=(self.n + 1,))
+        return np.ctypeslib.as_array(p, shape=(self.n + 1,))
 
     @property
     def adjncy(self):
+        if self.m == 0:
+            return np.zeros(0, dtype=np.uint32)  # null data ptr on empty CSR
         return np.ctypeslib.as_array(_lib.kmp_graph_adjncy(self._h), shape=(self.m,))
 
     @property
