@@ -5995,6 +5995,7 @@ i64 kmp_lp_refine_dist(
   unsigned long long moves_prev = 0;
   for (int iter = 0; iter < iters; ++iter) {
     const u64 iseed = iter_seed_of(e->seed, iter);
+    e->ev_used = 0; // per-sweep phase-A event pairs (bench roofline input)
     for (u32 chunk = 0; chunk < kmp::kNumChunks; ++chunk) {
       const u32 lo = chunk * e->C;
       const u32 hi = lo + e->C > e->P ? e->P : lo + e->C;
@@ -6011,7 +6012,15 @@ i64 kmp_lp_refine_dist(
       // stale proposals in inactive units' slots, which the v2 commit
       // gates by unit_active but a flat compaction cannot.
       HIP_CHECK(hipMemsetAsync(e->d_slots, 0xFF, sizeof(Prop) * (shi - slo), e->stream));
+      {
+        hipEvent_t a0 = ev_one(e);
+        HIP_CHECK(hipEventRecord(a0, e->stream));
+      }
       phase_a_v2(e, iter, slo, shi, lo);
+      {
+        hipEvent_t a1 = ev_one(e);
+        HIP_CHECK(hipEventRecord(a1, e->stream));
+      }
       {
         size_t tb = e->select_temp_bytes;
         HIP_CHECK(rocprim::select(
@@ -6133,10 +6142,25 @@ i64 kmp_lp_refine_dist(
     HIP_CHECK(hipMemcpyAsync(&e->h_moves[1], e->d_moves, sizeof(unsigned long long),
                              hipMemcpyDeviceToHost, e->stream));
     sync_spin(e);
+    for (size_t i = 0; i + 1 < e->ev_used; i += 2) {
+      float ms = 0;
+      HIP_CHECK(hipEventElapsedTime(&ms, e->ev_pool[i], e->ev_pool[i + 1]));
+      e->phase_a_ms += ms;
+    }
     if (e->h_moves[1] == moves_prev) {
       break;
     }
     moves_prev = e->h_moves[1];
+  }
+  // arcs are tallied per position slice: sum over ranks so stats report
+  // the whole job (every other counter is already rank-identical)
+  if (world > 1) {
+    NCCL_CHECK(ncclAllReduce(
+        reinterpret_cast<unsigned long long *>(e->d_arcs),
+        reinterpret_cast<unsigned long long *>(e->d_arcs), 1, ncclUint64, ncclSum, comm,
+        e->stream
+    ));
+    HIP_CHECK(hipStreamSynchronize(e->stream));
   }
   HIP_CHECK(hipFree(d_send));
   HIP_CHECK(hipFree(d_recv));
