@@ -133,3 +133,60 @@ def test_perm_cross_impl(oracle):
             valid = a[a < n]
             assert len(valid) == n
             assert len(np.unique(valid)) == n and valid.max() == n - 1
+
+
+@pytest.mark.parametrize("trial", range(6))
+def test_underload_invariants(oracle, trial):
+    """The oracle's underload balancer (underload_balancer.cc semantics):
+    fills underloaded blocks without dropping any feasible source below its
+    minimum or creating NEW maximum violations, over random drained
+    partitions (CPU-only twin of the GPU parity suite)."""
+    from helpers import oracle_underload
+
+    rng = np.random.default_rng(900 + trial)
+    g = ka.Graph.rmat(11, 8, seed=trial + 1)
+    k = int(rng.integers(2, 24))
+    total = g.n
+    mbw = np.full(k, int(total / k * 1.5) + 2, np.int64)
+    mnw = (rng.uniform(0.3, 0.8, k) * total / k).astype(np.int64)
+    part0 = ka.random_partition(g.n, k, seed=trial)
+    drain = int(rng.integers(0, k))
+    sel = part0 == drain
+    part0[sel] = (drain + 1 + (np.arange(g.n)[sel] % max(1, k - 1))).astype(
+        np.uint32) % k
+
+    cut, part, _ = oracle_underload(oracle, g, k, mbw, mnw, part0, seed=2,
+                                    iters=4)
+    assert cut == g.edge_cut(part)
+    bw0 = np.bincount(part0, minlength=k).astype(np.int64)
+    bw = np.bincount(part, minlength=k).astype(np.int64)
+    assert (bw <= np.maximum(bw0, mbw)).all()
+    feas0 = bw0 >= mnw
+    assert (bw[feas0] >= mnw[feas0]).all()
+    deficit0 = int(np.maximum(mnw - bw0, 0).sum())
+    deficit1 = int(np.maximum(mnw - bw, 0).sum())
+    assert deficit1 <= deficit0
+
+    # determinism
+    cut2, part2, _ = oracle_underload(oracle, g, k, mbw, mnw, part0, seed=2,
+                                      iters=4)
+    assert cut2 == cut and np.array_equal(part2, part)
+
+
+@pytest.mark.parametrize("ncomm", [2, 5])
+def test_cluster_communities_invariants(oracle, ncomm):
+    """Clusterer::set_communities (clusterer.h:35): the oracle never merges
+    across community boundaries, and clearing communities restores the
+    unrestricted result (CPU-only)."""
+    from helpers import oracle_cluster_comm
+
+    rng = np.random.default_rng(ncomm)
+    g = ka.Graph.rmat(11, 8, seed=9)
+    comm = rng.integers(0, ncomm, g.n).astype(np.uint32)
+    nc, clus, _ = oracle_cluster_comm(oracle, g, 32, comm, seed=3, iters=5)
+    assert (comm[clus] == comm).all()
+    sizes = np.bincount(clus, minlength=g.n)
+    assert sizes.max() <= 32
+    # more communities -> at least as many clusters as one unrestricted run
+    nc0, _clus0, _ = oracle_cluster(oracle, g, 32, seed=3, iters=5)
+    assert nc >= nc0
