@@ -111,7 +111,7 @@ __device__ inline bool key_better(i32 g, u64 h, u32 c, const BestState &b) {
 // merged once per vertex. R is chosen so k * R stays within the LDS budget.
 __host__ __device__ inline u32 gain_replicas(u32 k) {
   if (k <= 64) {
-    return 16;
+    return 8;
   }
   if (k <= 256) {
     return 4;
